@@ -1,0 +1,90 @@
+"""End-to-end plumbing check — BASELINE config[0]:
+
+federated column-mean on synthetic CSV shards, 2 CPU stations, driven
+through the client API exactly as a researcher would (SURVEY.md §4.1 call
+stack: client → server → station → algorithm → subtasks → results).
+"""
+import numpy as np
+import pandas as pd
+import pytest
+
+from vantage6_amd.cli.dev import DevNetwork
+from vantage6_amd.common.config import DatabaseConfig
+
+
+@pytest.fixture
+def csv_shards(tmp_path):
+    rng = np.random.default_rng(0)
+    paths, frames = [], []
+    for i in range(2):
+        df = pd.DataFrame(
+            {"age": rng.integers(20, 90, 50 + 10 * i), "bmi": rng.normal(25, 4, 50 + 10 * i)}
+        )
+        p = tmp_path / f"shard{i}.csv"
+        df.to_csv(p, index=False)
+        paths.append(p)
+        frames.append(df)
+    return paths, frames
+
+
+def expected_means(frames):
+    all_df = pd.concat(frames)
+    return {c: float(all_df[c].mean()) for c in ("age", "bmi")}
+
+
+def run_column_mean(net, frames):
+    with net:
+        task = net.client.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],  # central runs on station 0
+            name="column-mean",
+            image="algorithms.column_mean",
+            input_={"method": "central_mean", "kwargs": {"columns": ["age", "bmi"]}},
+            databases=[{"label": "default"}],
+        )
+        results = net.client.wait_for_results(task["id"], interval=0.05, timeout=60)
+    assert len(results) == 1
+    got = results[0]
+    want = expected_means(frames)
+    for c in ("age", "bmi"):
+        assert got[c] == pytest.approx(want[c], rel=1e-9)
+
+
+def test_column_mean_unencrypted(csv_shards):
+    paths, frames = csv_shards
+    net = DevNetwork(
+        n_stations=2,
+        databases=[[DatabaseConfig(label="default", uri=str(p), type="csv")] for p in paths],
+    )
+    run_column_mean(net, frames)
+
+
+def test_column_mean_encrypted(csv_shards):
+    """Same flow with end-to-end hybrid RSA encryption enabled."""
+    paths, frames = csv_shards
+    net = DevNetwork(
+        n_stations=2,
+        encrypted=True,
+        key_bits=1024,  # fast keygen for tests; production default is 2048
+        databases=[[DatabaseConfig(label="default", uri=str(p), type="csv")] for p in paths],
+    )
+    run_column_mean(net, frames)
+
+
+def test_partial_direct_with_mock_client(csv_shards):
+    """Algorithm unit-testing via MockAlgorithmClient (reference test style)."""
+    paths, frames = csv_shards
+    from vantage6_amd.algorithm import MockAlgorithmClient
+
+    mock = MockAlgorithmClient(
+        datasets=[[{"label": "default", "database": str(p), "type": "csv"}] for p in paths],
+        module="algorithms.column_mean",
+    )
+    task = mock.task.create(
+        input_={"method": "central_mean", "kwargs": {"columns": ["age", "bmi"]}},
+        organizations=[1],
+    )
+    got = mock.result.from_task(task["id"])[0]
+    want = expected_means(frames)
+    for c in ("age", "bmi"):
+        assert got[c] == pytest.approx(want[c], rel=1e-9)

@@ -1,0 +1,282 @@
+"""Station daemon — one worker per data station (per GPU in production).
+
+Reference parity: ``vantage6-node/vantage6/node/__init__.py::Node``
+ — authenticate with api key, subscribe to task
+events, sync open runs on boot (crash recovery), a listening worker thread
+pops the task queue and executes runs, results are patched back with status
+and captured logs; kill events cancel queued/active work.
+
+MI355X mapping: instead of spawning a Docker container per run, the run's
+algorithm package executes in-process on the station's pinned HIP device
+(``torch.cuda.set_device``); the I/O file contract (INPUT_FILE/OUTPUT_FILE/
+TOKEN_FILE + env map, Appendix A.3) is materialized under a per-job
+temporary folder for algorithm-code portability.
+"""
+from __future__ import annotations
+
+import io
+import json
+import logging
+import queue
+import tempfile
+import threading
+import traceback
+from contextlib import redirect_stderr, redirect_stdout
+from pathlib import Path
+
+from ..common import base64s_to_bytes, bytes_to_base64s
+from ..common.config import DatabaseConfig, StationConfig
+from ..common.encryption import CryptorBase, RSACryptor
+from ..common.serialization import deserialize, serialize
+from ..common.task_status import RunStatus
+from ..server.core import Identity, ServerCore
+from ..server.model import Organization
+from ..algorithm.client import AlgorithmClient
+from ..algorithm.exceptions import AlgorithmError, InputError
+from ..algorithm.wrap import dispatch, resolve_module
+from .proxy import NodeProxy
+
+log = logging.getLogger(__name__)
+
+_STOP = object()
+
+
+class Station:
+    """In-process data-station worker."""
+
+    def __init__(
+        self,
+        core: ServerCore,
+        config: StationConfig,
+        api_key: str,
+        private_key_file: str | None = None,
+        key_bits: int = 1024,
+    ):
+        self.core = core
+        self.config = config
+        self.api_key = api_key
+        self.queue: "queue.Queue" = queue.Queue()
+        self.identity: Identity | None = None
+        self._thread: threading.Thread | None = None
+        self._pool = None  # ThreadPoolExecutor: tasks run concurrently, like
+        # the reference's one-Docker-container-per-run (a central task blocks
+        # in wait_for_results while its partial runs on the same station)
+        self._kill_flags: set[int] = set()  # run ids killed while queued/active
+        self._lock = threading.Lock()
+        self._inflight = 0
+        if private_key_file:
+            self.cryptor: CryptorBase = RSACryptor(private_key_file, bits=key_bits)
+        else:
+            self.cryptor = CryptorBase()
+        # TEMPORARY_FOLDER root, shared across runs of the same job_id
+        self._tmp_root = Path(tempfile.mkdtemp(prefix=f"v6amd_{config.name}_"))
+
+    # ------------------------------------------------------------- lifecycle
+    def start(self) -> None:
+        token = self.core.authenticate_node(self.api_key)
+        self.identity = self.core.identity_from_token(token["access_token"])
+        # upload our public key so clients/nodes can encrypt for our org
+        if isinstance(self.cryptor, RSACryptor):
+            org = self.core.session.get(Organization, self.identity.organization_id)
+            org.public_key = self.cryptor.public_key_str
+            self.core.session.commit()
+        cid, oid = self.identity.collaboration_id, self.identity.organization_id
+        self.core.events.subscribe(f"collaboration_{cid}_organization_{oid}", self._on_event)
+        self.core.events.subscribe(f"collaboration_{cid}", self._on_event)
+        # crash recovery: re-queue open runs assigned to us
+        for run in self.core.node_sync_open_runs(self.identity):
+            self.queue.put(run["task_id"])
+        from concurrent.futures import ThreadPoolExecutor
+
+        self._pool = ThreadPoolExecutor(
+            max_workers=8, thread_name_prefix=f"station-{self.config.name}-exec"
+        )
+        self._thread = threading.Thread(
+            target=self._listening_worker, name=f"station-{self.config.name}", daemon=True
+        )
+        self._thread.start()
+
+    def stop(self, timeout: float = 10.0) -> None:
+        self.queue.put(_STOP)
+        if self._thread:
+            self._thread.join(timeout)
+        if self._pool:
+            self._pool.shutdown(wait=False)
+        if self.identity:
+            self.core.node_set_offline(self.identity.id)
+
+    def join_idle(self, timeout: float = 60.0) -> bool:
+        """Testing/ops helper: wait until the queue is drained."""
+        import time
+
+        t0 = time.monotonic()
+        while time.monotonic() - t0 < timeout:
+            with self._lock:
+                busy = self._inflight
+            if self.queue.empty() and busy == 0:
+                return True
+            time.sleep(0.01)
+        return False
+
+    # ---------------------------------------------------------------- events
+    def _on_event(self, event: str, data: dict) -> None:
+        if event == "new_task":
+            if data.get("organization_id") == self.identity.organization_id:
+                self.queue.put(data["task_id"])
+        elif event == "kill_containers":
+            with self._lock:
+                for item in data.get("kill_list", []):
+                    if item.get("organization_id") == self.identity.organization_id:
+                        self._kill_flags.add(item["run_id"])
+
+    # ------------------------------------------------------------- hot loop
+    def _listening_worker(self) -> None:
+        while True:
+            item = self.queue.get()
+            if item is _STOP:
+                return
+            with self._lock:
+                self._inflight += 1
+            self._pool.submit(self._run_one, item)
+
+    def _run_one(self, task_id: int) -> None:
+        try:
+            self._execute_task(task_id)
+        except Exception:  # noqa: BLE001 — worker must survive anything
+            log.exception("station %s: task %s execution failed", self.config.name, task_id)
+        finally:
+            with self._lock:
+                self._inflight -= 1
+
+    def _my_run_for_task(self, task_id: int) -> dict | None:
+        runs = self.core.list_runs(self.identity, task_id=task_id)["data"]
+        for r in runs:
+            if r["node_id"] == self.identity.id:
+                return r
+        return None
+
+    def _execute_task(self, task_id: int) -> None:
+        run = self._my_run_for_task(task_id)
+        if run is None:
+            return
+        run_id = run["id"]
+        with self._lock:
+            if run_id in self._kill_flags:
+                self._kill_flags.discard(run_id)
+                return
+        if run["status"] != RunStatus.PENDING.value:
+            return  # already handled (e.g. killed while queued)
+        task = self.core.get_task(self.identity, task_id)
+        self.core.patch_run(self.identity, run_id, status=RunStatus.INITIALIZING.value)
+
+        # -- policy check (reference: DockerManager allowed-images policy)
+        allowed = self.config.policies.get("allowed_algorithms")
+        if allowed is not None and task["image"] not in allowed:
+            self.core.patch_run(
+                self.identity, run_id, status=RunStatus.NOT_ALLOWED.value,
+                log=f"algorithm {task['image']!r} not in station allow-list",
+            )
+            return
+        try:
+            module = resolve_module(task["image"])
+        except InputError as exc:
+            self.core.patch_run(
+                self.identity, run_id, status=RunStatus.NO_IMAGE.value, log=str(exc)
+            )
+            return
+
+        # -- decrypt input (encrypted to this org's public key)
+        try:
+            payload = deserialize(self.cryptor.decrypt_str_to_bytes(run["input"]))
+        except Exception as exc:  # noqa: BLE001
+            self.core.patch_run(
+                self.identity, run_id, status=RunStatus.START_FAILED.value,
+                log=f"input decryption/deserialization failed: {exc}",
+            )
+            return
+
+        # -- build runtime contract (Appendix A.3)
+        token = self.core.token_for_container(self.identity, task_id, task["image"])
+        container_ident = self.core.identity_from_token(token["container_token"])
+        job_dir = self._tmp_root / f"job_{task['job_id']}"
+        job_dir.mkdir(parents=True, exist_ok=True)
+        run_dir = job_dir / f"run_{run_id}"
+        run_dir.mkdir(exist_ok=True)
+        input_file = run_dir / "input"
+        output_file = run_dir / "output"
+        token_file = run_dir / "token"
+        input_file.write_bytes(serialize(payload))
+        token_file.write_text(token["container_token"])
+        env = {
+            "INPUT_FILE": str(input_file),
+            "OUTPUT_FILE": str(output_file),
+            "TOKEN_FILE": str(token_file),
+            "TEMPORARY_FOLDER": str(job_dir),
+            "TASK_ID": str(task_id),
+            "NODE_ID": str(self.identity.id),
+            "ORGANIZATION_ID": str(self.identity.organization_id),
+            "COLLABORATION_ID": str(self.identity.collaboration_id),
+            "PKG_NAME": module.__name__,
+            "HOST": "in-process", "PORT": "0", "API_PATH": "/api",
+        }
+        requested = [d.get("label", "default") for d in task.get("databases", [])]
+        if requested:
+            env["USER_REQUESTED_DATABASE_LABELS"] = ",".join(requested)
+        for db in self.config.databases:
+            env[f"{db.label.upper()}_DATABASE_URI"] = db.uri
+            env[f"{db.label.upper()}_DATABASE_TYPE"] = db.type
+            if db.label == "default":
+                env["DATABASE_URI"] = db.uri
+                env["DATABASE_TYPE"] = db.type
+
+        collab_encrypted = isinstance(self.cryptor, RSACryptor)
+        proxy = NodeProxy(self.core, container_ident, self.cryptor,
+                          collab_encrypted, task["image"])
+        client = AlgorithmClient(proxy)
+
+        # -- execute on the pinned device, capturing stdout/stderr as the log
+        self.core.patch_run(self.identity, run_id, status=RunStatus.ACTIVE.value)
+        buf = io.StringIO()
+        try:
+            with redirect_stdout(buf), redirect_stderr(buf):
+                self._bind_device()
+                result = dispatch(module, payload, env, client)
+            blob = serialize(result)
+            output_file.write_bytes(blob)
+            # encrypt result for the task's initiating organization
+            result_str = self._encrypt_result(blob, task["init_org_id"])
+            self.core.patch_run(
+                self.identity, run_id, status=RunStatus.COMPLETED.value,
+                result=result_str, log=buf.getvalue(),
+            )
+        except AlgorithmError as exc:
+            self.core.patch_run(
+                self.identity, run_id, status=RunStatus.FAILED.value,
+                log=buf.getvalue() + f"\nAlgorithmError: {exc}",
+            )
+        except Exception:  # noqa: BLE001
+            self.core.patch_run(
+                self.identity, run_id, status=RunStatus.CRASHED.value,
+                log=buf.getvalue() + "\n" + traceback.format_exc(),
+            )
+
+    def _encrypt_result(self, blob: bytes, init_org_id: int | None) -> str:
+        if isinstance(self.cryptor, RSACryptor) and init_org_id is not None:
+            pubkey = self.core.organization_public_key(init_org_id)
+            if pubkey:
+                return self.cryptor.encrypt_bytes_to_str(blob, pubkey)
+        return bytes_to_base64s(blob)
+
+    def _bind_device(self) -> None:
+        dev = self.config.device
+        if dev == "auto":
+            try:
+                import torch
+
+                dev = "cuda:0" if torch.cuda.is_available() else "cpu"
+            except ImportError:
+                dev = "cpu"
+        if dev.startswith("cuda"):
+            import torch
+
+            torch.cuda.set_device(dev)

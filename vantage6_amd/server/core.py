@@ -1,0 +1,779 @@
+"""Orchestrator service layer — the collapsed vantage6 server.
+
+Reference parity: the Flask REST resources
+(``vantage6-server/vantage6/server/resource/*.py``), JWT token minting
+(``resource/token.py``), task brokering (``resource/task.py``), run patching
+(``resource/run.py``) and the Socket.IO eventing (``websockets.py``) are all
+here as plain method calls on :class:`ServerCore`. The FastAPI façade
+(:mod:`vantage6_amd.server.api`) and the in-process client both drive this
+object, so HTTP is an optional transport instead of a mandatory boundary —
+that is the MI355X-first design decision: stations live in-process (or
+one-per-GPU via torchrun), not behind HTTPS + Docker.
+
+Events that the reference pushes over Socket.IO (``new_task``,
+``algorithm_status_change``, ``kill_containers``, ``node-info-update``) are
+delivered through :class:`EventBus` callbacks/queues.
+"""
+from __future__ import annotations
+
+import datetime as dt
+import json
+import logging
+import secrets
+import threading
+from dataclasses import dataclass, field
+from typing import Any, Callable, Optional
+
+from ..common import jwt as jwtlib
+from ..common.task_status import RunStatus, has_task_finished, task_status_from_runs
+from .model import (
+    Collaboration,
+    Database,
+    Node,
+    Organization,
+    Role,
+    Run,
+    Study,
+    Task,
+    User,
+    utcnow,
+)
+from .permission import (
+    Operation,
+    PermissionError_,
+    PermissionManager,
+    RuleCollection,
+    Scope,
+    seed_default_roles,
+    seed_rules,
+)
+
+log = logging.getLogger(__name__)
+
+
+@dataclass
+class Identity:
+    """Decoded token identity: a user, a node, or an algorithm container."""
+
+    type: str  # "user" | "node" | "container"
+    id: int  # user id or node id
+    organization_id: int | None = None
+    collaboration_id: int | None = None
+    task_id: int | None = None  # container tokens only
+    image: str | None = None  # container tokens only
+
+
+class EventBus:
+    """In-process analog of the Socket.IO rooms.
+
+    Subscribers register a callback keyed by room; ``emit`` fans out
+    synchronously (station queues do their own buffering).
+    """
+
+    def __init__(self):
+        self._subs: dict[str, list[Callable[[str, dict], None]]] = {}
+        self._lock = threading.Lock()
+
+    def subscribe(self, room: str, callback: Callable[[str, dict], None]) -> None:
+        with self._lock:
+            self._subs.setdefault(room, []).append(callback)
+
+    def unsubscribe(self, room: str, callback) -> None:
+        with self._lock:
+            if room in self._subs and callback in self._subs[room]:
+                self._subs[room].remove(callback)
+
+    def emit(self, event: str, data: dict, room: str) -> None:
+        with self._lock:
+            subs = list(self._subs.get(room, []))
+        for cb in subs:
+            try:
+                cb(event, data)
+            except Exception:  # noqa: BLE001 — a bad subscriber must not kill dispatch
+                log.exception("event subscriber failed for %s/%s", room, event)
+
+
+class NotFound(Exception):
+    status_code = 404
+
+
+class BadRequest(Exception):
+    status_code = 400
+
+
+class ServerCore:
+    """The orchestrator. One instance per deployment (tests create many)."""
+
+    def __init__(
+        self,
+        uri: str = "sqlite://",
+        jwt_secret: str | None = None,
+        token_expires_seconds: float = 21600.0,
+    ):
+        self.db = Database(uri)
+        self.jwt_secret = jwt_secret or secrets.token_hex(32)
+        self.token_expires = token_expires_seconds
+        self.events = EventBus()
+        self._lock = threading.RLock()
+        self._job_counter_lock = threading.Lock()
+        s = self.session
+        seed_rules(s)
+        seed_default_roles(s)
+
+    # ------------------------------------------------------------------ infra
+    @property
+    def session(self):
+        return self.db.session()
+
+    def close(self):
+        self.db.close()
+
+    # ------------------------------------------------------------- bootstrap
+    def create_root_user(self, username: str = "root", password: str = "root") -> dict:
+        s = self.session
+        if s.query(User).filter_by(username=username).first():
+            raise BadRequest(f"user {username!r} exists")
+        org = s.query(Organization).filter_by(name="root").first()
+        if org is None:
+            org = Organization(name="root")
+            s.add(org)
+            s.flush()
+        user = User(username=username, organization_id=org.id)
+        user.set_password(password)
+        root_role = s.query(Role).filter_by(name="Root").first()
+        user.roles = [root_role]
+        s.add(user)
+        s.commit()
+        return self._user_json(user)
+
+    # ------------------------------------------------------------------ auth
+    def authenticate_user(self, username: str, password: str) -> dict:
+        s = self.session
+        user = s.query(User).filter_by(username=username).first()
+        if user is None or not user.check_password(password):
+            if user is not None:
+                user.failed_login_attempts = (user.failed_login_attempts or 0) + 1
+                user.last_login_attempt = utcnow()
+                s.commit()
+            raise PermissionError_("invalid username or password", 401)
+        user.failed_login_attempts = 0
+        user.last_seen = utcnow()
+        s.commit()
+        payload = {"sub": user.id, "type": "user", "org": user.organization_id}
+        return {
+            "access_token": jwtlib.encode(payload, self.jwt_secret, self.token_expires),
+            "refresh_token": jwtlib.encode(
+                {**payload, "refresh": True}, self.jwt_secret, self.token_expires * 4
+            ),
+            "user_url": f"/api/user/{user.id}",
+            "refresh_url": "/api/token/refresh",
+        }
+
+    def authenticate_node(self, api_key: str) -> dict:
+        s = self.session
+        for node in s.query(Node).all():
+            if node.check_api_key(api_key):
+                node.last_seen = utcnow()
+                node.status = "online"
+                s.commit()
+                self.events.emit(
+                    "node-status-change",
+                    {"id": node.id, "name": node.name, "status": "online"},
+                    room=f"collaboration_{node.collaboration_id}",
+                )
+                payload = {
+                    "sub": node.id,
+                    "type": "node",
+                    "org": node.organization_id,
+                    "collab": node.collaboration_id,
+                }
+                return {
+                    "access_token": jwtlib.encode(payload, self.jwt_secret, self.token_expires),
+                    "refresh_token": jwtlib.encode(
+                        {**payload, "refresh": True}, self.jwt_secret, self.token_expires * 4
+                    ),
+                    "node_url": f"/api/node/{node.id}",
+                }
+        raise PermissionError_("invalid api key", 401)
+
+    def token_for_container(self, node_identity: Identity, task_id: int, image: str) -> dict:
+        """Node forges a scoped token for a running algorithm
+        (``resource/token.py::ContainerToken``)."""
+        if node_identity.type != "node":
+            raise PermissionError_("only nodes may request container tokens")
+        s = self.session
+        task = s.get(Task, task_id)
+        if task is None:
+            raise NotFound(f"task {task_id} not found")
+        if task.collaboration_id != node_identity.collaboration_id:
+            raise PermissionError_("task is not in this node's collaboration")
+        payload = {
+            "sub": node_identity.id,
+            "type": "container",
+            "org": node_identity.organization_id,
+            "collab": node_identity.collaboration_id,
+            "task_id": task_id,
+            "image": image,
+        }
+        return {"container_token": jwtlib.encode(payload, self.jwt_secret, self.token_expires)}
+
+    def refresh_token(self, refresh_token: str) -> dict:
+        payload = jwtlib.decode(refresh_token, self.jwt_secret)
+        if not payload.get("refresh"):
+            raise PermissionError_("not a refresh token", 401)
+        payload = {k: v for k, v in payload.items() if k in ("sub", "type", "org", "collab")}
+        return {
+            "access_token": jwtlib.encode(payload, self.jwt_secret, self.token_expires),
+            "refresh_token": jwtlib.encode(
+                {**payload, "refresh": True}, self.jwt_secret, self.token_expires * 4
+            ),
+        }
+
+    def identity_from_token(self, token: str) -> Identity:
+        p = jwtlib.decode(token, self.jwt_secret)
+        return Identity(
+            type=p["type"],
+            id=p["sub"],
+            organization_id=p.get("org"),
+            collaboration_id=p.get("collab"),
+            task_id=p.get("task_id"),
+            image=p.get("image"),
+        )
+
+    # ----------------------------------------------------------- permissions
+    def _user(self, ident: Identity) -> User:
+        user = self.session.get(User, ident.id)
+        if user is None:
+            raise PermissionError_("unknown user", 401)
+        return user
+
+    def _rc(self, resource: str, ident: Identity) -> RuleCollection:
+        return PermissionManager(self.session).collection(resource, self._user(ident))
+
+    def _check_user_can(
+        self, resource: str, op: Operation, ident: Identity, *, org_id=None, collab_id=None
+    ) -> None:
+        """Raise unless `ident` (a user) may perform op on a resource owned by
+        org_id / inside collab_id."""
+        rc = self._rc(resource, ident)
+        scope = rc.allowed_scope(op)
+        if scope is None:
+            raise PermissionError_(f"user lacks {op.value} permission on {resource}")
+        if scope == Scope.GLOBAL:
+            return
+        if scope == Scope.COLLABORATION:
+            if collab_id is None:
+                if org_id is not None and org_id == ident.organization_id:
+                    return
+                raise PermissionError_(f"{op.value} on {resource}: collaboration scope exceeded")
+            s = self.session
+            collab = s.get(Collaboration, collab_id)
+            if collab and any(o.id == ident.organization_id for o in collab.organizations):
+                return
+            raise PermissionError_(f"{op.value} on {resource}: not in collaboration {collab_id}")
+        # ORGANIZATION / OWN
+        if org_id is not None and org_id == ident.organization_id:
+            return
+        raise PermissionError_(f"{op.value} on {resource}: organization scope exceeded")
+
+    # ------------------------------------------------------ entity management
+    def create_organization(self, ident: Identity, name: str, **kw) -> dict:
+        self._check_user_can("organization", Operation.CREATE, ident)
+        s = self.session
+        if s.query(Organization).filter_by(name=name).first():
+            raise BadRequest(f"organization {name!r} exists")
+        org = Organization(name=name, **{k: v for k, v in kw.items() if k in (
+            "domain", "address1", "country", "public_key")})
+        s.add(org)
+        s.commit()
+        return self._org_json(org)
+
+    def set_organization_public_key(self, ident: Identity, org_id: int, public_key: str) -> dict:
+        # uploading your OWN organization's public key is part of
+        # setup_encryption and needs no org-edit permission (reference:
+        # organization PATCH at OWN scope)
+        if not (ident.type == "user" and org_id == ident.organization_id):
+            self._check_user_can("organization", Operation.EDIT, ident, org_id=org_id)
+        s = self.session
+        org = s.get(Organization, org_id)
+        if org is None:
+            raise NotFound(f"organization {org_id} not found")
+        org.public_key = public_key
+        s.commit()
+        return self._org_json(org)
+
+    def create_collaboration(
+        self, ident: Identity, name: str, organization_ids: list[int], encrypted: bool = False
+    ) -> dict:
+        self._check_user_can("collaboration", Operation.CREATE, ident)
+        s = self.session
+        if s.query(Collaboration).filter_by(name=name).first():
+            raise BadRequest(f"collaboration {name!r} exists")
+        collab = Collaboration(name=name, encrypted=encrypted)
+        collab.organizations = [s.get(Organization, i) for i in organization_ids]
+        if any(o is None for o in collab.organizations):
+            raise NotFound("unknown organization in collaboration")
+        s.add(collab)
+        s.commit()
+        return self._collab_json(collab)
+
+    def create_study(self, ident: Identity, name: str, collaboration_id: int,
+                     organization_ids: list[int]) -> dict:
+        self._check_user_can("study", Operation.CREATE, ident, collab_id=collaboration_id)
+        s = self.session
+        collab = s.get(Collaboration, collaboration_id)
+        if collab is None:
+            raise NotFound(f"collaboration {collaboration_id} not found")
+        member_ids = {o.id for o in collab.organizations}
+        if not set(organization_ids) <= member_ids:
+            raise BadRequest("study organizations must be collaboration members")
+        study = Study(name=name, collaboration_id=collaboration_id)
+        study.organizations = [s.get(Organization, i) for i in organization_ids]
+        s.add(study)
+        s.commit()
+        return {"id": study.id, "name": study.name,
+                "collaboration_id": collaboration_id,
+                "organizations": [{"id": o.id} for o in study.organizations]}
+
+    def create_node(
+        self,
+        ident: Identity,
+        name: str,
+        collaboration_id: int,
+        organization_id: int,
+        device: str = "cpu",
+    ) -> dict:
+        self._check_user_can("node", Operation.CREATE, ident, org_id=organization_id,
+                             collab_id=collaboration_id)
+        s = self.session
+        api_key = secrets.token_urlsafe(32)
+        node = Node(
+            name=name,
+            collaboration_id=collaboration_id,
+            organization_id=organization_id,
+            device=device,
+        )
+        node.set_api_key(api_key)
+        s.add(node)
+        s.commit()
+        out = self._node_json(node)
+        out["api_key"] = api_key  # only returned at creation, like the reference
+        return out
+
+    def create_user(
+        self,
+        ident: Identity,
+        username: str,
+        password: str,
+        organization_id: int,
+        roles: list[str] | None = None,
+        **kw,
+    ) -> dict:
+        self._check_user_can("user", Operation.CREATE, ident, org_id=organization_id)
+        s = self.session
+        if s.query(User).filter_by(username=username).first():
+            raise BadRequest(f"user {username!r} exists")
+        user = User(username=username, organization_id=organization_id,
+                    **{k: v for k, v in kw.items() if k in ("firstname", "lastname", "email")})
+        user.set_password(password)
+        role_objs = []
+        for rname in roles or []:
+            role = s.query(Role).filter_by(name=rname).first()
+            if role is None:
+                raise NotFound(f"role {rname!r} not found")
+            role_objs.append(role)
+        user.roles = role_objs
+        s.add(user)
+        s.commit()
+        return self._user_json(user)
+
+    # -------------------------------------------------------------- listings
+    def _paginate(self, items: list, page: int = 1, per_page: int = 20) -> dict:
+        start = (page - 1) * per_page
+        return {
+            "data": items[start : start + per_page],
+            "links": {"total": len(items), "page": page, "per_page": per_page,
+                      "pages": max(1, -(-len(items) // per_page))},
+        }
+
+    def list_organizations(self, ident: Identity, page=1, per_page=20) -> dict:
+        self._check_user_can_view_any("organization", ident)
+        orgs = self.session.query(Organization).order_by(Organization.id).all()
+        return self._paginate([self._org_json(o) for o in orgs], page, per_page)
+
+    def list_collaborations(self, ident: Identity, page=1, per_page=20) -> dict:
+        self._check_user_can_view_any("collaboration", ident)
+        cs = self.session.query(Collaboration).order_by(Collaboration.id).all()
+        rc = self._rc("collaboration", ident)
+        if rc.allowed_scope(Operation.VIEW) != Scope.GLOBAL:
+            cs = [c for c in cs if any(o.id == ident.organization_id for o in c.organizations)]
+        return self._paginate([self._collab_json(c) for c in cs], page, per_page)
+
+    def list_nodes(self, ident: Identity, page=1, per_page=20, collaboration_id=None) -> dict:
+        self._check_user_can_view_any("node", ident)
+        q = self.session.query(Node).order_by(Node.id)
+        if collaboration_id is not None:
+            q = q.filter(Node.collaboration_id == collaboration_id)
+        return self._paginate([self._node_json(n) for n in q.all()], page, per_page)
+
+    def _check_user_can_view_any(self, resource: str, ident: Identity) -> None:
+        if ident.type != "user":
+            return  # nodes/containers have implicit view of their collaboration
+        rc = self._rc(resource, ident)
+        if rc.allowed_scope(Operation.VIEW) is None:
+            raise PermissionError_(f"user lacks view permission on {resource}")
+
+    def get_organization(self, ident: Identity, org_id: int) -> dict:
+        org = self.session.get(Organization, org_id)
+        if org is None:
+            raise NotFound(f"organization {org_id} not found")
+        return self._org_json(org)
+
+    # ------------------------------------------------------------------ tasks
+    def _next_job_id(self) -> int:
+        with self._job_counter_lock:
+            s = self.session
+            max_job = s.query(Task).order_by(Task.job_id.desc()).first()
+            return (max_job.job_id if max_job else 0) + 1
+
+    def create_task(
+        self,
+        ident: Identity,
+        *,
+        name: str = "",
+        image: str,
+        collaboration_id: int | None = None,
+        organizations: list[dict],
+        description: str = "",
+        databases: list[dict] | None = None,
+        study_id: int | None = None,
+        method: str = "",
+    ) -> dict:
+        """POST /api/task.
+
+        ``organizations`` is a list of ``{"id": org_id, "input": b64-str}``;
+        the input is already per-org encrypted by the client (or plain b64 in
+        unencrypted collaborations) — the server never sees plaintext.
+        """
+        with self._lock:
+            s = self.session
+            parent: Task | None = None
+            init_user_id = None
+            if ident.type == "user":
+                self._check_user_can("task", Operation.CREATE, ident, collab_id=collaboration_id)
+                init_user_id = ident.id
+            elif ident.type == "container":
+                # subtask fan-out: same collaboration, live parent
+                parent = s.get(Task, ident.task_id)
+                if parent is None:
+                    raise PermissionError_("container's parent task vanished")
+                if collaboration_id is None:
+                    collaboration_id = parent.collaboration_id
+                if collaboration_id != parent.collaboration_id:
+                    raise PermissionError_("container may only create tasks in its collaboration")
+                if image != parent.image:
+                    raise PermissionError_("subtask image must match parent task image")
+                parent_statuses = [r.status for r in parent.runs]
+                if any(RunStatus(st) == RunStatus.KILLED for st in parent_statuses):
+                    raise PermissionError_("parent task was killed")
+            else:
+                raise PermissionError_("nodes cannot create tasks")
+
+            collab = s.get(Collaboration, collaboration_id)
+            if collab is None:
+                raise NotFound(f"collaboration {collaboration_id} not found")
+            member_ids = {o.id for o in collab.organizations}
+            org_inputs = {o["id"]: o.get("input", "") for o in organizations}
+            if not org_inputs:
+                raise BadRequest("task needs at least one target organization")
+            if not set(org_inputs) <= member_ids:
+                raise BadRequest("target organizations must be collaboration members")
+            if study_id is not None:
+                study = s.get(Study, study_id)
+                if study is None or study.collaboration_id != collaboration_id:
+                    raise BadRequest("invalid study for this collaboration")
+                study_member_ids = {o.id for o in study.organizations}
+                if not set(org_inputs) <= study_member_ids:
+                    raise BadRequest("target organizations must be study members")
+
+            task = Task(
+                name=name,
+                description=description,
+                image=image,
+                method=method,
+                collaboration_id=collaboration_id,
+                study_id=study_id,
+                parent_id=parent.id if parent else None,
+                job_id=parent.job_id if parent else self._next_job_id(),
+                init_org_id=ident.organization_id,
+                init_user_id=init_user_id,
+                databases=json.dumps(databases or []),
+            )
+            s.add(task)
+            s.flush()
+            for org_id, input_b64 in org_inputs.items():
+                node = (
+                    s.query(Node)
+                    .filter_by(collaboration_id=collaboration_id, organization_id=org_id)
+                    .first()
+                )
+                run = Run(
+                    task_id=task.id,
+                    organization_id=org_id,
+                    node_id=node.id if node else None,
+                    status=RunStatus.PENDING.value,
+                    input=input_b64,
+                )
+                s.add(run)
+            s.commit()
+            task_json = self._task_json(task)
+
+        # socketio.emit('new_task', room=collaboration_X_organization_Y)
+        for org_id in org_inputs:
+            self.events.emit(
+                "new_task",
+                {"task_id": task.id, "collaboration_id": collaboration_id, "organization_id": org_id},
+                room=f"collaboration_{collaboration_id}_organization_{org_id}",
+            )
+        return task_json
+
+    def get_task(self, ident: Identity, task_id: int) -> dict:
+        task = self.session.get(Task, task_id)
+        if task is None:
+            raise NotFound(f"task {task_id} not found")
+        if ident.type == "user":
+            self._check_user_can("task", Operation.VIEW, ident,
+                                 org_id=task.init_org_id, collab_id=task.collaboration_id)
+        return self._task_json(task)
+
+    def list_tasks(self, ident: Identity, page=1, per_page=20, **filters) -> dict:
+        self._check_user_can_view_any("task", ident)
+        q = self.session.query(Task).order_by(Task.id)
+        if filters.get("parent_id") is not None:
+            q = q.filter(Task.parent_id == filters["parent_id"])
+        if filters.get("job_id") is not None:
+            q = q.filter(Task.job_id == filters["job_id"])
+        if filters.get("collaboration_id") is not None:
+            q = q.filter(Task.collaboration_id == filters["collaboration_id"])
+        return self._paginate([self._task_json(t) for t in q.all()], page, per_page)
+
+    def kill_task(self, ident: Identity, task_id: int) -> dict:
+        """DELETE /api/task/<id> — emit kill_containers to the nodes."""
+        s = self.session
+        task = s.get(Task, task_id)
+        if task is None:
+            raise NotFound(f"task {task_id} not found")
+        if ident.type == "user":
+            self._check_user_can("task", Operation.DELETE, ident,
+                                 org_id=task.init_org_id, collab_id=task.collaboration_id)
+        kill_list = []
+        for run in task.runs:
+            if not has_task_finished(run.status):
+                run.status = RunStatus.KILLED.value
+                run.finished_at = utcnow()
+                kill_list.append({"run_id": run.id, "task_id": task.id,
+                                  "organization_id": run.organization_id})
+        s.commit()
+        self.events.emit(
+            "kill_containers",
+            {"task_id": task.id, "kill_list": kill_list},
+            room=f"collaboration_{task.collaboration_id}",
+        )
+        return {"id": task.id, "status": "killed"}
+
+    # ------------------------------------------------------------------- runs
+    def list_runs(self, ident: Identity, task_id: int | None = None,
+                  node_id: int | None = None, state: str | None = None,
+                  include_data: bool = True, page=1, per_page=50) -> dict:
+        q = self.session.query(Run).order_by(Run.id)
+        if task_id is not None:
+            q = q.filter(Run.task_id == task_id)
+        if node_id is not None:
+            q = q.filter(Run.node_id == node_id)
+        runs = q.all()
+        if state == "open":
+            runs = [r for r in runs if not has_task_finished(r.status)]
+        if ident.type == "node":
+            runs = [r for r in runs if r.node_id == ident.id]
+        return self._paginate([self._run_json(r, include_data) for r in runs], page, per_page)
+
+    def get_run(self, ident: Identity, run_id: int) -> dict:
+        run = self.session.get(Run, run_id)
+        if run is None:
+            raise NotFound(f"run {run_id} not found")
+        return self._run_json(run, True)
+
+    def patch_run(self, ident: Identity, run_id: int, **fields) -> dict:
+        """PATCH /api/run/<id> — node reports status/result/log."""
+        s = self.session
+        run = s.get(Run, run_id)
+        if run is None:
+            raise NotFound(f"run {run_id} not found")
+        if ident.type == "node" and run.node_id != ident.id:
+            raise PermissionError_("node may only patch its own runs")
+        if ident.type == "user":
+            self._check_user_can("run", Operation.EDIT, ident,
+                                 org_id=run.organization_id)
+        old_status = run.status
+        for key in ("status", "result", "log", "input"):
+            if key in fields and fields[key] is not None:
+                setattr(run, key, str(fields[key]) if key == "status" else fields[key])
+        if "started_at" in fields:
+            run.started_at = fields["started_at"] or utcnow()
+        if fields.get("status") == RunStatus.ACTIVE.value and run.started_at is None:
+            run.started_at = utcnow()
+        if fields.get("status") is not None and has_task_finished(fields["status"]):
+            run.finished_at = utcnow()
+        s.commit()
+        if run.status != old_status:
+            task = run.task
+            self.events.emit(
+                "algorithm_status_change",
+                {
+                    "task_id": run.task_id,
+                    "run_id": run.id,
+                    "status": run.status,
+                    "organization_id": run.organization_id,
+                    "job_id": task.job_id if task else None,
+                    "parent_id": task.parent_id if task else None,
+                },
+                room=f"collaboration_{task.collaboration_id}" if task else "global",
+            )
+        return self._run_json(run, True)
+
+    def list_results(self, ident: Identity, task_id: int, page=1, per_page=50) -> dict:
+        """GET /api/result?task_id= — result field only (v4 split result/run)."""
+        runs = self.session.query(Run).filter(Run.task_id == task_id).order_by(Run.id).all()
+        items = [
+            {
+                "id": r.id,
+                "run": f"/api/run/{r.id}",
+                "task": f"/api/task/{r.task_id}",
+                "result": r.result,
+                "organization_id": r.organization_id,
+                "status": r.status,
+            }
+            for r in runs
+        ]
+        return self._paginate(items, page, per_page)
+
+    # ------------------------------------------------------- node-side helpers
+    def node_sync_open_runs(self, ident: Identity) -> list[dict]:
+        """Crash recovery: node re-fetches unfinished runs on boot
+        (``Node.sync_task_queue_with_server``)."""
+        if ident.type != "node":
+            raise PermissionError_("only nodes sync run queues")
+        runs = self.list_runs(ident, node_id=ident.id, state="open")["data"]
+        return runs
+
+    def node_set_offline(self, node_id: int) -> None:
+        s = self.session
+        node = s.get(Node, node_id)
+        if node:
+            node.status = "offline"
+            s.commit()
+            self.events.emit(
+                "node-status-change",
+                {"id": node.id, "name": node.name, "status": "offline"},
+                room=f"collaboration_{node.collaboration_id}",
+            )
+
+    def organization_public_key(self, org_id: int) -> str:
+        org = self.session.get(Organization, org_id)
+        if org is None:
+            raise NotFound(f"organization {org_id} not found")
+        return org.public_key or ""
+
+    # ------------------------------------------------------------- misc / ops
+    def version(self) -> dict:
+        from .. import __version__
+
+        return {"version": __version__}
+
+    def health(self) -> dict:
+        try:
+            self.session.query(Organization).first()
+            db_ok = True
+        except Exception:  # noqa: BLE001
+            db_ok = False
+        return {"status": "ok" if db_ok else "degraded", "database": db_ok}
+
+    # ---------------------------------------------------------- serialization
+    def _org_json(self, o: Organization) -> dict:
+        return {
+            "id": o.id,
+            "name": o.name,
+            "domain": o.domain,
+            "country": o.country,
+            "public_key": o.public_key,
+            "collaborations": [{"id": c.id} for c in o.collaborations],
+        }
+
+    def _collab_json(self, c: Collaboration) -> dict:
+        return {
+            "id": c.id,
+            "name": c.name,
+            "encrypted": bool(c.encrypted),
+            "organizations": [{"id": o.id, "name": o.name} for o in c.organizations],
+        }
+
+    def _node_json(self, n: Node) -> dict:
+        return {
+            "id": n.id,
+            "name": n.name,
+            "collaboration_id": n.collaboration_id,
+            "organization_id": n.organization_id,
+            "status": n.status,
+            "device": n.device,
+            "last_seen": n.last_seen.isoformat() if n.last_seen else None,
+        }
+
+    def _user_json(self, u: User) -> dict:
+        return {
+            "id": u.id,
+            "username": u.username,
+            "firstname": u.firstname,
+            "lastname": u.lastname,
+            "email": u.email,
+            "organization_id": u.organization_id,
+            "roles": [{"id": r.id, "name": r.name} for r in u.roles],
+        }
+
+    def _task_json(self, t: Task) -> dict:
+        statuses = [r.status for r in t.runs]
+        return {
+            "id": t.id,
+            "name": t.name,
+            "description": t.description,
+            "image": t.image,
+            "method": t.method,
+            "status": task_status_from_runs(statuses).value,
+            "collaboration_id": t.collaboration_id,
+            "study_id": t.study_id,
+            "parent_id": t.parent_id,
+            "job_id": t.job_id,
+            "init_org_id": t.init_org_id,
+            "init_user_id": t.init_user_id,
+            "databases": json.loads(t.databases or "[]"),
+            "runs": f"/api/run?task_id={t.id}",
+            "results": f"/api/result?task_id={t.id}",
+            "created_at": t.created_at.isoformat() if t.created_at else None,
+        }
+
+    def _run_json(self, r: Run, include_data: bool = True) -> dict:
+        out = {
+            "id": r.id,
+            "task_id": r.task_id,
+            "organization_id": r.organization_id,
+            "node_id": r.node_id,
+            "status": r.status,
+            "ports": json.loads(r.ports or "[]"),
+            "assigned_at": r.assigned_at.isoformat() if r.assigned_at else None,
+            "started_at": r.started_at.isoformat() if r.started_at else None,
+            "finished_at": r.finished_at.isoformat() if r.finished_at else None,
+            "log": r.log,
+        }
+        if include_data:
+            out["input"] = r.input
+            out["result"] = r.result
+        return out
