@@ -1,0 +1,163 @@
+#!/usr/bin/env python3
+"""Flagship benchmark — FedAvg rounds/sec (BASELINE.json metric).
+
+Default config (headline, BASELINE config[2]): FedAvg ResNet-18 bf16 on
+synthetic 224×224 shards, N stations = N GPUs (one rank per GPU over RCCL),
+local_steps SGD steps per round + weighted parameter all-reduce.
+
+Contract (driver): ``python bench.py --gpus N --steps K --warmup W``;
+for N>1 launched via torch.distributed.run with one rank per GPU. Does W
+untimed warmup rounds, times exactly K rounds between barrier+synchronize
+pairs, takes the MAX elapsed over ranks, and rank 0 prints ONE JSON line.
+
+A "step" here is one FedAvg ROUND (local_steps local minibatch SGD steps +
+aggregation) — the reference's unit of work (one central
+round-trip of subtask fan-out + fit + collect).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import sys
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent))
+
+import torch
+
+from algorithms.fedavg import make_batches, _build
+from vantage6_amd.parallel.comm import init_from_env
+from vantage6_amd.parallel.fedavg import FedAvgConfig, FedAvgTrainer
+
+
+def synthetic_dataset(model: str, n: int, seed: int, model_kwargs: dict):
+    g = torch.Generator().manual_seed(seed)
+    if model == "resnet18":
+        return {
+            "x": torch.randn(n, 3, 224, 224, generator=g),
+            "y": torch.randint(0, model_kwargs.get("num_classes", 1000), (n,), generator=g),
+        }
+    if model == "logreg":
+        d = model_kwargs.get("in_features", 128)
+        c = model_kwargs.get("num_classes", 2)
+        x = torch.randn(n, d, generator=g)
+        w = torch.randn(d, c, generator=g)
+        y = (x @ w).argmax(-1)
+        return {"x": x, "y": y}
+    # token models
+    seq = model_kwargs.pop("_seq_len", 512)
+    vocab = model_kwargs.get("vocab", 30522 if model.startswith("bert") else 32000)
+    return {"input_ids": torch.randint(0, vocab, (n, seq), generator=g)}
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20, help="timed FedAvg rounds")
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--model", default="resnet18",
+                    choices=["resnet18", "logreg", "bert_base", "llama_1b"])
+    ap.add_argument("--batch-size", type=int, default=None, help="per-station batch")
+    ap.add_argument("--local-steps", type=int, default=4)
+    ap.add_argument("--seq-len", type=int, default=512)
+    ap.add_argument("--shard-size", type=int, default=None, help="samples per station")
+    ap.add_argument("--secure-agg", action="store_true")
+    ap.add_argument("--dtype", default="bf16", choices=["bf16"])
+    ap.add_argument("--device", default=None, help="override (e.g. cpu for smoke)")
+    args = ap.parse_args()
+
+    comm = init_from_env()
+    device = torch.device(args.device) if args.device else comm.device
+    on_gpu = device.type == "cuda"
+    if on_gpu:
+        torch.backends.cudnn.benchmark = True  # MIOpen find-mode
+        torch.backends.cuda.matmul.allow_bf16_reduced_precision_reduction = True
+
+    defaults = {
+        "resnet18": {"batch": 32, "kwargs": {"num_classes": 1000}, "shard": 256},
+        "logreg": {"batch": 1024, "kwargs": {"in_features": 128, "num_classes": 2},
+                   "shard": 16384},
+        "bert_base": {"batch": 8, "kwargs": {}, "shard": 64},
+        "llama_1b": {"batch": 2, "kwargs": {}, "shard": 16},
+    }[args.model]
+    batch = args.batch_size or defaults["batch"]
+    shard = args.shard_size or max(defaults["shard"], batch)
+    mk = dict(defaults["kwargs"])
+    if args.model in ("bert_base", "llama_1b"):
+        mk["_seq_len"] = args.seq_len
+
+    dataset = synthetic_dataset(args.model, shard, seed=100 + comm.rank, model_kwargs=mk)
+    mk.pop("_seq_len", None)
+
+    net, loss_fn = _build(args.model, mk, device)
+    batches = make_batches(dataset, batch, device, seed=17 + comm.rank)
+    cfg = FedAvgConfig(
+        local_steps=args.local_steps,
+        lr=0.1 if args.model in ("resnet18", "logreg") else 1e-4,
+        optimizer="sgd" if args.model in ("resnet18", "logreg") else "adamw",
+        secure_agg=args.secure_agg,
+    )
+    trainer = FedAvgTrainer(net, loss_fn, batches, comm, cfg, n_samples=shard)
+
+    for _ in range(args.warmup):
+        trainer.run_round()
+
+    comm.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    losses = []
+    for _ in range(args.steps):
+        stats = trainer.run_round()
+        losses.append(stats["loss"])
+    comm.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # MAX elapsed over ranks
+    t = torch.tensor([elapsed], device=device if on_gpu else "cpu")
+    if comm.is_distributed:
+        import torch.distributed as dist
+
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    n = comm.world_size
+    rounds_per_sec = args.steps / elapsed
+    if comm.rank == 0:
+        out = {
+            "metric": "FedAvg rounds/sec (ResNet-18 @ N stations)"
+            if args.model == "resnet18" else f"FedAvg rounds/sec ({args.model})",
+            "value": rounds_per_sec,
+            "unit": "rounds/s",
+            "n_gpus": n,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1e3,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": batch * n,
+                "per_station_batch": batch,
+                "local_steps": args.local_steps,
+                "seq_len": args.seq_len if args.model in ("bert_base", "llama_1b") else None,
+                "parallelism": f"federated-dp{n}",
+                "samples_per_round": batch * args.local_steps * n,
+                "samples_per_sec": rounds_per_sec * batch * args.local_steps * n,
+                "secure_agg": args.secure_agg,
+                "final_loss": losses[-1],
+                "n_params": trainer.flat.numel,
+                "hip_ops": __import__("vantage6_amd.ops", fromlist=["has_hip_ops"]).has_hip_ops(),
+            },
+        }
+        print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,204 @@
+"""Numerics tests for the hand-written CDNA4 kernels vs plain PyTorch fp32
+references (tolerance-tiered for bf16 I/O). All @pytest.mark.gpu."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    DEV = torch.device("cuda:0")
+else:
+    pytest.skip("needs MI355X", allow_module_level=True)
+
+
+@pytest.fixture(scope="module")
+def ops():
+    from vantage6_amd.ops import hip_ops
+
+    return hip_ops()
+
+
+def bf16(x):
+    return x.to(torch.bfloat16)
+
+
+# --------------------------------------------------------------------- GEMM
+@pytest.mark.parametrize("m,n,k", [(128, 128, 64), (256, 512, 128),
+                                   (512, 1000, 512), (129, 257, 100),
+                                   (64, 64, 4096)])
+def test_gemm_nt(ops, m, n, k):
+    torch.manual_seed(0)
+    a = bf16(torch.randn(m, k)).to(DEV)
+    b = bf16(torch.randn(n, k)).to(DEV)
+    c = ops.gemm_bf16(a, b, False, True)
+    ref = (a.float() @ b.float().t())
+    err = (c.float() - ref).abs().max().item()
+    scale = ref.abs().max().item() + 1e-6
+    assert err / scale < 2e-2, f"rel err {err/scale}"
+
+
+def test_gemm_transpose_detecting(ops):
+    """Asymmetric operands so a row/col swap cannot pass (guide §5.4 #16)."""
+    m, n, k = 128, 64, 32
+    a = torch.zeros(m, k)
+    a[3, 5] = 1.0
+    b = torch.zeros(n, k)
+    b[7, 5] = 2.0
+    c = ops.gemm_bf16(bf16(a).to(DEV), bf16(b).to(DEV), False, True).float()
+    assert c[3, 7].item() == pytest.approx(2.0, rel=1e-2)
+    assert c.abs().sum().item() == pytest.approx(2.0, rel=1e-2)
+
+
+@pytest.mark.parametrize("ta,tb", [(False, False), (True, False), (False, True)])
+def test_gemm_layouts(ops, ta, tb):
+    torch.manual_seed(1)
+    m, n, k = 192, 320, 256
+    a_shape = (k, m) if ta else (m, k)
+    b_shape = (n, k) if tb else (k, n)
+    a = bf16(torch.randn(*a_shape)).to(DEV)
+    b = bf16(torch.randn(*b_shape)).to(DEV)
+    c = ops.gemm_bf16(a, b, ta, tb).float()
+    af = a.float().t() if ta else a.float()
+    bf = b.float().t() if tb else b.float()
+    ref = af @ bf
+    rel = (c - ref).abs().max().item() / (ref.abs().max().item() + 1e-6)
+    assert rel < 2e-2
+
+
+def test_transpose(ops):
+    x = bf16(torch.randn(130, 70)).to(DEV)
+    xt = ops.transpose_bf16(x)
+    assert torch.equal(xt, x.t().contiguous())
+
+
+# -------------------------------------------------------------------- optim
+def test_fused_sgd_matches_cpu_reference(ops):
+    from vantage6_amd.ops.optim import fused_sgd
+
+    torch.manual_seed(2)
+    n = 100_003
+    master = torch.randn(n)
+    param = bf16(master.clone())
+    grad = bf16(torch.randn(n))
+    mom = torch.randn(n)
+    # CPU reference
+    p_c, m_c, g_c, v_c = param.clone(), master.clone(), grad.clone(), mom.clone()
+    fused_sgd(p_c, m_c, g_c, v_c, lr=0.1, momentum=0.9, weight_decay=1e-4)
+    # GPU kernel
+    p_g, m_g, g_g, v_g = (t.to(DEV) for t in (param, master, grad, mom))
+    fused_sgd(p_g, m_g, g_g, v_g, lr=0.1, momentum=0.9, weight_decay=1e-4)
+    assert torch.allclose(m_g.cpu(), m_c, rtol=1e-5, atol=1e-6)
+    assert torch.allclose(v_g.cpu(), v_c, rtol=1e-5, atol=1e-6)
+    assert torch.equal(p_g.cpu().view(torch.uint16), p_c.view(torch.uint16))
+
+
+def test_fused_adamw_matches_cpu_reference(ops):
+    from vantage6_amd.ops.optim import fused_adamw
+
+    torch.manual_seed(3)
+    n = 65_537
+    master = torch.randn(n)
+    param = bf16(master.clone())
+    grad = bf16(torch.randn(n))
+    ea, es = torch.randn(n).abs(), torch.rand(n)
+    args = dict(lr=1e-3, beta1=0.9, beta2=0.999, eps=1e-8, weight_decay=0.01, step=3)
+    p_c, m_c = param.clone(), master.clone()
+    ea_c, es_c = ea.clone(), es.clone()
+    fused_adamw(p_c, m_c, grad.clone(), ea_c, es_c, **args)
+    p_g, m_g, g_g, ea_g, es_g = (t.to(DEV) for t in (param, master, grad, ea, es))
+    fused_adamw(p_g, m_g, g_g, ea_g, es_g, **args)
+    assert torch.allclose(m_g.cpu(), m_c, rtol=1e-4, atol=1e-6)
+    assert torch.allclose(ea_g.cpu(), ea_c, rtol=1e-5, atol=1e-6)
+    assert torch.allclose(es_g.cpu(), es_c, rtol=1e-5, atol=1e-6)
+
+
+# -------------------------------------------------------------------- norms
+@pytest.mark.parametrize("rows,h", [(64, 768), (37, 2048), (128, 8192)])
+def test_layernorm_fwd_bwd(ops, rows, h):
+    from vantage6_amd.ops.norms import layer_norm
+
+    torch.manual_seed(4)
+    x = bf16(torch.randn(rows, h)).to(DEV).requires_grad_(True)
+    w = bf16(torch.randn(h)).to(DEV).requires_grad_(True)
+    b = bf16(torch.randn(h)).to(DEV).requires_grad_(True)
+    y = layer_norm(x, w, b)
+    dy = bf16(torch.randn(rows, h)).to(DEV)
+    y.backward(dy)
+
+    xf = x.detach().float().cpu().requires_grad_(True)
+    wf = w.detach().float().cpu().requires_grad_(True)
+    bf_ = b.detach().float().cpu().requires_grad_(True)
+    yref = torch.nn.functional.layer_norm(xf, (h,), wf, bf_)
+    yref.backward(dy.float().cpu())
+
+    assert (y.float().cpu() - yref).abs().max() < 5e-2
+    assert (x.grad.float().cpu() - xf.grad).abs().max() < 5e-2
+    tol = 0.5 + 2e-3 * rows ** 0.5 * 3  # dw/db accumulate over rows (bf16 out)
+    assert (w.grad.float().cpu() - wf.grad).abs().max() < tol
+    assert (b.grad.float().cpu() - bf_.grad).abs().max() < tol
+
+
+@pytest.mark.parametrize("rows,h", [(64, 2048), (33, 4096)])
+def test_rmsnorm_fwd_bwd(ops, rows, h):
+    from vantage6_amd.ops.norms import rms_norm
+
+    torch.manual_seed(5)
+    x = bf16(torch.randn(rows, h)).to(DEV).requires_grad_(True)
+    w = bf16(torch.randn(h)).to(DEV).requires_grad_(True)
+    y = rms_norm(x, w)
+    dy = bf16(torch.randn(rows, h)).to(DEV)
+    y.backward(dy)
+
+    xf = x.detach().float().cpu().requires_grad_(True)
+    wf = w.detach().float().cpu().requires_grad_(True)
+    rstd = (xf.pow(2).mean(-1, keepdim=True) + 1e-6).rsqrt()
+    yref = xf * rstd * wf
+    yref.backward(dy.float().cpu())
+    assert (y.float().cpu() - yref).abs().max() < 5e-2
+    assert (x.grad.float().cpu() - xf.grad).abs().max() < 5e-2
+    assert (w.grad.float().cpu() - wf.grad).abs().max() < 0.5 + 6e-3 * rows ** 0.5
+
+
+# ------------------------------------------------------------------- secagg
+def test_secagg_masks_cancel(ops):
+    """Sum of all ranks' masked buffers == sum of quantized values exactly."""
+    torch.manual_seed(6)
+    n = 50_001
+    world = 4
+    master = torch.randn(n).to(DEV)
+    scale = float(1 << 32)
+    total = torch.zeros(n, dtype=torch.int64, device=DEV)
+    for rank in range(world):
+        q = torch.empty(n, dtype=torch.int64, device=DEV)
+        ops.secagg_quantize_mask(q, master, 1.0 / world, scale, 1234, rank, world)
+        total += q  # int64 wraparound == modular sum
+    expect = (master.double() * (1.0 / world) * scale).round().long() * world
+    assert torch.equal(total, expect)
+
+
+def test_secagg_single_mask_is_random(ops):
+    n = 1024
+    master = torch.zeros(n).to(DEV)
+    q = torch.empty(n, dtype=torch.int64, device=DEV)
+    ops.secagg_quantize_mask(q, master, 1.0, float(1 << 32), 99, 0, 2)
+    # masked zeros should look like full-range noise, not zeros
+    assert (q == 0).sum().item() < 5
+
+
+# ----------------------------------------------------------------------- dp
+def test_dp_noise_statistics(ops):
+    torch.manual_seed(7)
+    n = 1 << 20
+    g = torch.zeros(n, dtype=torch.bfloat16, device=DEV)
+    ops.dp_scale_noise(g, 1.0, 0.5, 42)
+    gf = g.float()
+    assert abs(gf.mean().item()) < 5e-3
+    assert abs(gf.std().item() - 0.5) < 1e-2
+
+
+def test_grad_norm2(ops):
+    torch.manual_seed(8)
+    x = bf16(torch.randn(1_000_003)).to(DEV)
+    n2 = ops.grad_norm2_partials(x).sum().item()
+    ref = x.float().pow(2).sum().item()
+    assert n2 == pytest.approx(ref, rel=1e-3)

@@ -1,0 +1,51 @@
+"""Model + loss factory used by the fedavg algorithm packages and bench."""
+from __future__ import annotations
+
+from typing import Callable
+
+import torch
+import torch.nn.functional as F
+
+
+def build_model(name: str, device="cpu", dtype=torch.bfloat16, **kw) -> torch.nn.Module:
+    name = name.lower()
+    if name in ("logreg", "logistic_regression"):
+        from .logreg import logreg
+
+        return logreg(device=device, dtype=dtype, **kw)
+    if name in ("resnet18", "resnet-18"):
+        from .resnet import resnet18
+
+        return resnet18(device=device, dtype=dtype, **kw)
+    if name in ("bert", "bert_base", "bert-base"):
+        from .transformer import bert_base
+
+        return bert_base(device=device, dtype=dtype, **kw)
+    if name in ("llama", "llama_1b", "llama-1b", "llama-3-1b"):
+        from .transformer import llama_1b
+
+        return llama_1b(device=device, dtype=dtype, **kw)
+    raise ValueError(f"unknown model {name!r}")
+
+
+def model_loss(name: str) -> Callable:
+    """(model, batch) -> scalar loss, fp32 loss computation."""
+    name = name.lower()
+    if name in ("bert", "bert_base", "bert-base", "llama", "llama_1b", "llama-1b",
+                "llama-3-1b"):
+
+        def lm_loss(model, batch):
+            ids = batch["input_ids"]
+            logits = model(ids[:, :-1])
+            return F.cross_entropy(
+                logits.float().reshape(-1, logits.shape[-1]),
+                ids[:, 1:].reshape(-1),
+            )
+
+        return lm_loss
+
+    def cls_loss(model, batch):
+        logits = model(batch["x"])
+        return F.cross_entropy(logits.float(), batch["y"])
+
+    return cls_loss

@@ -1,0 +1,105 @@
+"""ResNet-18 (bf16) — the headline FedAvg benchmark model.
+
+Hand-written (no torchvision in the image): standard BasicBlock ResNet-18
+for 224×224 inputs. Convolutions run through MIOpen (library path — plain
+conv is a library GEMM-class op); the classifier head runs on the
+hand-written MFMA GEMM (ops.gemm.Linear) so the fit path exercises the
+native kernels; BatchNorm stays fp32-stats via torch.
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+from ..ops.gemm import Linear as MfmaLinear
+
+
+class BatchNorm2dBF16(nn.Module):
+    """BatchNorm with bf16 affine params (flat-param-space compatible) and
+    fp32 statistics/compute (stability). Forward casts around F.batch_norm."""
+
+    def __init__(self, num_features: int, eps: float = 1e-5, momentum: float = 0.1):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(num_features, dtype=torch.bfloat16))
+        self.bias = nn.Parameter(torch.zeros(num_features, dtype=torch.bfloat16))
+        self.register_buffer("running_mean", torch.zeros(num_features))
+        self.register_buffer("running_var", torch.ones(num_features))
+        self.eps, self.momentum = eps, momentum
+
+    def forward(self, x):
+        y = torch.nn.functional.batch_norm(
+            x.float(), self.running_mean, self.running_var,
+            self.weight.float(), self.bias.float(),
+            self.training, self.momentum, self.eps,
+        )
+        return y.to(x.dtype)
+
+
+def conv3x3(cin, cout, stride=1):
+    return nn.Conv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
+
+
+class BasicBlock(nn.Module):
+    expansion = 1
+
+    def __init__(self, cin, cout, stride=1):
+        super().__init__()
+        self.conv1 = conv3x3(cin, cout, stride)
+        self.bn1 = BatchNorm2dBF16(cout)
+        self.conv2 = conv3x3(cout, cout)
+        self.bn2 = BatchNorm2dBF16(cout)
+        self.relu = nn.ReLU(inplace=True)
+        self.down = None
+        if stride != 1 or cin != cout:
+            self.down = nn.Sequential(
+                nn.Conv2d(cin, cout, 1, stride=stride, bias=False),
+                BatchNorm2dBF16(cout),
+            )
+
+    def forward(self, x):
+        idt = x if self.down is None else self.down(x)
+        out = self.relu(self.bn1(self.conv1(x)))
+        out = self.bn2(self.conv2(out))
+        return self.relu(out + idt)
+
+
+class ResNet18(nn.Module):
+    def __init__(self, num_classes: int = 1000, use_mfma_head: bool = True):
+        super().__init__()
+        self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
+        self.bn1 = BatchNorm2dBF16(64)
+        self.relu = nn.ReLU(inplace=True)
+        self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
+        layers = []
+        cin = 64
+        for cout, stride in ((64, 1), (128, 2), (256, 2), (512, 2)):
+            layers.append(BasicBlock(cin, cout, stride))
+            layers.append(BasicBlock(cout, cout))
+            cin = cout
+        self.layers = nn.Sequential(*layers)
+        self.avgpool = nn.AdaptiveAvgPool2d(1)
+        if use_mfma_head:
+            self.fc = MfmaLinear(512, num_classes)
+        else:
+            self.fc = nn.Linear(512, num_classes)
+
+    def forward(self, x):
+        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.layers(x)
+        x = self.avgpool(x).flatten(1)
+        return self.fc(x)
+
+
+def resnet18(num_classes: int = 1000, dtype=torch.bfloat16, device="cpu",
+             use_mfma_head: bool | None = None) -> ResNet18:
+    if use_mfma_head is None:
+        # MFMA head requires CUDA(HIP) extension or CPU reference — both exist
+        use_mfma_head = True
+    m = ResNet18(num_classes, use_mfma_head=use_mfma_head)
+    m = m.to(device=device, dtype=dtype)
+    # keep BN statistics buffers fp32 regardless of model dtype
+    for mod in m.modules():
+        if isinstance(mod, BatchNorm2dBF16):
+            mod.running_mean.data = mod.running_mean.data.float()
+            mod.running_var.data = mod.running_var.data.float()
+    return m

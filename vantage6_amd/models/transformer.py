@@ -1,0 +1,138 @@
+"""Transformer models for configs 3 and 4.
+
+- ``bert_base``: 12-layer post-LN encoder (hidden 768, 12 heads, MLM-style
+  token-classification head) — the DP-SGD config.
+- ``llama_1b``: Llama-3.2-1B-shaped decoder (16 layers, hidden 2048, GQA
+  32/8 heads, SwiGLU 8192, RMSNorm, RoPE) — the secure-aggregation config.
+
+All Linear layers run on the hand-written MFMA GEMM (ops.gemm.Linear);
+norms run on the hand-written LayerNorm/RMSNorm HIP kernels; attention uses
+torch SDPA (library path). bf16 throughout.
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..ops.gemm import Linear as MfmaLinear
+from ..ops.norms import LayerNorm, RMSNorm
+
+
+# ---------------------------------------------------------------- BERT-base
+class BertLayer(nn.Module):
+    def __init__(self, hidden: int, heads: int, ffn: int):
+        super().__init__()
+        self.heads = heads
+        self.qkv = MfmaLinear(hidden, 3 * hidden)
+        self.proj = MfmaLinear(hidden, hidden)
+        self.ln1 = LayerNorm(hidden)
+        self.fc1 = MfmaLinear(hidden, ffn)
+        self.fc2 = MfmaLinear(ffn, hidden)
+        self.ln2 = LayerNorm(hidden)
+
+    def forward(self, x):
+        B, S, H = x.shape
+        qkv = self.qkv(x).reshape(B, S, 3, self.heads, H // self.heads)
+        q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))
+        att = F.scaled_dot_product_attention(q, k, v)
+        att = att.transpose(1, 2).reshape(B, S, H)
+        x = self.ln1(x + self.proj(att))
+        x = self.ln2(x + self.fc2(F.gelu(self.fc1(x))))
+        return x
+
+
+class BertBase(nn.Module):
+    def __init__(self, vocab: int = 30522, hidden: int = 768, layers: int = 12,
+                 heads: int = 12, ffn: int = 3072, max_seq: int = 512):
+        super().__init__()
+        self.tok = nn.Embedding(vocab, hidden)
+        self.pos = nn.Embedding(max_seq, hidden)
+        self.ln = LayerNorm(hidden)
+        self.blocks = nn.ModuleList(BertLayer(hidden, heads, ffn) for _ in range(layers))
+        self.head = MfmaLinear(hidden, vocab, bias=False)
+
+    def forward(self, input_ids):
+        B, S = input_ids.shape
+        pos = torch.arange(S, device=input_ids.device)
+        x = self.ln(self.tok(input_ids) + self.pos(pos)[None])
+        for blk in self.blocks:
+            x = blk(x)
+        return self.head(x)
+
+
+# ----------------------------------------------------------------- Llama-1B
+def _rope(q, k, base: float = 500000.0):
+    # rotary embeddings, applied per head-dim pair
+    B, Hn, S, D = q.shape
+    half = D // 2
+    inv = 1.0 / base ** (torch.arange(half, device=q.device, dtype=torch.float32) / half)
+    t = torch.arange(S, device=q.device, dtype=torch.float32)
+    freqs = torch.outer(t, inv)  # [S, half]
+    cos, sin = freqs.cos()[None, None], freqs.sin()[None, None]
+
+    def rot(x):
+        x1, x2 = x[..., :half].float(), x[..., half:].float()
+        return torch.cat([x1 * cos - x2 * sin, x1 * sin + x2 * cos], -1).to(x.dtype)
+
+    return rot(q), rot(k)
+
+
+class LlamaLayer(nn.Module):
+    def __init__(self, hidden: int, heads: int, kv_heads: int, ffn: int):
+        super().__init__()
+        self.heads, self.kv_heads = heads, kv_heads
+        self.head_dim = hidden // heads
+        self.q = MfmaLinear(hidden, hidden, bias=False)
+        self.k = MfmaLinear(hidden, kv_heads * self.head_dim, bias=False)
+        self.v = MfmaLinear(hidden, kv_heads * self.head_dim, bias=False)
+        self.o = MfmaLinear(hidden, hidden, bias=False)
+        self.gate = MfmaLinear(hidden, ffn, bias=False)
+        self.up = MfmaLinear(hidden, ffn, bias=False)
+        self.down = MfmaLinear(ffn, hidden, bias=False)
+        self.norm1 = RMSNorm(hidden)
+        self.norm2 = RMSNorm(hidden)
+
+    def forward(self, x):
+        B, S, H = x.shape
+        h = self.norm1(x)
+        q = self.q(h).reshape(B, S, self.heads, self.head_dim).transpose(1, 2)
+        k = self.k(h).reshape(B, S, self.kv_heads, self.head_dim).transpose(1, 2)
+        v = self.v(h).reshape(B, S, self.kv_heads, self.head_dim).transpose(1, 2)
+        q, k = _rope(q, k)
+        att = F.scaled_dot_product_attention(q, k, v, is_causal=True, enable_gqa=True)
+        att = att.transpose(1, 2).reshape(B, S, H)
+        x = x + self.o(att)
+        h = self.norm2(x)
+        x = x + self.down(F.silu(self.gate(h)) * self.up(h))
+        return x
+
+
+class Llama(nn.Module):
+    def __init__(self, vocab: int = 128256, hidden: int = 2048, layers: int = 16,
+                 heads: int = 32, kv_heads: int = 8, ffn: int = 8192):
+        super().__init__()
+        self.tok = nn.Embedding(vocab, hidden)
+        self.blocks = nn.ModuleList(
+            LlamaLayer(hidden, heads, kv_heads, ffn) for _ in range(layers)
+        )
+        self.norm = RMSNorm(hidden)
+        self.head = MfmaLinear(hidden, vocab, bias=False)
+
+    def forward(self, input_ids):
+        x = self.tok(input_ids)
+        for blk in self.blocks:
+            x = blk(x)
+        return self.head(self.norm(x))
+
+
+def bert_base(dtype=torch.bfloat16, device="cpu", **kw) -> BertBase:
+    m = BertBase(**kw)
+    return m.to(device=device, dtype=dtype)
+
+
+def llama_1b(dtype=torch.bfloat16, device="cpu", **kw) -> Llama:
+    m = Llama(**kw)
+    return m.to(device=device, dtype=dtype)

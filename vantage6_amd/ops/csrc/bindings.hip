@@ -1,0 +1,243 @@
+// Python bindings (torch extension) for the vantage6_amd CDNA4 kernels.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+#include "common.h"
+
+// kernels defined in sibling TUs
+extern "C" __global__ void fused_sgd_kernel(short*, float*, const short*, float*,
+                                            long long, float, float, float, int);
+extern "C" __global__ void fused_adamw_kernel(short*, float*, const short*, float*,
+                                              float*, long long, float, float, float,
+                                              float, float, float, float);
+extern "C" __global__ void secagg_quantize_mask_kernel(long long*, const float*,
+                                                       long long, double,
+                                                       unsigned long long, int, int);
+extern "C" __global__ void dp_scale_noise_kernel(short*, long long, float, float,
+                                                 unsigned long long);
+extern "C" __global__ void norm2_bf16_kernel(const short*, long long, float*);
+extern "C" __global__ void layernorm_fwd_kernel(const short*, const short*,
+                                                const short*, short*, float*, float*,
+                                                long long, int, float);
+extern "C" __global__ void layernorm_bwd_kernel(const short*, const short*,
+                                                const short*, const float*,
+                                                const float*, short*, float*, float*,
+                                                long long, int);
+extern "C" __global__ void rmsnorm_fwd_kernel(const short*, const short*, short*,
+                                              float*, long long, int, float);
+extern "C" __global__ void rmsnorm_bwd_kernel(const short*, const short*,
+                                              const short*, const float*, short*,
+                                              float*, long long, int);
+extern "C" __global__ void colsum_ws_kernel(const float*, short*, int, int);
+extern "C" __global__ void gemm_nt_bf16_kernel(const short*, const short*, short*,
+                                               int, int, int, int, int, int);
+extern "C" __global__ void transpose_bf16_kernel(const short*, short*, int, int);
+
+#define CHECK_HIP(t, d)                                              \
+  TORCH_CHECK((t).is_cuda(), #t " must be a HIP tensor");            \
+  TORCH_CHECK((t).is_contiguous(), #t " must be contiguous");        \
+  TORCH_CHECK((t).scalar_type() == (d), #t " has wrong dtype")
+
+static inline hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+static inline int ew_grid(long long n_items) {
+  long long blocks = (n_items + 255) / 256;
+  return (int)std::min<long long>(blocks, 2048);
+}
+
+// ----------------------------------------------------------------- optim
+void fused_sgd(torch::Tensor param, torch::Tensor master, torch::Tensor grad,
+               torch::Tensor mom, double lr, double momentum,
+               double weight_decay, bool nesterov) {
+  CHECK_HIP(param, torch::kBFloat16);
+  CHECK_HIP(master, torch::kFloat32);
+  CHECK_HIP(grad, torch::kBFloat16);
+  CHECK_HIP(mom, torch::kFloat32);
+  long long n = param.numel();
+  hipLaunchKernelGGL(fused_sgd_kernel, dim3(ew_grid((n + 7) / 8)), dim3(256), 0,
+                     cur_stream(), (short*)param.data_ptr(),
+                     master.data_ptr<float>(), (const short*)grad.data_ptr(),
+                     mom.data_ptr<float>(), n, (float)lr, (float)momentum,
+                     (float)weight_decay, nesterov ? 1 : 0);
+}
+
+void fused_adamw(torch::Tensor param, torch::Tensor master, torch::Tensor grad,
+                 torch::Tensor exp_avg, torch::Tensor exp_avg_sq, double lr,
+                 double beta1, double beta2, double eps, double weight_decay,
+                 int64_t step) {
+  CHECK_HIP(param, torch::kBFloat16);
+  CHECK_HIP(master, torch::kFloat32);
+  CHECK_HIP(grad, torch::kBFloat16);
+  long long n = param.numel();
+  float bc1 = 1.0f - powf((float)beta1, (float)step);
+  float bc2 = 1.0f - powf((float)beta2, (float)step);
+  hipLaunchKernelGGL(fused_adamw_kernel, dim3(ew_grid((n + 7) / 8)), dim3(256), 0,
+                     cur_stream(), (short*)param.data_ptr(),
+                     master.data_ptr<float>(), (const short*)grad.data_ptr(),
+                     exp_avg.data_ptr<float>(), exp_avg_sq.data_ptr<float>(), n,
+                     (float)lr, (float)beta1, (float)beta2, (float)eps,
+                     (float)weight_decay, bc1, bc2);
+}
+
+// ---------------------------------------------------------------- secagg
+void secagg_quantize_mask(torch::Tensor q, torch::Tensor master, double weight,
+                          double scale, int64_t seed, int64_t rank,
+                          int64_t world_size) {
+  CHECK_HIP(q, torch::kInt64);
+  CHECK_HIP(master, torch::kFloat32);
+  long long n = master.numel();
+  hipLaunchKernelGGL(secagg_quantize_mask_kernel, dim3(ew_grid(n)), dim3(256), 0,
+                     cur_stream(), (long long*)q.data_ptr(),
+                     master.data_ptr<float>(), n, weight * scale,
+                     (unsigned long long)seed, (int)rank, (int)world_size);
+}
+
+// -------------------------------------------------------------------- dp
+void dp_scale_noise(torch::Tensor grad, double scale, double noise_std,
+                    int64_t seed) {
+  CHECK_HIP(grad, torch::kBFloat16);
+  long long n = grad.numel();
+  hipLaunchKernelGGL(dp_scale_noise_kernel, dim3(ew_grid((n + 1) / 2)), dim3(256),
+                     0, cur_stream(), (short*)grad.data_ptr(), n, (float)scale,
+                     (float)noise_std, (unsigned long long)seed);
+}
+
+torch::Tensor grad_norm2_partials(torch::Tensor grad) {
+  CHECK_HIP(grad, torch::kBFloat16);
+  long long n = grad.numel();
+  int blocks = ew_grid((n + 7) / 8);
+  auto out = torch::empty({blocks}, grad.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(norm2_bf16_kernel, dim3(blocks), dim3(256), 0, cur_stream(),
+                     (const short*)grad.data_ptr(), n, out.data_ptr<float>());
+  return out;
+}
+
+// ----------------------------------------------------------------- norms
+static inline std::pair<long long, int> rows_cols(const torch::Tensor& x) {
+  int H = (int)x.size(-1);
+  long long rows = x.numel() / H;
+  return {rows, H};
+}
+
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor b, double eps) {
+  CHECK_HIP(x, torch::kBFloat16);
+  auto [rows, H] = rows_cols(x);
+  TORCH_CHECK(H % 2 == 0 && H <= 16384, "H must be even and <= 16384");
+  auto y = torch::empty_like(x);
+  auto mean = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+  auto rstd = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+  int grid = (int)std::min<long long>(rows, 2048);
+  hipLaunchKernelGGL(layernorm_fwd_kernel, dim3(grid), dim3(256), 0, cur_stream(),
+                     (const short*)x.data_ptr(), (const short*)w.data_ptr(),
+                     (const short*)b.data_ptr(), (short*)y.data_ptr(),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(), rows, H,
+                     (float)eps);
+  return {y, mean, rstd};
+}
+
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor w, torch::Tensor mean,
+                                         torch::Tensor rstd) {
+  CHECK_HIP(dy, torch::kBFloat16);
+  auto [rows, H] = rows_cols(x);
+  auto dx = torch::empty_like(x);
+  int grid = (int)std::min<long long>(rows, 1024);
+  auto dw_ws = torch::empty({grid, H}, x.options().dtype(torch::kFloat32));
+  auto db_ws = torch::empty({grid, H}, x.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(layernorm_bwd_kernel, dim3(grid), dim3(256), 0, cur_stream(),
+                     (const short*)dy.data_ptr(), (const short*)x.data_ptr(),
+                     (const short*)w.data_ptr(), mean.data_ptr<float>(),
+                     rstd.data_ptr<float>(), (short*)dx.data_ptr(),
+                     dw_ws.data_ptr<float>(), db_ws.data_ptr<float>(), rows, H);
+  auto dw = torch::empty({H}, x.options());
+  auto db = torch::empty({H}, x.options());
+  int cgrid = (H + 255) / 256;
+  hipLaunchKernelGGL(colsum_ws_kernel, dim3(cgrid), dim3(256), 0, cur_stream(),
+                     dw_ws.data_ptr<float>(), (short*)dw.data_ptr(), grid, H);
+  hipLaunchKernelGGL(colsum_ws_kernel, dim3(cgrid), dim3(256), 0, cur_stream(),
+                     db_ws.data_ptr<float>(), (short*)db.data_ptr(), grid, H);
+  return {dx, dw, db};
+}
+
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
+                                       double eps) {
+  CHECK_HIP(x, torch::kBFloat16);
+  auto [rows, H] = rows_cols(x);
+  TORCH_CHECK(H % 2 == 0 && H <= 16384, "H must be even and <= 16384");
+  auto y = torch::empty_like(x);
+  auto rstd = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+  int grid = (int)std::min<long long>(rows, 2048);
+  hipLaunchKernelGGL(rmsnorm_fwd_kernel, dim3(grid), dim3(256), 0, cur_stream(),
+                     (const short*)x.data_ptr(), (const short*)w.data_ptr(),
+                     (short*)y.data_ptr(), rstd.data_ptr<float>(), rows, H,
+                     (float)eps);
+  return {y, rstd};
+}
+
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                       torch::Tensor w, torch::Tensor rstd) {
+  CHECK_HIP(dy, torch::kBFloat16);
+  auto [rows, H] = rows_cols(x);
+  auto dx = torch::empty_like(x);
+  int grid = (int)std::min<long long>(rows, 1024);
+  auto dw_ws = torch::empty({grid, H}, x.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(rmsnorm_bwd_kernel, dim3(grid), dim3(256), 0, cur_stream(),
+                     (const short*)dy.data_ptr(), (const short*)x.data_ptr(),
+                     (const short*)w.data_ptr(), rstd.data_ptr<float>(),
+                     (short*)dx.data_ptr(), dw_ws.data_ptr<float>(), rows, H);
+  auto dw = torch::empty({H}, x.options());
+  int cgrid = (H + 255) / 256;
+  hipLaunchKernelGGL(colsum_ws_kernel, dim3(cgrid), dim3(256), 0, cur_stream(),
+                     dw_ws.data_ptr<float>(), (short*)dw.data_ptr(), grid, H);
+  return {dx, dw};
+}
+
+// ------------------------------------------------------------------ gemm
+torch::Tensor transpose_bf16(torch::Tensor a) {
+  CHECK_HIP(a, torch::kBFloat16);
+  TORCH_CHECK(a.dim() == 2, "transpose_bf16 expects 2-D");
+  int M = (int)a.size(0), N = (int)a.size(1);
+  auto out = torch::empty({N, M}, a.options());
+  dim3 grid((N + 63) / 64, (M + 63) / 64);
+  hipLaunchKernelGGL(transpose_bf16_kernel, grid, dim3(256), 0, cur_stream(),
+                     (const short*)a.data_ptr(), (short*)out.data_ptr(), M, N);
+  return out;
+}
+
+// C[M,N] = op(A)·op(B); inputs 2-D bf16 contiguous.
+torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b, bool trans_a,
+                        bool trans_b) {
+  CHECK_HIP(a, torch::kBFloat16);
+  CHECK_HIP(b, torch::kBFloat16);
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2, "gemm_bf16 expects 2-D");
+  // normalize to NT: C[M,N] = A'[M,K] · B'[N,K]^T
+  torch::Tensor A = trans_a ? transpose_bf16(a) : a;
+  torch::Tensor B = trans_b ? b : transpose_bf16(b);
+  int M = (int)A.size(0), K = (int)A.size(1), N = (int)B.size(0);
+  TORCH_CHECK((int)B.size(1) == K, "inner dims mismatch");
+  auto c = torch::empty({M, N}, a.options());
+  dim3 grid((N + 127) / 128, (M + 127) / 128);  // BN=BM=128 (gemm.hip)
+  hipLaunchKernelGGL(gemm_nt_bf16_kernel, grid, dim3(256), 0, cur_stream(),
+                     (const short*)A.data_ptr(), (const short*)B.data_ptr(),
+                     (short*)c.data_ptr(), M, N, K, K, K, N);
+  return c;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fused_sgd", &fused_sgd, "fused SGD-momentum on flat buffers");
+  m.def("fused_adamw", &fused_adamw, "fused AdamW on flat buffers");
+  m.def("secagg_quantize_mask", &secagg_quantize_mask,
+        "fixed-point quantize + pairwise Philox masks");
+  m.def("dp_scale_noise", &dp_scale_noise, "DP clip-scale + gaussian noise");
+  m.def("grad_norm2_partials", &grad_norm2_partials, "block partial sum of g^2");
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("gemm_bf16", &gemm_bf16, "bf16 MFMA GEMM, fp32 accumulate");
+  m.def("transpose_bf16", &transpose_bf16);
+}

@@ -1,0 +1,179 @@
+#include "hip/hip_runtime.h"
+// bf16 MFMA GEMM for gfx950 — C[M,N] = A[M,K] · B[N,K]^T  ("NT": both
+// operands K-contiguous), fp32 accumulate, bf16 output.
+//
+// Structure (cdna_hip_programming.md §5 ladder, step-2/3 class):
+//   128×128 tile, BK=64, 4 waves (2×2), 64×64 output per wave as 4×4
+//   fragments of v_mfma_f32_16x16x32_bf16; LDS double-buffered with
+//   +16B-padded rows (row stride 72 bf16 = 144 B → the 16-lane groups of
+//   ds_read_b128 hit 16 distinct 16-B slots: conflict-free); register
+//   staging global→LDS with 16-B vector accesses.
+//
+// Fragment layout (verified against rocm CK gfx950 attributes):
+//   A/B: lane l holds 8 contiguous K elems at row/col (l&15), k-group
+//        (l>>4)*8.   C/D: col = lane&15, row = (lane>>4)*4 + reg.
+//
+// The other Linear-layer products (dgrad/wgrad) are composed NT +
+// transpose_bf16 (transpose cost ~3-8% of GEMM; a tr_b16-read variant can
+// remove it later).
+#include "common.h"
+
+#define BM 128
+#define BN 128
+#define BK 64
+#define LDS_STRIDE 72  // 64 + 8 bf16 pad
+#define NTH 256
+
+typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
+
+extern "C" __global__ void __launch_bounds__(NTH)
+gemm_nt_bf16_kernel(const short* __restrict__ A, const short* __restrict__ B,
+                    short* __restrict__ C, int M, int N, int K,
+                    int lda, int ldb, int ldc) {
+  __shared__ short lds[2 * 2 * BM * LDS_STRIDE];  // [buf][op][row][LDS_STRIDE]
+  auto As = [&](int buf) -> short* { return lds + buf * 2 * BM * LDS_STRIDE; };
+  auto Bs = [&](int buf) -> short* {
+    return lds + (buf * 2 + 1) * BM * LDS_STRIDE;
+  };
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = (wave >> 1) * 64;  // wave row offset in tile
+  const int wc = (wave & 1) * 64;   // wave col offset in tile
+
+  const int row0 = blockIdx.y * BM;
+  const int col0 = blockIdx.x * BN;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int nk = (K + BK - 1) / BK;
+
+  // ---- staging helper: load [128 rows][64 cols] K-slab of src into dst LDS
+  // chunk c in [0,1024): row = c>>3, col16 = (c&7)*8 ; thread t does
+  // c = t + 256*i (8 consecutive threads cover one 128-B row: coalesced).
+  auto stage = [&](const short* __restrict__ src, int ld, int base_row,
+                   int max_row, int k0, short* dst) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int c = tid + NTH * i;
+      int r = c >> 3;
+      int col = (c & 7) * 8;
+      int gr = base_row + r;
+      int gk = k0 + col;
+      s16x8 v;
+      if (gr < max_row && gk + 8 <= K) {
+        v = *(const s16x8*)(src + (long long)gr * ld + gk);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int kk = gk + j;
+          v[j] = (gr < max_row && kk < K) ? src[(long long)gr * ld + kk] : (short)0;
+        }
+      }
+      *(s16x8*)(dst + r * LDS_STRIDE + col) = v;
+    }
+  };
+
+  stage(A, lda, row0, M, 0, As(0));
+  stage(B, ldb, col0, N, 0, Bs(0));
+  __syncthreads();
+
+  for (int kt = 0; kt < nk; ++kt) {
+    int cur = kt & 1, nxt = cur ^ 1;
+    if (kt + 1 < nk) {  // prefetch next K-slab while computing this one
+      stage(A, lda, row0, M, (kt + 1) * BK, As(nxt));
+      stage(B, ldb, col0, N, (kt + 1) * BK, Bs(nxt));
+    }
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {  // two K=32 sub-steps per BK=64
+      const int koff = kk * 32 + (lane >> 4) * 8;
+      bf16x8_t a_frag[4], b_frag[4];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        a_frag[mi] = *(const bf16x8_t*)(As(cur) + (wr + mi * 16 + (lane & 15)) * LDS_STRIDE + koff);
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        b_frag[ni] = *(const bf16x8_t*)(Bs(cur) + (wc + ni * 16 + (lane & 15)) * LDS_STRIDE + koff);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: C[row][col] ; row = row0+wr+mi*16+(lane>>4)*4+r,
+  //                col = col0+wc+ni*16+(lane&15)
+  const int crow_base = row0 + wr + (lane >> 4) * 4;
+  const int ccol_base = col0 + wc + (lane & 15);
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = crow_base + mi * 16 + r;
+      if (row >= M) continue;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int col = ccol_base + ni * 16;
+        if (col < N) C[(long long)row * ldc + col] = f32_to_bf16(acc[mi][ni][r]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------- transpose
+// out[N,M] = in[M,N]^T, bf16. 64×64 LDS tiles, padded stride.
+#define TDIM 64
+#define TSTRIDE 72
+extern "C" __global__ void __launch_bounds__(256)
+transpose_bf16_kernel(const short* __restrict__ in, short* __restrict__ out,
+                      int M, int N) {
+  __shared__ short tile[TDIM * TSTRIDE];
+  const int tid = threadIdx.x;
+  const int tile_r = blockIdx.y * TDIM;  // row offset in `in`
+  const int tile_c = blockIdx.x * TDIM;  // col offset in `in`
+  // load: each thread 2 chunks of 8 (64*64/8/256 = 2)
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    int c = tid + 256 * i;          // chunk id in [0,512)
+    int r = c >> 3;                 // row in tile
+    int col = (c & 7) * 8;
+    int gr = tile_r + r, gc = tile_c + col;
+    s16x8 v;
+    if (gr < M && gc + 8 <= N) {
+      v = *(const s16x8*)(in + (long long)gr * N + gc);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        v[j] = (gr < M && gc + j < N) ? in[(long long)gr * N + gc + j] : (short)0;
+    }
+    *(s16x8*)(tile + r * TSTRIDE + col) = v;
+  }
+  __syncthreads();
+  // store: out row = in col. thread t writes chunks of out rows.
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    int c = tid + 256 * i;
+    int r = c >> 3;                 // row in OUT tile (= col in in)
+    int col = (c & 7) * 8;          // col in OUT tile (= row in in)
+    int gr = tile_c + r, gc = tile_r + col;
+    if (gr >= N) continue;
+    s16x8 v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) v[j] = tile[(col + j) * TSTRIDE + r];
+    if (gc + 8 <= M) {
+      *(s16x8*)(out + (long long)gr * M + gc) = v;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (gc + j < M) out[(long long)gr * M + gc + j] = v[j];
+    }
+  }
+}

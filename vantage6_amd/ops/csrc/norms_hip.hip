@@ -1,0 +1,253 @@
+#include "hip/hip_runtime.h"
+// LayerNorm / RMSNorm forward + backward (gfx950).
+//
+// One 256-thread workgroup per row-group (grid-stride over rows), bf16 I/O,
+// fp32 accumulation, 16-B vector accesses. A row (<= 16 KB bf16 at H=8192)
+// is L1-resident between the two passes of forward. Backward accumulates
+// per-block dweight/dbias partials into an fp32 workspace; a small second
+// kernel column-reduces the workspace (deterministic — no atomics).
+#include "common.h"
+
+#define NTHREADS 256
+// max row chunks a thread tracks in registers: H <= NTHREADS*8*MAXCH
+#define MAXCH 8
+
+DEVINL float block_reduce_sum(float v, float* red) {
+  red[threadIdx.x] = v;
+  __syncthreads();
+#pragma unroll
+  for (int s = NTHREADS / 2; s > 0; s >>= 1) {
+    if (threadIdx.x < (unsigned)s) red[threadIdx.x] += red[threadIdx.x + s];
+    __syncthreads();
+  }
+  float out = red[0];
+  __syncthreads();
+  return out;
+}
+
+// ---------------------------------------------------------------- LN fwd
+extern "C" __global__ void __launch_bounds__(NTHREADS)
+layernorm_fwd_kernel(const short* __restrict__ x, const short* __restrict__ w,
+                     const short* __restrict__ b, short* __restrict__ y,
+                     float* __restrict__ mean_out, float* __restrict__ rstd_out,
+                     long long nrows, int H, float eps) {
+  __shared__ float red[NTHREADS];
+  for (long long row = blockIdx.x; row < nrows; row += gridDim.x) {
+    const short* xr = x + row * (long long)H;
+    short* yr = y + row * (long long)H;
+    float s1 = 0.f, s2 = 0.f;
+    for (int i = threadIdx.x * 8; i < H; i += NTHREADS * 8) {
+      if (i + 8 <= H) {
+        s16x8 v = *(const s16x8*)(xr + i);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float f = bf16_to_f32(v[j]);
+          s1 += f; s2 = fmaf(f, f, s2);
+        }
+      } else {
+        for (int k = i; k < H; ++k) { float f = bf16_to_f32(xr[k]); s1 += f; s2 = fmaf(f, f, s2); }
+      }
+    }
+    s1 = block_reduce_sum(s1, red);
+    s2 = block_reduce_sum(s2, red);
+    float mean = s1 / H;
+    float var = s2 / H - mean * mean;
+    float rstd = rsqrtf(var + eps);
+    if (threadIdx.x == 0) { mean_out[row] = mean; rstd_out[row] = rstd; }
+    for (int i = threadIdx.x * 8; i < H; i += NTHREADS * 8) {
+      if (i + 8 <= H) {
+        s16x8 v = *(const s16x8*)(xr + i);
+        s16x8 wv = *(const s16x8*)(w + i);
+        s16x8 bv = *(const s16x8*)(b + i);
+        s16x8 o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float xh = (bf16_to_f32(v[j]) - mean) * rstd;
+          o[j] = f32_to_bf16(fmaf(xh, bf16_to_f32(wv[j]), bf16_to_f32(bv[j])));
+        }
+        *(s16x8*)(yr + i) = o;
+      } else {
+        for (int k = i; k < H; ++k) {
+          float xh = (bf16_to_f32(xr[k]) - mean) * rstd;
+          yr[k] = f32_to_bf16(fmaf(xh, bf16_to_f32(w[k]), bf16_to_f32(b[k])));
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------- LN bwd
+extern "C" __global__ void __launch_bounds__(NTHREADS)
+layernorm_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
+                     const short* __restrict__ w, const float* __restrict__ mean,
+                     const float* __restrict__ rstd, short* __restrict__ dx,
+                     float* __restrict__ dw_ws, float* __restrict__ db_ws,
+                     long long nrows, int H) {
+  __shared__ float red[NTHREADS];
+  float dw_acc[MAXCH * 8];
+  float db_acc[MAXCH * 8];
+  const int nch = (H + NTHREADS * 8 - 1) / (NTHREADS * 8);
+#pragma unroll
+  for (int c = 0; c < MAXCH * 8; ++c) { dw_acc[c] = 0.f; db_acc[c] = 0.f; }
+  for (long long row = blockIdx.x; row < nrows; row += gridDim.x) {
+    const short* xr = x + row * (long long)H;
+    const short* dyr = dy + row * (long long)H;
+    short* dxr = dx + row * (long long)H;
+    float mu = mean[row], rs = rstd[row];
+    float c1 = 0.f, c2 = 0.f;  // mean(wdy), mean(wdy*xhat)
+    for (int i = threadIdx.x * 8, c = 0; i < H; i += NTHREADS * 8, ++c) {
+      for (int j = 0; j < 8; ++j) {
+        int k = i + j;
+        if (k >= H) break;
+        float xh = (bf16_to_f32(xr[k]) - mu) * rs;
+        float d = bf16_to_f32(dyr[k]);
+        float wdy = bf16_to_f32(w[k]) * d;
+        c1 += wdy; c2 = fmaf(wdy, xh, c2);
+      }
+    }
+    c1 = block_reduce_sum(c1, red) / H;
+    c2 = block_reduce_sum(c2, red) / H;
+#pragma unroll MAXCH
+    for (int c = 0; c < MAXCH; ++c) {
+      int i = threadIdx.x * 8 + c * NTHREADS * 8;
+      if (i >= H) break;
+      if (i + 8 <= H) {
+        s16x8 xv = *(const s16x8*)(xr + i);
+        s16x8 dv = *(const s16x8*)(dyr + i);
+        s16x8 wv = *(const s16x8*)(w + i);
+        s16x8 o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float xh = (bf16_to_f32(xv[j]) - mu) * rs;
+          float d = bf16_to_f32(dv[j]);
+          float wdy = bf16_to_f32(wv[j]) * d;
+          o[j] = f32_to_bf16((wdy - c1 - xh * c2) * rs);
+          dw_acc[c * 8 + j] = fmaf(d, xh, dw_acc[c * 8 + j]);
+          db_acc[c * 8 + j] += d;
+        }
+        *(s16x8*)(dxr + i) = o;
+      } else {
+        for (int j = 0; i + j < H && j < 8; ++j) {
+          int k = i + j;
+          float xh = (bf16_to_f32(xr[k]) - mu) * rs;
+          float d = bf16_to_f32(dyr[k]);
+          float wdy = bf16_to_f32(w[k]) * d;
+          dxr[k] = f32_to_bf16((wdy - c1 - xh * c2) * rs);
+          dw_acc[c * 8 + j] = fmaf(d, xh, dw_acc[c * 8 + j]);
+          db_acc[c * 8 + j] += d;
+        }
+      }
+    }
+  }
+  // write this block's partials: workspace [gridDim.x][H]
+  float* dwr = dw_ws + (long long)blockIdx.x * H;
+  float* dbr = db_ws + (long long)blockIdx.x * H;
+#pragma unroll MAXCH
+  for (int c = 0; c < MAXCH; ++c) {
+    int i = threadIdx.x * 8 + c * NTHREADS * 8;
+    if (i >= H) break;
+    for (int j = 0; j < 8 && i + j < H; ++j) {
+      dwr[i + j] = dw_acc[c * 8 + j];
+      dbr[i + j] = db_acc[c * 8 + j];
+    }
+  }
+}
+
+// -------------------------------------------------------------- RMS fwd
+extern "C" __global__ void __launch_bounds__(NTHREADS)
+rmsnorm_fwd_kernel(const short* __restrict__ x, const short* __restrict__ w,
+                   short* __restrict__ y, float* __restrict__ rstd_out,
+                   long long nrows, int H, float eps) {
+  __shared__ float red[NTHREADS];
+  for (long long row = blockIdx.x; row < nrows; row += gridDim.x) {
+    const short* xr = x + row * (long long)H;
+    short* yr = y + row * (long long)H;
+    float s2 = 0.f;
+    for (int i = threadIdx.x * 8; i < H; i += NTHREADS * 8) {
+      if (i + 8 <= H) {
+        s16x8 v = *(const s16x8*)(xr + i);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) { float f = bf16_to_f32(v[j]); s2 = fmaf(f, f, s2); }
+      } else {
+        for (int k = i; k < H; ++k) { float f = bf16_to_f32(xr[k]); s2 = fmaf(f, f, s2); }
+      }
+    }
+    s2 = block_reduce_sum(s2, red);
+    float rstd = rsqrtf(s2 / H + eps);
+    if (threadIdx.x == 0) rstd_out[row] = rstd;
+    for (int i = threadIdx.x * 8; i < H; i += NTHREADS * 8) {
+      if (i + 8 <= H) {
+        s16x8 v = *(const s16x8*)(xr + i);
+        s16x8 wv = *(const s16x8*)(w + i);
+        s16x8 o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          o[j] = f32_to_bf16(bf16_to_f32(v[j]) * rstd * bf16_to_f32(wv[j]));
+        *(s16x8*)(yr + i) = o;
+      } else {
+        for (int k = i; k < H; ++k)
+          yr[k] = f32_to_bf16(bf16_to_f32(xr[k]) * rstd * bf16_to_f32(w[k]));
+      }
+    }
+  }
+}
+
+// -------------------------------------------------------------- RMS bwd
+extern "C" __global__ void __launch_bounds__(NTHREADS)
+rmsnorm_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
+                   const short* __restrict__ w, const float* __restrict__ rstd,
+                   short* __restrict__ dx, float* __restrict__ dw_ws,
+                   long long nrows, int H) {
+  __shared__ float red[NTHREADS];
+  float dw_acc[MAXCH * 8];
+#pragma unroll
+  for (int c = 0; c < MAXCH * 8; ++c) dw_acc[c] = 0.f;
+  for (long long row = blockIdx.x; row < nrows; row += gridDim.x) {
+    const short* xr = x + row * (long long)H;
+    const short* dyr = dy + row * (long long)H;
+    short* dxr = dx + row * (long long)H;
+    float rs = rstd[row];
+    float c2 = 0.f;  // mean(wdy * xhat)
+    for (int i = threadIdx.x * 8; i < H; i += NTHREADS * 8) {
+      for (int j = 0; j < 8; ++j) {
+        int k = i + j;
+        if (k >= H) break;
+        float xh = bf16_to_f32(xr[k]) * rs;
+        float wdy = bf16_to_f32(w[k]) * bf16_to_f32(dyr[k]);
+        c2 = fmaf(wdy, xh, c2);
+      }
+    }
+    c2 = block_reduce_sum(c2, red) / H;
+#pragma unroll MAXCH
+    for (int c = 0; c < MAXCH; ++c) {
+      int i = threadIdx.x * 8 + c * NTHREADS * 8;
+      if (i >= H) break;
+      for (int j = 0; j < 8 && i + j < H; ++j) {
+        int k = i + j;
+        float xh = bf16_to_f32(xr[k]) * rs;
+        float d = bf16_to_f32(dyr[k]);
+        float wdy = bf16_to_f32(w[k]) * d;
+        dxr[k] = f32_to_bf16((wdy - xh * c2) * rs);
+        dw_acc[c * 8 + j] = fmaf(d, xh, dw_acc[c * 8 + j]);
+      }
+    }
+  }
+  float* dwr = dw_ws + (long long)blockIdx.x * H;
+#pragma unroll MAXCH
+  for (int c = 0; c < MAXCH; ++c) {
+    int i = threadIdx.x * 8 + c * NTHREADS * 8;
+    if (i >= H) break;
+    for (int j = 0; j < 8 && i + j < H; ++j) dwr[i + j] = dw_acc[c * 8 + j];
+  }
+}
+
+// column-sum of an fp32 workspace [nblocks][H] -> bf16 [H]
+extern "C" __global__ void __launch_bounds__(NTHREADS)
+colsum_ws_kernel(const float* __restrict__ ws, short* __restrict__ out,
+                 int nblocks, int H) {
+  int h = blockIdx.x * NTHREADS + threadIdx.x;
+  if (h >= H) return;
+  float acc = 0.f;
+  for (int b = 0; b < nblocks; ++b) acc += ws[(long long)b * H + h];
+  out[h] = f32_to_bf16(acc);
+}
