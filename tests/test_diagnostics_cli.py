@@ -1,0 +1,27 @@
+"""Diagnostics workload + CLI tests (click CliRunner, reference test style)."""
+from click.testing import CliRunner
+
+from vantage6_amd.cli.main import cli
+
+
+def test_diagnostics_via_cli():
+    r = CliRunner().invoke(cli, ["test", "--stations", "2"])
+    assert r.exit_code == 0, r.output
+    assert '"ok": true' in r.output
+
+
+def test_new_config_template(tmp_path):
+    p = tmp_path / "net.yaml"
+    r = CliRunner().invoke(cli, ["new", "--config", str(p), "--stations", "3"])
+    assert r.exit_code == 0, r.output
+    from vantage6_amd.common.config import load_config
+
+    cfg = load_config(p)
+    assert len(cfg.stations) == 3
+    assert cfg.stations[0].databases[0].type == "synthetic"
+
+
+def test_demo_network_cli():
+    r = CliRunner().invoke(cli, ["dev", "demo", "--stations", "2"])
+    assert r.exit_code == 0, r.output
+    assert "column_mean" in r.output

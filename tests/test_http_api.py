@@ -1,0 +1,119 @@
+"""REST façade tests: the vantage6 client schema over real HTTP.
+
+Boots the orchestrator + stations in-process, serves the FastAPI façade on
+a uvicorn thread, and drives the full federated column-mean through
+``UserClient(url=...)`` — the same researcher code as in-process mode.
+"""
+import threading
+import time
+
+import numpy as np
+import pandas as pd
+import pytest
+import requests
+import uvicorn
+
+from vantage6_amd.cli.dev import DevNetwork
+from vantage6_amd.client import UserClient
+from vantage6_amd.common.config import DatabaseConfig
+from vantage6_amd.server.api import create_app
+
+PORT = 7655
+
+
+@pytest.fixture
+def net(tmp_path):
+    rng = np.random.default_rng(3)
+    paths = []
+    for i in range(2):
+        df = pd.DataFrame({"v": rng.normal(10, 2, 40)})
+        p = tmp_path / f"s{i}.csv"
+        df.to_csv(p, index=False)
+        paths.append(p)
+    network = DevNetwork(
+        n_stations=2,
+        databases=[[DatabaseConfig(label="default", uri=str(p), type="csv")]
+                   for p in paths],
+        name="httpnet",
+    )
+    network.start()
+    yield network
+    network.stop()
+
+
+@pytest.fixture
+def server(net):
+    app = create_app(net.core)
+    config = uvicorn.Config(app, host="127.0.0.1", port=PORT, log_level="error")
+    srv = uvicorn.Server(config)
+    th = threading.Thread(target=srv.run, daemon=True)
+    th.start()
+    for _ in range(100):
+        try:
+            requests.get(f"http://127.0.0.1:{PORT}/api/version", timeout=1)
+            break
+        except Exception:
+            time.sleep(0.05)
+    yield f"http://127.0.0.1:{PORT}"
+    srv.should_exit = True
+    th.join(timeout=5)
+
+
+def test_version_health(server):
+    v = requests.get(f"{server}/api/version").json()
+    assert "version" in v
+    h = requests.get(f"{server}/api/health").json()
+    assert h["status"] == "ok"
+
+
+def test_auth_required(server):
+    r = requests.get(f"{server}/api/task")
+    assert r.status_code == 401
+
+
+def test_bad_login(server):
+    r = requests.post(f"{server}/api/token/user",
+                      json={"username": "nobody", "password": "x"})
+    assert r.status_code == 401
+
+
+def test_e2e_column_mean_over_http(server, net):
+    client = UserClient(url=server)
+    client.authenticate("httpnet-researcher", "password")
+    task = client.task.create(
+        collaboration=net.collaboration["id"],
+        organizations=[net.org_ids[0]],
+        name="http column mean",
+        image="algorithms.column_mean",
+        input_={"method": "central_mean", "kwargs": {"columns": ["v"]}},
+    )
+    assert task["status"] == "pending"
+    results = client.wait_for_results(task["id"], interval=0.05, timeout=60)
+    assert results[0]["v"] == pytest.approx(10, abs=2)
+    # run listing + logs over HTTP
+    runs = client.run.from_task(task["id"])
+    assert runs and runs[0]["status"] == "completed"
+
+
+def test_token_refresh_roundtrip(server):
+    r = requests.post(f"{server}/api/token/user",
+                      json={"username": "root", "password": "root"})
+    tokens = r.json()
+    rr = requests.post(f"{server}/api/token/refresh",
+                       json={"refresh_token": tokens["refresh_token"]})
+    assert rr.status_code == 200 and "access_token" in rr.json()
+
+
+def test_node_token_and_open_runs(server, net):
+    # nodes authenticate with api keys over HTTP too
+    root = UserClient(url=server)
+    root.authenticate("root", "root")
+    node = root.node.create(name="extra-node",
+                            collaboration_id=net.collaboration["id"],
+                            organization_id=net.org_ids[0])
+    r = requests.post(f"{server}/api/token/node", json={"api_key": node["api_key"]})
+    assert r.status_code == 200
+    tok = r.json()["access_token"]
+    rr = requests.get(f"{server}/api/run", params={"state": "open"},
+                      headers={"Authorization": f"Bearer {tok}"})
+    assert rr.status_code == 200

@@ -1,0 +1,220 @@
+"""Server-core unit tests: permission matrix, task brokering, entity rules.
+
+Mirrors the reference's largest suite (in-memory DB + every
+resource exercised across permission scopes).
+"""
+import pytest
+
+from vantage6_amd.client import UserClient
+from vantage6_amd.common.task_status import RunStatus
+from vantage6_amd.server.core import BadRequest, NotFound, ServerCore
+from vantage6_amd.server.permission import PermissionError_
+
+
+@pytest.fixture
+def core():
+    c = ServerCore()
+    c.create_root_user("root", "root")
+    yield c
+    c.close()
+
+
+@pytest.fixture
+def root(core):
+    cl = UserClient(core=core)
+    cl.authenticate("root", "root")
+    return cl
+
+
+def make_collab(root, n_orgs=2, encrypted=False, name="c1"):
+    orgs = [root.organization.create(name=f"{name}-org{i}")["id"] for i in range(n_orgs)]
+    collab = root.collaboration.create(name=name, organization_ids=orgs,
+                                       encrypted=encrypted)
+    return orgs, collab
+
+
+class TestAuth:
+    def test_wrong_password(self, core):
+        with pytest.raises(PermissionError_):
+            core.authenticate_user("root", "wrong")
+
+    def test_node_auth_bad_key(self, core):
+        with pytest.raises(PermissionError_):
+            core.authenticate_node("not-a-key")
+
+    def test_refresh_requires_refresh_token(self, core, root):
+        with pytest.raises(PermissionError_):
+            core.refresh_token(root.gw.token)  # access token, not refresh
+
+
+class TestPermissions:
+    def test_researcher_cannot_create_organizations(self, core, root):
+        orgs, collab = make_collab(root)
+        root.user.create("res", "pw", organization_id=orgs[0], roles=["Researcher"])
+        res = UserClient(core=core)
+        res.authenticate("res", "pw")
+        with pytest.raises(PermissionError_):
+            res.organization.create(name="sneaky")
+
+    def test_viewer_cannot_create_tasks(self, core, root):
+        orgs, collab = make_collab(root)
+        root.user.create("viewer", "pw", organization_id=orgs[0], roles=["Viewer"])
+        v = UserClient(core=core)
+        v.authenticate("viewer", "pw")
+        with pytest.raises(PermissionError_):
+            v.task.create(collaboration=collab["id"], organizations=[orgs[0]],
+                          name="t", image="algorithms.column_mean",
+                          input_={"method": "x"})
+
+    def test_researcher_cannot_create_task_outside_own_collab(self, core, root):
+        orgs1, collab1 = make_collab(root, name="c1")
+        orgs2, collab2 = make_collab(root, name="c2")
+        root.user.create("res1", "pw", organization_id=orgs1[0], roles=["Researcher"])
+        res = UserClient(core=core)
+        res.authenticate("res1", "pw")
+        with pytest.raises(PermissionError_):
+            res.task.create(collaboration=collab2["id"], organizations=[orgs2[0]],
+                            name="t", image="algorithms.column_mean",
+                            input_={"method": "x"})
+
+    def test_org_admin_scoped_to_own_org(self, core, root):
+        orgs, collab = make_collab(root)
+        root.user.create("admin0", "pw", organization_id=orgs[0],
+                         roles=["Organization-Admin"])
+        adm = UserClient(core=core)
+        adm.authenticate("admin0", "pw")
+        # can create users in own org
+        adm.user.create("local", "pw", organization_id=orgs[0])
+        # cannot create users in the sibling org
+        with pytest.raises(PermissionError_):
+            adm.user.create("foreign", "pw", organization_id=orgs[1])
+
+
+class TestTasks:
+    def test_task_targets_must_be_members(self, core, root):
+        orgs, collab = make_collab(root)
+        outsider = root.organization.create(name="outsider")["id"]
+        with pytest.raises(BadRequest):
+            root.task.create(collaboration=collab["id"], organizations=[outsider],
+                             name="t", image="algorithms.column_mean",
+                             input_={"method": "x"})
+
+    def test_task_status_lifecycle_and_kill(self, core, root):
+        orgs, collab = make_collab(root)
+        root.node.create(name="n0", collaboration_id=collab["id"],
+                         organization_id=orgs[0])
+        task = root.task.create(collaboration=collab["id"], organizations=[orgs[0]],
+                                name="t", image="algorithms.column_mean",
+                                input_={"method": "partial_mean"})
+        assert task["status"] == "pending"
+        out = root.task.delete(task["id"])
+        assert out["status"] == "killed"
+        t = root.task.get(task["id"])
+        assert t["status"] == RunStatus.KILLED.value
+
+    def test_job_id_groups_task_trees(self, core, root):
+        orgs, collab = make_collab(root)
+        t1 = root.task.create(collaboration=collab["id"], organizations=[orgs[0]],
+                              name="a", image="x", input_={"method": "m"})
+        t2 = root.task.create(collaboration=collab["id"], organizations=[orgs[0]],
+                              name="b", image="x", input_={"method": "m"})
+        assert t1["job_id"] != t2["job_id"]
+
+    def test_study_restricts_targets(self, core, root):
+        orgs, collab = make_collab(root, n_orgs=3, name="s")
+        study = root.study.create(name="sub", collaboration_id=collab["id"],
+                                  organization_ids=orgs[:2])
+        with pytest.raises(BadRequest):
+            root.task.create(collaboration=collab["id"], organizations=[orgs[2]],
+                             name="t", image="x", input_={"method": "m"},
+                             study=study["id"])
+        t = root.task.create(collaboration=collab["id"], organizations=[orgs[0]],
+                             name="t", image="x", input_={"method": "m"},
+                             study=study["id"])
+        assert t["study_id"] == study["id"]
+
+
+class TestContainerIdentity:
+    def test_container_subtask_image_must_match(self, core, root):
+        orgs, collab = make_collab(root)
+        node = root.node.create(name="n0", collaboration_id=collab["id"],
+                                organization_id=orgs[0])
+        task = root.task.create(collaboration=collab["id"], organizations=[orgs[0]],
+                                name="t", image="algorithms.column_mean",
+                                input_={"method": "m"})
+        node_tok = core.authenticate_node(node["api_key"])
+        node_ident = core.identity_from_token(node_tok["access_token"])
+        ctok = core.token_for_container(node_ident, task["id"],
+                                        "algorithms.column_mean")
+        cident = core.identity_from_token(ctok["container_token"])
+        with pytest.raises(PermissionError_):
+            core.create_task(cident, image="algorithms.other",
+                             collaboration_id=collab["id"],
+                             organizations=[{"id": orgs[0], "input": ""}])
+
+    def test_container_cannot_create_after_parent_killed(self, core, root):
+        orgs, collab = make_collab(root)
+        node = root.node.create(name="n0", collaboration_id=collab["id"],
+                                organization_id=orgs[0])
+        task = root.task.create(collaboration=collab["id"], organizations=[orgs[0]],
+                                name="t", image="img", input_={"method": "m"})
+        node_tok = core.authenticate_node(node["api_key"])
+        node_ident = core.identity_from_token(node_tok["access_token"])
+        ctok = core.token_for_container(node_ident, task["id"], "img")
+        cident = core.identity_from_token(ctok["container_token"])
+        root.task.delete(task["id"])
+        with pytest.raises(PermissionError_):
+            core.create_task(cident, image="img", collaboration_id=collab["id"],
+                             organizations=[{"id": orgs[0], "input": ""}])
+
+    def test_node_cannot_patch_foreign_run(self, core, root):
+        orgs, collab = make_collab(root)
+        n0 = root.node.create(name="n0", collaboration_id=collab["id"],
+                              organization_id=orgs[0])
+        n1 = root.node.create(name="n1", collaboration_id=collab["id"],
+                              organization_id=orgs[1])
+        task = root.task.create(collaboration=collab["id"], organizations=[orgs[0]],
+                                name="t", image="img", input_={"method": "m"})
+        runs = core.list_runs(core.identity_from_token(
+            core.authenticate_node(n0["api_key"])["access_token"]),
+            task_id=task["id"])["data"]
+        foreign = core.identity_from_token(
+            core.authenticate_node(n1["api_key"])["access_token"])
+        with pytest.raises(PermissionError_):
+            core.patch_run(foreign, runs[0]["id"], status="active")
+
+
+class TestNodeLifecycle:
+    def test_node_online_offline_tracking(self, core, root):
+        orgs, collab = make_collab(root)
+        node = root.node.create(name="n0", collaboration_id=collab["id"],
+                                organization_id=orgs[0])
+        core.authenticate_node(node["api_key"])
+        nodes = root.node.list()["data"]
+        assert nodes[0]["status"] == "online"
+        core.node_set_offline(nodes[0]["id"])
+        assert root.node.list()["data"][0]["status"] == "offline"
+
+    def test_open_run_resync(self, core, root):
+        """Crash recovery: unfinished runs are returned to a re-auth'ing node."""
+        orgs, collab = make_collab(root)
+        node = root.node.create(name="n0", collaboration_id=collab["id"],
+                                organization_id=orgs[0])
+        root.task.create(collaboration=collab["id"], organizations=[orgs[0]],
+                         name="t", image="img", input_={"method": "m"})
+        ident = core.identity_from_token(
+            core.authenticate_node(node["api_key"])["access_token"])
+        open_runs = core.node_sync_open_runs(ident)
+        assert len(open_runs) == 1
+        assert open_runs[0]["status"] == "pending"
+
+
+class TestPagination:
+    def test_paginated_listing(self, core, root):
+        for i in range(25):
+            root.organization.create(name=f"porg{i}")
+        page1 = root.organization.list(page=1, per_page=10)
+        assert len(page1["data"]) == 10
+        assert page1["links"]["total"] == 26  # + root org
+        page3 = root.organization.list(page=3, per_page=10)
+        assert len(page3["data"]) == 6

@@ -1,0 +1,256 @@
+"""FastAPI façade — the vantage6 REST surface over the in-process core.
+
+Reference parity: SURVEY.md Appendix A.1/A.2 — ``POST /api/token/user``,
+``/api/token/node``, ``/api/token/refresh``, task CRUD + kill, runs,
+results, organizations, collaborations, nodes, users, studies, version,
+health; bearer-JWT auth; pagination via ``page``/``per_page``.
+
+The reference uses Flask + flask-restful + marshmallow; this image has
+FastAPI + pydantic, speaking the same JSON shapes (SURVEY.md §8 stack note).
+"""
+from __future__ import annotations
+
+from typing import Any, Optional
+
+from fastapi import Depends, FastAPI, HTTPException, Request
+from pydantic import BaseModel, Field
+
+from ..common import jwt as jwtlib
+from .core import BadRequest, Identity, NotFound, ServerCore
+from .permission import PermissionError_
+
+
+# ------------------------------------------------------------ request bodies
+class UserTokenBody(BaseModel):
+    username: str
+    password: str
+
+
+class NodeTokenBody(BaseModel):
+    api_key: str
+
+
+class RefreshBody(BaseModel):
+    refresh_token: str
+
+
+class ContainerTokenBody(BaseModel):
+    task_id: int
+    image: str
+
+
+class TaskOrg(BaseModel):
+    id: int
+    input: str = ""
+
+
+class TaskCreateBody(BaseModel):
+    name: str = ""
+    image: str
+    collaboration_id: int
+    description: str = ""
+    databases: list[dict] = Field(default_factory=list)
+    study_id: Optional[int] = None
+    organizations: list[TaskOrg]
+    method: str = ""
+
+
+class OrgCreateBody(BaseModel):
+    name: str
+    domain: str = ""
+    country: str = ""
+    public_key: str = ""
+
+
+class OrgPatchBody(BaseModel):
+    public_key: Optional[str] = None
+
+
+class CollabCreateBody(BaseModel):
+    name: str
+    organization_ids: list[int]
+    encrypted: bool = False
+
+
+class NodeCreateBody(BaseModel):
+    name: str
+    collaboration_id: int
+    organization_id: int
+    device: str = "cpu"
+
+
+class UserCreateBody(BaseModel):
+    username: str
+    password: str
+    organization_id: int
+    roles: list[str] = Field(default_factory=list)
+    firstname: str = ""
+    lastname: str = ""
+    email: str = ""
+
+
+class StudyCreateBody(BaseModel):
+    name: str
+    collaboration_id: int
+    organization_ids: list[int]
+
+
+class RunPatchBody(BaseModel):
+    status: Optional[str] = None
+    result: Optional[str] = None
+    log: Optional[str] = None
+
+
+def create_app(core: ServerCore) -> FastAPI:
+    app = FastAPI(title="vantage6_amd orchestrator", docs_url="/api/docs")
+
+    def ident(request: Request) -> Identity:
+        auth = request.headers.get("Authorization", "")
+        if not auth.startswith("Bearer "):
+            raise HTTPException(401, "missing bearer token")
+        try:
+            return core.identity_from_token(auth[len("Bearer ") :])
+        except jwtlib.JWTError as exc:
+            raise HTTPException(401, str(exc))
+
+    def wrap(fn, *args, **kwargs):
+        try:
+            return fn(*args, **kwargs)
+        except PermissionError_ as exc:
+            raise HTTPException(exc.status_code, str(exc))
+        except NotFound as exc:
+            raise HTTPException(404, str(exc))
+        except BadRequest as exc:
+            raise HTTPException(400, str(exc))
+        except jwtlib.JWTError as exc:
+            raise HTTPException(401, str(exc))
+
+    # ------------------------------------------------------------- tokens
+    @app.post("/api/token/user")
+    def token_user(body: UserTokenBody):
+        return wrap(core.authenticate_user, body.username, body.password)
+
+    @app.post("/api/token/node")
+    def token_node(body: NodeTokenBody):
+        return wrap(core.authenticate_node, body.api_key)
+
+    @app.post("/api/token/refresh")
+    def token_refresh(body: RefreshBody):
+        return wrap(core.refresh_token, body.refresh_token)
+
+    @app.post("/api/token/container")
+    def token_container(body: ContainerTokenBody, who: Identity = Depends(ident)):
+        return wrap(core.token_for_container, who, body.task_id, body.image)
+
+    # -------------------------------------------------------------- tasks
+    @app.post("/api/task", status_code=201)
+    def task_create(body: TaskCreateBody, who: Identity = Depends(ident)):
+        return wrap(
+            core.create_task, who, name=body.name, image=body.image,
+            collaboration_id=body.collaboration_id, description=body.description,
+            databases=body.databases, study_id=body.study_id,
+            organizations=[o.model_dump() for o in body.organizations],
+            method=body.method,
+        )
+
+    @app.get("/api/task/{task_id}")
+    def task_get(task_id: int, who: Identity = Depends(ident)):
+        return wrap(core.get_task, who, task_id)
+
+    @app.get("/api/task")
+    def task_list(who: Identity = Depends(ident), page: int = 1, per_page: int = 20,
+                  parent_id: Optional[int] = None, job_id: Optional[int] = None,
+                  collaboration_id: Optional[int] = None):
+        return wrap(core.list_tasks, who, page=page, per_page=per_page,
+                    parent_id=parent_id, job_id=job_id,
+                    collaboration_id=collaboration_id)
+
+    @app.delete("/api/task/{task_id}")
+    def task_kill(task_id: int, who: Identity = Depends(ident)):
+        return wrap(core.kill_task, who, task_id)
+
+    # --------------------------------------------------------------- runs
+    @app.get("/api/run")
+    def run_list(who: Identity = Depends(ident), task_id: Optional[int] = None,
+                 node_id: Optional[int] = None, state: Optional[str] = None,
+                 page: int = 1, per_page: int = 50):
+        return wrap(core.list_runs, who, task_id=task_id, node_id=node_id,
+                    state=state, page=page, per_page=per_page)
+
+    @app.get("/api/run/{run_id}")
+    def run_get(run_id: int, who: Identity = Depends(ident)):
+        return wrap(core.get_run, who, run_id)
+
+    @app.patch("/api/run/{run_id}")
+    def run_patch(run_id: int, body: RunPatchBody, who: Identity = Depends(ident)):
+        return wrap(core.patch_run, who, run_id,
+                    **{k: v for k, v in body.model_dump().items() if v is not None})
+
+    @app.get("/api/result")
+    def result_list(who: Identity = Depends(ident), task_id: int = 0,
+                    page: int = 1, per_page: int = 50):
+        return wrap(core.list_results, who, task_id=task_id, page=page,
+                    per_page=per_page)
+
+    # ------------------------------------------------------------ entities
+    @app.post("/api/organization", status_code=201)
+    def org_create(body: OrgCreateBody, who: Identity = Depends(ident)):
+        return wrap(core.create_organization, who, body.name, domain=body.domain,
+                    country=body.country, public_key=body.public_key)
+
+    @app.get("/api/organization")
+    def org_list(who: Identity = Depends(ident), page: int = 1, per_page: int = 20):
+        return wrap(core.list_organizations, who, page=page, per_page=per_page)
+
+    @app.get("/api/organization/{org_id}")
+    def org_get(org_id: int, who: Identity = Depends(ident)):
+        return wrap(core.get_organization, who, org_id)
+
+    @app.patch("/api/organization/{org_id}")
+    def org_patch(org_id: int, body: OrgPatchBody, who: Identity = Depends(ident)):
+        if body.public_key is None:
+            raise HTTPException(400, "nothing to patch")
+        return wrap(core.set_organization_public_key, who, org_id, body.public_key)
+
+    @app.post("/api/collaboration", status_code=201)
+    def collab_create(body: CollabCreateBody, who: Identity = Depends(ident)):
+        return wrap(core.create_collaboration, who, body.name,
+                    body.organization_ids, body.encrypted)
+
+    @app.get("/api/collaboration")
+    def collab_list(who: Identity = Depends(ident), page: int = 1,
+                    per_page: int = 20):
+        return wrap(core.list_collaborations, who, page=page, per_page=per_page)
+
+    @app.post("/api/node", status_code=201)
+    def node_create(body: NodeCreateBody, who: Identity = Depends(ident)):
+        return wrap(core.create_node, who, body.name, body.collaboration_id,
+                    body.organization_id, body.device)
+
+    @app.get("/api/node")
+    def node_list(who: Identity = Depends(ident), page: int = 1, per_page: int = 20,
+                  collaboration_id: Optional[int] = None):
+        return wrap(core.list_nodes, who, page=page, per_page=per_page,
+                    collaboration_id=collaboration_id)
+
+    @app.post("/api/user", status_code=201)
+    def user_create(body: UserCreateBody, who: Identity = Depends(ident)):
+        return wrap(core.create_user, who, body.username, body.password,
+                    body.organization_id, body.roles, firstname=body.firstname,
+                    lastname=body.lastname, email=body.email)
+
+    @app.post("/api/study", status_code=201)
+    def study_create(body: StudyCreateBody, who: Identity = Depends(ident)):
+        return wrap(core.create_study, who, body.name, body.collaboration_id,
+                    body.organization_ids)
+
+    # ----------------------------------------------------------------- ops
+    @app.get("/api/version")
+    def version():
+        return core.version()
+
+    @app.get("/api/health")
+    def health():
+        return core.health()
+
+    return app
