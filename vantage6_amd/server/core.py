@@ -101,6 +101,21 @@ class BadRequest(Exception):
     status_code = 400
 
 
+def _locked(fn):
+    """Serialize service methods: stations/clients run on many threads but
+    share one SQLite connection (StaticPool); transactions must not
+    interleave. RLock → re-entrant from within locked methods."""
+
+    import functools
+
+    @functools.wraps(fn)
+    def wrapper(self, *args, **kwargs):
+        with self._lock:
+            return fn(self, *args, **kwargs)
+
+    return wrapper
+
+
 class ServerCore:
     """The orchestrator. One instance per deployment (tests create many)."""
 
@@ -129,6 +144,7 @@ class ServerCore:
         self.db.close()
 
     # ------------------------------------------------------------- bootstrap
+    @_locked
     def create_root_user(self, username: str = "root", password: str = "root") -> dict:
         s = self.session
         if s.query(User).filter_by(username=username).first():
@@ -147,6 +163,7 @@ class ServerCore:
         return self._user_json(user)
 
     # ------------------------------------------------------------------ auth
+    @_locked
     def authenticate_user(self, username: str, password: str) -> dict:
         s = self.session
         user = s.query(User).filter_by(username=username).first()
@@ -169,6 +186,7 @@ class ServerCore:
             "refresh_url": "/api/token/refresh",
         }
 
+    @_locked
     def authenticate_node(self, api_key: str) -> dict:
         s = self.session
         for node in s.query(Node).all():
@@ -196,6 +214,7 @@ class ServerCore:
                 }
         raise PermissionError_("invalid api key", 401)
 
+    @_locked
     def token_for_container(self, node_identity: Identity, task_id: int, image: str) -> dict:
         """Node forges a scoped token for a running algorithm
         (``resource/token.py::ContainerToken``)."""
@@ -217,6 +236,7 @@ class ServerCore:
         }
         return {"container_token": jwtlib.encode(payload, self.jwt_secret, self.token_expires)}
 
+    @_locked
     def refresh_token(self, refresh_token: str) -> dict:
         payload = jwtlib.decode(refresh_token, self.jwt_secret)
         if not payload.get("refresh"):
@@ -277,6 +297,7 @@ class ServerCore:
         raise PermissionError_(f"{op.value} on {resource}: organization scope exceeded")
 
     # ------------------------------------------------------ entity management
+    @_locked
     def create_organization(self, ident: Identity, name: str, **kw) -> dict:
         self._check_user_can("organization", Operation.CREATE, ident)
         s = self.session
@@ -288,6 +309,7 @@ class ServerCore:
         s.commit()
         return self._org_json(org)
 
+    @_locked
     def set_organization_public_key(self, ident: Identity, org_id: int, public_key: str) -> dict:
         # uploading your OWN organization's public key is part of
         # setup_encryption and needs no org-edit permission (reference:
@@ -302,6 +324,7 @@ class ServerCore:
         s.commit()
         return self._org_json(org)
 
+    @_locked
     def create_collaboration(
         self, ident: Identity, name: str, organization_ids: list[int], encrypted: bool = False
     ) -> dict:
@@ -317,6 +340,7 @@ class ServerCore:
         s.commit()
         return self._collab_json(collab)
 
+    @_locked
     def create_study(self, ident: Identity, name: str, collaboration_id: int,
                      organization_ids: list[int]) -> dict:
         self._check_user_can("study", Operation.CREATE, ident, collab_id=collaboration_id)
@@ -335,6 +359,7 @@ class ServerCore:
                 "collaboration_id": collaboration_id,
                 "organizations": [{"id": o.id} for o in study.organizations]}
 
+    @_locked
     def create_node(
         self,
         ident: Identity,
@@ -360,6 +385,7 @@ class ServerCore:
         out["api_key"] = api_key  # only returned at creation, like the reference
         return out
 
+    @_locked
     def create_user(
         self,
         ident: Identity,
@@ -396,11 +422,13 @@ class ServerCore:
                       "pages": max(1, -(-len(items) // per_page))},
         }
 
+    @_locked
     def list_organizations(self, ident: Identity, page=1, per_page=20) -> dict:
         self._check_user_can_view_any("organization", ident)
         orgs = self.session.query(Organization).order_by(Organization.id).all()
         return self._paginate([self._org_json(o) for o in orgs], page, per_page)
 
+    @_locked
     def list_collaborations(self, ident: Identity, page=1, per_page=20) -> dict:
         self._check_user_can_view_any("collaboration", ident)
         cs = self.session.query(Collaboration).order_by(Collaboration.id).all()
@@ -409,6 +437,7 @@ class ServerCore:
             cs = [c for c in cs if any(o.id == ident.organization_id for o in c.organizations)]
         return self._paginate([self._collab_json(c) for c in cs], page, per_page)
 
+    @_locked
     def list_nodes(self, ident: Identity, page=1, per_page=20, collaboration_id=None) -> dict:
         self._check_user_can_view_any("node", ident)
         q = self.session.query(Node).order_by(Node.id)
@@ -423,6 +452,7 @@ class ServerCore:
         if rc.allowed_scope(Operation.VIEW) is None:
             raise PermissionError_(f"user lacks view permission on {resource}")
 
+    @_locked
     def get_organization(self, ident: Identity, org_id: int) -> dict:
         org = self.session.get(Organization, org_id)
         if org is None:
@@ -537,6 +567,7 @@ class ServerCore:
             )
         return task_json
 
+    @_locked
     def get_task(self, ident: Identity, task_id: int) -> dict:
         task = self.session.get(Task, task_id)
         if task is None:
@@ -546,6 +577,7 @@ class ServerCore:
                                  org_id=task.init_org_id, collab_id=task.collaboration_id)
         return self._task_json(task)
 
+    @_locked
     def list_tasks(self, ident: Identity, page=1, per_page=20, **filters) -> dict:
         self._check_user_can_view_any("task", ident)
         q = self.session.query(Task).order_by(Task.id)
@@ -557,6 +589,7 @@ class ServerCore:
             q = q.filter(Task.collaboration_id == filters["collaboration_id"])
         return self._paginate([self._task_json(t) for t in q.all()], page, per_page)
 
+    @_locked
     def kill_task(self, ident: Identity, task_id: int) -> dict:
         """DELETE /api/task/<id> — emit kill_containers to the nodes."""
         s = self.session
@@ -582,6 +615,7 @@ class ServerCore:
         return {"id": task.id, "status": "killed"}
 
     # ------------------------------------------------------------------- runs
+    @_locked
     def list_runs(self, ident: Identity, task_id: int | None = None,
                   node_id: int | None = None, state: str | None = None,
                   include_data: bool = True, page=1, per_page=50) -> dict:
@@ -597,12 +631,14 @@ class ServerCore:
             runs = [r for r in runs if r.node_id == ident.id]
         return self._paginate([self._run_json(r, include_data) for r in runs], page, per_page)
 
+    @_locked
     def get_run(self, ident: Identity, run_id: int) -> dict:
         run = self.session.get(Run, run_id)
         if run is None:
             raise NotFound(f"run {run_id} not found")
         return self._run_json(run, True)
 
+    @_locked
     def patch_run(self, ident: Identity, run_id: int, **fields) -> dict:
         """PATCH /api/run/<id> — node reports status/result/log."""
         s = self.session
@@ -641,6 +677,7 @@ class ServerCore:
             )
         return self._run_json(run, True)
 
+    @_locked
     def list_results(self, ident: Identity, task_id: int, page=1, per_page=50) -> dict:
         """GET /api/result?task_id= — result field only (v4 split result/run)."""
         runs = self.session.query(Run).filter(Run.task_id == task_id).order_by(Run.id).all()
@@ -658,6 +695,7 @@ class ServerCore:
         return self._paginate(items, page, per_page)
 
     # ------------------------------------------------------- node-side helpers
+    @_locked
     def node_sync_open_runs(self, ident: Identity) -> list[dict]:
         """Crash recovery: node re-fetches unfinished runs on boot
         (``Node.sync_task_queue_with_server``)."""
@@ -666,6 +704,7 @@ class ServerCore:
         runs = self.list_runs(ident, node_id=ident.id, state="open")["data"]
         return runs
 
+    @_locked
     def node_set_offline(self, node_id: int) -> None:
         s = self.session
         node = s.get(Node, node_id)
@@ -678,6 +717,7 @@ class ServerCore:
                 room=f"collaboration_{node.collaboration_id}",
             )
 
+    @_locked
     def organization_public_key(self, org_id: int) -> str:
         org = self.session.get(Organization, org_id)
         if org is None:
@@ -690,6 +730,7 @@ class ServerCore:
 
         return {"version": __version__}
 
+    @_locked
     def health(self) -> dict:
         try:
             self.session.query(Organization).first()
