@@ -63,6 +63,8 @@ def main() -> None:
     ap.add_argument("--seq-len", type=int, default=512)
     ap.add_argument("--shard-size", type=int, default=None, help="samples per station")
     ap.add_argument("--secure-agg", action="store_true")
+    ap.add_argument("--channels-last", action="store_true",
+                    help="NHWC layout for conv models (MIOpen fast path)")
     ap.add_argument("--dtype", default="bf16", choices=["bf16"])
     ap.add_argument("--device", default=None, help="override (e.g. cpu for smoke)")
     args = ap.parse_args()
@@ -91,6 +93,15 @@ def main() -> None:
     mk.pop("_seq_len", None)
 
     net, loss_fn = _build(args.model, mk, device)
+    if args.channels_last and args.model == "resnet18":
+        net = net.to(memory_format=torch.channels_last)
+        inner = loss_fn
+
+        def loss_fn(model, batch):  # noqa: F811
+            batch = dict(batch, x=batch["x"].contiguous(
+                memory_format=torch.channels_last))
+            return inner(model, batch)
+
     batches = make_batches(dataset, batch, device, seed=17 + comm.rank)
     cfg = FedAvgConfig(
         local_steps=args.local_steps,

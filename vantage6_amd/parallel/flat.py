@@ -31,12 +31,25 @@ class FlatParamSpace:
         self._specs: list[tuple[torch.nn.Parameter, int, int]] = []
         for p in params:
             n = p.numel()
-            self.param_flat[offset : offset + n].copy_(
-                p.detach().reshape(-1).to(device=self.device, dtype=dtype)
-            )
-            # re-point the parameter at the flat slice (no copy from here on)
-            p.data = self.param_flat[offset : offset + n].view(p.shape)
-            p.grad = self.grad_flat[offset : offset + n].view(p.shape)
+            # preserve channels_last (NHWC) layout for conv weights: the flat
+            # slice stores the tensor in its PHYSICAL order, the view+permute
+            # reconstructs the same strides backed by the flat buffer
+            nhwc = p.dim() == 4 and p.is_contiguous(memory_format=torch.channels_last) \
+                and not p.is_contiguous()
+            src = p.detach().to(device=self.device, dtype=dtype)
+            if nhwc:
+                phys = src.permute(0, 2, 3, 1).reshape(-1)  # physical NHWC order
+                self.param_flat[offset : offset + n].copy_(phys)
+                N, C, H, W = p.shape
+                p.data = (self.param_flat[offset : offset + n]
+                          .view(N, H, W, C).permute(0, 3, 1, 2))
+                p.grad = (self.grad_flat[offset : offset + n]
+                          .view(N, H, W, C).permute(0, 3, 1, 2))
+            else:
+                self.param_flat[offset : offset + n].copy_(src.reshape(-1))
+                # re-point the parameter at the flat slice (no copy from here on)
+                p.data = self.param_flat[offset : offset + n].view(p.shape)
+                p.grad = self.grad_flat[offset : offset + n].view(p.shape)
             self._specs.append((p, offset, n))
             offset += n
         self.master_flat = self.param_flat.to(torch.float32)
