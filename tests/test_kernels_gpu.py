@@ -202,3 +202,76 @@ def test_grad_norm2(ops):
     n2 = ops.grad_norm2_partials(x).sum().item()
     ref = x.float().pow(2).sum().item()
     assert n2 == pytest.approx(ref, rel=1e-3)
+
+
+# ----------------------------------------------------------------- fused BN
+class TestFusedBN:
+    def _ref(self, x, w, b, rm, rv, relu=False, res=None, training=True):
+        y = torch.nn.functional.batch_norm(
+            x.float(), rm, rv, w.float(), b.float(), training, 0.1, 1e-5)
+        if res is not None:
+            y = y + res.float()
+        if relu:
+            y = torch.relu(y)
+        return y
+
+    @pytest.mark.parametrize("c,relu,res", [(64, False, False), (128, True, False),
+                                            (256, True, True), (512, False, True)])
+    def test_bn_fwd_bwd(self, ops, c, relu, res):
+        from vantage6_amd.ops.batchnorm import FusedBatchNorm2d
+
+        torch.manual_seed(10 + c)
+        n, h, wdt = 4, 14, 14
+        x = bf16(torch.randn(n, c, h, wdt)).to(DEV)
+        x = x.contiguous(memory_format=torch.channels_last).requires_grad_(True)
+        residual = None
+        if res:
+            residual = bf16(torch.randn(n, c, h, wdt)).to(DEV)
+            residual = residual.contiguous(
+                memory_format=torch.channels_last).requires_grad_(True)
+        bn = FusedBatchNorm2d(c).to(DEV)
+        bn.weight.data = bf16(torch.randn(c)).to(DEV)
+        bn.bias.data = bf16(torch.randn(c)).to(DEV)
+        y = bn(x, residual=residual, relu=relu)
+        dy = bf16(torch.randn(n, c, h, wdt)).to(DEV).contiguous(
+            memory_format=torch.channels_last)
+        y.backward(dy)
+
+        # fp32 CPU reference
+        xf = x.detach().float().cpu().requires_grad_(True)
+        rf = None
+        if res:
+            rf = residual.detach().float().cpu().requires_grad_(True)
+        rm, rv = torch.zeros(c), torch.ones(c)
+        wf = bn.weight.detach().float().cpu().requires_grad_(True)
+        bff = bn.bias.detach().float().cpu().requires_grad_(True)
+        yref = self._ref(xf, wf, bff, rm, rv, relu, rf)
+        yref.backward(dy.float().cpu())
+
+        assert (y.float().cpu() - yref).abs().max() < 8e-2
+        assert (x.grad.float().cpu() - xf.grad).abs().max() < 8e-2
+        R = n * h * wdt
+        tol = 0.5 + 0.05 * R ** 0.5
+        assert (bn.weight.grad.float().cpu() - wf.grad).abs().max() < tol
+        assert (bn.bias.grad.float().cpu() - bff.grad).abs().max() < tol
+        if res:
+            assert (residual.grad.float().cpu() - rf.grad).abs().max() < 8e-2
+        # running stats updated like torch's
+        assert (bn.running_mean.cpu() - rm).abs().max() < 2e-2
+        assert (bn.running_var.cpu() - rv).abs().max() < 5e-2
+
+    def test_bn_eval_mode(self, ops):
+        from vantage6_amd.ops.batchnorm import FusedBatchNorm2d
+
+        c = 64
+        bn = FusedBatchNorm2d(c).to(DEV)
+        bn.running_mean.copy_(torch.randn(c).to(DEV))
+        bn.running_var.copy_(torch.rand(c).to(DEV) + 0.5)
+        bn.eval()
+        x = bf16(torch.randn(2, c, 8, 8)).to(DEV).contiguous(
+            memory_format=torch.channels_last)
+        y = bn(x)
+        yref = torch.nn.functional.batch_norm(
+            x.float(), bn.running_mean, bn.running_var,
+            bn.weight.float(), bn.bias.float(), False, 0.1, 1e-5)
+        assert (y.float() - yref).abs().max().item() < 5e-2

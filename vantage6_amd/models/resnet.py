@@ -12,6 +12,7 @@ import torch
 import torch.nn as nn
 
 from ..ops.gemm import Linear as MfmaLinear
+from ..ops.batchnorm import FusedBatchNorm2d
 
 
 class BatchNorm2dBF16(nn.Module):
@@ -27,12 +28,14 @@ class BatchNorm2dBF16(nn.Module):
         self.eps, self.momentum = eps, momentum
 
     def forward(self, x):
-        y = torch.nn.functional.batch_norm(
-            x.float(), self.running_mean, self.running_var,
+        # bf16 activations stay bf16 (native_batch_norm takes fp32 affine
+        # params with bf16 input — autocast's convention); only the tiny
+        # per-channel params are cast. Avoids 2 full fp32 copies of x.
+        return torch.nn.functional.batch_norm(
+            x, self.running_mean, self.running_var,
             self.weight.float(), self.bias.float(),
             self.training, self.momentum, self.eps,
         )
-        return y.to(x.dtype)
 
 
 def conv3x3(cin, cout, stride=1):
@@ -45,30 +48,29 @@ class BasicBlock(nn.Module):
     def __init__(self, cin, cout, stride=1):
         super().__init__()
         self.conv1 = conv3x3(cin, cout, stride)
-        self.bn1 = BatchNorm2dBF16(cout)
+        self.bn1 = FusedBatchNorm2d(cout)
         self.conv2 = conv3x3(cout, cout)
-        self.bn2 = BatchNorm2dBF16(cout)
-        self.relu = nn.ReLU(inplace=True)
-        self.down = None
+        self.bn2 = FusedBatchNorm2d(cout)
+        self.down_conv = None
         if stride != 1 or cin != cout:
-            self.down = nn.Sequential(
-                nn.Conv2d(cin, cout, 1, stride=stride, bias=False),
-                BatchNorm2dBF16(cout),
-            )
+            self.down_conv = nn.Conv2d(cin, cout, 1, stride=stride, bias=False)
+            self.down_bn = FusedBatchNorm2d(cout)
 
     def forward(self, x):
-        idt = x if self.down is None else self.down(x)
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.bn2(self.conv2(out))
-        return self.relu(out + idt)
+        if self.down_conv is None:
+            idt = x
+        else:
+            idt = self.down_bn(self.down_conv(x))
+        out = self.bn1(self.conv1(x), relu=True)
+        # bn2 + residual add + relu fused into one pass
+        return self.bn2(self.conv2(out), residual=idt, relu=True)
 
 
 class ResNet18(nn.Module):
     def __init__(self, num_classes: int = 1000, use_mfma_head: bool = True):
         super().__init__()
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = BatchNorm2dBF16(64)
-        self.relu = nn.ReLU(inplace=True)
+        self.bn1 = FusedBatchNorm2d(64)
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         layers = []
         cin = 64
@@ -84,7 +86,7 @@ class ResNet18(nn.Module):
             self.fc = nn.Linear(512, num_classes)
 
     def forward(self, x):
-        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.maxpool(self.bn1(self.conv1(x), relu=True))
         x = self.layers(x)
         x = self.avgpool(x).flatten(1)
         return self.fc(x)
@@ -99,7 +101,7 @@ def resnet18(num_classes: int = 1000, dtype=torch.bfloat16, device="cpu",
     m = m.to(device=device, dtype=dtype)
     # keep BN statistics buffers fp32 regardless of model dtype
     for mod in m.modules():
-        if isinstance(mod, BatchNorm2dBF16):
+        if isinstance(mod, (BatchNorm2dBF16, FusedBatchNorm2d)):
             mod.running_mean.data = mod.running_mean.data.float()
             mod.running_var.data = mod.running_var.data.float()
     return m

@@ -33,6 +33,27 @@ extern "C" __global__ void colsum_ws_kernel(const float*, short*, int, int);
 extern "C" __global__ void gemm_nt_bf16_kernel(const short*, const short*, short*,
                                                int, int, int, int, int, int);
 extern "C" __global__ void transpose_bf16_kernel(const short*, short*, int, int);
+extern "C" __global__ void bn_fwd_partials_kernel(const short*, float*, long long,
+                                                  int, int);
+extern "C" __global__ void bn_fwd_finalize_kernel(const float*, int, int, long long,
+                                                  const short*, const short*, float*,
+                                                  float*, float*, float*, float*,
+                                                  float*, float, float, int);
+extern "C" __global__ void bn_fwd_apply_kernel(const short*, short*, const short*,
+                                               const float*, const float*, long long,
+                                               int, int);
+extern "C" __global__ void bn_bwd_partials_kernel(const short*, const short*,
+                                                  const short*, const float*,
+                                                  const float*, float*, long long,
+                                                  int, int, int);
+extern "C" __global__ void bn_bwd_finalize_kernel(const float*, int, int, long long,
+                                                  const short*, const float*, short*,
+                                                  short*, float*, float*, float*);
+extern "C" __global__ void bn_bwd_apply_kernel(const short*, const short*,
+                                               const short*, const float*,
+                                               const float*, const float*,
+                                               const float*, const float*, short*,
+                                               short*, long long, int, int);
 
 #define CHECK_HIP(t, d)                                              \
   TORCH_CHECK((t).is_cuda(), #t " must be a HIP tensor");            \
@@ -196,6 +217,108 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
   return {dx, dw};
 }
 
+// -------------------------------------------------------------------- bn
+static inline int bn_rows_pb(int C) {
+  int chunks = C / 8;
+  int r = 256 / chunks;
+  return r < 1 ? 1 : r;
+}
+
+std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor w,
+                                  torch::Tensor b, torch::Tensor running_mean,
+                                  torch::Tensor running_var, bool training,
+                                  double momentum, double eps, bool relu,
+                                  c10::optional<torch::Tensor> residual) {
+  CHECK_HIP(x, torch::kBFloat16);
+  int C = (int)x.size(-1);
+  TORCH_CHECK(C % 8 == 0 && C <= 2048, "BN channels must be %8==0 and <=2048");
+  long long R = x.numel() / C;
+  auto fopts = x.options().dtype(torch::kFloat32);
+  auto mean = torch::empty({C}, fopts);
+  auto rstd = torch::empty({C}, fopts);
+  auto scale = torch::empty({C}, fopts);
+  auto shift = torch::empty({C}, fopts);
+  int rows_pb = bn_rows_pb(C);
+  int nblocks = (int)std::min<long long>((R + rows_pb - 1) / rows_pb, 1024);
+  torch::Tensor part;
+  if (training) {
+    part = torch::empty({2LL * nblocks, C}, fopts);
+    hipLaunchKernelGGL(bn_fwd_partials_kernel, dim3(nblocks), dim3(256), 0,
+                       cur_stream(), (const short*)x.data_ptr(),
+                       part.data_ptr<float>(), R, C, nblocks);
+  } else {
+    part = torch::empty({1}, fopts);
+  }
+  hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0,
+                     cur_stream(), part.data_ptr<float>(), nblocks, C, R,
+                     (const short*)w.data_ptr(), (const short*)b.data_ptr(),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     scale.data_ptr<float>(), shift.data_ptr<float>(),
+                     running_mean.data_ptr<float>(),
+                     running_var.data_ptr<float>(), (float)momentum, (float)eps,
+                     training ? 1 : 0);
+  auto y = torch::empty_like(x);
+  const short* res_ptr = nullptr;
+  if (residual.has_value()) {
+    CHECK_HIP(residual.value(), torch::kBFloat16);
+    res_ptr = (const short*)residual->data_ptr();
+  }
+  long long total_chunks = R * (C / 8);
+  int agrid = (int)std::min<long long>((total_chunks + 255) / 256, 4096);
+  hipLaunchKernelGGL(bn_fwd_apply_kernel, dim3(agrid), dim3(256), 0,
+                     cur_stream(), (const short*)x.data_ptr(),
+                     (short*)y.data_ptr(), res_ptr, scale.data_ptr<float>(),
+                     shift.data_ptr<float>(), R, C, relu ? 1 : 0);
+  return {y, mean, rstd};
+}
+
+std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
+                                  torch::Tensor x, torch::Tensor w,
+                                  torch::Tensor mean, torch::Tensor rstd,
+                                  bool relu, bool want_dres) {
+  CHECK_HIP(dy, torch::kBFloat16);
+  int C = (int)x.size(-1);
+  long long R = x.numel() / C;
+  auto fopts = x.options().dtype(torch::kFloat32);
+  int rows_pb = bn_rows_pb(C);
+  int nblocks = (int)std::min<long long>((R + rows_pb - 1) / rows_pb, 1024);
+  auto part = torch::empty({2LL * nblocks, C}, fopts);
+  hipLaunchKernelGGL(bn_bwd_partials_kernel, dim3(nblocks), dim3(256), 0,
+                     cur_stream(), (const short*)dy.data_ptr(),
+                     (const short*)y.data_ptr(), (const short*)x.data_ptr(),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     part.data_ptr<float>(), R, C, nblocks, relu ? 1 : 0);
+  auto dw = torch::empty({C}, x.options());
+  auto db = torch::empty({C}, x.options());
+  auto c_scale = torch::empty({C}, fopts);
+  auto c_db = torch::empty({C}, fopts);
+  auto c_dw = torch::empty({C}, fopts);
+  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0,
+                     cur_stream(), part.data_ptr<float>(), nblocks, C, R,
+                     (const short*)w.data_ptr(), rstd.data_ptr<float>(),
+                     (short*)dw.data_ptr(), (short*)db.data_ptr(),
+                     c_scale.data_ptr<float>(), c_db.data_ptr<float>(),
+                     c_dw.data_ptr<float>());
+  auto dx = torch::empty_like(x);
+  torch::Tensor dres;
+  short* dres_ptr = nullptr;
+  if (want_dres) {
+    dres = torch::empty_like(x);
+    dres_ptr = (short*)dres.data_ptr();
+  }
+  long long total_chunks = R * (C / 8);
+  int agrid = (int)std::min<long long>((total_chunks + 255) / 256, 4096);
+  hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3(agrid), dim3(256), 0,
+                     cur_stream(), (const short*)dy.data_ptr(),
+                     (const short*)y.data_ptr(), (const short*)x.data_ptr(),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     c_scale.data_ptr<float>(), c_db.data_ptr<float>(),
+                     c_dw.data_ptr<float>(), (short*)dx.data_ptr(), dres_ptr, R,
+                     C, relu ? 1 : 0);
+  if (want_dres) return {dx, dw, db, dres};
+  return {dx, dw, db};
+}
+
 // ------------------------------------------------------------------ gemm
 torch::Tensor transpose_bf16(torch::Tensor a) {
   CHECK_HIP(a, torch::kBFloat16);
@@ -238,6 +361,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("bn_fwd", &bn_fwd, "fused NHWC BatchNorm fwd (+relu/+residual)");
+  m.def("bn_bwd", &bn_bwd, "fused NHWC BatchNorm bwd");
   m.def("gemm_bf16", &gemm_bf16, "bf16 MFMA GEMM, fp32 accumulate");
   m.def("transpose_bf16", &transpose_bf16);
 }

@@ -1,0 +1,280 @@
+#include "hip/hip_runtime.h"
+// Fused BatchNorm2d (NHWC / channels_last, bf16 I/O, fp32 stats) for gfx950.
+//
+// The ResNet hot path spends more time in BatchNorm + eager elementwise
+// glue than in the (MFMA) convolutions themselves (profiles/
+// resnet18_bench_steady_r01.txt), so BN is fused end-to-end:
+//   forward  = stats partials → finalize (scale/shift precompute, running
+//              stats update) → one apply pass with optional residual-add
+//              and ReLU folded in (replaces 5 MIOpen kernels + 2 casts +
+//              add + relu eager kernels per block)
+//   backward = mask(dy by y>0) partials → finalize → one dx (+dresidual)
+//              pass
+// Layout: x is [R, C] with C contiguous (R = N*H*W), C % 8 == 0.
+#include "common.h"
+
+#define BN_NTH 256
+
+// thread → (row-group, 8-channel chunk) decomposition
+DEVINL void bn_thread_map(int C, int* chunks, int* rows_pb, int* j, int* rr) {
+  *chunks = C >> 3;
+  *rows_pb = BN_NTH / *chunks;
+  if (*rows_pb < 1) *rows_pb = 1;
+  *j = threadIdx.x % *chunks;
+  *rr = threadIdx.x / *chunks;
+}
+
+// ---------------------------------------------------------- fwd: partials
+// part layout: [2][nblocks][C] (sum, sumsq)
+extern "C" __global__ void __launch_bounds__(BN_NTH)
+bn_fwd_partials_kernel(const short* __restrict__ x, float* __restrict__ part,
+                       long long R, int C, int nblocks) {
+  __shared__ float red[BN_NTH * 8 * 2];
+  int chunks, rows_pb, j, rr;
+  bn_thread_map(C, &chunks, &rows_pb, &j, &rr);
+  float s1[8] = {0}, s2[8] = {0};
+  if (threadIdx.x < (unsigned)(chunks * rows_pb)) {
+    for (long long row = (long long)blockIdx.x * rows_pb + rr; row < R;
+         row += (long long)gridDim.x * rows_pb) {
+      s16x8 v = *(const s16x8*)(x + row * C + j * 8);
+#pragma unroll
+      for (int q = 0; q < 8; ++q) {
+        float f = bf16_to_f32(v[q]);
+        s1[q] += f;
+        s2[q] = fmaf(f, f, s2[q]);
+      }
+    }
+  }
+#pragma unroll
+  for (int q = 0; q < 8; ++q) {
+    red[threadIdx.x * 8 + q] = s1[q];
+    red[(BN_NTH + threadIdx.x) * 8 + q] = s2[q];
+  }
+  __syncthreads();
+  // tree-reduce over the row-group dimension (threads strided by `chunks`)
+  for (int s = rows_pb >> 1; s > 0; s >>= 1) {
+    if (rr < s && threadIdx.x + s * chunks < BN_NTH) {
+#pragma unroll
+      for (int q = 0; q < 8; ++q) {
+        red[threadIdx.x * 8 + q] += red[(threadIdx.x + s * chunks) * 8 + q];
+        red[(BN_NTH + threadIdx.x) * 8 + q] +=
+            red[(BN_NTH + threadIdx.x + s * chunks) * 8 + q];
+      }
+    }
+    __syncthreads();
+  }
+  if (rr == 0) {
+#pragma unroll
+    for (int q = 0; q < 8; ++q) {
+      part[(long long)blockIdx.x * C + j * 8 + q] = red[threadIdx.x * 8 + q];
+      part[((long long)nblocks + blockIdx.x) * C + j * 8 + q] =
+          red[(BN_NTH + threadIdx.x) * 8 + q];
+    }
+  }
+}
+
+// --------------------------------------------------------- fwd: finalize
+// writes mean/rstd (saved for bwd) + scale/shift (apply pass) and updates
+// running stats in place.
+extern "C" __global__ void __launch_bounds__(BN_NTH)
+bn_fwd_finalize_kernel(const float* __restrict__ part, int nblocks, int C,
+                       long long R, const short* __restrict__ w,
+                       const short* __restrict__ b, float* __restrict__ mean_out,
+                       float* __restrict__ rstd_out, float* __restrict__ scale,
+                       float* __restrict__ shift, float* __restrict__ running_mean,
+                       float* __restrict__ running_var, float momentum,
+                       float eps, int training) {
+  for (int c = blockIdx.x * BN_NTH + threadIdx.x; c < C;
+       c += gridDim.x * BN_NTH) {
+    float mean, var;
+    if (training) {
+      float s1 = 0.f, s2 = 0.f;
+      for (int bkk = 0; bkk < nblocks; ++bkk) {
+        s1 += part[(long long)bkk * C + c];
+        s2 += part[((long long)nblocks + bkk) * C + c];
+      }
+      mean = s1 / R;
+      var = s2 / R - mean * mean;
+      if (var < 0.f) var = 0.f;
+      float unbiased = (R > 1) ? var * (float)R / (float)(R - 1) : var;
+      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+      running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+    } else {
+      mean = running_mean[c];
+      var = running_var[c];
+    }
+    float rstd = rsqrtf(var + eps);
+    float wf = bf16_to_f32(w[c]);
+    mean_out[c] = mean;
+    rstd_out[c] = rstd;
+    scale[c] = wf * rstd;
+    shift[c] = bf16_to_f32(b[c]) - mean * wf * rstd;
+  }
+}
+
+// ------------------------------------------------------------ fwd: apply
+// y = relu?( x*scale + shift (+ residual) )
+extern "C" __global__ void __launch_bounds__(BN_NTH)
+bn_fwd_apply_kernel(const short* __restrict__ x, short* __restrict__ y,
+                    const short* __restrict__ residual,
+                    const float* __restrict__ scale,
+                    const float* __restrict__ shift, long long R, int C,
+                    int relu) {
+  int chunks = C >> 3;
+  long long total_chunks = R * chunks;
+  long long i0 = (long long)blockIdx.x * BN_NTH + threadIdx.x;
+  for (long long i = i0; i < total_chunks; i += (long long)gridDim.x * BN_NTH) {
+    int j = (int)(i % chunks);
+    long long off = (i / chunks) * C + j * 8;
+    s16x8 v = *(const s16x8*)(x + off);
+    s16x8 o;
+    f32x4 sc0 = *(const f32x4*)(scale + j * 8);
+    f32x4 sc1 = *(const f32x4*)(scale + j * 8 + 4);
+    f32x4 sh0 = *(const f32x4*)(shift + j * 8);
+    f32x4 sh1 = *(const f32x4*)(shift + j * 8 + 4);
+    if (residual) {
+      s16x8 rv = *(const s16x8*)(residual + off);
+#pragma unroll
+      for (int q = 0; q < 8; ++q) {
+        float sc = q < 4 ? sc0[q] : sc1[q - 4];
+        float sh = q < 4 ? sh0[q] : sh1[q - 4];
+        float f = fmaf(bf16_to_f32(v[q]), sc, sh) + bf16_to_f32(rv[q]);
+        if (relu && f < 0.f) f = 0.f;
+        o[q] = f32_to_bf16(f);
+      }
+    } else {
+#pragma unroll
+      for (int q = 0; q < 8; ++q) {
+        float sc = q < 4 ? sc0[q] : sc1[q - 4];
+        float sh = q < 4 ? sh0[q] : sh1[q - 4];
+        float f = fmaf(bf16_to_f32(v[q]), sc, sh);
+        if (relu && f < 0.f) f = 0.f;
+        o[q] = f32_to_bf16(f);
+      }
+    }
+    *(s16x8*)(y + off) = o;
+  }
+}
+
+// ---------------------------------------------------------- bwd: partials
+// dyr = relu ? (y>0 ? dy : 0) : dy ; partials of db=Σdyr, dws=Σdyr*xhat
+extern "C" __global__ void __launch_bounds__(BN_NTH)
+bn_bwd_partials_kernel(const short* __restrict__ dy, const short* __restrict__ y,
+                       const short* __restrict__ x,
+                       const float* __restrict__ mean,
+                       const float* __restrict__ rstd, float* __restrict__ part,
+                       long long R, int C, int nblocks, int relu) {
+  __shared__ float red[BN_NTH * 8 * 2];
+  int chunks, rows_pb, j, rr;
+  bn_thread_map(C, &chunks, &rows_pb, &j, &rr);
+  float sdb[8] = {0}, sdw[8] = {0};
+  float mu[8], rs[8];
+#pragma unroll
+  for (int q = 0; q < 8; ++q) {
+    mu[q] = mean[j * 8 + q];
+    rs[q] = rstd[j * 8 + q];
+  }
+  if (threadIdx.x < (unsigned)(chunks * rows_pb)) {
+    for (long long row = (long long)blockIdx.x * rows_pb + rr; row < R;
+         row += (long long)gridDim.x * rows_pb) {
+      long long off = row * C + j * 8;
+      s16x8 dv = *(const s16x8*)(dy + off);
+      s16x8 xv = *(const s16x8*)(x + off);
+#pragma unroll
+      for (int q = 0; q < 8; ++q) {
+        float d = bf16_to_f32(dv[q]);
+        if (relu) {
+          float yy = bf16_to_f32(((const short*)(y + off))[q]);
+          if (yy <= 0.f) d = 0.f;
+        }
+        float xh = (bf16_to_f32(xv[q]) - mu[q]) * rs[q];
+        sdb[q] += d;
+        sdw[q] = fmaf(d, xh, sdw[q]);
+      }
+    }
+  }
+#pragma unroll
+  for (int q = 0; q < 8; ++q) {
+    red[threadIdx.x * 8 + q] = sdb[q];
+    red[(BN_NTH + threadIdx.x) * 8 + q] = sdw[q];
+  }
+  __syncthreads();
+  for (int s = rows_pb >> 1; s > 0; s >>= 1) {
+    if (rr < s && threadIdx.x + s * chunks < BN_NTH) {
+#pragma unroll
+      for (int q = 0; q < 8; ++q) {
+        red[threadIdx.x * 8 + q] += red[(threadIdx.x + s * chunks) * 8 + q];
+        red[(BN_NTH + threadIdx.x) * 8 + q] +=
+            red[(BN_NTH + threadIdx.x + s * chunks) * 8 + q];
+      }
+    }
+    __syncthreads();
+  }
+  if (rr == 0) {
+#pragma unroll
+    for (int q = 0; q < 8; ++q) {
+      part[(long long)blockIdx.x * C + j * 8 + q] = red[threadIdx.x * 8 + q];
+      part[((long long)nblocks + blockIdx.x) * C + j * 8 + q] =
+          red[(BN_NTH + threadIdx.x) * 8 + q];
+    }
+  }
+}
+
+// --------------------------------------------------------- bwd: finalize
+// db/dws column sums → dw,db (bf16 out) + dx coefficients
+extern "C" __global__ void __launch_bounds__(BN_NTH)
+bn_bwd_finalize_kernel(const float* __restrict__ part, int nblocks, int C,
+                       long long R, const short* __restrict__ w,
+                       const float* __restrict__ rstd, short* __restrict__ dw,
+                       short* __restrict__ db, float* __restrict__ c_scale,
+                       float* __restrict__ c_db, float* __restrict__ c_dw) {
+  for (int c = blockIdx.x * BN_NTH + threadIdx.x; c < C;
+       c += gridDim.x * BN_NTH) {
+    float sdb = 0.f, sdw = 0.f;
+    for (int bkk = 0; bkk < nblocks; ++bkk) {
+      sdb += part[(long long)bkk * C + c];
+      sdw += part[((long long)nblocks + bkk) * C + c];
+    }
+    dw[c] = f32_to_bf16(sdw);
+    db[c] = f32_to_bf16(sdb);
+    c_scale[c] = bf16_to_f32(w[c]) * rstd[c];
+    c_db[c] = sdb / R;
+    c_dw[c] = sdw / R;
+  }
+}
+
+// ------------------------------------------------------------ bwd: apply
+// dx = c_scale*(dyr − c_db − xhat*c_dw); optional dresidual = dyr
+extern "C" __global__ void __launch_bounds__(BN_NTH)
+bn_bwd_apply_kernel(const short* __restrict__ dy, const short* __restrict__ y,
+                    const short* __restrict__ x, const float* __restrict__ mean,
+                    const float* __restrict__ rstd,
+                    const float* __restrict__ c_scale,
+                    const float* __restrict__ c_db,
+                    const float* __restrict__ c_dw, short* __restrict__ dx,
+                    short* __restrict__ dres, long long R, int C, int relu) {
+  int chunks = C >> 3;
+  long long total_chunks = R * chunks;
+  long long i0 = (long long)blockIdx.x * BN_NTH + threadIdx.x;
+  for (long long i = i0; i < total_chunks; i += (long long)gridDim.x * BN_NTH) {
+    int j = (int)(i % chunks);
+    long long off = (i / chunks) * C + j * 8;
+    s16x8 dv = *(const s16x8*)(dy + off);
+    s16x8 xv = *(const s16x8*)(x + off);
+    s16x8 odx, odr;
+#pragma unroll
+    for (int q = 0; q < 8; ++q) {
+      int c = j * 8 + q;
+      float d = bf16_to_f32(dv[q]);
+      if (relu) {
+        float yy = bf16_to_f32(((const short*)(y + off))[q]);
+        if (yy <= 0.f) d = 0.f;
+      }
+      float xh = (bf16_to_f32(xv[q]) - mean[c]) * rstd[c];
+      odx[q] = f32_to_bf16(c_scale[c] * (d - c_db[c] - xh * c_dw[c]));
+      odr[q] = f32_to_bf16(d);
+    }
+    *(s16x8*)(dx + off) = odx;
+    if (dres) *(s16x8*)(dres + off) = odr;
+  }
+}
