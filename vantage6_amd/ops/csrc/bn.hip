@@ -75,6 +75,9 @@ bn_fwd_partials_kernel(const short* __restrict__ x, float* __restrict__ part,
 // --------------------------------------------------------- fwd: finalize
 // writes mean/rstd (saved for bwd) + scale/shift (apply pass) and updates
 // running stats in place.
+// grid: cdiv(C,64) blocks × 256 threads; 4 reduction slices per channel
+// (one mostly-idle block serializing a 1024-deep loop cost 199 µs/call —
+// the dominant kernel of the whole ResNet bench before this layout)
 extern "C" __global__ void __launch_bounds__(BN_NTH)
 bn_fwd_finalize_kernel(const float* __restrict__ part, int nblocks, int C,
                        long long R, const short* __restrict__ w,
@@ -83,15 +86,30 @@ bn_fwd_finalize_kernel(const float* __restrict__ part, int nblocks, int C,
                        float* __restrict__ shift, float* __restrict__ running_mean,
                        float* __restrict__ running_var, float momentum,
                        float eps, int training) {
-  for (int c = blockIdx.x * BN_NTH + threadIdx.x; c < C;
-       c += gridDim.x * BN_NTH) {
+  __shared__ float red[BN_NTH * 2];
+  int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  int slice = threadIdx.x >> 6;  // 0..3
+  {
     float mean, var;
     if (training) {
       float s1 = 0.f, s2 = 0.f;
-      for (int bkk = 0; bkk < nblocks; ++bkk) {
-        s1 += part[(long long)bkk * C + c];
-        s2 += part[((long long)nblocks + bkk) * C + c];
+      if (c < C) {
+        for (int bkk = slice; bkk < nblocks; bkk += 4) {
+          s1 += part[(long long)bkk * C + c];
+          s2 += part[((long long)nblocks + bkk) * C + c];
+        }
       }
+      red[threadIdx.x] = s1;
+      red[BN_NTH + threadIdx.x] = s2;
+      __syncthreads();
+      if (slice == 0) {
+#pragma unroll
+        for (int s = 1; s < 4; ++s) {
+          s1 += red[threadIdx.x + s * 64];
+          s2 += red[BN_NTH + threadIdx.x + s * 64];
+        }
+      }
+      if (slice != 0 || c >= C) return;
       mean = s1 / R;
       var = s2 / R - mean * mean;
       if (var < 0.f) var = 0.f;
@@ -99,6 +117,7 @@ bn_fwd_finalize_kernel(const float* __restrict__ part, int nblocks, int C,
       running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
       running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
     } else {
+      if (slice != 0 || c >= C) return;
       mean = running_mean[c];
       var = running_var[c];
     }
@@ -227,19 +246,30 @@ bn_bwd_finalize_kernel(const float* __restrict__ part, int nblocks, int C,
                        const float* __restrict__ rstd, short* __restrict__ dw,
                        short* __restrict__ db, float* __restrict__ c_scale,
                        float* __restrict__ c_db, float* __restrict__ c_dw) {
-  for (int c = blockIdx.x * BN_NTH + threadIdx.x; c < C;
-       c += gridDim.x * BN_NTH) {
-    float sdb = 0.f, sdw = 0.f;
-    for (int bkk = 0; bkk < nblocks; ++bkk) {
+  __shared__ float red[BN_NTH * 2];
+  int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  int slice = threadIdx.x >> 6;
+  float sdb = 0.f, sdw = 0.f;
+  if (c < C) {
+    for (int bkk = slice; bkk < nblocks; bkk += 4) {
       sdb += part[(long long)bkk * C + c];
       sdw += part[((long long)nblocks + bkk) * C + c];
     }
-    dw[c] = f32_to_bf16(sdw);
-    db[c] = f32_to_bf16(sdb);
-    c_scale[c] = bf16_to_f32(w[c]) * rstd[c];
-    c_db[c] = sdb / R;
-    c_dw[c] = sdw / R;
   }
+  red[threadIdx.x] = sdb;
+  red[BN_NTH + threadIdx.x] = sdw;
+  __syncthreads();
+  if (slice != 0 || c >= C) return;
+#pragma unroll
+  for (int s = 1; s < 4; ++s) {
+    sdb += red[threadIdx.x + s * 64];
+    sdw += red[BN_NTH + threadIdx.x + s * 64];
+  }
+  dw[c] = f32_to_bf16(sdw);
+  db[c] = f32_to_bf16(sdb);
+  c_scale[c] = bf16_to_f32(w[c]) * rstd[c];
+  c_db[c] = sdb / R;
+  c_dw[c] = sdw / R;
 }
 
 // ------------------------------------------------------------ bwd: apply
