@@ -32,6 +32,8 @@ extern "C" __global__ void rmsnorm_bwd_kernel(const short*, const short*,
 extern "C" __global__ void colsum_ws_kernel(const float*, short*, int, int);
 extern "C" __global__ void gemm_nt_bf16_kernel(const short*, const short*, short*,
                                                int, int, int, int, int, int);
+extern "C" __global__ void gemm_nt_bf16_v2_kernel(const short*, const short*,
+                                                  short*, int, int, int);
 extern "C" __global__ void transpose_bf16_kernel(const short*, short*, int, int);
 extern "C" __global__ void bn_fwd_partials_kernel(const short*, float*, long long,
                                                   int, int);
@@ -265,8 +267,8 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor w,
     CHECK_HIP(residual.value(), torch::kBFloat16);
     res_ptr = (const short*)residual->data_ptr();
   }
-  long long total_chunks = R * (C / 8);
-  int agrid = (int)std::min<long long>((total_chunks + 255) / 256, 4096);
+  int agrid = (int)std::max<long long>(
+      1, std::min<long long>((R + rows_pb - 1) / rows_pb, 4096));
   hipLaunchKernelGGL(bn_fwd_apply_kernel, dim3(agrid), dim3(256), 0,
                      cur_stream(), (const short*)x.data_ptr(),
                      (short*)y.data_ptr(), res_ptr, scale.data_ptr<float>(),
@@ -309,8 +311,8 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
     dres = torch::empty_like(x);
     dres_ptr = (short*)dres.data_ptr();
   }
-  long long total_chunks = R * (C / 8);
-  int agrid = (int)std::min<long long>((total_chunks + 255) / 256, 4096);
+  int agrid = (int)std::max<long long>(
+      1, std::min<long long>((R + rows_pb - 1) / rows_pb, 4096));
   hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3(agrid), dim3(256), 0,
                      cur_stream(), (const short*)dy.data_ptr(),
                      (const short*)y.data_ptr(), (const short*)x.data_ptr(),
@@ -346,10 +348,17 @@ torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b, bool trans_a,
   int M = (int)A.size(0), K = (int)A.size(1), N = (int)B.size(0);
   TORCH_CHECK((int)B.size(1) == K, "inner dims mismatch");
   auto c = torch::empty({M, N}, a.options());
-  dim3 grid((N + 127) / 128, (M + 127) / 128);  // BN=BM=128 (gemm.hip)
-  hipLaunchKernelGGL(gemm_nt_bf16_kernel, grid, dim3(256), 0, cur_stream(),
-                     (const short*)A.data_ptr(), (const short*)B.data_ptr(),
-                     (short*)c.data_ptr(), M, N, K, K, K, N);
+  if (M % 128 == 0 && N % 128 == 0 && K % 64 == 0) {
+    int nwg = (M / 128) * (N / 128);
+    hipLaunchKernelGGL(gemm_nt_bf16_v2_kernel, dim3(nwg), dim3(256), 0,
+                       cur_stream(), (const short*)A.data_ptr(),
+                       (const short*)B.data_ptr(), (short*)c.data_ptr(), M, N, K);
+  } else {
+    dim3 grid((N + 127) / 128, (M + 127) / 128);  // BN=BM=128 (gemm.hip)
+    hipLaunchKernelGGL(gemm_nt_bf16_kernel, grid, dim3(256), 0, cur_stream(),
+                       (const short*)A.data_ptr(), (const short*)B.data_ptr(),
+                       (short*)c.data_ptr(), M, N, K, K, K, N);
+  }
   return c;
 }
 

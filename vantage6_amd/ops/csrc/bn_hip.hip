@@ -132,41 +132,40 @@ bn_fwd_finalize_kernel(const float* __restrict__ part, int nblocks, int C,
 }
 
 // ------------------------------------------------------------ fwd: apply
-// y = relu?( x*scale + shift (+ residual) )
+// y = relu?( x*scale + shift (+ residual) ).  Affine addressing: thread owns
+// one 8-channel chunk (params preloaded to registers) and strides rows.
 extern "C" __global__ void __launch_bounds__(BN_NTH)
 bn_fwd_apply_kernel(const short* __restrict__ x, short* __restrict__ y,
                     const short* __restrict__ residual,
                     const float* __restrict__ scale,
                     const float* __restrict__ shift, long long R, int C,
                     int relu) {
-  int chunks = C >> 3;
-  long long total_chunks = R * chunks;
-  long long i0 = (long long)blockIdx.x * BN_NTH + threadIdx.x;
-  for (long long i = i0; i < total_chunks; i += (long long)gridDim.x * BN_NTH) {
-    int j = (int)(i % chunks);
-    long long off = (i / chunks) * C + j * 8;
+  int chunks, rows_pb, j, rr;
+  bn_thread_map(C, &chunks, &rows_pb, &j, &rr);
+  if (threadIdx.x >= (unsigned)(chunks * rows_pb)) return;
+  float sc[8], sh[8];
+#pragma unroll
+  for (int q = 0; q < 8; ++q) {
+    sc[q] = scale[j * 8 + q];
+    sh[q] = shift[j * 8 + q];
+  }
+  for (long long row = (long long)blockIdx.x * rows_pb + rr; row < R;
+       row += (long long)gridDim.x * rows_pb) {
+    long long off = row * C + j * 8;
     s16x8 v = *(const s16x8*)(x + off);
     s16x8 o;
-    f32x4 sc0 = *(const f32x4*)(scale + j * 8);
-    f32x4 sc1 = *(const f32x4*)(scale + j * 8 + 4);
-    f32x4 sh0 = *(const f32x4*)(shift + j * 8);
-    f32x4 sh1 = *(const f32x4*)(shift + j * 8 + 4);
     if (residual) {
       s16x8 rv = *(const s16x8*)(residual + off);
 #pragma unroll
       for (int q = 0; q < 8; ++q) {
-        float sc = q < 4 ? sc0[q] : sc1[q - 4];
-        float sh = q < 4 ? sh0[q] : sh1[q - 4];
-        float f = fmaf(bf16_to_f32(v[q]), sc, sh) + bf16_to_f32(rv[q]);
+        float f = fmaf(bf16_to_f32(v[q]), sc[q], sh[q]) + bf16_to_f32(rv[q]);
         if (relu && f < 0.f) f = 0.f;
         o[q] = f32_to_bf16(f);
       }
     } else {
 #pragma unroll
       for (int q = 0; q < 8; ++q) {
-        float sc = q < 4 ? sc0[q] : sc1[q - 4];
-        float sh = q < 4 ? sh0[q] : sh1[q - 4];
-        float f = fmaf(bf16_to_f32(v[q]), sc, sh);
+        float f = fmaf(bf16_to_f32(v[q]), sc[q], sh[q]);
         if (relu && f < 0.f) f = 0.f;
         o[q] = f32_to_bf16(f);
       }
@@ -283,26 +282,40 @@ bn_bwd_apply_kernel(const short* __restrict__ dy, const short* __restrict__ y,
                     const float* __restrict__ c_db,
                     const float* __restrict__ c_dw, short* __restrict__ dx,
                     short* __restrict__ dres, long long R, int C, int relu) {
-  int chunks = C >> 3;
-  long long total_chunks = R * chunks;
-  long long i0 = (long long)blockIdx.x * BN_NTH + threadIdx.x;
-  for (long long i = i0; i < total_chunks; i += (long long)gridDim.x * BN_NTH) {
-    int j = (int)(i % chunks);
-    long long off = (i / chunks) * C + j * 8;
+  int chunks, rows_pb, j, rr;
+  bn_thread_map(C, &chunks, &rows_pb, &j, &rr);
+  if (threadIdx.x >= (unsigned)(chunks * rows_pb)) return;
+  float mu[8], rs[8], cs[8], cdb[8], cdw[8];
+#pragma unroll
+  for (int q = 0; q < 8; ++q) {
+    int c = j * 8 + q;
+    mu[q] = mean[c]; rs[q] = rstd[c];
+    cs[q] = c_scale[c]; cdb[q] = c_db[c]; cdw[q] = c_dw[c];
+  }
+  for (long long row = (long long)blockIdx.x * rows_pb + rr; row < R;
+       row += (long long)gridDim.x * rows_pb) {
+    long long off = row * C + j * 8;
     s16x8 dv = *(const s16x8*)(dy + off);
     s16x8 xv = *(const s16x8*)(x + off);
     s16x8 odx, odr;
+    if (relu) {
+      s16x8 yv = *(const s16x8*)(y + off);
 #pragma unroll
-    for (int q = 0; q < 8; ++q) {
-      int c = j * 8 + q;
-      float d = bf16_to_f32(dv[q]);
-      if (relu) {
-        float yy = bf16_to_f32(((const short*)(y + off))[q]);
-        if (yy <= 0.f) d = 0.f;
+      for (int q = 0; q < 8; ++q) {
+        float d = bf16_to_f32(dv[q]);
+        if (bf16_to_f32(yv[q]) <= 0.f) d = 0.f;
+        float xh = (bf16_to_f32(xv[q]) - mu[q]) * rs[q];
+        odx[q] = f32_to_bf16(cs[q] * (d - cdb[q] - xh * cdw[q]));
+        odr[q] = f32_to_bf16(d);
       }
-      float xh = (bf16_to_f32(xv[q]) - mean[c]) * rstd[c];
-      odx[q] = f32_to_bf16(c_scale[c] * (d - c_db[c] - xh * c_dw[c]));
-      odr[q] = f32_to_bf16(d);
+    } else {
+#pragma unroll
+      for (int q = 0; q < 8; ++q) {
+        float d = bf16_to_f32(dv[q]);
+        float xh = (bf16_to_f32(xv[q]) - mu[q]) * rs[q];
+        odx[q] = f32_to_bf16(cs[q] * (d - cdb[q] - xh * cdw[q]));
+        odr[q] = f32_to_bf16(d);
+      }
     }
     *(s16x8*)(dx + off) = odx;
     if (dres) *(s16x8*)(dres + off) = odr;

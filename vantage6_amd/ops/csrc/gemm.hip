@@ -176,3 +176,114 @@ transpose_bf16_kernel(const short* __restrict__ in, short* __restrict__ out,
     }
   }
 }
+
+// ===========================================================================
+// GEMM v2 — glds-staged, K-major LDS image (full-tile shapes only:
+// M%128==0, N%128==0, K%64==0; the stride-general v1 kernel handles the rest)
+//
+// Design (cdna_hip_programming.md §5 step-3 class + T1 swizzle):
+//  * LDS image per operand = [8 kslots][128 rows] of 16-B slots (a "kslot"
+//    is 8 contiguous K elements). global_load_lds writes lane-linear, so a
+//    wave stages 64 rows of one kslot per instruction — and the MFMA
+//    fragment read (lane l: row (l&15), kslot (l>>4)) then reads 16
+//    CONSECUTIVE slots per 16-lane group: bank-conflict-free by
+//    construction, no XOR swizzle needed.
+//  * double-buffered; one __shared__ object (compiler-waitcnt trap);
+//    __syncthreads() drains the in-flight glds of the next tile.
+//  * XCD-aware bijective blockIdx remap (8 XCDs with private L2s).
+// ===========================================================================
+#define V2_SLOTS_PER_OP 1024  // 8 kslots * 128 rows
+#define V2_BUF_SLOTS 2048     // A + B
+
+typedef __attribute__((address_space(3))) unsigned int lds_uint;
+typedef __attribute__((address_space(1))) const unsigned int global_uint;
+
+extern "C" __global__ void __launch_bounds__(NTH)
+gemm_nt_bf16_v2_kernel(const short* __restrict__ A, const short* __restrict__ B,
+                       short* __restrict__ C, int M, int N, int K) {
+  __shared__ short lds[2 * V2_BUF_SLOTS * 8];  // 64 KB
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = (wave >> 1) * 64;
+  const int wc = (wave & 1) * 64;
+
+  // bijective XCD remap: contiguous grid chunk per XCD
+  const int nwg = gridDim.x;
+  int wgid = blockIdx.x;
+  {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = wgid & 7, idx = wgid >> 3;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int ntile_n = N / BN;
+  const int row0 = (wgid / ntile_n) * BM;
+  const int col0 = (wgid % ntile_n) * BN;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int nk = K / BK;
+
+  // wave stages 4 (kslot,half) pieces of A and of B per K-step
+  auto stage = [&](int buf, int kt) {
+    const long long k0 = (long long)kt * BK;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int idx = wave * 4 + i;
+      int kslot = idx >> 1, half = idx & 1;
+      const short* srcA = A + (long long)(row0 + half * 64 + lane) * K + k0 + kslot * 8;
+      int dstA = (buf * V2_BUF_SLOTS + kslot * 128 + half * 64) * 16;  // bytes
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcA, (lds_uint*)((char*)lds + dstA), 16, 0, 0);
+      const short* srcB = B + (long long)(col0 + half * 64 + lane) * K + k0 + kslot * 8;
+      int dstB = (buf * V2_BUF_SLOTS + V2_SLOTS_PER_OP + kslot * 128 + half * 64) * 16;
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcB, (lds_uint*)((char*)lds + dstB), 16, 0, 0);
+    }
+  };
+
+  stage(0, 0);
+  __syncthreads();  // vmcnt(0) drain + barrier (glds in flight)
+
+  for (int kt = 0; kt < nk; ++kt) {
+    int cur = kt & 1, nxt = cur ^ 1;
+    if (kt + 1 < nk) stage(nxt, kt + 1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      const int ks = kk * 4 + (lane >> 4);
+      bf16x8_t a_frag[4], b_frag[4];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        a_frag[mi] = *(const bf16x8_t*)(
+            lds + (cur * V2_BUF_SLOTS + ks * 128 + wr + mi * 16 + (lane & 15)) * 8);
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        b_frag[ni] = *(const bf16x8_t*)(
+            lds + (cur * V2_BUF_SLOTS + V2_SLOTS_PER_OP + ks * 128 + wc + ni * 16 +
+                   (lane & 15)) * 8);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();  // drains next tile's glds; releases cur for overwrite
+  }
+
+  const int crow_base = row0 + wr + (lane >> 4) * 4;
+  const int ccol_base = col0 + wc + (lane & 15);
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = crow_base + mi * 16 + r;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        C[(long long)row * N + ccol_base + ni * 16] = f32_to_bf16(acc[mi][ni][r]);
+    }
+}
