@@ -34,6 +34,12 @@ extern "C" __global__ void gemm_nt_bf16_kernel(const short*, const short*, short
                                                int, int, int, int, int, int);
 extern "C" __global__ void gemm_nt_bf16_v2_kernel(const short*, const short*,
                                                   short*, int, int, int);
+extern "C" __global__ void gemm_nt_bf16_v3_kernel(const short*, const short*,
+                                                  short*, int, int, int);
+extern "C" __global__ void gemm_nt_bf16_v4_kernel(const short*, const short*,
+                                                  short*, int, int, int);
+extern "C" __global__ void gemm_nt_bf16_v5_kernel(const short*, const short*,
+                                                  short*, int, int, int);
 extern "C" __global__ void transpose_bf16_kernel(const short*, short*, int, int);
 extern "C" __global__ void bn_fwd_partials_kernel(const short*, float*, long long,
                                                   int, int);
@@ -350,9 +356,17 @@ torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b, bool trans_a,
   auto c = torch::empty({M, N}, a.options());
   if (M % 128 == 0 && N % 128 == 0 && K % 64 == 0) {
     int nwg = (M / 128) * (N / 128);
-    hipLaunchKernelGGL(gemm_nt_bf16_v2_kernel, dim3(nwg), dim3(256), 0,
-                       cur_stream(), (const short*)A.data_ptr(),
-                       (const short*)B.data_ptr(), (short*)c.data_ptr(), M, N, K);
+    static int variant = [] {
+      const char* v = getenv("V6AMD_GEMM_KERNEL");
+      return v ? atoi(v) : 5;  // v5 (dbuf glds + src swizzle) default
+    }();
+    auto kfn = gemm_nt_bf16_v4_kernel;
+    if (variant == 2) kfn = gemm_nt_bf16_v2_kernel;
+    if (variant == 3) kfn = gemm_nt_bf16_v3_kernel;
+    if (variant == 5) kfn = gemm_nt_bf16_v5_kernel;
+    hipLaunchKernelGGL(kfn, dim3(nwg), dim3(256), 0, cur_stream(),
+                       (const short*)A.data_ptr(), (const short*)B.data_ptr(),
+                       (short*)c.data_ptr(), M, N, K);
   } else {
     dim3 grid((N + 127) / 128, (M + 127) / 128);  // BN=BM=128 (gemm.hip)
     hipLaunchKernelGGL(gemm_nt_bf16_kernel, grid, dim3(256), 0, cur_stream(),

@@ -288,3 +288,272 @@ gemm_nt_bf16_v2_kernel(const short* __restrict__ A, const short* __restrict__ B,
         C[(long long)row * N + ccol_base + ni * 16] = f32_to_bf16(acc[mi][ni][r]);
     }
 }
+
+
+// v3 — single-buffer, two barriers per K-step: 32 KB LDS → ~5 blocks/CU;
+// inter-block overlap hides the glds drain (the ladder's step-3 shape).
+extern "C" __global__ void __launch_bounds__(NTH)
+gemm_nt_bf16_v3_kernel(const short* __restrict__ A, const short* __restrict__ B,
+                       short* __restrict__ C, int M, int N, int K) {
+  __shared__ short lds[V2_BUF_SLOTS * 8];  // 32 KB
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = (wave >> 1) * 64;
+  const int wc = (wave & 1) * 64;
+  const int nwg = gridDim.x;
+  int wgid = blockIdx.x;
+  {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = wgid & 7, idx = wgid >> 3;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int ntile_n = N / BN;
+  const int row0 = (wgid / ntile_n) * BM;
+  const int col0 = (wgid % ntile_n) * BN;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int nk = K / BK;
+  for (int kt = 0; kt < nk; ++kt) {
+    const long long k0 = (long long)kt * BK;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int idx = wave * 4 + i;
+      int kslot = idx >> 1, half = idx & 1;
+      const short* srcA = A + (long long)(row0 + half * 64 + lane) * K + k0 + kslot * 8;
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcA,
+          (lds_uint*)((char*)lds + (kslot * 128 + half * 64) * 16), 16, 0, 0);
+      const short* srcB = B + (long long)(col0 + half * 64 + lane) * K + k0 + kslot * 8;
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcB,
+          (lds_uint*)((char*)lds + (V2_SLOTS_PER_OP + kslot * 128 + half * 64) * 16),
+          16, 0, 0);
+    }
+    __syncthreads();  // drain glds + barrier
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      const int ks = kk * 4 + (lane >> 4);
+      bf16x8_t a_frag[4], b_frag[4];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        a_frag[mi] = *(const bf16x8_t*)(
+            lds + (ks * 128 + wr + mi * 16 + (lane & 15)) * 8);
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        b_frag[ni] = *(const bf16x8_t*)(
+            lds + (V2_SLOTS_PER_OP + ks * 128 + wc + ni * 16 + (lane & 15)) * 8);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();  // everyone done reading before next overwrite
+  }
+  const int crow_base = row0 + wr + (lane >> 4) * 4;
+  const int ccol_base = col0 + wc + (lane & 15);
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = crow_base + mi * 16 + r;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        C[(long long)row * N + ccol_base + ni * 16] = f32_to_bf16(acc[mi][ni][r]);
+    }
+}
+
+// v4 — full-cacheline glds + source-XOR swizzle (rule 21).
+// Each glds instruction covers 8 consecutive rows × the full 128-B K-slab
+// (coalesced: 8 whole lines per instruction), with the kslot index of each
+// lane's 16-B piece XOR-permuted by (row&7) on the SOURCE address; the
+// fragment read applies the same XOR. Residual 2-way conflict (rows r,r+8)
+// is below the MFMA-issue bound. Single LDS buffer (32 KB → 3 waves/SIMD).
+extern "C" __global__ void __launch_bounds__(NTH)
+gemm_nt_bf16_v4_kernel(const short* __restrict__ A, const short* __restrict__ B,
+                       short* __restrict__ C, int M, int N, int K) {
+  __shared__ short lds[V2_BUF_SLOTS * 8];  // 32 KB: A tile + B tile
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = (wave >> 1) * 64;
+  const int wc = (wave & 1) * 64;
+  const int nwg = gridDim.x;
+  int wgid = blockIdx.x;
+  {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = wgid & 7, idx = wgid >> 3;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int ntile_n = N / BN;
+  const int row0 = (wgid / ntile_n) * BM;
+  const int col0 = (wgid % ntile_n) * BN;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // per-lane source column offset (elements): XOR-permuted kslot within row
+  const int src_koff = (((lane & 7) ^ (lane >> 3)) & 7) * 8;
+  const int row_in_piece = lane >> 3;
+
+  const int nk = K / BK;
+  for (int kt = 0; kt < nk; ++kt) {
+    const long long k0 = (long long)kt * BK;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int pa = wave * 4 + i;  // A piece: rows pa*8 .. pa*8+7
+      const short* srcA =
+          A + (long long)(row0 + pa * 8 + row_in_piece) * K + k0 + src_koff;
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcA, (lds_uint*)((char*)lds + pa * 1024), 16, 0, 0);
+      const short* srcB =
+          B + (long long)(col0 + pa * 8 + row_in_piece) * K + k0 + src_koff;
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcB, (lds_uint*)((char*)lds + 16384 + pa * 1024), 16, 0,
+          0);
+    }
+    __syncthreads();  // drain glds + barrier
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      const int ks = kk * 4 + (lane >> 4);
+      bf16x8_t a_frag[4], b_frag[4];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        int row = wr + mi * 16 + (lane & 15);
+        a_frag[mi] = *(const bf16x8_t*)((char*)lds + row * 128 +
+                                        ((ks ^ (row & 7)) & 7) * 16);
+      }
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int row = wc + ni * 16 + (lane & 15);
+        b_frag[ni] = *(const bf16x8_t*)((char*)lds + 16384 + row * 128 +
+                                        ((ks ^ (row & 7)) & 7) * 16);
+      }
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+  const int crow_base = row0 + wr + (lane >> 4) * 4;
+  const int ccol_base = col0 + wc + (lane & 15);
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = crow_base + mi * 16 + r;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        C[(long long)row * N + ccol_base + ni * 16] = f32_to_bf16(acc[mi][ni][r]);
+    }
+}
+
+// v5 — v4 + double-buffered glds (overlap staging of tile k+1 with compute).
+// Each glds instruction covers 8 consecutive rows × the full 128-B K-slab
+// (coalesced: 8 whole lines per instruction), with the kslot index of each
+// lane's 16-B piece XOR-permuted by (row&7) on the SOURCE address; the
+// fragment read applies the same XOR. Residual 2-way conflict (rows r,r+8)
+// is below the MFMA-issue bound. Single LDS buffer (32 KB → 3 waves/SIMD).
+extern "C" __global__ void __launch_bounds__(NTH)
+gemm_nt_bf16_v5_kernel(const short* __restrict__ A, const short* __restrict__ B,
+                       short* __restrict__ C, int M, int N, int K) {
+  __shared__ short lds[2 * V2_BUF_SLOTS * 8];  // 64 KB double-buffered
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = (wave >> 1) * 64;
+  const int wc = (wave & 1) * 64;
+  const int nwg = gridDim.x;
+  int wgid = blockIdx.x;
+  {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = wgid & 7, idx = wgid >> 3;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int ntile_n = N / BN;
+  const int row0 = (wgid / ntile_n) * BM;
+  const int col0 = (wgid % ntile_n) * BN;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // per-lane source column offset (elements): XOR-permuted kslot within row
+  const int src_koff = (((lane & 7) ^ (lane >> 3)) & 7) * 8;
+  const int row_in_piece = lane >> 3;
+
+  const int nk = K / BK;
+  auto stage = [&](int buf, int kt) {
+    const long long k0 = (long long)kt * BK;
+    char* base = (char*)lds + buf * 32768;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int pa = wave * 4 + i;
+      const short* srcA =
+          A + (long long)(row0 + pa * 8 + row_in_piece) * K + k0 + src_koff;
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcA, (lds_uint*)(base + pa * 1024), 16, 0, 0);
+      const short* srcB =
+          B + (long long)(col0 + pa * 8 + row_in_piece) * K + k0 + src_koff;
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcB, (lds_uint*)(base + 16384 + pa * 1024), 16, 0, 0);
+    }
+  };
+  stage(0, 0);
+  __syncthreads();
+  for (int kt = 0; kt < nk; ++kt) {
+    int cur = kt & 1;
+    if (kt + 1 < nk) stage(cur ^ 1, kt + 1);
+    char* base = (char*)lds + cur * 32768;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      const int ks = kk * 4 + (lane >> 4);
+      bf16x8_t a_frag[4], b_frag[4];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        int row = wr + mi * 16 + (lane & 15);
+        a_frag[mi] = *(const bf16x8_t*)(base + row * 128 +
+                                        ((ks ^ (row & 7)) & 7) * 16);
+      }
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int row = wc + ni * 16 + (lane & 15);
+        b_frag[ni] = *(const bf16x8_t*)(base + 16384 + row * 128 +
+                                        ((ks ^ (row & 7)) & 7) * 16);
+      }
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+  const int crow_base = row0 + wr + (lane >> 4) * 4;
+  const int ccol_base = col0 + wc + (lane & 15);
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = crow_base + mi * 16 + r;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        C[(long long)row * N + ccol_base + ni * 16] = f32_to_bf16(acc[mi][ni][r]);
+    }
+}
