@@ -67,6 +67,7 @@ def fit_rounds(
     weight_decay: float = 5e-4,
     optimizer: str = "sgd",
     secure_agg: bool = False,
+    dp: dict | None = None,
     comm: CommContext | None = None,
     device=None,
     round_hook=None,
@@ -75,12 +76,22 @@ def fit_rounds(
     device = device or comm.device
     net, loss_fn = _build(model, model_kwargs, device)
     batches = make_batches(dataset, batch_size, device, seed=17 + comm.rank)
+    dp_cfg = None
+    if dp:
+        from vantage6_amd.parallel.dp import DPConfig
+
+        dp_cfg = DPConfig(**dp)
     cfg = FedAvgConfig(
         local_steps=local_steps, lr=lr, momentum=momentum,
         weight_decay=weight_decay, optimizer=optimizer, secure_agg=secure_agg,
+        dp=dp_cfg,
     )
     n = len(dataset[list(dataset.keys())[0]])
     trainer = FedAvgTrainer(net, loss_fn, batches, comm, cfg, n_samples=n)
+    if dp_cfg is not None:
+        from vantage6_amd.models.factory import model_loss_per_sample
+
+        trainer.per_sample_loss_fn = model_loss_per_sample(model)
     history = []
     for _ in range(rounds):
         stats = trainer.run_round()

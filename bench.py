@@ -63,6 +63,7 @@ def main() -> None:
     ap.add_argument("--seq-len", type=int, default=512)
     ap.add_argument("--shard-size", type=int, default=None, help="samples per station")
     ap.add_argument("--secure-agg", action="store_true")
+    ap.add_argument("--dp", action="store_true", help="DP-SGD (ghost clipping)")
     ap.add_argument("--channels-last", action="store_true",
                     help="NHWC layout for conv models (MIOpen fast path)")
     ap.add_argument("--dtype", default="bf16", choices=["bf16"])
@@ -103,13 +104,23 @@ def main() -> None:
             return inner(model, batch)
 
     batches = make_batches(dataset, batch, device, seed=17 + comm.rank)
+    dp_cfg = None
+    if args.dp:
+        from vantage6_amd.parallel.dp import DPConfig
+
+        dp_cfg = DPConfig(max_grad_norm=1.0, noise_multiplier=1.0)
     cfg = FedAvgConfig(
         local_steps=args.local_steps,
         lr=0.1 if args.model in ("resnet18", "logreg") else 1e-4,
         optimizer="sgd" if args.model in ("resnet18", "logreg") else "adamw",
         secure_agg=args.secure_agg,
+        dp=dp_cfg,
     )
     trainer = FedAvgTrainer(net, loss_fn, batches, comm, cfg, n_samples=shard)
+    if dp_cfg is not None:
+        from vantage6_amd.models.factory import model_loss_per_sample
+
+        trainer.per_sample_loss_fn = model_loss_per_sample(args.model)
 
     for _ in range(args.warmup):
         trainer.run_round()
@@ -162,6 +173,7 @@ def main() -> None:
                 "samples_per_round": batch * args.local_steps * n,
                 "samples_per_sec": rounds_per_sec * batch * args.local_steps * n,
                 "secure_agg": args.secure_agg,
+                "dp": args.dp,
                 "final_loss": losses[-1],
                 "n_params": trainer.flat.numel,
                 "hip_ops": __import__("vantage6_amd.ops", fromlist=["has_hip_ops"]).has_hip_ops(),

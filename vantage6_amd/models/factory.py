@@ -49,3 +49,27 @@ def model_loss(name: str) -> Callable:
         return F.cross_entropy(logits.float(), batch["y"])
 
     return cls_loss
+
+
+def model_loss_per_sample(name: str) -> Callable:
+    """(model, batch) -> per-sample loss vector [B] (DP-SGD needs this)."""
+    name = name.lower()
+    if name in ("bert", "bert_base", "bert-base", "llama", "llama_1b", "llama-1b",
+                "llama-3-1b"):
+
+        def lm_loss_ps(model, batch):
+            ids = batch["input_ids"]
+            logits = model(ids[:, :-1])
+            losses = F.cross_entropy(
+                logits.float().reshape(-1, logits.shape[-1]),
+                ids[:, 1:].reshape(-1), reduction="none",
+            )
+            return losses.reshape(ids.shape[0], -1).mean(-1)
+
+        return lm_loss_ps
+
+    def cls_loss_ps(model, batch):
+        logits = model(batch["x"])
+        return F.cross_entropy(logits.float(), batch["y"], reduction="none")
+
+    return cls_loss_ps

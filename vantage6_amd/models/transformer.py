@@ -56,8 +56,9 @@ class BertBase(nn.Module):
 
     def forward(self, input_ids):
         B, S = input_ids.shape
-        pos = torch.arange(S, device=input_ids.device)
-        x = self.ln(self.tok(input_ids) + self.pos(pos)[None])
+        # batch-shaped position ids so per-sample grads stay separable (DP)
+        pos = torch.arange(S, device=input_ids.device)[None].expand(B, S)
+        x = self.ln(self.tok(input_ids) + self.pos(pos))
         for blk in self.blocks:
             x = blk(x)
         return self.head(x)

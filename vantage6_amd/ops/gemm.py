@@ -23,7 +23,7 @@ def matmul_bf16(a: torch.Tensor, b: torch.Tensor, trans_a: bool, trans_b: bool
         return hip_ops().gemm_bf16(a.contiguous(), b.contiguous(), trans_a, trans_b)
     af = (a.float().t() if trans_a else a.float())
     bf = (b.float().t() if trans_b else b.float())
-    return (af @ bf).to(torch.bfloat16)
+    return (af @ bf).to(a.dtype)  # bf16 for bf16 inputs (mirrors the kernel)
 
 
 class _LinearFn(torch.autograd.Function):

@@ -25,6 +25,7 @@ import torch
 import torch.distributed as dist
 
 from .comm import CommContext
+from .dp import DPConfig, DPEngine
 from .flat import FlatParamSpace
 
 _FIXED_POINT_BITS = 32  # fraction bits for secure-agg quantization
@@ -42,6 +43,7 @@ class FedAvgConfig:
     secure_agg: bool = False
     secagg_seed: int = 0x5ECA66
     lr_schedule: Optional[Callable[[int], float]] = None
+    dp: Optional[DPConfig] = None  # per-sample clip + noise (ghost clipping)
 
 
 class FedAvgTrainer:
@@ -73,10 +75,15 @@ class FedAvgTrainer:
             torch.cuda.Stream() if comm.device.type == "cuda" else None
         )
         self.last_stats: dict = {}
+        self.dp_engine = DPEngine(model, cfg.dp) if cfg.dp else None
+        self.per_sample_loss_fn: Optional[Callable] = None  # set when dp on
+        self.last_dp_stats: dict = {}
 
     # ------------------------------------------------------------ local fit
     def local_step(self) -> float:
         batch = next(self.batches)
+        if self.dp_engine is not None:
+            return self._dp_local_step(batch)
         self.flat.zero_grad()
         loss = self.loss_fn(self.model, batch)
         loss.backward()
@@ -89,6 +96,23 @@ class FedAvgTrainer:
                 weight_decay=self.cfg.weight_decay, nesterov=self.cfg.nesterov,
             )
         return float(loss.detach())
+
+    def _dp_local_step(self, batch) -> float:
+        if self.per_sample_loss_fn is None:
+            raise RuntimeError("DP-SGD needs trainer.per_sample_loss_fn set")
+        losses = self.per_sample_loss_fn(self.model, batch)
+        B = losses.shape[0]
+        self.last_dp_stats = self.dp_engine.compute_clipped_grads(
+            losses, self.flat.zero_grad, B
+        )
+        self.dp_engine.add_noise_and_scale(self.flat.grad_flat, B)
+        lr = self.cfg.lr_schedule(self.round_idx) if self.cfg.lr_schedule else self.cfg.lr
+        if self.cfg.optimizer == "adamw":
+            self.flat.adamw_step(lr=lr, weight_decay=self.cfg.weight_decay)
+        else:
+            self.flat.sgd_step(lr=lr, momentum=self.cfg.momentum,
+                               weight_decay=self.cfg.weight_decay)
+        return float(losses.detach().mean())
 
     # ----------------------------------------------------------- aggregation
     def aggregate(self) -> None:
