@@ -64,8 +64,10 @@ def main() -> None:
     ap.add_argument("--shard-size", type=int, default=None, help="samples per station")
     ap.add_argument("--secure-agg", action="store_true")
     ap.add_argument("--dp", action="store_true", help="DP-SGD (ghost clipping)")
-    ap.add_argument("--channels-last", action="store_true",
-                    help="NHWC layout for conv models (MIOpen fast path)")
+    ap.add_argument("--channels-last", dest="channels_last", action="store_true",
+                    default=True, help="NHWC layout for conv models (default)")
+    ap.add_argument("--no-channels-last", dest="channels_last",
+                    action="store_false")
     ap.add_argument("--dtype", default="bf16", choices=["bf16"])
     ap.add_argument("--device", default=None, help="override (e.g. cpu for smoke)")
     args = ap.parse_args()
