@@ -68,6 +68,9 @@ def main() -> None:
                     default=True, help="NHWC layout for conv models (default)")
     ap.add_argument("--no-channels-last", dest="channels_last",
                     action="store_false")
+    ap.add_argument("--graph", dest="graph", action="store_true", default=True,
+                    help="hipGraph-capture the local step (default)")
+    ap.add_argument("--no-graph", dest="graph", action="store_false")
     ap.add_argument("--dtype", default="bf16", choices=["bf16"])
     ap.add_argument("--device", default=None, help="override (e.g. cpu for smoke)")
     args = ap.parse_args()
@@ -126,6 +129,7 @@ def main() -> None:
 
     for _ in range(args.warmup):
         trainer.run_round()
+    used_graph = trainer.enable_graph() if (on_gpu and args.graph) else False
 
     comm.barrier()
     if on_gpu:
@@ -176,6 +180,7 @@ def main() -> None:
                 "samples_per_sec": rounds_per_sec * batch * args.local_steps * n,
                 "secure_agg": args.secure_agg,
                 "dp": args.dp,
+                "hip_graph": used_graph,
                 "final_loss": losses[-1],
                 "n_params": trainer.flat.numel,
                 "hip_ops": __import__("vantage6_amd.ops", fromlist=["has_hip_ops"]).has_hip_ops(),

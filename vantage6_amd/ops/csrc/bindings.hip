@@ -252,14 +252,15 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor w,
       1, std::min<long long>((R + rows_pb * 4 - 1) / (rows_pb * 4), 512));
   torch::Tensor part;
   if (training) {
-    part = torch::empty({2LL * nblocks, C}, fopts);
+    part = torch::empty({2, C}, fopts);
+    hipMemsetAsync(part.data_ptr(), 0, 2 * C * sizeof(float), cur_stream());
     hipLaunchKernelGGL(bn_fwd_partials_kernel, dim3(nblocks), dim3(256), 0,
                        cur_stream(), (const short*)x.data_ptr(),
                        part.data_ptr<float>(), R, C, nblocks);
   } else {
-    part = torch::empty({1}, fopts);
+    part = torch::empty({2, C}, fopts);
   }
-  hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3((C + 63) / 64), dim3(256), 0,
+  hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0,
                      cur_stream(), part.data_ptr<float>(), nblocks, C, R,
                      (const short*)w.data_ptr(), (const short*)b.data_ptr(),
                      mean.data_ptr<float>(), rstd.data_ptr<float>(),
@@ -293,7 +294,8 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
   int rows_pb = bn_rows_pb(C);
   int nblocks = (int)std::max<long long>(
       1, std::min<long long>((R + rows_pb * 4 - 1) / (rows_pb * 4), 512));
-  auto part = torch::empty({2LL * nblocks, C}, fopts);
+  auto part = torch::empty({2, C}, fopts);
+  hipMemsetAsync(part.data_ptr(), 0, 2 * C * sizeof(float), cur_stream());
   hipLaunchKernelGGL(bn_bwd_partials_kernel, dim3(nblocks), dim3(256), 0,
                      cur_stream(), (const short*)dy.data_ptr(),
                      (const short*)y.data_ptr(), (const short*)x.data_ptr(),
@@ -304,7 +306,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
   auto c_scale = torch::empty({C}, fopts);
   auto c_db = torch::empty({C}, fopts);
   auto c_dw = torch::empty({C}, fopts);
-  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + 63) / 64), dim3(256), 0,
+  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0,
                      cur_stream(), part.data_ptr<float>(), nblocks, C, R,
                      (const short*)w.data_ptr(), rstd.data_ptr<float>(),
                      (short*)dw.data_ptr(), (short*)db.data_ptr(),

@@ -65,9 +65,8 @@ bn_fwd_partials_kernel(const short* __restrict__ x, float* __restrict__ part,
   if (rr == 0) {
 #pragma unroll
     for (int q = 0; q < 8; ++q) {
-      part[(long long)blockIdx.x * C + j * 8 + q] = red[threadIdx.x * 8 + q];
-      part[((long long)nblocks + blockIdx.x) * C + j * 8 + q] =
-          red[(BN_NTH + threadIdx.x) * 8 + q];
+      atomicAdd(&part[j * 8 + q], red[threadIdx.x * 8 + q]);
+      atomicAdd(&part[C + j * 8 + q], red[(BN_NTH + threadIdx.x) * 8 + q]);
     }
   }
 }
@@ -75,9 +74,10 @@ bn_fwd_partials_kernel(const short* __restrict__ x, float* __restrict__ part,
 // --------------------------------------------------------- fwd: finalize
 // writes mean/rstd (saved for bwd) + scale/shift (apply pass) and updates
 // running stats in place.
-// grid: cdiv(C,64) blocks × 256 threads; 4 reduction slices per channel
-// (one mostly-idle block serializing a 1024-deep loop cost 199 µs/call —
-// the dominant kernel of the whole ResNet bench before this layout)
+// part is now the fp32 accumulator [2][C] (partials blocks atomicAdd into
+// it; the binding memsets it before the launch). finalize is pure
+// per-channel math — the earlier reduction-loop versions serialized one
+// near-idle block for 199 us (v1) / 38 us (v2) per call.
 extern "C" __global__ void __launch_bounds__(BN_NTH)
 bn_fwd_finalize_kernel(const float* __restrict__ part, int nblocks, int C,
                        long long R, const short* __restrict__ w,
@@ -86,48 +86,27 @@ bn_fwd_finalize_kernel(const float* __restrict__ part, int nblocks, int C,
                        float* __restrict__ shift, float* __restrict__ running_mean,
                        float* __restrict__ running_var, float momentum,
                        float eps, int training) {
-  __shared__ float red[BN_NTH * 2];
-  int c = blockIdx.x * 64 + (threadIdx.x & 63);
-  int slice = threadIdx.x >> 6;  // 0..3
-  {
-    float mean, var;
-    if (training) {
-      float s1 = 0.f, s2 = 0.f;
-      if (c < C) {
-        for (int bkk = slice; bkk < nblocks; bkk += 4) {
-          s1 += part[(long long)bkk * C + c];
-          s2 += part[((long long)nblocks + bkk) * C + c];
-        }
-      }
-      red[threadIdx.x] = s1;
-      red[BN_NTH + threadIdx.x] = s2;
-      __syncthreads();
-      if (slice == 0) {
-#pragma unroll
-        for (int s = 1; s < 4; ++s) {
-          s1 += red[threadIdx.x + s * 64];
-          s2 += red[BN_NTH + threadIdx.x + s * 64];
-        }
-      }
-      if (slice != 0 || c >= C) return;
-      mean = s1 / R;
-      var = s2 / R - mean * mean;
-      if (var < 0.f) var = 0.f;
-      float unbiased = (R > 1) ? var * (float)R / (float)(R - 1) : var;
-      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
-      running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
-    } else {
-      if (slice != 0 || c >= C) return;
-      mean = running_mean[c];
-      var = running_var[c];
-    }
-    float rstd = rsqrtf(var + eps);
-    float wf = bf16_to_f32(w[c]);
-    mean_out[c] = mean;
-    rstd_out[c] = rstd;
-    scale[c] = wf * rstd;
-    shift[c] = bf16_to_f32(b[c]) - mean * wf * rstd;
+  int c = blockIdx.x * BN_NTH + threadIdx.x;
+  if (c >= C) return;
+  float mean, var;
+  if (training) {
+    float s1 = part[c], s2 = part[C + c];
+    mean = s1 / R;
+    var = s2 / R - mean * mean;
+    if (var < 0.f) var = 0.f;
+    float unbiased = (R > 1) ? var * (float)R / (float)(R - 1) : var;
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+  } else {
+    mean = running_mean[c];
+    var = running_var[c];
   }
+  float rstd = rsqrtf(var + eps);
+  float wf = bf16_to_f32(w[c]);
+  mean_out[c] = mean;
+  rstd_out[c] = rstd;
+  scale[c] = wf * rstd;
+  shift[c] = bf16_to_f32(b[c]) - mean * wf * rstd;
 }
 
 // ------------------------------------------------------------ fwd: apply
@@ -230,9 +209,8 @@ bn_bwd_partials_kernel(const short* __restrict__ dy, const short* __restrict__ y
   if (rr == 0) {
 #pragma unroll
     for (int q = 0; q < 8; ++q) {
-      part[(long long)blockIdx.x * C + j * 8 + q] = red[threadIdx.x * 8 + q];
-      part[((long long)nblocks + blockIdx.x) * C + j * 8 + q] =
-          red[(BN_NTH + threadIdx.x) * 8 + q];
+      atomicAdd(&part[j * 8 + q], red[threadIdx.x * 8 + q]);
+      atomicAdd(&part[C + j * 8 + q], red[(BN_NTH + threadIdx.x) * 8 + q]);
     }
   }
 }
@@ -245,25 +223,9 @@ bn_bwd_finalize_kernel(const float* __restrict__ part, int nblocks, int C,
                        const float* __restrict__ rstd, short* __restrict__ dw,
                        short* __restrict__ db, float* __restrict__ c_scale,
                        float* __restrict__ c_db, float* __restrict__ c_dw) {
-  __shared__ float red[BN_NTH * 2];
-  int c = blockIdx.x * 64 + (threadIdx.x & 63);
-  int slice = threadIdx.x >> 6;
-  float sdb = 0.f, sdw = 0.f;
-  if (c < C) {
-    for (int bkk = slice; bkk < nblocks; bkk += 4) {
-      sdb += part[(long long)bkk * C + c];
-      sdw += part[((long long)nblocks + bkk) * C + c];
-    }
-  }
-  red[threadIdx.x] = sdb;
-  red[BN_NTH + threadIdx.x] = sdw;
-  __syncthreads();
-  if (slice != 0 || c >= C) return;
-#pragma unroll
-  for (int s = 1; s < 4; ++s) {
-    sdb += red[threadIdx.x + s * 64];
-    sdw += red[BN_NTH + threadIdx.x + s * 64];
-  }
+  int c = blockIdx.x * BN_NTH + threadIdx.x;
+  if (c >= C) return;
+  float sdb = part[c], sdw = part[C + c];
   dw[c] = f32_to_bf16(sdw);
   db[c] = f32_to_bf16(sdb);
   c_scale[c] = bf16_to_f32(w[c]) * rstd[c];
