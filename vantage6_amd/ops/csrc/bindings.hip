@@ -43,6 +43,7 @@ extern "C" __global__ void gemm_nt_bf16_v5_kernel(const short*, const short*,
 extern "C" __global__ void transpose_bf16_kernel(const short*, short*, int, int);
 extern "C" __global__ void bn_fwd_partials_kernel(const short*, float*, long long,
                                                   int, int);
+extern "C" __global__ void bn_zero_acc_kernel(float*, int);
 extern "C" __global__ void bn_fwd_finalize_kernel(const float*, int, int, long long,
                                                   const short*, const short*, float*,
                                                   float*, float*, float*, float*,
@@ -252,13 +253,14 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor w,
       1, std::min<long long>((R + rows_pb * 4 - 1) / (rows_pb * 4), 512));
   torch::Tensor part;
   if (training) {
-    part = torch::empty({2, C}, fopts);
-    hipMemsetAsync(part.data_ptr(), 0, 2 * C * sizeof(float), cur_stream());
+    part = torch::empty({16, 2, C}, fopts);
+    hipLaunchKernelGGL(bn_zero_acc_kernel, dim3((32 * C + 255) / 256), dim3(256),
+                       0, cur_stream(), part.data_ptr<float>(), 32 * C);
     hipLaunchKernelGGL(bn_fwd_partials_kernel, dim3(nblocks), dim3(256), 0,
                        cur_stream(), (const short*)x.data_ptr(),
                        part.data_ptr<float>(), R, C, nblocks);
   } else {
-    part = torch::empty({2, C}, fopts);
+    part = torch::empty({16, 2, C}, fopts);
   }
   hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0,
                      cur_stream(), part.data_ptr<float>(), nblocks, C, R,
@@ -294,8 +296,9 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
   int rows_pb = bn_rows_pb(C);
   int nblocks = (int)std::max<long long>(
       1, std::min<long long>((R + rows_pb * 4 - 1) / (rows_pb * 4), 512));
-  auto part = torch::empty({2, C}, fopts);
-  hipMemsetAsync(part.data_ptr(), 0, 2 * C * sizeof(float), cur_stream());
+  auto part = torch::empty({16, 2, C}, fopts);
+  hipLaunchKernelGGL(bn_zero_acc_kernel, dim3((32 * C + 255) / 256), dim3(256),
+                     0, cur_stream(), part.data_ptr<float>(), 32 * C);
   hipLaunchKernelGGL(bn_bwd_partials_kernel, dim3(nblocks), dim3(256), 0,
                      cur_stream(), (const short*)dy.data_ptr(),
                      (const short*)y.data_ptr(), (const short*)x.data_ptr(),

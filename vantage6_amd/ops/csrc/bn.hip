@@ -23,6 +23,15 @@ DEVINL void bn_thread_map(int C, int* chunks, int* rows_pb, int* j, int* rr) {
   *rr = threadIdx.x / *chunks;
 }
 
+// zero the [16][2][C] atomic accumulator — a kernel, not hipMemsetAsync,
+// so it is always captured into hipGraphs (a non-captured memset leaves
+// stale sums on every replay → diverging BN stats)
+extern "C" __global__ void __launch_bounds__(BN_NTH)
+bn_zero_acc_kernel(float* __restrict__ part, int n) {
+  int i = blockIdx.x * BN_NTH + threadIdx.x;
+  if (i < n) part[i] = 0.f;
+}
+
 // ---------------------------------------------------------- fwd: partials
 // part layout: [2][nblocks][C] (sum, sumsq)
 extern "C" __global__ void __launch_bounds__(BN_NTH)
@@ -63,10 +72,13 @@ bn_fwd_partials_kernel(const short* __restrict__ x, float* __restrict__ part,
     __syncthreads();
   }
   if (rr == 0) {
+    // 16 contention slices: block b adds into slice b&15 (<=32 atomics per
+    // address instead of up to 512)
+    float* slice = part + (blockIdx.x & 15) * 2 * C;
 #pragma unroll
     for (int q = 0; q < 8; ++q) {
-      atomicAdd(&part[j * 8 + q], red[threadIdx.x * 8 + q]);
-      atomicAdd(&part[C + j * 8 + q], red[(BN_NTH + threadIdx.x) * 8 + q]);
+      atomicAdd(&slice[j * 8 + q], red[threadIdx.x * 8 + q]);
+      atomicAdd(&slice[C + j * 8 + q], red[(BN_NTH + threadIdx.x) * 8 + q]);
     }
   }
 }
@@ -90,7 +102,12 @@ bn_fwd_finalize_kernel(const float* __restrict__ part, int nblocks, int C,
   if (c >= C) return;
   float mean, var;
   if (training) {
-    float s1 = part[c], s2 = part[C + c];
+    float s1 = 0.f, s2 = 0.f;
+#pragma unroll
+    for (int sl = 0; sl < 16; ++sl) {
+      s1 += part[sl * 2 * C + c];
+      s2 += part[sl * 2 * C + C + c];
+    }
     mean = s1 / R;
     var = s2 / R - mean * mean;
     if (var < 0.f) var = 0.f;
@@ -207,10 +224,11 @@ bn_bwd_partials_kernel(const short* __restrict__ dy, const short* __restrict__ y
     __syncthreads();
   }
   if (rr == 0) {
+    float* slice = part + (blockIdx.x & 15) * 2 * C;
 #pragma unroll
     for (int q = 0; q < 8; ++q) {
-      atomicAdd(&part[j * 8 + q], red[threadIdx.x * 8 + q]);
-      atomicAdd(&part[C + j * 8 + q], red[(BN_NTH + threadIdx.x) * 8 + q]);
+      atomicAdd(&slice[j * 8 + q], red[threadIdx.x * 8 + q]);
+      atomicAdd(&slice[C + j * 8 + q], red[(BN_NTH + threadIdx.x) * 8 + q]);
     }
   }
 }
@@ -225,7 +243,12 @@ bn_bwd_finalize_kernel(const float* __restrict__ part, int nblocks, int C,
                        float* __restrict__ c_db, float* __restrict__ c_dw) {
   int c = blockIdx.x * BN_NTH + threadIdx.x;
   if (c >= C) return;
-  float sdb = part[c], sdw = part[C + c];
+  float sdb = 0.f, sdw = 0.f;
+#pragma unroll
+  for (int sl = 0; sl < 16; ++sl) {
+    sdb += part[sl * 2 * C + c];
+    sdw += part[sl * 2 * C + C + c];
+  }
   dw[c] = f32_to_bf16(sdw);
   db[c] = f32_to_bf16(sdb);
   c_scale[c] = bf16_to_f32(w[c]) * rstd[c];
