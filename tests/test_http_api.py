@@ -117,3 +117,46 @@ def test_node_token_and_open_runs(server, net):
     rr = requests.get(f"{server}/api/run", params={"state": "open"},
                       headers={"Authorization": f"Bearer {tok}"})
     assert rr.status_code == 200
+
+
+def test_websocket_event_push(net):
+    """Socket.IO-parity: a websocket subscriber receives new_task events."""
+    from starlette.testclient import TestClient
+
+    from vantage6_amd.server.api import create_app
+
+    app = create_app(net.core)
+    tok = net.client.gw.token if hasattr(net.client.gw, "token") else None
+    client = TestClient(app)
+    with client.websocket_connect(f"/api/ws?token={tok}") as ws:
+        task = net.client.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],
+            name="ws-test", image="algorithms.column_mean",
+            input_={"method": "partial_mean", "kwargs": {"columns": ["v"]}},
+        )
+        # researcher subscribes to the collaboration room: receives the run
+        # status stream (initializing -> active -> completed); new_task goes
+        # to the per-organization room (the stations'), like the reference
+        statuses = []
+        for _ in range(10):
+            msg = ws.receive_json()
+            if msg["event"] == "algorithm_status_change":
+                statuses.append(msg["data"]["status"])
+                if msg["data"]["status"] == "completed":
+                    break
+        assert "completed" in statuses
+        net.client.wait_for_results(task["id"], timeout=30)
+
+
+def test_websocket_rejects_bad_token(net):
+    from starlette.testclient import TestClient
+
+    from vantage6_amd.server.api import create_app
+
+    client = TestClient(create_app(net.core))
+    import pytest as _pytest
+
+    with _pytest.raises(Exception):
+        with client.websocket_connect("/api/ws?token=garbage") as ws:
+            ws.receive_json()

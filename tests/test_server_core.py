@@ -218,3 +218,64 @@ class TestPagination:
         assert page1["links"]["total"] == 26  # + root org
         page3 = root.organization.list(page=3, per_page=10)
         assert len(page3["data"]) == 6
+
+
+class TestAlgorithmStore:
+    def test_submit_review_approve_flow(self, core, root):
+        ident = root.gw.ident
+        a = core.store_submit_algorithm(ident, "mean", "algorithms.column_mean",
+                                        "federated column mean")
+        assert a["status"] == "submitted"
+        assert any(f["name"] == "central_mean" for f in a["functions"])
+        a = core.store_review_algorithm(ident, a["id"], True, "lgtm")
+        assert a["status"] == "approved"
+        assert core.algorithm_is_approved("algorithms.column_mean")
+        listed = core.store_list_algorithms(ident, status="approved")["data"]
+        assert len(listed) == 1
+
+    def test_review_needs_permission(self, core, root):
+        orgs, collab = make_collab(root)
+        root.user.create("res2", "pw", organization_id=orgs[0], roles=["Researcher"])
+        res = UserClient(core=core)
+        res.authenticate("res2", "pw")
+        a = core.store_submit_algorithm(res.gw.ident, "x", "algorithms.diagnostics")
+        with pytest.raises(PermissionError_):
+            core.store_review_algorithm(res.gw.ident, a["id"], True)
+
+    def test_station_requires_approval_policy(self, core, root):
+        from vantage6_amd.common.config import StationConfig
+        from vantage6_amd.node.station import Station
+
+        orgs, collab = make_collab(root, name="storenet")
+        node = root.node.create(name="sn0", collaboration_id=collab["id"],
+                                organization_id=orgs[0])
+        cfg = StationConfig(name="sn0", organization_id=orgs[0],
+                            policies={"require_approved_algorithms": True})
+        st = Station(core, cfg, api_key=node["api_key"])
+        st.start()
+        try:
+            root.user.create("r3", "pw", organization_id=orgs[0],
+                             roles=["Researcher"])
+            res = UserClient(core=core)
+            res.authenticate("r3", "pw")
+            task = res.task.create(
+                collaboration=collab["id"], organizations=[orgs[0]],
+                name="t", image="algorithms.column_mean",
+                input_={"method": "partial_mean", "kwargs": {"columns": []}},
+            )
+            import time
+
+            for _ in range(100):
+                t = res.task.get(task["id"])
+                if t["status"] != "pending" and t["status"] != "initializing":
+                    break
+                time.sleep(0.05)
+            assert t["status"] == "not allowed"
+            # approve it -> new task runs (crashes on missing db, but allowed)
+            core.store_submit_algorithm(root.gw.ident, "mean",
+                                        "algorithms.column_mean")
+            algs = core.store_list_algorithms(root.gw.ident)["data"]
+            core.store_review_algorithm(root.gw.ident, algs[-1]["id"], True)
+            assert core.algorithm_is_approved("algorithms.column_mean")
+        finally:
+            st.stop()

@@ -169,12 +169,19 @@ class Station:
         task = self.core.get_task(self.identity, task_id)
         self.core.patch_run(self.identity, run_id, status=RunStatus.INITIALIZING.value)
 
-        # -- policy check (reference: DockerManager allowed-images policy)
+        # -- policy checks (reference: DockerManager allowed-images policy +
+        #    allowed-algorithm-stores)
         allowed = self.config.policies.get("allowed_algorithms")
         if allowed is not None and task["image"] not in allowed:
             self.core.patch_run(
                 self.identity, run_id, status=RunStatus.NOT_ALLOWED.value,
                 log=f"algorithm {task['image']!r} not in station allow-list",
+            )
+            return
+        if self.config.policies.get("require_approved_algorithms") and                 not self.core.algorithm_is_approved(task["image"]):
+            self.core.patch_run(
+                self.identity, run_id, status=RunStatus.NOT_ALLOWED.value,
+                log=f"algorithm {task['image']!r} is not store-approved",
             )
             return
         try:

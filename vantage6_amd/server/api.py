@@ -12,7 +12,7 @@ from __future__ import annotations
 
 from typing import Any, Optional
 
-from fastapi import Depends, FastAPI, HTTPException, Request
+from fastapi import Depends, FastAPI, HTTPException, Request, WebSocket
 from pydantic import BaseModel, Field
 
 from ..common import jwt as jwtlib
@@ -93,6 +93,17 @@ class StudyCreateBody(BaseModel):
     name: str
     collaboration_id: int
     organization_ids: list[int]
+
+
+class AlgorithmSubmitBody(BaseModel):
+    name: str
+    image: str
+    description: str = ""
+
+
+class AlgorithmReviewBody(BaseModel):
+    approve: bool
+    comment: str = ""
 
 
 class RunPatchBody(BaseModel):
@@ -243,6 +254,78 @@ def create_app(core: ServerCore) -> FastAPI:
     def study_create(body: StudyCreateBody, who: Identity = Depends(ident)):
         return wrap(core.create_study, who, body.name, body.collaboration_id,
                     body.organization_ids)
+
+    # ------------------------------------------------------ algorithm store
+    @app.post("/api/algorithm", status_code=201)
+    def algorithm_submit(body: AlgorithmSubmitBody, who: Identity = Depends(ident)):
+        return wrap(core.store_submit_algorithm, who, body.name, body.image,
+                    body.description)
+
+    @app.get("/api/algorithm")
+    def algorithm_list(who: Identity = Depends(ident), status: Optional[str] = None,
+                       page: int = 1, per_page: int = 20):
+        return wrap(core.store_list_algorithms, who, status=status, page=page,
+                    per_page=per_page)
+
+    @app.post("/api/algorithm/{algorithm_id}/review")
+    def algorithm_review(algorithm_id: int, body: AlgorithmReviewBody,
+                         who: Identity = Depends(ident)):
+        return wrap(core.store_review_algorithm, who, algorithm_id,
+                    body.approve, body.comment)
+
+    # ------------------------------------------------------------ websocket
+    # Reference parity: the Socket.IO '/tasks' namespace (SURVEY.md §3.2 L3,
+    # A.4) — a client subscribes to its collaboration rooms and receives
+    # new_task / algorithm_status_change / kill_containers / node-status
+    # events pushed instead of polling. Rooms derive from the caller's
+    # identity like the reference's on_connect.
+    @app.websocket("/api/ws")
+    async def events_ws(ws: WebSocket):
+        import asyncio
+        import anyio
+
+        await ws.accept()
+        try:
+            token = ws.query_params.get("token") or (
+                ws.headers.get("Authorization", "").removeprefix("Bearer ")
+            )
+            who = core.identity_from_token(token)
+        except Exception:  # noqa: BLE001
+            await ws.close(code=4401)
+            return
+        rooms = []
+        if who.collaboration_id is not None:
+            rooms.append(f"collaboration_{who.collaboration_id}")
+            if who.organization_id is not None:
+                rooms.append(
+                    f"collaboration_{who.collaboration_id}"
+                    f"_organization_{who.organization_id}"
+                )
+        elif who.type == "user":
+            # users join every collaboration their org is in
+            for c in core.list_collaborations(who, per_page=1000)["data"]:
+                rooms.append(f"collaboration_{c['id']}")
+        queue: "asyncio.Queue" = asyncio.Queue()
+        loop = asyncio.get_event_loop()
+
+        def forward(event, data, _room):
+            loop.call_soon_threadsafe(queue.put_nowait,
+                                      {"event": event, "data": data})
+
+        subs = []
+        for room in rooms:
+            cb = (lambda e, d, r=room: forward(e, d, r))
+            core.events.subscribe(room, cb)
+            subs.append((room, cb))
+        try:
+            while True:
+                msg = await queue.get()
+                await ws.send_json(msg)
+        except Exception:  # noqa: BLE001 — disconnects
+            pass
+        finally:
+            for room, cb in subs:
+                core.events.unsubscribe(room, cb)
 
     # ----------------------------------------------------------------- ops
     @app.get("/api/version")

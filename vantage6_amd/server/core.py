@@ -818,3 +818,84 @@ class ServerCore:
             out["input"] = r.input
             out["result"] = r.result
         return out
+
+
+# ------------------------------------------------------------ algorithm store
+# Reference parity: vantage6-algorithm-store — curated
+# registry with review workflow, folded into the orchestrator (the reference
+# runs it as a second Flask app sharing the server skeleton).
+def _extend_server_core():
+    import inspect
+    import json as _json
+
+    from .model import Algorithm
+
+    def store_submit_algorithm(self, ident, name, image, description=""):
+        if ident.type != "user":
+            raise PermissionError_("only users submit algorithms")
+        s = self.session
+        if s.query(Algorithm).filter_by(image=image).first():
+            raise BadRequest(f"algorithm {image!r} already submitted")
+        # introspect the module for its function signatures (UI hints)
+        functions = []
+        try:
+            from ..algorithm.wrap import resolve_module
+
+            mod = resolve_module(image)
+            for fname, fn in vars(mod).items():
+                if callable(fn) and not fname.startswith("_") and \
+                        getattr(fn, "__module__", "") == mod.__name__:
+                    try:
+                        sig = str(inspect.signature(fn))
+                    except (TypeError, ValueError):
+                        sig = "(...)"
+                    functions.append({"name": fname, "signature": sig,
+                                      "doc": (fn.__doc__ or "").strip()[:200]})
+        except Exception:  # noqa: BLE001 — unresolvable module: still listable
+            pass
+        alg = Algorithm(name=name, image=image, description=description,
+                        functions=_json.dumps(functions),
+                        submitted_by=ident.id)
+        s.add(alg)
+        s.commit()
+        return self._algorithm_json(alg)
+
+    def store_review_algorithm(self, ident, algorithm_id, approve, comment=""):
+        self._check_user_can("algorithm", Operation.EDIT, ident)
+        s = self.session
+        alg = s.get(Algorithm, algorithm_id)
+        if alg is None:
+            raise NotFound(f"algorithm {algorithm_id} not found")
+        alg.status = "approved" if approve else "rejected"
+        alg.reviewed_by = ident.id
+        alg.review_comment = comment
+        s.commit()
+        return self._algorithm_json(alg)
+
+    def store_list_algorithms(self, ident, status=None, page=1, per_page=20):
+        q = self.session.query(Algorithm).order_by(Algorithm.id)
+        if status:
+            q = q.filter(Algorithm.status == status)
+        return self._paginate([self._algorithm_json(a) for a in q.all()],
+                              page, per_page)
+
+    def algorithm_is_approved(self, image: str) -> bool:
+        alg = self.session.query(Algorithm).filter_by(image=image).first()
+        return alg is not None and alg.status == "approved"
+
+    def _algorithm_json(self, a):
+        return {
+            "id": a.id, "name": a.name, "image": a.image,
+            "description": a.description,
+            "functions": _json.loads(a.functions or "[]"),
+            "status": a.status, "submitted_by": a.submitted_by,
+            "reviewed_by": a.reviewed_by, "review_comment": a.review_comment,
+        }
+
+    for fn in (store_submit_algorithm, store_review_algorithm,
+               store_list_algorithms, _algorithm_json):
+        setattr(ServerCore, fn.__name__, _locked(fn) if fn.__name__ != "_algorithm_json" else fn)
+    ServerCore.algorithm_is_approved = _locked(algorithm_is_approved)
+
+
+_extend_server_core()

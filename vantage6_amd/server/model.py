@@ -242,6 +242,24 @@ class Run(Base):
     organization = relationship("Organization")
 
 
+class Algorithm(Base):
+    """Algorithm-store entry (vantage6-algorithm-store, 4.2+):
+    curated registry of algorithm packages with a review workflow
+    (submitted -> approved/rejected); stations can require approval."""
+
+    __tablename__ = "algorithm"
+    id = Column(Integer, primary_key=True)
+    name = Column(String, nullable=False)
+    image = Column(String, unique=True, nullable=False)  # module path
+    description = Column(String, default="")
+    functions = Column(Text, default="[]")  # JSON: [{name, args, doc}]
+    status = Column(String, default="submitted")  # submitted|approved|rejected
+    submitted_by = Column(Integer, ForeignKey("user.id"), nullable=True)
+    reviewed_by = Column(Integer, ForeignKey("user.id"), nullable=True)
+    review_comment = Column(String, default="")
+    created_at = Column(DateTime, default=utcnow)
+
+
 class Database:
     """Engine/session holder (``vantage6-server/.../model/base.py::Database``)."""
 

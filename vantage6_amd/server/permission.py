@@ -42,6 +42,7 @@ RESOURCES = [
     "rule",
     "study",
     "event",
+    "algorithm",  # algorithm-store entries
 ]
 
 # scope ordering: a broader scope implies the narrower ones
