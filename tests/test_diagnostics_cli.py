@@ -25,3 +25,14 @@ def test_demo_network_cli():
     r = CliRunner().invoke(cli, ["dev", "demo", "--stations", "2"])
     assert r.exit_code == 0, r.output
     assert "column_mean" in r.output
+
+
+def test_algorithm_scaffold(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    (tmp_path / "tests").mkdir()
+    r = CliRunner().invoke(cli, ["algorithm", "create", "--name", "my_count"])
+    assert r.exit_code == 0, r.output
+    assert (tmp_path / "algorithms" / "my_count.py").exists()
+    assert (tmp_path / "tests" / "test_algo_my_count.py").exists()
+    src = (tmp_path / "algorithms" / "my_count.py").read_text()
+    assert "central_my_count" in src and "partial_my_count" in src

@@ -279,3 +279,40 @@ class TestAlgorithmStore:
             assert core.algorithm_is_approved("algorithms.column_mean")
         finally:
             st.stop()
+
+
+class TestMFAAndRecovery:
+    def test_mfa_login_flow(self, core, root):
+        from vantage6_amd.common import totp
+
+        out = core.enable_mfa(root.gw.ident)
+        assert out["uri"].startswith("otpauth://totp/")
+        with pytest.raises(PermissionError_):
+            core.authenticate_user("root", "root")  # code now required
+        code = totp.totp_code(out["secret"])
+        tokens = core.authenticate_user("root", "root", mfa_code=code)
+        assert "access_token" in tokens
+
+    def test_password_recovery_flow(self, core, root):
+        from vantage6_amd.server.mail import (MailService,
+                                              request_password_recovery,
+                                              reset_password)
+
+        orgs, collab = make_collab(root, name="rec")
+        root.user.create("alice", "old-pw", organization_id=orgs[0],
+                         roles=["Researcher"], email="alice@example.org")
+        mail = MailService()
+        request_password_recovery(core, mail, "alice")
+        assert len(mail.outbox) == 1
+        token = mail.outbox[0]["body"].split(": ")[-1]
+        reset_password(core, token, "new-pw")
+        with pytest.raises(PermissionError_):
+            core.authenticate_user("alice", "old-pw")
+        assert "access_token" in core.authenticate_user("alice", "new-pw")
+
+    def test_recovery_no_user_enumeration(self, core):
+        from vantage6_amd.server.mail import MailService, request_password_recovery
+
+        mail = MailService()
+        request_password_recovery(core, mail, "ghost")  # must not raise
+        assert mail.outbox == []

@@ -141,6 +141,63 @@ def test(stations):
     sys.exit(0 if report.get("ok") else 1)
 
 
+@cli.group()
+def algorithm():
+    """Algorithm-package tools (`v6 algorithm` analog)."""
+
+
+@algorithm.command("create")
+@click.option("--name", required=True, help="package name (python identifier)")
+@click.option("--dir", "out_dir", default="algorithms", type=click.Path())
+def algorithm_create(name, out_dir):
+    """Scaffold a new algorithm package (central/partial pair + mock test)."""
+    pkg = Path(out_dir) / f"{name}.py"
+    if pkg.exists():
+        raise click.ClickException(f"{pkg} exists")
+    pkg.parent.mkdir(parents=True, exist_ok=True)
+    pkg.write_text(f'''"""Algorithm package `{name}` — scaffolded by v6amd algorithm create."""
+from vantage6_amd.algorithm import algorithm_client, data
+
+
+@algorithm_client
+def central_{name}(client, **kwargs):
+    """Central part: fan out partials and aggregate."""
+    org_ids = [org["id"] for org in client.organization.list()]
+    task = client.task.create(
+        input_={{"method": "partial_{name}", "kwargs": kwargs}},
+        organizations=org_ids,
+    )
+    results = client.wait_for_results(task["id"])
+    return {{"n_partials": len(results), "partials": results}}
+
+
+@data(1)
+def partial_{name}(df, **kwargs):
+    """Partial part: computed against this station's local data."""
+    return {{"n_rows": int(len(df))}}
+''')
+    test = Path("tests") / f"test_algo_{name}.py"
+    if not test.exists() and Path("tests").is_dir():
+        test.write_text(f'''"""Mock-client test for `{name}` (scaffolded)."""
+from vantage6_amd.algorithm import MockAlgorithmClient
+
+
+def test_{name}_mock(tmp_path):
+    import pandas as pd
+
+    p = tmp_path / "d.csv"
+    pd.DataFrame({{"x": [1, 2, 3]}}).to_csv(p, index=False)
+    mock = MockAlgorithmClient(
+        datasets=[[{{"label": "default", "database": str(p), "type": "csv"}}]],
+        module="algorithms.{name}",
+    )
+    task = mock.task.create(input_={{"method": "central_{name}"}}, organizations=[1])
+    out = mock.result.from_task(task["id"])[0]
+    assert out["n_partials"] == 1
+''')
+    click.echo(f"scaffolded {pkg}" + (f" and {test}" if test.exists() else ""))
+
+
 @cli.command()
 @click.option("--model", default="resnet18")
 @click.option("--steps", default=20)

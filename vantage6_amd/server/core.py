@@ -164,7 +164,8 @@ class ServerCore:
 
     # ------------------------------------------------------------------ auth
     @_locked
-    def authenticate_user(self, username: str, password: str) -> dict:
+    def authenticate_user(self, username: str, password: str,
+                          mfa_code: str | None = None) -> dict:
         s = self.session
         user = s.query(User).filter_by(username=username).first()
         if user is None or not user.check_password(password):
@@ -173,6 +174,13 @@ class ServerCore:
                 user.last_login_attempt = utcnow()
                 s.commit()
             raise PermissionError_("invalid username or password", 401)
+        if user.mfa_secret:
+            from ..common import totp
+
+            if not mfa_code:
+                raise PermissionError_("mfa code required", 401)
+            if not totp.verify_code(user.mfa_secret, mfa_code):
+                raise PermissionError_("invalid mfa code", 401)
         user.failed_login_attempts = 0
         user.last_seen = utcnow()
         s.commit()
@@ -235,6 +243,18 @@ class ServerCore:
             "image": image,
         }
         return {"container_token": jwtlib.encode(payload, self.jwt_secret, self.token_expires)}
+
+    @_locked
+    def enable_mfa(self, ident: Identity) -> dict:
+        """Provision a TOTP secret for the calling user (vantage6 4.x MFA)."""
+        from ..common import totp
+
+        user = self._user(ident)
+        secret = totp.generate_secret()
+        user.mfa_secret = secret
+        self.session.commit()
+        return {"secret": secret,
+                "uri": totp.provisioning_uri(secret, user.username)}
 
     @_locked
     def refresh_token(self, refresh_token: str) -> dict:

@@ -134,6 +134,7 @@ class User(Authenticatable):
     email = Column(String, default="")
     failed_login_attempts = Column(Integer, default=0)
     last_login_attempt = Column(DateTime, nullable=True)
+    mfa_secret = Column(String, default="")  # TOTP secret (empty = MFA off)
     organization_id = Column(Integer, ForeignKey("organization.id"))
 
     organization = relationship("Organization", back_populates="users")
