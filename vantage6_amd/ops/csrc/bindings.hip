@@ -40,6 +40,10 @@ extern "C" __global__ void gemm_nt_bf16_v4_kernel(const short*, const short*,
                                                   short*, int, int, int);
 extern "C" __global__ void gemm_nt_bf16_v5_kernel(const short*, const short*,
                                                   short*, int, int, int);
+extern "C" __global__ void gemm_nt_bf16_v6_kernel(const short*, const short*,
+                                                  short*, int, int, int);
+extern "C" __global__ void gemm_nt_bf16_v7_kernel(const short*, const short*,
+                                                  short*, int, int, int);
 extern "C" __global__ void transpose_bf16_kernel(const short*, short*, int, int);
 extern "C" __global__ void bn_fwd_partials_kernel(const short*, float*, long long,
                                                   int, int);
@@ -363,12 +367,24 @@ torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b, bool trans_a,
     int nwg = (M / 128) * (N / 128);
     static int variant = [] {
       const char* v = getenv("V6AMD_GEMM_KERNEL");
-      return v ? atoi(v) : 5;  // v5 (dbuf glds + src swizzle) default
+      // default: shape-aware — v7 (256² counted-vmcnt) when divisible,
+      // v5 (128² dbuf) otherwise
+      return v ? atoi(v) : 0;
     }();
+    if ((variant == 0 || variant == 6 || variant == 7) && M % 256 == 0 &&
+        N % 256 == 0) {
+      int nwg6 = (M / 256) * (N / 256);
+      auto k6 = (variant == 6) ? gemm_nt_bf16_v6_kernel : gemm_nt_bf16_v7_kernel;
+      hipLaunchKernelGGL(k6, dim3(nwg6), dim3(512), 0, cur_stream(),
+                         (const short*)A.data_ptr(), (const short*)B.data_ptr(),
+                         (short*)c.data_ptr(), M, N, K);
+      return c;
+    }
     auto kfn = gemm_nt_bf16_v4_kernel;
     if (variant == 2) kfn = gemm_nt_bf16_v2_kernel;
     if (variant == 3) kfn = gemm_nt_bf16_v3_kernel;
-    if (variant == 5) kfn = gemm_nt_bf16_v5_kernel;
+    if (variant == 0 || variant == 5 || variant == 6 || variant == 7)
+      kfn = gemm_nt_bf16_v5_kernel;
     hipLaunchKernelGGL(kfn, dim3(nwg), dim3(256), 0, cur_stream(),
                        (const short*)A.data_ptr(), (const short*)B.data_ptr(),
                        (short*)c.data_ptr(), M, N, K);

@@ -557,3 +557,191 @@ gemm_nt_bf16_v5_kernel(const short* __restrict__ A, const short* __restrict__ B,
         C[(long long)row * N + ccol_base + ni * 16] = f32_to_bf16(acc[mi][ni][r]);
     }
 }
+
+// v6 — 256×256 tile, 8 waves, BK=64, v4-style full-line glds + src XOR
+// swizzle, single LDS buffer (64 KB → 2 blocks/CU). Twice the
+// compute-per-staged-byte of the 128² tiles.
+extern "C" __global__ void __launch_bounds__(512)
+gemm_nt_bf16_v6_kernel(const short* __restrict__ A, const short* __restrict__ B,
+                       short* __restrict__ C, int M, int N, int K) {
+  __shared__ short lds[2 * 2048 * 8];  // A[256][64] + B[256][64] bf16 = 64 KB
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;           // 0..7
+  const int wr = (wave >> 2) * 128;    // 2 M-groups of 128
+  const int wc = (wave & 3) * 64;      // 4 N-groups of 64
+  const int nwg = gridDim.x;
+  int wgid = blockIdx.x;
+  {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = wgid & 7, idx = wgid >> 3;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int ntile_n = N / 256;
+  const int row0 = (wgid / ntile_n) * 256;
+  const int col0 = (wgid % ntile_n) * 256;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int src_koff = (((lane & 7) ^ (lane >> 3)) & 7) * 8;
+  const int row_in_piece = lane >> 3;
+
+  const int nk = K / BK;
+  for (int kt = 0; kt < nk; ++kt) {
+    const long long k0 = (long long)kt * BK;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int pa = wave * 4 + i;  // 32 pieces of 8 rows cover 256 rows
+      const short* srcA =
+          A + (long long)(row0 + pa * 8 + row_in_piece) * K + k0 + src_koff;
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcA, (lds_uint*)((char*)lds + pa * 1024), 16, 0, 0);
+      const short* srcB =
+          B + (long long)(col0 + pa * 8 + row_in_piece) * K + k0 + src_koff;
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcB, (lds_uint*)((char*)lds + 32768 + pa * 1024), 16, 0,
+          0);
+    }
+    __syncthreads();  // drain glds + barrier
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      const int ks = kk * 4 + (lane >> 4);
+      bf16x8_t b_frag[4];
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int row = wc + ni * 16 + (lane & 15);
+        b_frag[ni] = *(const bf16x8_t*)((char*)lds + 32768 + row * 128 +
+                                        ((ks ^ (row & 7)) & 7) * 16);
+      }
+#pragma unroll
+      for (int mi = 0; mi < 8; ++mi) {
+        int row = wr + mi * 16 + (lane & 15);
+        bf16x8_t a_frag = *(const bf16x8_t*)((char*)lds + row * 128 +
+                                             ((ks ^ (row & 7)) & 7) * 16);
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag, b_frag[ni], acc[mi][ni], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+  const int crow_base = row0 + wr + (lane >> 4) * 4;
+  const int ccol_base = col0 + wc + (lane & 15);
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = crow_base + mi * 16 + r;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        C[(long long)row * N + ccol_base + ni * 16] = f32_to_bf16(acc[mi][ni][r]);
+    }
+}
+
+// v7 — 256² tile, double-buffered glds with COUNTED vmcnt + raw barriers.
+// Per K-tile: issue the 8 prefetch glds for tile t+1 into buf^1, then
+// `s_waitcnt vmcnt(8)` (= everything older, i.e. tile t's stages, has
+// landed), one barrier, free-running fragment compute, one barrier before
+// the next tile may overwrite. No vmcnt(0) drain in the loop (the
+// structural stall of the __syncthreads() variants — guide T3/T4).
+extern "C" __global__ void __launch_bounds__(512)
+gemm_nt_bf16_v7_kernel(const short* __restrict__ A, const short* __restrict__ B,
+                       short* __restrict__ C, int M, int N, int K) {
+  __shared__ short lds[2 * 2 * 2048 * 8];  // 2 buf × (A+B)[256][64] = 128 KB
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = (wave >> 2) * 128;
+  const int wc = (wave & 3) * 64;
+  const int nwg = gridDim.x;
+  int wgid = blockIdx.x;
+  {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = wgid & 7, idx = wgid >> 3;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int ntile_n = N / 256;
+  const int row0 = (wgid / ntile_n) * 256;
+  const int col0 = (wgid % ntile_n) * 256;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int src_koff = (((lane & 7) ^ (lane >> 3)) & 7) * 8;
+  const int row_in_piece = lane >> 3;
+
+  auto stage = [&](int buf, int kt) {
+    const long long k0 = (long long)kt * BK;
+    char* base = (char*)lds + buf * 65536;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int pa = wave * 4 + i;
+      const short* srcA =
+          A + (long long)(row0 + pa * 8 + row_in_piece) * K + k0 + src_koff;
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcA, (lds_uint*)(base + pa * 1024), 16, 0, 0);
+      const short* srcB =
+          B + (long long)(col0 + pa * 8 + row_in_piece) * K + k0 + src_koff;
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcB, (lds_uint*)(base + 32768 + pa * 1024), 16, 0, 0);
+    }
+  };
+
+  const int nk = K / BK;
+  stage(0, 0);
+  for (int kt = 0; kt < nk; ++kt) {
+    int cur = kt & 1;
+    if (kt + 1 < nk) {
+      stage(cur ^ 1, kt + 1);
+      // tile t's 8 stages are everything older than the 8 just issued
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();  // all waves' tile-t data visible
+    char* base = (char*)lds + cur * 65536;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      const int ks = kk * 4 + (lane >> 4);
+      bf16x8_t b_frag[4];
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int row = wc + ni * 16 + (lane & 15);
+        b_frag[ni] = *(const bf16x8_t*)(base + 32768 + row * 128 +
+                                        ((ks ^ (row & 7)) & 7) * 16);
+      }
+#pragma unroll
+      for (int mi = 0; mi < 8; ++mi) {
+        int row = wr + mi * 16 + (lane & 15);
+        bf16x8_t a_frag = *(const bf16x8_t*)(base + row * 128 +
+                                             ((ks ^ (row & 7)) & 7) * 16);
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag, b_frag[ni], acc[mi][ni], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+      }
+    }
+    __builtin_amdgcn_s_barrier();  // reads done before t+1 overwrites buf
+  }
+  const int crow_base = row0 + wr + (lane >> 4) * 4;
+  const int ccol_base = col0 + wc + (lane & 15);
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = crow_base + mi * 16 + r;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        C[(long long)row * N + ccol_base + ni * 16] = f32_to_bf16(acc[mi][ni][r]);
+    }
+}
