@@ -47,8 +47,10 @@ def synthetic_dataset(model: str, n: int, seed: int, model_kwargs: dict):
         return {"x": x, "y": y}
     # token models
     seq = model_kwargs.pop("_seq_len", 512)
-    vocab = model_kwargs.get("vocab", 30522 if model.startswith("bert") else 32000)
-    return {"input_ids": torch.randint(0, vocab, (n, seq), generator=g)}
+    vocab = model_kwargs.get("vocab", 30522 if model.startswith("bert") else 128256)
+    # seq+1 tokens: LM loss shifts by one, so the model still sees seq_len
+    # positions and GEMM M-dims stay tile-aligned (batch*seq multiples of 256)
+    return {"input_ids": torch.randint(0, vocab, (n, seq + 1), generator=g)}
 
 
 def main() -> None:

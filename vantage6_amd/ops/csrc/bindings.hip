@@ -362,6 +362,23 @@ torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b, bool trans_a,
   torch::Tensor B = trans_b ? b : transpose_bf16(b);
   int M = (int)A.size(0), K = (int)A.size(1), N = (int)B.size(0);
   TORCH_CHECK((int)B.size(1) == K, "inner dims mismatch");
+  // pad to tile multiples so the fast kernels always run (zero rows/cols
+  // are exact under GEMM); slice the result back
+  int M0 = M, N0 = N;
+  {
+    int Kp = (K + 63) / 64 * 64;
+    int align = (M >= 256 || N >= 256) ? 256 : 128;
+    int Mp = (M + align - 1) / align * align;
+    int Np = (N + align - 1) / align * align;
+    namespace F = torch::nn::functional;
+    if (Kp != K || Mp != M) {
+      A = torch::constant_pad_nd(A, {0, Kp - K, 0, Mp - M});
+    }
+    if (Kp != K || Np != N) {
+      B = torch::constant_pad_nd(B, {0, Kp - K, 0, Np - N});
+    }
+    M = Mp; N = Np; K = Kp;
+  }
   auto c = torch::empty({M, N}, a.options());
   if (M % 128 == 0 && N % 128 == 0 && K % 64 == 0) {
     int nwg = (M / 128) * (N / 128);
@@ -378,6 +395,8 @@ torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b, bool trans_a,
       hipLaunchKernelGGL(k6, dim3(nwg6), dim3(512), 0, cur_stream(),
                          (const short*)A.data_ptr(), (const short*)B.data_ptr(),
                          (short*)c.data_ptr(), M, N, K);
+      if (M0 != M || N0 != N)
+        return c.narrow(0, 0, M0).narrow(1, 0, N0).contiguous();
       return c;
     }
     auto kfn = gemm_nt_bf16_v4_kernel;
@@ -394,6 +413,8 @@ torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b, bool trans_a,
                        (const short*)A.data_ptr(), (const short*)B.data_ptr(),
                        (short*)c.data_ptr(), M, N, K, K, K, N);
   }
+  if (M0 != M || N0 != N)
+    return c.narrow(0, 0, M0).narrow(1, 0, N0).contiguous();
   return c;
 }
 
