@@ -120,8 +120,15 @@ def central_fedavg(
     local_steps: int = 2,
     batch_size: int = 16,
     lr: float = 0.1,
+    min_stations: int = 1,
+    fail_orgs: list | None = None,  # fault injection (tests): orgs that crash
 ):
-    """Central part: per round, fan out partial_fit and weighted-average."""
+    """Central part: per round, fan out partial_fit and weighted-average.
+
+    Elastic: a station whose run crashes or is killed
+    simply drops out of that round's weighted mean — FedAvg tolerates it;
+    the round fails only if fewer than ``min_stations`` complete.
+    """
     org_ids = [o["id"] for o in client.organization.list()]
     # round 0 params come from a deterministic seed on every station
     global_params: dict | None = None
@@ -134,12 +141,19 @@ def central_fedavg(
                     "model": model, "model_kwargs": model_kwargs,
                     "params": global_params, "local_steps": local_steps,
                     "batch_size": batch_size, "lr": lr, "seed": 1234,
+                    "fail_orgs": fail_orgs,
                 },
             },
             organizations=org_ids,
             name=f"fedavg round {r}",
         )
-        results = client.wait_for_results(task["id"], timeout=600)
+        results = client.wait_for_results(task["id"], timeout=600,
+                                          allow_failures=True)
+        if len(results) < min_stations:
+            raise RuntimeError(
+                f"round {r}: only {len(results)} stations completed "
+                f"(min_stations={min_stations})"
+            )
         total = sum(res["n_samples"] for res in results)
         avg = None
         for res in results:
@@ -156,8 +170,7 @@ def central_fedavg(
     return {"params": global_params, "history": history}
 
 
-@data(1)
-def partial_fit(
+def _partial_fit_impl(
     db,
     model: str = "logreg",
     model_kwargs: dict | None = None,
@@ -166,8 +179,13 @@ def partial_fit(
     batch_size: int = 16,
     lr: float = 0.1,
     seed: int = 1234,
+    fail_orgs: list | None = None,
+    _v6_env: dict | None = None,
 ):
     """Partial: local SGD steps on this station's shard; return new params."""
+    env = _v6_env or {}
+    if fail_orgs and int(env.get("ORGANIZATION_ID", -1) or -1) in fail_orgs:
+        raise RuntimeError("injected fault: station down mid-round")
     device = "cuda" if torch.cuda.is_available() else "cpu"
     net, loss_fn = _build(model, model_kwargs, device, seed=seed)
     if params is not None:
@@ -191,3 +209,7 @@ def partial_fit(
         "loss": sum(losses) / len(losses),
         "n_samples": n,
     }
+
+
+_partial_fit_impl._v6_wants_env = True  # @data forwards the env map through
+partial_fit = data(1)(_partial_fit_impl)

@@ -86,3 +86,38 @@ def test_fedavg_single_process_runs():
     )
     assert out["rounds"] == 2
     assert out["final_loss"] > 0
+
+
+def test_elastic_round_survives_station_failure(tmp_path):
+    """API-mode FedAvg with fault injection: one of 3 stations crashes every
+    round; the weighted mean proceeds over the 2 survivors (SURVEY.md §6.3
+    elastic recovery); run statuses record the crash."""
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+
+    dbs = [[DatabaseConfig(label="default",
+                           uri=f"synthetic://tabular?n=128&d=8&seed={i}",
+                           type="synthetic")] for i in range(3)]
+    with DevNetwork(n_stations=3, databases=dbs, name="elastic") as net:
+        fail_org = net.org_ids[2]
+        task = net.client.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],
+            name="elastic fedavg", image="algorithms.fedavg",
+            input_={"method": "central_fedavg", "kwargs": {
+                "model": "logreg",
+                "model_kwargs": {"in_features": 8, "num_classes": 2},
+                "rounds": 2, "local_steps": 1, "batch_size": 16,
+                "min_stations": 2, "fail_orgs": [fail_org],
+            }},
+        )
+        out = net.client.wait_for_results(task["id"], timeout=120)[0]
+        assert len(out["history"]) == 2
+        # the injected-fault station shows crashed runs in the task tree
+        tasks = net.client.task.list(per_page=100)["data"]
+        sub = [t for t in tasks if t["parent_id"] == task["id"]]
+        assert sub, "no subtasks recorded"
+        runs = net.client.run.from_task(sub[0]["id"])
+        statuses = {r["organization_id"]: r["status"] for r in runs}
+        assert statuses[fail_org] == "crashed"
+        assert sum(1 for s in statuses.values() if s == "completed") == 2

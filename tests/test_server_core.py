@@ -316,3 +316,29 @@ class TestMFAAndRecovery:
         mail = MailService()
         request_password_recovery(core, mail, "ghost")  # must not raise
         assert mail.outbox == []
+
+
+def test_node_heartbeat_updates_last_seen(core, root):
+    from vantage6_amd.common.config import StationConfig
+    from vantage6_amd.node.station import Station
+
+    orgs, collab = make_collab(root, name="hb")
+    node = root.node.create(name="hb0", collaboration_id=collab["id"],
+                            organization_id=orgs[0])
+    st = Station(core, StationConfig(name="hb0", organization_id=orgs[0]),
+                 api_key=node["api_key"])
+    st.heartbeat_interval = 0.05
+    events = []
+    core.events.subscribe(f"collaboration_{collab['id']}",
+                          lambda e, d: events.append((e, d)))
+    st.start()
+    try:
+        import time
+
+        time.sleep(0.3)
+        infos = [d for e, d in events if e == "node-info-update"]
+        assert infos and "queue_depth" in infos[-1]
+        nodes = root.node.list()["data"]
+        assert nodes[0]["last_seen"] is not None
+    finally:
+        st.stop()

@@ -35,23 +35,42 @@ class AlgorithmClient:
         return cls(HttpProxy(token=token, base_url=f"{host}:{port}{api_path}"))
 
     # ---------------------------------------------------------------- waiting
-    def wait_for_results(self, task_id: int, interval: float = 0.1, timeout: float = 300.0):
+    def wait_for_results(self, task_id: int, interval: float = 0.1,
+                         timeout: float = 300.0, allow_failures: bool = False):
         """Block until all runs of `task_id` finish; return decrypted results.
 
         Reference: ``AlgorithmClient.wait_for_results`` (poll loop). The
         in-process deployment polls the orchestrator directly, so `interval`
         can be far below the reference's ~1 s default.
+
+        With ``allow_failures=True`` (elastic federated rounds, SURVEY.md
+        §6.3: a crashed station simply drops out of the weighted mean) the
+        call waits until every run is terminal and returns only the
+        COMPLETED runs' results; inspect ``run.from_task`` for statuses.
         """
+        from ..common.task_status import RunStatus
+
         t0 = time.monotonic()
         while True:
-            task = self._proxy.get_task(task_id)
-            status = task["status"]
-            if has_task_finished(status):
-                if has_task_failed(status):
-                    raise CollectResultsError(
-                        f"task {task_id} ended with status {status!r}"
-                    )
-                return self.result.from_task(task_id)
+            if allow_failures:
+                runs = self._proxy.list_runs(task_id)
+                if runs and all(has_task_finished(r["status"]) for r in runs):
+                    completed = [r for r in runs
+                                 if r["status"] == RunStatus.COMPLETED.value]
+                    if not completed:
+                        raise CollectResultsError(
+                            f"task {task_id}: every run failed"
+                        )
+                    return [self._proxy.get_result(r["id"]) for r in completed]
+            else:
+                task = self._proxy.get_task(task_id)
+                status = task["status"]
+                if has_task_finished(status):
+                    if has_task_failed(status):
+                        raise CollectResultsError(
+                            f"task {task_id} ended with status {status!r}"
+                        )
+                    return self.result.from_task(task_id)
             if time.monotonic() - t0 > timeout:
                 raise CollectResultsError(f"timeout waiting for task {task_id}")
             time.sleep(interval)

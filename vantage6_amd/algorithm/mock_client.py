@@ -136,5 +136,9 @@ class MockAlgorithmClient:
         def get_collaboration_id(self) -> int:
             return self.parent.collaboration_id
 
-    def wait_for_results(self, task_id: int, interval: float = 0.0, timeout: float = 0.0):
-        return self.result.from_task(task_id)
+    def wait_for_results(self, task_id: int, interval: float = 0.0,
+                         timeout: float = 0.0, allow_failures: bool = False):
+        results = self.result.from_task(task_id)
+        if allow_failures:
+            return [r for r in results if r is not None]
+        return results

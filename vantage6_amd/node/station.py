@@ -95,8 +95,33 @@ class Station:
             target=self._listening_worker, name=f"station-{self.config.name}", daemon=True
         )
         self._thread.start()
+        self._stop_ping = threading.Event()
+        self._ping_thread = threading.Thread(
+            target=self._heartbeat, name=f"station-{self.config.name}-ping",
+            daemon=True,
+        )
+        self._ping_thread.start()
+
+    heartbeat_interval: float = 15.0
+
+    def _heartbeat(self) -> None:
+        """Periodic last_seen ping + node-info-update (reference: socket
+        ping loop + capability broadcast)."""
+        while not self._stop_ping.wait(self.heartbeat_interval):
+            try:
+                with self._lock:
+                    busy = self._inflight
+                self.core.node_ping(self.identity, {
+                    "device": self.config.device,
+                    "queue_depth": self.queue.qsize(),
+                    "inflight": busy,
+                })
+            except Exception:  # noqa: BLE001
+                log.exception("heartbeat failed")
 
     def stop(self, timeout: float = 10.0) -> None:
+        if getattr(self, "_stop_ping", None) is not None:
+            self._stop_ping.set()
         self.queue.put(_STOP)
         if self._thread:
             self._thread.join(timeout)

@@ -725,6 +725,26 @@ class ServerCore:
         return runs
 
     @_locked
+    def node_ping(self, ident: Identity, info: dict | None = None) -> None:
+        """Heartbeat: update last_seen + relay node-info-update (reference:
+        socket ping/last_seen + on_node_info_update)."""
+        if ident.type != "node":
+            raise PermissionError_("only nodes ping")
+        s = self.session
+        node = s.get(Node, ident.id)
+        if node is None:
+            raise NotFound(f"node {ident.id} not found")
+        node.last_seen = utcnow()
+        node.status = "online"
+        s.commit()
+        if info:
+            self.events.emit(
+                "node-info-update",
+                {"id": node.id, "name": node.name, **info},
+                room=f"collaboration_{node.collaboration_id}",
+            )
+
+    @_locked
     def node_set_offline(self, node_id: int) -> None:
         s = self.session
         node = s.get(Node, node_id)
