@@ -133,6 +133,9 @@ def main() -> None:
         trainer.run_round()
     used_graph = trainer.enable_graph() if (on_gpu and args.graph) else False
 
+    from vantage6_amd.utils.metrics import RoundMetrics
+
+    metrics = RoundMetrics()
     comm.barrier()
     if on_gpu:
         torch.cuda.synchronize()
@@ -140,6 +143,7 @@ def main() -> None:
     losses = []
     for _ in range(args.steps):
         stats = trainer.run_round()
+        metrics.record(stats)
         losses.append(stats["loss"])
     comm.barrier()
     if on_gpu:
@@ -188,6 +192,9 @@ def main() -> None:
                 "hip_ops": __import__("vantage6_amd.ops", fromlist=["has_hip_ops"]).has_hip_ops(),
             },
         }
+        # per-round timing table on stderr; stdout stays
+        # the single JSON contract line
+        print(metrics.table(last=min(args.steps, 12)), file=sys.stderr)
         print(json.dumps(out))
 
 

@@ -160,3 +160,39 @@ def test_websocket_rejects_bad_token(net):
     with _pytest.raises(Exception):
         with client.websocket_connect("/api/ws?token=garbage") as ws:
             ws.receive_json()
+
+
+def test_external_container_via_wrap_and_http(server, net, tmp_path):
+    """A.3 contract end-to-end as an *external* process would run it:
+    INPUT/TOKEN/OUTPUT files + env map + AlgorithmClient over HTTP."""
+    from vantage6_amd.algorithm.wrap import wrap_algorithm
+    from vantage6_amd.common.serialization import deserialize, serialize
+
+    # a node forges a container token for a running task (as the station does)
+    core = net.core
+    root = UserClient(core=core)
+    root.authenticate("root", "root")
+    task = net.client.task.create(
+        collaboration=net.collaboration["id"], organizations=[net.org_ids[0]],
+        name="host-task", image="algorithms.column_mean",
+        input_={"method": "partial_mean", "kwargs": {"columns": ["v"]}},
+    )
+    net.stations[0].join_idle()
+    node_ident = net.stations[0].identity
+    ctok = core.token_for_container(node_ident, task["id"],
+                                    "algorithms.column_mean")["container_token"]
+
+    input_file = tmp_path / "input"
+    output_file = tmp_path / "output"
+    token_file = tmp_path / "token"
+    input_file.write_bytes(serialize(
+        {"method": "central_mean", "kwargs": {"columns": ["v"]}}))
+    token_file.write_text(ctok)
+    env = {
+        "INPUT_FILE": str(input_file), "OUTPUT_FILE": str(output_file),
+        "TOKEN_FILE": str(token_file), "PKG_NAME": "algorithms.column_mean",
+        "HOST": "http://127.0.0.1", "PORT": str(PORT), "API_PATH": "/api",
+    }
+    wrap_algorithm(environ=env)
+    out = deserialize(output_file.read_bytes())
+    assert out["v"] == pytest.approx(10, abs=2)

@@ -21,7 +21,6 @@ import queue
 import tempfile
 import threading
 import traceback
-from contextlib import redirect_stderr, redirect_stdout
 from pathlib import Path
 
 from ..common import base64s_to_bytes, bytes_to_base64s
@@ -35,6 +34,7 @@ from ..algorithm.client import AlgorithmClient
 from ..algorithm.exceptions import AlgorithmError, InputError
 from ..algorithm.wrap import dispatch, resolve_module
 from .proxy import NodeProxy
+from ..utils.stdcapture import capture_thread_output
 
 log = logging.getLogger(__name__)
 
@@ -268,9 +268,8 @@ class Station:
 
         # -- execute on the pinned device, capturing stdout/stderr as the log
         self.core.patch_run(self.identity, run_id, status=RunStatus.ACTIVE.value)
-        buf = io.StringIO()
         try:
-            with redirect_stdout(buf), redirect_stderr(buf):
+            with capture_thread_output() as buf:
                 self._bind_device()
                 result = dispatch(module, payload, env, client)
             blob = serialize(result)

@@ -452,9 +452,14 @@ class ServerCore:
     def list_collaborations(self, ident: Identity, page=1, per_page=20) -> dict:
         self._check_user_can_view_any("collaboration", ident)
         cs = self.session.query(Collaboration).order_by(Collaboration.id).all()
-        rc = self._rc("collaboration", ident)
-        if rc.allowed_scope(Operation.VIEW) != Scope.GLOBAL:
-            cs = [c for c in cs if any(o.id == ident.organization_id for o in c.organizations)]
+        if ident.type == "user":
+            rc = self._rc("collaboration", ident)
+            if rc.allowed_scope(Operation.VIEW) != Scope.GLOBAL:
+                cs = [c for c in cs
+                      if any(o.id == ident.organization_id for o in c.organizations)]
+        else:
+            # nodes/containers see only their own collaboration
+            cs = [c for c in cs if c.id == ident.collaboration_id]
         return self._paginate([self._collab_json(c) for c in cs], page, per_page)
 
     @_locked
