@@ -66,6 +66,8 @@ def main() -> None:
     ap.add_argument("--shard-size", type=int, default=None, help="samples per station")
     ap.add_argument("--secure-agg", action="store_true")
     ap.add_argument("--dp", action="store_true", help="DP-SGD (ghost clipping)")
+    ap.add_argument("--agg-algo", default="allreduce",
+                    choices=["allreduce", "reduce_bcast"])
     ap.add_argument("--channels-last", dest="channels_last", action="store_true",
                     default=True, help="NHWC layout for conv models (default)")
     ap.add_argument("--no-channels-last", dest="channels_last",
@@ -122,6 +124,7 @@ def main() -> None:
         optimizer="sgd" if args.model in ("resnet18", "logreg") else "adamw",
         secure_agg=args.secure_agg,
         dp=dp_cfg,
+        agg_algo=args.agg_algo,
     )
     trainer = FedAvgTrainer(net, loss_fn, batches, comm, cfg, n_samples=shard)
     if dp_cfg is not None:

@@ -43,11 +43,27 @@ def _worker(rank: int, world: int, port: int, results):
             batch_size=32, lr=0.05, secure_agg=True, comm=comm,
             device=torch.device("cpu"),
         )
+        # reduce+broadcast aggregation must match all-reduce exactly-ish
+        from vantage6_amd.parallel.fedavg import FedAvgConfig, FedAvgTrainer
+        from algorithms.fedavg import _build, make_batches
+
+        def run_agg(algo):
+            net2, loss2 = _build("logreg", mk, "cpu")
+            tr = FedAvgTrainer(net2, loss2, make_batches(dataset, 32, "cpu", seed=3),
+                               comm, FedAvgConfig(local_steps=1, lr=0.05,
+                                                  agg_algo=algo), n_samples=256)
+            for _ in range(2):
+                tr.run_round()
+            return float(tr.flat.master_flat.double().sum())
+
+        ck_ar = run_agg("allreduce")
+        ck_rb = run_agg("reduce_bcast")
         results[rank] = {
             "plain_loss": out_plain["final_loss"],
             "sec_loss": out_sec["final_loss"],
             "plain_ck": out_plain["master_checksum"],
             "sec_ck": out_sec["master_checksum"],
+            "agg_ar": ck_ar, "agg_rb": ck_rb,
         }
     finally:
         dist.destroy_process_group()
@@ -69,6 +85,9 @@ def test_fedavg_gloo_two_ranks_plain_and_secure():
     # after aggregation every rank must hold the identical global model
     assert results[0]["plain_ck"] == pytest.approx(results[1]["plain_ck"], abs=1e-6)
     assert results[0]["sec_ck"] == pytest.approx(results[1]["sec_ck"], abs=1e-6)
+    # reduce_bcast == allreduce (same math, different collective schedule)
+    for r in (0, 1):
+        assert results[r]["agg_ar"] == pytest.approx(results[r]["agg_rb"], abs=1e-4)
 
 
 def test_fedavg_single_process_runs():

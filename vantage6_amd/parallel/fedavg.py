@@ -44,6 +44,11 @@ class FedAvgConfig:
     secagg_seed: int = 0x5ECA66
     lr_schedule: Optional[Callable[[int], float]] = None
     dp: Optional[DPConfig] = None  # per-sample clip + noise (ghost clipping)
+    # "allreduce": one bucketed all-reduce (ring; per-link bound on xGMI).
+    # "reduce_bcast": reduce-to-root + broadcast — on the fully-connected
+    # 7-link xGMI topology the root's links aggregate ~1 TB/s in/out
+    # (SURVEY.md §6.8); pick per measurement on the 8-GPU node.
+    agg_algo: str = "allreduce"
 
 
 class FedAvgTrainer:
@@ -179,9 +184,21 @@ class FedAvgTrainer:
         master.mul_(self.weight)
         bucket = max(1, self.cfg.bucket_bytes // 4)
         works = []
+        use_rb = self.cfg.agg_algo == "reduce_bcast"
         for off in range(0, master.numel(), bucket):
             chunk = master[off : off + bucket]
-            works.append((off, chunk, self.comm.all_reduce(chunk, async_op=True)))
+            if use_rb:
+                w = self.comm.reduce(chunk, dst=0, async_op=True)
+            else:
+                w = self.comm.all_reduce(chunk, async_op=True)
+            works.append((off, chunk, w))
+        if use_rb:
+            for off, chunk, work in works:
+                if work is not None:
+                    work.wait()
+            bworks = [(off, chunk, self.comm.broadcast(chunk, src=0, async_op=True))
+                      for off, chunk, _ in works]
+            works = bworks
         # overlap: convert each bucket to bf16 params as soon as it lands
         for off, chunk, work in works:
             if work is not None:
