@@ -340,3 +340,31 @@ def test_graph_step_matches_eager(ops):
     a, b = eager.flat.master_flat, graphed.flat.master_flat
     rel = (a - b).abs().max().item() / (a.abs().max().item() + 1e-9)
     assert rel < 1e-3, f"master diverged: rel {rel}"
+
+
+def test_scatter_grads_matches_flat(ops):
+    """Multi-tensor SGD over stolen scattered grads == flat-buffer SGD."""
+    from algorithms.fedavg import _build
+    from vantage6_amd.parallel.flat import FlatParamSpace
+
+    def run(scatter):
+        torch.manual_seed(21)
+        net, loss_fn = _build("resnet18", {"num_classes": 10}, DEV)
+        net = net.to(memory_format=torch.channels_last)
+        fs = FlatParamSpace(net, device=DEV, scatter_grads=scatter)
+        g = torch.Generator().manual_seed(3)
+        for _ in range(3):
+            x = torch.randn(4, 3, 64, 64, generator=g).to(
+                DEV, torch.bfloat16).contiguous(memory_format=torch.channels_last)
+            y = torch.randint(0, 10, (4,), generator=g).to(DEV)
+            fs.zero_grad()
+            loss = torch.nn.functional.cross_entropy(net(x).float(), y)
+            loss.backward()
+            fs.sgd_step(lr=0.05, momentum=0.9, weight_decay=1e-4)
+        return fs.master_flat.clone()
+
+    a = run(False)
+    b = run(True)
+    # BN batch stats are atomic-nondeterministic → bf16-chaos tolerance
+    rel = (a - b).abs().max().item() / (a.abs().max().item() + 1e-9)
+    assert rel < 5e-2, rel

@@ -16,7 +16,8 @@ from . import hip_ops
 class _FusedBNFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var, training,
-                momentum, eps, relu, residual):
+                momentum, eps, relu, residual, fwd_acc=None, bwd_acc=None):
+        ctx.fwd_acc, ctx.bwd_acc = fwd_acc, bwd_acc
         if x.is_cuda:
             xc = x.contiguous(memory_format=torch.channels_last)
             N, C, H, W = xc.shape
@@ -27,7 +28,7 @@ class _FusedBNFn(torch.autograd.Function):
                         .permute(0, 2, 3, 1).reshape(-1, C))
             y2, mean, rstd = hip_ops().bn_fwd(
                 x2, weight, bias, running_mean, running_var, training,
-                momentum, eps, relu, res2,
+                momentum, eps, relu, res2, ctx.fwd_acc,
             )
             y = y2.view(N, H, W, C).permute(0, 3, 1, 2)
             ctx.save_for_backward(x2, y2, weight, mean, rstd)
@@ -61,13 +62,14 @@ class _FusedBNFn(torch.autograd.Function):
             N, C, H, W = shape
             dy2 = (dy.contiguous(memory_format=torch.channels_last)
                    .permute(0, 2, 3, 1).reshape(-1, C))
-            outs = hip_ops().bn_bwd(dy2, y, x, weight, mean, rstd, relu, has_res)
+            outs = hip_ops().bn_bwd(dy2, y, x, weight, mean, rstd, relu,
+                                    has_res, ctx.bwd_acc)
             dx2, dw, db = outs[0], outs[1], outs[2]
             dx = dx2.view(N, H, W, C).permute(0, 3, 1, 2)
             dres = None
             if has_res:
                 dres = outs[3].view(N, H, W, C).permute(0, 3, 1, 2)
-            return dx, dw, db, None, None, None, None, None, None, dres
+            return dx, dw, db, None, None, None, None, None, None, dres, None, None
         # ---- CPU reference
         dyf = dy.float()
         if relu:
@@ -82,7 +84,7 @@ class _FusedBNFn(torch.autograd.Function):
             dyf - db_[None, :, None, None] / R - xhat * dw_[None, :, None, None] / R
         )
         return (dxf.to(x.dtype), dw_.to(weight.dtype), db_.to(weight.dtype),
-                None, None, None, None, None, None, dres)
+                None, None, None, None, None, None, dres, None, None)
 
 
 class FusedBatchNorm2d(torch.nn.Module):
@@ -95,9 +97,18 @@ class FusedBatchNorm2d(torch.nn.Module):
         self.register_buffer("running_mean", torch.zeros(num_features))
         self.register_buffer("running_var", torch.ones(num_features))
         self.eps, self.momentum = eps, momentum
+        # persistent atomic-stat accumulators (zeroed once; the finalize
+        # kernels re-zero after reading — saves a launch per call)
+        self._fwd_acc = None
+        self._bwd_acc = None
 
     def forward(self, x, residual=None, relu: bool = False):
+        if x.is_cuda and self._fwd_acc is None:
+            c = self.weight.numel()
+            self._fwd_acc = torch.zeros(16, 2, c, device=x.device)
+            self._bwd_acc = torch.zeros(16, 2, c, device=x.device)
         return _FusedBNFn.apply(
             x, self.weight, self.bias, self.running_mean, self.running_var,
             self.training, self.momentum, self.eps, relu, residual,
+            self._fwd_acc, self._bwd_acc,
         )

@@ -8,6 +8,10 @@
 // kernels defined in sibling TUs
 extern "C" __global__ void fused_sgd_kernel(short*, float*, const short*, float*,
                                             long long, float, float, float, int);
+struct SgdChunk { const short* grad; long long offset; long long n; };
+extern "C" __global__ void fused_sgd_mt_kernel(const SgdChunk*, int, short*,
+                                               float*, float*, float, float,
+                                               float, int);
 extern "C" __global__ void fused_adamw_kernel(short*, float*, const short*, float*,
                                               float*, long long, float, float, float,
                                               float, float, float, float);
@@ -114,6 +118,21 @@ void fused_adamw(torch::Tensor param, torch::Tensor master, torch::Tensor grad,
                      exp_avg.data_ptr<float>(), exp_avg_sq.data_ptr<float>(), n,
                      (float)lr, (float)beta1, (float)beta2, (float)eps,
                      (float)weight_decay, bc1, bc2);
+}
+
+// table: int64 [n_chunks, 3] = {grad ptr, flat offset, numel} (device)
+void fused_sgd_mt(torch::Tensor table, torch::Tensor param, torch::Tensor master,
+                  torch::Tensor mom, double lr, double momentum,
+                  double weight_decay, bool nesterov) {
+  CHECK_HIP(table, torch::kInt64);
+  CHECK_HIP(param, torch::kBFloat16);
+  int n_chunks = (int)table.size(0);
+  int grid = std::min(n_chunks, 2048);
+  hipLaunchKernelGGL(fused_sgd_mt_kernel, dim3(grid), dim3(256), 0,
+                     cur_stream(), (const SgdChunk*)table.data_ptr(), n_chunks,
+                     (short*)param.data_ptr(), master.data_ptr<float>(),
+                     mom.data_ptr<float>(), (float)lr, (float)momentum,
+                     (float)weight_decay, nesterov ? 1 : 0);
 }
 
 // ---------------------------------------------------------------- secagg
@@ -241,7 +260,8 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor w,
                                   torch::Tensor b, torch::Tensor running_mean,
                                   torch::Tensor running_var, bool training,
                                   double momentum, double eps, bool relu,
-                                  c10::optional<torch::Tensor> residual) {
+                                  c10::optional<torch::Tensor> residual,
+                                  c10::optional<torch::Tensor> acc) {
   CHECK_HIP(x, torch::kBFloat16);
   int C = (int)x.size(-1);
   TORCH_CHECK(C % 8 == 0 && C <= 2048, "BN channels must be %8==0 and <=2048");
@@ -256,15 +276,19 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor w,
   int nblocks = (int)std::max<long long>(
       1, std::min<long long>((R + rows_pb * 4 - 1) / (rows_pb * 4), 512));
   torch::Tensor part;
-  if (training) {
+  if (acc.has_value()) {
+    part = acc.value();  // persistent, zeroed at creation; finalize re-zeroes
+  } else {
     part = torch::empty({16, 2, C}, fopts);
-    hipLaunchKernelGGL(bn_zero_acc_kernel, dim3((32 * C + 255) / 256), dim3(256),
-                       0, cur_stream(), part.data_ptr<float>(), 32 * C);
+    if (training)
+      hipLaunchKernelGGL(bn_zero_acc_kernel, dim3((32 * C + 255) / 256),
+                         dim3(256), 0, cur_stream(), part.data_ptr<float>(),
+                         32 * C);
+  }
+  if (training) {
     hipLaunchKernelGGL(bn_fwd_partials_kernel, dim3(nblocks), dim3(256), 0,
                        cur_stream(), (const short*)x.data_ptr(),
                        part.data_ptr<float>(), R, C, nblocks);
-  } else {
-    part = torch::empty({16, 2, C}, fopts);
   }
   hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0,
                      cur_stream(), part.data_ptr<float>(), nblocks, C, R,
@@ -292,7 +316,8 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor w,
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
                                   torch::Tensor x, torch::Tensor w,
                                   torch::Tensor mean, torch::Tensor rstd,
-                                  bool relu, bool want_dres) {
+                                  bool relu, bool want_dres,
+                                  c10::optional<torch::Tensor> acc) {
   CHECK_HIP(dy, torch::kBFloat16);
   int C = (int)x.size(-1);
   long long R = x.numel() / C;
@@ -300,9 +325,14 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
   int rows_pb = bn_rows_pb(C);
   int nblocks = (int)std::max<long long>(
       1, std::min<long long>((R + rows_pb * 4 - 1) / (rows_pb * 4), 512));
-  auto part = torch::empty({16, 2, C}, fopts);
-  hipLaunchKernelGGL(bn_zero_acc_kernel, dim3((32 * C + 255) / 256), dim3(256),
-                     0, cur_stream(), part.data_ptr<float>(), 32 * C);
+  torch::Tensor part;
+  if (acc.has_value()) {
+    part = acc.value();
+  } else {
+    part = torch::empty({16, 2, C}, fopts);
+    hipLaunchKernelGGL(bn_zero_acc_kernel, dim3((32 * C + 255) / 256), dim3(256),
+                       0, cur_stream(), part.data_ptr<float>(), 32 * C);
+  }
   hipLaunchKernelGGL(bn_bwd_partials_kernel, dim3(nblocks), dim3(256), 0,
                      cur_stream(), (const short*)dy.data_ptr(),
                      (const short*)y.data_ptr(), (const short*)x.data_ptr(),
@@ -421,6 +451,7 @@ torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b, bool trans_a,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_sgd", &fused_sgd, "fused SGD-momentum on flat buffers");
   m.def("fused_adamw", &fused_adamw, "fused AdamW on flat buffers");
+  m.def("fused_sgd_mt", &fused_sgd_mt, "multi-tensor SGD, scattered grads");
   m.def("secagg_quantize_mask", &secagg_quantize_mask,
         "fixed-point quantize + pairwise Philox masks");
   m.def("dp_scale_noise", &dp_scale_noise, "DP clip-scale + gaussian noise");

@@ -111,6 +111,12 @@ bn_fwd_finalize_kernel(const float* __restrict__ part, int nblocks, int C,
     }
     mean = s1 / R;
     var = s2 / R - mean * mean;
+    // reset the accumulator for the next call (saves a zeroing launch)
+#pragma unroll
+    for (int sl = 0; sl < 16; ++sl) {
+      ((float*)part)[sl * 2 * C + c] = 0.f;
+      ((float*)part)[sl * 2 * C + C + c] = 0.f;
+    }
     if (var < 0.f) var = 0.f;
     float unbiased = (R > 1) ? var * (float)R / (float)(R - 1) : var;
     running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
@@ -249,6 +255,8 @@ bn_bwd_finalize_kernel(const float* __restrict__ part, int nblocks, int C,
   for (int sl = 0; sl < 16; ++sl) {
     sdb += part[sl * 2 * C + c];
     sdw += part[sl * 2 * C + C + c];
+    ((float*)part)[sl * 2 * C + c] = 0.f;
+    ((float*)part)[sl * 2 * C + C + c] = 0.f;
   }
   dw[c] = f32_to_bf16(sdw);
   db[c] = f32_to_bf16(sdb);

@@ -181,3 +181,66 @@ norm2_bf16_kernel(const short* __restrict__ x, long long n,
   }
   if (threadIdx.x == 0) out_partial[blockIdx.x] = red[0];
 }
+
+// ------------------------------------------------- multi-tensor fused SGD
+// Scattered-gradient variant: autograd "steals" each param's grad into its
+// own tensor (no pre-assigned flat views), which removes BOTH the
+// grad-buffer zeroing pass and the AccumulateGrad add pass. The chunk
+// table maps {grad ptr, flat offset, n} per <=16Ki-element chunk; master/
+// momentum/param stay flat. Grad physical element order must equal the
+// flat order (host side validates layouts).
+struct SgdChunk {
+  const short* grad;
+  long long offset;
+  long long n;
+};
+
+extern "C" __global__ void __launch_bounds__(256)
+fused_sgd_mt_kernel(const SgdChunk* __restrict__ table, int n_chunks,
+                    short* __restrict__ param, float* __restrict__ master,
+                    float* __restrict__ mom, float lr, float momentum,
+                    float weight_decay, int nesterov) {
+  for (int c = blockIdx.x; c < n_chunks; c += gridDim.x) {
+    SgdChunk ch = table[c];
+    for (long long i = (long long)threadIdx.x * 8; i < ch.n; i += 256 * 8) {
+      long long gi = i;
+      long long fi = ch.offset + i;
+      if (gi + 8 <= ch.n) {
+        s16x8 g8 = *(const s16x8*)(ch.grad + gi);
+        f32x4 m0 = *(const f32x4*)(master + fi);
+        f32x4 m1 = *(const f32x4*)(master + fi + 4);
+        f32x4 v0 = *(const f32x4*)(mom + fi);
+        f32x4 v1 = *(const f32x4*)(mom + fi + 4);
+        s16x8 p8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float g = bf16_to_f32(g8[j]);
+          float m = (j < 4) ? m0[j] : m1[j - 4];
+          float v = (j < 4) ? v0[j] : v1[j - 4];
+          g = fmaf(weight_decay, m, g);
+          v = fmaf(momentum, v, g);
+          float step = nesterov ? fmaf(momentum, v, g) : v;
+          m = fmaf(-lr, step, m);
+          if (j < 4) { m0[j] = m; v0[j] = v; } else { m1[j - 4] = m; v1[j - 4] = v; }
+          p8[j] = f32_to_bf16(m);
+        }
+        *(f32x4*)(master + fi) = m0;
+        *(f32x4*)(master + fi + 4) = m1;
+        *(f32x4*)(mom + fi) = v0;
+        *(f32x4*)(mom + fi + 4) = v1;
+        *(s16x8*)(param + fi) = p8;
+      } else {
+        for (long long k = gi; k < ch.n; ++k) {
+          float g = bf16_to_f32(ch.grad[k]);
+          long long f = ch.offset + k;
+          float m = master[f], v = mom[f];
+          g = fmaf(weight_decay, m, g);
+          v = fmaf(momentum, v, g);
+          float step = nesterov ? fmaf(momentum, v, g) : v;
+          m = fmaf(-lr, step, m);
+          master[f] = m; mom[f] = v; param[f] = f32_to_bf16(m);
+        }
+      }
+    }
+  }
+}

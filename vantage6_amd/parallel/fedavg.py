@@ -66,7 +66,12 @@ class FedAvgTrainer:
         self.batches = batches
         self.comm = comm
         self.cfg = cfg
-        self.flat = FlatParamSpace(model, device=comm.device)
+        # scattered grads (no zero+accumulate passes) on GPU unless DP
+        # needs the flat grad buffer for clip/noise
+        scatter = comm.device.type == "cuda" and cfg.dp is None and \
+            cfg.optimizer == "sgd"
+        self.flat = FlatParamSpace(model, device=comm.device,
+                                   scatter_grads=scatter)
         self.round_idx = 0
         # FedAvg weight w_i = n_i / sum_j n_j
         n = torch.tensor([float(n_samples)], device=comm.device)
@@ -108,6 +113,11 @@ class FedAvgTrainer:
             self._graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(self._graph):
                 self._static_loss = self._graph_body()
+            if self.flat.scatter_grads:
+                # replays reuse the capture pool's grad addresses: point the
+                # persistent chunk table at them now (outside capture);
+                # strict: any layout conversion would break replay
+                self.flat.refresh_sgd_table(strict=True)
             return True
         except Exception:  # noqa: BLE001 — fall back to eager
             import logging
