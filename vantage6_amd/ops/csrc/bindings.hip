@@ -12,6 +12,10 @@ struct SgdChunk { const short* grad; long long offset; long long n; };
 extern "C" __global__ void fused_sgd_mt_kernel(const SgdChunk*, int, short*,
                                                float*, float*, float, float,
                                                float, int);
+extern "C" __global__ void fused_adamw_mt_kernel(const SgdChunk*, int, short*,
+                                                 float*, float*, float*, float,
+                                                 float, float, float, float,
+                                                 float, float);
 extern "C" __global__ void fused_adamw_kernel(short*, float*, const short*, float*,
                                               float*, long long, float, float, float,
                                               float, float, float, float);
@@ -34,6 +38,7 @@ extern "C" __global__ void rmsnorm_bwd_kernel(const short*, const short*,
                                               const short*, const float*, short*,
                                               float*, long long, int);
 extern "C" __global__ void colsum_ws_kernel(const float*, short*, int, int);
+extern "C" __global__ void zero_f32_kernel(float*, long long);
 extern "C" __global__ void gemm_nt_bf16_kernel(const short*, const short*, short*,
                                                int, int, int, int, int, int);
 extern "C" __global__ void gemm_nt_bf16_v2_kernel(const short*, const short*,
@@ -135,6 +140,24 @@ void fused_sgd_mt(torch::Tensor table, torch::Tensor param, torch::Tensor master
                      (float)weight_decay, nesterov ? 1 : 0);
 }
 
+void fused_adamw_mt(torch::Tensor table, torch::Tensor param,
+                    torch::Tensor master, torch::Tensor exp_avg,
+                    torch::Tensor exp_avg_sq, double lr, double beta1,
+                    double beta2, double eps, double weight_decay,
+                    int64_t step) {
+  CHECK_HIP(table, torch::kInt64);
+  int n_chunks = (int)table.size(0);
+  int grid = std::min(n_chunks, 2048);
+  float bc1 = 1.0f - powf((float)beta1, (float)step);
+  float bc2 = 1.0f - powf((float)beta2, (float)step);
+  hipLaunchKernelGGL(fused_adamw_mt_kernel, dim3(grid), dim3(256), 0,
+                     cur_stream(), (const SgdChunk*)table.data_ptr(), n_chunks,
+                     (short*)param.data_ptr(), master.data_ptr<float>(),
+                     exp_avg.data_ptr<float>(), exp_avg_sq.data_ptr<float>(),
+                     (float)lr, (float)beta1, (float)beta2, (float)eps,
+                     (float)weight_decay, bc1, bc2);
+}
+
 // ---------------------------------------------------------------- secagg
 void secagg_quantize_mask(torch::Tensor q, torch::Tensor master, double weight,
                           double scale, int64_t seed, int64_t rank,
@@ -199,8 +222,13 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto [rows, H] = rows_cols(x);
   auto dx = torch::empty_like(x);
   int grid = (int)std::min<long long>(rows, 1024);
-  auto dw_ws = torch::empty({grid, H}, x.options().dtype(torch::kFloat32));
-  auto db_ws = torch::empty({grid, H}, x.options().dtype(torch::kFloat32));
+  auto dw_ws = torch::empty({16, H}, x.options().dtype(torch::kFloat32));
+  auto db_ws = torch::empty({16, H}, x.options().dtype(torch::kFloat32));
+  int zgrid = (16 * H + 255) / 256;
+  hipLaunchKernelGGL(zero_f32_kernel, dim3(zgrid), dim3(256), 0, cur_stream(),
+                     dw_ws.data_ptr<float>(), 16LL * H);
+  hipLaunchKernelGGL(zero_f32_kernel, dim3(zgrid), dim3(256), 0, cur_stream(),
+                     db_ws.data_ptr<float>(), 16LL * H);
   hipLaunchKernelGGL(layernorm_bwd_kernel, dim3(grid), dim3(256), 0, cur_stream(),
                      (const short*)dy.data_ptr(), (const short*)x.data_ptr(),
                      (const short*)w.data_ptr(), mean.data_ptr<float>(),
@@ -210,9 +238,9 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto db = torch::empty({H}, x.options());
   int cgrid = (H + 255) / 256;
   hipLaunchKernelGGL(colsum_ws_kernel, dim3(cgrid), dim3(256), 0, cur_stream(),
-                     dw_ws.data_ptr<float>(), (short*)dw.data_ptr(), grid, H);
+                     dw_ws.data_ptr<float>(), (short*)dw.data_ptr(), 16, H);
   hipLaunchKernelGGL(colsum_ws_kernel, dim3(cgrid), dim3(256), 0, cur_stream(),
-                     db_ws.data_ptr<float>(), (short*)db.data_ptr(), grid, H);
+                     db_ws.data_ptr<float>(), (short*)db.data_ptr(), 16, H);
   return {dx, dw, db};
 }
 
@@ -237,7 +265,10 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto [rows, H] = rows_cols(x);
   auto dx = torch::empty_like(x);
   int grid = (int)std::min<long long>(rows, 1024);
-  auto dw_ws = torch::empty({grid, H}, x.options().dtype(torch::kFloat32));
+  auto dw_ws = torch::empty({16, H}, x.options().dtype(torch::kFloat32));
+  int zgrid = (16 * H + 255) / 256;
+  hipLaunchKernelGGL(zero_f32_kernel, dim3(zgrid), dim3(256), 0, cur_stream(),
+                     dw_ws.data_ptr<float>(), 16LL * H);
   hipLaunchKernelGGL(rmsnorm_bwd_kernel, dim3(grid), dim3(256), 0, cur_stream(),
                      (const short*)dy.data_ptr(), (const short*)x.data_ptr(),
                      (const short*)w.data_ptr(), rstd.data_ptr<float>(),
@@ -245,7 +276,7 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto dw = torch::empty({H}, x.options());
   int cgrid = (H + 255) / 256;
   hipLaunchKernelGGL(colsum_ws_kernel, dim3(cgrid), dim3(256), 0, cur_stream(),
-                     dw_ws.data_ptr<float>(), (short*)dw.data_ptr(), grid, H);
+                     dw_ws.data_ptr<float>(), (short*)dw.data_ptr(), 16, H);
   return {dx, dw};
 }
 
@@ -452,6 +483,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_sgd", &fused_sgd, "fused SGD-momentum on flat buffers");
   m.def("fused_adamw", &fused_adamw, "fused AdamW on flat buffers");
   m.def("fused_sgd_mt", &fused_sgd_mt, "multi-tensor SGD, scattered grads");
+  m.def("fused_adamw_mt", &fused_adamw_mt, "multi-tensor AdamW, scattered grads");
   m.def("secagg_quantize_mask", &secagg_quantize_mask,
         "fixed-point quantize + pairwise Philox masks");
   m.def("dp_scale_noise", &dp_scale_noise, "DP clip-scale + gaussian noise");

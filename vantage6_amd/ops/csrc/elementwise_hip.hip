@@ -245,3 +245,58 @@ fused_sgd_mt_kernel(const SgdChunk* __restrict__ table, int n_chunks,
     }
   }
 }
+
+extern "C" __global__ void __launch_bounds__(256)
+fused_adamw_mt_kernel(const SgdChunk* __restrict__ table, int n_chunks,
+                      short* __restrict__ param, float* __restrict__ master,
+                      float* __restrict__ exp_avg, float* __restrict__ exp_avg_sq,
+                      float lr, float beta1, float beta2, float eps,
+                      float weight_decay, float bc1, float bc2) {
+  float decay = 1.0f - lr * weight_decay;
+  float step_lr = lr / bc1;
+  float inv_bc2 = 1.0f / bc2;
+  for (int c = blockIdx.x; c < n_chunks; c += gridDim.x) {
+    SgdChunk ch = table[c];
+    for (long long i = (long long)threadIdx.x * 8; i < ch.n; i += 256 * 8) {
+      long long fi = ch.offset + i;
+      if (i + 8 <= ch.n) {
+        s16x8 g8 = *(const s16x8*)(ch.grad + i);
+        f32x4 m0 = *(const f32x4*)(master + fi), m1 = *(const f32x4*)(master + fi + 4);
+        f32x4 a0 = *(const f32x4*)(exp_avg + fi), a1 = *(const f32x4*)(exp_avg + fi + 4);
+        f32x4 s0 = *(const f32x4*)(exp_avg_sq + fi), s1 = *(const f32x4*)(exp_avg_sq + fi + 4);
+        s16x8 p8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float g = bf16_to_f32(g8[j]);
+          float m = (j < 4) ? m0[j] : m1[j - 4];
+          float a = (j < 4) ? a0[j] : a1[j - 4];
+          float s = (j < 4) ? s0[j] : s1[j - 4];
+          m *= decay;
+          a = fmaf(beta1, a, (1.0f - beta1) * g);
+          s = fmaf(beta2, s, (1.0f - beta2) * g * g);
+          float denom = sqrtf(s * inv_bc2) + eps;
+          m = fmaf(-step_lr, a / denom, m);
+          if (j < 4) { m0[j] = m; a0[j] = a; s0[j] = s; }
+          else { m1[j - 4] = m; a1[j - 4] = a; s1[j - 4] = s; }
+          p8[j] = f32_to_bf16(m);
+        }
+        *(f32x4*)(master + fi) = m0; *(f32x4*)(master + fi + 4) = m1;
+        *(f32x4*)(exp_avg + fi) = a0; *(f32x4*)(exp_avg + fi + 4) = a1;
+        *(f32x4*)(exp_avg_sq + fi) = s0; *(f32x4*)(exp_avg_sq + fi + 4) = s1;
+        *(s16x8*)(param + fi) = p8;
+      } else {
+        for (long long k = i; k < ch.n; ++k) {
+          float g = bf16_to_f32(ch.grad[k]);
+          long long f = ch.offset + k;
+          float m = master[f] * decay;
+          float a = fmaf(beta1, exp_avg[f], (1.0f - beta1) * g);
+          float s = fmaf(beta2, exp_avg_sq[f], (1.0f - beta2) * g * g);
+          float denom = sqrtf(s * inv_bc2) + eps;
+          m = fmaf(-step_lr, a / denom, m);
+          master[f] = m; exp_avg[f] = a; exp_avg_sq[f] = s;
+          param[f] = f32_to_bf16(m);
+        }
+      }
+    }
+  }
+}

@@ -139,16 +139,17 @@ layernorm_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
       }
     }
   }
-  // write this block's partials: workspace [gridDim.x][H]
-  float* dwr = dw_ws + (long long)blockIdx.x * H;
-  float* dbr = db_ws + (long long)blockIdx.x * H;
+  // accumulate into 16 contention slices [16][H] (serial column sums over
+  // [grid][H] workspaces cost 234 us/call — cross-XCD reads; same fix as BN)
+  float* dwr = dw_ws + (long long)(blockIdx.x & 15) * H;
+  float* dbr = db_ws + (long long)(blockIdx.x & 15) * H;
 #pragma unroll MAXCH
   for (int c = 0; c < MAXCH; ++c) {
     int i = threadIdx.x * 8 + c * NTHREADS * 8;
     if (i >= H) break;
     for (int j = 0; j < 8 && i + j < H; ++j) {
-      dwr[i + j] = dw_acc[c * 8 + j];
-      dbr[i + j] = db_acc[c * 8 + j];
+      atomicAdd(&dwr[i + j], dw_acc[c * 8 + j]);
+      atomicAdd(&dbr[i + j], db_acc[c * 8 + j]);
     }
   }
 }
@@ -232,16 +233,18 @@ rmsnorm_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
       }
     }
   }
-  float* dwr = dw_ws + (long long)blockIdx.x * H;
+  float* dwr = dw_ws + (long long)(blockIdx.x & 15) * H;
 #pragma unroll MAXCH
   for (int c = 0; c < MAXCH; ++c) {
     int i = threadIdx.x * 8 + c * NTHREADS * 8;
     if (i >= H) break;
-    for (int j = 0; j < 8 && i + j < H; ++j) dwr[i + j] = dw_acc[c * 8 + j];
+    for (int j = 0; j < 8 && i + j < H; ++j)
+      atomicAdd(&dwr[i + j], dw_acc[c * 8 + j]);
   }
 }
 
 // column-sum of an fp32 workspace [nblocks][H] -> bf16 [H]
+// (nblocks is 16 with the atomic-slice layout)
 extern "C" __global__ void __launch_bounds__(NTHREADS)
 colsum_ws_kernel(const float* __restrict__ ws, short* __restrict__ out,
                  int nblocks, int H) {
@@ -250,4 +253,11 @@ colsum_ws_kernel(const float* __restrict__ ws, short* __restrict__ out,
   float acc = 0.f;
   for (int b = 0; b < nblocks; ++b) acc += ws[(long long)b * H + h];
   out[h] = f32_to_bf16(acc);
+}
+
+extern "C" __global__ void __launch_bounds__(NTHREADS)
+zero_f32_kernel(float* __restrict__ p, long long n) {
+  long long i = (long long)blockIdx.x * NTHREADS + threadIdx.x;
+  long long stride = (long long)gridDim.x * NTHREADS;
+  for (; i < n; i += stride) p[i] = 0.f;
 }

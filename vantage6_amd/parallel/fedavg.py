@@ -68,8 +68,7 @@ class FedAvgTrainer:
         self.cfg = cfg
         # scattered grads (no zero+accumulate passes) on GPU unless DP
         # needs the flat grad buffer for clip/noise
-        scatter = comm.device.type == "cuda" and cfg.dp is None and \
-            cfg.optimizer == "sgd"
+        scatter = comm.device.type == "cuda" and cfg.dp is None
         self.flat = FlatParamSpace(model, device=comm.device,
                                    scatter_grads=scatter)
         self.round_idx = 0

@@ -151,6 +151,31 @@ class FlatParamSpace:
         if self.exp_avg_sq is None:
             self.exp_avg_sq = torch.zeros_like(self.master_flat)
         self._step += 1
+        if self.scatter_grads:
+            if self.param_flat.is_cuda:
+                from ..ops import hip_ops
+
+                self.refresh_sgd_table()
+                hip_ops().fused_adamw_mt(
+                    self._sgd_table, self.param_flat, self.master_flat,
+                    self.momentum_flat, self.exp_avg_sq, lr, beta1, beta2,
+                    eps, weight_decay, self._step,
+                )
+            else:
+                bc1 = 1 - beta1 ** self._step
+                bc2 = 1 - beta2 ** self._step
+                for p, offset, n, nhwc in self._specs:
+                    g = self._grad_phys(p, nhwc).reshape(-1).float()
+                    m = self.master_flat[offset : offset + n]
+                    a = self.momentum_flat[offset : offset + n]
+                    v = self.exp_avg_sq[offset : offset + n]
+                    m.mul_(1.0 - lr * weight_decay)
+                    a.mul_(beta1).add_(g, alpha=1 - beta1)
+                    v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+                    denom = (v / bc2).sqrt().add_(eps)
+                    m.addcdiv_(a, denom, value=-lr / bc1)
+                    self.param_flat[offset : offset + n].copy_(m.to(self.dtype))
+            return
         fused_optim.fused_adamw(
             self.param_flat, self.master_flat, self.grad_flat,
             self.momentum_flat, self.exp_avg_sq,
