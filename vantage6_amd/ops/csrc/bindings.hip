@@ -53,6 +53,8 @@ extern "C" __global__ void gemm_nt_bf16_v6_kernel(const short*, const short*,
                                                   short*, int, int, int);
 extern "C" __global__ void gemm_nt_bf16_v7_kernel(const short*, const short*,
                                                   short*, int, int, int);
+extern "C" __global__ void gemm_nt_bf16_v8_kernel(const short*, const short*,
+                                                  short*, int, int, int);
 extern "C" __global__ void transpose_bf16_kernel(const short*, short*, int, int);
 extern "C" __global__ void bn_fwd_partials_kernel(const short*, float*, long long,
                                                   int, int);
@@ -449,10 +451,13 @@ torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b, bool trans_a,
       // v5 (128² dbuf) otherwise
       return v ? atoi(v) : 0;
     }();
-    if ((variant == 0 || variant == 6 || variant == 7) && M % 256 == 0 &&
-        N % 256 == 0) {
+    if ((variant == 0 || variant >= 6) && M % 256 == 0 && N % 256 == 0) {
       int nwg6 = (M / 256) * (N / 256);
-      auto k6 = (variant == 6) ? gemm_nt_bf16_v6_kernel : gemm_nt_bf16_v7_kernel;
+      // default: v8 (phase-interleaved) amortizes only at deep K
+      auto k6 = (variant == 8 || (variant == 0 && K >= 2560))
+                    ? gemm_nt_bf16_v8_kernel
+                    : gemm_nt_bf16_v7_kernel;
+      if (variant == 6) k6 = gemm_nt_bf16_v6_kernel;
       hipLaunchKernelGGL(k6, dim3(nwg6), dim3(512), 0, cur_stream(),
                          (const short*)A.data_ptr(), (const short*)B.data_ptr(),
                          (short*)c.data_ptr(), M, N, K);

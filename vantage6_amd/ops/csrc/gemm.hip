@@ -744,3 +744,122 @@ gemm_nt_bf16_v7_kernel(const short* __restrict__ A, const short* __restrict__ B,
         C[(long long)row * N + ccol_base + ni * 16] = f32_to_bf16(acc[mi][ni][r]);
     }
 }
+
+// v8 — v7 geometry + the guide's phase interleave (T3/T4/T5): 4 phases per
+// K-tile, each = {stage 1 piece of tile t+1 ∥ ds_read one fragment
+// quadrant → raw barrier → MFMA×16 under setprio}; counted vmcnt(2) once
+// per K-tile (at phase 0, after issuing that phase's 2 glds). B fragments
+// are re-used across the two phases that share an N-half.
+extern "C" __global__ void __launch_bounds__(512)
+gemm_nt_bf16_v8_kernel(const short* __restrict__ A, const short* __restrict__ B,
+                       short* __restrict__ C, int M, int N, int K) {
+  __shared__ short lds[2 * 2 * 2048 * 8];  // 2 buf × (A+B)[256][64] = 128 KB
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = (wave >> 2) * 128;
+  const int wc = (wave & 3) * 64;
+  const int nwg = gridDim.x;
+  int wgid = blockIdx.x;
+  {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = wgid & 7, idx = wgid >> 3;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int ntile_n = N / 256;
+  const int row0 = (wgid / ntile_n) * 256;
+  const int col0 = (wgid % ntile_n) * 256;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int src_koff = (((lane & 7) ^ (lane >> 3)) & 7) * 8;
+  const int row_in_piece = lane >> 3;
+
+  // stage ONE (A,B) piece pair: piece index p in [0,4) per wave
+  auto stage_piece = [&](int buf, int kt, int p) {
+    const long long k0 = (long long)kt * BK;
+    char* base = (char*)lds + buf * 65536;
+    int pa = wave * 4 + p;
+    const short* srcA =
+        A + (long long)(row0 + pa * 8 + row_in_piece) * K + k0 + src_koff;
+    __builtin_amdgcn_global_load_lds(
+        (global_uint*)srcA, (lds_uint*)(base + pa * 1024), 16, 0, 0);
+    const short* srcB =
+        B + (long long)(col0 + pa * 8 + row_in_piece) * K + k0 + src_koff;
+    __builtin_amdgcn_global_load_lds(
+        (global_uint*)srcB, (lds_uint*)(base + 32768 + pa * 1024), 16, 0, 0);
+  };
+
+  const int nk = K / BK;
+#pragma unroll
+  for (int p = 0; p < 4; ++p) stage_piece(0, 0, p);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (int kt = 0; kt < nk; ++kt) {
+    int cur = kt & 1;
+    char* base = (char*)lds + cur * 65536;
+    bf16x8_t b_frag[2][2];  // [ni within half][kk]
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      const int mh = p & 1, nh = p >> 1;
+      if (kt + 1 < nk) stage_piece(cur ^ 1, kt + 1, p);
+      if (p == 0 && kt > 0) {
+        // tile t's 8 stages are everything older than this phase's 2
+        asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+      }
+      // B quadrant: read on even phases, reuse on odd
+      if (mh == 0) {
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+          for (int kk = 0; kk < 2; ++kk) {
+            int row = wc + (nh * 2 + ni) * 16 + (lane & 15);
+            int ks = kk * 4 + (lane >> 4);
+            b_frag[ni][kk] = *(const bf16x8_t*)(
+                base + 32768 + row * 128 + ((ks ^ (row & 7)) & 7) * 16);
+          }
+      }
+      bf16x8_t a_frag[4][2];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+          int row = wr + (mh * 4 + mi) * 16 + (lane & 15);
+          int ks = kk * 4 + (lane >> 4);
+          a_frag[mi][kk] = *(const bf16x8_t*)(
+              base + row * 128 + ((ks ^ (row & 7)) & 7) * 16);
+        }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+          for (int kk = 0; kk < 2; ++kk)
+            acc[mh * 4 + mi][nh * 2 + ni] =
+                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a_frag[mi][kk], b_frag[ni][kk],
+                    acc[mh * 4 + mi][nh * 2 + ni], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+    if (kt + 1 == nk) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  }
+  const int crow_base = row0 + wr + (lane >> 4) * 4;
+  const int ccol_base = col0 + wc + (lane & 15);
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = crow_base + mi * 16 + r;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        C[(long long)row * N + ccol_base + ni * 16] = f32_to_bf16(acc[mi][ni][r]);
+    }
+}
