@@ -115,7 +115,15 @@ def data(number_of_databases: int = 1):
                         f"no database uri for label {label!r} (looked for {key})"
                     )
                 dbtype = env.get(f"{label.upper()}_DATABASE_TYPE", env.get("DATABASE_TYPE", "csv"))
-                frames.append(_load_database(uri, dbtype, env))
+                frame = _load_database(uri, dbtype, env)
+                prep = env.get(f"{label.upper()}_PREPROCESSING")
+                if prep:
+                    import json as _json
+
+                    from .preprocessing import apply_preprocessing
+
+                    frame = apply_preprocessing(frame, _json.loads(prep))
+                frames.append(frame)
             return func(*frames, *args, **kwargs)
 
         wrapper._v6_wants_env = True

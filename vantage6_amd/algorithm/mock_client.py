@@ -55,6 +55,11 @@ class MockAlgorithmClient:
             label = db.get("label", "default" if i == 0 else f"db{i}")
             env[f"{label.upper()}_DATABASE_URI"] = str(db.get("database", ""))
             env[f"{label.upper()}_DATABASE_TYPE"] = db.get("type", "csv")
+            if db.get("preprocessing"):
+                import json as _json
+
+                env[f"{label.upper()}_PREPROCESSING"] = _json.dumps(
+                    db["preprocessing"])
             if label == "default":
                 env["DATABASE_URI"] = str(db.get("database", ""))
                 env["DATABASE_TYPE"] = db.get("type", "csv")

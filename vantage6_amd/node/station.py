@@ -254,6 +254,10 @@ class Station:
         requested = [d.get("label", "default") for d in task.get("databases", [])]
         if requested:
             env["USER_REQUESTED_DATABASE_LABELS"] = ",".join(requested)
+        for d in task.get("databases", []):
+            if d.get("preprocessing"):
+                env[f"{d.get('label', 'default').upper()}_PREPROCESSING"] = \
+                    json.dumps(d["preprocessing"])
         for db in self.config.databases:
             env[f"{db.label.upper()}_DATABASE_URI"] = db.uri
             env[f"{db.label.upper()}_DATABASE_TYPE"] = db.type

@@ -111,7 +111,16 @@ def _locked(fn):
     @functools.wraps(fn)
     def wrapper(self, *args, **kwargs):
         with self._lock:
-            return fn(self, *args, **kwargs)
+            try:
+                return fn(self, *args, **kwargs)
+            except Exception:
+                # an exception mid-transaction leaves the SHARED session
+                # dirty; every later call would hit PendingRollbackError
+                try:
+                    self.session.rollback()
+                except Exception:  # noqa: BLE001
+                    pass
+                raise
 
     return wrapper
 
