@@ -451,8 +451,14 @@ torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b, bool trans_a,
       // v5 (128² dbuf) otherwise
       return v ? atoi(v) : 0;
     }();
-    if ((variant == 0 || variant >= 6) && M % 256 == 0 && N % 256 == 0) {
-      int nwg6 = (M / 256) * (N / 256);
+    int nwg6_pre = (M / 256) * (N / 256);
+    // a 256² grid smaller than ~half the CUs starves the chip — the 128²
+    // tiles give 4× the workgroups (llama/bert wgrad shapes: 2048×2048
+    // outputs = 64 workgroups at 256²)
+    bool big_enough = nwg6_pre >= 128;
+    if ((variant >= 6 || (variant == 0 && big_enough)) && M % 256 == 0 &&
+        N % 256 == 0) {
+      int nwg6 = nwg6_pre;
       // default: v8 (phase-interleaved) amortizes only at deep K
       auto k6 = (variant == 8 || (variant == 0 && K >= 2560))
                     ? gemm_nt_bf16_v8_kernel
