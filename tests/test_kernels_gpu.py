@@ -383,3 +383,49 @@ def test_scatter_grads_matches_flat(ops):
     b = run(True)
     assert torch.allclose(a, b, rtol=1e-4, atol=1e-5), \
         (a - b).abs().max().item()
+
+
+def test_swiglu_fwd_bwd(ops):
+    from vantage6_amd.ops.activations import swiglu
+
+    torch.manual_seed(30)
+    g = bf16(torch.randn(64, 512)).to(DEV).requires_grad_(True)
+    u = bf16(torch.randn(64, 512)).to(DEV).requires_grad_(True)
+    y = swiglu(g, u)
+    dy = bf16(torch.randn(64, 512)).to(DEV)
+    y.backward(dy)
+    gf = g.detach().float().cpu().requires_grad_(True)
+    uf = u.detach().float().cpu().requires_grad_(True)
+    yref = torch.nn.functional.silu(gf) * uf
+    yref.backward(dy.float().cpu())
+    assert (y.float().cpu() - yref).abs().max() < 3e-2
+    assert (g.grad.float().cpu() - gf.grad).abs().max() < 3e-2
+    assert (u.grad.float().cpu() - uf.grad).abs().max() < 3e-2
+
+
+def test_rope_fwd_bwd(ops):
+    from vantage6_amd.ops.activations import rope, rope_tables, _rope_ref
+
+    torch.manual_seed(31)
+    B, H, S, D = 2, 4, 16, 64
+    x = bf16(torch.randn(B, H, S, D)).to(DEV).requires_grad_(True)
+    coss, sins = rope_tables(S, D, 500000.0, DEV)
+    y = rope(x, coss, sins)
+    dy = bf16(torch.randn(B, H, S, D)).to(DEV)
+    y.backward(dy)
+
+    xf = x.detach().float().cpu().requires_grad_(True)
+    yref = _rope_ref(xf, coss.cpu(), sins.cpu(), 1)
+    yref.backward(dy.float().cpu())
+    assert (y.float().cpu() - yref.float()).abs().max() < 3e-2
+    assert (x.grad.float().cpu() - xf.grad).abs().max() < 3e-2
+    # rotation inverse property: bwd(fwd(x)) ~= x for unit dy chains
+    z = rope(x.detach(), coss, sins)
+    zz = hip_ops_mod().rope_apply(z.contiguous(), coss, sins, -1)
+    assert (zz.float() - x.detach().float()).abs().max().item() < 3e-2
+
+
+def hip_ops_mod():
+    from vantage6_amd.ops import hip_ops
+
+    return hip_ops()

@@ -17,6 +17,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.activations import rope, rope_tables, swiglu
 from ..ops.gemm import Linear as MfmaLinear
 from ..ops.norms import LayerNorm, RMSNorm
 
@@ -65,20 +66,18 @@ class BertBase(nn.Module):
 
 
 # ----------------------------------------------------------------- Llama-1B
+_ROPE_CACHE: dict = {}
+
+
 def _rope(q, k, base: float = 500000.0):
-    # rotary embeddings, applied per head-dim pair
+    """Rotary embeddings via the hand-written HIP kernel; cos/sin tables
+    cached per (seq, head_dim, device)."""
     B, Hn, S, D = q.shape
-    half = D // 2
-    inv = 1.0 / base ** (torch.arange(half, device=q.device, dtype=torch.float32) / half)
-    t = torch.arange(S, device=q.device, dtype=torch.float32)
-    freqs = torch.outer(t, inv)  # [S, half]
-    cos, sin = freqs.cos()[None, None], freqs.sin()[None, None]
-
-    def rot(x):
-        x1, x2 = x[..., :half].float(), x[..., half:].float()
-        return torch.cat([x1 * cos - x2 * sin, x1 * sin + x2 * cos], -1).to(x.dtype)
-
-    return rot(q), rot(k)
+    key = (S, D, q.device)
+    if key not in _ROPE_CACHE:
+        _ROPE_CACHE[key] = rope_tables(S, D, base, q.device)
+    coss, sins = _ROPE_CACHE[key]
+    return rope(q, coss, sins), rope(k, coss, sins)
 
 
 class LlamaLayer(nn.Module):
@@ -107,7 +106,7 @@ class LlamaLayer(nn.Module):
         att = att.transpose(1, 2).reshape(B, S, H)
         x = x + self.o(att)
         h = self.norm2(x)
-        x = x + self.down(F.silu(self.gate(h)) * self.up(h))
+        x = x + self.down(swiglu(self.gate(h), self.up(h)))
         return x
 
 

@@ -25,6 +25,13 @@ extern "C" __global__ void secagg_quantize_mask_kernel(long long*, const float*,
 extern "C" __global__ void dp_scale_noise_kernel(short*, long long, float, float,
                                                  unsigned long long);
 extern "C" __global__ void norm2_bf16_kernel(const short*, long long, float*);
+extern "C" __global__ void swiglu_fwd_kernel(const short*, const short*, short*,
+                                             long long);
+extern "C" __global__ void rope_kernel(const short*, short*, const float*,
+                                       const float*, long long, int, int, int);
+extern "C" __global__ void swiglu_bwd_kernel(const short*, const short*,
+                                             const short*, short*, short*,
+                                             long long);
 extern "C" __global__ void layernorm_fwd_kernel(const short*, const short*,
                                                 const short*, short*, float*, float*,
                                                 long long, int, float);
@@ -190,6 +197,44 @@ torch::Tensor grad_norm2_partials(torch::Tensor grad) {
   auto out = torch::empty({blocks}, grad.options().dtype(torch::kFloat32));
   hipLaunchKernelGGL(norm2_bf16_kernel, dim3(blocks), dim3(256), 0, cur_stream(),
                      (const short*)grad.data_ptr(), n, out.data_ptr<float>());
+  return out;
+}
+
+torch::Tensor swiglu_fwd(torch::Tensor g, torch::Tensor u) {
+  CHECK_HIP(g, torch::kBFloat16);
+  auto out = torch::empty_like(g);
+  long long n = g.numel();
+  hipLaunchKernelGGL(swiglu_fwd_kernel, dim3(ew_grid((n + 7) / 8)), dim3(256),
+                     0, cur_stream(), (const short*)g.data_ptr(),
+                     (const short*)u.data_ptr(), (short*)out.data_ptr(), n);
+  return out;
+}
+
+std::vector<torch::Tensor> swiglu_bwd(torch::Tensor dout, torch::Tensor g,
+                                      torch::Tensor u) {
+  CHECK_HIP(dout, torch::kBFloat16);
+  auto dg = torch::empty_like(g);
+  auto du = torch::empty_like(u);
+  long long n = g.numel();
+  hipLaunchKernelGGL(swiglu_bwd_kernel, dim3(ew_grid((n + 7) / 8)), dim3(256),
+                     0, cur_stream(), (const short*)dout.data_ptr(),
+                     (const short*)g.data_ptr(), (const short*)u.data_ptr(),
+                     (short*)dg.data_ptr(), (short*)du.data_ptr(), n);
+  return {dg, du};
+}
+
+torch::Tensor rope_apply(torch::Tensor x, torch::Tensor coss, torch::Tensor sins,
+                         int64_t dir) {
+  CHECK_HIP(x, torch::kBFloat16);
+  int D = (int)x.size(-1);
+  int S = (int)x.size(-2);
+  long long rows = x.numel() / D;
+  auto out = torch::empty_like(x);
+  long long total = rows * (D / 2);
+  hipLaunchKernelGGL(rope_kernel, dim3(ew_grid(total)), dim3(256), 0,
+                     cur_stream(), (const short*)x.data_ptr(),
+                     (short*)out.data_ptr(), coss.data_ptr<float>(),
+                     sins.data_ptr<float>(), rows, S, D, (int)dir);
   return out;
 }
 
@@ -499,6 +544,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fixed-point quantize + pairwise Philox masks");
   m.def("dp_scale_noise", &dp_scale_noise, "DP clip-scale + gaussian noise");
   m.def("grad_norm2_partials", &grad_norm2_partials, "block partial sum of g^2");
+  m.def("rope_apply", &rope_apply);
+  m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_bwd", &swiglu_bwd);
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("rmsnorm_fwd", &rmsnorm_fwd);

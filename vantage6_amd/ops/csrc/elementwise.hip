@@ -299,3 +299,98 @@ fused_adamw_mt_kernel(const SgdChunk* __restrict__ table, int n_chunks,
     }
   }
 }
+
+// ------------------------------------------------------ fused SwiGLU (llama)
+// fwd: out = silu(g) * u           (2 reads + 1 write vs 3R+2W eager)
+// bwd: dg = dout * u * silu'(g); du = dout * silu(g)
+extern "C" __global__ void __launch_bounds__(256)
+swiglu_fwd_kernel(const short* __restrict__ g, const short* __restrict__ u,
+                  short* __restrict__ out, long long n) {
+  long long i0 = ((long long)blockIdx.x * 256 + threadIdx.x) * 8;
+  long long stride = (long long)gridDim.x * 256 * 8;
+  for (long long i = i0; i < n; i += stride) {
+    if (i + 8 <= n) {
+      s16x8 gv = *(const s16x8*)(g + i);
+      s16x8 uv = *(const s16x8*)(u + i);
+      s16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float x = bf16_to_f32(gv[j]);
+        float s = x / (1.0f + __expf(-x));
+        o[j] = f32_to_bf16(s * bf16_to_f32(uv[j]));
+      }
+      *(s16x8*)(out + i) = o;
+    } else {
+      for (long long k = i; k < n; ++k) {
+        float x = bf16_to_f32(g[k]);
+        float s = x / (1.0f + __expf(-x));
+        out[k] = f32_to_bf16(s * bf16_to_f32(u[k]));
+      }
+    }
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+swiglu_bwd_kernel(const short* __restrict__ dout, const short* __restrict__ g,
+                  const short* __restrict__ u, short* __restrict__ dg,
+                  short* __restrict__ du, long long n) {
+  long long i0 = ((long long)blockIdx.x * 256 + threadIdx.x) * 8;
+  long long stride = (long long)gridDim.x * 256 * 8;
+  for (long long i = i0; i < n; i += stride) {
+    long long lim = (i + 8 <= n) ? 8 : (n - i);
+    if (lim == 8) {
+      s16x8 dv = *(const s16x8*)(dout + i);
+      s16x8 gv = *(const s16x8*)(g + i);
+      s16x8 uv = *(const s16x8*)(u + i);
+      s16x8 odg, odu;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float x = bf16_to_f32(gv[j]);
+        float sig = 1.0f / (1.0f + __expf(-x));
+        float s = x * sig;
+        float ds = sig * (1.0f + x * (1.0f - sig));  // d silu(x)/dx
+        float d = bf16_to_f32(dv[j]);
+        odg[j] = f32_to_bf16(d * bf16_to_f32(uv[j]) * ds);
+        odu[j] = f32_to_bf16(d * s);
+      }
+      *(s16x8*)(dg + i) = odg;
+      *(s16x8*)(du + i) = odu;
+    } else {
+      for (long long k = i; k < n; ++k) {
+        float x = bf16_to_f32(g[k]);
+        float sig = 1.0f / (1.0f + __expf(-x));
+        float s = x * sig;
+        float ds = sig * (1.0f + x * (1.0f - sig));
+        float d = bf16_to_f32(dout[k]);
+        dg[k] = f32_to_bf16(d * bf16_to_f32(u[k]) * ds);
+        du[k] = f32_to_bf16(d * s);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------- RoPE
+// x [*, S, D] bf16, cos/sin [S, D/2] fp32 (precomputed once).
+// out1 = x1*cos ∓ x2*sin ; out2 = x1*sin ± x2*cos  (dir=-1 is the inverse
+// rotation = the backward pass). Grid-stride over rows of D.
+extern "C" __global__ void __launch_bounds__(256)
+rope_kernel(const short* __restrict__ x, short* __restrict__ out,
+            const float* __restrict__ coss, const float* __restrict__ sins,
+            long long rows, int S, int D, int dir) {
+  int half = D >> 1;
+  long long total = rows * half;  // one thread per rotation pair
+  long long i0 = (long long)blockIdx.x * 256 + threadIdx.x;
+  long long stride = (long long)gridDim.x * 256;
+  for (long long i = i0; i < total; i += stride) {
+    long long row = i / half;
+    int d = (int)(i % half);
+    int s = (int)(row % S);
+    float c = coss[(long long)s * half + d];
+    float sn = sins[(long long)s * half + d] * dir;
+    long long base = row * D;
+    float x1 = bf16_to_f32(x[base + d]);
+    float x2 = bf16_to_f32(x[base + half + d]);
+    out[base + d] = f32_to_bf16(x1 * c - x2 * sn);
+    out[base + half + d] = f32_to_bf16(x1 * sn + x2 * c);
+  }
+}
