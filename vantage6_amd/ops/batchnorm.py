@@ -26,12 +26,12 @@ class _FusedBNFn(torch.autograd.Function):
             if residual is not None:
                 res2 = (residual.contiguous(memory_format=torch.channels_last)
                         .permute(0, 2, 3, 1).reshape(-1, C))
-            y2, mean, rstd = hip_ops().bn_fwd(
+            y2, mean, rstd, msk = hip_ops().bn_fwd(
                 x2, weight, bias, running_mean, running_var, training,
                 momentum, eps, relu, res2, ctx.fwd_acc,
             )
             y = y2.view(N, H, W, C).permute(0, 3, 1, 2)
-            ctx.save_for_backward(x2, y2, weight, mean, rstd)
+            ctx.save_for_backward(x2, msk, weight, mean, rstd)
             ctx.meta = (relu, residual is not None, (N, C, H, W), True)
             return y
         # ---- CPU reference (fp32 math)
@@ -56,13 +56,13 @@ class _FusedBNFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        x, y, weight, mean, rstd = ctx.saved_tensors
+        x, y_or_mask, weight, mean, rstd = ctx.saved_tensors
         relu, has_res, shape, on_gpu = ctx.meta
         if on_gpu:
             N, C, H, W = shape
             dy2 = (dy.contiguous(memory_format=torch.channels_last)
                    .permute(0, 2, 3, 1).reshape(-1, C))
-            outs = hip_ops().bn_bwd(dy2, y, x, weight, mean, rstd, relu,
+            outs = hip_ops().bn_bwd(dy2, y_or_mask, x, weight, mean, rstd, relu,
                                     has_res, ctx.bwd_acc)
             dx2, dw, db = outs[0], outs[1], outs[2]
             dx = dx2.view(N, H, W, C).permute(0, 3, 1, 2)
@@ -70,7 +70,8 @@ class _FusedBNFn(torch.autograd.Function):
             if has_res:
                 dres = outs[3].view(N, H, W, C).permute(0, 3, 1, 2)
             return dx, dw, db, None, None, None, None, None, None, dres, None, None
-        # ---- CPU reference
+        # ---- CPU reference (y_or_mask holds y on CPU)
+        y = y_or_mask
         dyf = dy.float()
         if relu:
             dyf = dyf * (y.float() > 0)

@@ -71,16 +71,18 @@ extern "C" __global__ void bn_fwd_finalize_kernel(const float*, int, int, long l
                                                   float*, float*, float*, float*,
                                                   float*, float, float, int);
 extern "C" __global__ void bn_fwd_apply_kernel(const short*, short*, const short*,
-                                               const float*, const float*, long long,
-                                               int, int);
-extern "C" __global__ void bn_bwd_partials_kernel(const short*, const short*,
+                                               const float*, const float*,
+                                               unsigned char*, long long, int, int);
+extern "C" __global__ void bn_bwd_partials_kernel(const short*,
+                                                  const unsigned char*,
                                                   const short*, const float*,
                                                   const float*, float*, long long,
                                                   int, int, int);
 extern "C" __global__ void bn_bwd_finalize_kernel(const float*, int, int, long long,
                                                   const short*, const float*, short*,
                                                   short*, float*, float*, float*);
-extern "C" __global__ void bn_bwd_apply_kernel(const short*, const short*,
+extern "C" __global__ void bn_bwd_apply_kernel(const short*,
+                                               const unsigned char*,
                                                const short*, const float*,
                                                const float*, const float*,
                                                const float*, const float*, short*,
@@ -382,16 +384,24 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor w,
     CHECK_HIP(residual.value(), torch::kBFloat16);
     res_ptr = (const short*)residual->data_ptr();
   }
+  torch::Tensor msk;
+  unsigned char* mask_ptr = nullptr;
+  if (relu) {  // relu bitmask: backward never re-reads y
+    msk = torch::empty({R * C / 8}, x.options().dtype(torch::kUInt8));
+    mask_ptr = (unsigned char*)msk.data_ptr();
+  } else {
+    msk = torch::empty({0}, x.options().dtype(torch::kUInt8));
+  }
   int agrid = (int)std::max<long long>(
       1, std::min<long long>((R + rows_pb - 1) / rows_pb, 4096));
   hipLaunchKernelGGL(bn_fwd_apply_kernel, dim3(agrid), dim3(256), 0,
                      cur_stream(), (const short*)x.data_ptr(),
                      (short*)y.data_ptr(), res_ptr, scale.data_ptr<float>(),
-                     shift.data_ptr<float>(), R, C, relu ? 1 : 0);
-  return {y, mean, rstd};
+                     shift.data_ptr<float>(), mask_ptr, R, C, relu ? 1 : 0);
+  return {y, mean, rstd, msk};
 }
 
-std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
+std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor mask,
                                   torch::Tensor x, torch::Tensor w,
                                   torch::Tensor mean, torch::Tensor rstd,
                                   bool relu, bool want_dres,
@@ -413,7 +423,8 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
   }
   hipLaunchKernelGGL(bn_bwd_partials_kernel, dim3(nblocks), dim3(256), 0,
                      cur_stream(), (const short*)dy.data_ptr(),
-                     (const short*)y.data_ptr(), (const short*)x.data_ptr(),
+                     (const unsigned char*)mask.data_ptr(),
+                     (const short*)x.data_ptr(),
                      mean.data_ptr<float>(), rstd.data_ptr<float>(),
                      part.data_ptr<float>(), R, C, nblocks, relu ? 1 : 0);
   auto dw = torch::empty({C}, x.options());
@@ -438,7 +449,8 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
       1, std::min<long long>((R + rows_pb - 1) / rows_pb, 4096));
   hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3(agrid), dim3(256), 0,
                      cur_stream(), (const short*)dy.data_ptr(),
-                     (const short*)y.data_ptr(), (const short*)x.data_ptr(),
+                     (const unsigned char*)mask.data_ptr(),
+                     (const short*)x.data_ptr(),
                      mean.data_ptr<float>(), rstd.data_ptr<float>(),
                      c_scale.data_ptr<float>(), c_db.data_ptr<float>(),
                      c_dw.data_ptr<float>(), (short*)dx.data_ptr(), dres_ptr, R,

@@ -140,7 +140,8 @@ extern "C" __global__ void __launch_bounds__(BN_NTH)
 bn_fwd_apply_kernel(const short* __restrict__ x, short* __restrict__ y,
                     const short* __restrict__ residual,
                     const float* __restrict__ scale,
-                    const float* __restrict__ shift, long long R, int C,
+                    const float* __restrict__ shift,
+                    unsigned char* __restrict__ mask, long long R, int C,
                     int relu) {
   int chunks, rows_pb, j, rr;
   bn_thread_map(C, &chunks, &rows_pb, &j, &rr);
@@ -156,30 +157,33 @@ bn_fwd_apply_kernel(const short* __restrict__ x, short* __restrict__ y,
     long long off = row * C + j * 8;
     s16x8 v = *(const s16x8*)(x + off);
     s16x8 o;
+    unsigned int m = 0;
     if (residual) {
       s16x8 rv = *(const s16x8*)(residual + off);
 #pragma unroll
       for (int q = 0; q < 8; ++q) {
         float f = fmaf(bf16_to_f32(v[q]), sc[q], sh[q]) + bf16_to_f32(rv[q]);
-        if (relu && f < 0.f) f = 0.f;
+        if (relu && f < 0.f) f = 0.f; else m |= (1u << q);
         o[q] = f32_to_bf16(f);
       }
     } else {
 #pragma unroll
       for (int q = 0; q < 8; ++q) {
         float f = fmaf(bf16_to_f32(v[q]), sc[q], sh[q]);
-        if (relu && f < 0.f) f = 0.f;
+        if (relu && f < 0.f) f = 0.f; else m |= (1u << q);
         o[q] = f32_to_bf16(f);
       }
     }
     *(s16x8*)(y + off) = o;
+    if (mask) mask[off >> 3] = (unsigned char)m;  // byte per 8-chunk
   }
 }
 
 // ---------------------------------------------------------- bwd: partials
 // dyr = relu ? (y>0 ? dy : 0) : dy ; partials of db=Σdyr, dws=Σdyr*xhat
 extern "C" __global__ void __launch_bounds__(BN_NTH)
-bn_bwd_partials_kernel(const short* __restrict__ dy, const short* __restrict__ y,
+bn_bwd_partials_kernel(const short* __restrict__ dy,
+                       const unsigned char* __restrict__ mask,
                        const short* __restrict__ x,
                        const float* __restrict__ mean,
                        const float* __restrict__ rstd, float* __restrict__ part,
@@ -200,13 +204,10 @@ bn_bwd_partials_kernel(const short* __restrict__ dy, const short* __restrict__ y
       long long off = row * C + j * 8;
       s16x8 dv = *(const s16x8*)(dy + off);
       s16x8 xv = *(const s16x8*)(x + off);
+      unsigned int m = relu ? mask[off >> 3] : 0xFFu;
 #pragma unroll
       for (int q = 0; q < 8; ++q) {
-        float d = bf16_to_f32(dv[q]);
-        if (relu) {
-          float yy = bf16_to_f32(((const short*)(y + off))[q]);
-          if (yy <= 0.f) d = 0.f;
-        }
+        float d = (m >> q) & 1u ? bf16_to_f32(dv[q]) : 0.f;
         float xh = (bf16_to_f32(xv[q]) - mu[q]) * rs[q];
         sdb[q] += d;
         sdw[q] = fmaf(d, xh, sdw[q]);
@@ -268,7 +269,8 @@ bn_bwd_finalize_kernel(const float* __restrict__ part, int nblocks, int C,
 // ------------------------------------------------------------ bwd: apply
 // dx = c_scale*(dyr − c_db − xhat*c_dw); optional dresidual = dyr
 extern "C" __global__ void __launch_bounds__(BN_NTH)
-bn_bwd_apply_kernel(const short* __restrict__ dy, const short* __restrict__ y,
+bn_bwd_apply_kernel(const short* __restrict__ dy,
+                    const unsigned char* __restrict__ mask,
                     const short* __restrict__ x, const float* __restrict__ mean,
                     const float* __restrict__ rstd,
                     const float* __restrict__ c_scale,
@@ -291,24 +293,13 @@ bn_bwd_apply_kernel(const short* __restrict__ dy, const short* __restrict__ y,
     s16x8 dv = *(const s16x8*)(dy + off);
     s16x8 xv = *(const s16x8*)(x + off);
     s16x8 odx, odr;
-    if (relu) {
-      s16x8 yv = *(const s16x8*)(y + off);
+    unsigned int m = relu ? mask[off >> 3] : 0xFFu;
 #pragma unroll
-      for (int q = 0; q < 8; ++q) {
-        float d = bf16_to_f32(dv[q]);
-        if (bf16_to_f32(yv[q]) <= 0.f) d = 0.f;
-        float xh = (bf16_to_f32(xv[q]) - mu[q]) * rs[q];
-        odx[q] = f32_to_bf16(cs[q] * (d - cdb[q] - xh * cdw[q]));
-        odr[q] = f32_to_bf16(d);
-      }
-    } else {
-#pragma unroll
-      for (int q = 0; q < 8; ++q) {
-        float d = bf16_to_f32(dv[q]);
-        float xh = (bf16_to_f32(xv[q]) - mu[q]) * rs[q];
-        odx[q] = f32_to_bf16(cs[q] * (d - cdb[q] - xh * cdw[q]));
-        odr[q] = f32_to_bf16(d);
-      }
+    for (int q = 0; q < 8; ++q) {
+      float d = (m >> q) & 1u ? bf16_to_f32(dv[q]) : 0.f;
+      float xh = (bf16_to_f32(xv[q]) - mu[q]) * rs[q];
+      odx[q] = f32_to_bf16(cs[q] * (d - cdb[q] - xh * cdw[q]));
+      odr[q] = f32_to_bf16(d);
     }
     *(s16x8*)(dx + off) = odx;
     if (dres) *(s16x8*)(dres + off) = odr;
