@@ -62,6 +62,8 @@ extern "C" __global__ void gemm_nt_bf16_v7_kernel(const short*, const short*,
                                                   short*, int, int, int);
 extern "C" __global__ void gemm_nt_bf16_v8_kernel(const short*, const short*,
                                                   short*, int, int, int);
+extern "C" __global__ void gemm_wgrad_bf16_kernel(const short*, const short*,
+                                                  short*, int, int, int);
 extern "C" __global__ void transpose_bf16_kernel(const short*, short*, int, int);
 extern "C" __global__ void bn_fwd_partials_kernel(const short*, float*, long long,
                                                   int, int);
@@ -471,12 +473,41 @@ torch::Tensor transpose_bf16(torch::Tensor a) {
   return out;
 }
 
+torch::Tensor gemm_wgrad(torch::Tensor a, torch::Tensor b) {
+  // C[Na,Nb] = A^T · B for A[M,Na], B[M,Nb] (reduction-strided operands,
+  // tr_b16 fragment reads — no transpose pre-passes)
+  CHECK_HIP(a, torch::kBFloat16);
+  CHECK_HIP(b, torch::kBFloat16);
+  int M = (int)a.size(0), Na = (int)a.size(1), Nb = (int)b.size(1);
+  TORCH_CHECK((int)b.size(0) == M, "reduction dims mismatch");
+  TORCH_CHECK(M % 64 == 0 && Na % 128 == 0 && Nb % 128 == 0,
+              "gemm_wgrad needs M%64==0, Na/Nb%128==0");
+  auto c = torch::empty({Na, Nb}, a.options());
+  int nwg = (Na / 128) * (Nb / 128);
+  hipLaunchKernelGGL(gemm_wgrad_bf16_kernel, dim3(nwg), dim3(256), 0,
+                     cur_stream(), (const short*)a.data_ptr(),
+                     (const short*)b.data_ptr(), (short*)c.data_ptr(), M, Na,
+                     Nb);
+  return c;
+}
+
 // C[M,N] = op(A)·op(B); inputs 2-D bf16 contiguous.
 torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b, bool trans_a,
                         bool trans_b) {
   CHECK_HIP(a, torch::kBFloat16);
   CHECK_HIP(b, torch::kBFloat16);
   TORCH_CHECK(a.dim() == 2 && b.dim() == 2, "gemm_bf16 expects 2-D");
+  if (trans_a && !trans_b) {
+    // wgrad form: A^T·B with both operands reduction-strided
+    int M_ = (int)a.size(0), Na_ = (int)a.size(1), Nb_ = (int)b.size(1);
+    static bool use_wgrad = [] {
+      const char* v = getenv("V6AMD_WGRAD_KERNEL");
+      return v ? atoi(v) != 0 : true;
+    }();
+    if (use_wgrad && M_ % 64 == 0 && Na_ % 128 == 0 && Nb_ % 128 == 0 &&
+        (long long)M_ * Na_ > 65536)
+      return gemm_wgrad(a, b);
+  }
   // normalize to NT: C[M,N] = A'[M,K] · B'[N,K]^T
   torch::Tensor A = trans_a ? transpose_bf16(a) : a;
   torch::Tensor B = trans_b ? b : transpose_bf16(b);
@@ -566,5 +597,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd", &bn_fwd, "fused NHWC BatchNorm fwd (+relu/+residual)");
   m.def("bn_bwd", &bn_bwd, "fused NHWC BatchNorm bwd");
   m.def("gemm_bf16", &gemm_bf16, "bf16 MFMA GEMM, fp32 accumulate");
+  m.def("gemm_wgrad", &gemm_wgrad, "A^T*B via tr_b16 reads (no transposes)");
   m.def("transpose_bf16", &transpose_bf16);
 }

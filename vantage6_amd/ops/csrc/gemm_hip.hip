@@ -864,3 +864,163 @@ gemm_nt_bf16_v8_kernel(const short* __restrict__ A, const short* __restrict__ B,
         C[(long long)row * N + ccol_base + ni * 16] = f32_to_bf16(acc[mi][ni][r]);
     }
 }
+
+// ===========================================================================
+// v9 — wgrad kernel: C[Na,Nb] = A^T·B for A[M,Na], B[M,Nb] (both row-major,
+// reduction = M). Both operands are reduction-STRIDED in memory; tiles are
+// staged in [m/4][out/16][4][16] bf16 subtiles (glds full-line coalesced,
+// lane-linear) and fragments are gathered with the gfx950 hardware
+// transpose read ds_read_b64_tr_b16 (guide T10): lane l, elem j reads
+// lds[(l&15) + j*16 + (l>>4)*64] relative to a per-lane base — one read
+// yields 4 consecutive reduction elements of the lane's output column.
+// Replaces the transpose_bf16(dY) + transpose_bf16(X) pre-passes.
+// Tile: 128×128 out, 64 reduction/step, 4 waves, double-buffered.
+// ===========================================================================
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4_t;
+
+DEVINL bf16x4_t tr_read_b64_issue(const char* addr) {
+  // issue WITHOUT a wait — caller batches one lgkmcnt(0) before the first
+  // consumer (guide §5.7 form iii; a wait per read serializes the 16-read
+  // fragment gather)
+  unsigned a32 =
+      (unsigned)(__SIZE_TYPE__)(__attribute__((address_space(3))) const char*)addr;
+  bf16x4_t out;
+  asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(out) : "v"(a32) : "memory");
+  return out;
+}
+
+DEVINL void tr_read_drain() {
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_sched_barrier(0);  // rule 18: MFMAs must not hoist past
+}
+
+extern "C" __global__ void __launch_bounds__(NTH)
+gemm_wgrad_bf16_kernel(const short* __restrict__ A, const short* __restrict__ B,
+                       short* __restrict__ C, int M, int Na, int Nb) {
+  // LDS: 2 buf × 2 ops × [64][128] bf16 subtiled = 64 KB
+  __shared__ short lds[2 * 2 * 8192];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = (wave >> 1) * 64;  // output rows (A columns)
+  const int wc = (wave & 1) * 64;   // output cols (B columns)
+  const int ntile_n = Nb / BN;
+  int wgid = blockIdx.x;
+  {
+    int nwg = gridDim.x;
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = wgid & 7, idx = wgid >> 3;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int row0 = (wgid / ntile_n) * BM;  // Na offset
+  const int col0 = (wgid % ntile_n) * BN;  // Nb offset
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // glds source pattern: one 1-KB instruction = one 4-row m-group × 128
+  // out cols in subtile order; lane l fetches (m = mb + ((l&7)>>1),
+  // col = (l>>3)*16 + (l&1)*8), 16 B each → 4 full 256-B rows/instr.
+  const int src_m_in_grp = (lane & 7) >> 1;
+  const int src_col = ((lane >> 3) * 16 + (lane & 1) * 8);
+
+  auto stage = [&](int buf, int mt) {
+    const long long m0 = (long long)mt * BK;
+    char* base = (char*)lds + buf * 32768;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int grp = wave * 4 + i;  // 16 m-groups of 4 cover 64 reduction rows
+      const short* srcA =
+          A + (m0 + grp * 4 + src_m_in_grp) * (long long)Na + row0 + src_col;
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcA, (lds_uint*)(base + grp * 1024), 16, 0, 0);
+      const short* srcB =
+          B + (m0 + grp * 4 + src_m_in_grp) * (long long)Nb + col0 + src_col;
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcB, (lds_uint*)(base + 16384 + grp * 1024), 16, 0, 0);
+    }
+  };
+
+  const int nk = M / BK;
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (int mt = 0; mt < nk; ++mt) {
+    int cur = mt & 1;
+    if (mt + 1 < nk) {
+      stage(cur ^ 1, mt + 1);
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    char* base = (char*)lds + cur * 32768;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      // within-buffer per-lane tr-read base for reduction group
+      // m-group index g = kk*8 + (lane>>4)*2 + off4 ; subtile (g, out>>4)
+      bf16x8_t a_frag[4], b_frag[4];
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi) {
+        int colA = wr + fi * 16;  // output-row block (Na side)
+        bf16x4_t lo, hi;
+        {
+          // per-lane addr = subtile base + (lane&15)*8B: the 16-lane group
+          // collectively loads the [4][16] subtile; HW returns lane l elem j
+          // = data[l + 16j] = (row j, col l) — the transpose we need
+          int g0 = kk * 8 + (lane >> 4) * 2;
+          const char* pa = base + ((g0 * 8 + (colA >> 4)) * 64) * 2 +
+                           (lane & 15) * 8;
+          lo = tr_read_b64_issue(pa);
+          const char* pa2 = base + (((g0 + 1) * 8 + (colA >> 4)) * 64) * 2 +
+                            (lane & 15) * 8;
+          hi = tr_read_b64_issue(pa2);
+        }
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          a_frag[fi][e] = lo[e];
+          a_frag[fi][e + 4] = hi[e];
+        }
+        int colB = wc + fi * 16;
+        {
+          int g0 = kk * 8 + (lane >> 4) * 2;
+          const char* pb = base + 16384 + ((g0 * 8 + (colB >> 4)) * 64) * 2 +
+                           (lane & 15) * 8;
+          lo = tr_read_b64_issue(pb);
+          const char* pb2 = base + 16384 +
+                            (((g0 + 1) * 8 + (colB >> 4)) * 64) * 2 +
+                            (lane & 15) * 8;
+          hi = tr_read_b64_issue(pb2);
+        }
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          b_frag[fi][e] = lo[e];
+          b_frag[fi][e + 4] = hi[e];
+        }
+      }
+      tr_read_drain();
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+  const int crow_base = row0 + wr + (lane >> 4) * 4;
+  const int ccol_base = col0 + wc + (lane & 15);
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = crow_base + mi * 16 + r;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        C[(long long)row * Nb + ccol_base + ni * 16] = f32_to_bf16(acc[mi][ni][r]);
+    }
+}
