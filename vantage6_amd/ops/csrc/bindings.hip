@@ -64,6 +64,8 @@ extern "C" __global__ void gemm_nt_bf16_v8_kernel(const short*, const short*,
                                                   short*, int, int, int);
 extern "C" __global__ void gemm_wgrad_bf16_kernel(const short*, const short*,
                                                   short*, int, int, int);
+extern "C" __global__ void gemm_dgrad_bf16_kernel(const short*, const short*,
+                                                  short*, int, int, int);
 extern "C" __global__ void transpose_bf16_kernel(const short*, short*, int, int);
 extern "C" __global__ void bn_fwd_partials_kernel(const short*, float*, long long,
                                                   int, int);
@@ -491,12 +493,38 @@ torch::Tensor gemm_wgrad(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
+torch::Tensor gemm_dgrad(torch::Tensor a, torch::Tensor b) {
+  // C[M,K] = A[M,R]·B[R,K]: B reduction-strided via tr_b16 (no transpose)
+  CHECK_HIP(a, torch::kBFloat16);
+  CHECK_HIP(b, torch::kBFloat16);
+  int M = (int)a.size(0), R = (int)a.size(1), K = (int)b.size(1);
+  TORCH_CHECK((int)b.size(0) == R, "inner dims mismatch");
+  TORCH_CHECK(M % 128 == 0 && K % 128 == 0 && R % 64 == 0,
+              "gemm_dgrad needs M/K%128==0, R%64==0");
+  auto c = torch::empty({M, K}, a.options());
+  int nwg = (M / 128) * (K / 128);
+  hipLaunchKernelGGL(gemm_dgrad_bf16_kernel, dim3(nwg), dim3(256), 0,
+                     cur_stream(), (const short*)a.data_ptr(),
+                     (const short*)b.data_ptr(), (short*)c.data_ptr(), M, R, K);
+  return c;
+}
+
 // C[M,N] = op(A)·op(B); inputs 2-D bf16 contiguous.
 torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b, bool trans_a,
                         bool trans_b) {
   CHECK_HIP(a, torch::kBFloat16);
   CHECK_HIP(b, torch::kBFloat16);
   TORCH_CHECK(a.dim() == 2 && b.dim() == 2, "gemm_bf16 expects 2-D");
+  if (!trans_a && !trans_b) {
+    // dgrad form: A·B with B reduction-strided
+    int M_ = (int)a.size(0), R_ = (int)a.size(1), K_ = (int)b.size(1);
+    static bool use_dgrad = [] {
+      const char* v = getenv("V6AMD_DGRAD_KERNEL");
+      return v ? atoi(v) != 0 : true;
+    }();
+    if (use_dgrad && M_ % 128 == 0 && K_ % 128 == 0 && R_ % 64 == 0)
+      return gemm_dgrad(a, b);
+  }
   if (trans_a && !trans_b) {
     // wgrad form: A^T·B with both operands reduction-strided
     int M_ = (int)a.size(0), Na_ = (int)a.size(1), Nb_ = (int)b.size(1);
@@ -598,5 +626,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_bwd", &bn_bwd, "fused NHWC BatchNorm bwd");
   m.def("gemm_bf16", &gemm_bf16, "bf16 MFMA GEMM, fp32 accumulate");
   m.def("gemm_wgrad", &gemm_wgrad, "A^T*B via tr_b16 reads (no transposes)");
+  m.def("gemm_dgrad", &gemm_dgrad, "A*B with strided B via tr_b16 reads");
   m.def("transpose_bf16", &transpose_bf16);
 }

@@ -1024,3 +1024,122 @@ gemm_wgrad_bf16_kernel(const short* __restrict__ A, const short* __restrict__ B,
         C[(long long)row * Nb + ccol_base + ni * 16] = f32_to_bf16(acc[mi][ni][r]);
     }
 }
+
+// v10 — dgrad kernel: C[M,K] = A[M,R]·B[R,K] (A reduction-contiguous like
+// the NT kernels; B reduction-strided like wgrad). A staged v4-style
+// (full-line glds + src XOR swizzle, b128 fragment reads); B staged in
+// [r/4][k/16][4][16] subtiles and gathered with tr_b16. Replaces the
+// transpose_bf16(W) pre-pass of every Linear dgrad.
+extern "C" __global__ void __launch_bounds__(NTH)
+gemm_dgrad_bf16_kernel(const short* __restrict__ A, const short* __restrict__ B,
+                       short* __restrict__ C, int M, int R, int K) {
+  __shared__ short lds[2 * 2 * 8192];  // 2 buf × (A 16KB + B 16KB)
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = (wave >> 1) * 64;
+  const int wc = (wave & 1) * 64;
+  const int ntile_n = K / BN;
+  int wgid = blockIdx.x;
+  {
+    int nwg = gridDim.x;
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = wgid & 7, idx = wgid >> 3;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int row0 = (wgid / ntile_n) * BM;
+  const int col0 = (wgid % ntile_n) * BN;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // A source (v4 pattern): rows of 128-B red-slabs, XOR-permuted kslot
+  const int a_koff = (((lane & 7) ^ (lane >> 3)) & 7) * 8;
+  const int a_row_in_piece = lane >> 3;
+  // B source (wgrad pattern): 4-row red groups × subtile order
+  const int b_m_in_grp = (lane & 7) >> 1;
+  const int b_col = ((lane >> 3) * 16 + (lane & 1) * 8);
+
+  auto stage = [&](int buf, int rt) {
+    const long long r0 = (long long)rt * BK;
+    char* base = (char*)lds + buf * 32768;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int pa = wave * 4 + i;  // A: 16 pieces of 8 rows
+      const short* srcA =
+          A + (long long)(row0 + pa * 8 + a_row_in_piece) * R + r0 + a_koff;
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcA, (lds_uint*)(base + pa * 1024), 16, 0, 0);
+      // B: 16 red-groups of 4 rows
+      const short* srcB =
+          B + (r0 + pa * 4 + b_m_in_grp) * (long long)K + col0 + b_col;
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcB, (lds_uint*)(base + 16384 + pa * 1024), 16, 0, 0);
+    }
+  };
+
+  const int nk = R / BK;
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (int rt = 0; rt < nk; ++rt) {
+    int cur = rt & 1;
+    if (rt + 1 < nk) {
+      stage(cur ^ 1, rt + 1);
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    char* base = (char*)lds + cur * 32768;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      const int ks = kk * 4 + (lane >> 4);
+      bf16x8_t a_frag[4], b_frag[4];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        int row = wr + mi * 16 + (lane & 15);
+        a_frag[mi] = *(const bf16x8_t*)(base + row * 128 +
+                                        ((ks ^ (row & 7)) & 7) * 16);
+      }
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int colB = wc + ni * 16;
+        int g0 = kk * 8 + (lane >> 4) * 2;
+        bf16x4_t lo = tr_read_b64_issue(
+            base + 16384 + ((g0 * 8 + (colB >> 4)) * 64) * 2 + (lane & 15) * 8);
+        bf16x4_t hi = tr_read_b64_issue(
+            base + 16384 + (((g0 + 1) * 8 + (colB >> 4)) * 64) * 2 +
+            (lane & 15) * 8);
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          b_frag[ni][e] = lo[e];
+          b_frag[ni][e + 4] = hi[e];
+        }
+      }
+      tr_read_drain();
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+  const int crow_base = row0 + wr + (lane >> 4) * 4;
+  const int ccol_base = col0 + wc + (lane & 15);
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = crow_base + mi * 16 + r;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        C[(long long)row * K + ccol_base + ni * 16] = f32_to_bf16(acc[mi][ni][r]);
+    }
+}
