@@ -14,9 +14,8 @@ random shard — used by the benchmarks, there is no dataset egress).
 from __future__ import annotations
 
 import functools
-import inspect
 from dataclasses import dataclass
-from typing import Any, Callable
+from typing import Callable
 
 from .exceptions import DataError, EnvironmentVariableError
 

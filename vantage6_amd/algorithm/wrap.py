@@ -13,13 +13,11 @@ algorithm standalone exactly like a reference container would.
 from __future__ import annotations
 
 import importlib
-import json
 import os
-import traceback
 from typing import Any
 
 from ..common.serialization import deserialize, serialize
-from .exceptions import AlgorithmError, InputError
+from .exceptions import InputError
 
 
 def resolve_module(image: str):

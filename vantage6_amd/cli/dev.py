@@ -9,7 +9,6 @@ and the diagnostics workload.
 """
 from __future__ import annotations
 
-import secrets
 import tempfile
 from pathlib import Path
 

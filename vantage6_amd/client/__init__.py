@@ -12,9 +12,9 @@ schema against the FastAPI façade — researcher code is identical either way.
 from __future__ import annotations
 
 import time
-from typing import Any, Optional
+from typing import Any
 
-from ..common import base64s_to_bytes, bytes_to_base64s
+from ..common import bytes_to_base64s
 from ..common.encryption import CryptorBase, RSACryptor
 from ..common.serialization import deserialize, serialize
 from ..common.task_status import has_task_failed, has_task_finished

@@ -8,7 +8,7 @@ image) instead of the reference's ``schema`` package.
 from __future__ import annotations
 
 from pathlib import Path
-from typing import Any, Optional
+from typing import Any
 
 import yaml
 from pydantic import BaseModel, Field

@@ -11,8 +11,6 @@ torch SDPA (library path). bf16 throughout.
 """
 from __future__ import annotations
 
-import math
-
 import torch
 import torch.nn as nn
 import torch.nn.functional as F

@@ -14,7 +14,6 @@ temporary folder for algorithm-code portability.
 """
 from __future__ import annotations
 
-import io
 import json
 import logging
 import queue
@@ -23,8 +22,8 @@ import threading
 import traceback
 from pathlib import Path
 
-from ..common import base64s_to_bytes, bytes_to_base64s
-from ..common.config import DatabaseConfig, StationConfig
+from ..common import bytes_to_base64s
+from ..common.config import StationConfig
 from ..common.encryption import CryptorBase, RSACryptor
 from ..common.serialization import deserialize, serialize
 from ..common.task_status import RunStatus

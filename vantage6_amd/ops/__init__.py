@@ -18,10 +18,6 @@ eager PyTorch; on CPU tensors a plain PyTorch reference implementation runs
 """
 from __future__ import annotations
 
-import os
-
-import torch
-
 _EXT = None
 _EXT_ERR: str | None = None
 

@@ -18,7 +18,7 @@ from __future__ import annotations
 
 import torch
 
-from . import has_hip_ops, hip_ops
+from . import hip_ops
 
 
 def fused_sgd(

@@ -22,7 +22,6 @@ from dataclasses import dataclass, field
 from typing import Callable, Iterator, Optional
 
 import torch
-import torch.distributed as dist
 
 from .comm import CommContext
 from .dp import DPConfig, DPEngine

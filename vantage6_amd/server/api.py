@@ -10,7 +10,7 @@ FastAPI + pydantic, speaking the same JSON shapes (SURVEY.md §8 stack note).
 """
 from __future__ import annotations
 
-from typing import Any, Optional
+from typing import Optional
 
 from fastapi import Depends, FastAPI, HTTPException, Request, WebSocket
 from pydantic import BaseModel, Field
@@ -282,7 +282,6 @@ def create_app(core: ServerCore) -> FastAPI:
     @app.websocket("/api/ws")
     async def events_ws(ws: WebSocket):
         import asyncio
-        import anyio
 
         await ws.accept()
         try:

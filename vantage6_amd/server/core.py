@@ -16,13 +16,12 @@ delivered through :class:`EventBus` callbacks/queues.
 """
 from __future__ import annotations
 
-import datetime as dt
 import json
 import logging
 import secrets
 import threading
-from dataclasses import dataclass, field
-from typing import Any, Callable, Optional
+from dataclasses import dataclass
+from typing import Callable
 
 from ..common import jwt as jwtlib
 from ..common.task_status import RunStatus, has_task_finished, task_status_from_runs
