@@ -29,6 +29,8 @@ def _load_extension():
     try:
         import importlib
 
+        import torch  # noqa: F401 — the extension links against torch libs
+
         _EXT = importlib.import_module("vantage6_amd.ops._hip_ops")
     except ImportError as exc:
         _EXT_ERR = str(exc)
