@@ -196,3 +196,27 @@ def test_external_container_via_wrap_and_http(server, net, tmp_path):
     wrap_algorithm(environ=env)
     out = deserialize(output_file.read_bytes())
     assert out["v"] == pytest.approx(10, abs=2)
+
+
+def test_role_and_rule_listings(server):
+    root = UserClient(url=server)
+    root.authenticate("root", "root")
+    roles = root.gw.request("GET", "role")
+    names = {r["name"] for r in roles["data"]}
+    assert {"Root", "Organization-Admin", "Researcher", "Viewer"} <= names
+    rules = root.gw.request("GET", "rule", params={"per_page": 500})
+    assert rules["links"]["total"] >= 200  # resources × scopes × operations
+    assert {"name", "scope", "operation"} <= set(rules["data"][0])
+
+
+def test_algorithm_store_http(server):
+    root = UserClient(url=server)
+    root.authenticate("root", "root")
+    a = root.gw.request("POST", "algorithm", json={
+        "name": "diag", "image": "algorithms.diagnostics"})
+    assert a["status"] == "submitted"
+    r = root.gw.request("POST", f"algorithm/{a['id']}/review",
+                        json={"approve": True, "comment": "ok"})
+    assert r["status"] == "approved"
+    listed = root.gw.request("GET", "algorithm", params={"status": "approved"})
+    assert any(x["image"] == "algorithms.diagnostics" for x in listed["data"])

@@ -12,15 +12,16 @@
 #define MAXCH 8
 
 DEVINL float block_reduce_sum(float v, float* red) {
-  red[threadIdx.x] = v;
-  __syncthreads();
+  // wave-level butterfly (no barriers), then one cross-wave LDS step —
+  // the 8-level barrier tree cost ~16 barriers per row (88 us/call LN bwd
+  // at H=768: mostly barrier latency, 160 of 256 threads idle)
 #pragma unroll
-  for (int s = NTHREADS / 2; s > 0; s >>= 1) {
-    if (threadIdx.x < (unsigned)s) red[threadIdx.x] += red[threadIdx.x + s];
-    __syncthreads();
-  }
-  float out = red[0];
+  for (int off = 32; off > 0; off >>= 1)
+    v += __shfl_xor(v, off, 64);
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = v;
   __syncthreads();
+  float out = red[0] + red[1] + red[2] + red[3];
+  __syncthreads();  // red[] reusable by the next reduction
   return out;
 }
 

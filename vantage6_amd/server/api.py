@@ -255,6 +255,16 @@ def create_app(core: ServerCore) -> FastAPI:
         return wrap(core.create_study, who, body.name, body.collaboration_id,
                     body.organization_ids)
 
+    @app.get("/api/role")
+    def role_list(who: Identity = Depends(ident), page: int = 1,
+                  per_page: int = 50):
+        return wrap(core.list_roles, who, page=page, per_page=per_page)
+
+    @app.get("/api/rule")
+    def rule_list(who: Identity = Depends(ident), page: int = 1,
+                  per_page: int = 200):
+        return wrap(core.list_rules, who, page=page, per_page=per_page)
+
     # ------------------------------------------------------ algorithm store
     @app.post("/api/algorithm", status_code=201)
     def algorithm_submit(body: AlgorithmSubmitBody, who: Identity = Depends(ident)):

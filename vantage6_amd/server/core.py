@@ -777,6 +777,27 @@ class ServerCore:
             raise NotFound(f"organization {org_id} not found")
         return org.public_key or ""
 
+    @_locked
+    def list_roles(self, ident: Identity, page=1, per_page=50) -> dict:
+        self._check_user_can_view_any("role", ident)
+        roles = self.session.query(Role).order_by(Role.id).all()
+        return self._paginate([
+            {"id": r.id, "name": r.name, "description": r.description,
+             "organization_id": r.organization_id,
+             "rules": [ru.id for ru in r.rules]}
+            for r in roles], page, per_page)
+
+    @_locked
+    def list_rules(self, ident: Identity, page=1, per_page=200) -> dict:
+        self._check_user_can_view_any("rule", ident)
+        from .model import Rule
+
+        rules = self.session.query(Rule).order_by(Rule.id).all()
+        return self._paginate([
+            {"id": r.id, "name": r.name, "scope": r.scope,
+             "operation": r.operation}
+            for r in rules], page, per_page)
+
     # ------------------------------------------------------------- misc / ops
     def version(self) -> dict:
         from .. import __version__
