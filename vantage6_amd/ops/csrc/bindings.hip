@@ -63,7 +63,10 @@ extern "C" __global__ void gemm_nt_bf16_v7_kernel(const short*, const short*,
 extern "C" __global__ void gemm_nt_bf16_v8_kernel(const short*, const short*,
                                                   short*, int, int, int);
 extern "C" __global__ void gemm_wgrad_bf16_kernel(const short*, const short*,
-                                                  short*, int, int, int);
+                                                  short*, float*, int, int, int,
+                                                  int);
+extern "C" __global__ void splitk_reduce_kernel(const float*, short*, long long,
+                                                int);
 extern "C" __global__ void gemm_dgrad_bf16_kernel(const short*, const short*,
                                                   short*, int, int, int);
 extern "C" __global__ void transpose_bf16_kernel(const short*, short*, int, int);
@@ -485,11 +488,28 @@ torch::Tensor gemm_wgrad(torch::Tensor a, torch::Tensor b) {
   TORCH_CHECK(M % 64 == 0 && Na % 128 == 0 && Nb % 128 == 0,
               "gemm_wgrad needs M%64==0, Na/Nb%128==0");
   auto c = torch::empty({Na, Nb}, a.options());
-  int nwg = (Na / 128) * (Nb / 128);
-  hipLaunchKernelGGL(gemm_wgrad_bf16_kernel, dim3(nwg), dim3(256), 0,
+  int tiles = (Na / 128) * (Nb / 128);
+  // split the long reduction when the tile grid starves the 256 CUs
+  int nsplit = 1;
+  while (tiles * nsplit < 384 && nsplit < 8 && (M / 64) / (nsplit * 2) >= 2)
+    nsplit *= 2;
+  if (nsplit == 1) {
+    hipLaunchKernelGGL(gemm_wgrad_bf16_kernel, dim3(tiles), dim3(256), 0,
+                       cur_stream(), (const short*)a.data_ptr(),
+                       (const short*)b.data_ptr(), (short*)c.data_ptr(),
+                       nullptr, M, Na, Nb, 1);
+    return c;
+  }
+  auto ws = torch::empty({nsplit, Na, Nb},
+                         a.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(gemm_wgrad_bf16_kernel, dim3(tiles * nsplit), dim3(256), 0,
                      cur_stream(), (const short*)a.data_ptr(),
-                     (const short*)b.data_ptr(), (short*)c.data_ptr(), M, Na,
-                     Nb);
+                     (const short*)b.data_ptr(), nullptr,
+                     ws.data_ptr<float>(), M, Na, Nb, nsplit);
+  long long n = (long long)Na * Nb;
+  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid(n)), dim3(256), 0,
+                     cur_stream(), ws.data_ptr<float>(), (short*)c.data_ptr(),
+                     n, nsplit);
   return c;
 }
 

@@ -895,7 +895,8 @@ DEVINL void tr_read_drain() {
 
 extern "C" __global__ void __launch_bounds__(NTH)
 gemm_wgrad_bf16_kernel(const short* __restrict__ A, const short* __restrict__ B,
-                       short* __restrict__ C, int M, int Na, int Nb) {
+                       short* __restrict__ C, float* __restrict__ Cws,
+                       int M, int Na, int Nb, int nsplit) {
   // LDS: 2 buf × 2 ops × [64][128] bf16 subtiled = 64 KB
   __shared__ short lds[2 * 2 * 8192];
   const int tid = threadIdx.x;
@@ -911,6 +912,8 @@ gemm_wgrad_bf16_kernel(const short* __restrict__ A, const short* __restrict__ B,
     int xcd = wgid & 7, idx = wgid >> 3;
     wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
   }
+  const int split = wgid % nsplit;  // split-K slice over the reduction
+  wgid /= nsplit;
   const int row0 = (wgid / ntile_n) * BM;  // Na offset
   const int col0 = (wgid % ntile_n) * BN;  // Nb offset
 
@@ -943,14 +946,35 @@ gemm_wgrad_bf16_kernel(const short* __restrict__ A, const short* __restrict__ B,
     }
   };
 
-  const int nk = M / BK;
-  stage(0, 0);
+  const int nk_total = M / BK;
+  const int per = (nk_total + nsplit - 1) / nsplit;
+  const int mt0 = split * per;
+  const int mt1 = (mt0 + per < nk_total) ? mt0 + per : nk_total;
+  const int nk = mt1 - mt0;
+  if (nk <= 0) {  // empty slice: still must contribute zeros
+    if (nsplit > 1) {
+      const int crow_base0 = row0 + wr + (lane >> 4) * 4;
+      const int ccol_base0 = col0 + wc + (lane & 15);
+      float* W = Cws + (long long)split * Na * Nb;
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+#pragma unroll
+          for (int ni = 0; ni < 4; ++ni)
+            W[(long long)(crow_base0 + mi * 16 + r) * Nb + ccol_base0 + ni * 16] =
+                0.f;
+    }
+    return;
+  }
+  stage(0, mt0);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
 
-  for (int mt = 0; mt < nk; ++mt) {
-    int cur = mt & 1;
-    if (mt + 1 < nk) {
+  for (int mti = 0; mti < nk; ++mti) {
+    int mt = mt0 + mti;
+    int cur = mti & 1;
+    if (mti + 1 < nk) {
       stage(cur ^ 1, mt + 1);
       asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
     } else {
@@ -1013,15 +1037,42 @@ gemm_wgrad_bf16_kernel(const short* __restrict__ A, const short* __restrict__ B,
   }
   const int crow_base = row0 + wr + (lane >> 4) * 4;
   const int ccol_base = col0 + wc + (lane & 15);
+  if (nsplit == 1) {
 #pragma unroll
-  for (int mi = 0; mi < 4; ++mi)
+    for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      int row = crow_base + mi * 16 + r;
+      for (int r = 0; r < 4; ++r) {
+        int row = crow_base + mi * 16 + r;
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni)
-        C[(long long)row * Nb + ccol_base + ni * 16] = f32_to_bf16(acc[mi][ni][r]);
-    }
+        for (int ni = 0; ni < 4; ++ni)
+          C[(long long)row * Nb + ccol_base + ni * 16] =
+              f32_to_bf16(acc[mi][ni][r]);
+      }
+  } else {
+    float* W = Cws + (long long)split * Na * Nb;
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = crow_base + mi * 16 + r;
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          W[(long long)row * Nb + ccol_base + ni * 16] = acc[mi][ni][r];
+      }
+  }
+}
+
+// reduce split-K fp32 slabs -> bf16
+extern "C" __global__ void __launch_bounds__(256)
+splitk_reduce_kernel(const float* __restrict__ ws, short* __restrict__ out,
+                     long long n, int nsplit) {
+  long long i = (long long)blockIdx.x * 256 + threadIdx.x;
+  long long stride = (long long)gridDim.x * 256;
+  for (; i < n; i += stride) {
+    float acc = 0.f;
+    for (int s = 0; s < nsplit; ++s) acc += ws[(long long)s * n + i];
+    out[i] = f32_to_bf16(acc);
+  }
 }
 
 // v10 — dgrad kernel: C[M,K] = A[M,R]·B[R,K] (A reduction-contiguous like
