@@ -68,7 +68,8 @@ extern "C" __global__ void gemm_wgrad_bf16_kernel(const short*, const short*,
 extern "C" __global__ void splitk_reduce_kernel(const float*, short*, long long,
                                                 int);
 extern "C" __global__ void gemm_dgrad_bf16_kernel(const short*, const short*,
-                                                  short*, int, int, int);
+                                                  short*, float*, int, int, int,
+                                                  int);
 extern "C" __global__ void transpose_bf16_kernel(const short*, short*, int, int);
 extern "C" __global__ void bn_fwd_partials_kernel(const short*, float*, long long,
                                                   int, int);
@@ -522,10 +523,26 @@ torch::Tensor gemm_dgrad(torch::Tensor a, torch::Tensor b) {
   TORCH_CHECK(M % 128 == 0 && K % 128 == 0 && R % 64 == 0,
               "gemm_dgrad needs M/K%128==0, R%64==0");
   auto c = torch::empty({M, K}, a.options());
-  int nwg = (M / 128) * (K / 128);
-  hipLaunchKernelGGL(gemm_dgrad_bf16_kernel, dim3(nwg), dim3(256), 0,
+  int tiles = (M / 128) * (K / 128);
+  int nsplit = 1;
+  while (tiles * nsplit < 384 && nsplit < 8 && (R / 64) / (nsplit * 2) >= 2)
+    nsplit *= 2;
+  if (nsplit == 1) {
+    hipLaunchKernelGGL(gemm_dgrad_bf16_kernel, dim3(tiles), dim3(256), 0,
+                       cur_stream(), (const short*)a.data_ptr(),
+                       (const short*)b.data_ptr(), (short*)c.data_ptr(),
+                       nullptr, M, R, K, 1);
+    return c;
+  }
+  auto ws = torch::empty({nsplit, M, K}, a.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(gemm_dgrad_bf16_kernel, dim3(tiles * nsplit), dim3(256), 0,
                      cur_stream(), (const short*)a.data_ptr(),
-                     (const short*)b.data_ptr(), (short*)c.data_ptr(), M, R, K);
+                     (const short*)b.data_ptr(), nullptr, ws.data_ptr<float>(),
+                     M, R, K, nsplit);
+  long long n = (long long)M * K;
+  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid(n)), dim3(256), 0,
+                     cur_stream(), ws.data_ptr<float>(), (short*)c.data_ptr(),
+                     n, nsplit);
   return c;
 }
 

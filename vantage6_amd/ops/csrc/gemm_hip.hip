@@ -1083,7 +1083,8 @@ splitk_reduce_kernel(const float* __restrict__ ws, short* __restrict__ out,
 // transpose_bf16(W) pre-pass of every Linear dgrad.
 extern "C" __global__ void __launch_bounds__(NTH)
 gemm_dgrad_bf16_kernel(const short* __restrict__ A, const short* __restrict__ B,
-                       short* __restrict__ C, int M, int R, int K) {
+                       short* __restrict__ C, float* __restrict__ Cws,
+                       int M, int R, int K, int nsplit) {
   __shared__ short lds[2 * 2 * 8192];  // 2 buf × (A 16KB + B 16KB)
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -1098,6 +1099,8 @@ gemm_dgrad_bf16_kernel(const short* __restrict__ A, const short* __restrict__ B,
     int xcd = wgid & 7, idx = wgid >> 3;
     wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
   }
+  const int split = wgid % nsplit;
+  wgid /= nsplit;
   const int row0 = (wgid / ntile_n) * BM;
   const int col0 = (wgid % ntile_n) * BN;
 
@@ -1132,14 +1135,34 @@ gemm_dgrad_bf16_kernel(const short* __restrict__ A, const short* __restrict__ B,
     }
   };
 
-  const int nk = R / BK;
-  stage(0, 0);
+  const int nk_total = R / BK;
+  const int per = (nk_total + nsplit - 1) / nsplit;
+  const int rt0 = split * per;
+  const int rt1 = (rt0 + per < nk_total) ? rt0 + per : nk_total;
+  const int nk = rt1 - rt0;
+  if (nk <= 0) {
+    if (nsplit > 1) {
+      const int crow0 = row0 + wr + (lane >> 4) * 4;
+      const int ccol0 = col0 + wc + (lane & 15);
+      float* W = Cws + (long long)split * M * K;
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+#pragma unroll
+          for (int ni = 0; ni < 4; ++ni)
+            W[(long long)(crow0 + mi * 16 + r) * K + ccol0 + ni * 16] = 0.f;
+    }
+    return;
+  }
+  stage(0, rt0);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
 
-  for (int rt = 0; rt < nk; ++rt) {
-    int cur = rt & 1;
-    if (rt + 1 < nk) {
+  for (int rti = 0; rti < nk; ++rti) {
+    int rt = rt0 + rti;
+    int cur = rti & 1;
+    if (rti + 1 < nk) {
       stage(cur ^ 1, rt + 1);
       asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
     } else {
@@ -1184,13 +1207,27 @@ gemm_dgrad_bf16_kernel(const short* __restrict__ A, const short* __restrict__ B,
   }
   const int crow_base = row0 + wr + (lane >> 4) * 4;
   const int ccol_base = col0 + wc + (lane & 15);
+  if (nsplit == 1) {
 #pragma unroll
-  for (int mi = 0; mi < 4; ++mi)
+    for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      int row = crow_base + mi * 16 + r;
+      for (int r = 0; r < 4; ++r) {
+        int row = crow_base + mi * 16 + r;
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni)
-        C[(long long)row * K + ccol_base + ni * 16] = f32_to_bf16(acc[mi][ni][r]);
-    }
+        for (int ni = 0; ni < 4; ++ni)
+          C[(long long)row * K + ccol_base + ni * 16] =
+              f32_to_bf16(acc[mi][ni][r]);
+      }
+  } else {
+    float* W = Cws + (long long)split * M * K;
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = crow_base + mi * 16 + r;
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          W[(long long)row * K + ccol_base + ni * 16] = acc[mi][ni][r];
+      }
+  }
 }
