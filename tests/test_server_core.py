@@ -342,3 +342,37 @@ def test_node_heartbeat_updates_last_seen(core, root):
         assert nodes[0]["last_seen"] is not None
     finally:
         st.stop()
+
+
+class TestCustomRoles:
+    def test_create_custom_role_and_assign(self, core, root):
+        ident = root.gw.ident
+        rules = core.list_rules(ident, per_page=500)["data"]
+        task_view = [r["id"] for r in rules
+                     if r["name"] == "task" and r["operation"] == "view"]
+        role = core.create_role(ident, "TaskWatcher", task_view,
+                                description="sees tasks only")
+        assert role["rules"] == task_view
+        orgs, collab = make_collab(root, name="cr")
+        root.user.create("watcher", "pw", organization_id=orgs[0],
+                         roles=["TaskWatcher"])
+        w = UserClient(core=core)
+        w.authenticate("watcher", "pw")
+        w.task.list()  # allowed
+        with pytest.raises(PermissionError_):
+            w.task.create(collaboration=collab["id"], organizations=[orgs[0]],
+                          name="t", image="x", input_={"method": "m"})
+
+    def test_cannot_grant_unheld_rules(self, core, root):
+        ident = root.gw.ident
+        orgs, collab = make_collab(root, name="esc")
+        root.user.create("lowpriv", "pw", organization_id=orgs[0],
+                         roles=["Researcher"])
+        res = UserClient(core=core)
+        res.authenticate("lowpriv", "pw")
+        rules = core.list_rules(ident, per_page=500)["data"]
+        user_global_create = [r["id"] for r in rules
+                              if r["name"] == "user" and r["scope"] == "global"
+                              and r["operation"] == "create"]
+        with pytest.raises(PermissionError_):
+            core.create_role(res.gw.ident, "Sneaky", user_global_create)

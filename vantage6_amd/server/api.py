@@ -106,6 +106,13 @@ class AlgorithmReviewBody(BaseModel):
     comment: str = ""
 
 
+class RoleCreateBody(BaseModel):
+    name: str
+    rule_ids: list[int]
+    organization_id: Optional[int] = None
+    description: str = ""
+
+
 class RunPatchBody(BaseModel):
     status: Optional[str] = None
     result: Optional[str] = None
@@ -254,6 +261,11 @@ def create_app(core: ServerCore) -> FastAPI:
     def study_create(body: StudyCreateBody, who: Identity = Depends(ident)):
         return wrap(core.create_study, who, body.name, body.collaboration_id,
                     body.organization_ids)
+
+    @app.post("/api/role", status_code=201)
+    def role_create(body: "RoleCreateBody", who: Identity = Depends(ident)):
+        return wrap(core.create_role, who, body.name, body.rule_ids,
+                    body.organization_id, body.description)
 
     @app.get("/api/role")
     def role_list(who: Identity = Depends(ident), page: int = 1,

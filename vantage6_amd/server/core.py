@@ -778,6 +778,39 @@ class ServerCore:
         return org.public_key or ""
 
     @_locked
+    def create_role(self, ident: Identity, name: str, rule_ids: list[int],
+                    organization_id: int | None = None,
+                    description: str = "") -> dict:
+        """Custom role from a rule set (reference: role resource CREATE).
+        A user may only grant rules they hold themselves (the reference's
+        anti-escalation check)."""
+        self._check_user_can("role", Operation.CREATE, ident,
+                             org_id=organization_id)
+        s = self.session
+        own = PermissionManager(s).rules_for_user(self._user(ident))
+        from .model import Rule
+
+        rules = []
+        for rid in rule_ids:
+            rule = s.get(Rule, rid)
+            if rule is None:
+                raise NotFound(f"rule {rid} not found")
+            if (rule.name, rule.scope, rule.operation) not in own:
+                raise PermissionError_(
+                    f"cannot grant rule you do not hold: "
+                    f"{rule.name}/{rule.scope}/{rule.operation}"
+                )
+            rules.append(rule)
+        role = Role(name=name, description=description,
+                    organization_id=organization_id)
+        role.rules = rules
+        s.add(role)
+        s.commit()
+        return {"id": role.id, "name": role.name,
+                "organization_id": role.organization_id,
+                "rules": [r.id for r in role.rules]}
+
+    @_locked
     def list_roles(self, ident: Identity, page=1, per_page=50) -> dict:
         self._check_user_can_view_any("role", ident)
         roles = self.session.query(Role).order_by(Role.id).all()
