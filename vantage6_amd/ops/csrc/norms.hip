@@ -261,3 +261,152 @@ zero_f32_kernel(float* __restrict__ p, long long n) {
   long long stride = (long long)gridDim.x * NTHREADS;
   for (; i < n; i += stride) p[i] = 0.f;
 }
+
+// ---------------------------------------------------------------------------
+// Row-per-WAVE backward variants (H <= 2048): 4 rows in flight per block,
+// all reductions are wave-local __shfl_xor butterflies — no barriers at
+// all. The block-per-row kernels above are latency-bound at small H
+// (83 us/call at H=768: 96 of 256 threads active, serial rows, barrier
+// trees); these run the practical transformer widths (768/1024/2048).
+// ---------------------------------------------------------------------------
+#define WMAXCH 4  // H <= 64*8*WMAXCH = 2048
+
+DEVINL float wave_reduce_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+extern "C" __global__ void __launch_bounds__(NTHREADS)
+layernorm_bwd_wave_kernel(const short* __restrict__ dy,
+                          const short* __restrict__ x,
+                          const short* __restrict__ w,
+                          const float* __restrict__ mean,
+                          const float* __restrict__ rstd,
+                          short* __restrict__ dx, float* __restrict__ dw_ws,
+                          float* __restrict__ db_ws, long long nrows, int H) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  float dw_acc[WMAXCH * 8];
+  float db_acc[WMAXCH * 8];
+#pragma unroll
+  for (int c = 0; c < WMAXCH * 8; ++c) { dw_acc[c] = 0.f; db_acc[c] = 0.f; }
+  for (long long row = (long long)blockIdx.x * 4 + wave; row < nrows;
+       row += (long long)gridDim.x * 4) {
+    const short* xr = x + row * (long long)H;
+    const short* dyr = dy + row * (long long)H;
+    short* dxr = dx + row * (long long)H;
+    float mu = mean[row], rs = rstd[row];
+    float c1 = 0.f, c2 = 0.f;
+#pragma unroll WMAXCH
+    for (int c = 0; c < WMAXCH; ++c) {
+      int i = lane * 8 + c * 512;
+      if (i >= H) break;
+      s16x8 xv = *(const s16x8*)(xr + i);
+      s16x8 dv = *(const s16x8*)(dyr + i);
+      s16x8 wv = *(const s16x8*)(w + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xh = (bf16_to_f32(xv[j]) - mu) * rs;
+        float wdy = bf16_to_f32(wv[j]) * bf16_to_f32(dv[j]);
+        c1 += wdy;
+        c2 = fmaf(wdy, xh, c2);
+      }
+    }
+    c1 = wave_reduce_sum(c1) / H;
+    c2 = wave_reduce_sum(c2) / H;
+#pragma unroll WMAXCH
+    for (int c = 0; c < WMAXCH; ++c) {
+      int i = lane * 8 + c * 512;
+      if (i >= H) break;
+      s16x8 xv = *(const s16x8*)(xr + i);
+      s16x8 dv = *(const s16x8*)(dyr + i);
+      s16x8 wv = *(const s16x8*)(w + i);
+      s16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xh = (bf16_to_f32(xv[j]) - mu) * rs;
+        float d = bf16_to_f32(dv[j]);
+        float wdy = bf16_to_f32(wv[j]) * d;
+        o[j] = f32_to_bf16((wdy - c1 - xh * c2) * rs);
+        dw_acc[c * 8 + j] = fmaf(d, xh, dw_acc[c * 8 + j]);
+        db_acc[c * 8 + j] += d;
+      }
+      *(s16x8*)(dxr + i) = o;
+    }
+  }
+  float* dwr = dw_ws + (long long)((blockIdx.x * 4 + wave) & 15) * H;
+  float* dbr = db_ws + (long long)((blockIdx.x * 4 + wave) & 15) * H;
+#pragma unroll WMAXCH
+  for (int c = 0; c < WMAXCH; ++c) {
+    int i = lane * 8 + c * 512;
+    if (i >= H) break;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(&dwr[i + j], dw_acc[c * 8 + j]);
+      atomicAdd(&dbr[i + j], db_acc[c * 8 + j]);
+    }
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(NTHREADS)
+rmsnorm_bwd_wave_kernel(const short* __restrict__ dy,
+                        const short* __restrict__ x,
+                        const short* __restrict__ w,
+                        const float* __restrict__ rstd, short* __restrict__ dx,
+                        float* __restrict__ dw_ws, long long nrows, int H) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  float dw_acc[WMAXCH * 8];
+#pragma unroll
+  for (int c = 0; c < WMAXCH * 8; ++c) dw_acc[c] = 0.f;
+  for (long long row = (long long)blockIdx.x * 4 + wave; row < nrows;
+       row += (long long)gridDim.x * 4) {
+    const short* xr = x + row * (long long)H;
+    const short* dyr = dy + row * (long long)H;
+    short* dxr = dx + row * (long long)H;
+    float rs = rstd[row];
+    float c2 = 0.f;
+#pragma unroll WMAXCH
+    for (int c = 0; c < WMAXCH; ++c) {
+      int i = lane * 8 + c * 512;
+      if (i >= H) break;
+      s16x8 xv = *(const s16x8*)(xr + i);
+      s16x8 dv = *(const s16x8*)(dyr + i);
+      s16x8 wv = *(const s16x8*)(w + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xh = bf16_to_f32(xv[j]) * rs;
+        float wdy = bf16_to_f32(wv[j]) * bf16_to_f32(dv[j]);
+        c2 = fmaf(wdy, xh, c2);
+      }
+    }
+    c2 = wave_reduce_sum(c2) / H;
+#pragma unroll WMAXCH
+    for (int c = 0; c < WMAXCH; ++c) {
+      int i = lane * 8 + c * 512;
+      if (i >= H) break;
+      s16x8 xv = *(const s16x8*)(xr + i);
+      s16x8 dv = *(const s16x8*)(dyr + i);
+      s16x8 wv = *(const s16x8*)(w + i);
+      s16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xh = bf16_to_f32(xv[j]) * rs;
+        float d = bf16_to_f32(dv[j]);
+        float wdy = bf16_to_f32(wv[j]) * d;
+        o[j] = f32_to_bf16((wdy - xh * c2) * rs);
+        dw_acc[c * 8 + j] = fmaf(d, xh, dw_acc[c * 8 + j]);
+      }
+      *(s16x8*)(dxr + i) = o;
+    }
+  }
+  float* dwr = dw_ws + (long long)((blockIdx.x * 4 + wave) & 15) * H;
+#pragma unroll WMAXCH
+  for (int c = 0; c < WMAXCH; ++c) {
+    int i = lane * 8 + c * 512;
+    if (i >= H) break;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) atomicAdd(&dwr[i + j], dw_acc[c * 8 + j]);
+  }
+}
