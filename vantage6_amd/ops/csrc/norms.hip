@@ -335,16 +335,49 @@ layernorm_bwd_wave_kernel(const short* __restrict__ dy,
       *(s16x8*)(dxr + i) = o;
     }
   }
-  float* dwr = dw_ws + (long long)((blockIdx.x * 4 + wave) & 15) * H;
-  float* dbr = db_ws + (long long)((blockIdx.x * 4 + wave) & 15) * H;
+  // block-level LDS combine first: one global atomic set per BLOCK
+  __shared__ float comb[2 * 2048];
+  if (wave == 0) {
 #pragma unroll WMAXCH
-  for (int c = 0; c < WMAXCH; ++c) {
-    int i = lane * 8 + c * 512;
-    if (i >= H) break;
+    for (int c = 0; c < WMAXCH; ++c) {
+      int i = lane * 8 + c * 512;
+      if (i >= H) break;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      atomicAdd(&dwr[i + j], dw_acc[c * 8 + j]);
-      atomicAdd(&dbr[i + j], db_acc[c * 8 + j]);
+      for (int j = 0; j < 8; ++j) {
+        comb[i + j] = dw_acc[c * 8 + j];
+        comb[2048 + i + j] = db_acc[c * 8 + j];
+      }
+    }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int wv = 1; wv < 4; ++wv) {
+    if (wave == wv) {
+#pragma unroll WMAXCH
+      for (int c = 0; c < WMAXCH; ++c) {
+        int i = lane * 8 + c * 512;
+        if (i >= H) break;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          comb[i + j] += dw_acc[c * 8 + j];
+          comb[2048 + i + j] += db_acc[c * 8 + j];
+        }
+      }
+    }
+    __syncthreads();
+  }
+  float* dwr = dw_ws + (long long)(blockIdx.x & 15) * H;
+  float* dbr = db_ws + (long long)(blockIdx.x & 15) * H;
+  if (wave == 0) {
+#pragma unroll WMAXCH
+    for (int c = 0; c < WMAXCH; ++c) {
+      int i = lane * 8 + c * 512;
+      if (i >= H) break;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        atomicAdd(&dwr[i + j], comb[i + j]);
+        atomicAdd(&dbr[i + j], comb[2048 + i + j]);
+      }
     }
   }
 }
@@ -401,12 +434,38 @@ rmsnorm_bwd_wave_kernel(const short* __restrict__ dy,
       *(s16x8*)(dxr + i) = o;
     }
   }
-  float* dwr = dw_ws + (long long)((blockIdx.x * 4 + wave) & 15) * H;
+  __shared__ float comb[2048];
+  if (wave == 0) {
 #pragma unroll WMAXCH
-  for (int c = 0; c < WMAXCH; ++c) {
-    int i = lane * 8 + c * 512;
-    if (i >= H) break;
+    for (int c = 0; c < WMAXCH; ++c) {
+      int i = lane * 8 + c * 512;
+      if (i >= H) break;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) atomicAdd(&dwr[i + j], dw_acc[c * 8 + j]);
+      for (int j = 0; j < 8; ++j) comb[i + j] = dw_acc[c * 8 + j];
+    }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int wv = 1; wv < 4; ++wv) {
+    if (wave == wv) {
+#pragma unroll WMAXCH
+      for (int c = 0; c < WMAXCH; ++c) {
+        int i = lane * 8 + c * 512;
+        if (i >= H) break;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) comb[i + j] += dw_acc[c * 8 + j];
+      }
+    }
+    __syncthreads();
+  }
+  float* dwr = dw_ws + (long long)(blockIdx.x & 15) * H;
+  if (wave == 0) {
+#pragma unroll WMAXCH
+    for (int c = 0; c < WMAXCH; ++c) {
+      int i = lane * 8 + c * 512;
+      if (i >= H) break;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) atomicAdd(&dwr[i + j], comb[i + j]);
+    }
   }
 }
