@@ -72,7 +72,9 @@ class TestPermissions:
         root.user.create("res1", "pw", organization_id=orgs1[0], roles=["Researcher"])
         res = UserClient(core=core)
         res.authenticate("res1", "pw")
-        with pytest.raises(PermissionError_):
+        # foreign collaboration is invisible (KeyError from client lookup) or
+        # rejected server-side (PermissionError_) — either denies the task
+        with pytest.raises((PermissionError_, KeyError)):
             res.task.create(collaboration=collab2["id"], organizations=[orgs2[0]],
                             name="t", image="algorithms.column_mean",
                             input_={"method": "x"})
@@ -376,3 +378,35 @@ class TestCustomRoles:
                               and r["operation"] == "create"]
         with pytest.raises(PermissionError_):
             core.create_role(res.gw.ident, "Sneaky", user_global_create)
+
+
+class TestReadScoping:
+    def test_runs_and_results_scoped_to_collaboration(self, core, root):
+        orgs1, c1 = make_collab(root, name="sc1")
+        orgs2, c2 = make_collab(root, name="sc2")
+        root.node.create(name="sc-n0", collaboration_id=c1["id"],
+                         organization_id=orgs1[0])
+        t = root.task.create(collaboration=c1["id"], organizations=[orgs1[0]],
+                             name="secret", image="x", input_={"method": "m"})
+        root.user.create("outsider", "pw", organization_id=orgs2[0],
+                         roles=["Researcher"])
+        out = UserClient(core=core)
+        out.authenticate("outsider", "pw")
+        # an outsider must see neither the runs nor the results
+        assert out.run.from_task(t["id"]) == []
+        assert out.result.from_task(t["id"]) == []
+        runs = root.run.from_task(t["id"])  # root (GLOBAL) still sees them
+        assert len(runs) == 1
+        with pytest.raises(PermissionError_):
+            core.get_run(out.gw.ident, runs[0]["id"])
+
+    def test_organization_view_scoped(self, core, root):
+        orgs1, c1 = make_collab(root, name="ov1")
+        orgs2, c2 = make_collab(root, name="ov2")
+        root.user.create("viewer2", "pw", organization_id=orgs1[0],
+                         roles=["Researcher"])
+        v = UserClient(core=core)
+        v.authenticate("viewer2", "pw")
+        v.organization.get(orgs1[1])  # same collaboration: allowed
+        with pytest.raises(PermissionError_):
+            v.organization.get(orgs2[0])  # foreign collaboration: denied

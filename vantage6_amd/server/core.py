@@ -485,11 +485,46 @@ class ServerCore:
         if rc.allowed_scope(Operation.VIEW) is None:
             raise PermissionError_(f"user lacks view permission on {resource}")
 
+    def _org_ids_in_my_collabs(self, ident: Identity) -> set:
+        out = {ident.organization_id}
+        me = self.session.get(Organization, ident.organization_id)
+        if me is not None:
+            for c in me.collaborations:
+                out |= {o.id for o in c.organizations}
+        return out
+
+    def _check_can_see_run(self, ident: Identity, run: Run) -> bool:
+        """Scope filter for runs/results (reference: run VIEW scopes)."""
+        if ident.type == "node":
+            return run.node_id == ident.id
+        if ident.type == "container":
+            task = run.task
+            return task is not None and \
+                task.collaboration_id == ident.collaboration_id
+        rc = self._rc("run", ident)
+        scope = rc.allowed_scope(Operation.VIEW)
+        if scope is None:
+            return False
+        if scope == Scope.GLOBAL:
+            return True
+        if scope == Scope.COLLABORATION:
+            return run.organization_id in self._org_ids_in_my_collabs(ident)
+        return run.organization_id == ident.organization_id
+
     @_locked
     def get_organization(self, ident: Identity, org_id: int) -> dict:
         org = self.session.get(Organization, org_id)
         if org is None:
             raise NotFound(f"organization {org_id} not found")
+        if ident.type == "user" and org_id != ident.organization_id:
+            rc = self._rc("organization", ident)
+            scope = rc.allowed_scope(Operation.VIEW)
+            if scope is None:
+                raise PermissionError_("no view permission on organization")
+            if scope != Scope.GLOBAL and \
+                    org_id not in self._org_ids_in_my_collabs(ident):
+                raise PermissionError_(
+                    "organization outside your collaborations")
         return self._org_json(org)
 
     # ------------------------------------------------------------------ tasks
@@ -660,8 +695,7 @@ class ServerCore:
         runs = q.all()
         if state == "open":
             runs = [r for r in runs if not has_task_finished(r.status)]
-        if ident.type == "node":
-            runs = [r for r in runs if r.node_id == ident.id]
+        runs = [r for r in runs if self._check_can_see_run(ident, r)]
         return self._paginate([self._run_json(r, include_data) for r in runs], page, per_page)
 
     @_locked
@@ -669,6 +703,8 @@ class ServerCore:
         run = self.session.get(Run, run_id)
         if run is None:
             raise NotFound(f"run {run_id} not found")
+        if not self._check_can_see_run(ident, run):
+            raise PermissionError_("no view permission on this run")
         return self._run_json(run, True)
 
     @_locked
@@ -714,6 +750,7 @@ class ServerCore:
     def list_results(self, ident: Identity, task_id: int, page=1, per_page=50) -> dict:
         """GET /api/result?task_id= — result field only (v4 split result/run)."""
         runs = self.session.query(Run).filter(Run.task_id == task_id).order_by(Run.id).all()
+        runs = [r for r in runs if self._check_can_see_run(ident, r)]
         items = [
             {
                 "id": r.id,

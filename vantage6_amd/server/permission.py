@@ -69,6 +69,7 @@ DEFAULT_ROLES = {
         or (
             r.operation == Operation.VIEW.value
             and r.name in ("organization", "collaboration", "node", "study")
+            and r.scope != Scope.GLOBAL.value
         )
         or (r.name == "event" and r.scope != Scope.GLOBAL.value)
     ),
