@@ -410,3 +410,25 @@ class TestReadScoping:
         v.organization.get(orgs1[1])  # same collaboration: allowed
         with pytest.raises(PermissionError_):
             v.organization.get(orgs2[0])  # foreign collaboration: denied
+
+
+class TestLockout:
+    def test_account_locks_after_failed_logins(self):
+        core = ServerCore(max_failed_logins=3, login_lockout_seconds=3600)
+        core.create_root_user("root", "root")
+        root = UserClient(core=core)
+        root.authenticate("root", "root")
+        org = root.organization.create(name="lk-org")
+        root.user.create("locky", "goodpw", organization_id=org["id"],
+                         roles=["Researcher"])
+        c = UserClient(core=core)
+        for _ in range(3):
+            with pytest.raises(PermissionError_):
+                c.authenticate("locky", "badpw")
+        # correct password is now also refused (locked)
+        with pytest.raises(PermissionError_, match="locked"):
+            c.authenticate("locky", "goodpw")
+        # expire the lockout window and the good password works again
+        core.login_lockout_seconds = 0.0
+        c.authenticate("locky", "goodpw")
+        core.close()

@@ -132,7 +132,11 @@ class ServerCore:
         uri: str = "sqlite://",
         jwt_secret: str | None = None,
         token_expires_seconds: float = 21600.0,
+        max_failed_logins: int = 5,
+        login_lockout_seconds: float = 3600.0,
     ):
+        self.max_failed_logins = max_failed_logins
+        self.login_lockout_seconds = login_lockout_seconds
         self.db = Database(uri)
         self.jwt_secret = jwt_secret or secrets.token_hex(32)
         self.token_expires = token_expires_seconds
@@ -176,6 +180,16 @@ class ServerCore:
                           mfa_code: str | None = None) -> dict:
         s = self.session
         user = s.query(User).filter_by(username=username).first()
+        # account lockout (reference: user blocked after max_number_of_attempts
+        # failed logins for the password-policy cooldown window)
+        if (user is not None
+                and (user.failed_login_attempts or 0) >= self.max_failed_logins
+                and user.last_login_attempt is not None
+                and (utcnow() - user.last_login_attempt).total_seconds()
+                < self.login_lockout_seconds):
+            raise PermissionError_(
+                "account temporarily locked after too many failed login "
+                "attempts; try again later", 401)
         if user is None or not user.check_password(password):
             if user is not None:
                 user.failed_login_attempts = (user.failed_login_attempts or 0) + 1
