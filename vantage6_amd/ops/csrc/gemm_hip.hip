@@ -373,7 +373,9 @@ gemm_nt_bf16_v3_kernel(const short* __restrict__ A, const short* __restrict__ B,
 // v4 — full-cacheline glds + source-XOR swizzle (rule 21).
 // Each glds instruction covers 8 consecutive rows × the full 128-B K-slab
 // (coalesced: 8 whole lines per instruction), with the kslot index of each
-// lane's 16-B piece XOR-permuted by (row&7) on the SOURCE address; the
+// lane's 16-B piece XOR-permuted by ((row>>1)&7) on the SOURCE address
+// (bijective over each 16-row fragment: parity bit picks the 128-B half of
+// the 256-B bank line, (row>>1)&7 covers its 8 slots exactly once); the
 // fragment read applies the same XOR. Residual 2-way conflict (rows r,r+8)
 // is below the MFMA-issue bound. Single LDS buffer (32 KB → 3 waves/SIMD).
 extern "C" __global__ void __launch_bounds__(NTH)
@@ -403,7 +405,6 @@ gemm_nt_bf16_v4_kernel(const short* __restrict__ A, const short* __restrict__ B,
     for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   // per-lane source column offset (elements): XOR-permuted kslot within row
-  const int src_koff = (((lane & 7) ^ (lane >> 3)) & 7) * 8;
   const int row_in_piece = lane >> 3;
 
   const int nk = K / BK;
@@ -412,6 +413,8 @@ gemm_nt_bf16_v4_kernel(const short* __restrict__ A, const short* __restrict__ B,
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       int pa = wave * 4 + i;  // A piece: rows pa*8 .. pa*8+7
+      const int src_koff =
+          (((lane & 7) ^ (pa * 4 + (lane >> 4))) & 7) * 8;
       const short* srcA =
           A + (long long)(row0 + pa * 8 + row_in_piece) * K + k0 + src_koff;
       __builtin_amdgcn_global_load_lds(
@@ -431,13 +434,13 @@ gemm_nt_bf16_v4_kernel(const short* __restrict__ A, const short* __restrict__ B,
       for (int mi = 0; mi < 4; ++mi) {
         int row = wr + mi * 16 + (lane & 15);
         a_frag[mi] = *(const bf16x8_t*)((char*)lds + row * 128 +
-                                        ((ks ^ (row & 7)) & 7) * 16);
+                                        ((ks ^ (row >> 1)) & 7) * 16);
       }
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
         int row = wc + ni * 16 + (lane & 15);
         b_frag[ni] = *(const bf16x8_t*)((char*)lds + 16384 + row * 128 +
-                                        ((ks ^ (row & 7)) & 7) * 16);
+                                        ((ks ^ (row >> 1)) & 7) * 16);
       }
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
@@ -464,7 +467,9 @@ gemm_nt_bf16_v4_kernel(const short* __restrict__ A, const short* __restrict__ B,
 // v5 — v4 + double-buffered glds (overlap staging of tile k+1 with compute).
 // Each glds instruction covers 8 consecutive rows × the full 128-B K-slab
 // (coalesced: 8 whole lines per instruction), with the kslot index of each
-// lane's 16-B piece XOR-permuted by (row&7) on the SOURCE address; the
+// lane's 16-B piece XOR-permuted by ((row>>1)&7) on the SOURCE address
+// (bijective over each 16-row fragment: parity bit picks the 128-B half of
+// the 256-B bank line, (row>>1)&7 covers its 8 slots exactly once); the
 // fragment read applies the same XOR. Residual 2-way conflict (rows r,r+8)
 // is below the MFMA-issue bound. Single LDS buffer (32 KB → 3 waves/SIMD).
 extern "C" __global__ void __launch_bounds__(NTH)
@@ -494,7 +499,6 @@ gemm_nt_bf16_v5_kernel(const short* __restrict__ A, const short* __restrict__ B,
     for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   // per-lane source column offset (elements): XOR-permuted kslot within row
-  const int src_koff = (((lane & 7) ^ (lane >> 3)) & 7) * 8;
   const int row_in_piece = lane >> 3;
 
   const int nk = K / BK;
@@ -504,6 +508,8 @@ gemm_nt_bf16_v5_kernel(const short* __restrict__ A, const short* __restrict__ B,
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       int pa = wave * 4 + i;
+      const int src_koff =
+          (((lane & 7) ^ (pa * 4 + (lane >> 4))) & 7) * 8;
       const short* srcA =
           A + (long long)(row0 + pa * 8 + row_in_piece) * K + k0 + src_koff;
       __builtin_amdgcn_global_load_lds(
@@ -528,13 +534,13 @@ gemm_nt_bf16_v5_kernel(const short* __restrict__ A, const short* __restrict__ B,
       for (int mi = 0; mi < 4; ++mi) {
         int row = wr + mi * 16 + (lane & 15);
         a_frag[mi] = *(const bf16x8_t*)(base + row * 128 +
-                                        ((ks ^ (row & 7)) & 7) * 16);
+                                        ((ks ^ (row >> 1)) & 7) * 16);
       }
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
         int row = wc + ni * 16 + (lane & 15);
         b_frag[ni] = *(const bf16x8_t*)(base + 16384 + row * 128 +
-                                        ((ks ^ (row & 7)) & 7) * 16);
+                                        ((ks ^ (row >> 1)) & 7) * 16);
       }
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
@@ -587,7 +593,6 @@ gemm_nt_bf16_v6_kernel(const short* __restrict__ A, const short* __restrict__ B,
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int src_koff = (((lane & 7) ^ (lane >> 3)) & 7) * 8;
   const int row_in_piece = lane >> 3;
 
   const int nk = K / BK;
@@ -596,6 +601,8 @@ gemm_nt_bf16_v6_kernel(const short* __restrict__ A, const short* __restrict__ B,
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       int pa = wave * 4 + i;  // 32 pieces of 8 rows cover 256 rows
+      const int src_koff =
+          (((lane & 7) ^ (pa * 4 + (lane >> 4))) & 7) * 8;
       const short* srcA =
           A + (long long)(row0 + pa * 8 + row_in_piece) * K + k0 + src_koff;
       __builtin_amdgcn_global_load_lds(
@@ -615,13 +622,13 @@ gemm_nt_bf16_v6_kernel(const short* __restrict__ A, const short* __restrict__ B,
       for (int ni = 0; ni < 4; ++ni) {
         int row = wc + ni * 16 + (lane & 15);
         b_frag[ni] = *(const bf16x8_t*)((char*)lds + 32768 + row * 128 +
-                                        ((ks ^ (row & 7)) & 7) * 16);
+                                        ((ks ^ (row >> 1)) & 7) * 16);
       }
 #pragma unroll
       for (int mi = 0; mi < 8; ++mi) {
         int row = wr + mi * 16 + (lane & 15);
         bf16x8_t a_frag = *(const bf16x8_t*)((char*)lds + row * 128 +
-                                             ((ks ^ (row & 7)) & 7) * 16);
+                                             ((ks ^ (row >> 1)) & 7) * 16);
 #pragma unroll
         for (int ni = 0; ni < 4; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -675,7 +682,6 @@ gemm_nt_bf16_v7_kernel(const short* __restrict__ A, const short* __restrict__ B,
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int src_koff = (((lane & 7) ^ (lane >> 3)) & 7) * 8;
   const int row_in_piece = lane >> 3;
 
   auto stage = [&](int buf, int kt) {
@@ -684,6 +690,8 @@ gemm_nt_bf16_v7_kernel(const short* __restrict__ A, const short* __restrict__ B,
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       int pa = wave * 4 + i;
+      const int src_koff =
+          (((lane & 7) ^ (pa * 4 + (lane >> 4))) & 7) * 8;
       const short* srcA =
           A + (long long)(row0 + pa * 8 + row_in_piece) * K + k0 + src_koff;
       __builtin_amdgcn_global_load_lds(
@@ -716,13 +724,13 @@ gemm_nt_bf16_v7_kernel(const short* __restrict__ A, const short* __restrict__ B,
       for (int ni = 0; ni < 4; ++ni) {
         int row = wc + ni * 16 + (lane & 15);
         b_frag[ni] = *(const bf16x8_t*)(base + 32768 + row * 128 +
-                                        ((ks ^ (row & 7)) & 7) * 16);
+                                        ((ks ^ (row >> 1)) & 7) * 16);
       }
 #pragma unroll
       for (int mi = 0; mi < 8; ++mi) {
         int row = wr + mi * 16 + (lane & 15);
         bf16x8_t a_frag = *(const bf16x8_t*)(base + row * 128 +
-                                             ((ks ^ (row & 7)) & 7) * 16);
+                                             ((ks ^ (row >> 1)) & 7) * 16);
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int ni = 0; ni < 4; ++ni)
@@ -777,7 +785,6 @@ gemm_nt_bf16_v8_kernel(const short* __restrict__ A, const short* __restrict__ B,
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int src_koff = (((lane & 7) ^ (lane >> 3)) & 7) * 8;
   const int row_in_piece = lane >> 3;
 
   // stage ONE (A,B) piece pair: piece index p in [0,4) per wave
@@ -785,6 +792,8 @@ gemm_nt_bf16_v8_kernel(const short* __restrict__ A, const short* __restrict__ B,
     const long long k0 = (long long)kt * BK;
     char* base = (char*)lds + buf * 65536;
     int pa = wave * 4 + p;
+    const int src_koff =
+        (((lane & 7) ^ (pa * 4 + (lane >> 4))) & 7) * 8;
     const short* srcA =
         A + (long long)(row0 + pa * 8 + row_in_piece) * K + k0 + src_koff;
     __builtin_amdgcn_global_load_lds(
@@ -823,7 +832,7 @@ gemm_nt_bf16_v8_kernel(const short* __restrict__ A, const short* __restrict__ B,
             int row = wc + (nh * 2 + ni) * 16 + (lane & 15);
             int ks = kk * 4 + (lane >> 4);
             b_frag[ni][kk] = *(const bf16x8_t*)(
-                base + 32768 + row * 128 + ((ks ^ (row & 7)) & 7) * 16);
+                base + 32768 + row * 128 + ((ks ^ (row >> 1)) & 7) * 16);
           }
       }
       bf16x8_t a_frag[4][2];
@@ -834,7 +843,7 @@ gemm_nt_bf16_v8_kernel(const short* __restrict__ A, const short* __restrict__ B,
           int row = wr + (mh * 4 + mi) * 16 + (lane & 15);
           int ks = kk * 4 + (lane >> 4);
           a_frag[mi][kk] = *(const bf16x8_t*)(
-              base + row * 128 + ((ks ^ (row & 7)) & 7) * 16);
+              base + row * 128 + ((ks ^ (row >> 1)) & 7) * 16);
         }
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -1111,7 +1120,6 @@ gemm_dgrad_bf16_kernel(const short* __restrict__ A, const short* __restrict__ B,
     for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   // A source (v4 pattern): rows of 128-B red-slabs, XOR-permuted kslot
-  const int a_koff = (((lane & 7) ^ (lane >> 3)) & 7) * 8;
   const int a_row_in_piece = lane >> 3;
   // B source (wgrad pattern): 4-row red groups × subtile order
   const int b_m_in_grp = (lane & 7) >> 1;
@@ -1123,6 +1131,8 @@ gemm_dgrad_bf16_kernel(const short* __restrict__ A, const short* __restrict__ B,
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       int pa = wave * 4 + i;  // A: 16 pieces of 8 rows
+      const int a_koff =
+          (((lane & 7) ^ (pa * 4 + (lane >> 4))) & 7) * 8;
       const short* srcA =
           A + (long long)(row0 + pa * 8 + a_row_in_piece) * R + r0 + a_koff;
       __builtin_amdgcn_global_load_lds(
@@ -1178,7 +1188,7 @@ gemm_dgrad_bf16_kernel(const short* __restrict__ A, const short* __restrict__ B,
       for (int mi = 0; mi < 4; ++mi) {
         int row = wr + mi * 16 + (lane & 15);
         a_frag[mi] = *(const bf16x8_t*)(base + row * 128 +
-                                        ((ks ^ (row & 7)) & 7) * 16);
+                                        ((ks ^ (row >> 1)) & 7) * 16);
       }
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
