@@ -37,6 +37,24 @@ def test_gemm_nt(ops, m, n, k):
     assert err / scale < 2e-2, f"rel err {err/scale}"
 
 
+def test_gemm_deep_k_race_screen(ops):
+    """v8 (phase-interleaved, dispatched at K>=2560) at deep K, 3 repeats.
+
+    Regression test for the last-K-tile vmcnt(2) underwait: with no prefetch
+    issued on the final tile, the newest 2 of its own stage loads could still
+    be in flight when fragments were read — stale LDS on a timing-dependent
+    subset of workgroups (seen as 3.7e-2 rel error at 8192**3, zero at 4096)."""
+    torch.manual_seed(3)
+    m, n, k = 4096, 4096, 8192  # 256 output tiles -> default dispatch = v8
+    a = bf16(torch.randn(m, k)).to(DEV)
+    b = bf16(torch.randn(n, k)).to(DEV)
+    ref = torch.matmul(a, b.t()).float()
+    for _ in range(3):
+        c = ops.gemm_bf16(a, b, False, True).float()
+        rel = (c - ref).norm().item() / (ref.norm().item() + 1e-9)
+        assert rel < 3e-3, f"rel {rel}"
+
+
 def test_gemm_transpose_detecting(ops):
     """Asymmetric operands so a row/col swap cannot pass (guide §5.4 #16)."""
     m, n, k = 128, 64, 32

@@ -819,8 +819,14 @@ gemm_nt_bf16_v8_kernel(const short* __restrict__ A, const short* __restrict__ B,
       const int mh = p & 1, nh = p >> 1;
       if (kt + 1 < nk) stage_piece(cur ^ 1, kt + 1, p);
       if (p == 0 && kt > 0) {
-        // tile t's 8 stages are everything older than this phase's 2
-        asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+        if (kt + 1 < nk) {
+          // tile t's 8 stages are everything older than this phase's 2
+          asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+        } else {
+          // last tile: NOTHING was issued this phase, so vmcnt(2) would
+          // leave the newest 2 of tile t's own stages in flight — drain
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
         __builtin_amdgcn_s_barrier();
       }
       // B quadrant: read on even phases, reuse on odd
