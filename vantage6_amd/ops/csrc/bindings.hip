@@ -12,13 +12,14 @@ struct SgdChunk { const short* grad; long long offset; long long n; };
 extern "C" __global__ void fused_sgd_mt_kernel(const SgdChunk*, int, short*,
                                                float*, float*, float, float,
                                                float, int);
+extern "C" __global__ void adamw_tick_kernel(int*);
 extern "C" __global__ void fused_adamw_mt_kernel(const SgdChunk*, int, short*,
                                                  float*, float*, float*, float,
                                                  float, float, float, float,
-                                                 float, float);
+                                                 float, float, const int*);
 extern "C" __global__ void fused_adamw_kernel(short*, float*, const short*, float*,
                                               float*, long long, float, float, float,
-                                              float, float, float, float);
+                                              float, float, float, float, const int*);
 extern "C" __global__ void secagg_quantize_mask_kernel(long long*, const float*,
                                                        long long, double,
                                                        unsigned long long, int, int);
@@ -138,7 +139,13 @@ void fused_sgd(torch::Tensor param, torch::Tensor master, torch::Tensor grad,
 void fused_adamw(torch::Tensor param, torch::Tensor master, torch::Tensor grad,
                  torch::Tensor exp_avg, torch::Tensor exp_avg_sq, double lr,
                  double beta1, double beta2, double eps, double weight_decay,
-                 int64_t step) {
+                 int64_t step, c10::optional<torch::Tensor> step_dev) {
+  int* sd = nullptr;
+  if (step_dev.has_value()) {
+    TORCH_CHECK(step_dev->scalar_type() == torch::kInt32, "step_dev must be int32");
+    sd = step_dev->data_ptr<int>();
+    hipLaunchKernelGGL(adamw_tick_kernel, dim3(1), dim3(1), 0, cur_stream(), sd);
+  }
   CHECK_HIP(param, torch::kBFloat16);
   CHECK_HIP(master, torch::kFloat32);
   CHECK_HIP(grad, torch::kBFloat16);
@@ -150,7 +157,7 @@ void fused_adamw(torch::Tensor param, torch::Tensor master, torch::Tensor grad,
                      master.data_ptr<float>(), (const short*)grad.data_ptr(),
                      exp_avg.data_ptr<float>(), exp_avg_sq.data_ptr<float>(), n,
                      (float)lr, (float)beta1, (float)beta2, (float)eps,
-                     (float)weight_decay, bc1, bc2);
+                     (float)weight_decay, bc1, bc2, sd);
 }
 
 // table: int64 [n_chunks, 3] = {grad ptr, flat offset, numel} (device)
@@ -172,7 +179,13 @@ void fused_adamw_mt(torch::Tensor table, torch::Tensor param,
                     torch::Tensor master, torch::Tensor exp_avg,
                     torch::Tensor exp_avg_sq, double lr, double beta1,
                     double beta2, double eps, double weight_decay,
-                    int64_t step) {
+                    int64_t step, c10::optional<torch::Tensor> step_dev) {
+  int* sd = nullptr;
+  if (step_dev.has_value()) {
+    TORCH_CHECK(step_dev->scalar_type() == torch::kInt32, "step_dev must be int32");
+    sd = step_dev->data_ptr<int>();
+    hipLaunchKernelGGL(adamw_tick_kernel, dim3(1), dim3(1), 0, cur_stream(), sd);
+  }
   CHECK_HIP(table, torch::kInt64);
   int n_chunks = (int)table.size(0);
   int grid = std::min(n_chunks, 2048);
@@ -183,7 +196,7 @@ void fused_adamw_mt(torch::Tensor table, torch::Tensor param,
                      (short*)param.data_ptr(), master.data_ptr<float>(),
                      exp_avg.data_ptr<float>(), exp_avg_sq.data_ptr<float>(),
                      (float)lr, (float)beta1, (float)beta2, (float)eps,
-                     (float)weight_decay, bc1, bc2);
+                     (float)weight_decay, bc1, bc2, sd);
 }
 
 // ---------------------------------------------------------------- secagg
@@ -690,9 +703,17 @@ torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b, bool trans_a,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_sgd", &fused_sgd, "fused SGD-momentum on flat buffers");
-  m.def("fused_adamw", &fused_adamw, "fused AdamW on flat buffers");
+  m.def("fused_adamw", &fused_adamw, "fused AdamW on flat buffers",
+        py::arg("param"), py::arg("master"), py::arg("grad"), py::arg("exp_avg"),
+        py::arg("exp_avg_sq"), py::arg("lr"), py::arg("beta1"), py::arg("beta2"),
+        py::arg("eps"), py::arg("weight_decay"), py::arg("step"),
+        py::arg("step_dev") = py::none());
   m.def("fused_sgd_mt", &fused_sgd_mt, "multi-tensor SGD, scattered grads");
-  m.def("fused_adamw_mt", &fused_adamw_mt, "multi-tensor AdamW, scattered grads");
+  m.def("fused_adamw_mt", &fused_adamw_mt, "multi-tensor AdamW, scattered grads",
+        py::arg("table"), py::arg("param"), py::arg("master"), py::arg("exp_avg"),
+        py::arg("exp_avg_sq"), py::arg("lr"), py::arg("beta1"), py::arg("beta2"),
+        py::arg("eps"), py::arg("weight_decay"), py::arg("step"),
+        py::arg("step_dev") = py::none());
   m.def("secagg_quantize_mask", &secagg_quantize_mask,
         "fixed-point quantize + pairwise Philox masks");
   m.def("dp_scale_noise", &dp_scale_noise, "DP clip-scale + gaussian noise");

@@ -54,15 +54,28 @@ fused_sgd_kernel(short* __restrict__ param, float* __restrict__ master,
   }
 }
 
+
+// one-thread increment of the device step counter — captured in the hipGraph
+// so bias correction advances across replays (a host-computed bc1/bc2 would
+// freeze at capture-time values)
+extern "C" __global__ void adamw_tick_kernel(int* __restrict__ step) {
+  if (blockIdx.x == 0 && threadIdx.x == 0) ++*step;
+}
+
 // -------------------------------------------------------------- fused AdamW
 extern "C" __global__ void __launch_bounds__(256)
 fused_adamw_kernel(short* __restrict__ param, float* __restrict__ master,
                    const short* __restrict__ grad, float* __restrict__ exp_avg,
                    float* __restrict__ exp_avg_sq, long long n, float lr,
                    float beta1, float beta2, float eps, float weight_decay,
-                   float bc1, float bc2) {
+                   float bc1, float bc2, const int* __restrict__ step_dev) {
   long long i0 = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
   long long stride = (long long)gridDim.x * blockDim.x * 8;
+  if (step_dev) {
+    float t = (float)*step_dev;
+    bc1 = 1.0f - __powf(beta1, t);
+    bc2 = 1.0f - __powf(beta2, t);
+  }
   float decay = 1.0f - lr * weight_decay;
   float step_lr = lr / bc1;
   float inv_bc2 = 1.0f / bc2;
@@ -250,7 +263,13 @@ fused_adamw_mt_kernel(const SgdChunk* __restrict__ table, int n_chunks,
                       short* __restrict__ param, float* __restrict__ master,
                       float* __restrict__ exp_avg, float* __restrict__ exp_avg_sq,
                       float lr, float beta1, float beta2, float eps,
-                      float weight_decay, float bc1, float bc2) {
+                      float weight_decay, float bc1, float bc2,
+                      const int* __restrict__ step_dev) {
+  if (step_dev) {
+    float t = (float)*step_dev;
+    bc1 = 1.0f - __powf(beta1, t);
+    bc2 = 1.0f - __powf(beta2, t);
+  }
   float decay = 1.0f - lr * weight_decay;
   float step_lr = lr / bc1;
   float inv_bc2 = 1.0f / bc2;

@@ -61,12 +61,17 @@ def fused_adamw(
     eps: float = 1e-8,
     weight_decay: float = 0.01,
     step: int = 1,
+    step_dev: torch.Tensor | None = None,
 ) -> None:
-    """AdamW (decoupled weight decay), fp32 state, bf16 compute copy."""
+    """AdamW (decoupled weight decay), fp32 state, bf16 compute copy.
+
+    ``step_dev`` (int32 device scalar) makes bias correction device-side so
+    the step is hipGraph-capturable; the kernel pre-increments it each call.
+    """
     if param_bf16.is_cuda:
         hip_ops().fused_adamw(
             param_bf16, master_f32, grad_bf16, exp_avg, exp_avg_sq,
-            lr, beta1, beta2, eps, weight_decay, step,
+            lr, beta1, beta2, eps, weight_decay, step, step_dev,
         )
         return
     g = grad_bf16.float()

@@ -94,11 +94,11 @@ class FedAvgTrainer:
         launches/step). Flat grads make this safe — every gradient buffer
         address is static. Call AFTER warmup (MIOpen find must be done).
         Freezes lr/hyperparams at capture values; not for DP (dynamic
-        clipping control flow)."""
+        clipping control flow). AdamW is capturable because its bias
+        correction derives from an int32 DEVICE step counter that the
+        captured tick kernel advances on every replay."""
         if self.comm.device.type != "cuda" or self.dp_engine is not None:
             return False
-        if self.cfg.optimizer != "sgd":
-            return False  # AdamW bias-correction step counter is host-side
         try:
             batch = next(self.batches)
             self._static_batch = {k: v.clone() for k, v in batch.items()}

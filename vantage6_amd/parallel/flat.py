@@ -68,6 +68,10 @@ class FlatParamSpace:
         self.momentum_flat = torch.zeros_like(self.master_flat)
         self.exp_avg_sq: torch.Tensor | None = None  # lazily for AdamW
         self._step = 0
+        # int32 device step counter: the AdamW kernels pre-increment it and
+        # derive bias correction from it, so the optimizer step is
+        # hipGraph-capturable (a host-side counter freezes at capture time)
+        self._step_dev: torch.Tensor | None = None
 
     # ------------------------------------------------------------- optimizer
     def zero_grad(self) -> None:
@@ -150,6 +154,9 @@ class FlatParamSpace:
                    weight_decay=0.01) -> None:
         if self.exp_avg_sq is None:
             self.exp_avg_sq = torch.zeros_like(self.master_flat)
+        if self.param_flat.is_cuda and self._step_dev is None:
+            self._step_dev = torch.full((1,), self._step, dtype=torch.int32,
+                                        device=self.device)
         self._step += 1
         if self.scatter_grads:
             if self.param_flat.is_cuda:
@@ -159,7 +166,7 @@ class FlatParamSpace:
                 hip_ops().fused_adamw_mt(
                     self._sgd_table, self.param_flat, self.master_flat,
                     self.momentum_flat, self.exp_avg_sq, lr, beta1, beta2,
-                    eps, weight_decay, self._step,
+                    eps, weight_decay, self._step, self._step_dev,
                 )
             else:
                 bc1 = 1 - beta1 ** self._step
@@ -180,7 +187,7 @@ class FlatParamSpace:
             self.param_flat, self.master_flat, self.grad_flat,
             self.momentum_flat, self.exp_avg_sq,
             lr=lr, beta1=beta1, beta2=beta2, eps=eps, weight_decay=weight_decay,
-            step=self._step,
+            step=self._step, step_dev=self._step_dev,
         )
 
     # ------------------------------------------------------------ aggregation
@@ -192,6 +199,10 @@ class FlatParamSpace:
         self.sync_param_from_master()
 
     def state_dict(self) -> dict:
+        # under hipGraph replay the python counter goes stale — the device
+        # counter is authoritative
+        if self._step_dev is not None:
+            self._step = int(self._step_dev.item())
         return {
             "master": self.master_flat,
             "momentum": self.momentum_flat,
@@ -207,4 +218,6 @@ class FlatParamSpace:
                 self.exp_avg_sq = torch.zeros_like(self.master_flat)
             self.exp_avg_sq.copy_(sd["exp_avg_sq"])
         self._step = sd.get("step", 0)
+        if self._step_dev is not None:
+            self._step_dev.fill_(self._step)
         self.sync_param_from_master()
