@@ -512,4 +512,4 @@ def test_graph_adamw_matches_eager(ops):
     a, b = eager.flat.master_flat, graphed.flat.master_flat
     diff = (a - b).abs().max().item()
     assert diff < 1e-5, f"master weights diverged {diff}; losses {losses}"
-    assert int(graphed.flat._step_dev.item()) == 5 + 4  # 5 replays + warmup
+    assert int(graphed.flat._step_dev.item()) == 5  # reset post-capture, +1/replay
