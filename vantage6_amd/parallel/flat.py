@@ -162,7 +162,11 @@ class FlatParamSpace:
             if self.param_flat.is_cuda:
                 from ..ops import hip_ops
 
-                self.refresh_sgd_table()
+                if not torch.cuda.is_current_stream_capturing():
+                    # during capture the table tensor must already exist
+                    # (warmup built it); refreshing here would record an H2D
+                    # copy from a python temp that is freed before replay
+                    self.refresh_sgd_table()
                 hip_ops().fused_adamw_mt(
                     self._sgd_table, self.param_flat, self.master_flat,
                     self.momentum_flat, self.exp_avg_sq, lr, beta1, beta2,
