@@ -513,3 +513,14 @@ def test_graph_adamw_matches_eager(ops):
     diff = (a - b).abs().max().item()
     assert diff < 1e-5, f"master weights diverged {diff}; losses {losses}"
     assert int(graphed.flat._step_dev.item()) == 5  # reset post-capture, +1/replay
+
+
+def test_colsum_bias_grad(ops):
+    """colsum_bf16 (Linear bias grad) vs fp32 reference."""
+    torch.manual_seed(5)
+    for m, n in [(4096, 768), (4096, 3072), (128, 8), (1000, 30528)]:
+        dy = bf16(torch.randn(m, n)).to(DEV)
+        db = ops.colsum_bf16(dy).float()
+        ref = dy.float().sum(0)
+        err = (db - ref).abs().max().item() / (ref.abs().max().item() + 1e-6)
+        assert err < 2e-2, f"{m}x{n}: rel {err}"

@@ -56,6 +56,8 @@ extern "C" __global__ void rmsnorm_bwd_wave_kernel(const short*, const short*,
                                                    short*, float*, long long,
                                                    int);
 extern "C" __global__ void zero_f32_kernel(float*, long long);
+extern "C" __global__ void colsum_bf16_kernel(const short*, float*, long long,
+                                              int, int);
 extern "C" __global__ void gemm_nt_bf16_kernel(const short*, const short*, short*,
                                                int, int, int, int, int, int);
 extern "C" __global__ void gemm_nt_bf16_v2_kernel(const short*, const short*,
@@ -197,6 +199,30 @@ void fused_adamw_mt(torch::Tensor table, torch::Tensor param,
                      exp_avg.data_ptr<float>(), exp_avg_sq.data_ptr<float>(),
                      (float)lr, (float)beta1, (float)beta2, (float)eps,
                      (float)weight_decay, bc1, bc2, sd);
+}
+
+torch::Tensor colsum_bf16(torch::Tensor dy) {
+  CHECK_HIP(dy, torch::kBFloat16);
+  TORCH_CHECK(dy.dim() == 2 && dy.is_contiguous(), "colsum_bf16: 2-D contig");
+  long long M = dy.size(0);
+  int N = (int)dy.size(1);
+  TORCH_CHECK(N % 8 == 0, "colsum_bf16: N must be a multiple of 8");
+  auto ws = torch::empty({16, (long long)N}, dy.options().dtype(torch::kFloat32));
+  auto db = torch::empty({(long long)N}, dy.options());
+  int zgrid = (int)((16LL * N + 2047) / 2048);
+  hipLaunchKernelGGL(zero_f32_kernel, dim3(zgrid), dim3(256), 0, cur_stream(),
+                     ws.data_ptr<float>(), 16LL * N);
+  int stripes = (N + 2047) / 2048;
+  int rchunk = 64;
+  int gy = (int)((M + rchunk - 1) / rchunk);
+  if (gy > 1024) { rchunk = (int)((M + 1023) / 1024); gy = (int)((M + rchunk - 1) / rchunk); }
+  hipLaunchKernelGGL(colsum_bf16_kernel, dim3(stripes, gy), dim3(256), 0,
+                     cur_stream(), (const short*)dy.data_ptr(),
+                     ws.data_ptr<float>(), M, N, rchunk);
+  int cgrid = (N + 255) / 256;
+  hipLaunchKernelGGL(colsum_ws_kernel, dim3(cgrid), dim3(256), 0, cur_stream(),
+                     ws.data_ptr<float>(), (short*)db.data_ptr(), 16, N);
+  return db;
 }
 
 // ---------------------------------------------------------------- secagg
@@ -719,6 +745,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dp_scale_noise", &dp_scale_noise, "DP clip-scale + gaussian noise");
   m.def("grad_norm2_partials", &grad_norm2_partials, "block partial sum of g^2");
   m.def("rope_apply", &rope_apply);
+  m.def("colsum_bf16", &colsum_bf16, "column sum of a bf16 matrix (bias grad)");
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("layernorm_fwd", &layernorm_fwd);

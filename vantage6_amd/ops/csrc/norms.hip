@@ -469,3 +469,29 @@ rmsnorm_bwd_wave_kernel(const short* __restrict__ dy,
     }
   }
 }
+
+// ------------------------------------------------------------ bf16 colsum
+// db[N] = sum over M rows of dy[M,N] (Linear bias grad). Row-major bf16
+// reads are fully coalesced (vs torch's generic reduce, which is also NOT
+// hipGraph-replay-safe: its semaphore workspace is memset at allocation
+// time only). 16-slice fp32 workspace + colsum_ws finalize, all kernels —
+// zeroing included — so every byte of state is re-written on each replay.
+extern "C" __global__ void __launch_bounds__(256)
+colsum_bf16_kernel(const short* __restrict__ dy, float* __restrict__ ws,
+                   long long M, int N, int rchunk) {
+  const int col0 = blockIdx.x * 2048 + threadIdx.x * 8;
+  if (col0 + 8 > N) return;
+  const long long r0 = (long long)blockIdx.y * rchunk;
+  const long long r1 = (r0 + rchunk < M) ? r0 + rchunk : M;
+  float acc[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) acc[j] = 0.f;
+  for (long long r = r0; r < r1; ++r) {
+    s16x8 v = *(const s16x8*)(dy + r * N + col0);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] += bf16_to_f32(v[j]);
+  }
+  float* slice = ws + (long long)(blockIdx.y & 15) * N;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) atomicAdd(&slice[col0 + j], acc[j]);
+}

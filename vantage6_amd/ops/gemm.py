@@ -45,7 +45,16 @@ class _LinearFn(torch.autograd.Function):
         dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
         dx = matmul_bf16(dy2, weight, False, False).reshape(ctx.in_shape)
         dw = matmul_bf16(dy2, x2, True, False)
-        db = dy2.sum(0) if ctx.has_bias else None
+        if ctx.has_bias:
+            if dy2.is_cuda and dy2.shape[-1] % 8 == 0:
+                # hand-written colsum: coalesced bf16 reads, and (unlike
+                # torch's two-pass reduce) hipGraph-replay-safe — its
+                # workspace is zeroed by a kernel every call
+                db = hip_ops().colsum_bf16(dy2)
+            else:
+                db = dy2.sum(0)
+        else:
+            db = None
         return dx, dw, db
 
 
