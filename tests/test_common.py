@@ -148,3 +148,21 @@ class TestTOTP:
 
         secret = base64.b32encode(b"12345678901234567890").decode()
         assert totp.totp_code(secret, at=59, digits=8) == "94287082"
+
+
+def test_embedding_cpu_backward_matches_torch():
+    """ops.embedding.Embedding CPU path vs nn.Embedding."""
+    import torch
+    from vantage6_amd.ops.embedding import Embedding
+
+    torch.manual_seed(3)
+    emb = Embedding(50, 16)
+    ref = torch.nn.Embedding(50, 16)
+    with torch.no_grad():
+        ref.weight.copy_(emb.weight)
+    ids = torch.randint(0, 50, (8, 12))
+    dy = torch.randn(8, 12, 16)
+    emb(ids).backward(dy)
+    ref(ids).backward(dy)
+    assert torch.allclose(emb.weight.grad, ref.weight.grad, atol=1e-5)
+    assert isinstance(emb, torch.nn.Embedding)  # DP hooks use isinstance

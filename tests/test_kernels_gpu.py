@@ -524,3 +524,21 @@ def test_colsum_bias_grad(ops):
         ref = dy.float().sum(0)
         err = (db - ref).abs().max().item() / (ref.abs().max().item() + 1e-6)
         assert err < 2e-2, f"{m}x{n}: rel {err}"
+
+
+def test_emb_grad(ops):
+    """Graph-safe embedding backward vs nn.Embedding (fp32 scatter ref)."""
+    torch.manual_seed(7)
+    V, H, B, S = 1000, 768, 4, 64
+    ids = torch.randint(0, V, (B, S), device=DEV)
+    from vantage6_amd.ops.embedding import Embedding
+    emb = Embedding(V, H).to(DEV, torch.bfloat16)
+    ref = torch.nn.Embedding(V, H).to(DEV, torch.bfloat16)
+    with torch.no_grad():
+        ref.weight.copy_(emb.weight)
+    dy = bf16(torch.randn(B, S, H)).to(DEV)
+    emb(ids).backward(dy)
+    ref(ids).backward(dy)
+    d = (emb.weight.grad.float() - ref.weight.grad.float()).abs().max().item()
+    scale = ref.weight.grad.float().abs().max().item() + 1e-6
+    assert d / scale < 3e-2, f"rel {d/scale}"

@@ -16,6 +16,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops.activations import rope, rope_tables, swiglu
+from ..ops.embedding import Embedding as GraphSafeEmbedding
 from ..ops.gemm import Linear as MfmaLinear
 from ..ops.norms import LayerNorm, RMSNorm
 
@@ -47,8 +48,8 @@ class BertBase(nn.Module):
     def __init__(self, vocab: int = 30522, hidden: int = 768, layers: int = 12,
                  heads: int = 12, ffn: int = 3072, max_seq: int = 512):
         super().__init__()
-        self.tok = nn.Embedding(vocab, hidden)
-        self.pos = nn.Embedding(max_seq, hidden)
+        self.tok = GraphSafeEmbedding(vocab, hidden)
+        self.pos = GraphSafeEmbedding(max_seq, hidden)
         self.ln = LayerNorm(hidden)
         self.blocks = nn.ModuleList(BertLayer(hidden, heads, ffn) for _ in range(layers))
         self.head = MfmaLinear(hidden, vocab, bias=False)
@@ -112,7 +113,7 @@ class Llama(nn.Module):
     def __init__(self, vocab: int = 128256, hidden: int = 2048, layers: int = 16,
                  heads: int = 32, kv_heads: int = 8, ffn: int = 8192):
         super().__init__()
-        self.tok = nn.Embedding(vocab, hidden)
+        self.tok = GraphSafeEmbedding(vocab, hidden)
         self.blocks = nn.ModuleList(
             LlamaLayer(hidden, heads, kv_heads, ffn) for _ in range(layers)
         )

@@ -56,6 +56,9 @@ extern "C" __global__ void rmsnorm_bwd_wave_kernel(const short*, const short*,
                                                    short*, float*, long long,
                                                    int);
 extern "C" __global__ void zero_f32_kernel(float*, long long);
+extern "C" __global__ void emb_scatter_kernel(const long long*, const short*,
+                                              float*, long long, int);
+extern "C" __global__ void f32_to_bf16_kernel(const float*, short*, long long);
 extern "C" __global__ void colsum_bf16_kernel(const short*, float*, long long,
                                               int, int);
 extern "C" __global__ void gemm_nt_bf16_kernel(const short*, const short*, short*,
@@ -223,6 +226,32 @@ torch::Tensor colsum_bf16(torch::Tensor dy) {
   hipLaunchKernelGGL(colsum_ws_kernel, dim3(cgrid), dim3(256), 0, cur_stream(),
                      ws.data_ptr<float>(), (short*)db.data_ptr(), 16, N);
   return db;
+}
+
+torch::Tensor emb_grad_bf16(torch::Tensor ids, torch::Tensor dy,
+                            int64_t vocab) {
+  TORCH_CHECK(ids.scalar_type() == torch::kInt64 && ids.is_cuda() &&
+              ids.is_contiguous(), "emb_grad_bf16: ids int64 contig cuda");
+  CHECK_HIP(dy, torch::kBFloat16);
+  TORCH_CHECK(dy.dim() == 2 && dy.is_contiguous(), "emb_grad_bf16: dy 2-D");
+  long long n_idx = ids.numel();
+  int H = (int)dy.size(1);
+  TORCH_CHECK(dy.size(0) == n_idx, "ids/dy row mismatch");
+  auto ws = torch::empty({vocab, (long long)H},
+                         dy.options().dtype(torch::kFloat32));
+  auto dw = torch::empty({vocab, (long long)H}, dy.options());
+  long long n = (long long)vocab * H;
+  hipLaunchKernelGGL(zero_f32_kernel, dim3(ew_grid(n)), dim3(256), 0,
+                     cur_stream(), ws.data_ptr<float>(), n);
+  int grid = (int)std::min<long long>(n_idx, 8192);
+  hipLaunchKernelGGL(emb_scatter_kernel, dim3(grid), dim3(256), 0,
+                     cur_stream(), (const long long*)ids.data_ptr<int64_t>(),
+                     (const short*)dy.data_ptr(), ws.data_ptr<float>(), n_idx,
+                     H);
+  hipLaunchKernelGGL(f32_to_bf16_kernel, dim3(ew_grid(n)), dim3(256), 0,
+                     cur_stream(), ws.data_ptr<float>(), (short*)dw.data_ptr(),
+                     n);
+  return dw;
 }
 
 // ---------------------------------------------------------------- secagg
@@ -746,6 +775,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("grad_norm2_partials", &grad_norm2_partials, "block partial sum of g^2");
   m.def("rope_apply", &rope_apply);
   m.def("colsum_bf16", &colsum_bf16, "column sum of a bf16 matrix (bias grad)");
+  m.def("emb_grad_bf16", &emb_grad_bf16, "graph-safe embedding backward");
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("layernorm_fwd", &layernorm_fwd);

@@ -413,3 +413,34 @@ rope_kernel(const short* __restrict__ x, short* __restrict__ out,
     out[base + half + d] = f32_to_bf16(x1 * sn + x2 * c);
   }
 }
+
+// ------------------------------------------------- graph-safe embedding bwd
+// dW[v] += sum of dy rows with ids==v. torch's embedding_dense_backward
+// (cub sort + memset-at-alloc temp storage) aborts with an aperture fault
+// on the 2nd hipGraph replay on this ROCm build; this path is plain
+// kernels end-to-end: zero_f32 ws -> fp32 atomic scatter -> bf16 convert.
+extern "C" __global__ void __launch_bounds__(256)
+emb_scatter_kernel(const long long* __restrict__ ids,
+                   const short* __restrict__ dy, float* __restrict__ ws,
+                   long long n_idx, int H) {
+  for (long long r = blockIdx.x; r < n_idx; r += gridDim.x) {
+    long long v = ids[r];
+    const short* row = dy + r * (long long)H;
+    float* wrow = ws + v * (long long)H;
+    for (int c = threadIdx.x * 8; c + 8 <= H; c += 256 * 8) {
+      s16x8 g8 = *(const s16x8*)(row + c);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) atomicAdd(&wrow[c + j], bf16_to_f32(g8[j]));
+    }
+    for (int c = (H & ~7) + threadIdx.x; c < H; c += 256)
+      atomicAdd(&wrow[c], bf16_to_f32(row[c]));
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+f32_to_bf16_kernel(const float* __restrict__ src, short* __restrict__ dst,
+                   long long n) {
+  long long i = (long long)blockIdx.x * 256 + threadIdx.x;
+  long long stride = (long long)gridDim.x * 256;
+  for (; i < n; i += stride) dst[i] = f32_to_bf16(src[i]);
+}
