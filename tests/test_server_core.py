@@ -432,3 +432,83 @@ class TestLockout:
         core.login_lockout_seconds = 0.0
         c.authenticate("locky", "goodpw")
         core.close()
+
+
+class TestRound1Additions:
+    def test_mfa_reset_flow(self, core, root):
+        from vantage6_amd.common import totp
+        from vantage6_amd.server.mail import MailService, request_mfa_reset
+
+        orgs, _ = make_collab(root, name="mfr")
+        root.user.create("bob2fa", "pw", organization_id=orgs[0],
+                         roles=["Researcher"], email="bob@example.org")
+        bob = UserClient(core=core)
+        bob.authenticate("bob2fa", "pw")
+        core.enable_mfa(bob.gw.ident)
+        mail = MailService()
+        request_mfa_reset(core, mail, "bob2fa")
+        token = mail.outbox[0]["body"].split(": ")[-1]
+        out = core.reset_mfa_with_token(token)
+        code = totp.totp_code(out["mfa_secret"])
+        tokens = core.authenticate_user("bob2fa", "pw", mfa_code=code)
+        assert "access_token" in tokens
+
+    def test_role_rule_store_subclients(self, core, root):
+        rules = root.rule.list(per_page=500)["data"]
+        assert any(r["name"] == "task" for r in rules)
+        view_ids = [r["id"] for r in rules
+                    if r["operation"] == "view" and r["scope"] == "own"]
+        role = root.role.create("AuditorX", view_ids)
+        assert any(x["name"] == "AuditorX"
+                   for x in root.role.list(per_page=200)["data"])
+        alg = root.store.submit("meanx", "algorithms.column_mean")
+        root.store.review(alg["id"], approve=True)
+        assert any(a["name"] == "meanx"
+                   for a in root.store.list(status="approved")["data"])
+
+    def test_column_discovery(self, tmp_path):
+        import pandas as pd
+
+        from vantage6_amd.cli.dev import DevNetwork
+        from vantage6_amd.common.config import DatabaseConfig
+
+        csv = tmp_path / "d.csv"
+        pd.DataFrame({"age": [1, 2], "bmi": [20.0, 21.5]}).to_csv(csv, index=False)
+        net = DevNetwork(n_stations=1, databases=[[
+            DatabaseConfig(label="default", uri=str(csv), type="csv")]])
+        with net:
+            st = net.stations[0]
+            st.core.node_ping(st.identity, {"columns": st._database_columns()})
+            node_id = net.root_client.node.list()["data"][0]["id"]
+            cols = net.root_client.node.columns(node_id)
+            assert cols == ["age", "bmi"]
+
+    def test_import_cli(self, tmp_path):
+        import yaml
+        from click.testing import CliRunner
+
+        from vantage6_amd.cli.main import cli
+
+        fixture = {
+            "organizations": [{"name": "imp-a"}, {"name": "imp-b"}],
+            "collaborations": [{"name": "imp-c",
+                                "organizations": ["imp-a", "imp-b"]}],
+            "users": [{"username": "imp-u", "password": "pw",
+                       "organization": "imp-a"}],
+            "nodes": [{"name": "imp-n", "collaboration": "imp-c",
+                       "organization": "imp-b"}],
+        }
+        path = tmp_path / "fixture.yaml"
+        path.write_text(yaml.safe_dump(fixture))
+        res = CliRunner().invoke(cli, ["import", "--config", str(path)])
+        assert res.exit_code == 0, res.output
+        assert "api_key" in res.output
+
+    def test_algorithm_util_helpers(self, capsys):
+        from vantage6_amd.algorithm import get_env_var, info, warn
+
+        info("hello")
+        warn("careful")
+        out = capsys.readouterr().out
+        assert "info > hello" in out and "warn > careful" in out
+        assert get_env_var("V6AMD_NOT_SET_XYZ", "dflt") == "dflt"

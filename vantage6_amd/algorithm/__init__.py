@@ -22,3 +22,4 @@ from .exceptions import (  # noqa: F401
 from .decorators import algorithm_client, data, metadata  # noqa: F401
 from .client import AlgorithmClient  # noqa: F401
 from .mock_client import MockAlgorithmClient  # noqa: F401
+from .util import error, get_env_var, info, warn  # noqa: F401

@@ -123,6 +123,53 @@ def dev_demo(stations, encrypted):
     click.echo(json.dumps({"demo": "column_mean", "result": res[0]}, indent=2))
 
 
+@cli.command(name="import")
+@click.option("--config", "config_path", required=True,
+              type=click.Path(exists=True), help="entity fixture YAML")
+@click.option("--uri", default="sqlite://", help="orchestrator database URI")
+@click.option("--root-password", default="root")
+def import_entities(config_path, uri, root_password):
+    """Load organizations/collaborations/users/nodes from a YAML fixture
+    (reference: ``v6 server import`` — SURVEY.md §3.6). Prints created
+    node api-keys so stations can be configured."""
+    import yaml as _yaml
+
+    from ..client import UserClient
+    from ..server.core import ServerCore
+
+    with open(config_path) as fh:
+        fixture = _yaml.safe_load(fh) or {}
+    core = ServerCore(uri=uri)
+    core.create_root_user("root", root_password)
+    root = UserClient(core=core)
+    root.authenticate("root", root_password)
+    org_ids = {}
+    for org in fixture.get("organizations", []):
+        created = root.organization.create(name=org["name"],
+                                           public_key=org.get("public_key", ""))
+        org_ids[org["name"]] = created["id"]
+        click.echo(f"organization {org['name']!r}: id {created['id']}")
+    collab_ids = {}
+    for col in fixture.get("collaborations", []):
+        created = root.collaboration.create(
+            name=col["name"],
+            organization_ids=[org_ids[n] for n in col["organizations"]],
+            encrypted=col.get("encrypted", False))
+        collab_ids[col["name"]] = created["id"]
+        click.echo(f"collaboration {col['name']!r}: id {created['id']}")
+    for user in fixture.get("users", []):
+        root.user.create(username=user["username"], password=user["password"],
+                         organization_id=org_ids[user["organization"]],
+                         roles=user.get("roles", ["Researcher"]))
+        click.echo(f"user {user['username']!r} created")
+    for node in fixture.get("nodes", []):
+        created = root.node.create(name=node["name"],
+                                   collaboration_id=collab_ids[node["collaboration"]],
+                                   organization_id=org_ids[node["organization"]])
+        click.echo(f"node {node['name']!r}: api_key {created['api_key']}")
+    core.close()
+
+
 @cli.command()
 @click.option("--stations", default=2)
 def test(stations):

@@ -118,6 +118,9 @@ class UserClient:
         self.node = self.NodeSub(self)
         self.user = self.UserSub(self)
         self.study = self.StudySub(self)
+        self.role = self.RoleSub(self)
+        self.rule = self.RuleSub(self)
+        self.store = self.StoreSub(self)
 
     # ----------------------------------------------------------------- auth
     def authenticate(self, username: str, password: str) -> dict:
@@ -280,6 +283,12 @@ class UserClient:
         def list(self, **params) -> dict:
             return self.parent._call("list_nodes", "GET", "node", params=params, **params)
 
+        def columns(self, node_id: int, label: str = "default") -> list[str]:
+            """Data-column discovery (reference /api/column)."""
+            return self.parent._call("get_columns", "GET", f"column/{node_id}",
+                                     params={"label": label}, node_id=node_id,
+                                     label=label)
+
         def create(self, name: str, collaboration_id: int, organization_id: int,
                    device: str = "cpu") -> dict:
             body = {"name": name, "collaboration_id": collaboration_id,
@@ -292,6 +301,38 @@ class UserClient:
             body = {"username": username, "password": password,
                     "organization_id": organization_id, "roles": roles or [], **kw}
             return self.parent._call("create_user", "POST", "user", json=body, **body)
+
+    class RoleSub(_Sub):
+        def list(self, **params) -> dict:
+            return self.parent._call("list_roles", "GET", "role", params=params, **params)
+
+        def create(self, name: str, rule_ids: list[int],
+                   organization_id: int | None = None) -> dict:
+            body = {"name": name, "rule_ids": rule_ids,
+                    "organization_id": organization_id}
+            return self.parent._call("create_role", "POST", "role", json=body, **body)
+
+    class RuleSub(_Sub):
+        def list(self, **params) -> dict:
+            return self.parent._call("list_rules", "GET", "rule", params=params, **params)
+
+    class StoreSub(_Sub):
+        """Algorithm-store sub-client (reference ``client.store``, 4.2+)."""
+
+        def submit(self, name: str, image: str, description: str = "") -> dict:
+            body = {"name": name, "image": image, "description": description}
+            return self.parent._call("store_submit_algorithm", "POST", "algorithm",
+                                     json=body, **body)
+
+        def review(self, algorithm_id: int, approve: bool, comment: str = "") -> dict:
+            body = {"approve": approve, "comment": comment}
+            return self.parent._call("store_review_algorithm", "PATCH",
+                                     f"algorithm/{algorithm_id}/review", json=body,
+                                     algorithm_id=algorithm_id, **body)
+
+        def list(self, **params) -> dict:
+            return self.parent._call("store_list_algorithms", "GET", "algorithm",
+                                     params=params, **params)
 
     class StudySub(_Sub):
         def create(self, name: str, collaboration_id: int,

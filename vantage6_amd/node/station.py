@@ -114,9 +114,29 @@ class Station:
                     "device": self.config.device,
                     "queue_depth": self.queue.qsize(),
                     "inflight": busy,
+                    "columns": self._database_columns(),
                 })
             except Exception:  # noqa: BLE001
                 log.exception("heartbeat failed")
+
+    def _database_columns(self) -> dict:
+        """Column names per database label (reference: /api/column — data
+        discovery for the UI/researcher). Cached; only tabular sources
+        report columns, tensor/synthetic report none."""
+        if getattr(self, "_columns_cache", None) is not None:
+            return self._columns_cache
+        out: dict[str, list[str]] = {}
+        for db in self.config.databases:
+            try:
+                if db.type in ("csv", "parquet", "excel"):
+                    from ..algorithm.decorators import _load_database
+
+                    df = _load_database(db.uri, db.type, {})
+                    out[db.label] = [str(c) for c in df.columns]
+            except Exception:  # noqa: BLE001 — discovery must not kill pings
+                log.exception("column discovery failed for %s", db.label)
+        self._columns_cache = out
+        return out
 
     def stop(self, timeout: float = 10.0) -> None:
         if getattr(self, "_stop_ping", None) is not None:

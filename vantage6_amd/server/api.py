@@ -348,6 +348,40 @@ def create_app(core: ServerCore) -> FastAPI:
             for room, cb in subs:
                 core.events.unsubscribe(room, cb)
 
+    # ------------------------------------------------------------- recovery
+    from . import mail as mail_mod
+
+    mail_service = mail_mod.MailService()
+
+    class RecoverBody(BaseModel):
+        username: str = ""
+        token: str = ""
+        password: str = ""
+
+    @app.post("/api/recover/lost")
+    def recover_lost(body: RecoverBody):
+        mail_mod.request_password_recovery(core, mail_service, body.username)
+        return {"msg": "if the user exists, a recovery email was sent"}
+
+    @app.post("/api/recover/reset")
+    def recover_reset(body: RecoverBody):
+        mail_mod.reset_password(core, body.token, body.password)
+        return {"msg": "password reset"}
+
+    @app.post("/api/recover/2fa/lost")
+    def recover_2fa_lost(body: RecoverBody):
+        mail_mod.request_mfa_reset(core, mail_service, body.username)
+        return {"msg": "if the user exists, an mfa-reset email was sent"}
+
+    @app.post("/api/recover/2fa/reset")
+    def recover_2fa_reset(body: RecoverBody):
+        return core.reset_mfa_with_token(body.token)
+
+    @app.get("/api/column/{node_id}")
+    def get_columns(node_id: int, label: str = "default",
+                    who: Identity = Depends(ident)):
+        return core.get_columns(who, node_id, label)
+
     # ----------------------------------------------------------------- ops
     @app.get("/api/version")
     def version():

@@ -267,6 +267,24 @@ class ServerCore:
         return {"container_token": jwtlib.encode(payload, self.jwt_secret, self.token_expires)}
 
     @_locked
+    def reset_mfa_with_token(self, token: str) -> dict:
+        """Complete an emailed MFA-reset: issue a fresh TOTP secret."""
+        import vantage6_amd.common.jwt as jwtlib_
+
+        payload = jwtlib_.decode(token, self.jwt_secret)
+        if payload.get("purpose") != "mfa_reset":
+            raise PermissionError_("not an mfa-reset token", 401)
+        user = self.session.get(User, payload["sub"])
+        if user is None:
+            raise NotFound("user not found")
+        from ..common import totp as totp_
+
+        user.mfa_secret = totp_.generate_secret()
+        self.session.commit()
+        return {"mfa_secret": user.mfa_secret,
+                "otp_uri": totp_.provisioning_uri(user.mfa_secret, user.username)}
+
+    @_locked
     def enable_mfa(self, ident: Identity) -> dict:
         """Provision a TOTP secret for the calling user (vantage6 4.x MFA)."""
         from ..common import totp
@@ -800,6 +818,10 @@ class ServerCore:
             raise NotFound(f"node {ident.id} not found")
         node.last_seen = utcnow()
         node.status = "online"
+        if info:
+            import json as _json
+
+            node.info_json = _json.dumps(info)
         s.commit()
         if info:
             self.events.emit(
@@ -807,6 +829,33 @@ class ServerCore:
                 {"id": node.id, "name": node.name, **info},
                 room=f"collaboration_{node.collaboration_id}",
             )
+
+    @_locked
+    def get_columns(self, ident: Identity, node_id: int,
+                    label: str = "default") -> list[str]:
+        """Column names of a node-mounted database (reference: /api/column —
+        the UI asks nodes for available data columns). Stations report
+        columns with their first heartbeat (node-info-update)."""
+        import json as _json
+
+        node = self.session.get(Node, node_id)
+        if node is None:
+            raise NotFound(f"node {node_id} not found")
+        if ident.type == "user":
+            rc = self._rc("node", ident)
+            scope = rc.allowed_scope(Operation.VIEW)
+            if scope is None:
+                raise PermissionError_("no view permission on node")
+            if scope != Scope.GLOBAL and \
+                    node.organization_id not in self._org_ids_in_my_collabs(ident):
+                raise PermissionError_("node outside your collaborations")
+        info = _json.loads(node.info_json or "{}")
+        cols = (info.get("columns") or {}).get(label)
+        if cols is None:
+            raise NotFound(
+                f"no column info for database {label!r} on node {node_id} "
+                "(station offline or database has no tabular columns)")
+        return cols
 
     @_locked
     def node_set_offline(self, node_id: int) -> None:

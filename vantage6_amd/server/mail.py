@@ -55,3 +55,17 @@ def reset_password(core, token: str, new_password: str) -> None:
     user.set_password(new_password)
     user.failed_login_attempts = 0
     core.session.commit()
+
+
+def request_mfa_reset(core, mail: MailService, username: str) -> None:
+    """POST /api/recover/2fa/lost — emails an MFA-reset token (reference:
+    2FA reset flow). Always succeeds: no user enumeration."""
+    from .model import User
+
+    user = core.session.query(User).filter_by(username=username).first()
+    if user is None:
+        return
+    token = jwtlib.encode({"sub": user.id, "purpose": "mfa_reset"},
+                          core.jwt_secret, expires_in=3600)
+    mail.send(user.email or user.username, "mfa reset",
+              f"mfa reset token (valid 1h): {token}")

@@ -162,6 +162,9 @@ class Node(Authenticatable):
     collaboration_id = Column(Integer, ForeignKey("collaboration.id"))
     organization_id = Column(Integer, ForeignKey("organization.id"))
     device = Column(String, default="cpu")  # "cpu" or "cuda:<i>" (HIP device)
+    # node-info-update payload (reference: shared node config/capabilities;
+    # here: database labels -> column names, device info)
+    info_json = Column(Text, default="")
 
     collaboration = relationship("Collaboration", back_populates="nodes")
     organization = relationship("Organization", back_populates="nodes")
