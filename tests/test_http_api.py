@@ -44,6 +44,7 @@ def net(tmp_path):
 @pytest.fixture
 def server(net):
     app = create_app(net.core)
+    net.http_app = app  # tests reach app.state (mail outbox)
     config = uvicorn.Config(app, host="127.0.0.1", port=PORT, log_level="error")
     srv = uvicorn.Server(config)
     th = threading.Thread(target=srv.run, daemon=True)
@@ -220,3 +221,36 @@ def test_algorithm_store_http(server):
     assert r["status"] == "approved"
     listed = root.gw.request("GET", "algorithm", params={"status": "approved"})
     assert any(x["image"] == "algorithms.diagnostics" for x in listed["data"])
+
+
+def test_recover_password_and_mfa_over_http(server, net):
+    net.root_client.user.create("webrec", "old", organization_id=net.org_ids[0],
+                                roles=["Researcher"], email="w@example.org")
+    r = requests.post(f"{server}/api/recover/lost", json={"username": "webrec"})
+    assert r.status_code == 200
+    outbox = net.http_app.state.mail.outbox
+    token = outbox[-1]["body"].split(": ")[-1]
+    r = requests.post(f"{server}/api/recover/reset",
+                      json={"token": token, "password": "new"})
+    assert r.status_code == 200
+    r = requests.post(f"{server}/api/token/user",
+                      json={"username": "webrec", "password": "new"})
+    assert r.status_code == 200
+    # 2FA reset round-trip
+    r = requests.post(f"{server}/api/recover/2fa/lost", json={"username": "webrec"})
+    assert r.status_code == 200
+    token = net.http_app.state.mail.outbox[-1]["body"].split(": ")[-1]
+    r = requests.post(f"{server}/api/recover/2fa/reset", json={"token": token})
+    assert r.status_code == 200 and "mfa_secret" in r.json()
+
+
+def test_column_discovery_over_http(server, net):
+    # trigger an immediate heartbeat so column info is on the Node row
+    for st in net.stations:
+        st.core.node_ping(st.identity, {"columns": st._database_columns()})
+    tok = requests.post(f"{server}/api/token/user",
+                        json={"username": "root", "password": "root"}).json()
+    h = {"Authorization": f"Bearer {tok['access_token']}"}
+    nodes = requests.get(f"{server}/api/node", headers=h).json()["data"]
+    r = requests.get(f"{server}/api/column/{nodes[0]['id']}", headers=h)
+    assert r.status_code == 200 and r.json() == ["v"]

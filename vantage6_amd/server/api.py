@@ -30,6 +30,12 @@ class NodeTokenBody(BaseModel):
     api_key: str
 
 
+class RecoverBody(BaseModel):
+    username: str = ""
+    token: str = ""
+    password: str = ""
+
+
 class RefreshBody(BaseModel):
     refresh_token: str
 
@@ -352,11 +358,7 @@ def create_app(core: ServerCore) -> FastAPI:
     from . import mail as mail_mod
 
     mail_service = mail_mod.MailService()
-
-    class RecoverBody(BaseModel):
-        username: str = ""
-        token: str = ""
-        password: str = ""
+    app.state.mail = mail_service  # tests read the outbox
 
     @app.post("/api/recover/lost")
     def recover_lost(body: RecoverBody):
