@@ -147,3 +147,50 @@ def test_preprocessing_mock_client(csv_shards):
     )
     out = mock.result.from_task(task["id"])[0]
     assert out["counts"]["bmi"] == len(frames[0])
+
+
+def test_summary_statistics(tmp_path):
+    """Federated summary stats match pandas on the pooled data; a privacy
+    threshold above the shard size makes the task fail cleanly."""
+    import numpy as np
+    import pandas as pd
+
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+
+    rng = np.random.default_rng(11)
+    frames, paths = [], []
+    for i in range(2):
+        df = pd.DataFrame({"v": rng.normal(5, 3, 30 + 10 * i)})
+        df.loc[df.index[:3], "v"] = np.nan
+        p = tmp_path / f"sum{i}.csv"
+        df.to_csv(p, index=False)
+        frames.append(df)
+        paths.append(p)
+    pooled = pd.concat(frames)["v"]
+    net = DevNetwork(n_stations=2, databases=[
+        [DatabaseConfig(label="default", uri=str(p), type="csv")] for p in paths])
+    with net:
+        task = net.client.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],
+            name="summary", image="algorithms.summary_stats",
+            input_={"method": "central_summary",
+                    "kwargs": {"columns": ["v"]}})
+        (res,) = net.client.wait_for_results(task["id"])
+        st = res["v"]
+        assert st["n"] == int(pooled.count())
+        assert st["na"] == int(pooled.isna().sum())
+        assert st["mean"] == pytest.approx(pooled.mean(), rel=1e-9)
+        assert st["std"] == pytest.approx(pooled.std(), rel=1e-9)
+        assert st["min"] == pytest.approx(pooled.min())
+        assert st["max"] == pytest.approx(pooled.max())
+        # privacy threshold: shards are 30/40 rows — 1000 must fail
+        t2 = net.client.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],
+            name="summary-blocked", image="algorithms.summary_stats",
+            input_={"method": "central_summary",
+                    "kwargs": {"columns": ["v"], "min_rows": 1000}})
+        with pytest.raises(RuntimeError, match="PrivacyThresholdViolation"):
+            net.client.wait_for_results(t2["id"])

@@ -67,8 +67,21 @@ class AlgorithmClient:
                 status = task["status"]
                 if has_task_finished(status):
                     if has_task_failed(status):
+                        # surface the failed subtask logs so the parent
+                        # run's log names the real error (reference: child
+                        # logs are fetched by the researcher; algorithms
+                        # only see the status otherwise)
+                        logs = ""
+                        try:
+                            runs = self._proxy.list_runs(task_id)
+                            logs = "\n---\n".join(
+                                (r.get("log") or "") for r in runs
+                                if has_task_failed(r["status"]))
+                        except Exception:  # noqa: BLE001
+                            pass
                         raise CollectResultsError(
-                            f"task {task_id} ended with status {status!r}"
+                            f"task {task_id} ended with status {status!r}; "
+                            f"subtask logs:\n{logs}"
                         )
                     return self.result.from_task(task_id)
             if time.monotonic() - t0 > timeout:

@@ -28,3 +28,9 @@ class DataError(AlgorithmError):
 
 class CollectResultsError(AlgorithmError):
     """Subtask result collection failed (a partial crashed or was killed)."""
+
+
+class PrivacyThresholdViolation(AlgorithmError):
+    """A station refused to answer because its shard is below the privacy
+    threshold (reference: disclosure checks in community algorithms such as
+    v6-summary-statistics)."""

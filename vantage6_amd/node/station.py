@@ -306,7 +306,7 @@ class Station:
         except AlgorithmError as exc:
             self.core.patch_run(
                 self.identity, run_id, status=RunStatus.FAILED.value,
-                log=buf.getvalue() + f"\nAlgorithmError: {exc}",
+                log=buf.getvalue() + f"\n{type(exc).__name__}: {exc}",
             )
         except Exception:  # noqa: BLE001
             self.core.patch_run(
