@@ -194,3 +194,51 @@ def test_summary_statistics(tmp_path):
                     "kwargs": {"columns": ["v"], "min_rows": 1000}})
         with pytest.raises(RuntimeError, match="PrivacyThresholdViolation"):
             net.client.wait_for_results(t2["id"])
+
+
+def test_kaplan_meier(tmp_path):
+    """Federated KM equals the pooled product-limit estimator."""
+    import numpy as np
+    import pandas as pd
+
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+
+    rng = np.random.default_rng(21)
+    frames, paths = [], []
+    for i in range(2):
+        df = pd.DataFrame({
+            "t": rng.integers(1, 15, 50).astype(float),
+            "e": rng.integers(0, 2, 50),
+        })
+        p = tmp_path / f"km{i}.csv"
+        df.to_csv(p, index=False)
+        frames.append(df)
+        paths.append(p)
+    pooled = pd.concat(frames, ignore_index=True)
+
+    # plain pooled product-limit reference
+    at_risk = len(pooled)
+    surv_ref = {}
+    s = 1.0
+    for t in sorted(pooled["t"].unique()):
+        d = int(((pooled["t"] == t) & (pooled["e"] == 1)).sum())
+        c = int(((pooled["t"] == t) & (pooled["e"] == 0)).sum())
+        if d:
+            s *= 1.0 - d / at_risk
+        surv_ref[t] = s
+        at_risk -= d + c
+
+    net = DevNetwork(n_stations=2, databases=[
+        [DatabaseConfig(label="default", uri=str(p), type="csv")] for p in paths])
+    with net:
+        task = net.client.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],
+            name="km", image="algorithms.kaplan_meier",
+            input_={"method": "central_kaplan_meier",
+                    "kwargs": {"time_col": "t", "event_col": "e"}})
+        (res,) = net.client.wait_for_results(task["id"])
+    assert res["n"] == len(pooled)
+    for row in res["curve"]:
+        assert row["survival"] == pytest.approx(surv_ref[row["time"]], rel=1e-12)
