@@ -216,9 +216,13 @@ torch::Tensor colsum_bf16(torch::Tensor dy) {
   hipLaunchKernelGGL(zero_f32_kernel, dim3(zgrid), dim3(256), 0, cur_stream(),
                      ws.data_ptr<float>(), 16LL * N);
   int stripes = (N + 2047) / 2048;
-  int rchunk = 64;
+  // fill the chip (~256 workgroups) without piling atomics onto the
+  // 16-slice workspace: more stripes -> wider row-chunks
+  int target_gy = 256 / stripes > 0 ? 256 / stripes : 1;
+  int rchunk = (int)((M + target_gy - 1) / target_gy);
+  rchunk = ((rchunk + 15) / 16) * 16;
+  if (rchunk < 16) rchunk = 16;
   int gy = (int)((M + rchunk - 1) / rchunk);
-  if (gy > 1024) { rchunk = (int)((M + 1023) / 1024); gy = (int)((M + rchunk - 1) / rchunk); }
   hipLaunchKernelGGL(colsum_bf16_kernel, dim3(stripes, gy), dim3(256), 0,
                      cur_stream(), (const short*)dy.data_ptr(),
                      ws.data_ptr<float>(), M, N, rchunk);
