@@ -512,3 +512,62 @@ class TestRound1Additions:
         out = capsys.readouterr().out
         assert "info > hello" in out and "warn > careful" in out
         assert get_env_var("V6AMD_NOT_SET_XYZ", "dflt") == "dflt"
+
+
+class TestUserNodeCrud:
+    def test_user_crud_and_self_service(self, core, root):
+        orgs, _ = make_collab(root, name="crud")
+        root.user.create("crud-u", "pw1", organization_id=orgs[0],
+                         roles=["Researcher"])
+        u = UserClient(core=core)
+        u.authenticate("crud-u", "pw1")
+        me = [x for x in root.user.list(per_page=500)["data"]
+              if x["username"] == "crud-u"][0]
+        # self-service password change
+        u.user.update(me["id"], password="pw2", email="new@example.org")
+        u2 = UserClient(core=core)
+        u2.authenticate("crud-u", "pw2")
+        # a researcher may NOT delete other users
+        other = root.user.create("crud-v", "pw", organization_id=orgs[0],
+                                 roles=["Researcher"])
+        with pytest.raises(PermissionError_):
+            u.user.delete(other["id"])
+        root.user.delete(other["id"])
+        assert all(x["username"] != "crud-v"
+                   for x in root.user.list(per_page=500)["data"])
+
+    def test_role_escalation_blocked(self, core, root):
+        orgs, _ = make_collab(root, name="esc")
+        root.user.create("esc-admin", "pw", organization_id=orgs[0],
+                         roles=["Organization-Admin"])
+        adm = UserClient(core=core)
+        adm.authenticate("esc-admin", "pw")
+        # an org-admin (scope <= organization) must not mint a Root user
+        with pytest.raises(PermissionError_, match="do not hold"):
+            adm.user.create("evil", "pw", organization_id=orgs[0],
+                            roles=["Root"])
+        # nor grant Root to an existing user
+        target = root.user.create("esc-t", "pw", organization_id=orgs[0],
+                                  roles=["Researcher"])
+        with pytest.raises(PermissionError_):
+            adm.user.update(target["id"], roles=["Root"])
+
+    def test_node_rename_delete_and_collab_get(self, core, root):
+        orgs, collab = make_collab(root, name="nodecrud")
+        node = root.node.create(name="nc-n0", collaboration_id=collab["id"],
+                                organization_id=orgs[0])
+        out = root.node.update(node["id"], name="nc-renamed")
+        assert out["name"] == "nc-renamed"
+        c = root.collaboration.get(collab["id"])
+        assert c["id"] == collab["id"]
+        # outsider cannot fetch the collaboration
+        orgs2, _ = make_collab(root, name="nodecrud2")
+        root.user.create("nc-out", "pw", organization_id=orgs2[0],
+                         roles=["Researcher"])
+        out_c = UserClient(core=core)
+        out_c.authenticate("nc-out", "pw")
+        with pytest.raises(PermissionError_):
+            out_c.collaboration.get(collab["id"])
+        root.node.delete(node["id"])
+        assert all(n["name"] != "nc-renamed"
+                   for n in root.node.list(per_page=500)["data"])

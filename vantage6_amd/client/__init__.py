@@ -269,10 +269,8 @@ class UserClient:
                                      params=params, **params)
 
         def get(self, cid: int) -> dict:
-            for c in self.list(per_page=1000)["data"]:
-                if c["id"] == cid:
-                    return c
-            raise KeyError(cid)
+            return self.parent._call("get_collaboration", "GET",
+                                     f"collaboration/{cid}", collab_id=cid)
 
         def create(self, name: str, organization_ids: list[int], encrypted=False) -> dict:
             body = {"name": name, "organization_ids": organization_ids, "encrypted": encrypted}
@@ -282,6 +280,14 @@ class UserClient:
     class NodeSub(_Sub):
         def list(self, **params) -> dict:
             return self.parent._call("list_nodes", "GET", "node", params=params, **params)
+
+        def update(self, node_id: int, **fields) -> dict:
+            return self.parent._call("update_node", "PATCH", f"node/{node_id}",
+                                     json=fields, node_id=node_id, **fields)
+
+        def delete(self, node_id: int) -> dict:
+            return self.parent._call("delete_node", "DELETE", f"node/{node_id}",
+                                     node_id=node_id)
 
         def columns(self, node_id: int, label: str = "default") -> list[str]:
             """Data-column discovery (reference /api/column)."""
@@ -301,6 +307,22 @@ class UserClient:
             body = {"username": username, "password": password,
                     "organization_id": organization_id, "roles": roles or [], **kw}
             return self.parent._call("create_user", "POST", "user", json=body, **body)
+
+        def list(self, **params) -> dict:
+            return self.parent._call("list_users", "GET", "user", params=params,
+                                     **params)
+
+        def get(self, user_id: int) -> dict:
+            return self.parent._call("get_user", "GET", f"user/{user_id}",
+                                     user_id=user_id)
+
+        def update(self, user_id: int, **fields) -> dict:
+            return self.parent._call("update_user", "PATCH", f"user/{user_id}",
+                                     json=fields, user_id=user_id, **fields)
+
+        def delete(self, user_id: int) -> dict:
+            return self.parent._call("delete_user", "DELETE", f"user/{user_id}",
+                                     user_id=user_id)
 
     class RoleSub(_Sub):
         def list(self, **params) -> dict:

@@ -257,6 +257,38 @@ def create_app(core: ServerCore) -> FastAPI:
         return wrap(core.list_nodes, who, page=page, per_page=per_page,
                     collaboration_id=collaboration_id)
 
+    @app.get("/api/user")
+    def user_list(who: Identity = Depends(ident), page: int = 1,
+                  per_page: int = 20, organization_id: Optional[int] = None):
+        return core.list_users(who, page=page, per_page=per_page,
+                               organization_id=organization_id)
+
+    @app.get("/api/user/{user_id}")
+    def user_get(user_id: int, who: Identity = Depends(ident)):
+        return core.get_user(who, user_id)
+
+    @app.patch("/api/user/{user_id}")
+    def user_patch(user_id: int, body: dict, who: Identity = Depends(ident)):
+        allowed = {k: v for k, v in (body or {}).items() if k in (
+            "password", "email", "firstname", "lastname", "roles")}
+        return core.update_user(who, user_id, **allowed)
+
+    @app.delete("/api/user/{user_id}")
+    def user_delete(user_id: int, who: Identity = Depends(ident)):
+        return core.delete_user(who, user_id)
+
+    @app.patch("/api/node/{node_id}")
+    def node_patch(node_id: int, body: dict, who: Identity = Depends(ident)):
+        return core.update_node(who, node_id, name=(body or {}).get("name"))
+
+    @app.delete("/api/node/{node_id}")
+    def node_delete(node_id: int, who: Identity = Depends(ident)):
+        return core.delete_node(who, node_id)
+
+    @app.get("/api/collaboration/{collab_id}")
+    def collaboration_get(collab_id: int, who: Identity = Depends(ident)):
+        return core.get_collaboration(who, collab_id)
+
     @app.post("/api/user", status_code=201)
     def user_create(body: UserCreateBody, who: Identity = Depends(ident)):
         return wrap(core.create_user, who, body.username, body.password,

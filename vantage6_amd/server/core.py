@@ -343,7 +343,9 @@ class ServerCore:
             return
         if scope == Scope.COLLABORATION:
             if collab_id is None:
-                if org_id is not None and org_id == ident.organization_id:
+                if org_id is not None and (
+                        org_id == ident.organization_id
+                        or org_id in self._org_ids_in_my_collabs(ident)):
                     return
                 raise PermissionError_(f"{op.value} on {resource}: collaboration scope exceeded")
             s = self.session
@@ -463,15 +465,153 @@ class ServerCore:
                     **{k: v for k, v in kw.items() if k in ("firstname", "lastname", "email")})
         user.set_password(password)
         role_objs = []
-        for rname in roles or []:
-            role = s.query(Role).filter_by(name=rname).first()
-            if role is None:
-                raise NotFound(f"role {rname!r} not found")
-            role_objs.append(role)
+        if roles:
+            own = PermissionManager(s).rules_for_user(self._user(ident))
+            for rname in roles:
+                role = s.query(Role).filter_by(name=rname).first()
+                if role is None:
+                    raise NotFound(f"role {rname!r} not found")
+                for rule in role.rules:
+                    if (rule.name, rule.scope, rule.operation) not in own:
+                        raise PermissionError_(
+                            f"cannot assign role {rname!r}: it grants "
+                            f"{rule.name}/{rule.scope}/{rule.operation}, "
+                            "which you do not hold")
+                role_objs.append(role)
         user.roles = role_objs
         s.add(user)
         s.commit()
         return self._user_json(user)
+
+    def _check_can_touch_user(self, ident: Identity, target: User,
+                              op: Operation) -> None:
+        if ident.type != "user":
+            raise PermissionError_("users only")
+        if op == Operation.EDIT and target.id == ident.id:
+            return  # self-service (password/email change)
+        self._check_user_can("user", op, ident, org_id=target.organization_id)
+
+    @_locked
+    def list_users(self, ident: Identity, page=1, per_page=20,
+                   organization_id: int | None = None) -> dict:
+        rc = self._rc("user", ident)
+        scope = rc.allowed_scope(Operation.VIEW)
+        if scope is None:
+            raise PermissionError_("no view permission on user")
+        q = self.session.query(User).order_by(User.id)
+        users = q.all()
+        if scope == Scope.OWN:
+            users = [u for u in users if u.id == ident.id]
+        elif scope == Scope.ORGANIZATION:
+            users = [u for u in users
+                     if u.organization_id == ident.organization_id]
+        elif scope == Scope.COLLABORATION:
+            mine = self._org_ids_in_my_collabs(ident)
+            users = [u for u in users if u.organization_id in mine]
+        if organization_id is not None:
+            users = [u for u in users if u.organization_id == organization_id]
+        return self._paginate([self._user_json(u) for u in users], page, per_page)
+
+    @_locked
+    def get_user(self, ident: Identity, user_id: int) -> dict:
+        user = self.session.get(User, user_id)
+        if user is None:
+            raise NotFound(f"user {user_id} not found")
+        if user.id != ident.id:
+            self._check_user_can("user", Operation.VIEW, ident,
+                                 org_id=user.organization_id)
+        return self._user_json(user)
+
+    @_locked
+    def update_user(self, ident: Identity, user_id: int,
+                    password: str | None = None, email: str | None = None,
+                    firstname: str | None = None, lastname: str | None = None,
+                    roles: list[str] | None = None) -> dict:
+        user = self.session.get(User, user_id)
+        if user is None:
+            raise NotFound(f"user {user_id} not found")
+        self._check_can_touch_user(ident, user, Operation.EDIT)
+        if password is not None:
+            user.set_password(password)
+        if email is not None:
+            user.email = email
+        if firstname is not None:
+            user.firstname = firstname
+        if lastname is not None:
+            user.lastname = lastname
+        if roles is not None:
+            # role changes need user EDIT on the target org (not self-service)
+            # plus the anti-escalation check
+            self._check_user_can("user", Operation.EDIT, ident,
+                                 org_id=user.organization_id)
+            s = self.session
+            own = PermissionManager(s).rules_for_user(self._user(ident))
+            role_objs = []
+            for rname in roles:
+                role = s.query(Role).filter_by(name=rname).first()
+                if role is None:
+                    raise NotFound(f"role {rname!r} not found")
+                for rule in role.rules:
+                    if (rule.name, rule.scope, rule.operation) not in own:
+                        raise PermissionError_(
+                            f"cannot assign role {rname!r}: grants rules "
+                            "you do not hold")
+                role_objs.append(role)
+            user.roles = role_objs
+        self.session.commit()
+        return self._user_json(user)
+
+    @_locked
+    def delete_user(self, ident: Identity, user_id: int) -> dict:
+        user = self.session.get(User, user_id)
+        if user is None:
+            raise NotFound(f"user {user_id} not found")
+        self._check_can_touch_user(ident, user, Operation.DELETE)
+        self.session.delete(user)
+        self.session.commit()
+        return {"msg": f"user {user_id} deleted"}
+
+    @_locked
+    def update_node(self, ident: Identity, node_id: int,
+                    name: str | None = None) -> dict:
+        node = self.session.get(Node, node_id)
+        if node is None:
+            raise NotFound(f"node {node_id} not found")
+        self._check_user_can("node", Operation.EDIT, ident,
+                             org_id=node.organization_id)
+        if name is not None:
+            node.name = name
+        self.session.commit()
+        return self._node_json(node)
+
+    @_locked
+    def delete_node(self, ident: Identity, node_id: int) -> dict:
+        node = self.session.get(Node, node_id)
+        if node is None:
+            raise NotFound(f"node {node_id} not found")
+        self._check_user_can("node", Operation.DELETE, ident,
+                             org_id=node.organization_id)
+        self.session.delete(node)
+        self.session.commit()
+        return {"msg": f"node {node_id} deleted"}
+
+    @_locked
+    def get_collaboration(self, ident: Identity, collab_id: int) -> dict:
+        collab = self.session.get(Collaboration, collab_id)
+        if collab is None:
+            raise NotFound(f"collaboration {collab_id} not found")
+        if ident.type == "user":
+            rc = self._rc("collaboration", ident)
+            scope = rc.allowed_scope(Operation.VIEW)
+            if scope is None:
+                raise PermissionError_("no view permission on collaboration")
+            if scope != Scope.GLOBAL and not any(
+                    o.id == ident.organization_id for o in collab.organizations):
+                raise PermissionError_("not a member of this collaboration")
+        elif ident.type in ("node", "container"):
+            if collab_id != ident.collaboration_id:
+                raise PermissionError_("collaboration mismatch")
+        return self._collab_json(collab)
 
     # -------------------------------------------------------------- listings
     def _paginate(self, items: list, page: int = 1, per_page: int = 20) -> dict:
