@@ -18,7 +18,7 @@ aggregate in Python}. Here a round is:
 from __future__ import annotations
 
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Callable, Iterator, Optional
 
 import torch
