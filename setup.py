@@ -7,7 +7,7 @@ the repo snapshot to GPU boxes (JIT caches under ~/.cache do not).
 import os
 from pathlib import Path
 
-from setuptools import setup
+from setuptools import find_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -20,7 +20,13 @@ sources = [str(CSRC / f) for f in
            ("bindings.hip", "elementwise.hip", "norms.hip", "gemm.hip", "bn.hip")]
 
 setup(
-    name="vantage6_amd_ops",
+    name="vantage6_amd",
+    version="0.1.0",
+    description="MI355X-native privacy-preserving federated learning "
+                "framework (vantage6-capability surface)",
+    packages=find_packages(include=["vantage6_amd", "vantage6_amd.*",
+                                    "algorithms"]),
+    entry_points={"console_scripts": ["v6amd = vantage6_amd.cli.main:cli"]},
     ext_modules=[
         CUDAExtension(
             name="vantage6_amd.ops._hip_ops",

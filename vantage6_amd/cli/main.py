@@ -123,6 +123,21 @@ def dev_demo(stations, encrypted):
     click.echo(json.dumps({"demo": "column_mean", "result": res[0]}, indent=2))
 
 
+@cli.command(name="create-private-key")
+@click.option("--output", "output_path", required=True, type=click.Path(),
+              help="where to write the RSA private key")
+@click.option("--bits", default=2048)
+def create_private_key(output_path, bits):
+    """Generate an organization RSA keypair (reference:
+    ``v6 node create-private-key``). The public key is printed so it can be
+    uploaded via ``client.organization`` / the station does it on start."""
+    from ..common.encryption import RSACryptor
+
+    cryptor = RSACryptor(output_path, bits=bits)
+    click.echo(f"private key written to {output_path}")
+    click.echo(f"public key (b64):\n{cryptor.public_key_str}")
+
+
 @cli.command(name="import")
 @click.option("--config", "config_path", required=True,
               type=click.Path(exists=True), help="entity fixture YAML")
