@@ -254,3 +254,66 @@ def test_column_discovery_over_http(server, net):
     nodes = requests.get(f"{server}/api/node", headers=h).json()["data"]
     r = requests.get(f"{server}/api/column/{nodes[0]['id']}", headers=h)
     assert r.status_code == 200 and r.json() == ["v"]
+
+
+def test_remote_station_over_http(tmp_path):
+    """Full reference deployment shape: orchestrator behind HTTP, a Station
+    attached via RemoteCore (api-key auth, event/polling pickup, result
+    PATCH), researcher over the HTTP client — column mean end-to-end with
+    NOTHING sharing memory with the server core."""
+    from vantage6_amd.client import UserClient
+    from vantage6_amd.common.config import DatabaseConfig, StationConfig
+    from vantage6_amd.node.remote import RemoteCore
+    from vantage6_amd.node.station import Station
+    from vantage6_amd.server.core import ServerCore
+
+    df = pd.DataFrame({"v": [1.0, 2.0, 3.0, 4.0]})
+    csv = tmp_path / "remote.csv"
+    df.to_csv(csv, index=False)
+
+    core = ServerCore()
+    core.create_root_user("root", "root")
+    root = UserClient(core=core)
+    root.authenticate("root", "root")
+    org = root.organization.create(name="remote-org")
+    collab = root.collaboration.create(name="remote-c",
+                                       organization_ids=[org["id"]])
+    node = root.node.create(name="remote-n", collaboration_id=collab["id"],
+                            organization_id=org["id"])
+
+    app = create_app(core)
+    port = PORT + 7
+    config = uvicorn.Config(app, host="127.0.0.1", port=port, log_level="error")
+    srv = uvicorn.Server(config)
+    th = threading.Thread(target=srv.run, daemon=True)
+    th.start()
+    url = f"http://127.0.0.1:{port}"
+    for _ in range(100):
+        try:
+            requests.get(f"{url}/api/version", timeout=1)
+            break
+        except Exception:
+            time.sleep(0.05)
+
+    remote = RemoteCore(url)
+    remote.events.interval = 0.1  # fast polling for the test
+    station = Station(remote, StationConfig(
+        name="remote-station", organization_id=org["id"], device="cpu",
+        databases=[DatabaseConfig(label="default", uri=str(csv), type="csv")]),
+        api_key=node["api_key"])
+    try:
+        station.start()
+        web = UserClient(url=url)
+        web.authenticate("root", "root")
+        task = web.task.create(
+            collaboration=collab["id"], organizations=[org["id"]],
+            name="remote-mean", image="algorithms.column_mean",
+            input_={"method": "partial_mean", "kwargs": {"columns": ["v"]}})
+        (res,) = web.wait_for_results(task["id"], timeout=30)
+        assert res["sums"]["v"] == pytest.approx(10.0)
+        assert res["counts"]["v"] == 4
+    finally:
+        station.stop()
+        remote.close()
+        srv.should_exit = True
+        th.join(timeout=5)

@@ -17,7 +17,6 @@ from ..common import bytes_to_base64s
 from ..common.encryption import CryptorBase
 from ..common.serialization import deserialize, serialize
 from ..server.core import Identity, ServerCore
-from ..server.model import Collaboration
 
 
 class NodeProxy:
@@ -101,7 +100,7 @@ class NodeProxy:
         return self._open_result(run["result"])
 
     def list_organizations(self) -> list[dict]:
-        collab = self.core.session.get(Collaboration, self.ident.collaboration_id)
-        if collab is None:
-            return []
-        return [{"id": o.id, "name": o.name} for o in collab.organizations]
+        collab = self.core.get_collaboration(self.ident,
+                                             self.ident.collaboration_id)
+        return [{"id": o["id"], "name": o["name"]}
+                for o in collab["organizations"]]

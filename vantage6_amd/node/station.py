@@ -28,7 +28,6 @@ from ..common.encryption import CryptorBase, RSACryptor
 from ..common.serialization import deserialize, serialize
 from ..common.task_status import RunStatus
 from ..server.core import Identity, ServerCore
-from ..server.model import Organization
 from ..algorithm.client import AlgorithmClient
 from ..algorithm.exceptions import AlgorithmError, InputError
 from ..algorithm.wrap import dispatch, resolve_module
@@ -76,9 +75,9 @@ class Station:
         self.identity = self.core.identity_from_token(token["access_token"])
         # upload our public key so clients/nodes can encrypt for our org
         if isinstance(self.cryptor, RSACryptor):
-            org = self.core.session.get(Organization, self.identity.organization_id)
-            org.public_key = self.cryptor.public_key_str
-            self.core.session.commit()
+            self.core.set_organization_public_key(
+                self.identity, self.identity.organization_id,
+                self.cryptor.public_key_str)
         cid, oid = self.identity.collaboration_id, self.identity.organization_id
         self.core.events.subscribe(f"collaboration_{cid}_organization_{oid}", self._on_event)
         self.core.events.subscribe(f"collaboration_{cid}", self._on_event)

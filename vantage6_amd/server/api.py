@@ -417,6 +417,14 @@ def create_app(core: ServerCore) -> FastAPI:
         return core.get_columns(who, node_id, label)
 
     # ----------------------------------------------------------------- ops
+    @app.post("/api/ping")
+    def ping(body: dict, who: Identity = Depends(ident)):
+        if (body or {}).get("offline"):
+            core.node_set_offline(who.id)
+        else:
+            core.node_ping(who, (body or {}).get("info") or None)
+        return {"msg": "pong"}
+
     @app.get("/api/version")
     def version():
         return core.version()

@@ -376,7 +376,8 @@ class ServerCore:
         # uploading your OWN organization's public key is part of
         # setup_encryption and needs no org-edit permission (reference:
         # organization PATCH at OWN scope)
-        if not (ident.type == "user" and org_id == ident.organization_id):
+        if not (ident.type in ("user", "node")
+                and org_id == ident.organization_id):
             self._check_user_can("organization", Operation.EDIT, ident, org_id=org_id)
         s = self.session
         org = s.get(Organization, org_id)
