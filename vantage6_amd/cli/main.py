@@ -86,6 +86,42 @@ def server(config_path, port, stations):
     uvicorn.run(app, host="127.0.0.1", port=port, log_level="info")
 
 
+@cli.command()
+@click.option("--config", "config_path", required=True,
+              type=click.Path(exists=True), help="StationConfig YAML")
+@click.option("--server", "server_url", required=True,
+              help="orchestrator URL, e.g. http://host:7601")
+@click.option("--api-key", envvar="V6AMD_API_KEY", required=True,
+              help="node api-key (or env V6AMD_API_KEY)")
+@click.option("--private-key", "private_key", type=click.Path(), default=None,
+              help="org RSA key file for encrypted collaborations")
+def node(config_path, server_url, api_key, private_key):
+    """Start a data station against a REMOTE orchestrator (reference:
+    ``v6 node start`` — the node-daemon deployment shape). Runs until
+    interrupted."""
+    import time as _time
+
+    from ..common.config import StationConfig, load_config
+    from ..node.remote import RemoteCore
+    from ..node.station import Station
+
+    cfg = load_config(config_path, model=StationConfig)
+    core = RemoteCore(server_url)
+    station = Station(core, cfg, api_key=api_key,
+                      private_key_file=private_key)
+    station.start()
+    click.echo(f"station {cfg.name!r} online against {server_url} "
+               f"(device {cfg.device}); ctrl-c to stop")
+    try:
+        while True:
+            _time.sleep(1.0)
+    except KeyboardInterrupt:
+        pass
+    finally:
+        station.stop()
+        core.close()
+
+
 @cli.group()
 def dev():
     """Demo-network commands (integration harness)."""
