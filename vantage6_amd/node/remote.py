@@ -102,27 +102,49 @@ class _PollingEvents(_WsEvents):
 
     interval = 1.0
 
+    def _dispatch(self, event: str, data: dict) -> None:
+        for cb in list(self._callbacks):
+            try:
+                cb(event, data)
+            except Exception:  # noqa: BLE001
+                log.exception("event callback failed")
+
     def _run(self) -> None:
         seen: set[int] = set()
+        open_now: dict[int, int] = {}  # run id -> task id
         while not self._stop.wait(self.interval):
             try:
                 runs = self._remote.gw.request(
                     "GET", "run", params={"state": "open"})["data"]
             except Exception:  # noqa: BLE001
                 continue
+            ids = {r["id"] for r in runs}
             for r in runs:
                 if r["id"] in seen:
                     continue
                 seen.add(r["id"])
-                for cb in list(self._callbacks):
-                    try:
-                        cb("new_task", {
-                            "task_id": r["task_id"], "run_id": r["id"],
-                            # node-token listings only contain OUR runs
-                            "organization_id": self._remote.org_id,
-                        })
-                    except Exception:  # noqa: BLE001
-                        log.exception("event callback failed")
+                open_now[r["id"]] = r["task_id"]
+                self._dispatch("new_task", {
+                    "task_id": r["task_id"], "run_id": r["id"],
+                    # node-token listings only contain OUR runs
+                    "organization_id": self._remote.org_id,
+                })
+            # a run that LEFT the open set without us finishing it was
+            # terminated server-side — check for a user kill
+            for rid in [x for x in open_now if x not in ids]:
+                tid = open_now.pop(rid)
+                try:
+                    status = self._remote.gw.request(
+                        "GET", f"run/{rid}")["status"]
+                except Exception:  # noqa: BLE001
+                    continue
+                if status == "killed by user":
+                    self._dispatch("kill_containers", {
+                        "task_id": tid,
+                        "kill_list": [{
+                            "run_id": rid, "task_id": tid,
+                            "organization_id": self._remote.org_id}],
+                    })
 
 
 class RemoteCore:
