@@ -77,6 +77,9 @@ def main() -> None:
     ap.add_argument("--no-graph", dest="graph", action="store_false")
     ap.add_argument("--dtype", default="bf16", choices=["bf16"])
     ap.add_argument("--device", default=None, help="override (e.g. cpu for smoke)")
+    ap.add_argument("--target-loss", type=float, default=None,
+                    help="also report rounds/seconds until loss < target "
+                         "(BASELINE metric's wall-clock-to-target-loss half)")
     args = ap.parse_args()
 
     comm = init_from_env()
@@ -190,11 +193,19 @@ def main() -> None:
                 "secure_agg": args.secure_agg,
                 "dp": args.dp,
                 "hip_graph": used_graph,
+                "loss_first": losses[0],
                 "final_loss": losses[-1],
                 "n_params": trainer.flat.numel,
                 "hip_ops": __import__("vantage6_amd.ops", fromlist=["has_hip_ops"]).has_hip_ops(),
             },
         }
+        if args.target_loss is not None:
+            hit = next((i for i, l in enumerate(losses)
+                        if l == l and l < args.target_loss), None)
+            out["config"]["target_loss"] = args.target_loss
+            out["config"]["rounds_to_target"] = None if hit is None else hit + 1
+            out["config"]["secs_to_target"] = (
+                None if hit is None else (hit + 1) * elapsed / args.steps)
         # per-round timing table on stderr; stdout stays
         # the single JSON contract line
         print(metrics.table(last=min(args.steps, 12)), file=sys.stderr)
