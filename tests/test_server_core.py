@@ -571,3 +571,22 @@ class TestUserNodeCrud:
         root.node.delete(node["id"])
         assert all(n["name"] != "nc-renamed"
                    for n in root.node.list(per_page=500)["data"])
+
+
+def test_killed_run_is_terminal(core, root):
+    orgs, collab = make_collab(root, name="killterm")
+    node = root.node.create(name="kt-n", collaboration_id=collab["id"],
+                            organization_id=orgs[0])
+    t = root.task.create(collaboration=collab["id"], organizations=[orgs[0]],
+                         name="kt", image="x", input_={"method": "m"})
+    root.task.delete(t["id"])  # kill
+    run = root.run.from_task(t["id"])[0]
+    assert run["status"] == "killed by user"
+    # the node later reports completion — must NOT resurrect the run
+    tok = core.authenticate_node(node["api_key"])["access_token"]
+    nid = core.identity_from_token(tok)
+    core.patch_run(nid, run["id"], status="completed", result="xx",
+                   log="late finish")
+    run2 = core.get_run(root.gw.ident, run["id"])
+    assert run2["status"] == "killed by user"
+    assert "late finish" in (run2["log"] or "")

@@ -893,6 +893,11 @@ class ServerCore:
             self._check_user_can("run", Operation.EDIT, ident,
                                  org_id=run.organization_id)
         old_status = run.status
+        if (old_status == RunStatus.KILLED.value and ident.type != "user"
+                and fields.get("status") != RunStatus.KILLED.value):
+            # killed is terminal: a node finishing the (already-killed)
+            # computation later may append its log but not resurrect the run
+            fields = {k: v for k, v in fields.items() if k == "log"}
         for key in ("status", "result", "log", "input"):
             if key in fields and fields[key] is not None:
                 setattr(run, key, str(fields[key]) if key == "status" else fields[key])
