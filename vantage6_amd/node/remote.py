@@ -114,7 +114,7 @@ class _PollingEvents(_WsEvents):
         open_now: dict[int, int] = {}  # run id -> task id
         while not self._stop.wait(self.interval):
             try:
-                runs = self._remote.gw.request(
+                runs = self._remote._request(
                     "GET", "run", params={"state": "open"})["data"]
             except Exception:  # noqa: BLE001
                 continue
@@ -134,7 +134,7 @@ class _PollingEvents(_WsEvents):
             for rid in [x for x in open_now if x not in ids]:
                 tid = open_now.pop(rid)
                 try:
-                    status = self._remote.gw.request(
+                    status = self._remote._request(
                         "GET", f"run/{rid}")["status"]
                 except Exception:  # noqa: BLE001
                     continue
@@ -175,13 +175,28 @@ class RemoteCore:
 
     # ----------------------------------------------------------------- auth
     org_id: int | None = None  # set at authenticate; used by event synth
+    _api_key: str | None = None
 
     def authenticate_node(self, api_key: str) -> dict:
+        self._api_key = api_key
         out = self.gw.request("POST", "token/node", json={"api_key": api_key})
         self.gw.token = out["access_token"]
         self.gw.refresh_token_value = out.get("refresh_token")
         self.org_id = _decode_payload(out["access_token"]).get("org")
         return out
+
+    def _request(self, verb: str, path: str, **kw):
+        """Gateway call that survives full token expiry: on a 401 (even
+        after the gateway's own refresh attempt), re-authenticate with the
+        stored api-key and retry once (reference: node re-auths on the
+        ``expired_token`` socket event)."""
+        try:
+            return self.gw.request(verb, path, **kw)
+        except RuntimeError as exc:
+            if "401" not in str(exc) or self._api_key is None:
+                raise
+            self.authenticate_node(self._api_key)
+            return self.gw.request(verb, path, **kw)
 
     def identity_from_token(self, token: str) -> Identity:
         p = _decode_payload(token)
@@ -197,52 +212,52 @@ class RemoteCore:
 
     # ---------------------------------------------------------- node runtime
     def node_ping(self, ident, info: dict | None = None) -> None:
-        self.gw.request("POST", "ping", json={"info": info or {}})
+        self._request("POST", "ping", json={"info": info or {}})
 
     def node_set_offline(self, node_id: int) -> None:
         try:
-            self.gw.request("POST", "ping", json={"info": {}, "offline": True})
+            self._request("POST", "ping", json={"info": {}, "offline": True})
         except Exception:  # noqa: BLE001 — best-effort during shutdown
             pass
 
     def node_sync_open_runs(self, ident) -> list[dict]:
-        return self.gw.request("GET", "run", params={"state": "open"})["data"]
+        return self._request("GET", "run", params={"state": "open"})["data"]
 
     def set_organization_public_key(self, ident, org_id: int,
                                     public_key: str) -> dict:
-        return self.gw.request("PATCH", f"organization/{org_id}",
+        return self._request("PATCH", f"organization/{org_id}",
                                json={"public_key": public_key})
 
     # ------------------------------------------------------- task/run plane
     def get_task(self, ident, task_id: int) -> dict:
-        return self.gw.request("GET", f"task/{task_id}")
+        return self._request("GET", f"task/{task_id}")
 
     def list_runs(self, ident, **params) -> dict:
         params = {k: v for k, v in params.items() if v is not None}
-        return self.gw.request("GET", "run", params=params)
+        return self._request("GET", "run", params=params)
 
     def get_run(self, ident, run_id: int) -> dict:
-        return self.gw.request("GET", f"run/{run_id}")
+        return self._request("GET", f"run/{run_id}")
 
     def patch_run(self, ident, run_id: int, **fields) -> dict:
-        return self.gw.request("PATCH", f"run/{run_id}", json=fields)
+        return self._request("PATCH", f"run/{run_id}", json=fields)
 
     def list_results(self, ident, task_id: int, **kw) -> dict:
-        return self.gw.request("GET", "result", params={"task_id": task_id})
+        return self._request("GET", "result", params={"task_id": task_id})
 
     def create_task(self, ident, **body) -> dict:
-        return self.gw.request("POST", "task", json=body)
+        return self._request("POST", "task", json=body)
 
     # --------------------------------------------------------------- lookups
     def organization_public_key(self, org_id: int) -> str:
-        return self.gw.request("GET", f"organization/{org_id}").get(
+        return self._request("GET", f"organization/{org_id}").get(
             "public_key") or ""
 
     def get_collaboration(self, ident, collab_id: int) -> dict:
-        return self.gw.request("GET", f"collaboration/{collab_id}")
+        return self._request("GET", f"collaboration/{collab_id}")
 
     def algorithm_is_approved(self, image: str) -> bool:
-        rows = self.gw.request("GET", "algorithm",
+        rows = self._request("GET", "algorithm",
                                params={"status": "approved",
                                        "per_page": 1000})["data"]
         return any(a["image"] == image for a in rows)
