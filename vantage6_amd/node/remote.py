@@ -207,8 +207,8 @@ class RemoteCore:
         )
 
     def token_for_container(self, ident, task_id: int, image: str) -> dict:
-        return self.gw.request("POST", "token/container",
-                               json={"task_id": task_id, "image": image})
+        return self._request("POST", "token/container",
+                             json={"task_id": task_id, "image": image})
 
     # ---------------------------------------------------------- node runtime
     def node_ping(self, ident, info: dict | None = None) -> None:
