@@ -62,3 +62,67 @@ def test_api_mode_central_fedavg_on_gpu():
     assert len(out["history"]) == 3
     # loss should drop across rounds on this separable synthetic data
     assert out["history"][-1]["mean_loss"] < out["history"][0]["mean_loss"]
+
+
+def test_remote_station_partial_fit_on_gpu():
+    """Remote deployment shape ON the GPU: orchestrator behind uvicorn, a
+    cuda:0 station attached over HTTP via RemoteCore, partial_fit runs on
+    the HIP kernels — nothing shares memory with the server core."""
+    import threading
+    import time
+
+    import requests
+    import uvicorn
+
+    from vantage6_amd.client import UserClient
+    from vantage6_amd.common.config import DatabaseConfig, StationConfig
+    from vantage6_amd.node.remote import RemoteCore
+    from vantage6_amd.node.station import Station
+    from vantage6_amd.server.api import create_app
+    from vantage6_amd.server.core import ServerCore
+
+    core = ServerCore()
+    core.create_root_user("root", "root")
+    root = UserClient(core=core)
+    root.authenticate("root", "root")
+    org = root.organization.create(name="gpu-remote-org")
+    collab = root.collaboration.create(name="gpu-remote-c",
+                                       organization_ids=[org["id"]])
+    node = root.node.create(name="gpu-remote-n",
+                            collaboration_id=collab["id"],
+                            organization_id=org["id"])
+    app = create_app(core)
+    srv = uvicorn.Server(uvicorn.Config(app, host="127.0.0.1", port=7741,
+                                        log_level="error"))
+    threading.Thread(target=srv.run, daemon=True).start()
+    url = "http://127.0.0.1:7741"
+    for _ in range(100):
+        try:
+            requests.get(f"{url}/api/version", timeout=1)
+            break
+        except Exception:
+            time.sleep(0.05)
+    remote = RemoteCore(url)
+    remote.events.interval = 0.1
+    station = Station(remote, StationConfig(
+        name="gpu-remote-station", organization_id=org["id"], device="cuda:0",
+        databases=[DatabaseConfig(
+            label="default", uri="synthetic://tabular?n=256&d=16&seed=4",
+            type="synthetic")]), api_key=node["api_key"])
+    try:
+        station.start()
+        web = UserClient(url=url)
+        web.authenticate("root", "root")
+        task = web.task.create(
+            collaboration=collab["id"], organizations=[org["id"]],
+            name="gpu remote partial", image="algorithms.fedavg",
+            input_={"method": "partial_fit", "kwargs": {
+                "model": "logreg",
+                "model_kwargs": {"in_features": 16, "num_classes": 2},
+                "local_steps": 2, "batch_size": 32}})
+        (res,) = web.wait_for_results(task["id"], timeout=120)
+        assert res["n_samples"] == 256 and "fc.weight" in res["params"]
+    finally:
+        station.stop()
+        remote.close()
+        srv.should_exit = True
