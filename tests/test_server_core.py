@@ -590,3 +590,30 @@ def test_killed_run_is_terminal(core, root):
     run2 = core.get_run(root.gw.ident, run["id"])
     assert run2["status"] == "killed by user"
     assert "late finish" in (run2["log"] or "")
+
+
+def test_study_list_get_scoped(core, root):
+    orgs, collab = make_collab(root, n_orgs=3, name="stud")
+    st = root.study.create("substudy", collab["id"], orgs[:2])
+    got = root.study.get(st["id"])
+    assert {o["id"] for o in got["organizations"]} == set(orgs[:2])
+    assert any(x["id"] == st["id"]
+               for x in root.study.list(collaboration_id=collab["id"])["data"])
+    # member researcher sees it; outsider does not
+    root.user.create("stud-m", "pw", organization_id=orgs[0],
+                     roles=["Researcher"])
+    m = UserClient(core=core)
+    m.authenticate("stud-m", "pw")
+    assert m.study.get(st["id"])["id"] == st["id"]
+    orgs2, _ = make_collab(root, name="stud2")
+    root.user.create("stud-o", "pw", organization_id=orgs2[0],
+                     roles=["Researcher"])
+    o = UserClient(core=core)
+    o.authenticate("stud-o", "pw")
+    with pytest.raises(PermissionError_):
+        o.study.get(st["id"])
+    # task targeting a non-member of the study is rejected
+    with pytest.raises(Exception, match="study members"):
+        root.task.create(collaboration=collab["id"], organizations=[orgs[2]],
+                         name="t", image="x", input_={"method": "m"},
+                         study=st["id"])

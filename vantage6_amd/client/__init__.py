@@ -357,6 +357,14 @@ class UserClient:
                                      params=params, **params)
 
     class StudySub(_Sub):
+        def list(self, **params) -> dict:
+            return self.parent._call("list_studies", "GET", "study",
+                                     params=params, **params)
+
+        def get(self, study_id: int) -> dict:
+            return self.parent._call("get_study", "GET", f"study/{study_id}",
+                                     study_id=study_id)
+
         def create(self, name: str, collaboration_id: int,
                    organization_ids: list[int]) -> dict:
             body = {"name": name, "collaboration_id": collaboration_id,

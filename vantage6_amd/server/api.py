@@ -295,6 +295,17 @@ def create_app(core: ServerCore) -> FastAPI:
                     body.organization_id, body.roles, firstname=body.firstname,
                     lastname=body.lastname, email=body.email)
 
+    @app.get("/api/study")
+    def study_list(who: Identity = Depends(ident), page: int = 1,
+                   per_page: int = 20,
+                   collaboration_id: Optional[int] = None):
+        return core.list_studies(who, collaboration_id=collaboration_id,
+                                 page=page, per_page=per_page)
+
+    @app.get("/api/study/{study_id}")
+    def study_get(study_id: int, who: Identity = Depends(ident)):
+        return core.get_study(who, study_id)
+
     @app.post("/api/study", status_code=201)
     def study_create(body: StudyCreateBody, who: Identity = Depends(ident)):
         return wrap(core.create_study, who, body.name, body.collaboration_id,

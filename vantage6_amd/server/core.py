@@ -358,6 +358,47 @@ class ServerCore:
             return
         raise PermissionError_(f"{op.value} on {resource}: organization scope exceeded")
 
+    @_locked
+    def list_studies(self, ident: Identity, collaboration_id: int | None = None,
+                     page=1, per_page=20) -> dict:
+        rc = self._rc("study", ident)
+        scope = rc.allowed_scope(Operation.VIEW)
+        if scope is None:
+            raise PermissionError_("no view permission on study")
+        studies = self.session.query(Study).order_by(Study.id).all()
+        if scope != Scope.GLOBAL:
+            me = self.session.get(Organization, ident.organization_id)
+            my_collabs = {c.id for c in me.collaborations} if me else set()
+            studies = [st for st in studies
+                       if st.collaboration_id in my_collabs]
+        if collaboration_id is not None:
+            studies = [st for st in studies
+                       if st.collaboration_id == collaboration_id]
+        return self._paginate([self._study_json(st) for st in studies],
+                              page, per_page)
+
+    @_locked
+    def get_study(self, ident: Identity, study_id: int) -> dict:
+        study = self.session.get(Study, study_id)
+        if study is None:
+            raise NotFound(f"study {study_id} not found")
+        rc = self._rc("study", ident)
+        scope = rc.allowed_scope(Operation.VIEW)
+        if scope is None:
+            raise PermissionError_("no view permission on study")
+        if scope != Scope.GLOBAL:
+            collab = self.session.get(Collaboration, study.collaboration_id)
+            if not any(o.id == ident.organization_id
+                       for o in (collab.organizations if collab else [])):
+                raise PermissionError_("study outside your collaborations")
+        return self._study_json(study)
+
+    def _study_json(self, st: Study) -> dict:
+        return {"id": st.id, "name": st.name,
+                "collaboration_id": st.collaboration_id,
+                "organizations": [{"id": o.id, "name": o.name}
+                                  for o in st.organizations]}
+
     # ------------------------------------------------------ entity management
     @_locked
     def create_organization(self, ident: Identity, name: str, **kw) -> dict:
