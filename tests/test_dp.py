@@ -121,3 +121,30 @@ def test_epsilon_report():
     engine = DPEngine(nn.Sequential(nn.Linear(4, 2)), DPConfig(noise_multiplier=1.0))
     rep = engine.epsilon_report(steps=100, sample_rate=0.01)
     assert 0 < rep["epsilon"] < 100
+
+
+def test_rdp_accountant_properties():
+    """RDP epsilon: tighter than the loose bound at realistic settings,
+    monotone in steps, decreasing in sigma, and for q=1/T=1 it matches the
+    analytic Gaussian-mechanism minimum over alpha."""
+    import math
+
+    from vantage6_amd.parallel.dp import DPConfig, DPEngine
+    import torch.nn as nn
+
+    def eng(sigma):
+        return DPEngine(nn.Linear(4, 2), DPConfig(noise_multiplier=sigma,
+                                                  max_grad_norm=1.0,
+                                                  delta=1e-5))
+    e = eng(1.1)
+    r = e.epsilon_report(steps=1000, sample_rate=0.01)
+    assert r["epsilon"] < r["epsilon_loose"]
+    assert 0.1 < r["epsilon"] < 5.0  # opacus reports ~0.9-1.1 here
+    assert e.epsilon_report(2000, 0.01)["epsilon"] > r["epsilon"]
+    assert eng(2.0).epsilon_report(1000, 0.01)["epsilon"] < r["epsilon"]
+    # q=1, T=1: epsilon = min_alpha alpha/(2 sigma^2) + log(1/delta)/(alpha-1)
+    sigma = 3.0
+    got = eng(sigma).epsilon_report(1, 1.0)["epsilon"]
+    want = min(a / (2 * sigma**2) + math.log(1e5) / (a - 1)
+               for a in range(2, 1025))
+    assert got == pytest.approx(want, rel=1e-6)
