@@ -98,7 +98,7 @@ def fit_rounds(
         history.append(stats)
         if round_hook:
             round_hook(trainer, stats)
-    return {
+    out = {
         "rounds": rounds,
         "final_loss": history[-1]["loss"],
         "history": history,
@@ -106,6 +106,10 @@ def fit_rounds(
         # post-aggregation master checksum: identical across ranks by design
         "master_checksum": float(trainer.flat.master_flat.double().sum().item()),
     }
+    if dp_cfg is not None:
+        out["dp_privacy"] = trainer.dp_engine.epsilon_report(
+            steps=rounds * local_steps, sample_rate=min(1.0, batch_size / n))
+    return out
 
 
 # --------------------------------------------------------------------------

@@ -193,6 +193,10 @@ def main() -> None:
                 "secure_agg": args.secure_agg,
                 "dp": args.dp,
                 "hip_graph": used_graph,
+                "dp_privacy": (trainer.dp_engine.epsilon_report(
+                    steps=args.steps * args.local_steps,
+                    sample_rate=min(1.0, batch / shard))
+                    if args.dp else None),
                 "loss_first": losses[0],
                 "final_loss": losses[-1],
                 "n_params": trainer.flat.numel,
