@@ -223,6 +223,13 @@ class FedAvgTrainer:
         torch reference on CPU); all-reduce(SUM) wraps mod 2^64 so masks
         cancel exactly; dequantize. The server/other stations only ever see
         masked integers.
+
+        Limitation (same trade-off the reference ecosystem's basic
+        pairwise-mask schemes make): no dropout recovery — if a rank dies
+        MID-collective, its pair masks do not cancel and the round result
+        is garbage; the collective itself also fails, so in practice the
+        job restarts from the last checkpoint. Bonawitz-style secret-shared
+        mask recovery is a round-2+ item.
         """
         from ..ops import secagg as secagg_ops
 
