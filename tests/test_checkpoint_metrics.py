@@ -61,3 +61,16 @@ def test_metrics_table():
     table = m.table()
     assert "rounds/s" in table
     assert len(table.splitlines()) >= 5
+
+
+def test_prometheus_endpoint_scrape():
+    """RoundMetrics with a port exposes rounds/phase metrics over HTTP."""
+    import requests
+
+    from vantage6_amd.utils.metrics import RoundMetrics
+
+    m = RoundMetrics(prometheus_port=7791)
+    m.record({"loss": 1.5, "fit_s": 0.01, "agg_s": 0.002, "total_s": 0.012})
+    m.record({"loss": 1.2, "fit_s": 0.011, "agg_s": 0.002, "total_s": 0.013})
+    body = requests.get("http://127.0.0.1:7791/metrics", timeout=5).text
+    assert "v6amd_rounds_total" in body or "rounds" in body, body[:400]
