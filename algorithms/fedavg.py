@@ -71,7 +71,13 @@ def fit_rounds(
     comm: CommContext | None = None,
     device=None,
     round_hook=None,
+    checkpoint: str | None = None,
+    checkpoint_every: int = 0,
 ) -> dict:
+    """Engine mode. With ``checkpoint`` set, training resumes from the file
+    if it exists and (with ``checkpoint_every`` > 0) saves every N rounds —
+    the reference's long-running-task recovery story on
+    the engine path."""
     comm = comm or init_from_env()
     device = device or comm.device
     net, loss_fn = _build(model, model_kwargs, device)
@@ -92,12 +98,24 @@ def fit_rounds(
         from vantage6_amd.models.factory import model_loss_per_sample
 
         trainer.per_sample_loss_fn = model_loss_per_sample(model)
+    start_round = 0
+    if checkpoint is not None:
+        import os as _os
+
+        from vantage6_amd.parallel.checkpoint import (load_checkpoint,
+                                                      save_checkpoint)
+
+        if _os.path.exists(checkpoint):
+            start_round = load_checkpoint(trainer, checkpoint)
     history = []
-    for _ in range(rounds):
+    for r in range(start_round, rounds):
         stats = trainer.run_round()
         history.append(stats)
         if round_hook:
             round_hook(trainer, stats)
+        if checkpoint is not None and checkpoint_every and \
+                (r + 1) % checkpoint_every == 0:
+            save_checkpoint(trainer, checkpoint)
     out = {
         "rounds": rounds,
         "final_loss": history[-1]["loss"],
