@@ -203,6 +203,10 @@ class Station:
         if run is None:
             return
         run_id = run["id"]
+        # kill semantics: a queued run is dropped here; an ALREADY-RUNNING
+        # algorithm function cannot be pre-empted in-process (the reference
+        # kills the Docker container) — the server marks the run killed
+        # (terminal) and any late patch from us cannot resurrect it
         with self._lock:
             if run_id in self._kill_flags:
                 self._kill_flags.discard(run_id)
@@ -221,7 +225,8 @@ class Station:
                 log=f"algorithm {task['image']!r} not in station allow-list",
             )
             return
-        if self.config.policies.get("require_approved_algorithms") and                 not self.core.algorithm_is_approved(task["image"]):
+        if self.config.policies.get("require_approved_algorithms") and \
+                not self.core.algorithm_is_approved(task["image"]):
             self.core.patch_run(
                 self.identity, run_id, status=RunStatus.NOT_ALLOWED.value,
                 log=f"algorithm {task['image']!r} is not store-approved",
