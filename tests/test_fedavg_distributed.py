@@ -140,3 +140,21 @@ def test_elastic_round_survives_station_failure(tmp_path):
         statuses = {r["organization_id"]: r["status"] for r in runs}
         assert statuses[fail_org] == "crashed"
         assert sum(1 for s in statuses.values() if s == "completed") == 2
+
+
+@pytest.mark.timeout(240)
+def test_fedavg_gloo_four_ranks_secure_equals_plain():
+    """World size 4: the pairwise Philox masks (6 pairs) must still cancel
+    exactly and match the plain weighted mean — guards the pair indexing
+    beyond the trivial 2-rank case (the driver scales to 8)."""
+    port = 29621
+    results = mp.Manager().dict()
+    mp.spawn(_worker, args=(4, port, results), nprocs=4, join=True)
+    assert len(results) == 4
+    cks = {r["plain_ck"] for r in results.values()}
+    assert len(cks) == 1, "plain master diverged across ranks"
+    sks = {r["sec_ck"] for r in results.values()}
+    assert len(sks) == 1, "secure master diverged across ranks"
+    for r in results.values():
+        assert r["sec_ck"] == pytest.approx(r["plain_ck"], rel=1e-5)
+        assert r["agg_ar"] == pytest.approx(r["agg_rb"], rel=1e-6)
