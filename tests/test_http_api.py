@@ -317,3 +317,54 @@ def test_remote_station_over_http(tmp_path):
         remote.close()
         srv.should_exit = True
         th.join(timeout=5)
+
+
+def test_encrypted_e2e_over_http(tmp_path):
+    """Encrypted collaboration with the researcher on the HTTP client:
+    key upload via PATCH /organization, per-org hybrid-RSA task input,
+    station decrypt, result encrypted back to the initiating org."""
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.client import UserClient
+    from vantage6_amd.common.config import DatabaseConfig
+
+    df = pd.DataFrame({"v": [2.0, 4.0, 6.0]})
+    csv = tmp_path / "enc.csv"
+    df.to_csv(csv, index=False)
+    net = DevNetwork(
+        n_stations=1, encrypted=True,
+        databases=[[DatabaseConfig(label="default", uri=str(csv), type="csv")]],
+        name="enc-http")
+    net.start()
+    app = create_app(net.core)
+    port = PORT + 11
+    srv = uvicorn.Server(uvicorn.Config(app, host="127.0.0.1", port=port,
+                                        log_level="error"))
+    th = threading.Thread(target=srv.run, daemon=True)
+    th.start()
+    url = f"http://127.0.0.1:{port}"
+    for _ in range(100):
+        try:
+            requests.get(f"{url}/api/version", timeout=1)
+            break
+        except Exception:
+            time.sleep(0.05)
+    try:
+        web = UserClient(url=url)
+        web.authenticate("enc-http-researcher", "password")
+        web.setup_encryption(str(net._tmp / "station0.key"), key_bits=1024)
+        task = web.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],
+            name="enc-mean", image="algorithms.column_mean",
+            input_={"method": "partial_mean", "kwargs": {"columns": ["v"]}})
+        (res,) = web.wait_for_results(task["id"], timeout=30)
+        assert res["sums"]["v"] == pytest.approx(12.0)
+        # the stored result on the server is NOT plaintext-b64 (it is the
+        # hybrid envelope: key$nonce$mac$ciphertext)
+        raw = net.root_client.run.from_task(task["id"])[0]
+        run_full = net.root_client.run.get(raw["id"])
+        assert "$" in (run_full["result"] or "")
+    finally:
+        net.stop()
+        srv.should_exit = True
+        th.join(timeout=5)
