@@ -75,9 +75,25 @@ class HttpGateway:
                 msg = r.text
             raise RuntimeError(f"server returned {r.status_code}: {msg}")
 
+    max_retries = 3  # transient connection errors (reference ClientBase)
+
+    def _send(self, verb, url, **kw):
+        import time as _time
+
+        import requests
+
+        last = None
+        for attempt in range(self.max_retries):
+            try:
+                return self.http.request(verb, url, **kw)
+            except requests.exceptions.ConnectionError as exc:
+                last = exc
+                _time.sleep(0.2 * (attempt + 1))
+        raise last
+
     def request(self, verb: str, path: str, json: dict | None = None, params=None):
         headers = {"Authorization": f"Bearer {self.token}"} if self.token else {}
-        r = self.http.request(
+        r = self._send(
             verb, f"{self.base_url}/{path.lstrip('/')}", json=json, params=params,
             headers=headers,
         )
@@ -90,7 +106,7 @@ class HttpGateway:
                 self.token = rr.json()["access_token"]
                 self.refresh_token_value = rr.json().get("refresh_token")
                 headers = {"Authorization": f"Bearer {self.token}"}
-                r = self.http.request(
+                r = self._send(
                     verb, f"{self.base_url}/{path.lstrip('/')}", json=json, params=params,
                     headers=headers,
                 )
