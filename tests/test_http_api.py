@@ -368,3 +368,46 @@ def test_encrypted_e2e_over_http(tmp_path):
         net.stop()
         srv.should_exit = True
         th.join(timeout=5)
+
+
+def test_remote_core_reauth_after_token_loss(tmp_path):
+    """RemoteCore must recover from a dead/expired token by re-authenticating
+    with its api-key (reference: node re-auth on expired_token)."""
+    from vantage6_amd.client import UserClient
+    from vantage6_amd.node.remote import RemoteCore
+    from vantage6_amd.server.core import ServerCore
+
+    core = ServerCore()
+    core.create_root_user("root", "root")
+    root = UserClient(core=core)
+    root.authenticate("root", "root")
+    org = root.organization.create(name="reauth-org")
+    collab = root.collaboration.create(name="reauth-c",
+                                       organization_ids=[org["id"]])
+    node = root.node.create(name="reauth-n", collaboration_id=collab["id"],
+                            organization_id=org["id"])
+    app = create_app(core)
+    port = PORT + 13
+    srv = uvicorn.Server(uvicorn.Config(app, host="127.0.0.1", port=port,
+                                        log_level="error"))
+    th = threading.Thread(target=srv.run, daemon=True)
+    th.start()
+    url = f"http://127.0.0.1:{port}"
+    for _ in range(100):
+        try:
+            requests.get(f"{url}/api/version", timeout=1)
+            break
+        except Exception:
+            time.sleep(0.05)
+    try:
+        remote = RemoteCore(url)
+        remote.authenticate_node(node["api_key"])
+        assert remote.node_sync_open_runs(None) == []
+        # simulate token loss: both tokens invalidated
+        remote.gw.token = "garbage"
+        remote.gw.refresh_token_value = "garbage"
+        assert remote.node_sync_open_runs(None) == []  # re-auths transparently
+        assert remote.gw.token != "garbage"
+    finally:
+        srv.should_exit = True
+        th.join(timeout=5)
