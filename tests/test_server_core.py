@@ -617,3 +617,22 @@ def test_study_list_get_scoped(core, root):
         root.task.create(collaboration=collab["id"], organizations=[orgs[2]],
                          name="t", image="x", input_={"method": "m"},
                          study=st["id"])
+
+
+def test_task_listing_scoped_and_filtered(core, root):
+    orgs1, c1 = make_collab(root, name="tl1")
+    orgs2, c2 = make_collab(root, name="tl2")
+    root.node.create(name="tl-n", collaboration_id=c1["id"],
+                     organization_id=orgs1[0])
+    t = root.task.create(collaboration=c1["id"], organizations=[orgs1[0]],
+                         name="secret-listing", image="x",
+                         input_={"method": "m"})
+    root.user.create("tl-out", "pw", organization_id=orgs2[0],
+                     roles=["Researcher"])
+    out = UserClient(core=core)
+    out.authenticate("tl-out", "pw")
+    visible = out.task.list(per_page=500)["data"]
+    assert all(x["id"] != t["id"] for x in visible)
+    hits = root.task.list(name="secret", per_page=50)["data"]
+    assert any(x["id"] == t["id"] for x in hits)
+    assert root.task.list(status="pending", per_page=50)["data"]

@@ -869,7 +869,30 @@ class ServerCore:
             q = q.filter(Task.job_id == filters["job_id"])
         if filters.get("collaboration_id") is not None:
             q = q.filter(Task.collaboration_id == filters["collaboration_id"])
-        return self._paginate([self._task_json(t) for t in q.all()], page, per_page)
+        if filters.get("init_user_id") is not None:
+            q = q.filter(Task.init_user_id == filters["init_user_id"])
+        if filters.get("name"):
+            q = q.filter(Task.name.contains(filters["name"]))
+        tasks = q.all()
+        # scope filter (reference: task VIEW scopes) — mirror of run scoping
+        if ident.type == "user":
+            scope = self._rc("task", ident).allowed_scope(Operation.VIEW)
+            if scope == Scope.OWN:
+                tasks = [t for t in tasks if t.init_user_id == ident.id]
+            elif scope == Scope.ORGANIZATION:
+                tasks = [t for t in tasks
+                         if t.init_org_id == ident.organization_id]
+            elif scope == Scope.COLLABORATION:
+                me = self.session.get(Organization, ident.organization_id)
+                mine = {c.id for c in me.collaborations} if me else set()
+                tasks = [t for t in tasks if t.collaboration_id in mine]
+        elif ident.type in ("node", "container"):
+            tasks = [t for t in tasks
+                     if t.collaboration_id == ident.collaboration_id]
+        items = [self._task_json(t) for t in tasks]
+        if filters.get("status"):
+            items = [t for t in items if t["status"] == filters["status"]]
+        return self._paginate(items, page, per_page)
 
     @_locked
     def kill_task(self, ident: Identity, task_id: int) -> dict:
