@@ -636,3 +636,23 @@ def test_task_listing_scoped_and_filtered(core, root):
     hits = root.task.list(name="secret", per_page=50)["data"]
     assert any(x["id"] == t["id"] for x in hits)
     assert root.task.list(status="pending", per_page=50)["data"]
+
+
+def test_org_and_node_listings_scoped(core, root):
+    orgs1, c1 = make_collab(root, name="ls1")
+    orgs2, c2 = make_collab(root, name="ls2")
+    root.node.create(name="ls-n1", collaboration_id=c1["id"],
+                     organization_id=orgs1[0])
+    root.node.create(name="ls-n2", collaboration_id=c2["id"],
+                     organization_id=orgs2[0])
+    root.user.create("ls-r", "pw", organization_id=orgs1[0],
+                     roles=["Researcher"])
+    r = UserClient(core=core)
+    r.authenticate("ls-r", "pw")
+    org_ids = {o["id"] for o in r.organization.list(per_page=500)["data"]}
+    assert set(orgs1) <= org_ids and not (set(orgs2) & org_ids)
+    node_names = {n["name"] for n in r.node.list(per_page=500)["data"]}
+    assert "ls-n1" in node_names and "ls-n2" not in node_names
+    # root (GLOBAL) still sees everything
+    allnames = {n["name"] for n in root.node.list(per_page=500)["data"]}
+    assert {"ls-n1", "ls-n2"} <= allnames

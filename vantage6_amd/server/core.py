@@ -668,6 +668,16 @@ class ServerCore:
     def list_organizations(self, ident: Identity, page=1, per_page=20) -> dict:
         self._check_user_can_view_any("organization", ident)
         orgs = self.session.query(Organization).order_by(Organization.id).all()
+        if ident.type == "user":
+            if self._rc("organization", ident).allowed_scope(
+                    Operation.VIEW) != Scope.GLOBAL:
+                mine = self._org_ids_in_my_collabs(ident)
+                orgs = [o for o in orgs if o.id in mine]
+        else:
+            # nodes/containers see their collaboration's member orgs
+            collab = self.session.get(Collaboration, ident.collaboration_id)
+            member = {o.id for o in collab.organizations} if collab else set()
+            orgs = [o for o in orgs if o.id in member]
         return self._paginate([self._org_json(o) for o in orgs], page, per_page)
 
     @_locked
@@ -690,7 +700,16 @@ class ServerCore:
         q = self.session.query(Node).order_by(Node.id)
         if collaboration_id is not None:
             q = q.filter(Node.collaboration_id == collaboration_id)
-        return self._paginate([self._node_json(n) for n in q.all()], page, per_page)
+        nodes = q.all()
+        if ident.type == "user":
+            if self._rc("node", ident).allowed_scope(
+                    Operation.VIEW) != Scope.GLOBAL:
+                mine = self._org_ids_in_my_collabs(ident)
+                nodes = [n for n in nodes if n.organization_id in mine]
+        else:
+            nodes = [n for n in nodes
+                     if n.collaboration_id == ident.collaboration_id]
+        return self._paginate([self._node_json(n) for n in nodes], page, per_page)
 
     def _check_user_can_view_any(self, resource: str, ident: Identity) -> None:
         if ident.type != "user":
