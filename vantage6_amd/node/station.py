@@ -99,6 +99,17 @@ class Station:
             daemon=True,
         )
         self._ping_thread.start()
+        # immediate first node-info-update (columns etc.) — the heartbeat
+        # loop's first tick is heartbeat_interval away
+        try:
+            self.core.node_ping(self.identity, {
+                "device": self.config.device,
+                "queue_depth": self.queue.qsize(),
+                "inflight": 0,
+                "columns": self._database_columns(),
+            })
+        except Exception:  # noqa: BLE001
+            log.exception("initial node-info-update failed")
 
     heartbeat_interval: float = 15.0
 
