@@ -477,8 +477,7 @@ class TestRound1Additions:
         net = DevNetwork(n_stations=1, databases=[[
             DatabaseConfig(label="default", uri=str(csv), type="csv")]])
         with net:
-            st = net.stations[0]
-            st.core.node_ping(st.identity, {"columns": st._database_columns()})
+            # columns arrive with the station's IMMEDIATE first heartbeat
             node_id = net.root_client.node.list()["data"][0]["id"]
             cols = net.root_client.node.columns(node_id)
             assert cols == ["age", "bmi"]
