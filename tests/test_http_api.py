@@ -305,13 +305,14 @@ def test_remote_station_over_http(tmp_path):
         station.start()
         web = UserClient(url=url)
         web.authenticate("root", "root")
+        # CENTRAL task: exercises the proxy's subtask fan-out over HTTP
+        # with the CONTAINER token (a node bearer would be rejected)
         task = web.task.create(
             collaboration=collab["id"], organizations=[org["id"]],
             name="remote-mean", image="algorithms.column_mean",
-            input_={"method": "partial_mean", "kwargs": {"columns": ["v"]}})
+            input_={"method": "central_mean", "kwargs": {"columns": ["v"]}})
         (res,) = web.wait_for_results(task["id"], timeout=30)
-        assert res["sums"]["v"] == pytest.approx(10.0)
-        assert res["counts"]["v"] == 4
+        assert res["v"] == pytest.approx(2.5)
     finally:
         station.stop()
         remote.close()

@@ -185,11 +185,24 @@ class RemoteCore:
         self.org_id = _decode_payload(out["access_token"]).get("org")
         return out
 
-    def _request(self, verb: str, path: str, **kw):
+    def _request(self, verb: str, path: str, ident=None, **kw):
         """Gateway call that survives full token expiry: on a 401 (even
         after the gateway's own refresh attempt), re-authenticate with the
         stored api-key and retry once (reference: node re-auths on the
-        ``expired_token`` socket event)."""
+        ``expired_token`` socket event).
+
+        When ``ident`` carries a container token (proxy calls on behalf of
+        a running algorithm), the request is sent with THAT bearer — the
+        server enforces container semantics (same-collaboration subtasks,
+        image match) on it."""
+        token = getattr(ident, "token", None)
+        if token is not None and getattr(ident, "type", None) == "container":
+            r = self.gw._send(
+                verb, f"{self.gw.base_url}/{path.lstrip('/')}",
+                json=kw.get("json"), params=kw.get("params"),
+                headers={"Authorization": f"Bearer {token}"})
+            self.gw._raise_for(r)
+            return r.json() if r.text else None
         try:
             return self.gw.request(verb, path, **kw)
         except RuntimeError as exc:
@@ -200,11 +213,15 @@ class RemoteCore:
 
     def identity_from_token(self, token: str) -> Identity:
         p = _decode_payload(token)
-        return Identity(
+        ident = Identity(
             type=p["type"], id=p["sub"], organization_id=p.get("org"),
             collaboration_id=p.get("collab"), task_id=p.get("task_id"),
             image=p.get("image"),
         )
+        # keep the raw token so calls made AS this identity (the NodeProxy
+        # acting for a running algorithm container) carry the right bearer
+        ident.token = token
+        return ident
 
     def token_for_container(self, ident, task_id: int, image: str) -> dict:
         return self._request("POST", "token/container",
@@ -230,23 +247,24 @@ class RemoteCore:
 
     # ------------------------------------------------------- task/run plane
     def get_task(self, ident, task_id: int) -> dict:
-        return self._request("GET", f"task/{task_id}")
+        return self._request("GET", f"task/{task_id}", ident=ident)
 
     def list_runs(self, ident, **params) -> dict:
         params = {k: v for k, v in params.items() if v is not None}
-        return self._request("GET", "run", params=params)
+        return self._request("GET", "run", ident=ident, params=params)
 
     def get_run(self, ident, run_id: int) -> dict:
-        return self._request("GET", f"run/{run_id}")
+        return self._request("GET", f"run/{run_id}", ident=ident)
 
     def patch_run(self, ident, run_id: int, **fields) -> dict:
         return self._request("PATCH", f"run/{run_id}", json=fields)
 
     def list_results(self, ident, task_id: int, **kw) -> dict:
-        return self._request("GET", "result", params={"task_id": task_id})
+        return self._request("GET", "result", ident=ident,
+                             params={"task_id": task_id})
 
     def create_task(self, ident, **body) -> dict:
-        return self._request("POST", "task", json=body)
+        return self._request("POST", "task", ident=ident, json=body)
 
     # --------------------------------------------------------------- lookups
     def organization_public_key(self, org_id: int) -> str:
@@ -254,7 +272,7 @@ class RemoteCore:
             "public_key") or ""
 
     def get_collaboration(self, ident, collab_id: int) -> dict:
-        return self._request("GET", f"collaboration/{collab_id}")
+        return self._request("GET", f"collaboration/{collab_id}", ident=ident)
 
     def algorithm_is_approved(self, image: str) -> bool:
         rows = self._request("GET", "algorithm",
