@@ -409,6 +409,8 @@ def test_remote_core_reauth_after_token_loss(tmp_path):
         remote.gw.refresh_token_value = "garbage"
         assert remote.node_sync_open_runs(None) == []  # re-auths transparently
         assert remote.gw.token != "garbage"
+        # store policy check with a node bearer over HTTP
+        assert remote.algorithm_is_approved("algorithms.column_mean") is False
     finally:
         srv.should_exit = True
         th.join(timeout=5)
