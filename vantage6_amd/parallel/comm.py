@@ -63,6 +63,10 @@ def init_from_env(force_backend: str | None = None, timeout_s: float = 300.0) ->
         device = torch.device("cpu")
     backend = force_backend or ("nccl" if cuda else "gloo")  # nccl IS RCCL on ROCm
     if world > 1 and not dist.is_initialized():
+        # host driver supports only dmabuf IPC — without this RCCL's
+        # cross-process CUDA-tensor sharing fails with
+        # hipIpcGetMemHandle: invalid argument
+        os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29512")
         dist.init_process_group(
