@@ -7,7 +7,9 @@ add, for every pair (i, j) with i < j, a full-range 64-bit mask with sign
 +1 on rank i and −1 on rank j. All-reduce(SUM) wraps mod 2^64 ⇒ masks
 cancel exactly; the server and every other station only ever see masked
 integers. GPU path: one fused Philox kernel (csrc/secagg.hip); CPU path
-below is the reference.
+below is the reference. The CPU and GPU paths use different mask PRNGs,
+so a secure-aggregation job must be device-homogeneous across ranks (it
+always is: one torchrun launch) — masks only cancel against themselves.
 """
 from __future__ import annotations
 
