@@ -166,3 +166,30 @@ def test_embedding_cpu_backward_matches_torch():
     ref(ids).backward(dy)
     assert torch.allclose(emb.weight.grad, ref.weight.grad, atol=1e-5)
     assert isinstance(emb, torch.nn.Embedding)  # DP hooks use isinstance
+
+
+def test_resolve_module_name_forms():
+    """Algorithm "image" references: import path, local://, docker-style."""
+    import pytest as _pytest
+
+    from vantage6_amd.algorithm.exceptions import InputError
+    from vantage6_amd.algorithm.wrap import resolve_module
+
+    m1 = resolve_module("algorithms.column_mean")
+    m2 = resolve_module("local://algorithms.column_mean")
+    m3 = resolve_module("harbor2.vantage6.ai/algorithms.column_mean:latest")
+    assert m1 is m2 is m3
+    with _pytest.raises(InputError, match="cannot import"):
+        resolve_module("registry.example/nope.not_a_module:v1")
+
+
+def test_dispatch_rejects_unknown_method():
+    from vantage6_amd.algorithm.exceptions import InputError
+    from vantage6_amd.algorithm.wrap import dispatch, resolve_module
+    import pytest as _pytest
+
+    mod = resolve_module("algorithms.column_mean")
+    with _pytest.raises(InputError, match="not found"):
+        dispatch(mod, {"method": "does_not_exist"}, {})
+    with _pytest.raises(InputError, match="no 'method'"):
+        dispatch(mod, {}, {})
