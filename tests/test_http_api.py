@@ -414,3 +414,19 @@ def test_remote_core_reauth_after_token_loss(tmp_path):
     finally:
         srv.should_exit = True
         th.join(timeout=5)
+
+
+def test_pagination_link_headers(server, net):
+    tok = requests.post(f"{server}/api/token/user",
+                        json={"username": "root", "password": "root"}).json()
+    h = {"Authorization": f"Bearer {tok['access_token']}"}
+    for _ in range(3):
+        net.root_client.task.create(
+            collaboration=net.collaboration["id"], organizations=[net.org_ids[0]],
+            name="pg", image="algorithms.column_mean",
+            input_={"method": "partial_mean", "kwargs": {"columns": ["v"]}})
+    r = requests.get(f"{server}/api/task", params={"per_page": 2, "page": 1},
+                     headers=h)
+    assert r.status_code == 200
+    assert 'rel="next"' in r.headers.get("Link", "")
+    assert int(r.headers["Total-Count"]) >= 3

@@ -12,7 +12,7 @@ from __future__ import annotations
 
 from typing import Optional
 
-from fastapi import Depends, FastAPI, HTTPException, Request, WebSocket
+from fastapi import Depends, FastAPI, HTTPException, Request, Response, WebSocket
 from pydantic import BaseModel, Field
 
 from ..common import jwt as jwtlib
@@ -181,13 +181,40 @@ def create_app(core: ServerCore) -> FastAPI:
     def task_get(task_id: int, who: Identity = Depends(ident)):
         return wrap(core.get_task, who, task_id)
 
+    def paginate_headers(out, request: Request, response: Response):
+        """RFC-5988 ``Link`` pagination headers (reference A.1) alongside
+        the body's ``links`` block."""
+        links = out.get("links") if isinstance(out, dict) else None
+        if links:
+            from urllib.parse import urlencode
+
+            def u(p):
+                q = dict(request.query_params)
+                q["page"] = str(p)
+                q["per_page"] = str(links["per_page"])
+                return f"<{request.url.path}?{urlencode(q)}>"
+
+            parts = [f'{u(1)}; rel="first"', f'{u(links["pages"])}; rel="last"']
+            if links["page"] < links["pages"]:
+                parts.insert(0, f'{u(links["page"] + 1)}; rel="next"')
+            if links["page"] > 1:
+                parts.insert(0, f'{u(links["page"] - 1)}; rel="prev"')
+            response.headers["Link"] = ", ".join(parts)
+            response.headers["Total-Count"] = str(links["total"])
+        return out
+
     @app.get("/api/task")
-    def task_list(who: Identity = Depends(ident), page: int = 1, per_page: int = 20,
-                  parent_id: Optional[int] = None, job_id: Optional[int] = None,
-                  collaboration_id: Optional[int] = None):
-        return wrap(core.list_tasks, who, page=page, per_page=per_page,
-                    parent_id=parent_id, job_id=job_id,
-                    collaboration_id=collaboration_id)
+    def task_list(request: Request, response: Response,
+                  who: Identity = Depends(ident), page: int = 1,
+                  per_page: int = 20, parent_id: Optional[int] = None,
+                  job_id: Optional[int] = None,
+                  collaboration_id: Optional[int] = None,
+                  name: Optional[str] = None, status: Optional[str] = None):
+        out = wrap(core.list_tasks, who, page=page, per_page=per_page,
+                   parent_id=parent_id, job_id=job_id,
+                   collaboration_id=collaboration_id, name=name,
+                   status=status)
+        return paginate_headers(out, request, response)
 
     @app.delete("/api/task/{task_id}")
     def task_kill(task_id: int, who: Identity = Depends(ident)):
@@ -195,11 +222,13 @@ def create_app(core: ServerCore) -> FastAPI:
 
     # --------------------------------------------------------------- runs
     @app.get("/api/run")
-    def run_list(who: Identity = Depends(ident), task_id: Optional[int] = None,
+    def run_list(request: Request, response: Response,
+                 who: Identity = Depends(ident), task_id: Optional[int] = None,
                  node_id: Optional[int] = None, state: Optional[str] = None,
                  page: int = 1, per_page: int = 50):
-        return wrap(core.list_runs, who, task_id=task_id, node_id=node_id,
-                    state=state, page=page, per_page=per_page)
+        out = wrap(core.list_runs, who, task_id=task_id, node_id=node_id,
+                   state=state, page=page, per_page=per_page)
+        return paginate_headers(out, request, response)
 
     @app.get("/api/run/{run_id}")
     def run_get(run_id: int, who: Identity = Depends(ident)):
