@@ -242,3 +242,45 @@ def test_kaplan_meier(tmp_path):
     assert res["n"] == len(pooled)
     for row in res["curve"]:
         assert row["survival"] == pytest.approx(surv_ref[row["time"]], rel=1e-12)
+
+
+def test_federated_logistic_regression_matches_sklearn(tmp_path):
+    """Federated IRLS equals the pooled sklearn (unpenalized) fit."""
+    import numpy as np
+    import pandas as pd
+    from sklearn.linear_model import LogisticRegression
+
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+
+    rng = np.random.default_rng(31)
+    frames, paths = [], []
+    true_beta = np.array([0.8, -1.2])
+    for i in range(2):
+        X = rng.normal(0, 1, (120, 2))
+        logits = 0.3 + X @ true_beta
+        y = (rng.random(120) < 1 / (1 + np.exp(-logits))).astype(int)
+        df = pd.DataFrame({"a": X[:, 0], "b": X[:, 1], "y": y})
+        p = tmp_path / f"glm{i}.csv"
+        df.to_csv(p, index=False)
+        frames.append(df)
+        paths.append(p)
+    pooled = pd.concat(frames, ignore_index=True)
+    sk = LogisticRegression(penalty=None, tol=1e-10, max_iter=500)
+    sk.fit(pooled[["a", "b"]], pooled["y"])
+
+    net = DevNetwork(n_stations=2, databases=[
+        [DatabaseConfig(label="default", uri=str(p), type="csv")] for p in paths])
+    with net:
+        task = net.client.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],
+            name="glm", image="algorithms.glm_logistic",
+            input_={"method": "central_logistic_regression",
+                    "kwargs": {"feature_cols": ["a", "b"],
+                               "target_col": "y"}})
+        (res,) = net.client.wait_for_results(task["id"], timeout=60)
+    assert res["converged"] and res["n"] == 240
+    assert res["intercept"] == pytest.approx(sk.intercept_[0], abs=1e-5)
+    assert res["coefficients"]["a"] == pytest.approx(sk.coef_[0][0], abs=1e-5)
+    assert res["coefficients"]["b"] == pytest.approx(sk.coef_[0][1], abs=1e-5)
