@@ -357,3 +357,33 @@ def test_federated_coxph_matches_pooled_newton(tmp_path):
     assert res["coefficients"]["b"] == pytest.approx(ref_beta[1], abs=1e-6)
     # directional recovery of the simulated effects
     assert res["coefficients"]["a"] > 0.3 and res["coefficients"]["b"] < -0.2
+
+
+def test_federated_crosstab(tmp_path):
+    import pandas as pd
+
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+
+    frames, paths = [], []
+    for i, rows in enumerate([[("m", "yes")] * 6 + [("f", "no")] * 3,
+                              [("m", "yes")] * 2 + [("f", "yes")] * 1]):
+        df = pd.DataFrame(rows, columns=["sex", "treated"])
+        p = tmp_path / f"ct{i}.csv"
+        df.to_csv(p, index=False)
+        frames.append(df)
+        paths.append(p)
+    net = DevNetwork(n_stations=2, databases=[
+        [DatabaseConfig(label="default", uri=str(p), type="csv")] for p in paths])
+    with net:
+        task = net.client.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],
+            name="ct", image="algorithms.crosstab",
+            input_={"method": "central_crosstab",
+                    "kwargs": {"row_col": "sex", "col_col": "treated",
+                               "suppress_below": 2}})
+        (res,) = net.client.wait_for_results(task["id"], timeout=30)
+    assert res["counts"]["m|yes"] == 8
+    assert res["counts"]["f|no"] == 3
+    assert res["counts"]["f|yes"] == "<2"  # suppressed cell
