@@ -386,4 +386,6 @@ def test_federated_crosstab(tmp_path):
         (res,) = net.client.wait_for_results(task["id"], timeout=30)
     assert res["counts"]["m|yes"] == 8
     assert res["counts"]["f|no"] == 3
-    assert res["counts"]["f|yes"] == "<2"  # suppressed cell
+    # the 1-count f|yes cell never left station 1
+    assert "f|yes" not in res["counts"]
+    assert res["suppressed_cells"] == 1
