@@ -389,3 +389,48 @@ def test_federated_crosstab(tmp_path):
     # the 1-count f|yes cell never left station 1
     assert "f|yes" not in res["counts"]
     assert res["suppressed_cells"] == 1
+
+
+def test_multi_database_task(tmp_path):
+    """A task naming two database labels gets BOTH shards injected, in the
+    requested order (@data(2) + USER_REQUESTED_DATABASE_LABELS contract)."""
+    import pandas as pd
+
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+
+    a = tmp_path / "a.csv"
+    b = tmp_path / "b.csv"
+    pd.DataFrame({"v": [1.0, 2.0]}).to_csv(a, index=False)
+    pd.DataFrame({"v": [10.0, 20.0, 30.0]}).to_csv(b, index=False)
+    net = DevNetwork(n_stations=1, databases=[[
+        DatabaseConfig(label="alpha", uri=str(a), type="csv"),
+        DatabaseConfig(label="beta", uri=str(b), type="csv"),
+    ]])
+    # ad-hoc two-db algorithm: register a module object on the fly
+    import sys
+    import types
+
+    from vantage6_amd.algorithm import data
+
+    mod = types.ModuleType("algos_twodb")
+
+    @data(2)
+    def row_counts(df1, df2):
+        return {"n1": int(len(df1)), "n2": int(len(df2)),
+                "sum2": float(df2["v"].sum())}
+
+    mod.row_counts = row_counts
+    sys.modules["algos_twodb"] = mod
+    try:
+        with net:
+            task = net.client.task.create(
+                collaboration=net.collaboration["id"],
+                organizations=[net.org_ids[0]],
+                name="twodb", image="algos_twodb",
+                input_={"method": "row_counts"},
+                databases=[{"label": "alpha"}, {"label": "beta"}])
+            (res,) = net.client.wait_for_results(task["id"], timeout=30)
+        assert res == {"n1": 2, "n2": 3, "sum2": 60.0}
+    finally:
+        del sys.modules["algos_twodb"]
