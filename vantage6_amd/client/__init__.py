@@ -142,6 +142,14 @@ class UserClient:
     def authenticate(self, username: str, password: str) -> dict:
         out = self.gw.authenticate(username, password)
         self.whoami = {"username": username}
+        try:
+            me = self._my_org_id()
+            self.whoami["organization_id"] = me
+            if isinstance(self.gw, CoreGateway):
+                ident = self.gw.ident
+                self.whoami["user_id"] = ident.id
+        except Exception:  # noqa: BLE001 — whoami is best-effort metadata
+            pass
         return out
 
     def setup_encryption(self, private_key_file: str | None, key_bits: int = 1024) -> None:
