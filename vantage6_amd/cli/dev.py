@@ -29,7 +29,7 @@ class DevNetwork:
         devices: list[str] | None = None,
         uri: str = "sqlite://",
         name: str = "demo",
-        key_bits: int = 1024,
+        key_bits: int = 2048,
     ):
         self.core = ServerCore(uri=uri)
         self.encrypted = encrypted

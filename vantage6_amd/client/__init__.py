@@ -152,7 +152,7 @@ class UserClient:
             pass
         return out
 
-    def setup_encryption(self, private_key_file: str | None, key_bits: int = 1024) -> None:
+    def setup_encryption(self, private_key_file: str | None, key_bits: int = 2048) -> None:
         """Load/create the researcher org's private key & upload public key."""
         if private_key_file is None:
             self.cryptor = CryptorBase()

@@ -16,7 +16,16 @@ here directly —
   encrypt-then-MAC tag instead of AES-CTR (same interface and security
   role; AES is unavailable without OpenSSL bindings).
 
-Wire format: ``b64(rsa(shared_key)) $ b64(nonce) $ b64(ct) $ b64(tag)``.
+Wire format (reference-shaped, 3 ``$``-joined fields like upstream's
+``b64(enc_key)$b64(iv)$b64(ciphertext)``):
+
+    ``b64(rsa(shared_key)) $ b64(nonce) $ b64(ct || tag)``
+
+with the 32-byte encrypt-then-MAC tag folded into the tail of the third
+field (upstream AES-CTR has no tag; the extra 32 bytes are this
+implementation's authenticity deviation, visible only at this boundary).
+Decryption also accepts the legacy round-1 4-field envelope
+``... $ b64(ct) $ b64(tag)``.
 """
 from __future__ import annotations
 
@@ -248,12 +257,18 @@ class RSACryptor(CryptorBase):
         shared = secrets.token_bytes(32)
         enc_key = pub.encrypt(shared)
         nonce, ct, tag = _sym_encrypt(shared, data)
-        return SEP.join(bytes_to_base64s(x) for x in (enc_key, nonce, ct, tag))
+        return SEP.join(bytes_to_base64s(x) for x in (enc_key, nonce, ct + tag))
 
     def decrypt_str_to_bytes(self, data: str) -> bytes:
-        try:
-            enc_key, nonce, ct, tag = (base64s_to_bytes(x) for x in data.split(SEP))
-        except ValueError as exc:
-            raise ValueError("malformed encryption envelope") from exc
+        fields = [base64s_to_bytes(x) for x in data.split(SEP)]
+        if len(fields) == 3:  # reference-shaped: tag folded into field 3
+            enc_key, nonce, body = fields
+            if len(body) < _HLEN:
+                raise ValueError("malformed encryption envelope")
+            ct, tag = body[:-_HLEN], body[-_HLEN:]
+        elif len(fields) == 4:  # legacy round-1 envelope
+            enc_key, nonce, ct, tag = fields
+        else:
+            raise ValueError("malformed encryption envelope")
         shared = self.key.decrypt(enc_key)
         return _sym_decrypt(shared, nonce, ct, tag)

@@ -48,7 +48,7 @@ class Station:
         config: StationConfig,
         api_key: str,
         private_key_file: str | None = None,
-        key_bits: int = 1024,
+        key_bits: int = 2048,
     ):
         self.core = core
         self.config = config
@@ -330,10 +330,24 @@ class Station:
             )
 
     def _encrypt_result(self, blob: bytes, init_org_id: int | None) -> str:
-        if isinstance(self.cryptor, RSACryptor) and init_org_id is not None:
+        """Encrypt the result for the task's initiating organization.
+
+        In an encrypted collaboration a missing recipient key is an ERROR
+        (fail the run), never a silent downgrade to plaintext — the same
+        contract NodeProxy._encrypt_for_org enforces on the input path.
+        """
+        if isinstance(self.cryptor, RSACryptor):
+            if init_org_id is None:
+                raise RuntimeError(
+                    "encrypted collaboration but task has no initiating "
+                    "organization to encrypt the result for")
             pubkey = self.core.organization_public_key(init_org_id)
-            if pubkey:
-                return self.cryptor.encrypt_bytes_to_str(blob, pubkey)
+            if not pubkey:
+                raise RuntimeError(
+                    f"encrypted collaboration but organization {init_org_id} "
+                    "has no public key uploaded — refusing to send the "
+                    "result in plaintext")
+            return self.cryptor.encrypt_bytes_to_str(blob, pubkey)
         return bytes_to_base64s(blob)
 
     def _bind_device(self) -> None:

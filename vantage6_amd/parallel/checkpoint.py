@@ -25,6 +25,8 @@ def save_checkpoint(trainer, path: str | Path) -> None:
         if trainer.flat.exp_avg_sq is not None:
             tensors["exp_avg_sq"] = trainer.flat.exp_avg_sq.cpu().contiguous()
         save_file(tensors, str(path))
+        if trainer.flat._step_dev is not None:  # device counter is authoritative
+            trainer.flat._step = int(trainer.flat._step_dev.item())
         meta = {"round": trainer.round_idx, "numel": trainer.flat.numel,
                 "optimizer": trainer.cfg.optimizer, "step": trainer.flat._step}
         Path(str(path) + ".json").write_text(json.dumps(meta))
@@ -49,9 +51,17 @@ def load_checkpoint(trainer, path: str | Path) -> int:
         trainer.flat.exp_avg_sq.copy_(tensors["exp_avg_sq"].to(dev))
     trainer.flat._step = meta.get("step", 0)
     # all ranks read the same file in tests; in production rank 0's state is
-    # authoritative — broadcast to make divergence impossible
+    # authoritative — broadcast ALL optimizer state (incl. second moment and
+    # the bias-correction step counter) to make divergence impossible
     trainer.comm.broadcast(trainer.flat.master_flat, src=0)
     trainer.comm.broadcast(trainer.flat.momentum_flat, src=0)
+    if trainer.flat.exp_avg_sq is not None:
+        trainer.comm.broadcast(trainer.flat.exp_avg_sq, src=0)
+    step_t = torch.tensor([trainer.flat._step], dtype=torch.long, device=dev)
+    trainer.comm.broadcast(step_t, src=0)
+    trainer.flat._step = int(step_t.item())
+    if trainer.flat._step_dev is not None:
+        trainer.flat._step_dev.fill_(trainer.flat._step)
     trainer.flat.sync_param_from_master()
     trainer.round_idx = meta["round"]
     return trainer.round_idx

@@ -32,7 +32,12 @@ import torch.nn as nn
 class DPConfig:
     max_grad_norm: float = 1.0  # C
     noise_multiplier: float = 1.0  # sigma
-    seed: int = 0xD9
+    # None (the default, and the only safe production value) → each DPEngine
+    # draws its noise seed from the OS CSPRNG, so no other party — including
+    # sibling stations running the same config dict — can predict or cancel
+    # the noise. An explicit int is for reproducible unit tests ONLY, and even
+    # then the engine decorrelates ranks by mixing the distributed rank in.
+    seed: Optional[int] = None
     delta: float = 1e-5
 
 
@@ -43,6 +48,17 @@ class DPEngine:
     def __init__(self, model: nn.Module, cfg: DPConfig):
         self.model = model
         self.cfg = cfg
+        if cfg.seed is None:
+            import secrets
+
+            self._noise_seed = secrets.randbits(62)
+        else:
+            # test-only deterministic path; still never bit-identical across
+            # ranks (identical noise would cancel in the FedAvg difference)
+            rank = 0
+            if torch.distributed.is_available() and torch.distributed.is_initialized():
+                rank = torch.distributed.get_rank()
+            self._noise_seed = (cfg.seed + rank * 0x9E3779B97F4A7C15) & ((1 << 62) - 1)
         self._norms: Optional[torch.Tensor] = None  # [B] accumulating norm^2
         self._capturing = False
         self._step = 0
@@ -164,9 +180,9 @@ class DPEngine:
             from ..ops import hip_ops
 
             hip_ops().dp_scale_noise(grad_flat, 1.0 / batch_size, std,
-                                     cfg.seed + self._step)
+                                     self._noise_seed + self._step)
         else:
-            g = torch.Generator().manual_seed(cfg.seed + self._step)
+            g = torch.Generator().manual_seed(self._noise_seed + self._step)
             noise = torch.randn(grad_flat.shape, generator=g) * std
             grad_flat.mul_(1.0 / batch_size).add_(noise.to(grad_flat.dtype))
 

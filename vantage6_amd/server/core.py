@@ -311,6 +311,11 @@ class ServerCore:
 
     def identity_from_token(self, token: str) -> Identity:
         p = jwtlib.decode(token, self.jwt_secret)
+        # only ACCESS tokens are bearer credentials: the longer-lived refresh
+        # token is exchangeable via /token/refresh only, and purpose-tokens
+        # (e.g. mfa_reset) unlock exactly their one flow
+        if p.get("refresh") or p.get("purpose"):
+            raise PermissionError_("not an access token", 401)
         return Identity(
             type=p["type"],
             id=p["sub"],
@@ -331,16 +336,37 @@ class ServerCore:
         return PermissionManager(self.session).collection(resource, self._user(ident))
 
     def _check_user_can(
-        self, resource: str, op: Operation, ident: Identity, *, org_id=None, collab_id=None
+        self, resource: str, op: Operation, ident: Identity, *, org_id=None,
+        collab_id=None, owner_id=None,
     ) -> None:
         """Raise unless `ident` (a user) may perform op on a resource owned by
-        org_id / inside collab_id."""
+        org_id / inside collab_id. ``owner_id`` is the id of the USER who owns
+        the resource (task initiator, target user account): an effective scope
+        of OWN grants access only when that owner is the caller — never the
+        whole organization."""
         rc = self._rc(resource, ident)
         scope = rc.allowed_scope(op)
         if scope is None:
             raise PermissionError_(f"user lacks {op.value} permission on {resource}")
         if scope == Scope.GLOBAL:
             return
+        if scope == Scope.OWN:
+            if owner_id is None or owner_id != ident.id:
+                raise PermissionError_(
+                    f"{op.value} on {resource}: own-scope covers only your own resources")
+            # ownership proven; still enforce containment (own-scope is a
+            # strict subset of organization scope, never broader)
+            if collab_id is not None:
+                collab = self.session.get(Collaboration, collab_id)
+                if collab and any(o.id == ident.organization_id
+                                  for o in collab.organizations):
+                    return
+                raise PermissionError_(
+                    f"{op.value} on {resource}: not in collaboration {collab_id}")
+            if org_id is None or org_id == ident.organization_id:
+                return
+            raise PermissionError_(
+                f"{op.value} on {resource}: organization scope exceeded")
         if scope == Scope.COLLABORATION:
             if collab_id is None:
                 if org_id is not None and (
@@ -353,7 +379,7 @@ class ServerCore:
             if collab and any(o.id == ident.organization_id for o in collab.organizations):
                 return
             raise PermissionError_(f"{op.value} on {resource}: not in collaboration {collab_id}")
-        # ORGANIZATION / OWN
+        # ORGANIZATION
         if org_id is not None and org_id == ident.organization_id:
             return
         raise PermissionError_(f"{op.value} on {resource}: organization scope exceeded")
@@ -531,7 +557,8 @@ class ServerCore:
             raise PermissionError_("users only")
         if op == Operation.EDIT and target.id == ident.id:
             return  # self-service (password/email change)
-        self._check_user_can("user", op, ident, org_id=target.organization_id)
+        self._check_user_can("user", op, ident, org_id=target.organization_id,
+                             owner_id=target.id)
 
     @_locked
     def list_users(self, ident: Identity, page=1, per_page=20,
@@ -561,7 +588,7 @@ class ServerCore:
             raise NotFound(f"user {user_id} not found")
         if user.id != ident.id:
             self._check_user_can("user", Operation.VIEW, ident,
-                                 org_id=user.organization_id)
+                                 org_id=user.organization_id, owner_id=user.id)
         return self._user_json(user)
 
     @_locked
@@ -791,7 +818,10 @@ class ServerCore:
             parent: Task | None = None
             init_user_id = None
             if ident.type == "user":
-                self._check_user_can("task", Operation.CREATE, ident, collab_id=collaboration_id)
+                # the creator owns the task — own-scope CREATE is permitted
+                # (containment to the collaboration still applies)
+                self._check_user_can("task", Operation.CREATE, ident,
+                                     collab_id=collaboration_id, owner_id=ident.id)
                 init_user_id = ident.id
             elif ident.type == "container":
                 # subtask fan-out: same collaboration, live parent
@@ -875,7 +905,8 @@ class ServerCore:
             raise NotFound(f"task {task_id} not found")
         if ident.type == "user":
             self._check_user_can("task", Operation.VIEW, ident,
-                                 org_id=task.init_org_id, collab_id=task.collaboration_id)
+                                 org_id=task.init_org_id, collab_id=task.collaboration_id,
+                                 owner_id=task.init_user_id)
         return self._task_json(task)
 
     @_locked
@@ -922,7 +953,8 @@ class ServerCore:
             raise NotFound(f"task {task_id} not found")
         if ident.type == "user":
             self._check_user_can("task", Operation.DELETE, ident,
-                                 org_id=task.init_org_id, collab_id=task.collaboration_id)
+                                 org_id=task.init_org_id, collab_id=task.collaboration_id,
+                                 owner_id=task.init_user_id)
         kill_list = []
         for run in task.runs:
             if not has_task_finished(run.status):
@@ -974,7 +1006,8 @@ class ServerCore:
             raise PermissionError_("node may only patch its own runs")
         if ident.type == "user":
             self._check_user_can("run", Operation.EDIT, ident,
-                                 org_id=run.organization_id)
+                                 org_id=run.organization_id,
+                                 owner_id=run.task.init_user_id if run.task else None)
         old_status = run.status
         if (old_status == RunStatus.KILLED.value and ident.type != "user"
                 and fields.get("status") != RunStatus.KILLED.value):
