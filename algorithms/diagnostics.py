@@ -79,3 +79,18 @@ def partial_read_marker(meta):
         marker = Path(meta.temporary_directory) / f"diag_marker_{meta.organization_id}"
         seen = marker.exists()
     return {"marker_seen": seen, "organization_id": meta.organization_id}
+
+
+def partial_spin(seconds: float = 300.0, marker_file: str | None = None):
+    """Busy-loop for the kill-preemption test: burns CPU in a Python loop
+    until ``seconds`` elapse (far longer than any test timeout). Writes
+    ``marker_file`` once started so the test knows the loop is live."""
+    import time
+
+    t0 = time.monotonic()
+    if marker_file:
+        Path(marker_file).write_text("spinning")
+    x = 0
+    while time.monotonic() - t0 < seconds:
+        x += 1  # pure-Python loop: interruptible at every bytecode
+    return {"iterations": x}

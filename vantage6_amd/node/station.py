@@ -39,6 +39,29 @@ log = logging.getLogger(__name__)
 _STOP = object()
 
 
+class KilledInterrupt(BaseException):
+    """Raised asynchronously inside an algorithm thread on kill_containers.
+
+    BaseException so a broad ``except Exception`` inside user algorithm code
+    cannot swallow the kill. The in-process analog of the reference's
+    container kill (``DockerManager.kill_tasks``,
+    ``vantage6-node/vantage6/node/docker/docker_manager.py``): the algorithm
+    stops at its next Python bytecode boundary, the station thread and GPU
+    are freed for the next task.
+    """
+
+
+def _async_raise(thread_ident: int, exc_type) -> None:
+    """Inject ``exc_type`` into a running thread (CPython API)."""
+    import ctypes
+
+    res = ctypes.pythonapi.PyThreadState_SetAsyncExc(
+        ctypes.c_ulong(thread_ident), ctypes.py_object(exc_type))
+    if res > 1:  # "injected into more than one thread" — undo, per CPython docs
+        ctypes.pythonapi.PyThreadState_SetAsyncExc(
+            ctypes.c_ulong(thread_ident), None)
+
+
 class Station:
     """In-process data-station worker."""
 
@@ -60,6 +83,7 @@ class Station:
         # the reference's one-Docker-container-per-run (a central task blocks
         # in wait_for_results while its partial runs on the same station)
         self._kill_flags: set[int] = set()  # run ids killed while queued/active
+        self._active: dict[int, int] = {}  # run id -> executing thread ident
         self._lock = threading.Lock()
         self._inflight = 0
         if private_key_file:
@@ -151,6 +175,14 @@ class Station:
     def stop(self, timeout: float = 10.0) -> None:
         if getattr(self, "_stop_ping", None) is not None:
             self._stop_ping.set()
+        # preempt anything still executing (reference: daemon shutdown kills
+        # its task containers); without this, non-daemon pool threads keep
+        # the interpreter alive at exit
+        with self._lock:
+            active_tids = list(self._active.values())
+            self._kill_flags.update(self._active.keys())
+        for tid in active_tids:
+            _async_raise(tid, KilledInterrupt)
         self.queue.put(_STOP)
         if self._thread:
             self._thread.join(timeout)
@@ -178,10 +210,18 @@ class Station:
             if data.get("organization_id") == self.identity.organization_id:
                 self.queue.put(data["task_id"])
         elif event == "kill_containers":
+            to_interrupt = []
             with self._lock:
                 for item in data.get("kill_list", []):
                     if item.get("organization_id") == self.identity.organization_id:
-                        self._kill_flags.add(item["run_id"])
+                        rid = item["run_id"]
+                        self._kill_flags.add(rid)
+                        tid = self._active.get(rid)
+                        if tid is not None:
+                            to_interrupt.append(tid)
+            # preempt RUNNING algorithms (the reference kills the container)
+            for tid in to_interrupt:
+                _async_raise(tid, KilledInterrupt)
 
     # ------------------------------------------------------------- hot loop
     def _listening_worker(self) -> None:
@@ -214,10 +254,11 @@ class Station:
         if run is None:
             return
         run_id = run["id"]
-        # kill semantics: a queued run is dropped here; an ALREADY-RUNNING
-        # algorithm function cannot be pre-empted in-process (the reference
-        # kills the Docker container) — the server marks the run killed
-        # (terminal) and any late patch from us cannot resurrect it
+        # kill semantics: a queued run is dropped here; a RUNNING algorithm
+        # is pre-empted via an async KilledInterrupt injected into its
+        # executor thread (see _on_event) — the in-process analog of the
+        # reference killing the Docker container. The server marks the run
+        # killed (terminal) and any late patch from us cannot resurrect it.
         with self._lock:
             if run_id in self._kill_flags:
                 self._kill_flags.discard(run_id)
@@ -306,6 +347,14 @@ class Station:
 
         # -- execute on the pinned device, capturing stdout/stderr as the log
         self.core.patch_run(self.identity, run_id, status=RunStatus.ACTIVE.value)
+        with self._lock:
+            if run_id in self._kill_flags:  # killed between queue-pop and start
+                self._kill_flags.discard(run_id)
+                return
+            self._active[run_id] = threading.get_ident()
+        import io
+
+        buf = io.StringIO()  # in case the kill lands before capture starts
         try:
             with capture_thread_output() as buf:
                 self._bind_device()
@@ -318,6 +367,16 @@ class Station:
                 self.identity, run_id, status=RunStatus.COMPLETED.value,
                 result=result_str, log=buf.getvalue(),
             )
+        except KilledInterrupt:
+            # status already KILLED server-side; log the preemption (the
+            # terminal-killed guard keeps the status from changing)
+            try:
+                self.core.patch_run(
+                    self.identity, run_id, status=RunStatus.KILLED.value,
+                    log=buf.getvalue() + "\nalgorithm pre-empted by kill request",
+                )
+            except Exception:  # noqa: BLE001
+                log.exception("post-kill patch failed for run %s", run_id)
         except AlgorithmError as exc:
             self.core.patch_run(
                 self.identity, run_id, status=RunStatus.FAILED.value,
@@ -328,6 +387,10 @@ class Station:
                 self.identity, run_id, status=RunStatus.CRASHED.value,
                 log=buf.getvalue() + "\n" + traceback.format_exc(),
             )
+        finally:
+            with self._lock:
+                self._active.pop(run_id, None)
+                self._kill_flags.discard(run_id)
 
     def _encrypt_result(self, blob: bytes, init_org_id: int | None) -> str:
         """Encrypt the result for the task's initiating organization.
