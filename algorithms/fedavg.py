@@ -116,6 +116,7 @@ def fit_rounds(
         if checkpoint is not None and checkpoint_every and \
                 (r + 1) % checkpoint_every == 0:
             save_checkpoint(trainer, checkpoint)
+    trainer.finalize()  # join any in-flight overlapped aggregation
     out = {
         "rounds": rounds,
         "final_loss": history[-1]["loss"],

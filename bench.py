@@ -68,6 +68,9 @@ def main() -> None:
     ap.add_argument("--dp", action="store_true", help="DP-SGD (ghost clipping)")
     ap.add_argument("--agg-algo", default="allreduce",
                     choices=["allreduce", "reduce_bcast"])
+    ap.add_argument("--overlap", dest="overlap", action="store_true", default=True,
+                    help="overlap round-k collective with round-k+1 forward (default)")
+    ap.add_argument("--no-overlap", dest="overlap", action="store_false")
     ap.add_argument("--channels-last", dest="channels_last", action="store_true",
                     default=True, help="NHWC layout for conv models (default)")
     ap.add_argument("--no-channels-last", dest="channels_last",
@@ -79,8 +82,17 @@ def main() -> None:
     ap.add_argument("--device", default=None, help="override (e.g. cpu for smoke)")
     ap.add_argument("--target-loss", type=float, default=None,
                     help="also report rounds/seconds until loss < target "
-                         "(BASELINE metric's wall-clock-to-target-loss half)")
+                         "(BASELINE metric's wall-clock-to-target-loss half; "
+                         "defaults to a per-model target below random-init loss)")
     args = ap.parse_args()
+    if args.target_loss is None:
+        # BASELINE.json names "wall-clock-to-target-loss" as half the
+        # headline metric — emit it by DEFAULT. Targets sit meaningfully
+        # below the random-init loss of each model (ln(1000)=6.9,
+        # ln(30522)=10.3, ln(128256)=11.8) so hitting them requires real
+        # optimization progress on the synthetic shards.
+        args.target_loss = {"resnet18": 6.0, "logreg": 0.45,
+                            "bert_base": 9.5, "llama_1b": 11.0}[args.model]
 
     comm = init_from_env()
     device = torch.device(args.device) if args.device else comm.device
@@ -128,6 +140,7 @@ def main() -> None:
         secure_agg=args.secure_agg,
         dp=dp_cfg,
         agg_algo=args.agg_algo,
+        overlap=args.overlap,
     )
     trainer = FedAvgTrainer(net, loss_fn, batches, comm, cfg, n_samples=shard)
     if dp_cfg is not None:
@@ -137,6 +150,7 @@ def main() -> None:
 
     for _ in range(args.warmup):
         trainer.run_round()
+    trainer.finalize()  # drain warmup's in-flight aggregation
     used_graph = trainer.enable_graph() if (on_gpu and args.graph) else False
 
     from vantage6_amd.utils.metrics import RoundMetrics
@@ -151,6 +165,7 @@ def main() -> None:
         stats = trainer.run_round()
         metrics.record(stats)
         losses.append(stats["loss"])
+    trainer.finalize()  # join in-flight overlapped aggregation
     comm.barrier()
     if on_gpu:
         torch.cuda.synchronize()
@@ -193,6 +208,9 @@ def main() -> None:
                 "secure_agg": args.secure_agg,
                 "dp": args.dp,
                 "hip_graph": used_graph,
+                "overlap": bool(args.overlap and comm.is_distributed
+                                and not args.secure_agg and on_gpu),
+                "overlap_stats": trainer.overlap_stats or None,
                 "dp_privacy": (trainer.dp_engine.epsilon_report(
                     steps=args.steps * args.local_steps,
                     sample_rate=min(1.0, batch / shard))

@@ -17,6 +17,8 @@ from safetensors.torch import load_file, save_file
 def save_checkpoint(trainer, path: str | Path) -> None:
     path = Path(path)
     path.parent.mkdir(parents=True, exist_ok=True)
+    if hasattr(trainer, "finalize"):
+        trainer.finalize()  # drain any in-flight overlapped aggregation
     if trainer.comm.rank == 0:
         tensors = {
             "master": trainer.flat.master_flat.cpu().contiguous(),

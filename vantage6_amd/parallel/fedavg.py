@@ -48,6 +48,14 @@ class FedAvgConfig:
     # 7-link xGMI topology the root's links aggregate ~1 TB/s in/out
     # (SURVEY.md §6.8); pick per measurement on the 8-GPU node.
     agg_algo: str = "allreduce"
+    # fit/agg overlap: the bucketed collective of round k
+    # runs on a dedicated HIP stream while round k+1's forward begins;
+    # per-bucket events gate each module's forward on exactly the buckets
+    # that hold its parameters, so the semantics are EXACTLY FedAvg — the
+    # forward simply starts before the full collective has landed. GPU +
+    # distributed + non-secure only; CPU/gloo and secure-agg stay
+    # synchronous.
+    overlap: bool = True
 
 
 class FedAvgTrainer:
@@ -86,6 +94,105 @@ class FedAvgTrainer:
         self.dp_engine = DPEngine(model, cfg.dp) if cfg.dp else None
         self.per_sample_loss_fn: Optional[Callable] = None  # set when dp on
         self.last_dp_stats: dict = {}
+        # ---- fit/agg overlap state --------------------------------------
+        self._agg_flat: Optional[torch.Tensor] = None  # collective workspace
+        self._agg_events: list = []  # [(lo, hi, cuda.Event)] per bucket
+        self._bucket_waited: list = []
+        self._agg_pending = False
+        self._stall_pairs: list = []  # (ev_a, ev_b) compute-stream stalls
+        self._ev_agg_start: Optional[torch.cuda.Event] = None
+        self._ev_agg_end: Optional[torch.cuda.Event] = None
+        self.overlap_stats: dict = {}  # comm_ms / exposed_ms / overlap_pct
+        if self._overlap_capable():
+            self._install_overlap_hooks()
+
+    # ------------------------------------------------- fit/agg overlap plumbing
+    def _overlap_capable(self) -> bool:
+        return (self.cfg.overlap and self.comm.device.type == "cuda"
+                and self.comm.is_distributed and not self.cfg.secure_agg)
+
+    def _install_overlap_hooks(self) -> None:
+        """Per-module forward-pre-hooks: wait on exactly the collective
+        buckets that hold this module's parameters (timed, to report the
+        exposed fraction of the collective)."""
+        by_param = {id(p): (off, n) for p, off, n, _ in self.flat._specs}
+        for mod in self.model.modules():
+            lo, hi = None, 0
+            for p in mod.parameters(recurse=False):
+                if id(p) in by_param:
+                    off, n = by_param[id(p)]
+                    lo = off if lo is None else min(lo, off)
+                    hi = max(hi, off + n)
+            if lo is None:
+                continue
+            mod.register_forward_pre_hook(self._make_overlap_hook(lo, hi))
+
+    def _make_overlap_hook(self, mlo: int, mhi: int):
+        def hook(mod, inputs):
+            if not self._agg_pending:
+                return
+            cur = torch.cuda.current_stream()
+            waited = False
+            ev_a = ev_b = None
+            for i, (lo, hi, ev) in enumerate(self._agg_events):
+                if not self._bucket_waited[i] and lo < mhi and hi > mlo:
+                    if not waited:
+                        ev_a = torch.cuda.Event(enable_timing=True)
+                        ev_a.record(cur)
+                        waited = True
+                    cur.wait_event(ev)
+                    self._bucket_waited[i] = True
+            if waited:
+                ev_b = torch.cuda.Event(enable_timing=True)
+                ev_b.record(cur)
+                self._stall_pairs.append((ev_a, ev_b))
+                if all(self._bucket_waited):
+                    self._agg_pending = False
+
+        return hook
+
+    def _join_agg(self) -> None:
+        """Make the compute stream wait for every outstanding collective
+        bucket (called before anything writes master/params: the optimizer
+        step, a graph replay, checkpointing)."""
+        if not self._agg_pending:
+            return
+        cur = torch.cuda.current_stream()
+        ev_a = torch.cuda.Event(enable_timing=True)
+        ev_a.record(cur)
+        for i, (lo, hi, ev) in enumerate(self._agg_events):
+            if not self._bucket_waited[i]:
+                cur.wait_event(ev)
+                self._bucket_waited[i] = True
+        ev_b = torch.cuda.Event(enable_timing=True)
+        ev_b.record(cur)
+        self._stall_pairs.append((ev_a, ev_b))
+        self._agg_pending = False
+
+    def finalize(self) -> None:
+        """Join any in-flight aggregation and drain the device (call before
+        reading final parameters or exiting the round loop)."""
+        self._join_agg()
+        if self.comm.device.type == "cuda":
+            torch.cuda.synchronize()
+
+    def _collect_overlap_stats(self) -> None:
+        """Called at the next aggregate launch: by then last round's events
+        are complete, so elapsed_time is free."""
+        if self._ev_agg_start is None:
+            return
+        try:
+            comm_ms = self._ev_agg_start.elapsed_time(self._ev_agg_end)
+            exposed_ms = sum(a.elapsed_time(b) for a, b in self._stall_pairs)
+        except RuntimeError:  # events not yet complete (shouldn't happen)
+            return
+        self.overlap_stats = {
+            "agg_comm_ms": comm_ms,
+            "agg_exposed_ms": exposed_ms,
+            "overlap_pct": max(0.0, 100.0 * (1.0 - exposed_ms / comm_ms))
+            if comm_ms > 0 else 0.0,
+        }
+        self._stall_pairs = []
 
     # -------------------------------------------------- hipGraph local step
     def enable_graph(self) -> bool:
@@ -100,6 +207,7 @@ class FedAvgTrainer:
         if self.comm.device.type != "cuda" or self.dp_engine is not None:
             return False
         try:
+            self.finalize()  # no in-flight collective may touch params mid-capture
             batch = next(self.batches)
             self._static_batch = {k: v.clone() for k, v in batch.items()}
             side = torch.cuda.Stream()
@@ -143,6 +251,7 @@ class FedAvgTrainer:
         if getattr(self, "_graph", None) is not None:
             for k, v in batch.items():
                 self._static_batch[k].copy_(v, non_blocking=True)
+            self._join_agg()  # replay reads AND writes params/master
             self._graph.replay()
             return float(self._static_loss.detach())
         if self.dp_engine is not None:
@@ -150,6 +259,7 @@ class FedAvgTrainer:
         self.flat.zero_grad()
         loss = self.loss_fn(self.model, batch)
         loss.backward()
+        self._join_agg()  # optimizer writes master/params
         lr = self.cfg.lr_schedule(self.round_idx) if self.cfg.lr_schedule else self.cfg.lr
         if self.cfg.optimizer == "adamw":
             self.flat.adamw_step(lr=lr, weight_decay=self.cfg.weight_decay)
@@ -169,6 +279,7 @@ class FedAvgTrainer:
             losses, self.flat.zero_grad, B
         )
         self.dp_engine.add_noise_and_scale(self.flat.grad_flat, B)
+        self._join_agg()  # optimizer writes master/params
         lr = self.cfg.lr_schedule(self.round_idx) if self.cfg.lr_schedule else self.cfg.lr
         if self.cfg.optimizer == "adamw":
             self.flat.adamw_step(lr=lr, weight_decay=self.cfg.weight_decay)
@@ -188,6 +299,9 @@ class FedAvgTrainer:
         master = self.flat.master_flat
         if not self.comm.is_distributed:
             self.flat.sync_param_from_master()
+            return
+        if self._overlap_capable():
+            self._aggregate_overlapped()
             return
         master.mul_(self.weight)
         bucket = max(1, self.cfg.bucket_bytes // 4)
@@ -214,6 +328,53 @@ class FedAvgTrainer:
             self.flat.param_flat[off : off + chunk.numel()].copy_(
                 chunk.to(self.flat.dtype)
             )
+
+    def _aggregate_overlapped(self) -> None:
+        """Launch the weighted FedAvg collective on the comm stream and
+        return immediately. Per-bucket, on the comm stream:
+
+            agg = weight * master  →  all-reduce(agg[b])  →
+            master[b] = agg[b]  →  param[b] = bf16(agg[b])  →  event[b]
+
+        Round k+1's forward waits per-module on exactly the buckets holding
+        its params (hooks), and the optimizer step joins the rest — so the
+        result is bit-identical to the synchronous path while the collective
+        hides behind the next forward. The separate ``agg`` workspace keeps
+        the collective's reads/writes off master until each bucket commits,
+        which is what makes the early forward safe."""
+        master = self.flat.master_flat
+        self._collect_overlap_stats()
+        if self._agg_flat is None:
+            self._agg_flat = torch.empty_like(master)
+            self._ev_agg_start = torch.cuda.Event(enable_timing=True)
+            self._ev_agg_end = torch.cuda.Event(enable_timing=True)
+        agg = self._agg_flat
+        bucket = max(1, self.cfg.bucket_bytes // 4)
+        use_rb = self.cfg.agg_algo == "reduce_bcast"
+        self.comm_stream.wait_stream(torch.cuda.current_stream())
+        events = []
+        with torch.cuda.stream(self.comm_stream):
+            self._ev_agg_start.record(self.comm_stream)
+            torch.mul(master, self.weight, out=agg)
+            for off in range(0, master.numel(), bucket):
+                chunk = agg[off : off + bucket]
+                if use_rb:
+                    w = self.comm.reduce(chunk, dst=0, async_op=True)
+                    w.wait()
+                    w = self.comm.broadcast(chunk, src=0, async_op=True)
+                else:
+                    w = self.comm.all_reduce(chunk, async_op=True)
+                w.wait()  # comm stream (not host) waits for this bucket
+                hi = off + chunk.numel()
+                master[off:hi].copy_(chunk)
+                self.flat.param_flat[off:hi].copy_(chunk.to(self.flat.dtype))
+                ev = torch.cuda.Event()
+                ev.record(self.comm_stream)
+                events.append((off, hi, ev))
+            self._ev_agg_end.record(self.comm_stream)
+        self._agg_events = events
+        self._bucket_waited = [False] * len(events)
+        self._agg_pending = True
 
     def _aggregate_secure(self) -> None:
         """Fixed-point + pairwise-mask secure aggregation.
@@ -250,13 +411,16 @@ class FedAvgTrainer:
 
     # ---------------------------------------------------------------- rounds
     def run_round(self) -> dict:
+        overlapping = self._overlap_capable()
         t0 = time.perf_counter()
         losses = [self.local_step() for _ in range(self.cfg.local_steps)]
         if self.comm.device.type == "cuda":
-            torch.cuda.synchronize()
+            # compute stream only: the in-flight collective (overlap mode)
+            # must keep running on the comm stream
+            torch.cuda.current_stream().synchronize()
         t1 = time.perf_counter()
         self.aggregate()
-        if self.comm.device.type == "cuda":
+        if self.comm.device.type == "cuda" and not overlapping:
             torch.cuda.synchronize()
         t2 = time.perf_counter()
         self.round_idx += 1
@@ -264,7 +428,11 @@ class FedAvgTrainer:
             "round": self.round_idx,
             "loss": sum(losses) / len(losses),
             "fit_ms": (t1 - t0) * 1e3,
+            # overlap mode: launch cost only — the collective itself runs
+            # under the next round's forward; see agg_comm_ms/overlap_pct
             "agg_ms": (t2 - t1) * 1e3,
             "round_ms": (t2 - t0) * 1e3,
         }
+        if overlapping and self.overlap_stats:
+            self.last_stats.update(self.overlap_stats)
         return self.last_stats

@@ -39,14 +39,21 @@ class RoundMetrics:
             self._prom["loss"].set(stats.get("loss", 0))
 
     def table(self, last: int = 20) -> str:
+        overlap = any("overlap_pct" in r for r in self.rows)
         hdr = f"{'round':>6} {'loss':>10} {'fit ms':>9} {'agg ms':>9} {'total ms':>9}"
+        if overlap:
+            hdr += f" {'comm ms':>9} {'overlap%':>9}"
         lines = [hdr, "-" * len(hdr)]
         for r in self.rows[-last:]:
-            lines.append(
+            line = (
                 f"{r.get('round', 0):>6} {r.get('loss', 0):>10.4f} "
                 f"{r.get('fit_ms', 0):>9.2f} {r.get('agg_ms', 0):>9.2f} "
                 f"{r.get('round_ms', 0):>9.2f}"
             )
+            if overlap:
+                line += (f" {r.get('agg_comm_ms', 0):>9.2f} "
+                         f"{r.get('overlap_pct', 0):>9.1f}")
+            lines.append(line)
         if self.rows:
             import statistics
 
