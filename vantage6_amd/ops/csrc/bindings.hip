@@ -77,6 +77,18 @@ extern "C" __global__ void gemm_nt_bf16_v7_kernel(const short*, const short*,
                                                   short*, int, int, int);
 extern "C" __global__ void gemm_nt_bf16_v8_kernel(const short*, const short*,
                                                   short*, int, int, int);
+extern "C" __global__ void gemm_nt_bf16_v9_kernel(const short*, const short*,
+                                                  short*, float*, int, int, int,
+                                                  int);
+extern "C" __global__ void gemm_nt_bf16_v9sk_kernel(const short*, const short*,
+                                                    short*, float*, int, int,
+                                                    int, int);
+extern "C" __global__ void gemm_nt_bf16_v10_kernel(const short*, const short*,
+                                                   short*, float*, int, int,
+                                                   int, int);
+extern "C" __global__ void gemm_nt_bf16_v10sk_kernel(const short*, const short*,
+                                                     short*, float*, int, int,
+                                                     int, int);
 extern "C" __global__ void gemm_wgrad_bf16_kernel(const short*, const short*,
                                                   short*, float*, int, int, int,
                                                   int);
@@ -663,6 +675,63 @@ torch::Tensor gemm_dgrad(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
+// 256²-tile NT with split-K: fills the chip when the tile grid is starved
+// (e.g. llama hidden GEMMs: 1024×2048 = 32 tiles over 256 CUs).
+static torch::Tensor gemm_nt_splitk(torch::Tensor A, torch::Tensor B, int M,
+                                    int N, int K, int nsplit, int variant) {
+  auto c = torch::empty({M, N}, A.options());
+  int tiles = (M / 256) * (N / 256);
+  auto kfn = variant == 10 ? gemm_nt_bf16_v10sk_kernel : gemm_nt_bf16_v9sk_kernel;
+  if (nsplit == 1) {
+    auto k1 = variant == 10 ? gemm_nt_bf16_v10_kernel : gemm_nt_bf16_v9_kernel;
+    hipLaunchKernelGGL(k1, dim3(tiles), dim3(512), 0, cur_stream(),
+                       (const short*)A.data_ptr(), (const short*)B.data_ptr(),
+                       (short*)c.data_ptr(), nullptr, M, N, K, 1);
+    return c;
+  }
+  auto ws = torch::empty({nsplit, M, N}, A.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(kfn, dim3(tiles * nsplit), dim3(512), 0, cur_stream(),
+                     (const short*)A.data_ptr(), (const short*)B.data_ptr(),
+                     nullptr, ws.data_ptr<float>(), M, N, K, nsplit);
+  long long n = (long long)M * N;
+  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid(n)), dim3(256), 0,
+                     cur_stream(), ws.data_ptr<float>(), (short*)c.data_ptr(),
+                     n, nsplit);
+  return c;
+}
+
+// direct variant launcher for A/B probes (no dispatch heuristics):
+// variant ∈ {5,6,7,8,9,10}; nsplit only for 9/10.
+torch::Tensor gemm_nt_probe(torch::Tensor a, torch::Tensor b, int variant,
+                            int nsplit) {
+  CHECK_HIP(a, torch::kBFloat16);
+  CHECK_HIP(b, torch::kBFloat16);
+  int M = (int)a.size(0), K = (int)a.size(1), N = (int)b.size(0);
+  TORCH_CHECK((int)b.size(1) == K, "inner dims mismatch");
+  if (variant >= 9) {
+    TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 64 == 0, "shape");
+    TORCH_CHECK((K / 64) % nsplit == 0, "nsplit must divide K-tiles");
+    return gemm_nt_splitk(a, b, M, N, K, nsplit, variant);
+  }
+  auto c = torch::empty({M, N}, a.options());
+  if (variant >= 6) {
+    TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 64 == 0, "shape");
+    auto k6 = variant == 8   ? gemm_nt_bf16_v8_kernel
+              : variant == 7 ? gemm_nt_bf16_v7_kernel
+                             : gemm_nt_bf16_v6_kernel;
+    hipLaunchKernelGGL(k6, dim3((M / 256) * (N / 256)), dim3(512), 0,
+                       cur_stream(), (const short*)a.data_ptr(),
+                       (const short*)b.data_ptr(), (short*)c.data_ptr(), M, N,
+                       K);
+    return c;
+  }
+  TORCH_CHECK(M % 128 == 0 && N % 128 == 0 && K % 64 == 0, "shape");
+  hipLaunchKernelGGL(gemm_nt_bf16_v5_kernel, dim3((M / 128) * (N / 128)),
+                     dim3(256), 0, cur_stream(), (const short*)a.data_ptr(),
+                     (const short*)b.data_ptr(), (short*)c.data_ptr(), M, N, K);
+  return c;
+}
+
 // C[M,N] = op(A)·op(B); inputs 2-D bf16 contiguous.
 torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b, bool trans_a,
                         bool trans_b) {
@@ -789,6 +858,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd", &bn_fwd, "fused NHWC BatchNorm fwd (+relu/+residual)");
   m.def("bn_bwd", &bn_bwd, "fused NHWC BatchNorm bwd");
   m.def("gemm_bf16", &gemm_bf16, "bf16 MFMA GEMM, fp32 accumulate");
+  m.def("gemm_nt_probe", &gemm_nt_probe, "direct NT kernel-variant launcher",
+        py::arg("a"), py::arg("b"), py::arg("variant"), py::arg("nsplit") = 1);
   m.def("gemm_wgrad", &gemm_wgrad, "A^T*B via tr_b16 reads (no transposes)");
   m.def("gemm_dgrad", &gemm_dgrad, "A*B with strided B via tr_b16 reads");
   m.def("transpose_bf16", &transpose_bf16);

@@ -1246,3 +1246,272 @@ gemm_dgrad_bf16_kernel(const short* __restrict__ A, const short* __restrict__ B,
       }
   }
 }
+
+// ===========================================================================
+// v9/v10 — the guide's two-barrier phase discipline on the 256² tile
+// (cdna_hip_programming.md §5 "256² 8-phase template"), with optional
+// split-K so starved grids (few 256² tiles) still fill 256 CUs. Each phase:
+//
+//   { staged glds piece ∥ fragment ds_reads } → s_barrier →
+//   lgkmcnt(0) → setprio(1) → 16 MFMA → setprio(0) → s_barrier
+//
+// The mid-phase barrier separates the memory phase from the MFMA phase
+// chip-wide (T5 stagger); vmcnt stays counted (never 0 mid-loop).
+//
+// STAGE_MODE 0 (v9): v8's per-wave piece schedule — phase p stages piece p
+//   of tile t+1 into buf^1; boundary wait vmcnt(2) after p0's own issue.
+// STAGE_MODE 1 (v10): group-aligned liveness schedule — strips are issued
+//   in the order their LDS slots die, which legalizes staging two tiles
+//   ahead and a boundary wait of vmcnt(4) with issue→consume distance of
+//   4-6 phases:
+//     p0: A strips a2,a3 of t+1 (G1: rows 64-127/192-255, buf^1)
+//     p1: B strips h2,h3 of t+1 (H1: cols (c&63)>=32, buf^1)
+//     p2: B strips h0,h1 of t+2 (H0 of THIS buf — dead since p0)
+//     p3: A strips a0,a1 of t+2 (G0 of THIS buf — dead since p2)
+// ===========================================================================
+template <int STAGE_MODE, int SPLITK>
+__device__ __forceinline__ void gemm_nt_v9_body(
+    const short* __restrict__ A, const short* __restrict__ B,
+    short* __restrict__ C, float* __restrict__ Cws, int M, int N, int K,
+    int nsplit) {
+  if (!SPLITK) nsplit = 1;  // compile-time: no workspace path at all
+  __shared__ short lds[2 * 2 * 2048 * 8];  // 2 buf × (A+B)[256][64] = 128 KB
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = (wave >> 2) * 128;
+  const int wc = (wave & 3) * 64;
+  const int nwg = gridDim.x;
+  int wgid = blockIdx.x;
+  {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = wgid & 7, idx = wgid >> 3;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int split = SPLITK ? wgid % nsplit : 0;
+  if (SPLITK) wgid /= nsplit;
+  const int ntile_n = N / 256;
+  const int row0 = (wgid / ntile_n) * 256;
+  const int col0 = (wgid % ntile_n) * 256;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int row_in_piece = lane >> 3;
+
+  // stage ONE 8-row strip of operand `op` (0=A at strip*1024, 1=B at
+  // 32768+strip*1024) for K-tile kt into buffer buf. The XOR source key
+  // matches the ds_read side's ((ks ^ (row>>1)) & 7) for any strip index.
+  auto stage_strip = [&](int buf, int kt, int strip, int op) {
+    const long long k0 = (long long)kt * BK;
+    char* base = (char*)lds + buf * 65536 + op * 32768;
+    const int src_koff = (((lane & 7) ^ (strip * 4 + (lane >> 4))) & 7) * 8;
+    const short* src = (op ? B : A) +
+        (long long)((op ? col0 : row0) + strip * 8 + row_in_piece) * K + k0 +
+        src_koff;
+    __builtin_amdgcn_global_load_lds((global_uint*)src,
+                                     (lds_uint*)(base + strip * 1024), 16, 0, 0);
+  };
+
+  const int nk_total = K / BK;
+  const int per = (nk_total + nsplit - 1) / nsplit;
+  const int kt0 = split * per;
+  const int kt1 = (kt0 + per < nk_total) ? kt0 + per : nk_total;
+  const int nk = kt1 - kt0;
+  const int crow_base = row0 + wr + (lane >> 4) * 4;
+  const int ccol_base = col0 + wc + (lane & 15);
+  // dispatch guarantees nk_total % nsplit == 0, so no slice is empty
+
+  // per-wave strip sets (STAGE_MODE 1): a0,a1 ∈ G0 (rows 0-63,128-191),
+  // a2,a3 = +8 ∈ G1; h0,h1 ∈ H0 (strips with (s&7)<4), h2,h3 = +4 ∈ H1.
+  const int aG0 = (wave >> 2) * 16 + (wave & 3) * 2;
+  const int hH0 = (wave >> 1) * 8 + (wave & 1) * 2;
+
+  if (STAGE_MODE == 0) {
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      stage_strip(0, kt0, wave * 4 + p, 0);
+      stage_strip(0, kt0, wave * 4 + p, 1);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  } else {
+    // prologue mirrors the steady-state issue order; tile kt0+1's H0/G0
+    // ride across the first boundary in flight (vmcnt(4))
+    stage_strip(0, kt0, hH0, 1);
+    stage_strip(0, kt0, hH0 + 1, 1);
+    stage_strip(0, kt0, aG0, 0);
+    stage_strip(0, kt0, aG0 + 1, 0);
+    stage_strip(0, kt0, aG0 + 8, 0);
+    stage_strip(0, kt0, aG0 + 9, 0);
+    stage_strip(0, kt0, hH0 + 4, 1);
+    stage_strip(0, kt0, hH0 + 5, 1);
+    if (nk > 1) {
+      stage_strip(1, kt0 + 1, hH0, 1);
+      stage_strip(1, kt0 + 1, hH0 + 1, 1);
+      stage_strip(1, kt0 + 1, aG0, 0);
+      stage_strip(1, kt0 + 1, aG0 + 1, 0);
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+  }
+  __builtin_amdgcn_s_barrier();
+
+  for (int kti = 0; kti < nk; ++kti) {
+    const int kt = kt0 + kti;
+    const int cur = kti & 1;
+    char* base = (char*)lds + cur * 65536;
+    bf16x8_t b_frag[2][2];
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      const int mh = p & 1, nh = p >> 1;
+      // ---- memory phase: staged glds piece, then this phase's ds_reads
+      if (STAGE_MODE == 0) {
+        if (kti + 1 < nk) {
+          stage_strip(cur ^ 1, kt + 1, wave * 4 + p, 0);
+          stage_strip(cur ^ 1, kt + 1, wave * 4 + p, 1);
+        }
+        if (p == 0 && kti > 0) {
+          // everything except p0's own 2 strips must have landed
+          if (kti + 1 < nk)
+            asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+          else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+          __builtin_amdgcn_s_barrier();
+        }
+      } else {
+        if (p == 0 && kti > 0) {
+          // in flight: the 4 strips issued at p2/p3 of the previous tile
+          // (tile kt+1's H0/G0); everything of tile kt has landed
+          if (kti + 1 < nk)
+            asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+          else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+          __builtin_amdgcn_s_barrier();
+        }
+        switch (p) {
+          case 0:
+            if (kti + 1 < nk) {
+              stage_strip(cur ^ 1, kt + 1, aG0 + 8, 0);
+              stage_strip(cur ^ 1, kt + 1, aG0 + 9, 0);
+            }
+            break;
+          case 1:
+            if (kti + 1 < nk) {
+              stage_strip(cur ^ 1, kt + 1, hH0 + 4, 1);
+              stage_strip(cur ^ 1, kt + 1, hH0 + 5, 1);
+            }
+            break;
+          case 2:
+            if (kti + 2 < nk) {
+              stage_strip(cur, kt + 2, hH0, 1);
+              stage_strip(cur, kt + 2, hH0 + 1, 1);
+            }
+            break;
+          case 3:
+            if (kti + 2 < nk) {
+              stage_strip(cur, kt + 2, aG0, 0);
+              stage_strip(cur, kt + 2, aG0 + 1, 0);
+            }
+            break;
+        }
+      }
+      if (mh == 0) {
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+          for (int kk = 0; kk < 2; ++kk) {
+            int row = wc + (nh * 2 + ni) * 16 + (lane & 15);
+            int ks = kk * 4 + (lane >> 4);
+            b_frag[ni][kk] = *(const bf16x8_t*)(
+                base + 32768 + row * 128 + ((ks ^ (row >> 1)) & 7) * 16);
+          }
+      }
+      bf16x8_t a_frag[4][2];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+          int row = wr + (mh * 4 + mi) * 16 + (lane & 15);
+          int ks = kk * 4 + (lane >> 4);
+          a_frag[mi][kk] = *(const bf16x8_t*)(
+              base + row * 128 + ((ks ^ (row >> 1)) & 7) * 16);
+        }
+      // ---- mid-phase barrier: memory phase ends, MFMA phase begins ----
+      __builtin_amdgcn_sched_barrier(0);
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+          for (int kk = 0; kk < 2; ++kk)
+            acc[mh * 4 + mi][nh * 2 + ni] =
+                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a_frag[mi][kk], b_frag[ni][kk],
+                    acc[mh * 4 + mi][nh * 2 + ni], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+    if (kti + 1 == nk) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  }
+  if (!SPLITK) {
+#pragma unroll
+    for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = crow_base + mi * 16 + r;
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          C[(long long)row * N + ccol_base + ni * 16] =
+              f32_to_bf16(acc[mi][ni][r]);
+      }
+  } else {
+    float* W = Cws + (long long)split * M * N;
+#pragma unroll
+    for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = crow_base + mi * 16 + r;
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          W[(long long)row * N + ccol_base + ni * 16] = acc[mi][ni][r];
+      }
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_nt_bf16_v9_kernel(const short* __restrict__ A, const short* __restrict__ B,
+                       short* __restrict__ C, float* __restrict__ Cws, int M,
+                       int N, int K, int nsplit) {
+  gemm_nt_v9_body<0, 0>(A, B, C, nullptr, M, N, K, 1);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_nt_bf16_v9sk_kernel(const short* __restrict__ A,
+                         const short* __restrict__ B, short* __restrict__ C,
+                         float* __restrict__ Cws, int M, int N, int K,
+                         int nsplit) {
+  gemm_nt_v9_body<0, 1>(A, B, C, Cws, M, N, K, nsplit);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_nt_bf16_v10_kernel(const short* __restrict__ A,
+                        const short* __restrict__ B, short* __restrict__ C,
+                        float* __restrict__ Cws, int M, int N, int K,
+                        int nsplit) {
+  gemm_nt_v9_body<1, 0>(A, B, C, nullptr, M, N, K, 1);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+gemm_nt_bf16_v10sk_kernel(const short* __restrict__ A,
+                          const short* __restrict__ B, short* __restrict__ C,
+                          float* __restrict__ Cws, int M, int N, int K,
+                          int nsplit) {
+  gemm_nt_v9_body<1, 1>(A, B, C, Cws, M, N, K, nsplit);
+}
