@@ -786,24 +786,65 @@ torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b, bool trans_a,
     int nwg = (M / 128) * (N / 128);
     static int variant = [] {
       const char* v = getenv("V6AMD_GEMM_KERNEL");
-      // default: shape-aware — v7 (256² counted-vmcnt) when divisible,
-      // v5 (128² dbuf) otherwise
-      return v ? atoi(v) : 0;
+      return v ? atoi(v) : 0;  // 0 = measured shape-aware dispatch below
     }();
     int nwg6_pre = (M / 256) * (N / 256);
-    // a 256² grid smaller than ~half the CUs starves the chip — the 128²
-    // tiles give 4× the workgroups (llama/bert wgrad shapes: 2048×2048
-    // outputs = 64 workgroups at 256²)
-    bool big_enough = nwg6_pre >= 128;
-    if ((variant >= 6 || (variant == 0 && big_enough)) && M % 256 == 0 &&
-        N % 256 == 0) {
-      int nwg6 = nwg6_pre;
-      // default: v8 (phase-interleaved) amortizes only at deep K
-      auto k6 = (variant == 8 || (variant == 0 && K >= 2560))
-                    ? gemm_nt_bf16_v8_kernel
-                    : gemm_nt_bf16_v7_kernel;
-      if (variant == 6) k6 = gemm_nt_bf16_v6_kernel;
-      hipLaunchKernelGGL(k6, dim3(nwg6), dim3(512), 0, cur_stream(),
+    bool div256 = M % 256 == 0 && N % 256 == 0;
+    // measured dispatch (tools/gemm_v9_ab.py, profiles/kernel_perf_r02.md):
+    //  big grid + deep K        → v8   (1054@4k³, 1116@8k³)
+    //  huge grid, mid K         → v10  (lm_head 1024×128256×2048: 978)
+    //  huge grid, shallow K     → v7   (bert_head 4096×30720×768: 807)
+    //  starved grid + deep K    → v10 split-K (llama down 1024×2048×8192:
+    //                             480 vs 340 on 128² tiles)
+    //  starved grid, shallow K  → v5 128² (llama gate 1024×8192×2048: 625
+    //                             beats every 256² variant)
+    if (variant == 0 && div256) {
+      if (nwg6_pre >= 128 && K >= 2560) {
+        hipLaunchKernelGGL(gemm_nt_bf16_v8_kernel, dim3(nwg6_pre), dim3(512),
+                           0, cur_stream(), (const short*)A.data_ptr(),
+                           (const short*)B.data_ptr(), (short*)c.data_ptr(), M,
+                           N, K);
+        if (M0 != M || N0 != N)
+          return c.narrow(0, 0, M0).narrow(1, 0, N0).contiguous();
+        return c;
+      }
+      if (nwg6_pre >= 384) {
+        if (K >= 1536) {
+          auto out = gemm_nt_splitk(A, B, M, N, K, 1, 10);
+          if (M0 != M || N0 != N)
+            return out.narrow(0, 0, M0).narrow(1, 0, N0).contiguous();
+          return out;
+        }
+        hipLaunchKernelGGL(gemm_nt_bf16_v7_kernel, dim3(nwg6_pre), dim3(512),
+                           0, cur_stream(), (const short*)A.data_ptr(),
+                           (const short*)B.data_ptr(), (short*)c.data_ptr(), M,
+                           N, K);
+        if (M0 != M || N0 != N)
+          return c.narrow(0, 0, M0).narrow(1, 0, N0).contiguous();
+        return c;
+      }
+      if (K >= 4096) {
+        int nk_tiles = K / 64;
+        int ns = 8;
+        while (ns > 1 && (nk_tiles % ns != 0 || nk_tiles / ns < 4)) ns >>= 1;
+        auto out = gemm_nt_splitk(A, B, M, N, K, ns, 10);
+        if (M0 != M || N0 != N)
+          return out.narrow(0, 0, M0).narrow(1, 0, N0).contiguous();
+        return out;
+      }
+      // fall through to v5 128² tiles (starved grid, shallow K)
+    }
+    if ((variant >= 6 || variant == 9 || variant == 10) && div256) {
+      if (variant >= 9) {
+        auto out = gemm_nt_splitk(A, B, M, N, K, 1, variant);
+        if (M0 != M || N0 != N)
+          return out.narrow(0, 0, M0).narrow(1, 0, N0).contiguous();
+        return out;
+      }
+      auto k6 = variant == 8   ? gemm_nt_bf16_v8_kernel
+                : variant == 7 ? gemm_nt_bf16_v7_kernel
+                               : gemm_nt_bf16_v6_kernel;
+      hipLaunchKernelGGL(k6, dim3(nwg6_pre), dim3(512), 0, cur_stream(),
                          (const short*)A.data_ptr(), (const short*)B.data_ptr(),
                          (short*)c.data_ptr(), M, N, K);
       if (M0 != M || N0 != N)
@@ -813,7 +854,8 @@ torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b, bool trans_a,
     auto kfn = gemm_nt_bf16_v4_kernel;
     if (variant == 2) kfn = gemm_nt_bf16_v2_kernel;
     if (variant == 3) kfn = gemm_nt_bf16_v3_kernel;
-    if (variant == 0 || variant == 5 || variant == 6 || variant == 7)
+    if (variant == 0 || variant == 5 || variant == 6 || variant == 7 ||
+        variant >= 9)
       kfn = gemm_nt_bf16_v5_kernel;
     hipLaunchKernelGGL(kfn, dim3(nwg), dim3(256), 0, cur_stream(),
                        (const short*)A.data_ptr(), (const short*)B.data_ptr(),
