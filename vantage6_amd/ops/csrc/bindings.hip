@@ -24,7 +24,8 @@ extern "C" __global__ void secagg_quantize_mask_kernel(long long*, const float*,
                                                        long long, double,
                                                        unsigned long long, int, int);
 extern "C" __global__ void dp_scale_noise_kernel(short*, long long, float, float,
-                                                 unsigned long long);
+                                                 unsigned long long,
+                                                 const int*);
 extern "C" __global__ void norm2_bf16_kernel(const short*, long long, float*);
 extern "C" __global__ void swiglu_fwd_kernel(const short*, const short*, short*,
                                              long long);
@@ -285,12 +286,20 @@ void secagg_quantize_mask(torch::Tensor q, torch::Tensor master, double weight,
 
 // -------------------------------------------------------------------- dp
 void dp_scale_noise(torch::Tensor grad, double scale, double noise_std,
-                    int64_t seed) {
+                    int64_t seed, c10::optional<torch::Tensor> step_dev) {
   CHECK_HIP(grad, torch::kBFloat16);
   long long n = grad.numel();
+  const int* sd = nullptr;
+  if (step_dev.has_value()) {
+    TORCH_CHECK(step_dev->scalar_type() == torch::kInt32, "step_dev int32");
+    sd = step_dev->data_ptr<int>();
+    // pre-increment (captured) so each graph replay draws fresh noise
+    hipLaunchKernelGGL(adamw_tick_kernel, dim3(1), dim3(1), 0, cur_stream(),
+                       const_cast<int*>(sd));
+  }
   hipLaunchKernelGGL(dp_scale_noise_kernel, dim3(ew_grid((n + 1) / 2)), dim3(256),
                      0, cur_stream(), (short*)grad.data_ptr(), n, (float)scale,
-                     (float)noise_std, (unsigned long long)seed);
+                     (float)noise_std, (unsigned long long)seed, sd);
 }
 
 torch::Tensor grad_norm2_partials(torch::Tensor grad) {
@@ -886,7 +895,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("step_dev") = py::none());
   m.def("secagg_quantize_mask", &secagg_quantize_mask,
         "fixed-point quantize + pairwise Philox masks");
-  m.def("dp_scale_noise", &dp_scale_noise, "DP clip-scale + gaussian noise");
+  m.def("dp_scale_noise", &dp_scale_noise, "DP clip-scale + gaussian noise",
+        py::arg("grad"), py::arg("scale"), py::arg("noise_std"), py::arg("seed"),
+        py::arg("step_dev") = py::none());
   m.def("grad_norm2_partials", &grad_norm2_partials, "block partial sum of g^2");
   m.def("rope_apply", &rope_apply);
   m.def("colsum_bf16", &colsum_bf16, "column sum of a bf16 matrix (bias grad)");

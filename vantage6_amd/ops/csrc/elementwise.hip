@@ -153,7 +153,12 @@ secagg_quantize_mask_kernel(long long* __restrict__ q,
 // applies scale + adds gaussian noise in one pass over the flat grad.
 extern "C" __global__ void __launch_bounds__(256)
 dp_scale_noise_kernel(short* __restrict__ grad, long long n, float scale,
-                      float noise_std, unsigned long long seed) {
+                      float noise_std, unsigned long long seed,
+                      const int* __restrict__ step_dev) {
+  // hipGraph-capturable: with step_dev set, the per-step seed component is
+  // read from a DEVICE counter (advanced by a captured tick kernel), so
+  // each replay draws fresh noise instead of freezing the capture seed
+  if (step_dev) seed += (unsigned long long)(unsigned)*step_dev;
   long long i0 = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 2;
   long long stride = (long long)gridDim.x * blockDim.x * 2;
   for (long long i = i0; i < n; i += stride) {

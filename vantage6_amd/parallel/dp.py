@@ -62,6 +62,8 @@ class DPEngine:
         self._norms: Optional[torch.Tensor] = None  # [B] accumulating norm^2
         self._capturing = False
         self._step = 0
+        self._step_dev: Optional[torch.Tensor] = None  # int32 device counter
+        self._stats_dev: Optional[torch.Tensor] = None  # [median, max, frac]
         self._hooked = []
         for mod in model.modules():
             has_params = any(True for _ in mod.parameters(recurse=False))
@@ -149,9 +151,14 @@ class DPEngine:
 
     # -------------------------------------------------------------- step
     def compute_clipped_grads(self, per_sample_losses: torch.Tensor,
-                              zero_grad_fn, batch_size: int) -> dict:
+                              zero_grad_fn, batch_size: int,
+                              sync_stats: bool = True) -> dict:
         """Run the two-pass ghost-clipping protocol; leaves the clipped,
-        noised MEAN gradient in the parameters' .grad buffers."""
+        noised MEAN gradient in the parameters' .grad buffers.
+
+        ``sync_stats=False`` is the hipGraph-capturable form: every op is
+        device-side; per-step stats land in a persistent device buffer
+        (read them with :meth:`stats`) instead of forcing a host sync."""
         cfg = self.cfg
         B = per_sample_losses.shape[0]
         # pass 1: norms
@@ -166,21 +173,40 @@ class DPEngine:
         zero_grad_fn()
         (per_sample_losses * clip.detach()).sum().backward()
         self._step += 1
-        return {
-            "median_grad_norm": float(norms.median()),
-            "max_grad_norm": float(norms.max()),
-            "clipped_fraction": float((clip < 1.0).float().mean()),
-        }
+        if self._stats_dev is None or self._stats_dev.device != norms.device:
+            # allocated during eager warmup (i.e. BEFORE any graph capture)
+            # so it is a persistent buffer, not a capture-pool tensor
+            self._stats_dev = torch.zeros(3, device=norms.device)
+        self._stats_dev[0].copy_(norms.median())
+        self._stats_dev[1].copy_(norms.max())
+        self._stats_dev[2].copy_((clip < 1.0).float().mean())
+        if not sync_stats:
+            return {}
+        return self.stats()
+
+    def stats(self) -> dict:
+        """Read the last step's clip stats (host sync; call off the hot
+        path — under graph replay the device buffer is updated per replay)."""
+        if self._stats_dev is None:
+            return {}
+        med, mx, frac = self._stats_dev.tolist()
+        return {"median_grad_norm": med, "max_grad_norm": mx,
+                "clipped_fraction": frac}
 
     def add_noise_and_scale(self, grad_flat: torch.Tensor, batch_size: int) -> None:
-        """grad = grad/B + N(0, sigma*C/B). One fused pass on GPU."""
+        """grad = grad/B + N(0, sigma*C/B). One fused pass on GPU; the
+        per-step seed component comes from an int32 DEVICE counter so the
+        pass is hipGraph-capturable (each replay draws fresh noise)."""
         cfg = self.cfg
         std = cfg.noise_multiplier * cfg.max_grad_norm / batch_size
         if grad_flat.is_cuda:
             from ..ops import hip_ops
 
+            if self._step_dev is None:
+                self._step_dev = torch.zeros(1, dtype=torch.int32,
+                                             device=grad_flat.device)
             hip_ops().dp_scale_noise(grad_flat, 1.0 / batch_size, std,
-                                     self._noise_seed + self._step)
+                                     self._noise_seed, self._step_dev)
         else:
             g = torch.Generator().manual_seed(self._noise_seed + self._step)
             noise = torch.randn(grad_flat.shape, generator=g) * std

@@ -200,11 +200,14 @@ class FedAvgTrainer:
         hipGraph: the ResNet round is ~19% launch-gap idle eager (≈250
         launches/step). Flat grads make this safe — every gradient buffer
         address is static. Call AFTER warmup (MIOpen find must be done).
-        Freezes lr/hyperparams at capture values; not for DP (dynamic
-        clipping control flow). AdamW is capturable because its bias
-        correction derives from an int32 DEVICE step counter that the
-        captured tick kernel advances on every replay."""
-        if self.comm.device.type != "cuda" or self.dp_engine is not None:
+        Freezes lr/hyperparams at capture values. AdamW is capturable
+        because its bias correction derives from an int32 DEVICE step
+        counter that the captured tick kernel advances on every replay.
+        DP-SGD is capturable too: the clip factors are a device tensor
+        multiplied into the per-sample losses (no host control flow), the
+        noise seed advances via a device counter, and clip stats land in a
+        persistent device buffer (dp_engine.stats())."""
+        if self.comm.device.type != "cuda":
             return False
         try:
             self.finalize()  # no in-flight collective may touch params mid-capture
@@ -233,9 +236,17 @@ class FedAvgTrainer:
             return False
 
     def _graph_body(self):
-        self.flat.zero_grad()
-        loss = self.loss_fn(self.model, self._static_batch)
-        loss.backward()
+        if self.dp_engine is not None:
+            losses = self.per_sample_loss_fn(self.model, self._static_batch)
+            B = losses.shape[0]
+            self.dp_engine.compute_clipped_grads(
+                losses, self.flat.zero_grad, B, sync_stats=False)
+            self.dp_engine.add_noise_and_scale(self.flat.grad_flat, B)
+            loss = losses.detach().mean()
+        else:
+            self.flat.zero_grad()
+            loss = self.loss_fn(self.model, self._static_batch)
+            loss.backward()
         lr = self.cfg.lr_schedule(self.round_idx) if self.cfg.lr_schedule else self.cfg.lr
         if self.cfg.optimizer == "adamw":
             self.flat.adamw_step(lr=lr, weight_decay=self.cfg.weight_decay)
