@@ -68,6 +68,11 @@ def main():
         ("llama_gate", 1024, 8192, 2048, [(5, 1), (7, 1), (9, 2), (9, 4), (10, 4)]),
         ("bert_head", 4096, 30720, 768, [(5, 1), (7, 1), (9, 2), (10, 2)]),
         ("lm_head", 1024, 128256, 2048, [(5, 1), (7, 1), (9, 1), (9, 2), (10, 1)]),
+        # bert fwd shapes (M=4096): nwg6 between the 128/384 thresholds
+        ("bert_qkv", 4096, 2304, 768, [(5, 1), (7, 1), (10, 1)]),
+        ("bert_fc1", 4096, 3072, 768, [(5, 1), (7, 1), (10, 1)]),
+        ("bert_fc2", 4096, 768, 3072, [(5, 1), (7, 1), (9, 2), (10, 2)]),
+        ("bert_proj", 4096, 768, 768, [(5, 1), (7, 1), (10, 1)]),
     ]
     for name, m, n, k, variants in shapes:
         a = torch.randn(m, k, device=dev, dtype=torch.bfloat16)
