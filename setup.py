@@ -17,7 +17,8 @@ ROOT = Path(__file__).resolve().parent
 CSRC = ROOT / "vantage6_amd" / "ops" / "csrc"
 
 sources = [str(CSRC / f) for f in
-           ("bindings.hip", "elementwise.hip", "norms.hip", "gemm.hip", "bn.hip")]
+           ("bindings.hip", "elementwise.hip", "norms.hip", "gemm.hip", "bn.hip",
+            "conv.hip")]
 
 setup(
     name="vantage6_amd",

@@ -18,10 +18,18 @@ def _make_trainer(dev, use_graph: bool):
     mk = {"in_features": 64, "num_classes": 4}
     net, loss_fn = _build("logreg", mk, dev)
     g = torch.Generator().manual_seed(11)
-    x = torch.randn(512, 64, generator=g)
-    dataset = {"x": x, "y": (x.sum(-1) > 0).long()}
+    x = torch.randn(64, 64, generator=g)
+    # CONSTANT batch stream: enable_graph consumes one batch for its static
+    # buffers, which would otherwise offset the graph trainer's data stream
+    # relative to the eager one
+    batch = {"x": x.to(dev), "y": (x.sum(-1) > 0).long().to(dev)}
+
+    def const_batches():
+        while True:
+            yield batch
+
     tr = FedAvgTrainer(
-        net, loss_fn, make_batches(dataset, 64, dev, seed=5),
+        net, loss_fn, const_batches(),
         CommContext(device=torch.device(dev)),
         FedAvgConfig(local_steps=2, lr=0.05,
                      dp=DPConfig(max_grad_norm=0.7, noise_multiplier=0.8,

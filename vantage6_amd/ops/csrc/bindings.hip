@@ -99,6 +99,14 @@ extern "C" __global__ void gemm_dgrad_bf16_kernel(const short*, const short*,
                                                   short*, float*, int, int, int,
                                                   int);
 extern "C" __global__ void transpose_bf16_kernel(const short*, short*, int, int);
+extern "C" __global__ void conv_nhwc_fwd64_kernel(const short*, const short*,
+                                                  short*, const short*, int,
+                                                  int, int, int, int, int, int,
+                                                  int, int, int, int);
+extern "C" __global__ void conv_nhwc_fwd128_kernel(const short*, const short*,
+                                                   short*, const short*, int,
+                                                   int, int, int, int, int,
+                                                   int, int, int, int, int);
 extern "C" __global__ void bn_fwd_partials_kernel(const short*, float*, long long,
                                                   int, int);
 extern "C" __global__ void bn_zero_acc_kernel(float*, int);
@@ -741,6 +749,41 @@ torch::Tensor gemm_nt_probe(torch::Tensor a, torch::Tensor b, int variant,
   return c;
 }
 
+// Direct NHWC conv forward (conv.hip): x [N,C,H,W] channels_last bf16,
+// w [Cout,Cin,KH,KW] channels_last bf16 → out [N,Cout,OH,OW] channels_last.
+torch::Tensor conv_nhwc_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
+                            int64_t pad) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "x NHWC");
+  TORCH_CHECK(w.is_contiguous(at::MemoryFormat::ChannelsLast), "w NHWC");
+  int Nb = (int)x.size(0), Cin = (int)x.size(1), H = (int)x.size(2),
+      Wd = (int)x.size(3);
+  int Cout = (int)w.size(0), KH = (int)w.size(2), KW = (int)w.size(3);
+  TORCH_CHECK((int)w.size(1) == Cin, "channel mismatch");
+  TORCH_CHECK(Cin % 64 == 0, "conv_nhwc_fwd needs Cin%64==0");
+  TORCH_CHECK(Cout % 64 == 0, "conv_nhwc_fwd needs Cout%64==0");
+  int OH = (H + 2 * (int)pad - KH) / (int)stride + 1;
+  int OW = (Wd + 2 * (int)pad - KW) / (int)stride + 1;
+  auto out = torch::empty({Nb, Cout, OH, OW},
+                          x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  // 128-short zero page for out-of-bounds gather lanes (halo)
+  static torch::Tensor zero_page;
+  if (!zero_page.defined() || zero_page.device() != x.device())
+    zero_page = torch::zeros({128}, x.options());
+  int M = Nb * OH * OW;
+  int mtiles = (M + 127) / 128;
+  bool wide = Cout % 128 == 0 && Cout > 64;
+  int tbn = wide ? 128 : 64;
+  dim3 grid(mtiles * (Cout / tbn));
+  auto kfn = wide ? conv_nhwc_fwd128_kernel : conv_nhwc_fwd64_kernel;
+  hipLaunchKernelGGL(kfn, grid, dim3(256), 0, cur_stream(),
+                     (const short*)x.data_ptr(), (const short*)w.data_ptr(),
+                     (short*)out.data_ptr(), (const short*)zero_page.data_ptr(),
+                     Nb, H, Wd, Cin, Cout, OH, OW, KH, KW, (int)stride,
+                     (int)pad);
+  return out;
+}
+
 // C[M,N] = op(A)·op(B); inputs 2-D bf16 contiguous.
 torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b, bool trans_a,
                         bool trans_b) {
@@ -911,6 +954,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd", &bn_fwd, "fused NHWC BatchNorm fwd (+relu/+residual)");
   m.def("bn_bwd", &bn_bwd, "fused NHWC BatchNorm bwd");
   m.def("gemm_bf16", &gemm_bf16, "bf16 MFMA GEMM, fp32 accumulate");
+  m.def("conv_nhwc_fwd", &conv_nhwc_fwd, "direct NHWC conv forward",
+        py::arg("x"), py::arg("w"), py::arg("stride") = 1, py::arg("pad") = 1);
   m.def("gemm_nt_probe", &gemm_nt_probe, "direct NT kernel-variant launcher",
         py::arg("a"), py::arg("b"), py::arg("variant"), py::arg("nsplit") = 1);
   m.def("gemm_wgrad", &gemm_wgrad, "A^T*B via tr_b16 reads (no transposes)");

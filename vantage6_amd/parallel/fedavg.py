@@ -211,6 +211,16 @@ class FedAvgTrainer:
             return False
         try:
             self.finalize()  # no in-flight collective may touch params mid-capture
+            # warmup + capture run ~4 real optimizer steps — snapshot the
+            # training state so enabling the graph does not perturb the
+            # trajectory (exact eager↔graph equivalence)
+            snap = {k: (v.clone() if torch.is_tensor(v) else v)
+                    for k, v in self.flat.state_dict().items()
+                    if v is not None}
+            dp_tick_snap = None
+            if self.dp_engine is not None and self.dp_engine._step_dev is not None:
+                dp_tick_snap = (self.dp_engine._step_dev.clone(),
+                                self.dp_engine._step)
             batch = next(self.batches)
             self._static_batch = {k: v.clone() for k, v in batch.items()}
             side = torch.cuda.Stream()
@@ -222,6 +232,10 @@ class FedAvgTrainer:
             self._graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(self._graph):
                 self._static_loss = self._graph_body()
+            self.flat.load_state_dict(snap)
+            if dp_tick_snap is not None:
+                self.dp_engine._step_dev.copy_(dp_tick_snap[0])
+                self.dp_engine._step = dp_tick_snap[1]
             if self.flat.scatter_grads:
                 # replays reuse the capture pool's grad addresses: point the
                 # persistent chunk table at them now (outside capture);
