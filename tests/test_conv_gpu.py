@@ -1,0 +1,66 @@
+"""Direct NHWC conv kernel numerics vs fp32 torch conv (tolerance-tiered
+bf16, same style as test_kernels_gpu.py)."""
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("cin,h,cout,k,s,p", [
+    (64, 56, 64, 3, 1, 1),
+    (128, 28, 128, 3, 1, 1),
+    (256, 14, 256, 3, 1, 1),
+    (512, 7, 512, 3, 1, 1),     # split-K path (starved grid)
+    (64, 56, 128, 3, 2, 1),
+    (256, 14, 512, 1, 2, 0),
+])
+def test_conv_fwd_matches_fp32(cin, h, cout, k, s, p):
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    from vantage6_amd.ops import hip_ops
+
+    torch.manual_seed(0)
+    x = torch.randn(8, cin, h, h, device="cuda", dtype=torch.bfloat16).to(
+        memory_format=torch.channels_last)
+    w = (torch.randn(cout, cin, k, k, device="cuda", dtype=torch.bfloat16)
+         * (2.0 / (cin * k * k)) ** 0.5).to(memory_format=torch.channels_last)
+    ref = F.conv2d(x.float(), w.float(), stride=s, padding=p)
+    got = hip_ops().conv_nhwc_fwd(x, w, s, p).float()
+    scale = ref.abs().max().item()
+    assert (got - ref).abs().max().item() <= 0.03 * scale + 0.05
+
+
+@pytest.mark.timeout(300)
+def test_conv_module_backward_matches_library():
+    """Conv2d wrapper: gradients equal the all-library path."""
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    import torch.nn as nn
+
+    from vantage6_amd.ops.conv import Conv2d as HipConv2d
+
+    torch.manual_seed(1)
+    x0 = torch.randn(4, 64, 28, 28, device="cuda", dtype=torch.bfloat16).to(
+        memory_format=torch.channels_last)
+    ours = HipConv2d(64, 128, 3, padding=1, bias=False).cuda().bfloat16().to(
+        memory_format=torch.channels_last)
+    lib = nn.Conv2d(64, 128, 3, padding=1, bias=False).cuda().bfloat16().to(
+        memory_format=torch.channels_last)
+    with torch.no_grad():
+        lib.weight.copy_(ours.weight)
+    xa = x0.clone().requires_grad_(True)
+    xb = x0.clone().requires_grad_(True)
+    ya = ours(xa)
+    yb = lib(xb)
+    scale = yb.float().abs().max().item()
+    assert (ya.float() - yb.float()).abs().max().item() <= 0.03 * scale + 0.05
+    ya.float().pow(2).mean().backward()
+    yb.float().pow(2).mean().backward()
+    gs = xb.grad.float().abs().max().item()
+    assert (xa.grad.float() - xb.grad.float()).abs().max().item() \
+        <= 0.05 * gs + 0.05
+    ws = lib.weight.grad.float().abs().max().item()
+    assert (ours.weight.grad.float() - lib.weight.grad.float()).abs().max() \
+        .item() <= 0.05 * ws + 0.05

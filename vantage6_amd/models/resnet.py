@@ -1,15 +1,18 @@
 """ResNet-18 (bf16) — the headline FedAvg benchmark model.
 
 Hand-written (no torchvision in the image): standard BasicBlock ResNet-18
-for 224×224 inputs. Convolutions run through MIOpen (library path — plain
-conv is a library GEMM-class op); the classifier head runs on the
-hand-written MFMA GEMM (ops.gemm.Linear) so the fit path exercises the
-native kernels; BatchNorm stays fp32-stats via torch.
+for 224×224 inputs. The residual 3×3 convs and 1×1 downsamples run on the
+hand-written NHWC implicit-GEMM kernel (ops/conv.py, forward; backward is
+MIOpen for now), the 3-channel 7×7 stem stays on MIOpen; the classifier
+head runs on the hand-written MFMA GEMM (ops.gemm.Linear); BatchNorm is
+the fused NHWC kernel set (ops/batchnorm.py).
 """
 from __future__ import annotations
 
 import torch
 import torch.nn as nn
+
+from ..ops.conv import Conv2d as HipConv2d
 
 from ..ops.gemm import Linear as MfmaLinear
 from ..ops.batchnorm import FusedBatchNorm2d
@@ -39,7 +42,8 @@ class BatchNorm2dBF16(nn.Module):
 
 
 def conv3x3(cin, cout, stride=1):
-    return nn.Conv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
+    # hand-written NHWC implicit-GEMM kernel on GPU (ops/conv.py)
+    return HipConv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
 
 
 class BasicBlock(nn.Module):
@@ -53,7 +57,7 @@ class BasicBlock(nn.Module):
         self.bn2 = FusedBatchNorm2d(cout)
         self.down_conv = None
         if stride != 1 or cin != cout:
-            self.down_conv = nn.Conv2d(cin, cout, 1, stride=stride, bias=False)
+            self.down_conv = HipConv2d(cin, cout, 1, stride=stride, bias=False)
             self.down_bn = FusedBatchNorm2d(cout)
 
     def forward(self, x):

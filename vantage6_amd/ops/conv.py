@@ -1,0 +1,52 @@
+"""Direct NHWC conv on the hand-written implicit-GEMM kernel (csrc/conv.hip).
+
+Forward runs the CDNA4 kernel for channels-last bf16 convs with
+Cin/Cout % 64 == 0 (every ResNet-18 conv except the 3-channel stem);
+backward delegates to torch's convolution_backward (MIOpen igemm) —
+replacing dgrad/wgrad with tr_b16-read kernels is a follow-up.
+
+Reference: the upstream framework ships no conv code (compute lives in
+user algorithm images); this is net-new MI355X work.
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+from . import hip_ops
+
+
+class _ConvFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, stride, pad):
+        y = hip_ops().conv_nhwc_fwd(x, w, stride, pad)
+        ctx.save_for_backward(x, w)
+        ctx.stride, ctx.pad = stride, pad
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        dx, dw, _ = torch.ops.aten.convolution_backward(
+            dy, x, w, None, [ctx.stride, ctx.stride], [ctx.pad, ctx.pad],
+            [1, 1], False, [0, 0], 1, [True, True, False])
+        return dx, dw, None, None
+
+
+class Conv2d(nn.Conv2d):
+    """Drop-in Conv2d: hand-written NHWC kernel on the eligible path,
+    library conv otherwise (3-channel stem, CPU, non-NHWC)."""
+
+    def forward(self, x):
+        w = self.weight
+        if (x.is_cuda and x.dtype == torch.bfloat16
+                and self.in_channels % 64 == 0 and self.out_channels % 64 == 0
+                and self.groups == 1 and self.dilation == (1, 1)
+                and self.bias is None
+                and self.stride[0] == self.stride[1]
+                and self.padding[0] == self.padding[1]
+                and x.is_contiguous(memory_format=torch.channels_last)
+                and w.is_contiguous(memory_format=torch.channels_last)):
+            return _ConvFn.apply(x, w, self.stride[0], self.padding[0])
+        return super().forward(x)

@@ -107,6 +107,16 @@ extern "C" __global__ void conv_nhwc_fwd128_kernel(const short*, const short*,
                                                    short*, const short*, int,
                                                    int, int, int, int, int,
                                                    int, int, int, int, int);
+extern "C" __global__ void conv_nhwc_fwd64sk_kernel(const short*, const short*,
+                                                    float*, const short*, int,
+                                                    int, int, int, int, int,
+                                                    int, int, int, int, int,
+                                                    int);
+extern "C" __global__ void conv_nhwc_fwd128sk_kernel(const short*, const short*,
+                                                     float*, const short*, int,
+                                                     int, int, int, int, int,
+                                                     int, int, int, int, int,
+                                                     int);
 extern "C" __global__ void bn_fwd_partials_kernel(const short*, float*, long long,
                                                   int, int);
 extern "C" __global__ void bn_zero_acc_kernel(float*, int);
@@ -774,9 +784,37 @@ torch::Tensor conv_nhwc_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
   int mtiles = (M + 127) / 128;
   bool wide = Cout % 128 == 0 && Cout > 64;
   int tbn = wide ? 128 : 64;
-  dim3 grid(mtiles * (Cout / tbn));
+  int tiles = mtiles * (Cout / tbn);
+  int nk = KH * KW * (Cin / 64);
+  // split the reduction when the tile grid starves the 256 CUs (deep
+  // layers: 7×7×512 outputs = 52 tiles); no empty slices allowed
+  int ns = 1;
+  if (tiles < 200) {
+    for (int cand = 8; cand > 1; cand >>= 1) {
+      int per = (nk + cand - 1) / cand;
+      bool no_empty_slice = (cand - 1) * per < nk;
+      if (no_empty_slice && tiles * cand >= 200 && tiles * cand <= 1024) {
+        ns = cand;
+        break;
+      }
+    }
+  }
+  if (ns > 1) {
+    auto ws = torch::empty({ns, M, Cout}, x.options().dtype(torch::kFloat32));
+    auto kfn = wide ? conv_nhwc_fwd128sk_kernel : conv_nhwc_fwd64sk_kernel;
+    hipLaunchKernelGGL(kfn, dim3(tiles * ns), dim3(256), 0, cur_stream(),
+                       (const short*)x.data_ptr(), (const short*)w.data_ptr(),
+                       ws.data_ptr<float>(),
+                       (const short*)zero_page.data_ptr(), Nb, H, Wd, Cin,
+                       Cout, OH, OW, KH, KW, (int)stride, (int)pad, ns);
+    long long n = (long long)M * Cout;
+    hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid(n)), dim3(256), 0,
+                       cur_stream(), ws.data_ptr<float>(),
+                       (short*)out.data_ptr(), n, ns);
+    return out;
+  }
   auto kfn = wide ? conv_nhwc_fwd128_kernel : conv_nhwc_fwd64_kernel;
-  hipLaunchKernelGGL(kfn, grid, dim3(256), 0, cur_stream(),
+  hipLaunchKernelGGL(kfn, dim3(tiles), dim3(256), 0, cur_stream(),
                      (const short*)x.data_ptr(), (const short*)w.data_ptr(),
                      (short*)out.data_ptr(), (const short*)zero_page.data_ptr(),
                      Nb, H, Wd, Cin, Cout, OH, OW, KH, KW, (int)stride,

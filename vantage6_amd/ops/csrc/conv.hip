@@ -22,12 +22,14 @@ typedef __bf16 cbf16x8_t __attribute__((ext_vector_type(8)));
 typedef __attribute__((address_space(3))) unsigned int lds_uint;
 typedef __attribute__((address_space(1))) const unsigned int global_uint;
 
-template <int TBN>
+template <int TBN, int SPLITK>
 __device__ __forceinline__ void conv_nhwc_fwd_body(
     const short* __restrict__ X, const short* __restrict__ Wt,
-    short* __restrict__ Out, const short* __restrict__ zero_page, int Nb,
-    int H, int Wd, int Cin, int Cout, int OH, int OW, int KH, int KW,
-    int stride, int pad) {
+    short* __restrict__ Out, float* __restrict__ Ws,
+    const short* __restrict__ zero_page, int Nb, int H, int Wd, int Cin,
+    int Cout, int OH, int OW, int KH, int KW, int stride, int pad,
+    int nsplit) {
+  if (!SPLITK) nsplit = 1;
   __shared__ short lds[2 * (128 + TBN) * 64];  // dbuf × (A[128][64]+B[TBN][64])
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -43,6 +45,8 @@ __device__ __forceinline__ void conv_nhwc_fwd_body(
     int xcd = wgid & 7, idx = wgid >> 3;
     wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
   }
+  const int split = SPLITK ? wgid % nsplit : 0;
+  if (SPLITK) wgid /= nsplit;
   const int ntile_n = Cout / TBN;
   const int row0 = (wgid / ntile_n) * 128;
   const int col0 = (wgid % ntile_n) * TBN;
@@ -72,7 +76,13 @@ __device__ __forceinline__ void conv_nhwc_fwd_body(
   }
 
   const int ck = Cin / 64;   // channel K-tiles per tap
-  const int nk = KH * KW * ck;
+  const int nk_total = KH * KW * ck;
+  // split-K over the (tap, channel) reduction; dispatch guarantees no
+  // empty slice ((nsplit-1)*ceil(nk/nsplit) < nk)
+  const int per = (nk_total + nsplit - 1) / nsplit;
+  const int kt0 = split * per;
+  const int kt1 = (kt0 + per < nk_total) ? kt0 + per : nk_total;
+  const int nk = kt1 - kt0;
   const int ldsB = 128 * 64;  // B image offset (shorts)
 
   auto stage = [&](int buf, int kt) {
@@ -109,11 +119,12 @@ __device__ __forceinline__ void conv_nhwc_fwd_body(
     }
   };
 
-  stage(0, 0);
+  stage(0, kt0);
   __syncthreads();
-  for (int kt = 0; kt < nk; ++kt) {
-    int cur = kt & 1;
-    if (kt + 1 < nk) stage(cur ^ 1, kt + 1);
+  for (int kti = 0; kti < nk; ++kti) {
+    int kt = kt0 + kti;
+    int cur = kti & 1;
+    if (kti + 1 < nk) stage(cur ^ 1, kt + 1);
     char* base = (char*)lds + cur * (128 + TBN) * 128;
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
@@ -148,10 +159,17 @@ __device__ __forceinline__ void conv_nhwc_fwd_body(
     for (int r = 0; r < 4; ++r) {
       int row = crow_base + mi * 16 + r;
       if (row < M) {
+        if (!SPLITK) {
 #pragma unroll
-        for (int ni = 0; ni < NFRAG; ++ni)
-          Out[(long long)row * Cout + ccol_base + ni * 16] =
-              f32_to_bf16(acc[mi][ni][r]);
+          for (int ni = 0; ni < NFRAG; ++ni)
+            Out[(long long)row * Cout + ccol_base + ni * 16] =
+                f32_to_bf16(acc[mi][ni][r]);
+        } else {
+          float* W = Ws + (long long)split * M * Cout;
+#pragma unroll
+          for (int ni = 0; ni < NFRAG; ++ni)
+            W[(long long)row * Cout + ccol_base + ni * 16] = acc[mi][ni][r];
+        }
       }
     }
 }
@@ -161,8 +179,8 @@ conv_nhwc_fwd64_kernel(const short* X, const short* Wt, short* Out,
                        const short* zero_page, int Nb, int H, int Wd, int Cin,
                        int Cout, int OH, int OW, int KH, int KW, int stride,
                        int pad) {
-  conv_nhwc_fwd_body<64>(X, Wt, Out, zero_page, Nb, H, Wd, Cin, Cout, OH, OW,
-                         KH, KW, stride, pad);
+  conv_nhwc_fwd_body<64, 0>(X, Wt, Out, nullptr, zero_page, Nb, H, Wd, Cin,
+                            Cout, OH, OW, KH, KW, stride, pad, 1);
 }
 
 extern "C" __global__ void __launch_bounds__(256)
@@ -170,6 +188,24 @@ conv_nhwc_fwd128_kernel(const short* X, const short* Wt, short* Out,
                         const short* zero_page, int Nb, int H, int Wd, int Cin,
                         int Cout, int OH, int OW, int KH, int KW, int stride,
                         int pad) {
-  conv_nhwc_fwd_body<128>(X, Wt, Out, zero_page, Nb, H, Wd, Cin, Cout, OH, OW,
-                          KH, KW, stride, pad);
+  conv_nhwc_fwd_body<128, 0>(X, Wt, Out, nullptr, zero_page, Nb, H, Wd, Cin,
+                             Cout, OH, OW, KH, KW, stride, pad, 1);
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+conv_nhwc_fwd64sk_kernel(const short* X, const short* Wt, float* Ws,
+                         const short* zero_page, int Nb, int H, int Wd,
+                         int Cin, int Cout, int OH, int OW, int KH, int KW,
+                         int stride, int pad, int nsplit) {
+  conv_nhwc_fwd_body<64, 1>(X, Wt, nullptr, Ws, zero_page, Nb, H, Wd, Cin,
+                            Cout, OH, OW, KH, KW, stride, pad, nsplit);
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+conv_nhwc_fwd128sk_kernel(const short* X, const short* Wt, float* Ws,
+                          const short* zero_page, int Nb, int H, int Wd,
+                          int Cin, int Cout, int OH, int OW, int KH, int KW,
+                          int stride, int pad, int nsplit) {
+  conv_nhwc_fwd_body<128, 1>(X, Wt, nullptr, Ws, zero_page, Nb, H, Wd, Cin,
+                             Cout, OH, OW, KH, KW, stride, pad, nsplit);
 }
