@@ -22,7 +22,8 @@ def _make_trainer(dev, use_graph: bool):
     # CONSTANT batch stream: enable_graph consumes one batch for its static
     # buffers, which would otherwise offset the graph trainer's data stream
     # relative to the eager one
-    batch = {"x": x.to(dev), "y": (x.sum(-1) > 0).long().to(dev)}
+    batch = {"x": x.to(dev, torch.bfloat16),
+             "y": (x.sum(-1) > 0).long().to(dev)}
 
     def const_batches():
         while True:

@@ -793,7 +793,10 @@ torch::Tensor conv_nhwc_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
     for (int cand = 8; cand > 1; cand >>= 1) {
       int per = (nk + cand - 1) / cand;
       bool no_empty_slice = (cand - 1) * per < nk;
-      if (no_empty_slice && tiles * cand >= 200 && tiles * cand <= 1024) {
+      // each slice needs >= 4 K-tiles or the double-buffer pipeline is
+      // all prologue (measured: 1x1 convs regress at per-slice nk 1)
+      if (no_empty_slice && nk / cand >= 4 && tiles * cand >= 200 &&
+          tiles * cand <= 1024) {
         ns = cand;
         break;
       }
