@@ -789,7 +789,10 @@ torch::Tensor conv_nhwc_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
   // split the reduction when the tile grid starves the 256 CUs (deep
   // layers: 7×7×512 outputs = 52 tiles); no empty slices allowed
   int ns = 1;
-  if (tiles < 200) {
+  // only truly starved grids: measured in-graph, the fp32-workspace +
+  // reduce overhead beats the fill gain above ~80 tiles (l2/l3 ResNet
+  // shapes regressed; l4 at 52 tiles gains 30-55%)
+  if (tiles < 80) {
     for (int cand = 8; cand > 1; cand >>= 1) {
       int per = (nk + cand - 1) / cand;
       bool no_empty_slice = (cand - 1) * per < nk;
