@@ -28,9 +28,22 @@ class _ConvFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, w = ctx.saved_tensors
         dy = dy.contiguous(memory_format=torch.channels_last)
+        s, p = ctx.stride, ctx.pad
+        k = w.shape[2]
+        if s == 1 and p == (k - 1) // 2 and w.shape[0] % 64 == 0:
+            # stride-1 dgrad IS a forward conv of dy with the spatially
+            # flipped, channel-transposed weights — reuse the hand-written
+            # kernel instead of MIOpen igemm_bwd
+            w_flip = w.flip(2, 3).transpose(0, 1).contiguous(
+                memory_format=torch.channels_last)
+            dx = hip_ops().conv_nhwc_fwd(dy, w_flip, 1, p)
+            _, dw, _ = torch.ops.aten.convolution_backward(
+                dy, x, w, None, [s, s], [p, p], [1, 1], False, [0, 0], 1,
+                [False, True, False])
+            return dx, dw, None, None
         dx, dw, _ = torch.ops.aten.convolution_backward(
-            dy, x, w, None, [ctx.stride, ctx.stride], [ctx.pad, ctx.pad],
-            [1, 1], False, [0, 0], 1, [True, True, False])
+            dy, x, w, None, [s, s], [p, p], [1, 1], False, [0, 0], 1,
+            [True, True, False])
         return dx, dw, None, None
 
 
