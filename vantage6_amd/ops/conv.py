@@ -34,8 +34,7 @@ class _ConvFn(torch.autograd.Function):
             # stride-1 dgrad IS a forward conv of dy with the spatially
             # flipped, channel-transposed weights — reuse the hand-written
             # kernel instead of MIOpen igemm_bwd
-            w_flip = w.flip(2, 3).transpose(0, 1).contiguous(
-                memory_format=torch.channels_last)
+            w_flip = hip_ops().conv_flip_weights(w)
             dx = hip_ops().conv_nhwc_fwd(dy, w_flip, 1, p)
             _, dw, _ = torch.ops.aten.convolution_backward(
                 dy, x, w, None, [s, s], [p, p], [1, 1], False, [0, 0], 1,

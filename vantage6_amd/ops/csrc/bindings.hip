@@ -107,6 +107,8 @@ extern "C" __global__ void conv_nhwc_fwd128_kernel(const short*, const short*,
                                                    short*, const short*, int,
                                                    int, int, int, int, int,
                                                    int, int, int, int, int);
+extern "C" __global__ void conv_flip_weights_kernel(const short*, short*, int,
+                                                    int, int, int);
 extern "C" __global__ void conv_nhwc_fwd64sk_kernel(const short*, const short*,
                                                     float*, const short*, int,
                                                     int, int, int, int, int,
@@ -834,6 +836,22 @@ torch::Tensor conv_nhwc_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
   return out;
 }
 
+// Weight transform for stride-1 dgrad (one kernel: flip + channel
+// transpose into channels-last layout for conv_nhwc_fwd).
+torch::Tensor conv_flip_weights(torch::Tensor w) {
+  TORCH_CHECK(w.is_cuda() && w.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(w.is_contiguous(at::MemoryFormat::ChannelsLast), "w NHWC");
+  int Cout = (int)w.size(0), Cin = (int)w.size(1), KH = (int)w.size(2),
+      KW = (int)w.size(3);
+  auto wd = torch::empty({Cin, Cout, KH, KW},
+                         w.options().memory_format(at::MemoryFormat::ChannelsLast));
+  int blocks = KH * KW * ((Cin + 63) / 64) * ((Cout + 63) / 64);
+  hipLaunchKernelGGL(conv_flip_weights_kernel, dim3(blocks), dim3(256), 0,
+                     cur_stream(), (const short*)w.data_ptr(),
+                     (short*)wd.data_ptr(), Cout, Cin, KH, KW);
+  return wd;
+}
+
 // C[M,N] = op(A)·op(B); inputs 2-D bf16 contiguous.
 torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b, bool trans_a,
                         bool trans_b) {
@@ -1004,6 +1022,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd", &bn_fwd, "fused NHWC BatchNorm fwd (+relu/+residual)");
   m.def("bn_bwd", &bn_bwd, "fused NHWC BatchNorm bwd");
   m.def("gemm_bf16", &gemm_bf16, "bf16 MFMA GEMM, fp32 accumulate");
+  m.def("conv_flip_weights", &conv_flip_weights,
+        "flip+transpose conv weights for stride-1 dgrad");
   m.def("conv_nhwc_fwd", &conv_nhwc_fwd, "direct NHWC conv forward",
         py::arg("x"), py::arg("w"), py::arg("stride") = 1, py::arg("pad") = 1);
   m.def("gemm_nt_probe", &gemm_nt_probe, "direct NT kernel-variant launcher",

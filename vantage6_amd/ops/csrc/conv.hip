@@ -209,3 +209,40 @@ conv_nhwc_fwd128sk_kernel(const short* X, const short* Wt, float* Ws,
   conv_nhwc_fwd_body<128, 1>(X, Wt, nullptr, Ws, zero_page, Nb, H, Wd, Cin,
                              Cout, OH, OW, KH, KW, stride, pad, nsplit);
 }
+
+// Weight transform for stride-1 dgrad: Wd[cin][KH-1-ky][KW-1-kx][cout] =
+// W[cout][ky][kx][cin]. One LDS-tiled transpose per (tap, 64×64 block):
+// reads and writes both coalesce on their channel-contiguous axes.
+extern "C" __global__ void __launch_bounds__(256)
+conv_flip_weights_kernel(const short* __restrict__ W, short* __restrict__ Wd,
+                         int Cout, int Cin, int KH, int KW) {
+  __shared__ short tile[64][64 + 8];  // +8 pad: conflict-free transpose
+  const int taps = KH * KW;
+  const int cot = (Cout + 63) >> 6, cit = (Cin + 63) >> 6;
+  int b = blockIdx.x;
+  const int tap = b % taps;
+  b /= taps;
+  const int ci0 = (b % cit) << 6;
+  const int co0 = (b / cit) << 6;
+  const int lane = threadIdx.x & 7;    // 8 threads × 8 shorts = one 64-row
+  const int row = threadIdx.x >> 3;    // 32 rows per pass
+  const int ftap = taps - 1 - tap;     // spatial flip
+  for (int rr = row; rr < 64; rr += 32) {
+    int co = co0 + rr;
+    if (co < Cout && ci0 + lane * 8 < Cin) {
+      *(s16x8*)&tile[rr][lane * 8] =
+          *(const s16x8*)(W + ((long long)co * taps + tap) * Cin + ci0 +
+                          lane * 8);
+    }
+  }
+  __syncthreads();
+  for (int rr = row; rr < 64; rr += 32) {
+    int ci = ci0 + rr;
+    if (ci < Cin && co0 + lane * 8 < Cout) {
+      s16x8 v;
+#pragma unroll
+      for (int q = 0; q < 8; ++q) v[q] = tile[lane * 8 + q][rr];
+      *(s16x8*)(Wd + ((long long)ci * taps + ftap) * Cout + co0 + lane * 8) = v;
+    }
+  }
+}
