@@ -64,3 +64,33 @@ def test_conv_module_backward_matches_library():
     ws = lib.weight.grad.float().abs().max().item()
     assert (ours.weight.grad.float() - lib.weight.grad.float()).abs().max() \
         .item() <= 0.05 * ws + 0.05
+
+
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("cin,h,cout,k,s,p", [
+    (64, 56, 64, 3, 1, 1),
+    (128, 28, 128, 3, 1, 1),
+    (512, 7, 512, 3, 1, 1),
+    (64, 56, 128, 3, 2, 1),
+    (256, 14, 512, 1, 2, 0),
+])
+def test_conv_wgrad_matches_fp32(cin, h, cout, k, s, p):
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    from vantage6_amd.ops import hip_ops
+
+    torch.manual_seed(2)
+    x = torch.randn(8, cin, h, h, device="cuda", dtype=torch.bfloat16).to(
+        memory_format=torch.channels_last)
+    oh = (h + 2 * p - k) // s + 1
+    dy = torch.randn(8, cout, oh, oh, device="cuda", dtype=torch.bfloat16).to(
+        memory_format=torch.channels_last) * 0.1
+    # fp32 reference weight grad
+    xr = x.float().requires_grad_(True)
+    wr = torch.zeros(cout, cin, k, k, device="cuda", requires_grad=True)
+    F.conv2d(xr, wr, stride=s, padding=p).backward(dy.float())
+    ref = wr.grad
+    got = hip_ops().conv_nhwc_wgrad(dy, x, k, k, s, p).float()
+    scale = ref.abs().max().item()
+    err = (got - ref).abs().max().item()
+    assert err <= 0.05 * scale + 0.05, (err, scale)

@@ -30,20 +30,25 @@ class _ConvFn(torch.autograd.Function):
         dy = dy.contiguous(memory_format=torch.channels_last)
         s, p = ctx.stride, ctx.pad
         k = w.shape[2]
+        cin = w.shape[1]
+        own_wgrad = cin >= 64 and (cin & (cin - 1)) == 0
+        if own_wgrad:
+            dw = hip_ops().conv_nhwc_wgrad(dy, x, k, k, s, p)
         if s == 1 and p == (k - 1) // 2 and w.shape[0] % 64 == 0:
             # stride-1 dgrad IS a forward conv of dy with the spatially
             # flipped, channel-transposed weights — reuse the hand-written
             # kernel instead of MIOpen igemm_bwd
             w_flip = hip_ops().conv_flip_weights(w)
             dx = hip_ops().conv_nhwc_fwd(dy, w_flip, 1, p)
-            _, dw, _ = torch.ops.aten.convolution_backward(
-                dy, x, w, None, [s, s], [p, p], [1, 1], False, [0, 0], 1,
-                [False, True, False])
+            if not own_wgrad:
+                _, dw, _ = torch.ops.aten.convolution_backward(
+                    dy, x, w, None, [s, s], [p, p], [1, 1], False, [0, 0], 1,
+                    [False, True, False])
             return dx, dw, None, None
-        dx, dw, _ = torch.ops.aten.convolution_backward(
+        dx, dw_lib, _ = torch.ops.aten.convolution_backward(
             dy, x, w, None, [s, s], [p, p], [1, 1], False, [0, 0], 1,
-            [True, True, False])
-        return dx, dw, None, None
+            [True, not own_wgrad, False])
+        return dx, (dw if own_wgrad else dw_lib), None, None
 
 
 class Conv2d(nn.Conv2d):

@@ -109,6 +109,11 @@ extern "C" __global__ void conv_nhwc_fwd128_kernel(const short*, const short*,
                                                    int, int, int, int, int);
 extern "C" __global__ void conv_flip_weights_kernel(const short*, short*, int,
                                                     int, int, int);
+extern "C" __global__ void conv_nhwc_wgrad_kernel(const short*, const short*,
+                                                  float*, short*, const short*,
+                                                  int, int, int, int, int, int,
+                                                  int, int, int, int, int, int,
+                                                  int);
 extern "C" __global__ void conv_nhwc_fwd64sk_kernel(const short*, const short*,
                                                     float*, const short*, int,
                                                     int, int, int, int, int,
@@ -836,6 +841,64 @@ torch::Tensor conv_nhwc_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
   return out;
 }
 
+// conv wgrad: dW = per-tap dy^T · gathered-x (conv.hip); returns the
+// channels-last weight-grad tensor. Split-K over output pixels.
+torch::Tensor conv_nhwc_wgrad(torch::Tensor dy, torch::Tensor x, int64_t kh,
+                              int64_t kw, int64_t stride, int64_t pad) {
+  CHECK_HIP(dy, torch::kBFloat16);
+  CHECK_HIP(x, torch::kBFloat16);
+  TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast), "dy NHWC");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "x NHWC");
+  int Nb = (int)x.size(0), Cin = (int)x.size(1), H = (int)x.size(2),
+      Wd = (int)x.size(3);
+  int Cout = (int)dy.size(1), OH = (int)dy.size(2), OW = (int)dy.size(3);
+  TORCH_CHECK(Cout % 64 == 0 && Cin >= 64 && (Cin & (Cin - 1)) == 0,
+              "conv_nhwc_wgrad needs Cout%64==0 and power-of-2 Cin>=64");
+  int cin_log2 = 0;
+  while ((1 << cin_log2) < Cin) ++cin_log2;
+  int M = Nb * OH * OW;
+  int taps = (int)(kh * kw);
+  int NbCols = taps * Cin;
+  auto dw = torch::empty({Cout, Cin, (long)kh, (long)kw},
+                         x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  static torch::Tensor zero_page;
+  if (!zero_page.defined() || zero_page.device() != x.device())
+    zero_page = torch::zeros({128}, x.options());
+  int tiles = ((Cout + 127) / 128) * ((NbCols + 127) / 128);
+  int nk = (M + 63) / 64;
+  int ns = 1;
+  for (int cand = 64; cand > 1; cand >>= 1) {
+    int per = (nk + cand - 1) / cand;
+    if ((cand - 1) * per < nk && nk / cand >= 8 && tiles * cand <= 768) {
+      ns = cand;
+      break;
+    }
+  }
+  if (ns == 1) {
+    hipLaunchKernelGGL(conv_nhwc_wgrad_kernel, dim3(tiles), dim3(256), 0,
+                       cur_stream(), (const short*)dy.data_ptr(),
+                       (const short*)x.data_ptr(), nullptr,
+                       (short*)dw.data_ptr(),
+                       (const short*)zero_page.data_ptr(), M, NbCols, H, Wd,
+                       cin_log2, Cout, OH, OW, (int)kh, (int)kw, (int)stride,
+                       (int)pad, 1);
+    return dw;
+  }
+  auto ws = torch::empty({ns, (long)Cout * NbCols},
+                         x.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(conv_nhwc_wgrad_kernel, dim3(tiles * ns), dim3(256), 0,
+                     cur_stream(), (const short*)dy.data_ptr(),
+                     (const short*)x.data_ptr(), ws.data_ptr<float>(), nullptr,
+                     (const short*)zero_page.data_ptr(), M, NbCols, H, Wd,
+                     cin_log2, Cout, OH, OW, (int)kh, (int)kw, (int)stride,
+                     (int)pad, ns);
+  long long n = (long long)Cout * NbCols;
+  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid(n)), dim3(256), 0,
+                     cur_stream(), ws.data_ptr<float>(), (short*)dw.data_ptr(),
+                     n, ns);
+  return dw;
+}
+
 // Weight transform for stride-1 dgrad (one kernel: flip + channel
 // transpose into channels-last layout for conv_nhwc_fwd).
 torch::Tensor conv_flip_weights(torch::Tensor w) {
@@ -1022,6 +1085,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd", &bn_fwd, "fused NHWC BatchNorm fwd (+relu/+residual)");
   m.def("bn_bwd", &bn_bwd, "fused NHWC BatchNorm bwd");
   m.def("gemm_bf16", &gemm_bf16, "bf16 MFMA GEMM, fp32 accumulate");
+  m.def("conv_nhwc_wgrad", &conv_nhwc_wgrad, "direct NHWC conv weight grad",
+        py::arg("dy"), py::arg("x"), py::arg("kh"), py::arg("kw"),
+        py::arg("stride") = 1, py::arg("pad") = 1);
   m.def("conv_flip_weights", &conv_flip_weights,
         "flip+transpose conv weights for stride-1 dgrad");
   m.def("conv_nhwc_fwd", &conv_nhwc_fwd, "direct NHWC conv forward",

@@ -246,3 +246,188 @@ conv_flip_weights_kernel(const short* __restrict__ W, short* __restrict__ Wd,
     }
   }
 }
+
+// ---------------------------------------------------------------------------
+// conv wgrad: dW[cout][tap][cin] = Σ_m dy[m,cout] · x[n, oh·S+dy−P, ow·S+dx−P, cin]
+// = per tap an A^T·B with reduction over all output pixels. Structure =
+// gemm.hip::gemm_wgrad_bf16_kernel (subtiled glds staging + tr_b16
+// fragment reads + split-K fp32 workspace); the B stage gathers shifted
+// input rows per (tap, cin) with zero-page redirect for halo/tail, and
+// the A stage pads Cout<128 tiles the same way.
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(4))) __bf16 cbf16x4_t;
+
+__device__ __forceinline__ cbf16x4_t cv_tr_read_b64(const char* addr) {
+  unsigned a32 =
+      (unsigned)(__SIZE_TYPE__)(__attribute__((address_space(3))) const char*)addr;
+  cbf16x4_t out;
+  asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(out) : "v"(a32) : "memory");
+  return out;
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+conv_nhwc_wgrad_kernel(const short* __restrict__ dY,
+                       const short* __restrict__ X, float* __restrict__ Ws,
+                       short* __restrict__ dW,
+                       const short* __restrict__ zero_page, int M, int Nb_,
+                       int H, int Wd, int Cin_log2, int Cout, int OH, int OW,
+                       int KH, int KW, int stride, int pad, int nsplit) {
+  __shared__ short lds[2 * 2 * 8192];  // 2 buf × (A+B)[64][128] subtiled
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = (wave >> 1) * 64;  // dW rows (cout)
+  const int wc = (wave & 1) * 64;   // dW cols (tap*Cin + cin space)
+  const int Cin = 1 << Cin_log2;
+  const int taps = KH * KW;
+  const int NbCols = taps << Cin_log2;  // 9*Cin
+  const int ntile_col = (NbCols + 127) >> 7;
+  int wgid = blockIdx.x;
+  {
+    int nwg = gridDim.x;
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = wgid & 7, idx = wgid >> 3;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int split = wgid % nsplit;
+  wgid /= nsplit;
+  const int row0 = (wgid / ntile_col) * 128;  // cout tile base
+  const int col0 = (wgid % ntile_col) * 128;  // (tap,cin) tile base
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int src_m_in_grp = (lane & 7) >> 1;
+  const int src_col = ((lane >> 3) * 16 + (lane & 1) * 8);
+  // B gather: this lane's fixed (tap, cin, within-tile m offset)
+  const int bcol = col0 + src_col;
+  const int btap = bcol >> Cin_log2;
+  const int bdy = btap / KW, bdx = btap - bdy * KW;
+  const int bcin = bcol & (Cin - 1);
+  const bool bcol_ok = bcol < NbCols;
+  const bool acol_ok = row0 + src_col < Cout;
+
+  auto stage = [&](int buf, int mt) {
+    const int m0 = mt * 64;
+    char* base = (char*)lds + buf * 32768;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int grp = wave * 4 + i;
+      int m = m0 + grp * 4 + src_m_in_grp;
+      const short* srcA =
+          (m < M && acol_ok) ? dY + (long long)m * Cout + row0 + src_col
+                             : zero_page;
+      __builtin_amdgcn_global_load_lds((global_uint*)srcA,
+                                       (lds_uint*)(base + grp * 1024), 16, 0,
+                                       0);
+      const short* srcB = zero_page;
+      if (m < M && bcol_ok) {
+        int n = m / (OH * OW);
+        int r = m - n * (OH * OW);
+        int oh = r / OW, ow = r - (r / OW) * OW;
+        int ih = oh * stride + bdy - pad;
+        int iw = ow * stride + bdx - pad;
+        if ((unsigned)ih < (unsigned)H && (unsigned)iw < (unsigned)Wd)
+          srcB = X + (((long long)(n * H + ih) * Wd + iw) << Cin_log2) + bcin;
+      }
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcB, (lds_uint*)(base + 16384 + grp * 1024), 16, 0,
+          0);
+    }
+  };
+
+  const int nk_total = (M + 63) / 64;
+  const int per = (nk_total + nsplit - 1) / nsplit;
+  const int mt0 = split * per;
+  const int mt1 = (mt0 + per < nk_total) ? mt0 + per : nk_total;
+  const int nk = mt1 - mt0;
+  stage(0, mt0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (int mti = 0; mti < nk; ++mti) {
+    int mt = mt0 + mti;
+    int cur = mti & 1;
+    if (mti + 1 < nk) {
+      stage(cur ^ 1, mt + 1);
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    char* base = (char*)lds + cur * 32768;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      cbf16x8_t a_frag[4], b_frag[4];
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi) {
+        int colA = wr + fi * 16;
+        cbf16x4_t lo, hi;
+        {
+          int g0 = kk * 8 + (lane >> 4) * 2;
+          const char* pa = base + ((g0 * 8 + (colA >> 4)) * 64) * 2 +
+                           (lane & 15) * 8;
+          lo = cv_tr_read_b64(pa);
+          const char* pa2 = base + (((g0 + 1) * 8 + (colA >> 4)) * 64) * 2 +
+                            (lane & 15) * 8;
+          hi = cv_tr_read_b64(pa2);
+        }
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          a_frag[fi][e] = lo[e];
+          a_frag[fi][e + 4] = hi[e];
+        }
+        int colB = wc + fi * 16;
+        {
+          int g0 = kk * 8 + (lane >> 4) * 2;
+          const char* pb = base + 16384 + ((g0 * 8 + (colB >> 4)) * 64) * 2 +
+                           (lane & 15) * 8;
+          lo = cv_tr_read_b64(pb);
+          const char* pb2 = base + 16384 +
+                            (((g0 + 1) * 8 + (colB >> 4)) * 64) * 2 +
+                            (lane & 15) * 8;
+          hi = cv_tr_read_b64(pb2);
+        }
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          b_frag[fi][e] = lo[e];
+          b_frag[fi][e + 4] = hi[e];
+        }
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+  // epilogue: map (tap,cin)-space cols back to dW[cout][tap][cin]
+  const int crow_base = row0 + wr + (lane >> 4) * 4;
+  const int ccol_base = col0 + wc + (lane & 15);
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int cout = crow_base + mi * 16 + r;
+      if (cout >= Cout) continue;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int c = ccol_base + ni * 16;
+        if (c >= NbCols) continue;
+        int tap = c >> Cin_log2;
+        int cin = c & (Cin - 1);
+        long long off = ((long long)cout * taps + tap) * (long long)Cin + cin;
+        if (nsplit == 1)
+          dW[off] = f32_to_bf16(acc[mi][ni][r]);
+        else
+          Ws[(long long)split * Cout * NbCols + off] = acc[mi][ni][r];
+      }
+    }
+}
