@@ -845,8 +845,8 @@ torch::Tensor conv_nhwc_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
 // channels-last weight-grad tensor. Split-K over output pixels.
 torch::Tensor conv_nhwc_wgrad(torch::Tensor dy, torch::Tensor x, int64_t kh,
                               int64_t kw, int64_t stride, int64_t pad) {
-  CHECK_HIP(dy, torch::kBFloat16);
-  CHECK_HIP(x, torch::kBFloat16);
+  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast), "dy NHWC");
   TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "x NHWC");
   int Nb = (int)x.size(0), Cin = (int)x.size(1), H = (int)x.size(2),
