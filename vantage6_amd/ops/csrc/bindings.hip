@@ -109,11 +109,10 @@ extern "C" __global__ void conv_nhwc_fwd128_kernel(const short*, const short*,
                                                    int, int, int, int, int);
 extern "C" __global__ void conv_flip_weights_kernel(const short*, short*, int,
                                                     int, int, int);
-extern "C" __global__ void conv_nhwc_wgrad_kernel(const short*, const short*,
-                                                  float*, short*, const short*,
-                                                  int, int, int, int, int, int,
-                                                  int, int, int, int, int, int,
-                                                  int);
+extern "C" __global__ void conv_nhwc_wgrad_kernel(
+    const short*, const short*, float*, short*, const short*, int, int, int,
+    int, int, int, int, int, int, int, int, int, int, unsigned long long,
+    unsigned long long);
 extern "C" __global__ void conv_nhwc_fwd64sk_kernel(const short*, const short*,
                                                     float*, const short*, int,
                                                     int, int, int, int, int,
@@ -866,6 +865,8 @@ torch::Tensor conv_nhwc_wgrad(torch::Tensor dy, torch::Tensor x, int64_t kh,
     zero_page = torch::zeros({128}, x.options());
   int tiles = ((Cout + 127) / 128) * ((NbCols + 127) / 128);
   int nk = (M + 63) / 64;
+  unsigned long long magic_ohow = ((1ULL << 40) / (OH * OW)) + 1;
+  unsigned long long magic_ow = ((1ULL << 40) / OW) + 1;
   int ns = 1;
   for (int cand = 64; cand > 1; cand >>= 1) {
     int per = (nk + cand - 1) / cand;
@@ -881,7 +882,7 @@ torch::Tensor conv_nhwc_wgrad(torch::Tensor dy, torch::Tensor x, int64_t kh,
                        (short*)dw.data_ptr(),
                        (const short*)zero_page.data_ptr(), M, NbCols, H, Wd,
                        cin_log2, Cout, OH, OW, (int)kh, (int)kw, (int)stride,
-                       (int)pad, 1);
+                       (int)pad, 1, magic_ohow, magic_ow);
     return dw;
   }
   auto ws = torch::empty({ns, (long)Cout * NbCols},
@@ -891,7 +892,7 @@ torch::Tensor conv_nhwc_wgrad(torch::Tensor dy, torch::Tensor x, int64_t kh,
                      (const short*)x.data_ptr(), ws.data_ptr<float>(), nullptr,
                      (const short*)zero_page.data_ptr(), M, NbCols, H, Wd,
                      cin_log2, Cout, OH, OW, (int)kh, (int)kw, (int)stride,
-                     (int)pad, ns);
+                     (int)pad, ns, magic_ohow, magic_ow);
   long long n = (long long)Cout * NbCols;
   hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid(n)), dim3(256), 0,
                      cur_stream(), ws.data_ptr<float>(), (short*)dw.data_ptr(),

@@ -265,13 +265,21 @@ __device__ __forceinline__ cbf16x4_t cv_tr_read_b64(const char* addr) {
   return out;
 }
 
+// division by the runtime constants OH*OW and OW via magic multiply:
+// floor(m/d) = (m * (2^40/d + 1)) >> 40, exact while m*d < 2^40.
+__device__ __forceinline__ int magic_div(int m, unsigned long long magic) {
+  return (int)(((unsigned long long)(unsigned)m * magic) >> 40);
+}
+
 extern "C" __global__ void __launch_bounds__(256)
 conv_nhwc_wgrad_kernel(const short* __restrict__ dY,
                        const short* __restrict__ X, float* __restrict__ Ws,
                        short* __restrict__ dW,
                        const short* __restrict__ zero_page, int M, int Nb_,
                        int H, int Wd, int Cin_log2, int Cout, int OH, int OW,
-                       int KH, int KW, int stride, int pad, int nsplit) {
+                       int KH, int KW, int stride, int pad, int nsplit,
+                       unsigned long long magic_ohow,
+                       unsigned long long magic_ow) {
   __shared__ short lds[2 * 2 * 8192];  // 2 buf × (A+B)[64][128] subtiled
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -325,9 +333,9 @@ conv_nhwc_wgrad_kernel(const short* __restrict__ dY,
                                        0);
       const short* srcB = zero_page;
       if (m < M && bcol_ok) {
-        int n = m / (OH * OW);
+        int n = magic_div(m, magic_ohow);
         int r = m - n * (OH * OW);
-        int oh = r / OW, ow = r - (r / OW) * OW;
+        int oh = magic_div(r, magic_ow), ow = r - oh * OW;
         int ih = oh * stride + bdy - pad;
         int iw = ow * stride + bdx - pad;
         if ((unsigned)ih < (unsigned)H && (unsigned)iw < (unsigned)Wd)
