@@ -18,7 +18,7 @@ CSRC = ROOT / "vantage6_amd" / "ops" / "csrc"
 
 sources = [str(CSRC / f) for f in
            ("bindings.hip", "elementwise.hip", "norms.hip", "gemm.hip", "bn.hip",
-            "conv.hip")]
+            "conv.hip", "attention.hip")]
 
 setup(
     name="vantage6_amd",

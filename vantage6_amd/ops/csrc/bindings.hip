@@ -109,6 +109,10 @@ extern "C" __global__ void conv_nhwc_fwd128_kernel(const short*, const short*,
                                                    int, int, int, int, int);
 extern "C" __global__ void conv_flip_weights_kernel(const short*, short*, int,
                                                     int, int, int);
+extern "C" __global__ void attn_fwd_d64_kernel(const short*, const short*,
+                                               const short*, short*, float*,
+                                               int, int, int, int, float,
+                                               int);
 extern "C" __global__ void conv_nhwc_wgrad_kernel(
     const short*, const short*, float*, short*, const short*, int, int, int,
     int, int, int, int, int, int, int, int, int, int, unsigned long long,
@@ -840,6 +844,37 @@ torch::Tensor conv_nhwc_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
   return out;
 }
 
+// flash-style attention forward, head_dim 64 (attention.hip).
+// q [B,H,S,64], k/v [B,Hkv,Skv,64] contiguous bf16 → (o, logsumexp)
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, bool causal,
+                                    double scale, bool want_lse) {
+  for (auto& t : {q, k, v}) {
+    TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kBFloat16 &&
+                t.is_contiguous(), "attn_fwd: contiguous bf16 cuda");
+  }
+  int B = (int)q.size(0), H = (int)q.size(1), S = (int)q.size(2);
+  int Hkv = (int)k.size(1), Skv = (int)k.size(2);
+  TORCH_CHECK(q.size(3) == 64 && k.size(3) == 64 && v.size(3) == 64,
+              "attn_fwd: head_dim must be 64");
+  TORCH_CHECK(H % Hkv == 0, "attn_fwd: H % Hkv");
+  auto o = torch::empty_like(q);
+  torch::Tensor lse;
+  float* lse_ptr = nullptr;
+  if (want_lse) {
+    lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat32));
+    lse_ptr = lse.data_ptr<float>();
+  } else {
+    lse = torch::empty({0}, q.options().dtype(torch::kFloat32));
+  }
+  dim3 grid((S + 63) / 64, B * H);
+  hipLaunchKernelGGL(attn_fwd_d64_kernel, grid, dim3(256), 0, cur_stream(),
+                     (const short*)q.data_ptr(), (const short*)k.data_ptr(),
+                     (const short*)v.data_ptr(), (short*)o.data_ptr(), lse_ptr,
+                     S, Skv, H, Hkv, (float)scale, causal ? 1 : 0);
+  return {o, lse};
+}
+
 // conv wgrad: dW = per-tap dy^T · gathered-x (conv.hip); returns the
 // channels-last weight-grad tensor. Split-K over output pixels.
 torch::Tensor conv_nhwc_wgrad(torch::Tensor dy, torch::Tensor x, int64_t kh,
@@ -1086,6 +1121,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd", &bn_fwd, "fused NHWC BatchNorm fwd (+relu/+residual)");
   m.def("bn_bwd", &bn_bwd, "fused NHWC BatchNorm bwd");
   m.def("gemm_bf16", &gemm_bf16, "bf16 MFMA GEMM, fp32 accumulate");
+  m.def("attn_fwd", &attn_fwd, "flash-style attention fwd (d=64)",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("causal") = false,
+        py::arg("scale") = 0.125, py::arg("want_lse") = true);
   m.def("conv_nhwc_wgrad", &conv_nhwc_wgrad, "direct NHWC conv weight grad",
         py::arg("dy"), py::arg("x"), py::arg("kh"), py::arg("kw"),
         py::arg("stride") = 1, py::arg("pad") = 1);
