@@ -35,11 +35,12 @@ __device__ __forceinline__ abf16x4_t at_tr_read_b64(const char* addr) {
 #define AV(buf) (lds + 8192 + (buf) * 4096)
 #define AP(w) (lds + 16384 + (w) * 1024)
 
-extern "C" __global__ void __launch_bounds__(512)
-attn_fwd_d64_kernel(const short* __restrict__ Q, const short* __restrict__ K,
-                    const short* __restrict__ V, short* __restrict__ O,
-                    float* __restrict__ Lse, int S, int Skv, int nheads,
-                    int kv_heads, float scale, int causal) {
+template <int PLIN>
+__device__ __forceinline__ void attn_fwd_d64_body(
+    const short* __restrict__ Q, const short* __restrict__ K,
+    const short* __restrict__ V, short* __restrict__ O,
+    float* __restrict__ Lse, int S, int Skv, int nheads, int kv_heads,
+    float scale, int causal) {
   __shared__ short lds[16384 + 8 * 1024];
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -168,32 +169,47 @@ attn_fwd_d64_kernel(const short* __restrict__ Q, const short* __restrict__ K,
 #pragma unroll
       for (int r = 0; r < 4; ++r) o_acc[ni][r] *= corr[r];
 
-    // ---- P → LDS transposed subtiles, packed b64 per ni ----------------
-    // value (key = ni*16+(lane&15), qrow = (lane>>4)*4+r) → subtile image
-    // [key>>2][key&3][qrow 0..15]; the 4 r-values are consecutive shorts
+    // ---- P → LDS ------------------------------------------------------
+    if (PLIN) {
+      // linear [16 q][64 key] rows, scalar scatter (debug/reference path)
 #pragma unroll
-    for (int ni = 0; ni < 4; ++ni) {
-      int key = ni * 16 + (lane & 15);
-      short4p_t packed;
+      for (int ni = 0; ni < 4; ++ni)
 #pragma unroll
-      for (int r = 0; r < 4; ++r) packed[r] = f32_to_bf16(s_acc[ni][r]);
-      *(short4p_t*)(AP(wave) + (key >> 2) * 64 + (key & 3) * 16 +
-                    (lane >> 4) * 4) = packed;
+        for (int r = 0; r < 4; ++r)
+          AP(wave)[((lane >> 4) * 4 + r) * 64 + ni * 16 + (lane & 15)] =
+              f32_to_bf16(s_acc[ni][r]);
+    } else {
+      // transposed subtiles [key>>2][key&3][16 q], packed b64 per ni
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int key = ni * 16 + (lane & 15);
+        short4p_t packed;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) packed[r] = f32_to_bf16(s_acc[ni][r]);
+        *(short4p_t*)(AP(wave) + (key >> 2) * 64 + (key & 3) * 16 +
+                      (lane >> 4) * 4) = packed;
+      }
     }
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    // ---- O += P·V (both operands via tr_b16 subtile reads) -------------
+    __builtin_amdgcn_sched_barrier(0);  // tr reads must not hoist above
+    // ---- O += P·V ------------------------------------------------------
 #pragma unroll
     for (int kkp = 0; kkp < 2; ++kkp) {
       int g0 = kkp * 8 + (lane >> 4) * 2;
-      abf16x4_t plo = at_tr_read_b64((const char*)AP(wave) + (g0 * 64) * 2 +
-                                     (lane & 15) * 8);
-      abf16x4_t phi = at_tr_read_b64((const char*)AP(wave) +
-                                     ((g0 + 1) * 64) * 2 + (lane & 15) * 8);
       abf16x8_t p_frag;
+      if (PLIN) {
+        p_frag = *(const abf16x8_t*)(AP(wave) + (lane & 15) * 64 + kkp * 32 +
+                                     (lane >> 4) * 8);
+      } else {
+        abf16x4_t plo = at_tr_read_b64((const char*)AP(wave) + (g0 * 64) * 2 +
+                                       (lane & 15) * 8);
+        abf16x4_t phi = at_tr_read_b64((const char*)AP(wave) +
+                                       ((g0 + 1) * 64) * 2 + (lane & 15) * 8);
 #pragma unroll
-      for (int e = 0; e < 4; ++e) {
-        p_frag[e] = plo[e];
-        p_frag[e + 4] = phi[e];
+        for (int e = 0; e < 4; ++e) {
+          p_frag[e] = plo[e];
+          p_frag[e + 4] = phi[e];
+        }
       }
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
@@ -210,6 +226,7 @@ attn_fwd_d64_kernel(const short* __restrict__ Q, const short* __restrict__ K,
           v_frag[e + 4] = hi[e];
         }
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_sched_barrier(0);  // MFMA must not hoist past wait
         o_acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             p_frag, v_frag, o_acc[ni], 0, 0, 0);
       }
@@ -230,4 +247,24 @@ attn_fwd_d64_kernel(const short* __restrict__ Q, const short* __restrict__ K,
       Lse[(long long)bh * S + qrow] =
           row_max[r] * 0.69314718f + logf(row_sum[r]);
   }
+}
+
+
+extern "C" __global__ void __launch_bounds__(512)
+attn_fwd_d64_kernel(const short* __restrict__ Q, const short* __restrict__ K,
+                    const short* __restrict__ V, short* __restrict__ O,
+                    float* __restrict__ Lse, int S, int Skv, int nheads,
+                    int kv_heads, float scale, int causal) {
+  attn_fwd_d64_body<0>(Q, K, V, O, Lse, S, Skv, nheads, kv_heads, scale,
+                       causal);
+}
+
+extern "C" __global__ void __launch_bounds__(512)
+attn_fwd_d64_plin_kernel(const short* __restrict__ Q,
+                         const short* __restrict__ K,
+                         const short* __restrict__ V, short* __restrict__ O,
+                         float* __restrict__ Lse, int S, int Skv, int nheads,
+                         int kv_heads, float scale, int causal) {
+  attn_fwd_d64_body<1>(Q, K, V, O, Lse, S, Skv, nheads, kv_heads, scale,
+                       causal);
 }

@@ -113,6 +113,10 @@ extern "C" __global__ void attn_fwd_d64_kernel(const short*, const short*,
                                                const short*, short*, float*,
                                                int, int, int, int, float,
                                                int);
+extern "C" __global__ void attn_fwd_d64_plin_kernel(const short*, const short*,
+                                                    const short*, short*,
+                                                    float*, int, int, int, int,
+                                                    float, int);
 extern "C" __global__ void conv_nhwc_wgrad_kernel(
     const short*, const short*, float*, short*, const short*, int, int, int,
     int, int, int, int, int, int, int, int, int, int, unsigned long long,
@@ -867,8 +871,13 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   } else {
     lse = torch::empty({0}, q.options().dtype(torch::kFloat32));
   }
-  dim3 grid((S + 63) / 64, B * H);
-  hipLaunchKernelGGL(attn_fwd_d64_kernel, grid, dim3(256), 0, cur_stream(),
+  dim3 grid((S + 127) / 128, B * H);
+  static int plin = [] {
+    const char* v = getenv("V6AMD_ATTN_PLIN");
+    return v ? atoi(v) : 0;
+  }();
+  auto kfn = plin ? attn_fwd_d64_plin_kernel : attn_fwd_d64_kernel;
+  hipLaunchKernelGGL(kfn, grid, dim3(512), 0, cur_stream(),
                      (const short*)q.data_ptr(), (const short*)k.data_ptr(),
                      (const short*)v.data_ptr(), (short*)o.data_ptr(), lse_ptr,
                      S, Skv, H, Hkv, (float)scale, causal ? 1 : 0);
