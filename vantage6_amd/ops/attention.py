@@ -1,0 +1,46 @@
+"""Hand-written flash-style attention (csrc/attention.hip), head_dim 64.
+
+Forward: online-softmax tiles, MFMA QKᵀ/PV, logsumexp saved. Backward:
+recompute-from-lse flash backward (dq kernel over q tiles, dk/dv kernel
+over kv tiles with the GQA group looped in-block). Numerics verified vs
+fp32 torch SDPA (tests/test_attention_gpu.py).
+
+The upstream framework has no attention code (compute lives in user
+images); aotriton SDPA was the round-1 library fallback.
+"""
+from __future__ import annotations
+
+import os
+
+import torch
+import torch.nn.functional as F
+
+from . import hip_ops
+
+
+class _AttnFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, causal, scale):
+        o, lse = hip_ops().attn_fwd(q, k, v, causal, scale, True)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.causal, ctx.scale = causal, scale
+        return o
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, o, lse = ctx.saved_tensors
+        dq, dk, dv = hip_ops().attn_bwd(q, k, v, o, dout, lse,
+                                        ctx.causal, ctx.scale)
+        return dq, dk, dv, None, None
+
+
+def sdpa(q, k, v, causal: bool = False):
+    """Drop-in scaled-dot-product attention: the hand-written kernel when
+    eligible (cuda bf16, head_dim 64, contiguous), library SDPA otherwise.
+    ``V6AMD_ATTN=lib`` forces the library path."""
+    if (q.is_cuda and q.dtype == torch.bfloat16 and q.shape[-1] == 64
+            and q.is_contiguous() and k.is_contiguous() and v.is_contiguous()
+            and os.environ.get("V6AMD_ATTN", "hip") != "lib"):
+        return _AttnFn.apply(q, k, v, causal, q.shape[-1] ** -0.5)
+    return F.scaled_dot_product_attention(q, k, v, is_causal=causal,
+                                          enable_gqa=True)

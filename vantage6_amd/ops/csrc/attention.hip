@@ -268,3 +268,416 @@ attn_fwd_d64_plin_kernel(const short* __restrict__ Q,
   attn_fwd_d64_body<1>(Q, K, V, O, Lse, S, Skv, nheads, kv_heads, scale,
                        causal);
 }
+
+// ===========================================================================
+// Backward. Split like flash2: a dQ kernel (blocks over q tiles) and a
+// dK/dV kernel (blocks over kv tiles), both recomputing P from the saved
+// logsumexp. D[q] = rowsum(dO ∘ O) is precomputed by attn_bwd_d_kernel.
+//
+//   dP = dO·Vᵀ ; dS = P ∘ (dP − D) · scale ; dQ = dS·K ; dK = dSᵀ·Q ;
+//   dV = Pᵀ·dO
+// ===========================================================================
+extern "C" __global__ void __launch_bounds__(256)
+attn_bwd_d_kernel(const short* __restrict__ dO, const short* __restrict__ O,
+                  float* __restrict__ D, long long n_rows) {
+  // one wave of 4 threads... simple: thread per row, 64-wide dot
+  long long row = (long long)blockIdx.x * 256 + threadIdx.x;
+  if (row >= n_rows) return;
+  const short* a = dO + row * 64;
+  const short* b = O + row * 64;
+  float acc = 0.f;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    s16x8 va = *(const s16x8*)(a + i * 8);
+    s16x8 vb = *(const s16x8*)(b + i * 8);
+#pragma unroll
+    for (int q = 0; q < 8; ++q)
+      acc = fmaf(bf16_to_f32(va[q]), bf16_to_f32(vb[q]), acc);
+  }
+  D[row] = acc;
+}
+
+// ---- dQ: grid (ceil(S/128), B*H); 8 waves, wave owns 16 q rows ----------
+// LDS: K swizzled-linear (B-frags of S) + K tr-subtiles (B-frags of
+// dQ GEMM) + V swizzled-linear (B-frags of dP) + per-wave dS tr image.
+#define BQ_KLIN (lds)
+#define BQ_KTR (lds + 4096)
+#define BQ_VLIN (lds + 8192)
+#define BQ_DS(w) (lds + 12288 + (w) * 1024)
+
+extern "C" __global__ void __launch_bounds__(512)
+attn_bwd_dq_d64_kernel(const short* __restrict__ Q, const short* __restrict__ K,
+                       const short* __restrict__ V, const short* __restrict__ dO,
+                       const float* __restrict__ Lse,
+                       const float* __restrict__ Dsum, short* __restrict__ dQ,
+                       int S, int Skv, int nheads, int kv_heads, float scale,
+                       int causal) {
+  __shared__ short lds[12288 + 8 * 1024];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int bh = blockIdx.y;
+  const int q0 = blockIdx.x * 128;
+  const int b = bh / nheads, head = bh - b * nheads;
+  const int kvbh = b * kv_heads + head / (nheads / kv_heads);
+  const short* Qb = Q + (long long)bh * S * 64;
+  const short* Kb = K + (long long)kvbh * Skv * 64;
+  const short* Vb = V + (long long)kvbh * Skv * 64;
+  const short* dOb = dO + (long long)bh * S * 64;
+
+  abf16x8_t q_frag[2], do_frag[2];
+  {
+    int row = q0 + wave * 16 + (lane & 15);
+    int rr = row < S ? row : S - 1;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      q_frag[kk] = *(const abf16x8_t*)(Qb + (long long)rr * 64 + kk * 32 +
+                                       (lane >> 4) * 8);
+      do_frag[kk] = *(const abf16x8_t*)(dOb + (long long)rr * 64 + kk * 32 +
+                                        (lane >> 4) * 8);
+    }
+  }
+  // per-lane row state (4 rows each)
+  float lse_r[4], d_r[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int qrow = q0 + wave * 16 + (lane >> 4) * 4 + r;
+    int rr = qrow < S ? qrow : S - 1;
+    lse_r[r] = Lse[(long long)bh * S + rr] * 1.44269504f;
+    d_r[r] = Dsum[(long long)bh * S + rr];
+  }
+  f32x4 dq_acc[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) dq_acc[i] = f32x4{0.f, 0.f, 0.f, 0.f};
+  const float l2e_scale = scale * 1.44269504f;
+
+  const int ntiles = (Skv + 63) / 64;
+  const int last_tile = causal ? (q0 + 127) / 64 : ntiles - 1;
+  const int nt = (last_tile < ntiles ? last_tile : ntiles - 1) + 1;
+
+  for (int kt = 0; kt < nt; ++kt) {
+    const int k0 = kt * 64;
+    // stage K into both images + V linear (512 threads, 16 B each)
+    {
+      int r = tid >> 3, slot = tid & 7;
+      int krow = k0 + r;
+      int src_slot = slot ^ ((r >> 1) & 7);
+      s16x8 zero = {0, 0, 0, 0, 0, 0, 0, 0};
+      *(s16x8*)(BQ_KLIN + r * 64 + slot * 8) =
+          krow < Skv ? *(const s16x8*)(Kb + (long long)krow * 64 + src_slot * 8)
+                     : zero;
+      *(s16x8*)(BQ_VLIN + r * 64 + slot * 8) =
+          krow < Skv ? *(const s16x8*)(Vb + (long long)krow * 64 + src_slot * 8)
+                     : zero;
+      short* ktr = BQ_KTR + ((r >> 2) * 4 + (slot >> 1)) * 64 + (r & 3) * 16 +
+                   (slot & 1) * 8;
+      *(s16x8*)ktr =
+          krow < Skv ? *(const s16x8*)(Kb + (long long)krow * 64 + slot * 8)
+                     : zero;
+    }
+    __syncthreads();
+
+    // S and dP fragments (B-frags from the linear swizzled images)
+    f32x4 s_acc[4], dp_acc[4];
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      s_acc[ni] = f32x4{0.f, 0.f, 0.f, 0.f};
+      dp_acc[ni] = f32x4{0.f, 0.f, 0.f, 0.f};
+    }
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      int row = ni * 16 + (lane & 15);
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        int ks = kk * 4 + (lane >> 4);
+        abf16x8_t k_frag = *(const abf16x8_t*)(
+            (const char*)BQ_KLIN + row * 128 + ((ks ^ (row >> 1)) & 7) * 16);
+        s_acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            q_frag[kk], k_frag, s_acc[ni], 0, 0, 0);
+        abf16x8_t v_frag = *(const abf16x8_t*)(
+            (const char*)BQ_VLIN + row * 128 + ((ks ^ (row >> 1)) & 7) * 16);
+        dp_acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            do_frag[kk], v_frag, dp_acc[ni], 0, 0, 0);
+      }
+    }
+    // dS = P ∘ (dP − D) · scale, written transposed-packed per wave
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      int col = k0 + ni * 16 + (lane & 15);
+      int key = ni * 16 + (lane & 15);
+      short4p_t packed;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int qrow = q0 + wave * 16 + (lane >> 4) * 4 + r;
+        float p = exp2f(s_acc[ni][r] * l2e_scale - lse_r[r]);
+        if (col >= Skv || qrow >= S || (causal && col > qrow)) p = 0.f;
+        float ds = p * (dp_acc[ni][r] - d_r[r]) * scale;
+        packed[r] = f32_to_bf16(ds);
+      }
+      *(short4p_t*)(BQ_DS(wave) + (key >> 2) * 64 + (key & 3) * 16 +
+                    (lane >> 4) * 4) = packed;
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
+    // dQ += dS · K  (A = dS via tr reads, B = Kᵀ via tr reads)
+#pragma unroll
+    for (int kkp = 0; kkp < 2; ++kkp) {
+      int g0 = kkp * 8 + (lane >> 4) * 2;
+      abf16x4_t plo = at_tr_read_b64((const char*)BQ_DS(wave) + g0 * 128 +
+                                     (lane & 15) * 8);
+      abf16x4_t phi = at_tr_read_b64((const char*)BQ_DS(wave) +
+                                     (g0 + 1) * 128 + (lane & 15) * 8);
+      abf16x8_t ds_frag;
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        ds_frag[e] = plo[e];
+        ds_frag[e + 4] = phi[e];
+      }
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        abf16x4_t lo = at_tr_read_b64((const char*)BQ_KTR +
+                                      (g0 * 4 + ni) * 128 + (lane & 15) * 8);
+        abf16x4_t hi = at_tr_read_b64((const char*)BQ_KTR +
+                                      ((g0 + 1) * 4 + ni) * 128 +
+                                      (lane & 15) * 8);
+        abf16x8_t k_frag;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          k_frag[e] = lo[e];
+          k_frag[e + 4] = hi[e];
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_sched_barrier(0);
+        dq_acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            ds_frag, k_frag, dq_acc[ni], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int qrow = q0 + wave * 16 + (lane >> 4) * 4 + r;
+    if (qrow >= S) continue;
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+      dQ[(long long)bh * S * 64 + (long long)qrow * 64 + ni * 16 +
+         (lane & 15)] = f32_to_bf16(dq_acc[ni][r]);
+  }
+}
+
+// ---- dK/dV: grid (ceil(Skv/128), B*Hkv); 8 waves, wave owns 16 kv rows.
+// GQA: the q heads sharing this kv head are looped INSIDE the block (dK/dV
+// accumulate in registers — no atomics). Per q tile:
+//   Sᵀ = K·Qᵀ ; Pᵀ = exp2(Sᵀ·scale·log2e − lse[col]) ; dPᵀ = V·dOᵀ ;
+//   dSᵀ = Pᵀ ∘ (dPᵀ − D[col]) · scale ; dV += Pᵀ·dO ; dK += dSᵀ·Q
+#define BK_QLIN (lds)
+#define BK_QTR (lds + 4096)
+#define BK_DOLIN (lds + 8192)
+#define BK_DOTR (lds + 12288)
+#define BK_PT(w) (lds + 16384 + (w) * 2048)
+#define BK_DST(w) (lds + 16384 + 8 * 2048 + (w) * 2048)
+
+extern "C" __global__ void __launch_bounds__(512)
+attn_bwd_dkdv_d64_kernel(const short* __restrict__ Q,
+                         const short* __restrict__ K,
+                         const short* __restrict__ V,
+                         const short* __restrict__ dO,
+                         const float* __restrict__ Lse,
+                         const float* __restrict__ Dsum,
+                         short* __restrict__ dK, short* __restrict__ dV,
+                         int S, int Skv, int nheads, int kv_heads, float scale,
+                         int causal) {
+  __shared__ short lds[16384 + 16 * 2048];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;  // wave owns kv rows [wave*16, +16)
+  const int bkv = blockIdx.y;
+  const int kv0 = blockIdx.x * 128;
+  const int b = bkv / kv_heads, kvh = bkv - b * kv_heads;
+  const int group = nheads / kv_heads;
+  const short* Kb = K + (long long)bkv * Skv * 64;
+  const short* Vb = V + (long long)bkv * Skv * 64;
+
+  // persistent K/V fragments for this wave's 16 kv rows
+  abf16x8_t k_frag[2], v_frag[2];
+  {
+    int row = kv0 + wave * 16 + (lane & 15);
+    int rr = row < Skv ? row : Skv - 1;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      k_frag[kk] = *(const abf16x8_t*)(Kb + (long long)rr * 64 + kk * 32 +
+                                       (lane >> 4) * 8);
+      v_frag[kk] = *(const abf16x8_t*)(Vb + (long long)rr * 64 + kk * 32 +
+                                       (lane >> 4) * 8);
+    }
+  }
+  f32x4 dk_acc[4], dv_acc[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    dk_acc[i] = f32x4{0.f, 0.f, 0.f, 0.f};
+    dv_acc[i] = f32x4{0.f, 0.f, 0.f, 0.f};
+  }
+  const float l2e_scale = scale * 1.44269504f;
+  const int nqt = (S + 63) / 64;
+  const int qt0 = causal ? kv0 >> 6 : 0;
+
+  for (int gh = 0; gh < group; ++gh) {
+    const int bh = b * nheads + kvh * group + gh;
+    const short* Qb = Q + (long long)bh * S * 64;
+    const short* dOb = dO + (long long)bh * S * 64;
+    for (int qt = qt0; qt < nqt; ++qt) {
+      const int q0 = qt * 64;
+      {  // stage Q and dO (linear-swizzled + tr images)
+        int r = tid >> 3, slot = tid & 7;
+        int qrow = q0 + r;
+        int src_slot = slot ^ ((r >> 1) & 7);
+        s16x8 zero = {0, 0, 0, 0, 0, 0, 0, 0};
+        *(s16x8*)(BK_QLIN + r * 64 + slot * 8) =
+            qrow < S
+                ? *(const s16x8*)(Qb + (long long)qrow * 64 + src_slot * 8)
+                : zero;
+        *(s16x8*)(BK_DOLIN + r * 64 + slot * 8) =
+            qrow < S
+                ? *(const s16x8*)(dOb + (long long)qrow * 64 + src_slot * 8)
+                : zero;
+        short* qtr = BK_QTR + ((r >> 2) * 4 + (slot >> 1)) * 64 +
+                     (r & 3) * 16 + (slot & 1) * 8;
+        *(s16x8*)qtr =
+            qrow < S ? *(const s16x8*)(Qb + (long long)qrow * 64 + slot * 8)
+                     : zero;
+        short* dotr = BK_DOTR + ((r >> 2) * 4 + (slot >> 1)) * 64 +
+                      (r & 3) * 16 + (slot & 1) * 8;
+        *(s16x8*)dotr =
+            qrow < S ? *(const s16x8*)(dOb + (long long)qrow * 64 + slot * 8)
+                     : zero;
+      }
+      __syncthreads();
+
+      // per-ni q-column state
+      float lse_c[4], d_c[4];
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int qcol = q0 + ni * 16 + (lane & 15);
+        int cc = qcol < S ? qcol : S - 1;
+        lse_c[ni] = Lse[(long long)bh * S + cc] * 1.44269504f;
+        d_c[ni] = Dsum[(long long)bh * S + cc];
+      }
+      // Sᵀ and dPᵀ
+      f32x4 st_acc[4], dpt_acc[4];
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        st_acc[ni] = f32x4{0.f, 0.f, 0.f, 0.f};
+        dpt_acc[ni] = f32x4{0.f, 0.f, 0.f, 0.f};
+      }
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int row = ni * 16 + (lane & 15);
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+          int ks = kk * 4 + (lane >> 4);
+          abf16x8_t qf = *(const abf16x8_t*)(
+              (const char*)BK_QLIN + row * 128 + ((ks ^ (row >> 1)) & 7) * 16);
+          st_acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              k_frag[kk], qf, st_acc[ni], 0, 0, 0);
+          abf16x8_t dof = *(const abf16x8_t*)(
+              (const char*)BK_DOLIN + row * 128 +
+              ((ks ^ (row >> 1)) & 7) * 16);
+          dpt_acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              v_frag[kk], dof, dpt_acc[ni], 0, 0, 0);
+        }
+      }
+      // Pᵀ and dSᵀ → per-wave tr images (packed over the key dim)
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int qcol = q0 + ni * 16 + (lane & 15);
+        int qloc = ni * 16 + (lane & 15);
+        short4p_t packp, packd;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int keyrow = kv0 + wave * 16 + (lane >> 4) * 4 + r;
+          float p = exp2f(st_acc[ni][r] * l2e_scale - lse_c[ni]);
+          if (qcol >= S || keyrow >= Skv || (causal && qcol < keyrow)) p = 0.f;
+          float ds = p * (dpt_acc[ni][r] - d_c[ni]) * scale;
+          packp[r] = f32_to_bf16(p);
+          packd[r] = f32_to_bf16(ds);
+        }
+        int off = (qloc >> 2) * 64 + (qloc & 3) * 16 + (lane >> 4) * 4;
+        *(short4p_t*)(BK_PT(wave) + off) = packp;
+        *(short4p_t*)(BK_DST(wave) + off) = packd;
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+      // dV += Pᵀ·dO ; dK += dSᵀ·Q   (A via tr reads of the per-wave
+      // images, B via tr reads of the dO/Q subtile images)
+#pragma unroll
+      for (int kkq = 0; kkq < 2; ++kkq) {
+        int g0 = kkq * 8 + (lane >> 4) * 2;
+        abf16x4_t lo = at_tr_read_b64((const char*)BK_PT(wave) + g0 * 128 +
+                                      (lane & 15) * 8);
+        abf16x4_t hi = at_tr_read_b64((const char*)BK_PT(wave) +
+                                      (g0 + 1) * 128 + (lane & 15) * 8);
+        abf16x8_t pt_frag;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          pt_frag[e] = lo[e];
+          pt_frag[e + 4] = hi[e];
+        }
+        lo = at_tr_read_b64((const char*)BK_DST(wave) + g0 * 128 +
+                            (lane & 15) * 8);
+        hi = at_tr_read_b64((const char*)BK_DST(wave) + (g0 + 1) * 128 +
+                            (lane & 15) * 8);
+        abf16x8_t dst_frag;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          dst_frag[e] = lo[e];
+          dst_frag[e + 4] = hi[e];
+        }
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni) {
+          abf16x4_t blo = at_tr_read_b64((const char*)BK_DOTR +
+                                         (g0 * 4 + ni) * 128 +
+                                         (lane & 15) * 8);
+          abf16x4_t bhi = at_tr_read_b64((const char*)BK_DOTR +
+                                         ((g0 + 1) * 4 + ni) * 128 +
+                                         (lane & 15) * 8);
+          abf16x8_t bf;
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            bf[e] = blo[e];
+            bf[e + 4] = bhi[e];
+          }
+          asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+          __builtin_amdgcn_sched_barrier(0);
+          dv_acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              pt_frag, bf, dv_acc[ni], 0, 0, 0);
+          blo = at_tr_read_b64((const char*)BK_QTR + (g0 * 4 + ni) * 128 +
+                               (lane & 15) * 8);
+          bhi = at_tr_read_b64((const char*)BK_QTR +
+                               ((g0 + 1) * 4 + ni) * 128 + (lane & 15) * 8);
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            bf[e] = blo[e];
+            bf[e + 4] = bhi[e];
+          }
+          asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+          __builtin_amdgcn_sched_barrier(0);
+          dk_acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              dst_frag, bf, dk_acc[ni], 0, 0, 0);
+        }
+      }
+      __syncthreads();
+    }
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int krow = kv0 + wave * 16 + (lane >> 4) * 4 + r;
+    if (krow >= Skv) continue;
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      dK[(long long)bkv * Skv * 64 + (long long)krow * 64 + ni * 16 +
+         (lane & 15)] = f32_to_bf16(dk_acc[ni][r]);
+      dV[(long long)bkv * Skv * 64 + (long long)krow * 64 + ni * 16 +
+         (lane & 15)] = f32_to_bf16(dv_acc[ni][r]);
+    }
+  }
+}

@@ -117,6 +117,16 @@ extern "C" __global__ void attn_fwd_d64_plin_kernel(const short*, const short*,
                                                     const short*, short*,
                                                     float*, int, int, int, int,
                                                     float, int);
+extern "C" __global__ void attn_bwd_d_kernel(const short*, const short*,
+                                             float*, long long);
+extern "C" __global__ void attn_bwd_dq_d64_kernel(const short*, const short*,
+                                                  const short*, const short*,
+                                                  const float*, const float*,
+                                                  short*, int, int, int, int,
+                                                  float, int);
+extern "C" __global__ void attn_bwd_dkdv_d64_kernel(
+    const short*, const short*, const short*, const short*, const float*,
+    const float*, short*, short*, int, int, int, int, float, int);
 extern "C" __global__ void conv_nhwc_wgrad_kernel(
     const short*, const short*, float*, short*, const short*, int, int, int,
     int, int, int, int, int, int, int, int, int, int, unsigned long long,
@@ -884,6 +894,39 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   return {o, lse};
 }
 
+// attention backward (attention.hip): recompute-from-lse flash backward.
+std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, torch::Tensor o,
+                                    torch::Tensor dout, torch::Tensor lse,
+                                    bool causal, double scale) {
+  int B = (int)q.size(0), H = (int)q.size(1), S = (int)q.size(2);
+  int Hkv = (int)k.size(1), Skv = (int)k.size(2);
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  auto dsum = torch::empty({B, H, S}, q.options().dtype(torch::kFloat32));
+  long long nrows = (long long)B * H * S;
+  auto dco = dout.contiguous();
+  hipLaunchKernelGGL(attn_bwd_d_kernel, dim3((nrows + 255) / 256), dim3(256),
+                     0, cur_stream(), (const short*)dco.data_ptr(),
+                     (const short*)o.data_ptr(), dsum.data_ptr<float>(), nrows);
+  hipLaunchKernelGGL(attn_bwd_dq_d64_kernel, dim3((S + 127) / 128, B * H),
+                     dim3(512), 0, cur_stream(), (const short*)q.data_ptr(),
+                     (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                     (const short*)dco.data_ptr(), lse.data_ptr<float>(),
+                     dsum.data_ptr<float>(), (short*)dq.data_ptr(), S, Skv, H,
+                     Hkv, (float)scale, causal ? 1 : 0);
+  hipLaunchKernelGGL(attn_bwd_dkdv_d64_kernel,
+                     dim3((Skv + 127) / 128, B * Hkv), dim3(512), 0,
+                     cur_stream(), (const short*)q.data_ptr(),
+                     (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                     (const short*)dco.data_ptr(), lse.data_ptr<float>(),
+                     dsum.data_ptr<float>(), (short*)dk.data_ptr(),
+                     (short*)dv.data_ptr(), S, Skv, H, Hkv, (float)scale,
+                     causal ? 1 : 0);
+  return {dq, dk, dv};
+}
+
 // conv wgrad: dW = per-tap dy^T · gathered-x (conv.hip); returns the
 // channels-last weight-grad tensor. Split-K over output pixels.
 torch::Tensor conv_nhwc_wgrad(torch::Tensor dy, torch::Tensor x, int64_t kh,
@@ -1130,6 +1173,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd", &bn_fwd, "fused NHWC BatchNorm fwd (+relu/+residual)");
   m.def("bn_bwd", &bn_bwd, "fused NHWC BatchNorm bwd");
   m.def("gemm_bf16", &gemm_bf16, "bf16 MFMA GEMM, fp32 accumulate");
+  m.def("attn_bwd", &attn_bwd, "flash-style attention bwd (d=64)",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("o"), py::arg("dout"),
+        py::arg("lse"), py::arg("causal") = false, py::arg("scale") = 0.125);
   m.def("attn_fwd", &attn_fwd, "flash-style attention fwd (d=64)",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("causal") = false,
         py::arg("scale") = 0.125, py::arg("want_lse") = true);
