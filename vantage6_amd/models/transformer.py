@@ -16,6 +16,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops.activations import rope, rope_tables, swiglu
+from ..ops.attention import sdpa as hip_sdpa
 from ..ops.embedding import Embedding as GraphSafeEmbedding
 from ..ops.gemm import Linear as MfmaLinear
 from ..ops.norms import LayerNorm, RMSNorm
@@ -37,7 +38,7 @@ class BertLayer(nn.Module):
         B, S, H = x.shape
         qkv = self.qkv(x).reshape(B, S, 3, self.heads, H // self.heads)
         q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))
-        att = F.scaled_dot_product_attention(q, k, v)
+        att = hip_sdpa(q.contiguous(), k.contiguous(), v.contiguous())
         att = att.transpose(1, 2).reshape(B, S, H)
         x = self.ln1(x + self.proj(att))
         x = self.ln2(x + self.fc2(F.gelu(self.fc1(x))))
@@ -101,7 +102,8 @@ class LlamaLayer(nn.Module):
         k = self.k(h).reshape(B, S, self.kv_heads, self.head_dim).transpose(1, 2)
         v = self.v(h).reshape(B, S, self.kv_heads, self.head_dim).transpose(1, 2)
         q, k = _rope(q, k)
-        att = F.scaled_dot_product_attention(q, k, v, is_causal=True, enable_gqa=True)
+        att = hip_sdpa(q.contiguous(), k.contiguous(), v.contiguous(),
+                       causal=True)
         att = att.transpose(1, 2).reshape(B, S, H)
         x = x + self.o(att)
         h = self.norm2(x)
