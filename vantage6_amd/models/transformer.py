@@ -38,7 +38,7 @@ class BertLayer(nn.Module):
         B, S, H = x.shape
         qkv = self.qkv(x).reshape(B, S, 3, self.heads, H // self.heads)
         q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))
-        att = hip_sdpa(q.contiguous(), k.contiguous(), v.contiguous())
+        att = hip_sdpa(q, k, v)
         att = att.transpose(1, 2).reshape(B, S, H)
         x = self.ln1(x + self.proj(att))
         x = self.ln2(x + self.fc2(F.gelu(self.fc1(x))))
@@ -102,8 +102,7 @@ class LlamaLayer(nn.Module):
         k = self.k(h).reshape(B, S, self.kv_heads, self.head_dim).transpose(1, 2)
         v = self.v(h).reshape(B, S, self.kv_heads, self.head_dim).transpose(1, 2)
         q, k = _rope(q, k)
-        att = hip_sdpa(q.contiguous(), k.contiguous(), v.contiguous(),
-                       causal=True)
+        att = hip_sdpa(q, k, v, causal=True)
         att = att.transpose(1, 2).reshape(B, S, H)
         x = x + self.o(att)
         h = self.norm2(x)

@@ -39,7 +39,8 @@ def sdpa(q, k, v, causal: bool = False):
     eligible (cuda bf16, head_dim 64, contiguous), library SDPA otherwise.
     ``V6AMD_ATTN=lib`` forces the library path."""
     if (q.is_cuda and q.dtype == torch.bfloat16 and q.shape[-1] == 64
-            and q.is_contiguous() and k.is_contiguous() and v.is_contiguous()
+            and q.stride(-1) == 1 and k.stride(-1) == 1
+            and k.stride() == v.stride()
             and os.environ.get("V6AMD_ATTN", "hip") != "lib"):
         return _AttnFn.apply(q, k, v, causal, q.shape[-1] ** -0.5)
     return F.scaled_dot_product_attention(q, k, v, is_causal=causal,

@@ -40,7 +40,8 @@ __device__ __forceinline__ void attn_fwd_d64_body(
     const short* __restrict__ Q, const short* __restrict__ K,
     const short* __restrict__ V, short* __restrict__ O,
     float* __restrict__ Lse, int S, int Skv, int nheads, int kv_heads,
-    float scale, int causal) {
+    float scale, int causal, long long qs_b, long long qs_h, long long qs_s,
+    long long ks_b, long long ks_h, long long ks_s) {
   __shared__ short lds[16384 + 8 * 1024];
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -48,10 +49,10 @@ __device__ __forceinline__ void attn_fwd_d64_body(
   const int bh = blockIdx.y;
   const int q0 = blockIdx.x * 128;
   const int b = bh / nheads, head = bh - b * nheads;
-  const int kvbh = b * kv_heads + head / (nheads / kv_heads);
-  const short* Qb = Q + (long long)bh * S * 64;
-  const short* Kb = K + (long long)kvbh * Skv * 64;
-  const short* Vb = V + (long long)kvbh * Skv * 64;
+  const int kvh = head / (nheads / kv_heads);
+  const short* Qb = Q + (long long)b * qs_b + (long long)head * qs_h;
+  const short* Kb = K + (long long)b * ks_b + (long long)kvh * ks_h;
+  const short* Vb = V + (long long)b * ks_b + (long long)kvh * ks_h;
 
   // this wave's persistent Q fragments
   abf16x8_t q_frag[2];
@@ -60,7 +61,7 @@ __device__ __forceinline__ void attn_fwd_d64_body(
     int rr = row < S ? row : S - 1;
 #pragma unroll
     for (int kkq = 0; kkq < 2; ++kkq)
-      q_frag[kkq] = *(const abf16x8_t*)(Qb + (long long)rr * 64 + kkq * 32 +
+      q_frag[kkq] = *(const abf16x8_t*)(Qb + (long long)rr * qs_s + kkq * 32 +
                                         (lane >> 4) * 8);
   }
 
@@ -84,12 +85,13 @@ __device__ __forceinline__ void attn_fwd_d64_body(
       int krow = k0 + r;
       int src_slot = slot ^ ((r >> 1) & 7);
       *(s16x8*)(AK(buf) + r * 64 + slot * 8) =
-          krow < Skv ? *(const s16x8*)(Kb + (long long)krow * 64 + src_slot * 8)
-                     : s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+          krow < Skv
+              ? *(const s16x8*)(Kb + (long long)krow * ks_s + src_slot * 8)
+              : s16x8{0, 0, 0, 0, 0, 0, 0, 0};
       short* vdst = AV(buf) + ((r >> 2) * 4 + (slot >> 1)) * 64 +
                     (r & 3) * 16 + (slot & 1) * 8;
       *(s16x8*)vdst =
-          krow < Skv ? *(const s16x8*)(Vb + (long long)krow * 64 + slot * 8)
+          krow < Skv ? *(const s16x8*)(Vb + (long long)krow * ks_s + slot * 8)
                      : s16x8{0, 0, 0, 0, 0, 0, 0, 0};
     }
   };
@@ -241,8 +243,9 @@ __device__ __forceinline__ void attn_fwd_d64_body(
     float inv = row_sum[r] > 0.f ? 1.f / row_sum[r] : 0.f;
 #pragma unroll
     for (int ni = 0; ni < 4; ++ni)
-      O[(long long)bh * S * 64 + (long long)qrow * 64 + ni * 16 +
-        (lane & 15)] = f32_to_bf16(o_acc[ni][r] * inv);
+      O[(long long)b * qs_b + (long long)head * qs_h +
+        (long long)qrow * qs_s + ni * 16 + (lane & 15)] =
+          f32_to_bf16(o_acc[ni][r] * inv);
     if (Lse && (lane & 15) == 0)
       Lse[(long long)bh * S + qrow] =
           row_max[r] * 0.69314718f + logf(row_sum[r]);
@@ -254,9 +257,11 @@ extern "C" __global__ void __launch_bounds__(512)
 attn_fwd_d64_kernel(const short* __restrict__ Q, const short* __restrict__ K,
                     const short* __restrict__ V, short* __restrict__ O,
                     float* __restrict__ Lse, int S, int Skv, int nheads,
-                    int kv_heads, float scale, int causal) {
+                    int kv_heads, float scale, int causal, long long qs_b,
+                    long long qs_h, long long qs_s, long long ks_b,
+                    long long ks_h, long long ks_s) {
   attn_fwd_d64_body<0>(Q, K, V, O, Lse, S, Skv, nheads, kv_heads, scale,
-                       causal);
+                       causal, qs_b, qs_h, qs_s, ks_b, ks_h, ks_s);
 }
 
 extern "C" __global__ void __launch_bounds__(512)
@@ -264,9 +269,11 @@ attn_fwd_d64_plin_kernel(const short* __restrict__ Q,
                          const short* __restrict__ K,
                          const short* __restrict__ V, short* __restrict__ O,
                          float* __restrict__ Lse, int S, int Skv, int nheads,
-                         int kv_heads, float scale, int causal) {
+                         int kv_heads, float scale, int causal, long long qs_b,
+                         long long qs_h, long long qs_s, long long ks_b,
+                         long long ks_h, long long ks_s) {
   attn_fwd_d64_body<1>(Q, K, V, O, Lse, S, Skv, nheads, kv_heads, scale,
-                       causal);
+                       causal, qs_b, qs_h, qs_s, ks_b, ks_h, ks_s);
 }
 
 // ===========================================================================
@@ -311,7 +318,9 @@ attn_bwd_dq_d64_kernel(const short* __restrict__ Q, const short* __restrict__ K,
                        const float* __restrict__ Lse,
                        const float* __restrict__ Dsum, short* __restrict__ dQ,
                        int S, int Skv, int nheads, int kv_heads, float scale,
-                       int causal) {
+                       int causal, long long qs_b, long long qs_h,
+                       long long qs_s, long long ks_b, long long ks_h,
+                       long long ks_s) {
   __shared__ short lds[12288 + 8 * 1024];
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -319,11 +328,11 @@ attn_bwd_dq_d64_kernel(const short* __restrict__ Q, const short* __restrict__ K,
   const int bh = blockIdx.y;
   const int q0 = blockIdx.x * 128;
   const int b = bh / nheads, head = bh - b * nheads;
-  const int kvbh = b * kv_heads + head / (nheads / kv_heads);
-  const short* Qb = Q + (long long)bh * S * 64;
-  const short* Kb = K + (long long)kvbh * Skv * 64;
-  const short* Vb = V + (long long)kvbh * Skv * 64;
-  const short* dOb = dO + (long long)bh * S * 64;
+  const int kvh = head / (nheads / kv_heads);
+  const short* Qb = Q + (long long)b * qs_b + (long long)head * qs_h;
+  const short* Kb = K + (long long)b * ks_b + (long long)kvh * ks_h;
+  const short* Vb = V + (long long)b * ks_b + (long long)kvh * ks_h;
+  const short* dOb = dO + (long long)b * qs_b + (long long)head * qs_h;
 
   abf16x8_t q_frag[2], do_frag[2];
   {
@@ -331,9 +340,9 @@ attn_bwd_dq_d64_kernel(const short* __restrict__ Q, const short* __restrict__ K,
     int rr = row < S ? row : S - 1;
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      q_frag[kk] = *(const abf16x8_t*)(Qb + (long long)rr * 64 + kk * 32 +
+      q_frag[kk] = *(const abf16x8_t*)(Qb + (long long)rr * qs_s + kk * 32 +
                                        (lane >> 4) * 8);
-      do_frag[kk] = *(const abf16x8_t*)(dOb + (long long)rr * 64 + kk * 32 +
+      do_frag[kk] = *(const abf16x8_t*)(dOb + (long long)rr * qs_s + kk * 32 +
                                         (lane >> 4) * 8);
     }
   }
@@ -364,15 +373,17 @@ attn_bwd_dq_d64_kernel(const short* __restrict__ Q, const short* __restrict__ K,
       int src_slot = slot ^ ((r >> 1) & 7);
       s16x8 zero = {0, 0, 0, 0, 0, 0, 0, 0};
       *(s16x8*)(BQ_KLIN + r * 64 + slot * 8) =
-          krow < Skv ? *(const s16x8*)(Kb + (long long)krow * 64 + src_slot * 8)
-                     : zero;
+          krow < Skv
+              ? *(const s16x8*)(Kb + (long long)krow * ks_s + src_slot * 8)
+              : zero;
       *(s16x8*)(BQ_VLIN + r * 64 + slot * 8) =
-          krow < Skv ? *(const s16x8*)(Vb + (long long)krow * 64 + src_slot * 8)
-                     : zero;
+          krow < Skv
+              ? *(const s16x8*)(Vb + (long long)krow * ks_s + src_slot * 8)
+              : zero;
       short* ktr = BQ_KTR + ((r >> 2) * 4 + (slot >> 1)) * 64 + (r & 3) * 16 +
                    (slot & 1) * 8;
       *(s16x8*)ktr =
-          krow < Skv ? *(const s16x8*)(Kb + (long long)krow * 64 + slot * 8)
+          krow < Skv ? *(const s16x8*)(Kb + (long long)krow * ks_s + slot * 8)
                      : zero;
     }
     __syncthreads();
@@ -460,8 +471,9 @@ attn_bwd_dq_d64_kernel(const short* __restrict__ Q, const short* __restrict__ K,
     if (qrow >= S) continue;
 #pragma unroll
     for (int ni = 0; ni < 4; ++ni)
-      dQ[(long long)bh * S * 64 + (long long)qrow * 64 + ni * 16 +
-         (lane & 15)] = f32_to_bf16(dq_acc[ni][r]);
+      dQ[(long long)b * qs_b + (long long)head * qs_h +
+         (long long)qrow * qs_s + ni * 16 + (lane & 15)] =
+          f32_to_bf16(dq_acc[ni][r]);
   }
 }
 
@@ -486,7 +498,9 @@ attn_bwd_dkdv_d64_kernel(const short* __restrict__ Q,
                          const float* __restrict__ Dsum,
                          short* __restrict__ dK, short* __restrict__ dV,
                          int S, int Skv, int nheads, int kv_heads, float scale,
-                         int causal) {
+                         int causal, long long qs_b, long long qs_h,
+                         long long qs_s, long long ks_b, long long ks_h,
+                         long long ks_s) {
   __shared__ short lds[16384 + 16 * 2048];
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -495,8 +509,8 @@ attn_bwd_dkdv_d64_kernel(const short* __restrict__ Q,
   const int kv0 = blockIdx.x * 128;
   const int b = bkv / kv_heads, kvh = bkv - b * kv_heads;
   const int group = nheads / kv_heads;
-  const short* Kb = K + (long long)bkv * Skv * 64;
-  const short* Vb = V + (long long)bkv * Skv * 64;
+  const short* Kb = K + (long long)b * ks_b + (long long)kvh * ks_h;
+  const short* Vb = V + (long long)b * ks_b + (long long)kvh * ks_h;
 
   // persistent K/V fragments for this wave's 16 kv rows
   abf16x8_t k_frag[2], v_frag[2];
@@ -505,9 +519,9 @@ attn_bwd_dkdv_d64_kernel(const short* __restrict__ Q,
     int rr = row < Skv ? row : Skv - 1;
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      k_frag[kk] = *(const abf16x8_t*)(Kb + (long long)rr * 64 + kk * 32 +
+      k_frag[kk] = *(const abf16x8_t*)(Kb + (long long)rr * ks_s + kk * 32 +
                                        (lane >> 4) * 8);
-      v_frag[kk] = *(const abf16x8_t*)(Vb + (long long)rr * 64 + kk * 32 +
+      v_frag[kk] = *(const abf16x8_t*)(Vb + (long long)rr * ks_s + kk * 32 +
                                        (lane >> 4) * 8);
     }
   }
@@ -523,8 +537,8 @@ attn_bwd_dkdv_d64_kernel(const short* __restrict__ Q,
 
   for (int gh = 0; gh < group; ++gh) {
     const int bh = b * nheads + kvh * group + gh;
-    const short* Qb = Q + (long long)bh * S * 64;
-    const short* dOb = dO + (long long)bh * S * 64;
+    const short* Qb = Q + (long long)b * qs_b + (long long)(kvh * group + gh) * qs_h;
+    const short* dOb = dO + (long long)b * qs_b + (long long)(kvh * group + gh) * qs_h;
     for (int qt = qt0; qt < nqt; ++qt) {
       const int q0 = qt * 64;
       {  // stage Q and dO (linear-swizzled + tr images)
@@ -534,21 +548,21 @@ attn_bwd_dkdv_d64_kernel(const short* __restrict__ Q,
         s16x8 zero = {0, 0, 0, 0, 0, 0, 0, 0};
         *(s16x8*)(BK_QLIN + r * 64 + slot * 8) =
             qrow < S
-                ? *(const s16x8*)(Qb + (long long)qrow * 64 + src_slot * 8)
+                ? *(const s16x8*)(Qb + (long long)qrow * qs_s + src_slot * 8)
                 : zero;
         *(s16x8*)(BK_DOLIN + r * 64 + slot * 8) =
             qrow < S
-                ? *(const s16x8*)(dOb + (long long)qrow * 64 + src_slot * 8)
+                ? *(const s16x8*)(dOb + (long long)qrow * qs_s + src_slot * 8)
                 : zero;
         short* qtr = BK_QTR + ((r >> 2) * 4 + (slot >> 1)) * 64 +
                      (r & 3) * 16 + (slot & 1) * 8;
         *(s16x8*)qtr =
-            qrow < S ? *(const s16x8*)(Qb + (long long)qrow * 64 + slot * 8)
+            qrow < S ? *(const s16x8*)(Qb + (long long)qrow * qs_s + slot * 8)
                      : zero;
         short* dotr = BK_DOTR + ((r >> 2) * 4 + (slot >> 1)) * 64 +
                       (r & 3) * 16 + (slot & 1) * 8;
         *(s16x8*)dotr =
-            qrow < S ? *(const s16x8*)(dOb + (long long)qrow * 64 + slot * 8)
+            qrow < S ? *(const s16x8*)(dOb + (long long)qrow * qs_s + slot * 8)
                      : zero;
       }
       __syncthreads();
@@ -674,10 +688,12 @@ attn_bwd_dkdv_d64_kernel(const short* __restrict__ Q,
     if (krow >= Skv) continue;
 #pragma unroll
     for (int ni = 0; ni < 4; ++ni) {
-      dK[(long long)bkv * Skv * 64 + (long long)krow * 64 + ni * 16 +
-         (lane & 15)] = f32_to_bf16(dk_acc[ni][r]);
-      dV[(long long)bkv * Skv * 64 + (long long)krow * 64 + ni * 16 +
-         (lane & 15)] = f32_to_bf16(dv_acc[ni][r]);
+      dK[(long long)b * ks_b + (long long)kvh * ks_h +
+         (long long)krow * ks_s + ni * 16 + (lane & 15)] =
+          f32_to_bf16(dk_acc[ni][r]);
+      dV[(long long)b * ks_b + (long long)kvh * ks_h +
+         (long long)krow * ks_s + ni * 16 + (lane & 15)] =
+          f32_to_bf16(dv_acc[ni][r]);
     }
   }
 }

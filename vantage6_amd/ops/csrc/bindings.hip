@@ -111,22 +111,22 @@ extern "C" __global__ void conv_flip_weights_kernel(const short*, short*, int,
                                                     int, int, int);
 extern "C" __global__ void attn_fwd_d64_kernel(const short*, const short*,
                                                const short*, short*, float*,
-                                               int, int, int, int, float,
-                                               int);
-extern "C" __global__ void attn_fwd_d64_plin_kernel(const short*, const short*,
-                                                    const short*, short*,
-                                                    float*, int, int, int, int,
-                                                    float, int);
-extern "C" __global__ void attn_bwd_d_kernel(const short*, const short*,
-                                             float*, long long);
-extern "C" __global__ void attn_bwd_dq_d64_kernel(const short*, const short*,
-                                                  const short*, const short*,
-                                                  const float*, const float*,
-                                                  short*, int, int, int, int,
-                                                  float, int);
+                                               int, int, int, int, float, int,
+                                               long long, long long, long long,
+                                               long long, long long,
+                                               long long);
+extern "C" __global__ void attn_fwd_d64_plin_kernel(
+    const short*, const short*, const short*, short*, float*, int, int, int,
+    int, float, int, long long, long long, long long, long long, long long,
+    long long);
+extern "C" __global__ void attn_bwd_dq_d64_kernel(
+    const short*, const short*, const short*, const short*, const float*,
+    const float*, short*, int, int, int, int, float, int, long long,
+    long long, long long, long long, long long, long long);
 extern "C" __global__ void attn_bwd_dkdv_d64_kernel(
     const short*, const short*, const short*, const short*, const float*,
-    const float*, short*, short*, int, int, int, int, float, int);
+    const float*, short*, short*, int, int, int, int, float, int, long long,
+    long long, long long, long long, long long, long long);
 extern "C" __global__ void conv_nhwc_wgrad_kernel(
     const short*, const short*, float*, short*, const short*, int, int, int,
     int, int, int, int, int, int, int, int, int, int, unsigned long long,
@@ -865,8 +865,9 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     double scale, bool want_lse) {
   for (auto& t : {q, k, v}) {
     TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kBFloat16 &&
-                t.is_contiguous(), "attn_fwd: contiguous bf16 cuda");
+                t.stride(3) == 1, "attn_fwd: bf16 cuda, head-dim contiguous");
   }
+  TORCH_CHECK(k.strides() == v.strides(), "attn_fwd: k/v layouts differ");
   int B = (int)q.size(0), H = (int)q.size(1), S = (int)q.size(2);
   int Hkv = (int)k.size(1), Skv = (int)k.size(2);
   TORCH_CHECK(q.size(3) == 64 && k.size(3) == 64 && v.size(3) == 64,
@@ -890,7 +891,9 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   hipLaunchKernelGGL(kfn, grid, dim3(512), 0, cur_stream(),
                      (const short*)q.data_ptr(), (const short*)k.data_ptr(),
                      (const short*)v.data_ptr(), (short*)o.data_ptr(), lse_ptr,
-                     S, Skv, H, Hkv, (float)scale, causal ? 1 : 0);
+                     S, Skv, H, Hkv, (float)scale, causal ? 1 : 0,
+                     q.stride(0), q.stride(1), q.stride(2), k.stride(0),
+                     k.stride(1), k.stride(2));
   return {o, lse};
 }
 
@@ -906,16 +909,23 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   auto dv = torch::empty_like(v);
   auto dsum = torch::empty({B, H, S}, q.options().dtype(torch::kFloat32));
   long long nrows = (long long)B * H * S;
-  auto dco = dout.contiguous();
-  hipLaunchKernelGGL(attn_bwd_d_kernel, dim3((nrows + 255) / 256), dim3(256),
-                     0, cur_stream(), (const short*)dco.data_ptr(),
-                     (const short*)o.data_ptr(), dsum.data_ptr<float>(), nrows);
+  // D = rowsum(dO ∘ O): torch handles arbitrary strides; tiny tensor
+  dsum.copy_((dout.to(torch::kFloat32) * o.to(torch::kFloat32)).sum(-1));
+  // the kernels take ONE stride set for the q-side tensors (q, dO, dq):
+  // relayout dO into q's layout when it differs (one copy, rare)
+  torch::Tensor dco = dout;
+  if (dout.strides() != q.strides()) {
+    dco = torch::empty_like(q);
+    dco.copy_(dout);
+  }
   hipLaunchKernelGGL(attn_bwd_dq_d64_kernel, dim3((S + 127) / 128, B * H),
                      dim3(512), 0, cur_stream(), (const short*)q.data_ptr(),
                      (const short*)k.data_ptr(), (const short*)v.data_ptr(),
                      (const short*)dco.data_ptr(), lse.data_ptr<float>(),
                      dsum.data_ptr<float>(), (short*)dq.data_ptr(), S, Skv, H,
-                     Hkv, (float)scale, causal ? 1 : 0);
+                     Hkv, (float)scale, causal ? 1 : 0, q.stride(0),
+                     q.stride(1), q.stride(2), k.stride(0), k.stride(1),
+                     k.stride(2));
   hipLaunchKernelGGL(attn_bwd_dkdv_d64_kernel,
                      dim3((Skv + 127) / 128, B * Hkv), dim3(512), 0,
                      cur_stream(), (const short*)q.data_ptr(),
@@ -923,7 +933,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                      (const short*)dco.data_ptr(), lse.data_ptr<float>(),
                      dsum.data_ptr<float>(), (short*)dk.data_ptr(),
                      (short*)dv.data_ptr(), S, Skv, H, Hkv, (float)scale,
-                     causal ? 1 : 0);
+                     causal ? 1 : 0, q.stride(0), q.stride(1), q.stride(2),
+                     k.stride(0), k.stride(1), k.stride(2));
   return {dq, dk, dv};
 }
 
