@@ -284,14 +284,9 @@ fused_adamw_mt_kernel(const SgdChunk* __restrict__ table, int n_chunks,
       long long fi = ch.offset + i;
       if (i + 8 <= ch.n) {
         s16x8 g8 = *(const s16x8*)(ch.grad + i);
-        // optimizer state is touched once per step (next use a whole round
-        // away): nontemporal keeps ~18 GB/step of it from thrashing L2/MALL
-        f32x4 m0 = __builtin_nontemporal_load((const f32x4*)(master + fi));
-        f32x4 m1 = __builtin_nontemporal_load((const f32x4*)(master + fi + 4));
-        f32x4 a0 = __builtin_nontemporal_load((const f32x4*)(exp_avg + fi));
-        f32x4 a1 = __builtin_nontemporal_load((const f32x4*)(exp_avg + fi + 4));
-        f32x4 s0 = __builtin_nontemporal_load((const f32x4*)(exp_avg_sq + fi));
-        f32x4 s1 = __builtin_nontemporal_load((const f32x4*)(exp_avg_sq + fi + 4));
+        f32x4 m0 = *(const f32x4*)(master + fi), m1 = *(const f32x4*)(master + fi + 4);
+        f32x4 a0 = *(const f32x4*)(exp_avg + fi), a1 = *(const f32x4*)(exp_avg + fi + 4);
+        f32x4 s0 = *(const f32x4*)(exp_avg_sq + fi), s1 = *(const f32x4*)(exp_avg_sq + fi + 4);
         s16x8 p8;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
@@ -308,13 +303,10 @@ fused_adamw_mt_kernel(const SgdChunk* __restrict__ table, int n_chunks,
           else { m1[j - 4] = m; a1[j - 4] = a; s1[j - 4] = s; }
           p8[j] = f32_to_bf16(m);
         }
-        __builtin_nontemporal_store(m0, (f32x4*)(master + fi));
-        __builtin_nontemporal_store(m1, (f32x4*)(master + fi + 4));
-        __builtin_nontemporal_store(a0, (f32x4*)(exp_avg + fi));
-        __builtin_nontemporal_store(a1, (f32x4*)(exp_avg + fi + 4));
-        __builtin_nontemporal_store(s0, (f32x4*)(exp_avg_sq + fi));
-        __builtin_nontemporal_store(s1, (f32x4*)(exp_avg_sq + fi + 4));
-        *(s16x8*)(param + fi) = p8;  // params ARE re-read next forward
+        *(f32x4*)(master + fi) = m0; *(f32x4*)(master + fi + 4) = m1;
+        *(f32x4*)(exp_avg + fi) = a0; *(f32x4*)(exp_avg + fi + 4) = a1;
+        *(f32x4*)(exp_avg_sq + fi) = s0; *(f32x4*)(exp_avg_sq + fi + 4) = s1;
+        *(s16x8*)(param + fi) = p8;
       } else {
         for (long long k = i; k < ch.n; ++k) {
           float g = bf16_to_f32(ch.grad[k]);
