@@ -40,8 +40,17 @@ def sdpa(q, k, v, causal: bool = False):
     ``V6AMD_ATTN=lib`` forces the library path."""
     if (q.is_cuda and q.dtype == torch.bfloat16 and q.shape[-1] == 64
             and q.stride(-1) == 1 and k.stride(-1) == 1
-            and k.stride() == v.stride()
             and os.environ.get("V6AMD_ATTN", "hip") != "lib"):
+        if k.stride() != v.stride():
+            # one stride set serves both k and v in-kernel; relayout v to
+            # k's layout (tiny for GQA kv tensors; hits when RoPE returned
+            # a contiguous k while v stayed a transposed view)
+            if k.is_contiguous():
+                v = v.contiguous()
+            else:
+                vk = torch.empty_like(k)
+                vk.copy_(v)
+                v = vk
         return _AttnFn.apply(q, k, v, causal, q.shape[-1] ** -0.5)
     return F.scaled_dot_product_attention(q, k, v, is_causal=causal,
                                           enable_gqa=True)
