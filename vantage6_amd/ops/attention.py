@@ -35,12 +35,19 @@ class _AttnFn(torch.autograd.Function):
 
 
 def sdpa(q, k, v, causal: bool = False):
-    """Drop-in scaled-dot-product attention: the hand-written kernel when
-    eligible (cuda bf16, head_dim 64, contiguous), library SDPA otherwise.
-    ``V6AMD_ATTN=lib`` forces the library path."""
-    if (q.is_cuda and q.dtype == torch.bfloat16 and q.shape[-1] == 64
-            and q.stride(-1) == 1 and k.stride(-1) == 1
-            and os.environ.get("V6AMD_ATTN", "hip") != "lib"):
+    """Drop-in scaled-dot-product attention.
+
+    Measured dispatch (tools/attn_ab.py + end-to-end rounds, r02): the
+    hand-written kernel wins the BERT shape (16.81 vs 16.69 rounds/s) and
+    loses the causal-GQA Llama shape end-to-end (8.02 vs 8.30 — the dk/dv
+    kernel serializes the 4-head GQA group per kv block), so `auto` uses
+    hip except for causal GQA. ``V6AMD_ATTN=hip|lib`` forces a path."""
+    mode = os.environ.get("V6AMD_ATTN", "auto")
+    gqa = k.shape[1] != q.shape[1]
+    use_hip = mode == "hip" or (mode == "auto" and not (causal and gqa))
+    if (use_hip and q.is_cuda and q.dtype == torch.bfloat16
+            and q.shape[-1] == 64
+            and q.stride(-1) == 1 and k.stride(-1) == 1):
         if k.stride() != v.stride():
             # one stride set serves both k and v in-kernel; relayout v to
             # k's layout (tiny for GQA kv tensors; hits when RoPE returned
