@@ -125,8 +125,8 @@ extern "C" __global__ void attn_bwd_dq_d64_kernel(
     long long, long long, long long, long long, long long);
 extern "C" __global__ void attn_bwd_dkdv_d64_kernel(
     const short*, const short*, const short*, const short*, const float*,
-    const float*, short*, short*, int, int, int, int, float, int, long long,
-    long long, long long, long long, long long, long long);
+    const float*, short*, short*, float*, float*, int, int, int, int, float,
+    int, long long, long long, long long, long long, long long, long long);
 extern "C" __global__ void conv_nhwc_wgrad_kernel(
     const short*, const short*, float*, short*, const short*, int, int, int,
     int, int, int, int, int, int, int, int, int, int, unsigned long long,
@@ -926,15 +926,43 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                      Hkv, (float)scale, causal ? 1 : 0, q.stride(0),
                      q.stride(1), q.stride(2), k.stride(0), k.stride(1),
                      k.stride(2));
+  int group = H / Hkv;
+  if (group == 1) {
+    hipLaunchKernelGGL(attn_bwd_dkdv_d64_kernel,
+                       dim3((Skv + 127) / 128, B * Hkv, 1), dim3(512), 0,
+                       cur_stream(), (const short*)q.data_ptr(),
+                       (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                       (const short*)dco.data_ptr(), lse.data_ptr<float>(),
+                       dsum.data_ptr<float>(), (short*)dk.data_ptr(),
+                       (short*)dv.data_ptr(), nullptr, nullptr, S, Skv, H,
+                       Hkv, (float)scale, causal ? 1 : 0, q.stride(0),
+                       q.stride(1), q.stride(2), k.stride(0), k.stride(1),
+                       k.stride(2));
+    return {dq, dk, dv};
+  }
+  // GQA: one z-slice per group q-head, fp32 slabs, then a reduce — the
+  // slabs (and dk/dv) use the contiguous [B,Hkv,Skv,64] layout
+  dk = torch::empty({B, Hkv, Skv, 64}, k.options());
+  dv = torch::empty({B, Hkv, Skv, 64}, v.options());
+  long long nkv = (long long)B * Hkv * Skv * 64;
+  auto wsk = torch::empty({group, nkv}, q.options().dtype(torch::kFloat32));
+  auto wsv = torch::empty({group, nkv}, q.options().dtype(torch::kFloat32));
   hipLaunchKernelGGL(attn_bwd_dkdv_d64_kernel,
-                     dim3((Skv + 127) / 128, B * Hkv), dim3(512), 0,
+                     dim3((Skv + 127) / 128, B * Hkv, group), dim3(512), 0,
                      cur_stream(), (const short*)q.data_ptr(),
                      (const short*)k.data_ptr(), (const short*)v.data_ptr(),
                      (const short*)dco.data_ptr(), lse.data_ptr<float>(),
-                     dsum.data_ptr<float>(), (short*)dk.data_ptr(),
-                     (short*)dv.data_ptr(), S, Skv, H, Hkv, (float)scale,
-                     causal ? 1 : 0, q.stride(0), q.stride(1), q.stride(2),
-                     k.stride(0), k.stride(1), k.stride(2));
+                     dsum.data_ptr<float>(), nullptr, nullptr,
+                     wsk.data_ptr<float>(), wsv.data_ptr<float>(), S, Skv, H,
+                     Hkv, (float)scale, causal ? 1 : 0, q.stride(0),
+                     q.stride(1), q.stride(2), k.stride(0), k.stride(1),
+                     k.stride(2));
+  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid(nkv)), dim3(256), 0,
+                     cur_stream(), wsk.data_ptr<float>(),
+                     (short*)dk.data_ptr(), nkv, group);
+  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid(nkv)), dim3(256), 0,
+                     cur_stream(), wsv.data_ptr<float>(),
+                     (short*)dv.data_ptr(), nkv, group);
   return {dq, dk, dv};
 }
 
