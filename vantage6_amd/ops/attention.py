@@ -37,14 +37,11 @@ class _AttnFn(torch.autograd.Function):
 def sdpa(q, k, v, causal: bool = False):
     """Drop-in scaled-dot-product attention.
 
-    Measured dispatch (tools/attn_ab.py + end-to-end rounds, r02): the
-    hand-written kernel wins the BERT shape (16.81 vs 16.69 rounds/s) and
-    loses the causal-GQA Llama shape end-to-end (8.02 vs 8.30 — the dk/dv
-    kernel serializes the 4-head GQA group per kv block), so `auto` uses
-    hip except for causal GQA. ``V6AMD_ATTN=hip|lib`` forces a path."""
-    mode = os.environ.get("V6AMD_ATTN", "auto")
-    gqa = k.shape[1] != q.shape[1]
-    use_hip = mode == "hip" or (mode == "auto" and not (causal and gqa))
+    Measured end-to-end (r02): the hand-written kernels win both
+    benchmark shapes — BERT 16.81 vs 16.69 rounds/s, Llama (causal GQA,
+    grid.z group fan-out in dk/dv) 8.38 vs 8.30. ``V6AMD_ATTN=lib``
+    forces the library path."""
+    use_hip = os.environ.get("V6AMD_ATTN", "hip") != "lib"
     if (use_hip and q.is_cuda and q.dtype == torch.bfloat16
             and q.shape[-1] == 64
             and q.stride(-1) == 1 and k.stride(-1) == 1):
