@@ -51,3 +51,32 @@ def test_attention_fwd_bwd_matches_fp32(B, H, Hkv, S, causal):
     check(qa.grad, qr.grad, "dq", 0.05)
     check(ka.grad, kr.grad, "dk", 0.05)
     check(va.grad, vr.grad, "dv", 0.05)
+
+
+@pytest.mark.timeout(300)
+def test_attention_fused_qkv_slice_views():
+    """q/k/v carved from ONE fused qkv projection (BERT pattern) are
+    non-dense slice views — the wrapper must densify them; regression for
+    an out-of-bounds write caught by rocprof."""
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    from vantage6_amd.ops.attention import sdpa
+
+    torch.manual_seed(3)
+    B, S, H, hd = 4, 256, 8, 64
+    qkv = (torch.randn(B, S, 3, H, hd, device="cuda", dtype=torch.bfloat16)
+           .requires_grad_(True))
+    q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))
+    out = sdpa(q, k, v)
+    ref = F.scaled_dot_product_attention(q.float(), k.float(), v.float())
+    scale = ref.abs().max().item()
+    assert (out.float() - ref).abs().max().item() <= 0.02 * scale + 0.02
+    g = torch.randn_like(out)
+    out.backward(g)
+    # fp32 reference grads through the same views
+    qkv2 = qkv.detach().float().requires_grad_(True)
+    q2, k2, v2 = (qkv2[:, :, i].transpose(1, 2) for i in range(3))
+    F.scaled_dot_product_attention(q2, k2, v2).backward(g.float())
+    gs = qkv2.grad.abs().max().item()
+    err = (qkv.grad.float() - qkv2.grad).abs().max().item()
+    assert err <= 0.05 * gs + 0.05, (err, gs)

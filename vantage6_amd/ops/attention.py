@@ -18,6 +18,20 @@ import torch.nn.functional as F
 from . import hip_ops
 
 
+def _is_dense(t: torch.Tensor) -> bool:
+    """Non-overlapping dense (empty_like would preserve strides). Slice
+    views (e.g. q/k/v carved out of one fused qkv tensor) are NOT dense:
+    empty_like silently compacts them, so a kernel writing with the view's
+    strides would write out of bounds."""
+    dims = sorted((st, sz) for sz, st in zip(t.shape, t.stride()) if sz != 1)
+    expect = 1
+    for st, sz in dims:
+        if st != expect:
+            return False
+        expect *= sz
+    return True
+
+
 class _AttnFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, causal, scale):

@@ -874,6 +874,9 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
               "attn_fwd: head_dim must be 64");
   TORCH_CHECK(H % Hkv == 0, "attn_fwd: H % Hkv");
   auto o = torch::empty_like(q);
+  TORCH_CHECK(o.strides() == q.strides(),
+              "attn_fwd: q must be non-overlapping dense (slice views of a "
+              "fused tensor are not — densify in the caller)");
   torch::Tensor lse;
   float* lse_ptr = nullptr;
   if (want_lse) {
@@ -907,6 +910,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   auto dq = torch::empty_like(q);
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
+  TORCH_CHECK(dq.strides() == q.strides() && dk.strides() == k.strides(),
+              "attn_bwd: q/k must be non-overlapping dense");
   auto dsum = torch::empty({B, H, S}, q.options().dtype(torch::kFloat32));
   long long nrows = (long long)B * H * S;
   // D = rowsum(dO ∘ O): torch handles arbitrary strides; tiny tensor
