@@ -32,6 +32,18 @@ def _is_dense(t: torch.Tensor) -> bool:
     return True
 
 
+def _is_dense(t: torch.Tensor) -> bool:
+    """Non-overlapping dense: empty_like preserves these strides. Slice
+    views of a bigger tensor are NOT dense."""
+    dims = sorted((st, sz) for sz, st in zip(t.shape, t.stride()) if sz != 1)
+    expect = 1
+    for st, sz in dims:
+        if st != expect:
+            return False
+        expect *= sz
+    return True
+
+
 class _AttnFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, causal, scale):
@@ -59,7 +71,15 @@ def sdpa(q, k, v, causal: bool = False):
     if (use_hip and q.is_cuda and q.dtype == torch.bfloat16
             and q.shape[-1] == 64
             and q.stride(-1) == 1 and k.stride(-1) == 1):
-        if k.stride() != v.stride():
+        # non-dense SLICE views (q/k/v carved from a fused qkv tensor)
+        # must be densified: empty_like silently compacts them, so kernels
+        # writing outputs/grads with the view's strides would write out of
+        # bounds (found via a rocprof "write to read-only page" fault)
+        if not _is_dense(q):
+            q = q.contiguous()
+        if not _is_dense(k):
+            k = k.contiguous()
+        if not _is_dense(v) or k.stride() != v.stride():
             # one stride set serves both k and v in-kernel; relayout v to
             # k's layout (tiny for GQA kv tensors; hits when RoPE returned
             # a contiguous k while v stayed a transposed view)
