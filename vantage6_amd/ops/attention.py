@@ -71,24 +71,18 @@ def sdpa(q, k, v, causal: bool = False):
     if (use_hip and q.is_cuda and q.dtype == torch.bfloat16
             and q.shape[-1] == 64
             and q.stride(-1) == 1 and k.stride(-1) == 1):
-        # non-dense SLICE views (q/k/v carved from a fused qkv tensor)
-        # must be densified: empty_like silently compacts them, so kernels
-        # writing outputs/grads with the view's strides would write out of
-        # bounds (found via a rocprof "write to read-only page" fault)
-        if not _is_dense(q):
-            q = q.contiguous()
-        if not _is_dense(k):
-            k = k.contiguous()
-        if not _is_dense(v) or k.stride() != v.stride():
+        # Non-dense SLICE views (q/k/v carved from a fused qkv tensor) are
+        # fine: outputs/grads are allocated with empty_strided spanning the
+        # view's full extent (an earlier empty_like allocation compacted
+        # them and wrote out of bounds — caught via a rocprof fault).
+        if k.stride() != v.stride():
             # one stride set serves both k and v in-kernel; relayout v to
             # k's layout (tiny for GQA kv tensors; hits when RoPE returned
             # a contiguous k while v stayed a transposed view)
-            if k.is_contiguous():
-                v = v.contiguous()
-            else:
-                vk = torch.empty_like(k)
-                vk.copy_(v)
-                v = vk
+            vk = torch.empty_strided(k.shape, k.stride(), dtype=k.dtype,
+                                     device=k.device)
+            vk.copy_(v)
+            v = vk
         return _AttnFn.apply(q, k, v, causal, q.shape[-1] ** -0.5)
     return F.scaled_dot_product_attention(q, k, v, is_causal=causal,
                                           enable_gqa=True)

@@ -873,10 +873,10 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   TORCH_CHECK(q.size(3) == 64 && k.size(3) == 64 && v.size(3) == 64,
               "attn_fwd: head_dim must be 64");
   TORCH_CHECK(H % Hkv == 0, "attn_fwd: H % Hkv");
-  auto o = torch::empty_like(q);
-  TORCH_CHECK(o.strides() == q.strides(),
-              "attn_fwd: q must be non-overlapping dense (slice views of a "
-              "fused tensor are not — densify in the caller)");
+  // match q's exact layout (slice views of a fused qkv tensor included):
+  // empty_strided allocates the full addressed span, so strided kernel
+  // writes stay in bounds; the gaps are never read back
+  auto o = torch::empty_strided(q.sizes(), q.strides(), q.options());
   torch::Tensor lse;
   float* lse_ptr = nullptr;
   if (want_lse) {
@@ -907,11 +907,9 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                                     bool causal, double scale) {
   int B = (int)q.size(0), H = (int)q.size(1), S = (int)q.size(2);
   int Hkv = (int)k.size(1), Skv = (int)k.size(2);
-  auto dq = torch::empty_like(q);
-  auto dk = torch::empty_like(k);
-  auto dv = torch::empty_like(v);
-  TORCH_CHECK(dq.strides() == q.strides() && dk.strides() == k.strides(),
-              "attn_bwd: q/k must be non-overlapping dense");
+  auto dq = torch::empty_strided(q.sizes(), q.strides(), q.options());
+  auto dk = torch::empty_strided(k.sizes(), k.strides(), k.options());
+  auto dv = torch::empty_strided(v.sizes(), v.strides(), v.options());
   auto dsum = torch::empty({B, H, S}, q.options().dtype(torch::kFloat32));
   long long nrows = (long long)B * H * S;
   // D = rowsum(dO ∘ O): torch handles arbitrary strides; tiny tensor
@@ -920,7 +918,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   // relayout dO into q's layout when it differs (one copy, rare)
   torch::Tensor dco = dout;
   if (dout.strides() != q.strides()) {
-    dco = torch::empty_like(q);
+    dco = torch::empty_strided(q.sizes(), q.strides(), q.options());
     dco.copy_(dout);
   }
   hipLaunchKernelGGL(attn_bwd_dq_d64_kernel, dim3((S + 127) / 128, B * H),
