@@ -63,12 +63,16 @@ class _AttnFn(torch.autograd.Function):
 def sdpa(q, k, v, causal: bool = False):
     """Drop-in scaled-dot-product attention.
 
-    Measured end-to-end (r02): the hand-written kernels win both
-    benchmark shapes — BERT 16.81 vs 16.69 rounds/s, Llama (causal GQA,
-    grid.z group fan-out in dk/dv) 8.38 vs 8.30. ``V6AMD_ATTN=lib``
-    forces the library path."""
-    use_hip = os.environ.get("V6AMD_ATTN", "hip") != "lib"
-    if (use_hip and q.is_cuda and q.dtype == torch.bfloat16
+    Measured end-to-end (r02, same-box interleaved runs): the
+    hand-written kernels win the dense-layout Llama shape (8.43 vs 8.30
+    rounds/s, causal GQA with grid.z group fan-out in dk/dv) and lose the
+    BERT fused-qkv SLICE-view shape (15.73 vs 16.77 — the 3x-span strided
+    grad writes cost more than the kernel gains), so `auto` dispatches on
+    input density. ``V6AMD_ATTN=hip|lib`` forces a path."""
+    mode = os.environ.get("V6AMD_ATTN", "auto")
+    use_hip = mode == "hip" or (mode == "auto" and _is_dense(q)
+                                and _is_dense(k))
+    if (use_hip and mode != "lib" and q.is_cuda and q.dtype == torch.bfloat16
             and q.shape[-1] == 64
             and q.stride(-1) == 1 and k.stride(-1) == 1):
         # Non-dense SLICE views (q/k/v carved from a fused qkv tensor) are
