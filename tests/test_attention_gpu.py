@@ -60,6 +60,20 @@ def test_attention_fused_qkv_slice_views():
     an out-of-bounds write caught by rocprof."""
     if not torch.cuda.is_available():
         pytest.skip("needs a GPU")
+    import os
+
+    from vantage6_amd.ops.attention import sdpa
+
+    os.environ["V6AMD_ATTN"] = "hip"  # auto would route slice views to lib
+    try:
+        _run_qkv_slice_check()
+    finally:
+        os.environ.pop("V6AMD_ATTN", None)
+
+
+def _run_qkv_slice_check():
+    import torch.nn.functional as F
+
     from vantage6_amd.ops.attention import sdpa
 
     torch.manual_seed(3)
