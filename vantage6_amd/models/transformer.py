@@ -16,7 +16,11 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops.activations import rope, rope_tables, swiglu
+import os as _os
+
 from ..ops.attention import sdpa as hip_sdpa
+
+_DENSE_QKV = _os.environ.get("V6AMD_BERT_DENSE_QKV", "0") == "1"
 from ..ops.embedding import Embedding as GraphSafeEmbedding
 from ..ops.gemm import Linear as MfmaLinear
 from ..ops.norms import LayerNorm, RMSNorm
@@ -37,7 +41,12 @@ class BertLayer(nn.Module):
     def forward(self, x):
         B, S, H = x.shape
         qkv = self.qkv(x).reshape(B, S, 3, self.heads, H // self.heads)
-        q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))
+        if _DENSE_QKV:
+            # one copy makes q/k/v dense so the hand-written attention's
+            # strided writes stay compact (A/B experiment, see NOTES)
+            q, k, v = qkv.permute(2, 0, 3, 1, 4).contiguous().unbind(0)
+        else:
+            q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))
         att = hip_sdpa(q, k, v)
         att = att.transpose(1, 2).reshape(B, S, H)
         x = self.ln1(x + self.proj(att))
