@@ -291,24 +291,39 @@ layernorm_bwd_wave_kernel(const short* __restrict__ dy,
   float db_acc[WMAXCH * 8];
 #pragma unroll
   for (int c = 0; c < WMAXCH * 8; ++c) { dw_acc[c] = 0.f; db_acc[c] = 0.f; }
+  // w is row-invariant: load it ONCE; x/dy are register-cached across the
+  // two passes (the re-reads + serial load→reduce→load chain made the
+  // original version latency-bound at ~0.5 TB/s)
+  s16x8 wreg[WMAXCH];
+#pragma unroll
+  for (int c = 0; c < WMAXCH; ++c) {
+    int i = lane * 8 + c * 512;
+    wreg[c] = i < H ? *(const s16x8*)(w + i) : s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+  }
   for (long long row = (long long)blockIdx.x * 4 + wave; row < nrows;
        row += (long long)gridDim.x * 4) {
     const short* xr = x + row * (long long)H;
     const short* dyr = dy + row * (long long)H;
     short* dxr = dx + row * (long long)H;
     float mu = mean[row], rs = rstd[row];
+    s16x8 xv[WMAXCH], dv[WMAXCH];
+#pragma unroll
+    for (int c = 0; c < WMAXCH; ++c) {  // all loads in flight together
+      int i = lane * 8 + c * 512;
+      if (i < H) {
+        xv[c] = *(const s16x8*)(xr + i);
+        dv[c] = *(const s16x8*)(dyr + i);
+      }
+    }
     float c1 = 0.f, c2 = 0.f;
 #pragma unroll WMAXCH
     for (int c = 0; c < WMAXCH; ++c) {
       int i = lane * 8 + c * 512;
       if (i >= H) break;
-      s16x8 xv = *(const s16x8*)(xr + i);
-      s16x8 dv = *(const s16x8*)(dyr + i);
-      s16x8 wv = *(const s16x8*)(w + i);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float xh = (bf16_to_f32(xv[j]) - mu) * rs;
-        float wdy = bf16_to_f32(wv[j]) * bf16_to_f32(dv[j]);
+        float xh = (bf16_to_f32(xv[c][j]) - mu) * rs;
+        float wdy = bf16_to_f32(wreg[c][j]) * bf16_to_f32(dv[c][j]);
         c1 += wdy;
         c2 = fmaf(wdy, xh, c2);
       }
@@ -319,15 +334,12 @@ layernorm_bwd_wave_kernel(const short* __restrict__ dy,
     for (int c = 0; c < WMAXCH; ++c) {
       int i = lane * 8 + c * 512;
       if (i >= H) break;
-      s16x8 xv = *(const s16x8*)(xr + i);
-      s16x8 dv = *(const s16x8*)(dyr + i);
-      s16x8 wv = *(const s16x8*)(w + i);
       s16x8 o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float xh = (bf16_to_f32(xv[j]) - mu) * rs;
-        float d = bf16_to_f32(dv[j]);
-        float wdy = bf16_to_f32(wv[j]) * d;
+        float xh = (bf16_to_f32(xv[c][j]) - mu) * rs;
+        float d = bf16_to_f32(dv[c][j]);
+        float wdy = bf16_to_f32(wreg[c][j]) * d;
         o[j] = f32_to_bf16((wdy - c1 - xh * c2) * rs);
         dw_acc[c * 8 + j] = fmaf(d, xh, dw_acc[c * 8 + j]);
         db_acc[c * 8 + j] += d;
@@ -393,24 +405,36 @@ rmsnorm_bwd_wave_kernel(const short* __restrict__ dy,
   float dw_acc[WMAXCH * 8];
 #pragma unroll
   for (int c = 0; c < WMAXCH * 8; ++c) dw_acc[c] = 0.f;
+  s16x8 wreg[WMAXCH];  // row-invariant; x/dy register-cached across passes
+#pragma unroll
+  for (int c = 0; c < WMAXCH; ++c) {
+    int i = lane * 8 + c * 512;
+    wreg[c] = i < H ? *(const s16x8*)(w + i) : s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+  }
   for (long long row = (long long)blockIdx.x * 4 + wave; row < nrows;
        row += (long long)gridDim.x * 4) {
     const short* xr = x + row * (long long)H;
     const short* dyr = dy + row * (long long)H;
     short* dxr = dx + row * (long long)H;
     float rs = rstd[row];
+    s16x8 xv[WMAXCH], dv[WMAXCH];
+#pragma unroll
+    for (int c = 0; c < WMAXCH; ++c) {
+      int i = lane * 8 + c * 512;
+      if (i < H) {
+        xv[c] = *(const s16x8*)(xr + i);
+        dv[c] = *(const s16x8*)(dyr + i);
+      }
+    }
     float c2 = 0.f;
 #pragma unroll WMAXCH
     for (int c = 0; c < WMAXCH; ++c) {
       int i = lane * 8 + c * 512;
       if (i >= H) break;
-      s16x8 xv = *(const s16x8*)(xr + i);
-      s16x8 dv = *(const s16x8*)(dyr + i);
-      s16x8 wv = *(const s16x8*)(w + i);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float xh = bf16_to_f32(xv[j]) * rs;
-        float wdy = bf16_to_f32(wv[j]) * bf16_to_f32(dv[j]);
+        float xh = bf16_to_f32(xv[c][j]) * rs;
+        float wdy = bf16_to_f32(wreg[c][j]) * bf16_to_f32(dv[c][j]);
         c2 = fmaf(wdy, xh, c2);
       }
     }
@@ -419,15 +443,12 @@ rmsnorm_bwd_wave_kernel(const short* __restrict__ dy,
     for (int c = 0; c < WMAXCH; ++c) {
       int i = lane * 8 + c * 512;
       if (i >= H) break;
-      s16x8 xv = *(const s16x8*)(xr + i);
-      s16x8 dv = *(const s16x8*)(dyr + i);
-      s16x8 wv = *(const s16x8*)(w + i);
       s16x8 o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float xh = bf16_to_f32(xv[j]) * rs;
-        float d = bf16_to_f32(dv[j]);
-        float wdy = bf16_to_f32(wv[j]) * d;
+        float xh = bf16_to_f32(xv[c][j]) * rs;
+        float d = bf16_to_f32(dv[c][j]);
+        float wdy = bf16_to_f32(wreg[c][j]) * d;
         o[j] = f32_to_bf16((wdy - xh * c2) * rs);
         dw_acc[c * 8 + j] = fmaf(d, xh, dw_acc[c * 8 + j]);
       }
