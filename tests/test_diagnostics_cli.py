@@ -36,3 +36,25 @@ def test_algorithm_scaffold(tmp_path, monkeypatch):
     assert (tmp_path / "tests" / "test_algo_my_count.py").exists()
     src = (tmp_path / "algorithms" / "my_count.py").read_text()
     assert "central_my_count" in src and "partial_my_count" in src
+
+
+def test_shell_command_exists_and_binds_namespace(monkeypatch):
+    """`v6amd shell` (reference: vserver-local shell) drops into an
+    interactive session with core/session/client/models bound."""
+    import code as code_module
+
+    from click.testing import CliRunner
+
+    from vantage6_amd.cli.main import cli
+
+    captured = {}
+
+    def fake_interact(banner=None, local=None):
+        captured.update(local or {})
+
+    monkeypatch.setattr(code_module, "interact", fake_interact)
+    r = CliRunner().invoke(cli, ["shell"])
+    assert r.exit_code == 0, r.output
+    assert {"core", "session", "client", "models", "Task", "Run"} <= set(captured)
+    # the bound client is authenticated and usable
+    assert captured["client"].whoami["username"] == "root"

@@ -159,6 +159,37 @@ def dev_demo(stations, encrypted):
     click.echo(json.dumps({"demo": "column_mean", "result": res[0]}, indent=2))
 
 
+@cli.command()
+@click.option("--uri", default="sqlite://", help="database URI")
+@click.option("--root-password", default="root")
+def shell(uri, root_password):
+    """Interactive shell with an orchestrator session bound (reference:
+    ``vserver-local shell`` — IPython with the models; here a stdlib
+    ``code.interact`` with ServerCore, the model classes and a root
+    client in scope)."""
+    import code
+
+    from ..client import UserClient
+    from ..server import model as model_module
+    from ..server.core import ServerCore
+
+    core = ServerCore(uri=uri)
+    try:
+        core.create_root_user("root", root_password)
+    except Exception:  # noqa: BLE001 — existing DB keeps its root
+        pass
+    client = UserClient(core=core)
+    client.authenticate("root", root_password)
+    ns = {"core": core, "session": core.session, "client": client,
+          "models": model_module}
+    for name in ("User", "Node", "Organization", "Collaboration", "Task",
+                 "Run", "Role", "Rule", "Study"):
+        ns[name] = getattr(model_module, name)
+    banner = ("vantage6_amd shell — objects: core, session, client, models, "
+              "User/Node/Organization/Collaboration/Task/Run/Role/Rule/Study")
+    code.interact(banner=banner, local=ns)
+
+
 @cli.command(name="create-private-key")
 @click.option("--output", "output_path", required=True, type=click.Path(),
               help="where to write the RSA private key")
