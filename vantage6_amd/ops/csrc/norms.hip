@@ -507,7 +507,18 @@ colsum_bf16_kernel(const short* __restrict__ dy, float* __restrict__ ws,
   float acc[8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) acc[j] = 0.f;
-  for (long long r = r0; r < r1; ++r) {
+  long long r = r0;
+  for (; r + 3 < r1; r += 4) {  // 4 independent loads in flight
+    s16x8 v0 = *(const s16x8*)(dy + r * N + col0);
+    s16x8 v1 = *(const s16x8*)(dy + (r + 1) * N + col0);
+    s16x8 v2 = *(const s16x8*)(dy + (r + 2) * N + col0);
+    s16x8 v3 = *(const s16x8*)(dy + (r + 3) * N + col0);
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      acc[j] += (bf16_to_f32(v0[j]) + bf16_to_f32(v1[j])) +
+                (bf16_to_f32(v2[j]) + bf16_to_f32(v3[j]));
+  }
+  for (; r < r1; ++r) {
     s16x8 v = *(const s16x8*)(dy + r * N + col0);
 #pragma unroll
     for (int j = 0; j < 8; ++j) acc[j] += bf16_to_f32(v[j]);
