@@ -93,6 +93,9 @@ extern "C" __global__ void gemm_nt_bf16_v10sk_kernel(const short*, const short*,
 extern "C" __global__ void gemm_wgrad_bf16_kernel(const short*, const short*,
                                                   short*, float*, int, int, int,
                                                   int);
+extern "C" __global__ void gemm_wgrad2_bf16_kernel(const short*, const short*,
+                                                   short*, float*, int, int,
+                                                   int, int);
 extern "C" __global__ void splitk_reduce_kernel(const float*, short*, long long,
                                                 int);
 extern "C" __global__ void gemm_dgrad_bf16_kernel(const short*, const short*,
@@ -675,13 +678,22 @@ torch::Tensor gemm_wgrad(torch::Tensor a, torch::Tensor b) {
   TORCH_CHECK(M % 64 == 0 && Na % 128 == 0 && Nb % 128 == 0,
               "gemm_wgrad needs M%64==0, Na/Nb%128==0");
   auto c = torch::empty({Na, Nb}, a.options());
-  int tiles = (Na / 128) * (Nb / 128);
+  // 256² tile (v2): twice the compute-per-staged-byte; measured A/B in
+  // tools/wgrad2_ab.py — dispatched when divisible, env overridable
+  static int use_v2 = [] {
+    const char* v = getenv("V6AMD_WGRAD2");
+    return v ? atoi(v) : 1;
+  }();
+  bool v2 = use_v2 && Na % 256 == 0 && Nb % 256 == 0;
+  int tiles = v2 ? (Na / 256) * (Nb / 256) : (Na / 128) * (Nb / 128);
+  auto kfn = v2 ? gemm_wgrad2_bf16_kernel : gemm_wgrad_bf16_kernel;
+  int nth = v2 ? 512 : 256;
   // split the long reduction when the tile grid starves the 256 CUs
   int nsplit = 1;
   while (tiles * nsplit < 384 && nsplit < 8 && (M / 64) / (nsplit * 2) >= 2)
     nsplit *= 2;
   if (nsplit == 1) {
-    hipLaunchKernelGGL(gemm_wgrad_bf16_kernel, dim3(tiles), dim3(256), 0,
+    hipLaunchKernelGGL(kfn, dim3(tiles), dim3(nth), 0,
                        cur_stream(), (const short*)a.data_ptr(),
                        (const short*)b.data_ptr(), (short*)c.data_ptr(),
                        nullptr, M, Na, Nb, 1);
@@ -689,7 +701,7 @@ torch::Tensor gemm_wgrad(torch::Tensor a, torch::Tensor b) {
   }
   auto ws = torch::empty({nsplit, Na, Nb},
                          a.options().dtype(torch::kFloat32));
-  hipLaunchKernelGGL(gemm_wgrad_bf16_kernel, dim3(tiles * nsplit), dim3(256), 0,
+  hipLaunchKernelGGL(kfn, dim3(tiles * nsplit), dim3(nth), 0,
                      cur_stream(), (const short*)a.data_ptr(),
                      (const short*)b.data_ptr(), nullptr,
                      ws.data_ptr<float>(), M, Na, Nb, nsplit);

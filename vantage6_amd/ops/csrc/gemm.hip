@@ -1515,3 +1515,165 @@ gemm_nt_bf16_v10sk_kernel(const short* __restrict__ A,
                           int nsplit) {
   gemm_nt_v9_body<1, 1>(A, B, C, Cws, M, N, K, nsplit);
 }
+
+// ===========================================================================
+// wgrad v2 — 256² output tile (8 waves as 2M×4N, wave = 128×64 out), BK=64
+// reduction rows, double-buffered glds staging in [m-grp][colsub][4][16]
+// subtile order, tr_b16 fragment gathers. Twice the compute-per-staged-byte
+// of the 128² wgrad; requires Na%256 && Nb%256. Split-K as in wgrad v1.
+// ===========================================================================
+extern "C" __global__ void __launch_bounds__(512)
+gemm_wgrad2_bf16_kernel(const short* __restrict__ A, const short* __restrict__ B,
+                        short* __restrict__ C, float* __restrict__ Cws,
+                        int M, int Na, int Nb, int nsplit) {
+  __shared__ short lds[2 * 2 * 16384];  // 2 buf × (A+B)[64][256] = 128 KB
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = (wave >> 2) * 128;  // output rows (Na side)
+  const int wc = (wave & 3) * 64;    // output cols (Nb side)
+  const int ntile_n = Nb / 256;
+  int wgid = blockIdx.x;
+  {
+    int nwg = gridDim.x;
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = wgid & 7, idx = wgid >> 3;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int split = wgid % nsplit;
+  wgid /= nsplit;
+  const int row0 = (wgid / ntile_n) * 256;
+  const int col0 = (wgid % ntile_n) * 256;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int src_m_in_grp = (lane & 7) >> 1;
+  const int src_col = ((lane >> 3) * 16 + (lane & 1) * 8);
+
+  // one glds = 4 m-rows × 128 out cols; per wave: 2 (grp,half) pairs × 2 ops
+  auto stage = [&](int buf, int mt) {
+    const long long m0 = (long long)mt * BK;
+    char* base = (char*)lds + buf * 65536;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int grp = wave * 2 + (i >> 1);  // m-group 0..15
+      int half = i & 1;               // 128-col half
+      const short* srcA = A + (m0 + grp * 4 + src_m_in_grp) * (long long)Na +
+                          row0 + half * 128 + src_col;
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcA,
+          (lds_uint*)(base + grp * 2048 + half * 1024), 16, 0, 0);
+      const short* srcB = B + (m0 + grp * 4 + src_m_in_grp) * (long long)Nb +
+                          col0 + half * 128 + src_col;
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcB,
+          (lds_uint*)(base + 32768 + grp * 2048 + half * 1024), 16, 0, 0);
+    }
+  };
+
+  const int nk_total = M / BK;
+  const int per = (nk_total + nsplit - 1) / nsplit;
+  const int mt0 = split * per;
+  const int mt1 = (mt0 + per < nk_total) ? mt0 + per : nk_total;
+  const int nk = mt1 - mt0;
+  const int crow_base = row0 + wr + (lane >> 4) * 4;
+  const int ccol_base = col0 + wc + (lane & 15);
+  if (nk <= 0) {
+    if (nsplit > 1) {
+      float* W = Cws + (long long)split * Na * Nb;
+#pragma unroll
+      for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+#pragma unroll
+          for (int ni = 0; ni < 4; ++ni)
+            W[(long long)(crow_base + mi * 16 + r) * Nb + ccol_base + ni * 16] =
+                0.f;
+    }
+    return;
+  }
+  stage(0, mt0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (int mti = 0; mti < nk; ++mti) {
+    int mt = mt0 + mti;
+    int cur = mti & 1;
+    if (mti + 1 < nk) {
+      stage(cur ^ 1, mt + 1);
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    char* base = (char*)lds + cur * 65536;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8_t a_frag[8], b_frag[4];
+      int g0 = kk * 8 + (lane >> 4) * 2;
+#pragma unroll
+      for (int fi = 0; fi < 8; ++fi) {
+        int colA = wr + fi * 16;
+        bf16x4_t lo = tr_read_b64_issue(
+            base + ((g0 * 16 + (colA >> 4)) * 64) * 2 + (lane & 15) * 8);
+        bf16x4_t hi = tr_read_b64_issue(
+            base + (((g0 + 1) * 16 + (colA >> 4)) * 64) * 2 + (lane & 15) * 8);
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          a_frag[fi][e] = lo[e];
+          a_frag[fi][e + 4] = hi[e];
+        }
+      }
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi) {
+        int colB = wc + fi * 16;
+        bf16x4_t lo = tr_read_b64_issue(
+            base + 32768 + ((g0 * 16 + (colB >> 4)) * 64) * 2 +
+            (lane & 15) * 8);
+        bf16x4_t hi = tr_read_b64_issue(
+            base + 32768 + (((g0 + 1) * 16 + (colB >> 4)) * 64) * 2 +
+            (lane & 15) * 8);
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          b_frag[fi][e] = lo[e];
+          b_frag[fi][e + 4] = hi[e];
+        }
+      }
+      tr_read_drain();
+#pragma unroll
+      for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+  if (nsplit == 1) {
+#pragma unroll
+    for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = crow_base + mi * 16 + r;
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          C[(long long)row * Nb + ccol_base + ni * 16] =
+              f32_to_bf16(acc[mi][ni][r]);
+      }
+  } else {
+    float* W = Cws + (long long)split * Na * Nb;
+#pragma unroll
+    for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = crow_base + mi * 16 + r;
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          W[(long long)row * Nb + ccol_base + ni * 16] = acc[mi][ni][r];
+      }
+  }
+}
