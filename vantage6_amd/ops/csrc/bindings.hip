@@ -678,11 +678,13 @@ torch::Tensor gemm_wgrad(torch::Tensor a, torch::Tensor b) {
   TORCH_CHECK(M % 64 == 0 && Na % 128 == 0 && Nb % 128 == 0,
               "gemm_wgrad needs M%64==0, Na/Nb%128==0");
   auto c = torch::empty({Na, Nb}, a.options());
-  // 256² tile (v2): twice the compute-per-staged-byte; measured A/B in
-  // tools/wgrad2_ab.py — dispatched when divisible, env overridable
+  // 256² tile (v2): measured SLOWER than the 128² tile on every training
+  // shape (tools/wgrad2_ab.py — e.g. 59 vs 26 µs at 2048²/M=1024: the
+  // tr_b16 gather latency doubles with the tile while hiding no more of
+  // it, and the grid shrinks 4×). Kept behind V6AMD_WGRAD2=1.
   static int use_v2 = [] {
     const char* v = getenv("V6AMD_WGRAD2");
-    return v ? atoi(v) : 1;
+    return v ? atoi(v) : 0;
   }();
   bool v2 = use_v2 && Na % 256 == 0 && Nb % 256 == 0;
   int tiles = v2 ? (Na / 256) * (Nb / 256) : (Na / 128) * (Nb / 128);
