@@ -286,17 +286,23 @@ attn_fwd_d64_plin_kernel(const short* __restrict__ Q,
 // ===========================================================================
 extern "C" __global__ void __launch_bounds__(256)
 attn_bwd_d_kernel(const short* __restrict__ dO, const short* __restrict__ O,
-                  float* __restrict__ D, long long n_rows) {
-  // one wave of 4 threads... simple: thread per row, 64-wide dot
+                  float* __restrict__ D, int nheads, int S, long long n_rows,
+                  long long qs_b, long long qs_h, long long qs_s) {
+  // thread per (b, h, s) row: 64-wide dot with arbitrary row strides
   long long row = (long long)blockIdx.x * 256 + threadIdx.x;
   if (row >= n_rows) return;
-  const short* a = dO + row * 64;
-  const short* b = O + row * 64;
+  int sidx = (int)(row % S);
+  long long bh = row / S;
+  int h = (int)(bh % nheads);
+  long long b = bh / nheads;
+  long long off = b * qs_b + (long long)h * qs_h + (long long)sidx * qs_s;
+  const short* a = dO + off;
+  const short* o = O + off;
   float acc = 0.f;
 #pragma unroll
   for (int i = 0; i < 8; ++i) {
     s16x8 va = *(const s16x8*)(a + i * 8);
-    s16x8 vb = *(const s16x8*)(b + i * 8);
+    s16x8 vb = *(const s16x8*)(o + i * 8);
 #pragma unroll
     for (int q = 0; q < 8; ++q)
       acc = fmaf(bf16_to_f32(va[q]), bf16_to_f32(vb[q]), acc);

@@ -122,6 +122,9 @@ extern "C" __global__ void attn_fwd_d64_plin_kernel(
     const short*, const short*, const short*, short*, float*, int, int, int,
     int, float, int, long long, long long, long long, long long, long long,
     long long);
+extern "C" __global__ void attn_bwd_d_kernel(const short*, const short*,
+                                             float*, int, int, long long,
+                                             long long, long long, long long);
 extern "C" __global__ void attn_bwd_dq_d64_kernel(
     const short*, const short*, const short*, const short*, const float*,
     const float*, short*, int, int, int, int, float, int, long long,
@@ -926,8 +929,6 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   auto dv = torch::empty_strided(v.sizes(), v.strides(), v.options());
   auto dsum = torch::empty({B, H, S}, q.options().dtype(torch::kFloat32));
   long long nrows = (long long)B * H * S;
-  // D = rowsum(dO ∘ O): torch handles arbitrary strides; tiny tensor
-  dsum.copy_((dout.to(torch::kFloat32) * o.to(torch::kFloat32)).sum(-1));
   // the kernels take ONE stride set for the q-side tensors (q, dO, dq):
   // relayout dO into q's layout when it differs (one copy, rare)
   torch::Tensor dco = dout;
@@ -935,6 +936,13 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
     dco = torch::empty_strided(q.sizes(), q.strides(), q.options());
     dco.copy_(dout);
   }
+  // D = rowsum(dO ∘ O) via the strided kernel (the torch-expression form
+  // materialised ~100 MB of fp32 temporaries per call)
+  hipLaunchKernelGGL(attn_bwd_d_kernel, dim3((int)((nrows + 255) / 256)),
+                     dim3(256), 0, cur_stream(),
+                     (const short*)dco.data_ptr(), (const short*)o.data_ptr(),
+                     dsum.data_ptr<float>(), H, S, nrows, q.stride(0),
+                     q.stride(1), q.stride(2));
   hipLaunchKernelGGL(attn_bwd_dq_d64_kernel, dim3((S + 127) / 128, B * H),
                      dim3(512), 0, cur_stream(), (const short*)q.data_ptr(),
                      (const short*)k.data_ptr(), (const short*)v.data_ptr(),
