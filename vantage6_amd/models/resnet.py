@@ -55,10 +55,16 @@ class BasicBlock(nn.Module):
         self.bn1 = FusedBatchNorm2d(cout)
         self.conv2 = conv3x3(cout, cout)
         self.bn2 = FusedBatchNorm2d(cout)
+        # fuse each BN's batch-stat accumulation into the producing conv's
+        # epilogue (skips BN-fwd's whole partials read pass); plain
+        # attribute set so the BN is not re-registered as a conv child
+        object.__setattr__(self.conv1, "stats_bn", self.bn1)
+        object.__setattr__(self.conv2, "stats_bn", self.bn2)
         self.down_conv = None
         if stride != 1 or cin != cout:
             self.down_conv = HipConv2d(cin, cout, 1, stride=stride, bias=False)
             self.down_bn = FusedBatchNorm2d(cout)
+            object.__setattr__(self.down_conv, "stats_bn", self.down_bn)
 
     def forward(self, x):
         if self.down_conv is None:

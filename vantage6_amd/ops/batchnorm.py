@@ -16,7 +16,8 @@ from . import hip_ops
 class _FusedBNFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var, training,
-                momentum, eps, relu, residual, fwd_acc=None, bwd_acc=None):
+                momentum, eps, relu, residual, fwd_acc=None, bwd_acc=None,
+                stats_precomputed=False):
         ctx.fwd_acc, ctx.bwd_acc = fwd_acc, bwd_acc
         if x.is_cuda:
             xc = x.contiguous(memory_format=torch.channels_last)
@@ -28,7 +29,7 @@ class _FusedBNFn(torch.autograd.Function):
                         .permute(0, 2, 3, 1).reshape(-1, C))
             y2, mean, rstd, msk = hip_ops().bn_fwd(
                 x2, weight, bias, running_mean, running_var, training,
-                momentum, eps, relu, res2, ctx.fwd_acc,
+                momentum, eps, relu, res2, ctx.fwd_acc, stats_precomputed,
             )
             y = y2.view(N, H, W, C).permute(0, 3, 1, 2)
             ctx.save_for_backward(x2, msk, weight, mean, rstd)
@@ -69,7 +70,8 @@ class _FusedBNFn(torch.autograd.Function):
             dres = None
             if has_res:
                 dres = outs[3].view(N, H, W, C).permute(0, 3, 1, 2)
-            return dx, dw, db, None, None, None, None, None, None, dres, None, None
+            return (dx, dw, db, None, None, None, None, None, None, dres,
+                    None, None, None)
         # ---- CPU reference (y_or_mask holds y on CPU)
         y = y_or_mask
         dyf = dy.float()
@@ -85,7 +87,7 @@ class _FusedBNFn(torch.autograd.Function):
             dyf - db_[None, :, None, None] / R - xhat * dw_[None, :, None, None] / R
         )
         return (dxf.to(x.dtype), dw_.to(weight.dtype), db_.to(weight.dtype),
-                None, None, None, None, None, None, dres, None, None)
+                None, None, None, None, None, None, dres, None, None, None)
 
 
 class FusedBatchNorm2d(torch.nn.Module):
@@ -103,13 +105,25 @@ class FusedBatchNorm2d(torch.nn.Module):
         self._fwd_acc = None
         self._bwd_acc = None
 
+    # set True by a producing Conv2d whose epilogue already accumulated
+    # this BN's batch sum/sumsq into _fwd_acc (ops/conv.py)
+    _pending_stats = False
+
+    def _ensure_fwd_acc(self, c, device):
+        if self._fwd_acc is None:
+            self._fwd_acc = torch.zeros(16, 2, c, device=device)
+            self._bwd_acc = torch.zeros(16, 2, c, device=device)
+        return self._fwd_acc
+
     def forward(self, x, residual=None, relu: bool = False):
         if x.is_cuda and self._fwd_acc is None:
             c = self.weight.numel()
             self._fwd_acc = torch.zeros(16, 2, c, device=x.device)
             self._bwd_acc = torch.zeros(16, 2, c, device=x.device)
+        pre = self._pending_stats
+        self._pending_stats = False
         return _FusedBNFn.apply(
             x, self.weight, self.bias, self.running_mean, self.running_var,
             self.training, self.momentum, self.eps, relu, residual,
-            self._fwd_acc, self._bwd_acc,
+            self._fwd_acc, self._bwd_acc, pre,
         )

@@ -18,8 +18,14 @@ from . import hip_ops
 
 class _ConvFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, stride, pad):
-        y = hip_ops().conv_nhwc_fwd(x, w, stride, pad)
+    def forward(ctx, x, w, stride, pad, bn_acc=None, stats_sink=None):
+        if bn_acc is not None:
+            y, done = hip_ops().conv_nhwc_fwd_stats(x, w, stride, pad, bn_acc)
+            if stats_sink is not None and bool(done.item() if done.is_cpu
+                                               else done.cpu().item()):
+                stats_sink._pending_stats = True
+        else:
+            y = hip_ops().conv_nhwc_fwd(x, w, stride, pad)
         ctx.save_for_backward(x, w)
         ctx.stride, ctx.pad = stride, pad
         return y
@@ -50,7 +56,7 @@ class _ConvFn(torch.autograd.Function):
                 _, dw, _ = torch.ops.aten.convolution_backward(
                     dy, x, w, None, [s, s], [p, p], [1, 1], False, [0, 0], 1,
                     [False, True, False])
-            return dx, dw, None, None
+            return dx, dw, None, None, None, None
         dx, dw_lib, _ = torch.ops.aten.convolution_backward(
             dy, x, w, None, [s, s], [p, p], [1, 1], False, [0, 0], 1,
             [True, not own_wgrad, False])
@@ -59,7 +65,13 @@ class _ConvFn(torch.autograd.Function):
 
 class Conv2d(nn.Conv2d):
     """Drop-in Conv2d: hand-written NHWC kernel on the eligible path,
-    library conv otherwise (3-channel stem, CPU, non-NHWC)."""
+    library conv otherwise (3-channel stem, CPU, non-NHWC).
+
+    When ``stats_bn`` is set to the downstream :class:`FusedBatchNorm2d`,
+    the conv epilogue accumulates that BN's batch sum/sumsq directly and
+    the BN forward skips its whole partials read pass."""
+
+    stats_bn = None  # wired by the model (e.g. ResNet BasicBlock)
 
     def forward(self, x):
         w = self.weight
@@ -71,5 +83,10 @@ class Conv2d(nn.Conv2d):
                 and self.padding[0] == self.padding[1]
                 and x.is_contiguous(memory_format=torch.channels_last)
                 and w.is_contiguous(memory_format=torch.channels_last)):
-            return _ConvFn.apply(x, w, self.stride[0], self.padding[0])
+            bn = self.stats_bn
+            acc = None
+            if bn is not None and bn.training and self.training:
+                acc = bn._ensure_fwd_acc(self.out_channels, x.device)
+            return _ConvFn.apply(x, w, self.stride[0], self.padding[0],
+                                 acc, bn)
         return super().forward(x)

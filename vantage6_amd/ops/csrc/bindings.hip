@@ -137,6 +137,16 @@ extern "C" __global__ void conv_nhwc_wgrad_kernel(
     const short*, const short*, float*, short*, const short*, int, int, int,
     int, int, int, int, int, int, int, int, int, int, unsigned long long,
     unsigned long long);
+extern "C" __global__ void conv_nhwc_fwd64st_kernel(const short*, const short*,
+                                                    short*, const short*,
+                                                    float*, int, int, int, int,
+                                                    int, int, int, int, int,
+                                                    int, int);
+extern "C" __global__ void conv_nhwc_fwd128st_kernel(const short*, const short*,
+                                                     short*, const short*,
+                                                     float*, int, int, int,
+                                                     int, int, int, int, int,
+                                                     int, int, int);
 extern "C" __global__ void conv_nhwc_fwd64sk_kernel(const short*, const short*,
                                                     float*, const short*, int,
                                                     int, int, int, int, int,
@@ -535,7 +545,8 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor w,
                                   torch::Tensor running_var, bool training,
                                   double momentum, double eps, bool relu,
                                   c10::optional<torch::Tensor> residual,
-                                  c10::optional<torch::Tensor> acc) {
+                                  c10::optional<torch::Tensor> acc,
+                                  bool stats_precomputed) {
   CHECK_HIP(x, torch::kBFloat16);
   int C = (int)x.size(-1);
   TORCH_CHECK(C % 8 == 0 && C <= 2048, "BN channels must be %8==0 and <=2048");
@@ -563,10 +574,14 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor w,
                          dim3(256), 0, cur_stream(), part.data_ptr<float>(),
                          32 * C);
   }
-  if (training) {
+  if (training && !stats_precomputed) {
     hipLaunchKernelGGL(bn_fwd_partials_kernel, dim3(nblocks), dim3(256), 0,
                        cur_stream(), (const short*)x.data_ptr(),
                        part.data_ptr<float>(), R, C, nblocks);
+  } else if (stats_precomputed) {
+    // the producing conv's epilogue already accumulated sum/sumsq
+    TORCH_CHECK(acc.has_value(),
+                "bn_fwd: stats_precomputed requires the shared accumulator");
   }
   hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0,
                      cur_stream(), part.data_ptr<float>(), nblocks, C, R,
@@ -808,6 +823,11 @@ torch::Tensor gemm_nt_probe(torch::Tensor a, torch::Tensor b, int variant,
 
 // Direct NHWC conv forward (conv.hip): x [N,C,H,W] channels_last bf16,
 // w [Cout,Cin,KH,KW] channels_last bf16 → out [N,Cout,OH,OW] channels_last.
+std::vector<torch::Tensor> conv_nhwc_fwd_stats(torch::Tensor x,
+                                               torch::Tensor w,
+                                               int64_t stride, int64_t pad,
+                                               torch::Tensor bn_acc);
+
 torch::Tensor conv_nhwc_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
                             int64_t pad) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
@@ -989,6 +1009,64 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                      cur_stream(), wsv.data_ptr<float>(),
                      (short*)dv.data_ptr(), nkv, group);
   return {dq, dk, dv};
+}
+
+// conv fwd + fused BN-stat accumulation (acc = the downstream BN's
+// [16][2][Cout] atomic accumulator). Returns (out, stats_done): the
+// split-K path does not produce stats (the caller then runs the separate
+// BN partials pass).
+std::vector<torch::Tensor> conv_nhwc_fwd_stats(torch::Tensor x,
+                                               torch::Tensor w,
+                                               int64_t stride, int64_t pad,
+                                               torch::Tensor bn_acc) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "x NHWC");
+  TORCH_CHECK(w.is_contiguous(at::MemoryFormat::ChannelsLast), "w NHWC");
+  int Nb = (int)x.size(0), Cin = (int)x.size(1), H = (int)x.size(2),
+      Wd = (int)x.size(3);
+  int Cout = (int)w.size(0), KH = (int)w.size(2), KW = (int)w.size(3);
+  TORCH_CHECK(Cin % 64 == 0 && Cout % 64 == 0);
+  TORCH_CHECK(bn_acc.numel() == 32LL * Cout, "bn_acc must be [16][2][Cout]");
+  int OH = (H + 2 * (int)pad - KH) / (int)stride + 1;
+  int OW = (Wd + 2 * (int)pad - KW) / (int)stride + 1;
+  int M = Nb * OH * OW;
+  int mtiles = (M + 127) / 128;
+  bool wide = Cout % 128 == 0 && Cout > 64;
+  int tbn = wide ? 128 : 64;
+  int tiles = mtiles * (Cout / tbn);
+  int nk = KH * KW * (Cin / 64);
+  bool starved = tiles < 80;
+  int ns = 1;
+  if (starved) {
+    for (int cand = 8; cand > 1; cand >>= 1) {
+      int per = (nk + cand - 1) / cand;
+      if ((cand - 1) * per < nk && nk / cand >= 4 && tiles * cand >= 200 &&
+          tiles * cand <= 1024) {
+        ns = cand;
+        break;
+      }
+    }
+  }
+  auto done = torch::zeros({1}, torch::TensorOptions().dtype(torch::kUInt8));
+  if (ns > 1) {
+    // split-K path has no stats — delegate and report stats_done = 0
+    auto out = conv_nhwc_fwd(x, w, stride, pad);
+    return {out, done};
+  }
+  auto out = torch::empty({Nb, Cout, OH, OW},
+                          x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  static torch::Tensor zero_page2;
+  if (!zero_page2.defined() || zero_page2.device() != x.device())
+    zero_page2 = torch::zeros({128}, x.options());
+  auto kfn = wide ? conv_nhwc_fwd128st_kernel : conv_nhwc_fwd64st_kernel;
+  hipLaunchKernelGGL(kfn, dim3(tiles), dim3(256), 0, cur_stream(),
+                     (const short*)x.data_ptr(), (const short*)w.data_ptr(),
+                     (short*)out.data_ptr(),
+                     (const short*)zero_page2.data_ptr(),
+                     bn_acc.data_ptr<float>(), Nb, H, Wd, Cin, Cout, OH, OW,
+                     KH, KW, (int)stride, (int)pad);
+  done.fill_(1);
+  return {out, done};
 }
 
 // conv wgrad: dW = per-tap dy^T · gathered-x (conv.hip); returns the
@@ -1234,7 +1312,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
-  m.def("bn_fwd", &bn_fwd, "fused NHWC BatchNorm fwd (+relu/+residual)");
+  m.def("bn_fwd", &bn_fwd, "fused NHWC BatchNorm fwd (+relu/+residual)",
+        py::arg("x"), py::arg("w"), py::arg("b"), py::arg("running_mean"),
+        py::arg("running_var"), py::arg("training"), py::arg("momentum"),
+        py::arg("eps"), py::arg("relu"), py::arg("residual") = py::none(),
+        py::arg("acc") = py::none(), py::arg("stats_precomputed") = false);
   m.def("bn_bwd", &bn_bwd, "fused NHWC BatchNorm bwd");
   m.def("gemm_bf16", &gemm_bf16, "bf16 MFMA GEMM, fp32 accumulate");
   m.def("attn_bwd", &attn_bwd, "flash-style attention bwd (d=64)",
@@ -1248,6 +1330,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("stride") = 1, py::arg("pad") = 1);
   m.def("conv_flip_weights", &conv_flip_weights,
         "flip+transpose conv weights for stride-1 dgrad");
+  m.def("conv_nhwc_fwd_stats", &conv_nhwc_fwd_stats,
+        "conv forward with fused BN-stat accumulation",
+        py::arg("x"), py::arg("w"), py::arg("stride"), py::arg("pad"),
+        py::arg("bn_acc"));
   m.def("conv_nhwc_fwd", &conv_nhwc_fwd, "direct NHWC conv forward",
         py::arg("x"), py::arg("w"), py::arg("stride") = 1, py::arg("pad") = 1);
   m.def("gemm_nt_probe", &gemm_nt_probe, "direct NT kernel-variant launcher",

@@ -22,13 +22,13 @@ typedef __bf16 cbf16x8_t __attribute__((ext_vector_type(8)));
 typedef __attribute__((address_space(3))) unsigned int lds_uint;
 typedef __attribute__((address_space(1))) const unsigned int global_uint;
 
-template <int TBN, int SPLITK>
+template <int TBN, int SPLITK, int STATS = 0>
 __device__ __forceinline__ void conv_nhwc_fwd_body(
     const short* __restrict__ X, const short* __restrict__ Wt,
     short* __restrict__ Out, float* __restrict__ Ws,
     const short* __restrict__ zero_page, int Nb, int H, int Wd, int Cin,
     int Cout, int OH, int OW, int KH, int KW, int stride, int pad,
-    int nsplit) {
+    int nsplit, float* __restrict__ bn_acc = nullptr) {
   if (!SPLITK) nsplit = 1;
   __shared__ short lds[2 * (128 + TBN) * 64];  // dbuf × (A[128][64]+B[TBN][64])
   const int tid = threadIdx.x;
@@ -172,6 +172,36 @@ __device__ __forceinline__ void conv_nhwc_fwd_body(
         }
       }
     }
+  if (STATS && !SPLITK) {
+    // feed the downstream BatchNorm's [16][2][C] atomic accumulator from
+    // the epilogue — eliminates BN-fwd's whole partials read pass. Stats
+    // use the bf16-ROUNDED output (what the separate pass would read).
+#pragma unroll
+    for (int ni = 0; ni < NFRAG; ++ni) {
+      float s1 = 0.f, s2 = 0.f;
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = crow_base + mi * 16 + r;
+          if (row < M) {
+            float v = bf16_to_f32(f32_to_bf16(acc[mi][ni][r]));
+            s1 += v;
+            s2 = fmaf(v, v, s2);
+          }
+        }
+#pragma unroll
+      for (int off = 16; off < 64; off <<= 1) {
+        s1 += __shfl_xor(s1, off, 64);
+        s2 += __shfl_xor(s2, off, 64);
+      }
+      if ((lane >> 4) == 0) {
+        float* slice = bn_acc + (long long)(blockIdx.x & 15) * 2 * Cout;
+        atomicAdd(&slice[ccol_base + ni * 16], s1);
+        atomicAdd(&slice[Cout + ccol_base + ni * 16], s2);
+      }
+    }
+  }
 }
 
 extern "C" __global__ void __launch_bounds__(256)
@@ -190,6 +220,26 @@ conv_nhwc_fwd128_kernel(const short* X, const short* Wt, short* Out,
                         int pad) {
   conv_nhwc_fwd_body<128, 0>(X, Wt, Out, nullptr, zero_page, Nb, H, Wd, Cin,
                              Cout, OH, OW, KH, KW, stride, pad, 1);
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+conv_nhwc_fwd64st_kernel(const short* X, const short* Wt, short* Out,
+                         const short* zero_page, float* bn_acc, int Nb, int H,
+                         int Wd, int Cin, int Cout, int OH, int OW, int KH,
+                         int KW, int stride, int pad) {
+  conv_nhwc_fwd_body<64, 0, 1>(X, Wt, Out, nullptr, zero_page, Nb, H, Wd,
+                               Cin, Cout, OH, OW, KH, KW, stride, pad, 1,
+                               bn_acc);
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+conv_nhwc_fwd128st_kernel(const short* X, const short* Wt, short* Out,
+                          const short* zero_page, float* bn_acc, int Nb,
+                          int H, int Wd, int Cin, int Cout, int OH, int OW,
+                          int KH, int KW, int stride, int pad) {
+  conv_nhwc_fwd_body<128, 0, 1>(X, Wt, Out, nullptr, zero_page, Nb, H, Wd,
+                                Cin, Cout, OH, OW, KH, KW, stride, pad, 1,
+                                bn_acc);
 }
 
 extern "C" __global__ void __launch_bounds__(256)
