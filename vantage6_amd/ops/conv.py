@@ -60,7 +60,7 @@ class _ConvFn(torch.autograd.Function):
         dx, dw_lib, _ = torch.ops.aten.convolution_backward(
             dy, x, w, None, [s, s], [p, p], [1, 1], False, [0, 0], 1,
             [True, not own_wgrad, False])
-        return dx, (dw if own_wgrad else dw_lib), None, None
+        return dx, (dw if own_wgrad else dw_lib), None, None, None, None
 
 
 class Conv2d(nn.Conv2d):
