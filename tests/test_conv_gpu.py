@@ -94,3 +94,39 @@ def test_conv_wgrad_matches_fp32(cin, h, cout, k, s, p):
     scale = ref.abs().max().item()
     err = (got - ref).abs().max().item()
     assert err <= 0.05 * scale + 0.05, (err, scale)
+
+
+@pytest.mark.timeout(300)
+def test_conv_bn_fused_stats_matches_plain():
+    """Conv-epilogue BN-stat accumulation must reproduce the separate
+    partials pass (same bf16-rounded inputs; only atomic add order
+    differs)."""
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    from vantage6_amd.ops.batchnorm import FusedBatchNorm2d
+    from vantage6_amd.ops.conv import Conv2d as HipConv2d
+
+    torch.manual_seed(0)
+    x = torch.randn(8, 64, 28, 28, device="cuda", dtype=torch.bfloat16).to(
+        memory_format=torch.channels_last)
+
+    def build(fused):
+        torch.manual_seed(1)
+        conv = HipConv2d(64, 128, 3, padding=1, bias=False).cuda().bfloat16() \
+            .to(memory_format=torch.channels_last)
+        bn = FusedBatchNorm2d(128).cuda()
+        if fused:
+            object.__setattr__(conv, "stats_bn", bn)
+        conv.train(), bn.train()
+        y = bn(conv(x), relu=True)
+        return y, bn
+
+    y_plain, bn_plain = build(False)
+    y_fused, bn_fused = build(True)
+    assert bn_fused._pending_stats is False  # consumed
+    scale = y_plain.float().abs().max().item()
+    err = (y_fused.float() - y_plain.float()).abs().max().item()
+    assert err <= 2e-2 * scale + 2e-2, err
+    rm = (bn_fused.running_mean - bn_plain.running_mean).abs().max().item()
+    rv = (bn_fused.running_var - bn_plain.running_var).abs().max().item()
+    assert rm < 1e-3 and rv < 1e-3, (rm, rv)
