@@ -13,6 +13,7 @@ import torch
 import torch.nn as nn
 
 from ..ops.conv import Conv2d as HipConv2d
+from ..ops.conv import MaxPool2d as HipMaxPool2d
 
 from ..ops.gemm import Linear as MfmaLinear
 from ..ops.batchnorm import FusedBatchNorm2d
@@ -81,7 +82,7 @@ class ResNet18(nn.Module):
         super().__init__()
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
         self.bn1 = FusedBatchNorm2d(64)
-        self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
+        self.maxpool = HipMaxPool2d()  # hand-written NHWC kernels
         layers = []
         cin = 64
         for cout, stride in ((64, 1), (128, 2), (256, 2), (512, 2)):

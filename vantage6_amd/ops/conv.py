@@ -90,3 +90,30 @@ class Conv2d(nn.Conv2d):
             return _ConvFn.apply(x, w, self.stride[0], self.padding[0],
                                  acc, bn)
         return super().forward(x)
+
+
+class _MaxPoolFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        y, idx = hip_ops().maxpool_nhwc_fwd(x)
+        ctx.save_for_backward(idx)
+        ctx.hw = (x.shape[2], x.shape[3])
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        h, w = ctx.hw
+        return hip_ops().maxpool_nhwc_bwd(dy, idx, h, w)
+
+
+class MaxPool2d(nn.Module):
+    """3×3 stride-2 pad-1 max pool: hand-written NHWC kernels with a
+    uint8 argmax (gather-based backward, no atomics); library fallback
+    off the eligible path."""
+
+    def forward(self, x):
+        if (x.is_cuda and x.dtype == torch.bfloat16 and x.shape[1] % 8 == 0
+                and x.is_contiguous(memory_format=torch.channels_last)):
+            return _MaxPoolFn.apply(x)
+        return torch.nn.functional.max_pool2d(x, 3, stride=2, padding=1)

@@ -112,6 +112,13 @@ extern "C" __global__ void conv_nhwc_fwd128_kernel(const short*, const short*,
                                                    int, int, int, int, int);
 extern "C" __global__ void conv_flip_weights_kernel(const short*, short*, int,
                                                     int, int, int);
+extern "C" __global__ void maxpool_nhwc_fwd_kernel(const short*, short*,
+                                                   unsigned char*, int, int,
+                                                   int, int, int, int);
+extern "C" __global__ void maxpool_nhwc_bwd_kernel(const short*,
+                                                   const unsigned char*,
+                                                   short*, int, int, int, int,
+                                                   int, int);
 extern "C" __global__ void attn_fwd_d64_kernel(const short*, const short*,
                                                const short*, short*, float*,
                                                int, int, int, int, float, int,
@@ -1129,6 +1136,44 @@ torch::Tensor conv_nhwc_wgrad(torch::Tensor dy, torch::Tensor x, int64_t kh,
   return dw;
 }
 
+// NHWC max-pool 3x3/s2/p1 forward (+argmax) and backward (conv.hip).
+std::vector<torch::Tensor> maxpool_nhwc_fwd(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "x NHWC");
+  int Nb = (int)x.size(0), C = (int)x.size(1), H = (int)x.size(2),
+      Wd = (int)x.size(3);
+  TORCH_CHECK(C % 8 == 0);
+  int OH = (H + 1) / 2, OW = (Wd + 1) / 2;
+  auto y = torch::empty({Nb, C, OH, OW},
+                        x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  auto idx = torch::empty({(long long)Nb * OH * OW * C},
+                          x.options().dtype(torch::kUInt8));
+  long long total = (long long)Nb * OH * OW * (C / 8);
+  hipLaunchKernelGGL(maxpool_nhwc_fwd_kernel,
+                     dim3((int)((total + 255) / 256)), dim3(256), 0,
+                     cur_stream(), (const short*)x.data_ptr(),
+                     (short*)y.data_ptr(), (unsigned char*)idx.data_ptr(), Nb,
+                     H, Wd, C, OH, OW);
+  return {y, idx};
+}
+
+torch::Tensor maxpool_nhwc_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
+                               int64_t Wd) {
+  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == torch::kBFloat16);
+  auto dyc = dy.contiguous(at::MemoryFormat::ChannelsLast);
+  int Nb = (int)dyc.size(0), C = (int)dyc.size(1), OH = (int)dyc.size(2),
+      OW = (int)dyc.size(3);
+  auto dx = torch::empty({Nb, C, (long)H, (long)Wd},
+                         dy.options().memory_format(at::MemoryFormat::ChannelsLast));
+  long long total = (long long)Nb * H * Wd * (C / 8);
+  hipLaunchKernelGGL(maxpool_nhwc_bwd_kernel,
+                     dim3((int)((total + 255) / 256)), dim3(256), 0,
+                     cur_stream(), (const short*)dyc.data_ptr(),
+                     (const unsigned char*)idx.data_ptr(),
+                     (short*)dx.data_ptr(), Nb, (int)H, (int)Wd, C, OH, OW);
+  return dx;
+}
+
 // Weight transform for stride-1 dgrad (one kernel: flip + channel
 // transpose into channels-last layout for conv_nhwc_fwd).
 torch::Tensor conv_flip_weights(torch::Tensor w) {
@@ -1328,6 +1373,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_nhwc_wgrad", &conv_nhwc_wgrad, "direct NHWC conv weight grad",
         py::arg("dy"), py::arg("x"), py::arg("kh"), py::arg("kw"),
         py::arg("stride") = 1, py::arg("pad") = 1);
+  m.def("maxpool_nhwc_fwd", &maxpool_nhwc_fwd, "NHWC maxpool 3x3/s2/p1 + argmax");
+  m.def("maxpool_nhwc_bwd", &maxpool_nhwc_bwd, "NHWC maxpool backward (gather)");
   m.def("conv_flip_weights", &conv_flip_weights,
         "flip+transpose conv weights for stride-1 dgrad");
   m.def("conv_nhwc_fwd_stats", &conv_nhwc_fwd_stats,

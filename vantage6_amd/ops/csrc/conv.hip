@@ -489,3 +489,104 @@ conv_nhwc_wgrad_kernel(const short* __restrict__ dY,
       }
     }
 }
+
+// ---------------------------------------------------------------------------
+// NHWC max-pool 3×3/s2/p1 (the ResNet stem): forward saves a per-channel
+// uint8 window-argmax; backward GATHERS from the ≤4 covering windows (no
+// atomics). torch's NHWC maxpool_backward measured 83 µs per call on the
+// headline shape; these are plain streaming kernels.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+maxpool_nhwc_fwd_kernel(const short* __restrict__ X, short* __restrict__ Y,
+                        unsigned char* __restrict__ Idx, int Nb, int H, int Wd,
+                        int C, int OH, int OW) {
+  // thread per (output pixel, 8-channel chunk)
+  long long t = (long long)blockIdx.x * 256 + threadIdx.x;
+  int cchunks = C >> 3;
+  long long total = (long long)Nb * OH * OW * cchunks;
+  if (t >= total) return;
+  int cc = (int)(t % cchunks);
+  long long p = t / cchunks;
+  int ow = (int)(p % OW);
+  long long q = p / OW;
+  int oh = (int)(q % OH);
+  int n = (int)(q / OH);
+  const int c0 = cc * 8;
+  float best[8];
+  unsigned char bidx[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    best[j] = -3e38f;
+    bidx[j] = 0;
+  }
+  for (int ky = 0; ky < 3; ++ky) {
+    int ih = oh * 2 + ky - 1;
+    if ((unsigned)ih >= (unsigned)H) continue;
+    for (int kx = 0; kx < 3; ++kx) {
+      int iw = ow * 2 + kx - 1;
+      if ((unsigned)iw >= (unsigned)Wd) continue;
+      s16x8 v = *(const s16x8*)(X + (((long long)(n * H + ih) * Wd + iw) * C) +
+                                c0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf16_to_f32(v[j]);
+        if (f > best[j]) {
+          best[j] = f;
+          bidx[j] = (unsigned char)(ky * 3 + kx);
+        }
+      }
+    }
+  }
+  s16x8 o;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) o[j] = f32_to_bf16(best[j]);
+  long long off = ((long long)(n * OH + oh) * OW + ow) * C + c0;
+  *(s16x8*)(Y + off) = o;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) Idx[off + j] = bidx[j];
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+maxpool_nhwc_bwd_kernel(const short* __restrict__ dY,
+                        const unsigned char* __restrict__ Idx,
+                        short* __restrict__ dX, int Nb, int H, int Wd, int C,
+                        int OH, int OW) {
+  // thread per (input pixel, 8-channel chunk): gather from covering windows
+  long long t = (long long)blockIdx.x * 256 + threadIdx.x;
+  int cchunks = C >> 3;
+  long long total = (long long)Nb * H * Wd * cchunks;
+  if (t >= total) return;
+  int cc = (int)(t % cchunks);
+  long long p = t / cchunks;
+  int iw = (int)(p % Wd);
+  long long q = p / Wd;
+  int ih = (int)(q % H);
+  int n = (int)(q / H);
+  const int c0 = cc * 8;
+  float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  // windows with oh*2 + ky - 1 == ih → oh = (ih + 1 - ky)/2 for ky 0..2
+#pragma unroll
+  for (int ky = 0; ky < 3; ++ky) {
+    int num = ih + 1 - ky;
+    if (num < 0 || (num & 1)) continue;
+    int oh = num >> 1;
+    if (oh >= OH) continue;
+#pragma unroll
+    for (int kx = 0; kx < 3; ++kx) {
+      int numx = iw + 1 - kx;
+      if (numx < 0 || (numx & 1)) continue;
+      int ow = numx >> 1;
+      if (ow >= OW) continue;
+      long long off = ((long long)(n * OH + oh) * OW + ow) * C + c0;
+      s16x8 g = *(const s16x8*)(dY + off);
+      unsigned char code = (unsigned char)(ky * 3 + kx);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (Idx[off + j] == code) acc[j] += bf16_to_f32(g[j]);
+    }
+  }
+  s16x8 o;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) o[j] = f32_to_bf16(acc[j]);
+  *(s16x8*)(dX + (((long long)(n * H + ih) * Wd + iw) * C) + c0) = o;
+}
