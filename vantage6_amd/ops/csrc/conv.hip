@@ -37,7 +37,6 @@ __device__ __forceinline__ void conv_nhwc_fwd_body(
   const int wr = (wave >> 1) * 64;          // wave's output-row group
   const int wc = (wave & 1) * (TBN / 2);    // wave's cout group
   const int M = Nb * OH * OW;
-  const int mtiles = (M + 127) / 128;
   const int nwg = gridDim.x;
   int wgid = blockIdx.x;
   {
