@@ -733,7 +733,7 @@ torch::Tensor gemm_wgrad(torch::Tensor a, torch::Tensor b) {
                      (const short*)b.data_ptr(), nullptr,
                      ws.data_ptr<float>(), M, Na, Nb, nsplit);
   long long n = (long long)Na * Nb;
-  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid(n)), dim3(256), 0,
+  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid((n + 3) / 4)), dim3(256), 0,
                      cur_stream(), ws.data_ptr<float>(), (short*)c.data_ptr(),
                      n, nsplit);
   return c;
@@ -765,7 +765,7 @@ torch::Tensor gemm_dgrad(torch::Tensor a, torch::Tensor b) {
                      (const short*)b.data_ptr(), nullptr, ws.data_ptr<float>(),
                      M, R, K, nsplit);
   long long n = (long long)M * K;
-  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid(n)), dim3(256), 0,
+  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid((n + 3) / 4)), dim3(256), 0,
                      cur_stream(), ws.data_ptr<float>(), (short*)c.data_ptr(),
                      n, nsplit);
   return c;
@@ -790,7 +790,7 @@ static torch::Tensor gemm_nt_splitk(torch::Tensor A, torch::Tensor B, int M,
                      (const short*)A.data_ptr(), (const short*)B.data_ptr(),
                      nullptr, ws.data_ptr<float>(), M, N, K, nsplit);
   long long n = (long long)M * N;
-  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid(n)), dim3(256), 0,
+  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid((n + 3) / 4)), dim3(256), 0,
                      cur_stream(), ws.data_ptr<float>(), (short*)c.data_ptr(),
                      n, nsplit);
   return c;
@@ -888,7 +888,7 @@ torch::Tensor conv_nhwc_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
                        (const short*)zero_page.data_ptr(), Nb, H, Wd, Cin,
                        Cout, OH, OW, KH, KW, (int)stride, (int)pad, ns);
     long long n = (long long)M * Cout;
-    hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid(n)), dim3(256), 0,
+    hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid((n + 3) / 4)), dim3(256), 0,
                        cur_stream(), ws.data_ptr<float>(),
                        (short*)out.data_ptr(), n, ns);
     return out;
@@ -1009,10 +1009,10 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                      Hkv, (float)scale, causal ? 1 : 0, q.stride(0),
                      q.stride(1), q.stride(2), k.stride(0), k.stride(1),
                      k.stride(2));
-  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid(nkv)), dim3(256), 0,
+  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid((nkv + 3) / 4)), dim3(256), 0,
                      cur_stream(), wsk.data_ptr<float>(),
                      (short*)dk.data_ptr(), nkv, group);
-  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid(nkv)), dim3(256), 0,
+  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid((nkv + 3) / 4)), dim3(256), 0,
                      cur_stream(), wsv.data_ptr<float>(),
                      (short*)dv.data_ptr(), nkv, group);
   return {dq, dk, dv};
@@ -1130,7 +1130,7 @@ torch::Tensor conv_nhwc_wgrad(torch::Tensor dy, torch::Tensor x, int64_t kh,
                      cin_log2, Cout, OH, OW, (int)kh, (int)kw, (int)stride,
                      (int)pad, ns, magic_ohow, magic_ow);
   long long n = (long long)Cout * NbCols;
-  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid(n)), dim3(256), 0,
+  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(ew_grid((n + 3) / 4)), dim3(256), 0,
                      cur_stream(), ws.data_ptr<float>(), (short*)dw.data_ptr(),
                      n, ns);
   return dw;

@@ -1081,12 +1081,28 @@ gemm_wgrad_bf16_kernel(const short* __restrict__ A, const short* __restrict__ B,
 extern "C" __global__ void __launch_bounds__(256)
 splitk_reduce_kernel(const float* __restrict__ ws, short* __restrict__ out,
                      long long n, int nsplit) {
-  long long i = (long long)blockIdx.x * 256 + threadIdx.x;
-  long long stride = (long long)gridDim.x * 256;
-  for (; i < n; i += stride) {
-    float acc = 0.f;
-    for (int s = 0; s < nsplit; ++s) acc += ws[(long long)s * n + i];
-    out[i] = f32_to_bf16(acc);
+  // 4-wide: f32x4 slab loads (all slabs' loads in flight) + short4 store;
+  // the scalar version measured 6.7 ms/round on the llama config
+  long long i4 = ((long long)blockIdx.x * 256 + threadIdx.x) * 4;
+  long long stride = (long long)gridDim.x * 256 * 4;
+  for (; i4 + 4 <= n; i4 += stride) {
+    f32x4 acc = *(const f32x4*)(ws + i4);
+    for (int s = 1; s < nsplit; ++s) {
+      f32x4 v = *(const f32x4*)(ws + (long long)s * n + i4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[j] += v[j];
+    }
+    short o4[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) o4[j] = f32_to_bf16(acc[j]);
+    *(unsigned long long*)(out + i4) = *(unsigned long long*)o4;
+  }
+  if (i4 < n) {
+    for (; i4 < n; ++i4) {
+      float acc = 0.f;
+      for (int s = 0; s < nsplit; ++s) acc += ws[(long long)s * n + i4];
+      out[i4] = f32_to_bf16(acc);
+    }
   }
 }
 
