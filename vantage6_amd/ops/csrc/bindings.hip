@@ -569,7 +569,7 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor w,
   // 1.4 TB/s on the partials pass)
   long long cover = (R + rows_pb - 1) / rows_pb;
   int nblocks = (int)std::min<long long>(
-      cover, std::max<long long>(128, cover / 8));
+      cover, std::max<long long>(128, cover / 16));
   nblocks = std::min(nblocks, 384);
   torch::Tensor part;
   if (acc.has_value()) {
@@ -633,7 +633,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor mask,
   int rows_pb = bn_rows_pb(C);
   long long cover = (R + rows_pb - 1) / rows_pb;
   int nblocks = (int)std::min<long long>(
-      cover, std::max<long long>(128, cover / 8));
+      cover, std::max<long long>(128, cover / 16));
   nblocks = std::min(nblocks, 384);
   torch::Tensor part;
   if (acc.has_value()) {
