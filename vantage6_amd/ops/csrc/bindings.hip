@@ -72,6 +72,8 @@ extern "C" __global__ void gemm_nt_bf16_v4_kernel(const short*, const short*,
                                                   short*, int, int, int);
 extern "C" __global__ void gemm_nt_bf16_v5_kernel(const short*, const short*,
                                                   short*, int, int, int);
+extern "C" __global__ void gemm_nt_bf16_v5b_kernel(const short*, const short*,
+                                                   short*, int, int, int);
 extern "C" __global__ void gemm_nt_bf16_v6_kernel(const short*, const short*,
                                                   short*, int, int, int);
 extern "C" __global__ void gemm_nt_bf16_v7_kernel(const short*, const short*,
@@ -822,7 +824,8 @@ torch::Tensor gemm_nt_probe(torch::Tensor a, torch::Tensor b, int variant,
     return c;
   }
   TORCH_CHECK(M % 128 == 0 && N % 128 == 0 && K % 64 == 0, "shape");
-  hipLaunchKernelGGL(gemm_nt_bf16_v5_kernel, dim3((M / 128) * (N / 128)),
+  auto k5 = variant == 55 ? gemm_nt_bf16_v5b_kernel : gemm_nt_bf16_v5_kernel;
+  hipLaunchKernelGGL(k5, dim3((M / 128) * (N / 128)),
                      dim3(256), 0, cur_stream(), (const short*)a.data_ptr(),
                      (const short*)b.data_ptr(), (short*)c.data_ptr(), M, N, K);
   return c;

@@ -563,6 +563,106 @@ gemm_nt_bf16_v5_kernel(const short* __restrict__ A, const short* __restrict__ B,
     }
 }
 
+// v5b — v5's 128² tile with the v7 barrier discipline: counted
+// s_waitcnt vmcnt(8) + raw s_barrier instead of __syncthreads' vmcnt(0)
+// drain. The drain is a fixed per-K-step cost, so this targets SHALLOW-K
+// shapes (BERT's K=768 classes) where it is the largest fraction.
+extern "C" __global__ void __launch_bounds__(NTH)
+gemm_nt_bf16_v5b_kernel(const short* __restrict__ A, const short* __restrict__ B,
+                        short* __restrict__ C, int M, int N, int K) {
+  __shared__ short lds[2 * V2_BUF_SLOTS * 8];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = (wave >> 1) * 64;
+  const int wc = (wave & 1) * 64;
+  const int nwg = gridDim.x;
+  int wgid = blockIdx.x;
+  {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = wgid & 7, idx = wgid >> 3;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int ntile_n = N / BN;
+  const int row0 = (wgid / ntile_n) * BM;
+  const int col0 = (wgid % ntile_n) * BN;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int row_in_piece = lane >> 3;
+  const int nk = K / BK;
+  auto stage = [&](int buf, int kt) {
+    const long long k0 = (long long)kt * BK;
+    char* base = (char*)lds + buf * 32768;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int pa = wave * 4 + i;
+      const int src_koff = (((lane & 7) ^ (pa * 4 + (lane >> 4))) & 7) * 8;
+      const short* srcA =
+          A + (long long)(row0 + pa * 8 + row_in_piece) * K + k0 + src_koff;
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcA, (lds_uint*)(base + pa * 1024), 16, 0, 0);
+      const short* srcB =
+          B + (long long)(col0 + pa * 8 + row_in_piece) * K + k0 + src_koff;
+      __builtin_amdgcn_global_load_lds(
+          (global_uint*)srcB, (lds_uint*)(base + 16384 + pa * 1024), 16, 0, 0);
+    }
+  };
+  stage(0, 0);
+  for (int kt = 0; kt < nk; ++kt) {
+    int cur = kt & 1;
+    if (kt + 1 < nk) {
+      stage(cur ^ 1, kt + 1);
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    char* base = (char*)lds + cur * 32768;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      const int ks = kk * 4 + (lane >> 4);
+      bf16x8_t a_frag[4], b_frag[4];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        int row = wr + mi * 16 + (lane & 15);
+        a_frag[mi] = *(const bf16x8_t*)(base + row * 128 +
+                                        ((ks ^ (row >> 1)) & 7) * 16);
+      }
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int row = wc + ni * 16 + (lane & 15);
+        b_frag[ni] = *(const bf16x8_t*)(base + 16384 + row * 128 +
+                                        ((ks ^ (row >> 1)) & 7) * 16);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+  const int crow_base = row0 + wr + (lane >> 4) * 4;
+  const int ccol_base = col0 + wc + (lane & 15);
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = crow_base + mi * 16 + r;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        C[(long long)row * N + ccol_base + ni * 16] = f32_to_bf16(acc[mi][ni][r]);
+    }
+}
+
 // v6 — 256×256 tile, 8 waves, BK=64, v4-style full-line glds + src XOR
 // swizzle, single LDS buffer (64 KB → 2 blocks/CU). Twice the
 // compute-per-staged-byte of the 128² tiles.
