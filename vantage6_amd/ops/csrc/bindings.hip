@@ -47,6 +47,9 @@ extern "C" __global__ void rmsnorm_bwd_kernel(const short*, const short*,
                                               const short*, const float*, short*,
                                               float*, long long, int);
 extern "C" __global__ void colsum_ws_kernel(const float*, short*, int, int);
+extern "C" __global__ void layernorm_bwd_wave2_kernel(
+    const short*, const short*, const short*, const float*, const float*,
+    short*, float*, float*, long long, int);
 extern "C" __global__ void layernorm_bwd_wave_kernel(const short*, const short*,
                                                      const short*, const float*,
                                                      const float*, short*,
@@ -464,7 +467,9 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   }();
   if (use_wave && H <= 2048) {  // row-per-wave, barrier-free reductions
     int wgrid = (int)std::min<long long>((rows + 3) / 4, 2048);
-    hipLaunchKernelGGL(layernorm_bwd_wave_kernel, dim3(wgrid), dim3(256), 0,
+    auto kw = H <= 1024 ? layernorm_bwd_wave2_kernel
+                        : layernorm_bwd_wave_kernel;
+    hipLaunchKernelGGL(kw, dim3(wgrid), dim3(256), 0,
                        cur_stream(), (const short*)dy.data_ptr(),
                        (const short*)x.data_ptr(), (const short*)w.data_ptr(),
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),

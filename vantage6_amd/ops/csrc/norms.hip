@@ -277,26 +277,25 @@ DEVINL float wave_reduce_sum(float v) {
   return v;
 }
 
-extern "C" __global__ void __launch_bounds__(NTHREADS)
-layernorm_bwd_wave_kernel(const short* __restrict__ dy,
-                          const short* __restrict__ x,
-                          const short* __restrict__ w,
-                          const float* __restrict__ mean,
-                          const float* __restrict__ rstd,
-                          short* __restrict__ dx, float* __restrict__ dw_ws,
-                          float* __restrict__ db_ws, long long nrows, int H) {
+template <int WCH>
+__device__ __forceinline__ void layernorm_bwd_wave_body(
+    const short* __restrict__ dy, const short* __restrict__ x,
+    const short* __restrict__ w, const float* __restrict__ mean,
+    const float* __restrict__ rstd, short* __restrict__ dx,
+    float* __restrict__ dw_ws, float* __restrict__ db_ws, long long nrows,
+    int H) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  float dw_acc[WMAXCH * 8];
-  float db_acc[WMAXCH * 8];
+  float dw_acc[WCH * 8];
+  float db_acc[WCH * 8];
 #pragma unroll
-  for (int c = 0; c < WMAXCH * 8; ++c) { dw_acc[c] = 0.f; db_acc[c] = 0.f; }
+  for (int c = 0; c < WCH * 8; ++c) { dw_acc[c] = 0.f; db_acc[c] = 0.f; }
   // w is row-invariant: load it ONCE; x/dy are register-cached across the
-  // two passes (the re-reads + serial load→reduce→load chain made the
-  // original version latency-bound at ~0.5 TB/s)
-  s16x8 wreg[WMAXCH];
+  // two passes; WCH is a compile-time chunk count so H=768 doesn't pay
+  // the H=2048 register footprint
+  s16x8 wreg[WCH];
 #pragma unroll
-  for (int c = 0; c < WMAXCH; ++c) {
+  for (int c = 0; c < WCH; ++c) {
     int i = lane * 8 + c * 512;
     wreg[c] = i < H ? *(const s16x8*)(w + i) : s16x8{0, 0, 0, 0, 0, 0, 0, 0};
   }
@@ -306,9 +305,9 @@ layernorm_bwd_wave_kernel(const short* __restrict__ dy,
     const short* dyr = dy + row * (long long)H;
     short* dxr = dx + row * (long long)H;
     float mu = mean[row], rs = rstd[row];
-    s16x8 xv[WMAXCH], dv[WMAXCH];
+    s16x8 xv[WCH], dv[WCH];
 #pragma unroll
-    for (int c = 0; c < WMAXCH; ++c) {  // all loads in flight together
+    for (int c = 0; c < WCH; ++c) {  // all loads in flight together
       int i = lane * 8 + c * 512;
       if (i < H) {
         xv[c] = *(const s16x8*)(xr + i);
@@ -316,8 +315,8 @@ layernorm_bwd_wave_kernel(const short* __restrict__ dy,
       }
     }
     float c1 = 0.f, c2 = 0.f;
-#pragma unroll WMAXCH
-    for (int c = 0; c < WMAXCH; ++c) {
+#pragma unroll WCH
+    for (int c = 0; c < WCH; ++c) {
       int i = lane * 8 + c * 512;
       if (i >= H) break;
 #pragma unroll
@@ -330,8 +329,8 @@ layernorm_bwd_wave_kernel(const short* __restrict__ dy,
     }
     c1 = wave_reduce_sum(c1) / H;
     c2 = wave_reduce_sum(c2) / H;
-#pragma unroll WMAXCH
-    for (int c = 0; c < WMAXCH; ++c) {
+#pragma unroll WCH
+    for (int c = 0; c < WCH; ++c) {
       int i = lane * 8 + c * 512;
       if (i >= H) break;
       s16x8 o;
@@ -350,8 +349,8 @@ layernorm_bwd_wave_kernel(const short* __restrict__ dy,
   // block-level LDS combine first: one global atomic set per BLOCK
   __shared__ float comb[2 * 2048];
   if (wave == 0) {
-#pragma unroll WMAXCH
-    for (int c = 0; c < WMAXCH; ++c) {
+#pragma unroll WCH
+    for (int c = 0; c < WCH; ++c) {
       int i = lane * 8 + c * 512;
       if (i >= H) break;
 #pragma unroll
@@ -365,8 +364,8 @@ layernorm_bwd_wave_kernel(const short* __restrict__ dy,
 #pragma unroll
   for (int wv = 1; wv < 4; ++wv) {
     if (wave == wv) {
-#pragma unroll WMAXCH
-      for (int c = 0; c < WMAXCH; ++c) {
+#pragma unroll WCH
+      for (int c = 0; c < WCH; ++c) {
         int i = lane * 8 + c * 512;
         if (i >= H) break;
 #pragma unroll
@@ -381,8 +380,8 @@ layernorm_bwd_wave_kernel(const short* __restrict__ dy,
   float* dwr = dw_ws + (long long)(blockIdx.x & 15) * H;
   float* dbr = db_ws + (long long)(blockIdx.x & 15) * H;
   if (wave == 0) {
-#pragma unroll WMAXCH
-    for (int c = 0; c < WMAXCH; ++c) {
+#pragma unroll WCH
+    for (int c = 0; c < WCH; ++c) {
       int i = lane * 8 + c * 512;
       if (i >= H) break;
 #pragma unroll
@@ -392,6 +391,28 @@ layernorm_bwd_wave_kernel(const short* __restrict__ dy,
       }
     }
   }
+}
+
+extern "C" __global__ void __launch_bounds__(NTHREADS)
+layernorm_bwd_wave_kernel(const short* __restrict__ dy,
+                          const short* __restrict__ x,
+                          const short* __restrict__ w,
+                          const float* __restrict__ mean,
+                          const float* __restrict__ rstd,
+                          short* __restrict__ dx, float* __restrict__ dw_ws,
+                          float* __restrict__ db_ws, long long nrows, int H) {
+  layernorm_bwd_wave_body<4>(dy, x, w, mean, rstd, dx, dw_ws, db_ws, nrows, H);
+}
+
+extern "C" __global__ void __launch_bounds__(NTHREADS)
+layernorm_bwd_wave2_kernel(const short* __restrict__ dy,
+                           const short* __restrict__ x,
+                           const short* __restrict__ w,
+                           const float* __restrict__ mean,
+                           const float* __restrict__ rstd,
+                           short* __restrict__ dx, float* __restrict__ dw_ws,
+                           float* __restrict__ db_ws, long long nrows, int H) {
+  layernorm_bwd_wave_body<2>(dy, x, w, mean, rstd, dx, dw_ws, db_ws, nrows, H);
 }
 
 extern "C" __global__ void __launch_bounds__(NTHREADS)
