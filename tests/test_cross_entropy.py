@@ -1,0 +1,66 @@
+"""Fused bf16 cross-entropy vs fp32 torch (mean / per-row / ignore_index)."""
+import pytest
+import torch
+import torch.nn.functional as F
+
+from vantage6_amd.ops.cross_entropy import cross_entropy
+
+
+def test_cpu_fallback_matches_torch():
+    torch.manual_seed(0)
+    logits = torch.randn(8, 4, 128)
+    t = torch.randint(0, 128, (8, 4))
+    t[0, 0] = -100
+    ref = F.cross_entropy(logits.reshape(-1, 128), t.reshape(-1))
+    assert torch.allclose(cross_entropy(logits, t), ref)
+
+
+@pytest.mark.gpu
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("R,V,reduction,with_ignore", [
+    (64, 8192, "mean", False),
+    (64, 8192, "mean", True),
+    (128, 128256, "mean", False),   # llama vocab
+    (64, 8192, "none", True),
+])
+def test_fused_ce_matches_fp32(R, V, reduction, with_ignore):
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    torch.manual_seed(1)
+    dev = "cuda:0"
+    logits = (torch.randn(R, V, device=dev) * 3).to(torch.bfloat16)
+    t = torch.randint(0, V, (R,), device=dev)
+    if with_ignore:
+        t[::7] = -100
+
+    la = logits.clone().requires_grad_(True)
+    out = cross_entropy(la, t, reduction=reduction)
+    lr = logits.float().requires_grad_(True)
+    ref = F.cross_entropy(lr, t, reduction=reduction)
+    if reduction == "none":
+        g = torch.rand_like(ref)
+        out.backward(g)
+        ref.backward(g)
+        assert (out - ref).abs().max().item() < 2e-2
+    else:
+        out.backward()
+        ref.backward()
+        assert abs(out.item() - ref.item()) < 2e-2
+    gs = lr.grad.abs().max().item()
+    err = (la.grad.float() - lr.grad).abs().max().item()
+    assert err <= 0.05 * gs + 1e-4, (err, gs)
+
+
+@pytest.mark.gpu
+@pytest.mark.timeout(300)
+def test_fused_ce_ignored_rows_zero_grad():
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    torch.manual_seed(2)
+    logits = torch.randn(16, 8192, device="cuda",
+                         dtype=torch.bfloat16).requires_grad_(True)
+    t = torch.randint(0, 8192, (16,), device="cuda")
+    t[3] = -100
+    cross_entropy(logits, t).backward()
+    assert logits.grad[3].abs().max().item() == 0.0
+    assert logits.grad[4].abs().max().item() > 0.0

@@ -35,12 +35,11 @@ def model_loss(name: str) -> Callable:
                 "llama-3-1b"):
 
         def lm_loss(model, batch):
+            from ..ops.cross_entropy import cross_entropy
+
             ids = batch["input_ids"]
             logits = model(ids[:, :-1])
-            return F.cross_entropy(
-                logits.float().reshape(-1, logits.shape[-1]),
-                ids[:, 1:].reshape(-1),
-            )
+            return cross_entropy(logits, ids[:, 1:])
 
         return lm_loss
 
@@ -58,12 +57,11 @@ def model_loss_per_sample(name: str) -> Callable:
                 "llama-3-1b"):
 
         def lm_loss_ps(model, batch):
+            from ..ops.cross_entropy import cross_entropy
+
             ids = batch["input_ids"]
             logits = model(ids[:, :-1])
-            losses = F.cross_entropy(
-                logits.float().reshape(-1, logits.shape[-1]),
-                ids[:, 1:].reshape(-1), reduction="none",
-            )
+            losses = cross_entropy(logits, ids[:, 1:], reduction="none")
             return losses.reshape(ids.shape[0], -1).mean(-1)
 
         return lm_loss_ps

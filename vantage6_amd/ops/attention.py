@@ -32,18 +32,6 @@ def _is_dense(t: torch.Tensor) -> bool:
     return True
 
 
-def _is_dense(t: torch.Tensor) -> bool:
-    """Non-overlapping dense: empty_like preserves these strides. Slice
-    views of a bigger tensor are NOT dense."""
-    dims = sorted((st, sz) for sz, st in zip(t.shape, t.stride()) if sz != 1)
-    expect = 1
-    for st, sz in dims:
-        if st != expect:
-            return False
-        expect *= sz
-    return True
-
-
 class _AttnFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, causal, scale):

@@ -55,6 +55,11 @@ extern "C" __global__ void layernorm_bwd_wave_kernel(const short*, const short*,
                                                      const float*, short*,
                                                      float*, float*, long long,
                                                      int);
+extern "C" __global__ void ce_fwd_kernel(const short*, const long long*,
+                                         float*, float*, long long, long long);
+extern "C" __global__ void ce_bwd_kernel(const short*, const long long*,
+                                         const float*, const float*, short*,
+                                         long long, long long);
 extern "C" __global__ void rmsnorm_bwd_wave_kernel(const short*, const short*,
                                                    const short*, const float*,
                                                    short*, float*, long long,
@@ -514,6 +519,39 @@ static bool use_wave_rms() {
     return e ? atoi(e) != 0 : true;
   }();
   return v;
+}
+
+// fused LM-head cross-entropy: fwd returns {loss[R] fp32, lse[R] fp32}.
+std::vector<torch::Tensor> ce_fwd(torch::Tensor x, torch::Tensor t,
+                                  long long ignore_index) {
+  CHECK_HIP(x, torch::kBFloat16);
+  CHECK_HIP(t, torch::kLong);
+  TORCH_CHECK(x.dim() == 2, "ce_fwd: x must be [R, V]");
+  long long R = x.size(0), V = x.size(1);
+  TORCH_CHECK(V % 8 == 0, "ce_fwd: V must be a multiple of 8");
+  TORCH_CHECK(t.numel() == R, "ce_fwd: targets/rows mismatch");
+  auto lse = torch::empty({R}, x.options().dtype(torch::kFloat32));
+  auto loss = torch::empty({R}, x.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(ce_fwd_kernel, dim3((unsigned)R), dim3(256), 0,
+                     cur_stream(), (const short*)x.data_ptr(),
+                     (const long long*)t.data_ptr(), lse.data_ptr<float>(),
+                     loss.data_ptr<float>(), V, ignore_index);
+  return {loss, lse};
+}
+
+torch::Tensor ce_bwd(torch::Tensor x, torch::Tensor t, torch::Tensor lse,
+                     torch::Tensor dlv, long long ignore_index) {
+  CHECK_HIP(x, torch::kBFloat16);
+  CHECK_HIP(dlv, torch::kFloat32);
+  long long R = x.size(0), V = x.size(1);
+  auto dx = torch::empty_like(x);
+  int gx = (int)std::min<long long>((V / 8 + 255) / 256, 4096);
+  hipLaunchKernelGGL(ce_bwd_kernel, dim3(gx, (unsigned)R), dim3(256), 0,
+                     cur_stream(), (const short*)x.data_ptr(),
+                     (const long long*)t.data_ptr(), lse.data_ptr<float>(),
+                     dlv.data_ptr<float>(), (short*)dx.data_ptr(), V,
+                     ignore_index);
+  return dx;
 }
 
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
@@ -1358,6 +1396,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("grad_norm2_partials", &grad_norm2_partials, "block partial sum of g^2");
   m.def("rope_apply", &rope_apply);
   m.def("colsum_bf16", &colsum_bf16, "column sum of a bf16 matrix (bias grad)");
+  m.def("ce_fwd", &ce_fwd, "fused bf16 cross-entropy fwd: {loss[R], lse[R]}");
+  m.def("ce_bwd", &ce_bwd, "fused bf16 cross-entropy bwd: dlogits");
   m.def("emb_grad_bf16", &emb_grad_bf16, "graph-safe embedding backward");
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);

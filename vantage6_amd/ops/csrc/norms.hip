@@ -548,3 +548,74 @@ colsum_bf16_kernel(const short* __restrict__ dy, float* __restrict__ ws,
 #pragma unroll
   for (int j = 0; j < 8; ++j) atomicAdd(&slice[col0 + j], acc[j]);
 }
+
+// -------------------------------------------- fused cross-entropy (LM head)
+// One block per row, two HBM-light passes (the second re-reads the row from
+// L2 — a 256 KB bf16 row stays resident). Writes lse[r] and
+// loss[r] = lse - x[target] (0 for ignored rows). Replaces the torch
+// fp32 softmax+nll chain on a 128k-vocab LM head, whose fp32 logits
+// materialisation alone moves 2x the row's bf16 bytes (profile: ~5 ms of
+// cunn_SoftMax* + bfloat16_copy per llama round). V must be % 8.
+extern "C" __global__ void __launch_bounds__(256)
+ce_fwd_kernel(const short* __restrict__ x, const long long* __restrict__ t,
+              float* __restrict__ lse, float* __restrict__ loss,
+              long long V, long long ignore_index) {
+  __shared__ float red[4];
+  int row = blockIdx.x;
+  const short* xr = x + (long long)row * V;
+  float m = -1e30f;
+  for (long long i = threadIdx.x * 8LL; i < V; i += 256 * 8) {
+    s16x8 v8 = *(const s16x8*)(xr + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) m = fmaxf(m, bf16_to_f32(v8[j]));
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) m = fmaxf(m, __shfl_xor(m, off, 64));
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = m;
+  __syncthreads();
+  m = fmaxf(fmaxf(red[0], red[1]), fmaxf(red[2], red[3]));
+  __syncthreads();
+  float s = 0.f;
+  for (long long i = threadIdx.x * 8LL; i < V; i += 256 * 8) {
+    s16x8 v8 = *(const s16x8*)(xr + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) s += __expf(bf16_to_f32(v8[j]) - m);
+  }
+  s = block_reduce_sum(s, red);
+  if (threadIdx.x == 0) {
+    float l = m + __logf(s);
+    lse[row] = l;
+    long long tr = t[row];
+    loss[row] = (tr == ignore_index) ? 0.f : l - bf16_to_f32(xr[tr]);
+  }
+}
+
+// dX[r][v] = dloss[r] * (exp(x - lse) - 1[v==target]); bf16 out, one pass.
+// dloss[r] is the autograd grad of the per-row loss vector (0 for ignored
+// rows, so those rows get exact zero grads). Grid (v-chunks, R).
+extern "C" __global__ void __launch_bounds__(256)
+ce_bwd_kernel(const short* __restrict__ x, const long long* __restrict__ t,
+              const float* __restrict__ lse, const float* __restrict__ dlv,
+              short* __restrict__ dx, long long V, long long ignore_index) {
+  int row = blockIdx.y;
+  float g = dlv[row], l = lse[row];
+  long long tr = t[row];
+  // an ignored row's loss is a constant 0 — its grad is exactly 0 even
+  // though autograd hands every row d(sum)/drow = 1
+  if (tr == ignore_index) g = 0.f;
+  const short* xr = x + (long long)row * V;
+  short* dr = dx + (long long)row * V;
+  long long stride = (long long)gridDim.x * 256 * 8;
+  for (long long i = ((long long)blockIdx.x * 256 + threadIdx.x) * 8;
+       i < V; i += stride) {
+    s16x8 v8 = *(const s16x8*)(xr + i);
+    s16x8 o8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float p = __expf(bf16_to_f32(v8[j]) - l);
+      if (i + j == tr) p -= 1.f;
+      o8[j] = f32_to_bf16(p * g);
+    }
+    *(s16x8*)(dr + i) = o8;
+  }
+}
