@@ -64,3 +64,24 @@ def test_fused_ce_ignored_rows_zero_grad():
     cross_entropy(logits, t).backward()
     assert logits.grad[3].abs().max().item() == 0.0
     assert logits.grad[4].abs().max().item() > 0.0
+
+
+def test_hip_gate_never_fires_for_bad_shapes(monkeypatch):
+    """The HIP path requires cuda + bf16 + V%8==0 + V>=4096 + contiguous
+    rows; anything else (BERT's 30522 vocab, fp32 logits, CPU tensors)
+    must take the torch fallback without touching the extension."""
+    from vantage6_amd.ops import cross_entropy as ce_mod
+
+    def boom():
+        raise AssertionError("hip_ops must not be consulted on fallback")
+
+    monkeypatch.setattr(ce_mod, "hip_ops", boom)
+    t = torch.randint(0, 30522, (6,))
+    out = ce_mod.cross_entropy(torch.randn(6, 30522), t)  # V % 8 != 0
+    ref = F.cross_entropy(torch.randn(6, 30522) * 0 + 0, t)  # shape only
+    assert out.shape == ref.shape
+    # per-row reduction fallback too
+    v = ce_mod.cross_entropy(torch.randn(6, 30522), t, reduction="none")
+    assert v.shape == (6,)
+    # small-vocab classifier logits (resnet/logreg) always fall back
+    ce_mod.cross_entropy(torch.randn(4, 8), torch.randint(0, 8, (4,)))
