@@ -434,3 +434,80 @@ def test_multi_database_task(tmp_path):
         assert res == {"n1": 2, "n2": 3, "sum2": 60.0}
     finally:
         del sys.modules["algos_twodb"]
+
+
+def test_federated_correlation_matrix(tmp_path):
+    """Federated Pearson correlation equals pandas .corr() on the pooled
+    complete-case data."""
+    import numpy as np
+    import pandas as pd
+
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+
+    rng = np.random.default_rng(23)
+    frames, paths = [], []
+    for i in range(3):
+        n = 40 + 15 * i
+        a = rng.normal(0, 1, n)
+        df = pd.DataFrame({
+            "a": a,
+            "b": 0.6 * a + rng.normal(0, 0.8, n),
+            "c": rng.normal(4, 2, n),
+        })
+        df.loc[df.index[:2], "b"] = np.nan  # complete-case handling
+        p = tmp_path / f"corr{i}.csv"
+        df.to_csv(p, index=False)
+        frames.append(df)
+        paths.append(p)
+    pooled = pd.concat(frames)[["a", "b", "c"]].dropna()
+    ref = pooled.corr().to_numpy()
+    net = DevNetwork(n_stations=3, databases=[
+        [DatabaseConfig(label="default", uri=str(p), type="csv")] for p in paths])
+    with net:
+        task = net.client.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],
+            name="corr", image="algorithms.correlation",
+            input_={"method": "central_correlation",
+                    "kwargs": {"columns": ["a", "b", "c"]}})
+        (res,) = net.client.wait_for_results(task["id"])
+    assert res["n"] == len(pooled)
+    got = np.array(res["matrix"], dtype=float)
+    assert np.allclose(got, ref, atol=1e-9)
+
+
+def test_federated_quantiles(tmp_path):
+    """Bisection quantiles equal numpy's method='lower' order statistics
+    on the pooled data; only counts ever leave a station."""
+    import numpy as np
+    import pandas as pd
+
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+
+    rng = np.random.default_rng(31)
+    vals, paths = [], []
+    for i in range(2):
+        v = rng.normal(10, 4, 60 + 20 * i)
+        p = tmp_path / f"q{i}.csv"
+        pd.DataFrame({"v": v}).to_csv(p, index=False)
+        vals.append(v)
+        paths.append(p)
+    pooled = np.concatenate(vals)
+    net = DevNetwork(n_stations=2, databases=[
+        [DatabaseConfig(label="default", uri=str(p), type="csv")] for p in paths])
+    qs = [0.1, 0.5, 0.9]
+    with net:
+        task = net.client.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],
+            name="quantiles", image="algorithms.quantiles",
+            input_={"method": "central_quantiles",
+                    "kwargs": {"column": "v", "q": qs}})
+        (res,) = net.client.wait_for_results(task["id"])
+    assert res["n"] == len(pooled)
+    span = pooled.max() - pooled.min()
+    for frac in qs:
+        ref = np.quantile(pooled, frac, method="lower")
+        assert abs(res["quantiles"][str(frac)] - ref) <= 2e-6 * span + 1e-12, frac
