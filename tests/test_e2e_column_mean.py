@@ -511,3 +511,55 @@ def test_federated_quantiles(tmp_path):
     for frac in qs:
         ref = np.quantile(pooled, frac, method="lower")
         assert abs(res["quantiles"][str(frac)] - ref) <= 2e-6 * span + 1e-12, frac
+
+
+def test_federated_pca_matches_sklearn(tmp_path):
+    """Federated PCA (pooled covariance + central eigh) equals sklearn
+    PCA fit on the pooled rows."""
+    import numpy as np
+    import pandas as pd
+    from sklearn.decomposition import PCA
+
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+
+    rng = np.random.default_rng(47)
+    frames, paths = [], []
+    base = rng.normal(0, 1, (0, 4))
+    for i in range(3):
+        n = 50 + 10 * i
+        z = rng.normal(0, 1, (n, 2))
+        x = np.column_stack([
+            z[:, 0] * 3.0,
+            z[:, 0] * 1.5 + z[:, 1],
+            rng.normal(0, 0.5, n),
+            z[:, 1] * 2.0 + rng.normal(0, 0.2, n),
+        ])
+        df = pd.DataFrame(x, columns=list("abcd"))
+        p = tmp_path / f"pca{i}.csv"
+        df.to_csv(p, index=False)
+        frames.append(df)
+        paths.append(p)
+    pooled = pd.concat(frames).to_numpy()
+    ref = PCA(n_components=3).fit(pooled)
+    net = DevNetwork(n_stations=3, databases=[
+        [DatabaseConfig(label="default", uri=str(p), type="csv")] for p in paths])
+    with net:
+        task = net.client.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],
+            name="pca", image="algorithms.pca",
+            input_={"method": "central_pca",
+                    "kwargs": {"columns": list("abcd"), "n_components": 3}})
+        (res,) = net.client.wait_for_results(task["id"])
+    assert res["n"] == len(pooled)
+    assert np.allclose(res["explained_variance"],
+                       ref.explained_variance_, rtol=1e-8)
+    assert np.allclose(res["explained_variance_ratio"],
+                       ref.explained_variance_ratio_, rtol=1e-8)
+    got = np.array(res["components"])
+    # same subspace and same svd_flip sign convention
+    assert np.allclose(np.abs(got), np.abs(ref.components_), atol=1e-8)
+    for i in range(3):
+        assert np.allclose(got[i], ref.components_[i], atol=1e-8) or \
+            np.allclose(got[i], -ref.components_[i], atol=1e-8)
