@@ -563,3 +563,46 @@ def test_federated_pca_matches_sklearn(tmp_path):
     for i in range(3):
         assert np.allclose(got[i], ref.components_[i], atol=1e-8) or \
             np.allclose(got[i], -ref.components_[i], atol=1e-8)
+
+
+def test_federated_ttest_matches_scipy(tmp_path):
+    """Federated Welch and Student t-tests equal scipy on pooled rows."""
+    import numpy as np
+    import pandas as pd
+    from scipy import stats
+
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+
+    rng = np.random.default_rng(59)
+    frames, paths = [], []
+    for i in range(2):
+        n = 45 + 20 * i
+        df = pd.DataFrame({
+            "y": np.concatenate([rng.normal(5.0, 2.0, n),
+                                 rng.normal(5.8, 3.0, n)]),
+            "arm": ["ctl"] * n + ["trt"] * n,
+        })
+        p = tmp_path / f"tt{i}.csv"
+        df.to_csv(p, index=False)
+        frames.append(df)
+        paths.append(p)
+    pooled = pd.concat(frames)
+    ctl = pooled[pooled.arm == "ctl"].y.to_numpy()
+    trt = pooled[pooled.arm == "trt"].y.to_numpy()
+    net = DevNetwork(n_stations=2, databases=[
+        [DatabaseConfig(label="default", uri=str(p), type="csv")] for p in paths])
+    with net:
+        for equal_var in (False, True):
+            task = net.client.task.create(
+                collaboration=net.collaboration["id"],
+                organizations=[net.org_ids[0]],
+                name="ttest", image="algorithms.ttest",
+                input_={"method": "central_ttest",
+                        "kwargs": {"value": "y", "group": "arm",
+                                   "equal_var": equal_var}})
+            (res,) = net.client.wait_for_results(task["id"])
+            ref = stats.ttest_ind(ctl, trt, equal_var=equal_var)
+            assert res["t"] == pytest.approx(ref.statistic, rel=1e-10)
+            assert res["p_value"] == pytest.approx(ref.pvalue, rel=1e-9)
+            assert res["groups"]["ctl"]["n"] == len(ctl)
