@@ -54,3 +54,33 @@ def test_hybrid_encryption_roundtrip(blob):
                               bits=1024)
     enc = _CRYPTOR.encrypt_bytes_to_str(blob, _CRYPTOR.public_key_str)
     assert _CRYPTOR.decrypt_str_to_bytes(enc) == blob
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    st.lists(st.floats(min_value=-1e3, max_value=1e3, allow_nan=False,
+                       width=32), min_size=1, max_size=32),
+    st.integers(min_value=2, max_value=5),
+    st.integers(min_value=0, max_value=2**31),
+)
+def test_secagg_masks_cancel_exactly(vals, world, seed):
+    """Pairwise int64 masks cancel bit-exactly under summation for any
+    world size / payload / round seed, and every single rank's masked
+    vector differs from its unmasked quantization (it reveals nothing)."""
+    import torch
+
+    from vantage6_amd.ops.secagg import quantize_mask
+
+    scale = float(1 << 16)
+    master = torch.tensor(vals, dtype=torch.float32)
+    unmasked = (master.double() * scale).round().to(torch.int64)
+    total = torch.zeros_like(unmasked)
+    any_masked_differs = False
+    for rank in range(world):
+        q = torch.empty_like(unmasked)
+        quantize_mask(q, master, 1.0, scale, seed=seed, rank=rank,
+                      world_size=world)
+        any_masked_differs |= bool((q != unmasked).any())
+        total = total + q  # int64 wraps mod 2^64 — by design
+    assert torch.equal(total, unmasked * world)
+    assert any_masked_differs
