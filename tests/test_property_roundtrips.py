@@ -84,3 +84,43 @@ def test_secagg_masks_cancel_exactly(vals, world, seed):
         total = total + q  # int64 wraps mod 2^64 — by design
     assert torch.equal(total, unmasked * world)
     assert any_masked_differs
+
+
+def test_totp_rfc6238_vectors():
+    """RFC 6238 Appendix B SHA-1 test vectors (8 digits)."""
+    import base64
+
+    from vantage6_amd.common.totp import totp_code, verify_code
+
+    secret = base64.b32encode(b"12345678901234567890").decode()
+    for at, expect in [(59, "94287082"), (1111111109, "07081804"),
+                       (1111111111, "14050471"), (1234567890, "89005924"),
+                       (2000000000, "69279037")]:
+        assert totp_code(secret, at=at, digits=8) == expect
+    # 6-digit verify: current and ±1 window accepted, ±3 rejected
+    now = 1700000000.0
+    from vantage6_amd.common.totp import totp_code as tc
+    assert verify_code(secret, tc(secret, at=now), at=now)
+    assert verify_code(secret, tc(secret, at=now - 30), at=now)
+    assert not verify_code(secret, tc(secret, at=now - 120), at=now)
+
+
+@settings(max_examples=30, deadline=None)
+@given(
+    st.lists(st.sampled_from(["a", "b", "c", "d"]), min_size=1, max_size=60),
+    st.integers(min_value=2, max_value=6),
+)
+def test_crosstab_station_never_leaks_small_cells(labels, threshold):
+    """Disclosure control is enforced AT THE STATION: no emitted count is
+    ever below the suppression threshold, and suppressed cells are
+    counted but not named."""
+    import pandas as pd
+
+    from algorithms.crosstab import partial_counts
+
+    df = pd.DataFrame({"g": labels})
+    out = partial_counts.__wrapped__(df, "g", None,
+                                     suppress_below=threshold, min_rows=1)
+    assert all(v >= threshold for v in out["counts"].values())
+    true_cells = df["g"].value_counts()
+    assert out["suppressed_cells"] == int((true_cells < threshold).sum())
