@@ -144,3 +144,43 @@ def test_envelope_is_three_field_and_legacy_four_field_decrypts(tmp_path):
     nonce, ct, tag = _sym_encrypt(shared, b"old-wire")
     legacy = SEP.join(bytes_to_base64s(x) for x in (enc_key, nonce, ct, tag))
     assert bob.decrypt_str_to_bytes(legacy) == b"old-wire"
+
+
+def test_garbage_bearer_tokens_are_401_not_500():
+    """Any malformed token (bad base64, bad JSON, non-dict payload, wrong
+    part count) must raise JWTError — the API maps exactly that to 401."""
+    import base64
+    import json as _json
+
+    import pytest
+
+    from vantage6_amd.common import jwt as jwtlib
+
+    good = jwtlib.encode({"sub": 1, "type": "user"}, "s3cret")
+    assert jwtlib.decode(good, "s3cret")["sub"] == 1
+    h, p, s = good.split(".")
+
+    def b64(obj):
+        return base64.urlsafe_b64encode(
+            _json.dumps(obj).encode()).rstrip(b"=").decode()
+
+    cases = [
+        "not-a-token",
+        "a.b",                       # wrong part count
+        "a.b.c.d",
+        f"{h}.!!!notb64!!!.{s}",     # invalid base64 payload
+        f"{h}.{p}.!!!notb64!!!",     # invalid base64 signature
+        good[:-4] + "AAAA",          # bit-flipped signature
+        f"{h}.{base64.urlsafe_b64encode(b'not json').rstrip(b'=').decode()}.{s}",
+    ]
+    # valid signature over a NON-DICT payload
+    arr = b64([1, 2, 3])
+    import hashlib
+    import hmac as _hmac
+    sig = base64.urlsafe_b64encode(_hmac.new(
+        b"s3cret", f"{h}.{arr}".encode(), hashlib.sha256).digest()
+    ).rstrip(b"=").decode()
+    cases.append(f"{h}.{arr}.{sig}")
+    for bad in cases:
+        with pytest.raises(jwtlib.JWTError):
+            jwtlib.decode(bad, "s3cret")

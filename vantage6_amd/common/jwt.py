@@ -42,15 +42,23 @@ def encode(payload: dict[str, Any], secret: str, expires_in: float | None = 2160
 
 
 def decode(token: str, secret: str, verify_exp: bool = True) -> dict[str, Any]:
+    # every parse failure maps to JWTError: callers catch exactly that to
+    # answer 401 — a stray binascii/json error from garbage input must not
+    # surface as a 500
     try:
         h, p, s = token.split(".")
-    except ValueError as exc:
+        signing = f"{h}.{p}"
+        expect = hmac.new(secret.encode(), signing.encode(),
+                          hashlib.sha256).digest()
+        if not hmac.compare_digest(expect, _unb64url(s)):
+            raise JWTError("bad signature")
+        payload = json.loads(_unb64url(p))
+    except JWTError:
+        raise
+    except Exception as exc:
         raise JWTError("malformed token") from exc
-    signing = f"{h}.{p}"
-    expect = hmac.new(secret.encode(), signing.encode(), hashlib.sha256).digest()
-    if not hmac.compare_digest(expect, _unb64url(s)):
-        raise JWTError("bad signature")
-    payload = json.loads(_unb64url(p))
+    if not isinstance(payload, dict):
+        raise JWTError("malformed token")
     if verify_exp and "exp" in payload and payload["exp"] < time.time():
         raise JWTError("token expired")
     return payload
