@@ -184,3 +184,29 @@ def test_garbage_bearer_tokens_are_401_not_500():
     for bad in cases:
         with pytest.raises(jwtlib.JWTError):
             jwtlib.decode(bad, "s3cret")
+
+
+def test_malformed_keys_and_envelopes_raise_valueerror(tmp_path):
+    """Corrupt public keys and envelopes fail as ValueError (what all
+    crypto call sites catch), never KeyError/TypeError."""
+    import base64
+
+    import pytest
+
+    from vantage6_amd.common.encryption import RSACryptor
+
+    c = RSACryptor(tmp_path / "k.pem", bits=1024)
+    junk_keys = [
+        base64.b64encode(b"not json").decode(),
+        base64.b64encode(b'{"kty": "RSA"}').decode(),          # missing n/e
+        base64.b64encode(b'{"n": 5, "e": "x"}').decode(),      # wrong types
+    ]
+    for bad in junk_keys:
+        with pytest.raises(ValueError):
+            c.encrypt_bytes_to_str(b"payload", bad)
+    good = c.encrypt_bytes_to_str(b"payload", c.public_key_str)
+    assert c.decrypt_str_to_bytes(good) == b"payload"
+    for bad_env in ["", "a$b", "a$b$c$d$e", "!!$??$%%",
+                    good[:-6] + "AAAAAA"]:  # tag tamper
+        with pytest.raises(ValueError):
+            c.decrypt_str_to_bytes(bad_env)

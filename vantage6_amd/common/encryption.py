@@ -170,8 +170,16 @@ class RSAKey:
 
     @classmethod
     def from_json(cls, blob: str) -> "RSAKey":
-        d = json.loads(blob)
-        return cls(int(d["n"], 16), int(d["e"]), int(d["d"], 16) if "d" in d else None)
+        # junk keys (e.g. a corrupted org public_key record) must fail as
+        # ValueError — that's what every crypto call site catches
+        try:
+            d = json.loads(blob)
+            return cls(int(d["n"], 16), int(d["e"]),
+                       int(d["d"], 16) if "d" in d else None)
+        except ValueError:
+            raise
+        except Exception as exc:
+            raise ValueError("malformed RSA key") from exc
 
     def public_base64s(self) -> str:
         return bytes_to_base64s(self.public_json().encode())
