@@ -210,3 +210,36 @@ def test_malformed_keys_and_envelopes_raise_valueerror(tmp_path):
                     good[:-6] + "AAAAAA"]:  # tag tamper
         with pytest.raises(ValueError):
             c.decrypt_str_to_bytes(bad_env)
+
+
+def test_dispatch_only_calls_module_defined_public_functions(tmp_path):
+    """Task 'method' names resolve only to public functions DEFINED in the
+    algorithm module — imported callables (from os import system) and
+    underscore helpers are rejected."""
+    import sys
+    import types
+
+    import pytest
+
+    from vantage6_amd.algorithm.exceptions import InputError
+    from vantage6_amd.algorithm.wrap import dispatch
+
+    mod = types.ModuleType("fake_algo")
+    src = (
+        "from os import system\n"
+        "import math\n"
+        "def good(x):\n"
+        "    return x + 1\n"
+        "def _helper():\n"
+        "    return 'no'\n"
+    )
+    exec(src, mod.__dict__)
+    mod.__dict__["__name__"] = "fake_algo"
+    sys.modules["fake_algo"] = mod
+    try:
+        assert dispatch(mod, {"method": "good", "args": [1]}, {}) == 2
+        for bad in ("system", "math", "_helper", "__builtins__", ""):
+            with pytest.raises(InputError):
+                dispatch(mod, {"method": bad, "args": ["true"]}, {})
+    finally:
+        del sys.modules["fake_algo"]

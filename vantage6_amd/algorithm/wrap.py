@@ -51,9 +51,24 @@ def dispatch(
     method_name = payload.get("method")
     if not method_name:
         raise InputError("input payload has no 'method'")
-    func = getattr(module, method_name, None)
+    # in-process stations have no container boundary, so the dispatch gate
+    # is tighter than the reference's: only PUBLIC functions DEFINED IN the
+    # algorithm module are callable — never imported names (a module-level
+    # `from os import system` must not become a remote method) and never
+    # underscore-prefixed helpers
+    func = getattr(module, method_name, None) \
+        if not method_name.startswith("_") else None
     if func is None or not callable(func):
         raise InputError(f"method {method_name!r} not found in {module.__name__}")
+    is_marked = (getattr(func, "_v6_wants_env", False)
+                 or getattr(func, "_v6_wants_client", False))
+    if not is_marked and getattr(func, "__module__", None) != module.__name__:
+        # @data/@algorithm_client-decorated functions are explicitly
+        # exported task methods wherever they were defined (re-export
+        # from a submodule is a normal algorithm-package layout)
+        raise InputError(
+            f"method {method_name!r} is not defined in {module.__name__} "
+            f"(imported names are not callable as task methods)")
     args = payload.get("args", []) or []
     kwargs = dict(payload.get("kwargs", {}) or {})
     if getattr(func, "_v6_wants_env", False):
