@@ -430,3 +430,43 @@ def test_pagination_link_headers(server, net):
     assert r.status_code == 200
     assert 'rel="next"' in r.headers.get("Link", "")
     assert int(r.headers["Total-Count"]) >= 3
+
+
+def test_pagination_link_headers(server):
+    """GET listings carry RFC 8288 Link headers (first/last/next/prev) and
+    X-Total style links block in the body, like the reference's Pagination
+    helper."""
+    client = UserClient(url=server)
+    client.authenticate("root", "root")
+    for i in range(5):
+        client.organization.create(name=f"pg-http-{i}")
+    tok = client.gw.token
+    r = requests.get(f"{server}/api/organization",
+                     params={"page": 2, "per_page": 2},
+                     headers={"Authorization": f"Bearer {tok}"}, timeout=5)
+    assert r.status_code == 200
+    body = r.json()
+    assert len(body["data"]) == 2
+    assert body["links"]["page"] == 2
+    link = r.headers.get("link", "")
+    assert 'rel="first"' in link and 'rel="last"' in link
+    assert 'rel="next"' in link and 'rel="prev"' in link
+    assert "page=3" in link and "page=1" in link
+
+
+def test_permission_errors_are_403_not_500_on_user_study_listings(server, net):
+    """user/study listings previously bypassed the exception-mapping
+    wrapper: a core PermissionError_ surfaced as 500."""
+    core = net.core
+    from vantage6_amd.client import UserClient as UC
+    root = UC(core=core)
+    root.authenticate("root", "root")
+    org = root.organization.create(name="noperm-org")
+    root.user.create("noperm", "pw", organization_id=org["id"], roles=[])
+    http = UserClient(url=server)
+    http.authenticate("noperm", "pw")
+    for path in ("user", "study", "study/999"):
+        r = requests.get(f"{server}/api/{path}",
+                         headers={"Authorization": f"Bearer {http.gw.token}"},
+                         timeout=5)
+        assert r.status_code in (401, 403, 404), (path, r.status_code)

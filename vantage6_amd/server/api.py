@@ -240,10 +240,12 @@ def create_app(core: ServerCore) -> FastAPI:
                     **{k: v for k, v in body.model_dump().items() if v is not None})
 
     @app.get("/api/result")
-    def result_list(who: Identity = Depends(ident), task_id: int = 0,
+    def result_list(request: Request, response: Response,
+                    who: Identity = Depends(ident), task_id: int = 0,
                     page: int = 1, per_page: int = 50):
-        return wrap(core.list_results, who, task_id=task_id, page=page,
-                    per_page=per_page)
+        out = wrap(core.list_results, who, task_id=task_id, page=page,
+                   per_page=per_page)
+        return paginate_headers(out, request, response)
 
     # ------------------------------------------------------------ entities
     @app.post("/api/organization", status_code=201)
@@ -252,8 +254,11 @@ def create_app(core: ServerCore) -> FastAPI:
                     country=body.country, public_key=body.public_key)
 
     @app.get("/api/organization")
-    def org_list(who: Identity = Depends(ident), page: int = 1, per_page: int = 20):
-        return wrap(core.list_organizations, who, page=page, per_page=per_page)
+    def org_list(request: Request, response: Response,
+                 who: Identity = Depends(ident), page: int = 1,
+                 per_page: int = 20):
+        out = wrap(core.list_organizations, who, page=page, per_page=per_page)
+        return paginate_headers(out, request, response)
 
     @app.get("/api/organization/{org_id}")
     def org_get(org_id: int, who: Identity = Depends(ident)):
@@ -271,9 +276,11 @@ def create_app(core: ServerCore) -> FastAPI:
                     body.organization_ids, body.encrypted)
 
     @app.get("/api/collaboration")
-    def collab_list(who: Identity = Depends(ident), page: int = 1,
+    def collab_list(request: Request, response: Response,
+                    who: Identity = Depends(ident), page: int = 1,
                     per_page: int = 20):
-        return wrap(core.list_collaborations, who, page=page, per_page=per_page)
+        out = wrap(core.list_collaborations, who, page=page, per_page=per_page)
+        return paginate_headers(out, request, response)
 
     @app.post("/api/node", status_code=201)
     def node_create(body: NodeCreateBody, who: Identity = Depends(ident)):
@@ -281,16 +288,21 @@ def create_app(core: ServerCore) -> FastAPI:
                     body.organization_id, body.device)
 
     @app.get("/api/node")
-    def node_list(who: Identity = Depends(ident), page: int = 1, per_page: int = 20,
+    def node_list(request: Request, response: Response,
+                  who: Identity = Depends(ident), page: int = 1,
+                  per_page: int = 20,
                   collaboration_id: Optional[int] = None):
-        return wrap(core.list_nodes, who, page=page, per_page=per_page,
-                    collaboration_id=collaboration_id)
+        out = wrap(core.list_nodes, who, page=page, per_page=per_page,
+                   collaboration_id=collaboration_id)
+        return paginate_headers(out, request, response)
 
     @app.get("/api/user")
-    def user_list(who: Identity = Depends(ident), page: int = 1,
+    def user_list(request: Request, response: Response,
+                  who: Identity = Depends(ident), page: int = 1,
                   per_page: int = 20, organization_id: Optional[int] = None):
-        return core.list_users(who, page=page, per_page=per_page,
-                               organization_id=organization_id)
+        out = wrap(core.list_users, who, page=page, per_page=per_page,
+                   organization_id=organization_id)
+        return paginate_headers(out, request, response)
 
     @app.get("/api/user/{user_id}")
     def user_get(user_id: int, who: Identity = Depends(ident)):
@@ -325,15 +337,17 @@ def create_app(core: ServerCore) -> FastAPI:
                     lastname=body.lastname, email=body.email)
 
     @app.get("/api/study")
-    def study_list(who: Identity = Depends(ident), page: int = 1,
+    def study_list(request: Request, response: Response,
+                   who: Identity = Depends(ident), page: int = 1,
                    per_page: int = 20,
                    collaboration_id: Optional[int] = None):
-        return core.list_studies(who, collaboration_id=collaboration_id,
-                                 page=page, per_page=per_page)
+        out = wrap(core.list_studies, who, collaboration_id=collaboration_id,
+                   page=page, per_page=per_page)
+        return paginate_headers(out, request, response)
 
     @app.get("/api/study/{study_id}")
     def study_get(study_id: int, who: Identity = Depends(ident)):
-        return core.get_study(who, study_id)
+        return wrap(core.get_study, who, study_id)
 
     @app.post("/api/study", status_code=201)
     def study_create(body: StudyCreateBody, who: Identity = Depends(ident)):
@@ -346,14 +360,18 @@ def create_app(core: ServerCore) -> FastAPI:
                     body.organization_id, body.description)
 
     @app.get("/api/role")
-    def role_list(who: Identity = Depends(ident), page: int = 1,
+    def role_list(request: Request, response: Response,
+                  who: Identity = Depends(ident), page: int = 1,
                   per_page: int = 50):
-        return wrap(core.list_roles, who, page=page, per_page=per_page)
+        out = wrap(core.list_roles, who, page=page, per_page=per_page)
+        return paginate_headers(out, request, response)
 
     @app.get("/api/rule")
-    def rule_list(who: Identity = Depends(ident), page: int = 1,
+    def rule_list(request: Request, response: Response,
+                  who: Identity = Depends(ident), page: int = 1,
                   per_page: int = 200):
-        return wrap(core.list_rules, who, page=page, per_page=per_page)
+        out = wrap(core.list_rules, who, page=page, per_page=per_page)
+        return paginate_headers(out, request, response)
 
     # ------------------------------------------------------ algorithm store
     @app.post("/api/algorithm", status_code=201)
