@@ -470,3 +470,27 @@ def test_permission_errors_are_403_not_500_on_user_study_listings(server, net):
                          headers={"Authorization": f"Bearer {http.gw.token}"},
                          timeout=5)
         assert r.status_code in (401, 403, 404), (path, r.status_code)
+
+
+def test_all_entity_routes_map_core_errors(server, net):
+    """Every entity route maps core exceptions: 404 for missing ids, never
+    500 (user/node patch+delete, collaboration get, 2fa reset, columns)."""
+    http = UserClient(url=server)
+    http.authenticate("root", "root")
+    hdr = {"Authorization": f"Bearer {http.gw.token}"}
+    checks = [
+        ("GET", "user/99999", None),
+        ("PATCH", "user/99999", {"email": "x@y.z"}),
+        ("DELETE", "user/99999", None),
+        ("PATCH", "node/99999", {"name": "n"}),
+        ("DELETE", "node/99999", None),
+        ("GET", "collaboration/99999", None),
+        ("GET", "column/99999?label=default", None),
+    ]
+    for verb, path, body in checks:
+        r = requests.request(verb, f"{server}/api/{path}", json=body,
+                             headers=hdr, timeout=5)
+        assert r.status_code in (400, 401, 403, 404), (verb, path, r.status_code)
+    r = requests.post(f"{server}/api/recover/2fa/reset",
+                      json={"token": "garbage.token.here"}, timeout=5)
+    assert r.status_code in (400, 401), r.status_code

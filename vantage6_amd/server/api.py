@@ -306,29 +306,30 @@ def create_app(core: ServerCore) -> FastAPI:
 
     @app.get("/api/user/{user_id}")
     def user_get(user_id: int, who: Identity = Depends(ident)):
-        return core.get_user(who, user_id)
+        return wrap(core.get_user, who, user_id)
 
     @app.patch("/api/user/{user_id}")
     def user_patch(user_id: int, body: dict, who: Identity = Depends(ident)):
         allowed = {k: v for k, v in (body or {}).items() if k in (
             "password", "email", "firstname", "lastname", "roles")}
-        return core.update_user(who, user_id, **allowed)
+        return wrap(core.update_user, who, user_id, **allowed)
 
     @app.delete("/api/user/{user_id}")
     def user_delete(user_id: int, who: Identity = Depends(ident)):
-        return core.delete_user(who, user_id)
+        return wrap(core.delete_user, who, user_id)
 
     @app.patch("/api/node/{node_id}")
     def node_patch(node_id: int, body: dict, who: Identity = Depends(ident)):
-        return core.update_node(who, node_id, name=(body or {}).get("name"))
+        return wrap(core.update_node, who, node_id,
+                    name=(body or {}).get("name"))
 
     @app.delete("/api/node/{node_id}")
     def node_delete(node_id: int, who: Identity = Depends(ident)):
-        return core.delete_node(who, node_id)
+        return wrap(core.delete_node, who, node_id)
 
     @app.get("/api/collaboration/{collab_id}")
     def collaboration_get(collab_id: int, who: Identity = Depends(ident)):
-        return core.get_collaboration(who, collab_id)
+        return wrap(core.get_collaboration, who, collab_id)
 
     @app.post("/api/user", status_code=201)
     def user_create(body: UserCreateBody, who: Identity = Depends(ident)):
@@ -467,12 +468,12 @@ def create_app(core: ServerCore) -> FastAPI:
 
     @app.post("/api/recover/2fa/reset")
     def recover_2fa_reset(body: RecoverBody):
-        return core.reset_mfa_with_token(body.token)
+        return wrap(core.reset_mfa_with_token, body.token)
 
     @app.get("/api/column/{node_id}")
     def get_columns(node_id: int, label: str = "default",
                     who: Identity = Depends(ident)):
-        return core.get_columns(who, node_id, label)
+        return wrap(core.get_columns, who, node_id, label)
 
     # ----------------------------------------------------------------- ops
     @app.post("/api/ping")
