@@ -675,3 +675,29 @@ def test_pagination_serializes_only_the_page(core, root, monkeypatch):
     # pages are disjoint and ordered
     p1 = root.organization.list(page=1, per_page=3)["data"]
     assert {o["id"] for o in p1}.isdisjoint({o["id"] for o in out["data"]})
+
+
+def test_node_api_key_rotation(core, root):
+    """reset_api_key invalidates the old key, returns the new one exactly
+    once, and is permission-gated (node EDIT on the node's org)."""
+    orgs, c = make_collab(root, name="rot1")
+    made = root.node.create(name="rot-n", collaboration_id=c["id"],
+                            organization_id=orgs[0])
+    old_key = made["api_key"]
+    assert core.authenticate_node(old_key)["node_url"].endswith(
+        f"/node/{made['id']}")
+    out = root.node.reset_api_key(made["id"])
+    new_key = out["api_key"]
+    assert new_key != old_key
+    assert core.authenticate_node(new_key)["node_url"].endswith(
+        f"/node/{made['id']}")
+    with pytest.raises(PermissionError_):
+        core.authenticate_node(old_key)
+    # a user without node EDIT cannot rotate
+    orgs2, _ = make_collab(root, name="rot2")
+    root.user.create("rot-viewer", "pw", organization_id=orgs2[0],
+                     roles=["Viewer"])
+    v = UserClient(core=core)
+    v.authenticate("rot-viewer", "pw")
+    with pytest.raises(PermissionError_):
+        v.node.reset_api_key(made["id"])

@@ -313,6 +313,13 @@ class UserClient:
             return self.parent._call("delete_node", "DELETE", f"node/{node_id}",
                                      node_id=node_id)
 
+        def reset_api_key(self, node_id: int) -> dict:
+            """Rotate the node's api key; the new key is returned once
+            (reference /api/recover/node)."""
+            return self.parent._call("reset_node_api_key", "POST",
+                                     "recover/node", json={"id": node_id},
+                                     node_id=node_id)
+
         def columns(self, node_id: int, label: str = "default") -> list[str]:
             """Data-column discovery (reference /api/column)."""
             return self.parent._call("get_columns", "GET", f"column/{node_id}",

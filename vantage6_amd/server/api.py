@@ -327,6 +327,10 @@ def create_app(core: ServerCore) -> FastAPI:
     def node_delete(node_id: int, who: Identity = Depends(ident)):
         return wrap(core.delete_node, who, node_id)
 
+    @app.post("/api/recover/node")
+    def node_recover(body: dict, who: Identity = Depends(ident)):
+        return wrap(core.reset_node_api_key, who, int(body["id"]))
+
     @app.get("/api/collaboration/{collab_id}")
     def collaboration_get(collab_id: int, who: Identity = Depends(ident)):
         return wrap(core.get_collaboration, who, collab_id)

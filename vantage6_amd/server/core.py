@@ -653,6 +653,21 @@ class ServerCore:
         return self._node_json(node)
 
     @_locked
+    def reset_node_api_key(self, ident: Identity, node_id: int) -> dict:
+        """Rotate a node's api key (reference: /api/recover/node). The new
+        key is returned ONCE and only its hash is stored; any running
+        station using the old key loses access at its next auth."""
+        node = self.session.get(Node, node_id)
+        if node is None:
+            raise NotFound(f"node {node_id} not found")
+        self._check_user_can("node", Operation.EDIT, ident,
+                             org_id=node.organization_id)
+        api_key = secrets.token_urlsafe(32)
+        node.set_api_key(api_key)
+        self.session.commit()
+        return {"id": node.id, "api_key": api_key}
+
+    @_locked
     def delete_node(self, ident: Identity, node_id: int) -> dict:
         node = self.session.get(Node, node_id)
         if node is None:
