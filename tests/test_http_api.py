@@ -271,7 +271,7 @@ def test_remote_station_over_http(tmp_path):
     csv = tmp_path / "remote.csv"
     df.to_csv(csv, index=False)
 
-    core = ServerCore()
+    core = ServerCore(password_policy=False)
     core.create_root_user("root", "root")
     root = UserClient(core=core)
     root.authenticate("root", "root")
@@ -378,7 +378,7 @@ def test_remote_core_reauth_after_token_loss(tmp_path):
     from vantage6_amd.node.remote import RemoteCore
     from vantage6_amd.server.core import ServerCore
 
-    core = ServerCore()
+    core = ServerCore(password_policy=False)
     core.create_root_user("root", "root")
     root = UserClient(core=core)
     root.authenticate("root", "root")

@@ -20,7 +20,7 @@ from vantage6_amd.server.permission import PermissionError_
 
 @pytest.fixture
 def core():
-    c = ServerCore()
+    c = ServerCore(password_policy=False)
     c.create_root_user("root", "root")
     yield c
     c.close()
@@ -243,3 +243,30 @@ def test_dispatch_only_calls_module_defined_public_functions(tmp_path):
                 dispatch(mod, {"method": bad, "args": ["true"]}, {})
     finally:
         del sys.modules["fake_algo"]
+
+
+def test_password_policy_enforced_by_default():
+    """Default ServerCore enforces the reference-shaped password policy on
+    create, update, and recovery reset; dev/test cores opt out."""
+    import pytest
+
+    from vantage6_amd.server.core import BadRequest, ServerCore
+    from vantage6_amd.client import UserClient
+
+    core = ServerCore()  # policy ON by default
+    core.create_root_user("root", "root")  # bootstrap is exempt
+    cl = UserClient(core=core)
+    cl.authenticate("root", "root")
+    org = cl.organization.create(name="pol-org")
+    for weak in ("short1A", "alllowercase1", "ALLUPPERCASE1", "NoDigitsHere"):
+        with pytest.raises((BadRequest, RuntimeError)):
+            cl.user.create("pol-u", weak, organization_id=org["id"],
+                           roles=["Researcher"])
+    made = cl.user.create("pol-u", "Str0ngEnough",
+                          organization_id=org["id"], roles=["Researcher"])
+    with pytest.raises((BadRequest, RuntimeError)):
+        cl.user.update(made["id"], password="weak")
+    cl.user.update(made["id"], password="An0therStrong")
+    u2 = UserClient(core=core)
+    u2.authenticate("pol-u", "An0therStrong")
+    core.close()

@@ -13,7 +13,7 @@ from vantage6_amd.server.permission import PermissionError_
 
 @pytest.fixture
 def core():
-    c = ServerCore()
+    c = ServerCore(password_policy=False)
     c.create_root_user("root", "root")
     yield c
     c.close()
@@ -414,7 +414,8 @@ class TestReadScoping:
 
 class TestLockout:
     def test_account_locks_after_failed_logins(self):
-        core = ServerCore(max_failed_logins=3, login_lockout_seconds=3600)
+        core = ServerCore(max_failed_logins=3, login_lockout_seconds=3600,
+                          password_policy=False)
         core.create_root_user("root", "root")
         root = UserClient(core=core)
         root.authenticate("root", "root")

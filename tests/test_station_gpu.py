@@ -81,7 +81,7 @@ def test_remote_station_partial_fit_on_gpu():
     from vantage6_amd.server.api import create_app
     from vantage6_amd.server.core import ServerCore
 
-    core = ServerCore()
+    core = ServerCore(password_policy=False)
     core.create_root_user("root", "root")
     root = UserClient(core=core)
     root.authenticate("root", "root")

@@ -31,7 +31,7 @@ class DevNetwork:
         name: str = "demo",
         key_bits: int = 2048,
     ):
-        self.core = ServerCore(uri=uri)
+        self.core = ServerCore(uri=uri, password_policy=False)
         self.encrypted = encrypted
         self.core.create_root_user("root", "root")
         root = UserClient(core=self.core)

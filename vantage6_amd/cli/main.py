@@ -221,7 +221,9 @@ def import_entities(config_path, uri, root_password):
 
     with open(config_path) as fh:
         fixture = _yaml.safe_load(fh) or {}
-    core = ServerCore(uri=uri)
+    # fixture import is admin-controlled bootstrap (reference `v6 server
+    # import` also seeds literal YAML passwords) — no interactive policy
+    core = ServerCore(uri=uri, password_policy=False)
     core.create_root_user("root", root_password)
     root = UserClient(core=core)
     root.authenticate("root", root_password)

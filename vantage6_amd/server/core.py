@@ -134,9 +134,11 @@ class ServerCore:
         token_expires_seconds: float = 21600.0,
         max_failed_logins: int = 5,
         login_lockout_seconds: float = 3600.0,
+        password_policy: bool = True,
     ):
         self.max_failed_logins = max_failed_logins
         self.login_lockout_seconds = login_lockout_seconds
+        self.password_policy = password_policy
         self.db = Database(uri)
         self.jwt_secret = jwt_secret or secrets.token_hex(32)
         self.token_expires = token_expires_seconds
@@ -515,6 +517,24 @@ class ServerCore:
         return out
 
     @_locked
+    def validate_password(self, password: str) -> None:
+        """Reference-shaped password policy (vantage6 4.x user schema):
+        >= 8 chars with upper, lower and digit. Off for dev/demo networks
+        (the reference's `v6 dev` also seeds simple credentials)."""
+        if not self.password_policy:
+            return
+        problems = []
+        if len(password) < 8:
+            problems.append("at least 8 characters")
+        if not any(c.isupper() for c in password):
+            problems.append("an uppercase letter")
+        if not any(c.islower() for c in password):
+            problems.append("a lowercase letter")
+        if not any(c.isdigit() for c in password):
+            problems.append("a digit")
+        if problems:
+            raise BadRequest("password too weak; needs " + ", ".join(problems))
+
     def create_user(
         self,
         ident: Identity,
@@ -525,6 +545,7 @@ class ServerCore:
         **kw,
     ) -> dict:
         self._check_user_can("user", Operation.CREATE, ident, org_id=organization_id)
+        self.validate_password(password)
         s = self.session
         if s.query(User).filter_by(username=username).first():
             raise BadRequest(f"user {username!r} exists")
@@ -600,6 +621,7 @@ class ServerCore:
             raise NotFound(f"user {user_id} not found")
         self._check_can_touch_user(ident, user, Operation.EDIT)
         if password is not None:
+            self.validate_password(password)
             user.set_password(password)
         if email is not None:
             user.email = email

@@ -52,6 +52,7 @@ def reset_password(core, token: str, new_password: str) -> None:
     user = core.session.get(User, payload["sub"])
     if user is None:
         raise PermissionError_("unknown user", 401)
+    core.validate_password(new_password)
     user.set_password(new_password)
     user.failed_login_attempts = 0
     core.session.commit()
