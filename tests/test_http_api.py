@@ -494,3 +494,24 @@ def test_all_entity_routes_map_core_errors(server, net):
     r = requests.post(f"{server}/api/recover/2fa/reset",
                       json={"token": "garbage.token.here"}, timeout=5)
     assert r.status_code in (400, 401), r.status_code
+
+
+def test_auth_rate_limit_429(net):
+    """Credential endpoints are per-IP rate limited (429 + Retry-After)
+    upstream of the per-account lockout."""
+    from starlette.testclient import TestClient
+
+    from vantage6_amd.server.api import create_app
+
+    app = create_app(net.core, auth_rate_limit=5, auth_rate_window=60.0)
+    client = TestClient(app)
+    codes = []
+    for _ in range(8):
+        r = client.post("/api/token/user",
+                        json={"username": "nosuch", "password": "x"})
+        codes.append(r.status_code)
+    assert codes[:5] == [401] * 5
+    assert codes[5:] == [429] * 3
+    assert "retry-after" in {k.lower() for k in r.headers}
+    # non-auth endpoints are not limited
+    assert client.get("/api/version").status_code == 200

@@ -10,6 +10,8 @@ FastAPI + pydantic, speaking the same JSON shapes (SURVEY.md §8 stack note).
 """
 from __future__ import annotations
 
+import threading
+import time
 from typing import Optional
 
 from fastapi import Depends, FastAPI, HTTPException, Request, Response, WebSocket
@@ -125,8 +127,37 @@ class RunPatchBody(BaseModel):
     log: Optional[str] = None
 
 
-def create_app(core: ServerCore) -> FastAPI:
+def create_app(core: ServerCore, auth_rate_limit: int = 30,
+               auth_rate_window: float = 60.0) -> FastAPI:
     app = FastAPI(title="vantage6_amd orchestrator", docs_url="/api/docs")
+
+    # per-IP sliding-window rate limit on credential endpoints (reference:
+    # flask-limiter on the server app) — brute-force protection upstream
+    # of the per-account lockout. In-memory per process, like the
+    # reference's default memory storage.
+    _rl_hits: dict[str, list[float]] = {}
+    _rl_lock = threading.Lock()
+
+    @app.middleware("http")
+    async def _auth_rate_limiter(request: Request, call_next):
+        path = request.url.path
+        if auth_rate_limit and (path.startswith("/api/token/")
+                                or path.startswith("/api/recover/")):
+            ip = request.client.host if request.client else "?"
+            now = time.monotonic()
+            with _rl_lock:
+                hits = _rl_hits.setdefault(ip, [])
+                while hits and now - hits[0] > auth_rate_window:
+                    hits.pop(0)
+                if len(hits) >= auth_rate_limit:
+                    from starlette.responses import JSONResponse
+
+                    return JSONResponse(
+                        {"detail": "too many authentication attempts"},
+                        status_code=429,
+                        headers={"Retry-After": str(int(auth_rate_window))})
+                hits.append(now)
+        return await call_next(request)
 
     def ident(request: Request) -> Identity:
         auth = request.headers.get("Authorization", "")
