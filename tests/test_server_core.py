@@ -702,3 +702,54 @@ def test_node_api_key_rotation(core, root):
     v.authenticate("rot-viewer", "pw")
     with pytest.raises(PermissionError_):
         v.node.reset_api_key(made["id"])
+
+
+def test_entity_crud_completion(core, root):
+    """org/collab/study/role update + delete with dependents guards and
+    default-role immutability."""
+    orgs, c = make_collab(root, name="crud1")
+    # --- org update
+    upd = root.organization.update(orgs[0], name="crud1-renamed",
+                                   country="NL")
+    assert upd["name"] == "crud1-renamed" and upd["country"] == "NL"
+    # --- collab update
+    cu = root.collaboration.update(c["id"], name="crud1-c2", encrypted=False)
+    assert cu["name"] == "crud1-c2"
+    # --- study lifecycle
+    st = root.study.create(name="crud-study", collaboration_id=c["id"],
+                           organization_ids=[orgs[0]])
+    st2 = root.study.update(st["id"], name="crud-study-2",
+                            organization_ids=orgs)
+    assert st2["name"] == "crud-study-2"
+    root.study.delete(st["id"])
+    with pytest.raises(Exception):
+        root.study.get(st["id"])
+    # --- role: default roles immutable
+    roles = root.role.list(per_page=100)["data"]
+    default = next(r for r in roles if r["organization_id"] is None)
+    with pytest.raises(Exception):
+        root.role.update(default["id"], name="hax")
+    with pytest.raises(Exception):
+        root.role.delete(default["id"])
+    mine = root.role.create("crud-role", rule_ids=default["rules"][:2],
+                            organization_id=orgs[0])
+    upd = root.role.update(mine["id"], description="d2",
+                           rule_ids=default["rules"][:1])
+    assert len(upd["rules"]) == 1
+    root.role.delete(mine["id"])
+    # --- collab delete: guarded, then forced
+    root.node.create(name="crud-n", collaboration_id=c["id"],
+                     organization_id=orgs[0])
+    with pytest.raises(Exception, match="dependents"):
+        root.collaboration.delete(c["id"])
+    root.collaboration.delete(c["id"], delete_dependents=True)
+    with pytest.raises(Exception):
+        root.collaboration.get(c["id"])
+    # --- org delete: users guard, then forced
+    root.user.create("crud-u", "pw", organization_id=orgs[0],
+                     roles=["Researcher"])
+    with pytest.raises(Exception, match="dependents"):
+        root.organization.delete(orgs[0])
+    root.organization.delete(orgs[0], delete_dependents=True)
+    with pytest.raises(Exception):
+        root.organization.get(orgs[0])

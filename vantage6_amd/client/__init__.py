@@ -282,6 +282,18 @@ class UserClient:
             return self.parent._call("get_organization", "GET", f"organization/{org_id}",
                                      org_id=org_id)
 
+        def update(self, org_id: int, **fields) -> dict:
+            return self.parent._call("update_organization", "PATCH",
+                                     f"organization/{org_id}", json=fields,
+                                     org_id=org_id, **fields)
+
+        def delete(self, org_id: int, delete_dependents: bool = False) -> dict:
+            return self.parent._call(
+                "delete_organization", "DELETE",
+                f"organization/{org_id}?delete_dependents="
+                f"{str(delete_dependents).lower()}",
+                org_id=org_id, delete_dependents=delete_dependents)
+
         def create(self, name: str, **kw) -> dict:
             body = {"name": name, **kw}
             return self.parent._call("create_organization", "POST", "organization",
@@ -295,6 +307,18 @@ class UserClient:
         def get(self, cid: int) -> dict:
             return self.parent._call("get_collaboration", "GET",
                                      f"collaboration/{cid}", collab_id=cid)
+
+        def update(self, cid: int, **fields) -> dict:
+            return self.parent._call("update_collaboration", "PATCH",
+                                     f"collaboration/{cid}", json=fields,
+                                     collab_id=cid, **fields)
+
+        def delete(self, cid: int, delete_dependents: bool = False) -> dict:
+            return self.parent._call(
+                "delete_collaboration", "DELETE",
+                f"collaboration/{cid}?delete_dependents="
+                f"{str(delete_dependents).lower()}",
+                collab_id=cid, delete_dependents=delete_dependents)
 
         def create(self, name: str, organization_ids: list[int], encrypted=False) -> dict:
             body = {"name": name, "organization_ids": organization_ids, "encrypted": encrypted}
@@ -365,6 +389,15 @@ class UserClient:
                     "organization_id": organization_id}
             return self.parent._call("create_role", "POST", "role", json=body, **body)
 
+        def update(self, role_id: int, **fields) -> dict:
+            return self.parent._call("update_role", "PATCH",
+                                     f"role/{role_id}", json=fields,
+                                     role_id=role_id, **fields)
+
+        def delete(self, role_id: int) -> dict:
+            return self.parent._call("delete_role", "DELETE",
+                                     f"role/{role_id}", role_id=role_id)
+
     class RuleSub(_Sub):
         def list(self, **params) -> dict:
             return self.parent._call("list_rules", "GET", "rule", params=params, **params)
@@ -401,6 +434,15 @@ class UserClient:
             body = {"name": name, "collaboration_id": collaboration_id,
                     "organization_ids": organization_ids}
             return self.parent._call("create_study", "POST", "study", json=body, **body)
+
+        def update(self, study_id: int, **fields) -> dict:
+            return self.parent._call("update_study", "PATCH",
+                                     f"study/{study_id}", json=fields,
+                                     study_id=study_id, **fields)
+
+        def delete(self, study_id: int) -> dict:
+            return self.parent._call("delete_study", "DELETE",
+                                     f"study/{study_id}", study_id=study_id)
 
 
 Client = UserClient

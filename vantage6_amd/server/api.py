@@ -72,6 +72,9 @@ class OrgCreateBody(BaseModel):
 
 class OrgPatchBody(BaseModel):
     public_key: Optional[str] = None
+    name: Optional[str] = None
+    domain: Optional[str] = None
+    country: Optional[str] = None
 
 
 class CollabCreateBody(BaseModel):
@@ -297,9 +300,57 @@ def create_app(core: ServerCore, auth_rate_limit: int = 30,
 
     @app.patch("/api/organization/{org_id}")
     def org_patch(org_id: int, body: OrgPatchBody, who: Identity = Depends(ident)):
-        if body.public_key is None:
+        out = None
+        if body.public_key is not None:
+            out = wrap(core.set_organization_public_key, who, org_id,
+                       body.public_key)
+        if (body.name is not None or body.domain is not None
+                or body.country is not None):
+            out = wrap(core.update_organization, who, org_id, name=body.name,
+                       domain=body.domain, country=body.country)
+        if out is None:
             raise HTTPException(400, "nothing to patch")
-        return wrap(core.set_organization_public_key, who, org_id, body.public_key)
+        return out
+
+    @app.delete("/api/organization/{org_id}")
+    def org_delete(org_id: int, who: Identity = Depends(ident),
+                   delete_dependents: bool = False):
+        return wrap(core.delete_organization, who, org_id,
+                    delete_dependents=delete_dependents)
+
+    @app.patch("/api/collaboration/{collab_id}")
+    def collab_patch(collab_id: int, body: dict,
+                     who: Identity = Depends(ident)):
+        return wrap(core.update_collaboration, who, collab_id,
+                    name=(body or {}).get("name"),
+                    encrypted=(body or {}).get("encrypted"))
+
+    @app.delete("/api/collaboration/{collab_id}")
+    def collab_delete(collab_id: int, who: Identity = Depends(ident),
+                      delete_dependents: bool = False):
+        return wrap(core.delete_collaboration, who, collab_id,
+                    delete_dependents=delete_dependents)
+
+    @app.patch("/api/study/{study_id}")
+    def study_patch(study_id: int, body: dict, who: Identity = Depends(ident)):
+        return wrap(core.update_study, who, study_id,
+                    name=(body or {}).get("name"),
+                    organization_ids=(body or {}).get("organization_ids"))
+
+    @app.delete("/api/study/{study_id}")
+    def study_delete(study_id: int, who: Identity = Depends(ident)):
+        return wrap(core.delete_study, who, study_id)
+
+    @app.patch("/api/role/{role_id}")
+    def role_patch(role_id: int, body: dict, who: Identity = Depends(ident)):
+        return wrap(core.update_role, who, role_id,
+                    name=(body or {}).get("name"),
+                    description=(body or {}).get("description"),
+                    rule_ids=(body or {}).get("rule_ids"))
+
+    @app.delete("/api/role/{role_id}")
+    def role_delete(role_id: int, who: Identity = Depends(ident)):
+        return wrap(core.delete_role, who, role_id)
 
     @app.post("/api/collaboration", status_code=201)
     def collab_create(body: CollabCreateBody, who: Identity = Depends(ident)):

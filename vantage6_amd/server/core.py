@@ -456,6 +456,132 @@ class ServerCore:
         return self._org_json(org)
 
     @_locked
+    def update_organization(self, ident: Identity, org_id: int,
+                            name: str | None = None,
+                            domain: str | None = None,
+                            country: str | None = None) -> dict:
+        """PATCH /organization fields beyond public_key (reference allows
+        name/domain/address/country edits with org-scoped EDIT)."""
+        org = self.session.get(Organization, org_id)
+        if org is None:
+            raise NotFound(f"organization {org_id} not found")
+        self._check_user_can("organization", Operation.EDIT, ident,
+                             org_id=org_id)
+        if name is not None and name != org.name:
+            if self.session.query(Organization).filter_by(name=name).first():
+                raise BadRequest(f"organization {name!r} exists")
+            org.name = name
+        if domain is not None:
+            org.domain = domain
+        if country is not None:
+            org.country = country
+        self.session.commit()
+        return self._org_json(org)
+
+    @_locked
+    def delete_organization(self, ident: Identity, org_id: int,
+                            delete_dependents: bool = False) -> dict:
+        """DELETE /organization with the reference's dependents guard:
+        refuses while users/nodes/memberships exist unless
+        ``delete_dependents`` (then they are removed too)."""
+        org = self.session.get(Organization, org_id)
+        if org is None:
+            raise NotFound(f"organization {org_id} not found")
+        self._check_user_can("organization", Operation.DELETE, ident,
+                             org_id=org_id)
+        deps = {"users": len(org.users), "nodes": len(org.nodes),
+                "collaborations": len(org.collaborations)}
+        if any(deps.values()) and not delete_dependents:
+            raise BadRequest(
+                f"organization {org_id} has dependents {deps}; pass "
+                f"delete_dependents=true to remove them")
+        for u in list(org.users):
+            self.session.delete(u)
+        for n in list(org.nodes):
+            self.session.delete(n)
+        org.collaborations = []
+        self.session.delete(org)
+        self.session.commit()
+        return {"msg": f"organization {org_id} deleted"}
+
+    @_locked
+    def update_collaboration(self, ident: Identity, collab_id: int,
+                             name: str | None = None,
+                             encrypted: bool | None = None) -> dict:
+        collab = self.session.get(Collaboration, collab_id)
+        if collab is None:
+            raise NotFound(f"collaboration {collab_id} not found")
+        self._check_user_can("collaboration", Operation.EDIT, ident,
+                             collab_id=collab_id)
+        if name is not None and name != collab.name:
+            if self.session.query(Collaboration).filter_by(name=name).first():
+                raise BadRequest(f"collaboration {name!r} exists")
+            collab.name = name
+        if encrypted is not None:
+            collab.encrypted = bool(encrypted)
+        self.session.commit()
+        return self._collab_json(collab)
+
+    @_locked
+    def delete_collaboration(self, ident: Identity, collab_id: int,
+                             delete_dependents: bool = False) -> dict:
+        collab = self.session.get(Collaboration, collab_id)
+        if collab is None:
+            raise NotFound(f"collaboration {collab_id} not found")
+        self._check_user_can("collaboration", Operation.DELETE, ident,
+                             collab_id=collab_id)
+        deps = {"tasks": len(collab.tasks), "nodes": len(collab.nodes),
+                "studies": len(collab.studies)}
+        if any(deps.values()) and not delete_dependents:
+            raise BadRequest(
+                f"collaboration {collab_id} has dependents {deps}; pass "
+                f"delete_dependents=true to remove them")
+        for t in list(collab.tasks):
+            for r in t.runs:
+                self.session.delete(r)
+            self.session.delete(t)
+        for n in list(collab.nodes):
+            self.session.delete(n)
+        for st in list(collab.studies):
+            self.session.delete(st)
+        collab.organizations = []
+        self.session.delete(collab)
+        self.session.commit()
+        return {"msg": f"collaboration {collab_id} deleted"}
+
+    @_locked
+    def update_study(self, ident: Identity, study_id: int,
+                     name: str | None = None,
+                     organization_ids: list[int] | None = None) -> dict:
+        study = self.session.get(Study, study_id)
+        if study is None:
+            raise NotFound(f"study {study_id} not found")
+        self._check_user_can("study", Operation.EDIT, ident,
+                             collab_id=study.collaboration_id)
+        if name is not None:
+            study.name = name
+        if organization_ids is not None:
+            member_ids = {o.id for o in study.collaboration.organizations}
+            if not set(organization_ids) <= member_ids:
+                raise BadRequest(
+                    "study organizations must be collaboration members")
+            study.organizations = [self.session.get(Organization, i)
+                                   for i in organization_ids]
+        self.session.commit()
+        return self._study_json(study)
+
+    @_locked
+    def delete_study(self, ident: Identity, study_id: int) -> dict:
+        study = self.session.get(Study, study_id)
+        if study is None:
+            raise NotFound(f"study {study_id} not found")
+        self._check_user_can("study", Operation.DELETE, ident,
+                             collab_id=study.collaboration_id)
+        self.session.delete(study)
+        self.session.commit()
+        return {"msg": f"study {study_id} deleted"}
+
+    @_locked
     def create_collaboration(
         self, ident: Identity, name: str, organization_ids: list[int], encrypted: bool = False
     ) -> dict:
@@ -1220,6 +1346,61 @@ class ServerCore:
         return {"id": role.id, "name": role.name,
                 "organization_id": role.organization_id,
                 "rules": [r.id for r in role.rules]}
+
+    @_locked
+    def update_role(self, ident: Identity, role_id: int,
+                    name: str | None = None,
+                    description: str | None = None,
+                    rule_ids: list[int] | None = None) -> dict:
+        """Default (system) roles are immutable, like the reference's; rule
+        changes re-run the anti-escalation check."""
+        from .model import Role, Rule
+
+        s = self.session
+        role = s.get(Role, role_id)
+        if role is None:
+            raise NotFound(f"role {role_id} not found")
+        if role.organization_id is None:
+            raise BadRequest("default roles cannot be modified")
+        self._check_user_can("role", Operation.EDIT, ident,
+                             org_id=role.organization_id)
+        if rule_ids is not None:
+            own = PermissionManager(s).rules_for_user(self._user(ident))
+            rules = []
+            for rid in rule_ids:
+                rule = s.get(Rule, rid)
+                if rule is None:
+                    raise NotFound(f"rule {rid} not found")
+                if (rule.name, rule.scope, rule.operation) not in own:
+                    raise PermissionError_(
+                        f"cannot grant rule you do not hold: "
+                        f"{rule.name}/{rule.scope}/{rule.operation}")
+                rules.append(rule)
+            role.rules = rules
+        if name is not None:
+            role.name = name
+        if description is not None:
+            role.description = description
+        s.commit()
+        return {"id": role.id, "name": role.name,
+                "organization_id": role.organization_id,
+                "rules": [r.id for r in role.rules]}
+
+    @_locked
+    def delete_role(self, ident: Identity, role_id: int) -> dict:
+        from .model import Role
+
+        role = self.session.get(Role, role_id)
+        if role is None:
+            raise NotFound(f"role {role_id} not found")
+        if role.organization_id is None:
+            raise BadRequest("default roles cannot be deleted")
+        self._check_user_can("role", Operation.DELETE, ident,
+                             org_id=role.organization_id)
+        role.users = []
+        self.session.delete(role)
+        self.session.commit()
+        return {"msg": f"role {role_id} deleted"}
 
     @_locked
     def list_roles(self, ident: Identity, page=1, per_page=50) -> dict:
