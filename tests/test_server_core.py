@@ -753,3 +753,16 @@ def test_entity_crud_completion(core, root):
     root.organization.delete(orgs[0], delete_dependents=True)
     with pytest.raises(Exception):
         root.organization.get(orgs[0])
+
+
+def test_role_rule_get_by_id(core, root):
+    roles = root.role.list(per_page=5)["data"]
+    r = root.role.get(roles[0]["id"])
+    assert r["id"] == roles[0]["id"] and "rules" in r
+    rules = root.rule.list(per_page=5)["data"]
+    ru = root.rule.get(rules[0]["id"])
+    assert ru["id"] == rules[0]["id"] and {"name", "scope", "operation"} <= set(ru)
+    with pytest.raises(Exception):
+        root.role.get(99999)
+    with pytest.raises(Exception):
+        root.rule.get(99999)

@@ -389,6 +389,10 @@ class UserClient:
                     "organization_id": organization_id}
             return self.parent._call("create_role", "POST", "role", json=body, **body)
 
+        def get(self, role_id: int) -> dict:
+            return self.parent._call("get_role", "GET", f"role/{role_id}",
+                                     role_id=role_id)
+
         def update(self, role_id: int, **fields) -> dict:
             return self.parent._call("update_role", "PATCH",
                                      f"role/{role_id}", json=fields,
@@ -401,6 +405,10 @@ class UserClient:
     class RuleSub(_Sub):
         def list(self, **params) -> dict:
             return self.parent._call("list_rules", "GET", "rule", params=params, **params)
+
+        def get(self, rule_id: int) -> dict:
+            return self.parent._call("get_rule", "GET", f"rule/{rule_id}",
+                                     rule_id=rule_id)
 
     class StoreSub(_Sub):
         """Algorithm-store sub-client (reference ``client.store``, 4.2+)."""

@@ -341,6 +341,14 @@ def create_app(core: ServerCore, auth_rate_limit: int = 30,
     def study_delete(study_id: int, who: Identity = Depends(ident)):
         return wrap(core.delete_study, who, study_id)
 
+    @app.get("/api/role/{role_id}")
+    def role_get(role_id: int, who: Identity = Depends(ident)):
+        return wrap(core.get_role, who, role_id)
+
+    @app.get("/api/rule/{rule_id}")
+    def rule_get(rule_id: int, who: Identity = Depends(ident)):
+        return wrap(core.get_rule, who, rule_id)
+
     @app.patch("/api/role/{role_id}")
     def role_patch(role_id: int, body: dict, who: Identity = Depends(ident)):
         return wrap(core.update_role, who, role_id,

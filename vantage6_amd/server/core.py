@@ -1348,6 +1348,30 @@ class ServerCore:
                 "rules": [r.id for r in role.rules]}
 
     @_locked
+    def get_role(self, ident: Identity, role_id: int) -> dict:
+        from .model import Role
+
+        self._check_user_can_view_any("role", ident)
+        role = self.session.get(Role, role_id)
+        if role is None:
+            raise NotFound(f"role {role_id} not found")
+        return {"id": role.id, "name": role.name,
+                "description": role.description,
+                "organization_id": role.organization_id,
+                "rules": [r.id for r in role.rules]}
+
+    @_locked
+    def get_rule(self, ident: Identity, rule_id: int) -> dict:
+        from .model import Rule
+
+        self._check_user_can_view_any("rule", ident)
+        rule = self.session.get(Rule, rule_id)
+        if rule is None:
+            raise NotFound(f"rule {rule_id} not found")
+        return {"id": rule.id, "name": rule.name, "scope": rule.scope,
+                "operation": rule.operation}
+
+    @_locked
     def update_role(self, ident: Identity, role_id: int,
                     name: str | None = None,
                     description: str | None = None,
