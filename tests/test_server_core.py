@@ -766,3 +766,49 @@ def test_role_rule_get_by_id(core, root):
         root.role.get(99999)
     with pytest.raises(Exception):
         root.rule.get(99999)
+
+
+def test_station_user_org_allow_list_policies(core, root):
+    """allowed_users / allowed_organizations node policies (reference 4.x)
+    reject tasks from other initiators with status 'not allowed'."""
+    import time
+
+    from vantage6_amd.common.config import StationConfig
+    from vantage6_amd.node.station import Station
+
+    orgs, collab = make_collab(root, name="polnet")
+    node = root.node.create(name="pn0", collaboration_id=collab["id"],
+                            organization_id=orgs[0])
+    root.user.create("pol-a", "pw", organization_id=orgs[0],
+                     roles=["Researcher"])
+    ua = UserClient(core=core)
+    ua.authenticate("pol-a", "pw")
+    allowed_id = ua.gw.ident.id
+    cfg = StationConfig(name="pn0", organization_id=orgs[0],
+                        policies={"allowed_users": [allowed_id]})
+    st = Station(core, cfg, api_key=node["api_key"])
+    st.start()
+
+    def wait_done(tid):
+        for _ in range(100):
+            t = root.task.get(tid)
+            if t["status"] not in ("pending", "initializing", "active"):
+                return t["status"]
+            time.sleep(0.05)
+        return t["status"]
+
+    try:
+        # root is NOT in allowed_users → rejected
+        t1 = root.task.create(collaboration=collab["id"],
+                              organizations=[orgs[0]], name="t1",
+                              image="algorithms.diagnostics",
+                              input_={"method": "partial_diagnostics"})
+        assert wait_done(t1["id"]) == "not allowed"
+        # allowed user passes the policy gate
+        t2 = ua.task.create(collaboration=collab["id"],
+                            organizations=[orgs[0]], name="t2",
+                            image="algorithms.diagnostics",
+                            input_={"method": "partial_diagnostics"})
+        assert wait_done(t2["id"]) == "completed"
+    finally:
+        st.stop()

@@ -284,6 +284,27 @@ class Station:
                 log=f"algorithm {task['image']!r} is not store-approved",
             )
             return
+        # reference 4.x node policies: restrict WHO may run here, not just
+        # what — allowed_users / allowed_organizations are id allow-lists
+        # against the task initiator
+        allowed_users = self.config.policies.get("allowed_users")
+        if allowed_users is not None and \
+                task.get("init_user_id") not in allowed_users:
+            self.core.patch_run(
+                self.identity, run_id, status=RunStatus.NOT_ALLOWED.value,
+                log=f"user {task.get('init_user_id')} not in station "
+                    f"allowed_users policy",
+            )
+            return
+        allowed_orgs = self.config.policies.get("allowed_organizations")
+        if allowed_orgs is not None and \
+                task.get("init_org_id") not in allowed_orgs:
+            self.core.patch_run(
+                self.identity, run_id, status=RunStatus.NOT_ALLOWED.value,
+                log=f"organization {task.get('init_org_id')} not in station "
+                    f"allowed_organizations policy",
+            )
+            return
         try:
             module = resolve_module(task["image"])
         except InputError as exc:
