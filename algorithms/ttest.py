@@ -64,6 +64,49 @@ def central_ttest(client, value: str, group: str, equal_var: bool = False,
     }
 
 
+@algorithm_client
+def central_anova(client, value: str, group: str, min_rows: int = 1):
+    """Federated one-way ANOVA — exact from the same per-group moment
+    sums (matches scipy.stats.f_oneway on the pooled rows)."""
+    org_ids = [org["id"] for org in client.organization.list()]
+    task = client.task.create(
+        input_={"method": "partial_ttest_moments",
+                "kwargs": {"value": value, "group": group,
+                           "min_rows": min_rows}},
+        organizations=org_ids,
+        name="partial anova moments",
+    )
+    results = client.wait_for_results(task["id"])
+    levels = sorted({lv for r in results for lv in r})
+    if len(levels) < 2:
+        raise InputError(f"group column {group!r} needs >= 2 levels")
+    n_tot, s_tot, ss_tot = 0, 0.0, 0.0
+    groups = {}
+    for lv in levels:
+        n = sum(r.get(lv, {"n": 0})["n"] for r in results)
+        s1 = sum(r.get(lv, {"sum": 0.0})["sum"] for r in results)
+        s2 = sum(r.get(lv, {"sum_sq": 0.0})["sum_sq"] for r in results)
+        if n < 1:
+            continue
+        groups[lv] = (n, s1, s2)
+        n_tot += n
+        s_tot += s1
+        ss_tot += s2
+    k = len(groups)
+    grand_mean = s_tot / n_tot
+    ss_between = sum(n * (s1 / n - grand_mean) ** 2
+                     for n, s1, s2 in groups.values())
+    ss_within = sum(s2 - s1 * s1 / n for n, s1, s2 in groups.values())
+    df_b, df_w = k - 1, n_tot - k
+    if df_w <= 0 or ss_within <= 0:
+        raise InputError("not enough data for ANOVA")
+    f = (ss_between / df_b) / (ss_within / df_w)
+    p = float(_st.f.sf(f, df_b, df_w))
+    return {"F": f, "p_value": p, "df_between": df_b, "df_within": df_w,
+            "groups": {str(lv): {"n": n, "mean": s1 / n}
+                       for lv, (n, s1, s2) in groups.items()}}
+
+
 @data(1)
 def partial_ttest_moments(df, value: str, group: str, min_rows: int = 1):
     if len(df) < min_rows:

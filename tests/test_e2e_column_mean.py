@@ -606,3 +606,43 @@ def test_federated_ttest_matches_scipy(tmp_path):
             assert res["t"] == pytest.approx(ref.statistic, rel=1e-10)
             assert res["p_value"] == pytest.approx(ref.pvalue, rel=1e-9)
             assert res["groups"]["ctl"]["n"] == len(ctl)
+
+
+def test_federated_anova_matches_scipy(tmp_path):
+    """Federated one-way ANOVA equals scipy.stats.f_oneway on pooled rows."""
+    import numpy as np
+    import pandas as pd
+    from scipy import stats
+
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+
+    rng = np.random.default_rng(67)
+    frames, paths = [], []
+    for i in range(2):
+        n = 30 + 10 * i
+        df = pd.DataFrame({
+            "y": np.concatenate([rng.normal(5, 2, n), rng.normal(6, 2, n),
+                                 rng.normal(5.5, 2.5, n)]),
+            "arm": ["a"] * n + ["b"] * n + ["c"] * n,
+        })
+        p = tmp_path / f"an{i}.csv"
+        df.to_csv(p, index=False)
+        frames.append(df)
+        paths.append(p)
+    pooled = pd.concat(frames)
+    samples = [pooled[pooled.arm == g].y.to_numpy() for g in ("a", "b", "c")]
+    ref = stats.f_oneway(*samples)
+    net = DevNetwork(n_stations=2, databases=[
+        [DatabaseConfig(label="default", uri=str(p), type="csv")] for p in paths])
+    with net:
+        task = net.client.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],
+            name="anova", image="algorithms.ttest",
+            input_={"method": "central_anova",
+                    "kwargs": {"value": "y", "group": "arm"}})
+        (res,) = net.client.wait_for_results(task["id"])
+    assert res["F"] == pytest.approx(ref.statistic, rel=1e-9)
+    assert res["p_value"] == pytest.approx(ref.pvalue, rel=1e-8)
+    assert res["df_between"] == 2
