@@ -158,3 +158,96 @@ def test_fedavg_gloo_four_ranks_secure_equals_plain():
     for r in results.values():
         assert r["sec_ck"] == pytest.approx(r["plain_ck"], rel=1e-5)
         assert r["agg_ar"] == pytest.approx(r["agg_rb"], rel=1e-6)
+
+
+def _prox_momentum_worker(rank: int, world: int, port: int, results):
+    os.environ.update(
+        RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank),
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+    )
+    import torch.distributed as dist
+
+    from algorithms.fedavg import _build, make_batches
+    from vantage6_amd.parallel.comm import init_from_env
+    from vantage6_amd.parallel.fedavg import FedAvgConfig, FedAvgTrainer
+
+    comm = init_from_env(force_backend="gloo")
+    try:
+        g = torch.Generator().manual_seed(500 + rank)
+        x = torch.randn(128, 8, generator=g)
+        dataset = {"x": x, "y": (x.sum(-1) > 0).long()}
+        mk = {"in_features": 8, "num_classes": 2}
+
+        def run(mu=0.0, beta=0.0, slr=1.0, rounds=3):
+            net, loss = _build("logreg", mk, "cpu")
+            tr = FedAvgTrainer(
+                net, loss, make_batches(dataset, 32, "cpu", seed=7), comm,
+                FedAvgConfig(local_steps=2, lr=0.1, momentum=0.0,
+                             weight_decay=0.0, fedprox_mu=mu,
+                             server_momentum=beta, server_lr=slr),
+                n_samples=128)
+            snaps = []
+            for _ in range(rounds):
+                tr.run_round()
+                snaps.append(tr.flat.master_flat.clone())
+            return snaps
+
+        base = run()
+        prox = run(mu=10.0)  # strong prox pulls harder toward round start
+        # manual FedAvgM check: replay plain rounds and apply the server
+        # momentum recurrence to the plain per-round MEANS is non-trivial
+        # (momentum changes the trajectory), so verify round-1 exactly:
+        # with v0=0, w1 = w0 + slr*(mean_1 - w0); slr=1 -> identical to
+        # plain FedAvg on round 1, and diverges after.
+        favgm = run(beta=0.9, slr=1.0)
+        results[rank] = {
+            "base_r1": float(base[0].double().sum()),
+            "favgm_r1": float(favgm[0].double().sum()),
+            "base_r3": float(base[-1].double().sum()),
+            "favgm_r3": float(favgm[-1].double().sum()),
+            "prox_drift": float((prox[0] - base[0]).abs().max()),
+        }
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_fedprox_and_server_momentum_gloo():
+    """FedProx changes the local trajectory; FedAvgM round 1 equals plain
+    FedAvg (v0=0, server_lr=1) and diverges afterwards; ranks agree."""
+    port = 29531
+    with mp.Manager() as man:
+        results = man.dict()
+        mp.spawn(_prox_momentum_worker, args=(2, port, results), nprocs=2,
+                 join=True)
+        r0, r1 = results[0], results[1]
+    assert r0 == r1  # bitwise identical across ranks (same global state)
+    assert abs(r0["favgm_r1"] - r0["base_r1"]) < 1e-9
+    assert abs(r0["favgm_r3"] - r0["base_r3"]) > 1e-9
+    assert r0["prox_drift"] > 1e-9
+
+
+def test_fedprox_gradient_math_single_rank():
+    """One local step with huge mu: grad = g + mu*(w - w0); with w=w0 at
+    the step the prox term is ZERO -> first step matches plain exactly;
+    the second step differs."""
+    from algorithms.fedavg import _build, make_batches
+    from vantage6_amd.parallel.comm import CommContext
+    from vantage6_amd.parallel.fedavg import FedAvgConfig, FedAvgTrainer
+
+    comm = CommContext()  # world 1, cpu
+    x = torch.randn(64, 8, generator=torch.Generator().manual_seed(3))
+    dataset = {"x": x, "y": (x.sum(-1) > 0).long()}
+    mk = {"in_features": 8, "num_classes": 2}
+
+    def run(mu, steps):
+        net, loss = _build("logreg", mk, "cpu")
+        tr = FedAvgTrainer(net, loss, make_batches(dataset, 32, "cpu", seed=5),
+                           comm, FedAvgConfig(local_steps=steps, lr=0.1,
+                                              momentum=0.0, weight_decay=0.0,
+                                              fedprox_mu=mu), n_samples=64)
+        tr.run_round()
+        return tr.flat.master_flat.clone()
+
+    assert torch.equal(run(0.0, 1), run(25.0, 1))   # prox term is 0 at w=w0
+    assert not torch.equal(run(0.0, 2), run(25.0, 2))

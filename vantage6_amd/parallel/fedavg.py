@@ -56,6 +56,16 @@ class FedAvgConfig:
     # distributed + non-secure only; CPU/gloo and secure-agg stay
     # synchronous.
     overlap: bool = True
+    # FedProx (Li et al. 2020): local objective gains mu/2*||w - w_round||^2,
+    # i.e. grad += mu*(w - w_round) before every optimizer step. 0 = plain
+    # FedAvg (default; the reference ecosystem's FedAvg algorithms).
+    fedprox_mu: float = 0.0
+    # FedAvgM (Hsu et al. 2019): server-side momentum over round deltas —
+    # v = beta*v + (mean_w - w_old); w = w_old + server_lr*v. beta 0 = plain
+    # averaging (default). Disables fit/agg overlap when on (the sync
+    # aggregation path applies the transform per bucket).
+    server_momentum: float = 0.0
+    server_lr: float = 1.0
 
 
 class FedAvgTrainer:
@@ -105,11 +115,24 @@ class FedAvgTrainer:
         self.overlap_stats: dict = {}  # comm_ms / exposed_ms / overlap_pct
         if self._overlap_capable():
             self._install_overlap_hooks()
+        # ---- FedProx / FedAvgM state ------------------------------------
+        self._prox_ref: list = []  # per-param round-start snapshots
+        if cfg.fedprox_mu > 0:
+            if cfg.dp is not None:
+                raise ValueError("fedprox_mu with DP-SGD is unsupported "
+                                 "(per-sample clipping of the prox term is "
+                                 "not defined here)")
+            self._prox_params = [p for p in self.model.parameters()
+                                 if p.requires_grad]
+            self._prox_ref = [p.detach().clone() for p in self._prox_params]
+        self._srv_v = None  # server momentum buffer
+        self._srv_old = None
 
     # ------------------------------------------------- fit/agg overlap plumbing
     def _overlap_capable(self) -> bool:
         return (self.cfg.overlap and self.comm.device.type == "cuda"
-                and self.comm.is_distributed and not self.cfg.secure_agg)
+                and self.comm.is_distributed and not self.cfg.secure_agg
+                and self.cfg.server_momentum == 0.0)
 
     def _install_overlap_hooks(self) -> None:
         """Per-module forward-pre-hooks: wait on exactly the collective
@@ -261,6 +284,8 @@ class FedAvgTrainer:
             self.flat.zero_grad()
             loss = self.loss_fn(self.model, self._static_batch)
             loss.backward()
+            if self.cfg.fedprox_mu > 0:
+                self._apply_prox_grad()
         lr = self.cfg.lr_schedule(self.round_idx) if self.cfg.lr_schedule else self.cfg.lr
         if self.cfg.optimizer == "adamw":
             self.flat.adamw_step(lr=lr, weight_decay=self.cfg.weight_decay)
@@ -269,6 +294,16 @@ class FedAvgTrainer:
                                weight_decay=self.cfg.weight_decay,
                                nesterov=self.cfg.nesterov)
         return loss
+
+    def _apply_prox_grad(self) -> None:
+        # grad += mu*(w - w_round); per-param so it works with both flat
+        # and scattered grad layouts (p.grad views the flat buffer in the
+        # former). Pure tensor ops -> hipGraph-capturable; the _prox_ref
+        # buffers are persistent and refreshed by copy_ at round start.
+        with torch.no_grad():
+            for p, r in zip(self._prox_params, self._prox_ref):
+                if p.grad is not None:
+                    p.grad.add_(p.detach() - r, alpha=self.cfg.fedprox_mu)
 
     # ------------------------------------------------------------ local fit
     def local_step(self) -> float:
@@ -284,6 +319,8 @@ class FedAvgTrainer:
         self.flat.zero_grad()
         loss = self.loss_fn(self.model, batch)
         loss.backward()
+        if self.cfg.fedprox_mu > 0:
+            self._apply_prox_grad()
         self._join_agg()  # optimizer writes master/params
         lr = self.cfg.lr_schedule(self.round_idx) if self.cfg.lr_schedule else self.cfg.lr
         if self.cfg.optimizer == "adamw":
@@ -328,6 +365,9 @@ class FedAvgTrainer:
         if self._overlap_capable():
             self._aggregate_overlapped()
             return
+        beta = self.cfg.server_momentum  # _srv_old snapshot: run_round start
+        if beta > 0 and self._srv_old is None:
+            beta = 0.0  # aggregate() called outside run_round: plain mean
         master.mul_(self.weight)
         bucket = max(1, self.cfg.bucket_bytes // 4)
         works = []
@@ -350,6 +390,14 @@ class FedAvgTrainer:
         for off, chunk, work in works:
             if work is not None:
                 work.wait()
+            if beta > 0:
+                # FedAvgM per bucket: v = beta*v + (mean - old);
+                # w = old + server_lr*v  (chunk aliases master)
+                hi = off + chunk.numel()
+                v = self._srv_v[off:hi]
+                old = self._srv_old[off:hi]
+                v.mul_(beta).add_(chunk).sub_(old)
+                chunk.copy_(old).add_(v, alpha=self.cfg.server_lr)
             self.flat.param_flat[off : off + chunk.numel()].copy_(
                 chunk.to(self.flat.dtype)
             )
@@ -431,12 +479,32 @@ class FedAvgTrainer:
             rank=self.comm.rank, world_size=self.comm.world_size,
         )
         self.comm.all_reduce(q)
+        beta = self.cfg.server_momentum  # _srv_old snapshot: run_round start
+        if beta > 0 and self._srv_old is None:
+            beta = 0.0  # aggregate() called outside run_round: plain mean
         master.copy_(q.to(torch.float64).mul_(1.0 / scale).to(torch.float32))
+        if beta > 0:
+            self._srv_v.mul_(beta).add_(master).sub_(self._srv_old)
+            master.copy_(self._srv_old).add_(self._srv_v,
+                                             alpha=self.cfg.server_lr)
         self.flat.sync_param_from_master()
 
     # ---------------------------------------------------------------- rounds
     def run_round(self) -> dict:
         overlapping = self._overlap_capable()
+        if self.cfg.fedprox_mu > 0:
+            # w_round := the post-aggregation weights (persistent buffers,
+            # refreshed in place so a captured graph keeps reading them)
+            with torch.no_grad():
+                for p, r in zip(self._prox_params, self._prox_ref):
+                    r.copy_(p.detach())
+        if self.cfg.server_momentum > 0:
+            # FedAvgM's w_old is the ROUND-START global state — by
+            # aggregate time master has drifted through the local steps
+            if self._srv_v is None:
+                self._srv_v = torch.zeros_like(self.flat.master_flat)
+                self._srv_old = torch.empty_like(self.flat.master_flat)
+            self._srv_old.copy_(self.flat.master_flat)
         t0 = time.perf_counter()
         losses = [self.local_step() for _ in range(self.cfg.local_steps)]
         if self.comm.device.type == "cuda":
