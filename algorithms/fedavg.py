@@ -67,6 +67,9 @@ def fit_rounds(
     weight_decay: float = 5e-4,
     optimizer: str = "sgd",
     secure_agg: bool = False,
+    fedprox_mu: float = 0.0,
+    server_momentum: float = 0.0,
+    server_lr: float = 1.0,
     dp: dict | None = None,
     comm: CommContext | None = None,
     device=None,
@@ -90,7 +93,8 @@ def fit_rounds(
     cfg = FedAvgConfig(
         local_steps=local_steps, lr=lr, momentum=momentum,
         weight_decay=weight_decay, optimizer=optimizer, secure_agg=secure_agg,
-        dp=dp_cfg,
+        fedprox_mu=fedprox_mu, server_momentum=server_momentum,
+        server_lr=server_lr, dp=dp_cfg,
     )
     n = len(dataset[list(dataset.keys())[0]])
     trainer = FedAvgTrainer(net, loss_fn, batches, comm, cfg, n_samples=n)
