@@ -69,6 +69,7 @@ def fit_rounds(
     secure_agg: bool = False,
     fedprox_mu: float = 0.0,
     server_momentum: float = 0.0,
+    server_adam: bool = False,
     server_lr: float = 1.0,
     dp: dict | None = None,
     comm: CommContext | None = None,
@@ -94,7 +95,7 @@ def fit_rounds(
         local_steps=local_steps, lr=lr, momentum=momentum,
         weight_decay=weight_decay, optimizer=optimizer, secure_agg=secure_agg,
         fedprox_mu=fedprox_mu, server_momentum=server_momentum,
-        server_lr=server_lr, dp=dp_cfg,
+        server_adam=server_adam, server_lr=server_lr, dp=dp_cfg,
     )
     n = len(dataset[list(dataset.keys())[0]])
     trainer = FedAvgTrainer(net, loss_fn, batches, comm, cfg, n_samples=n)

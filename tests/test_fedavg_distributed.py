@@ -251,3 +251,66 @@ def test_fedprox_gradient_math_single_rank():
 
     assert torch.equal(run(0.0, 1), run(25.0, 1))   # prox term is 0 at w=w0
     assert not torch.equal(run(0.0, 2), run(25.0, 2))
+
+
+def _fedadam_worker(rank, world, port, results):
+    os.environ.update(
+        RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank),
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+    )
+    import torch.distributed as dist
+
+    from algorithms.fedavg import _build, make_batches
+    from vantage6_amd.parallel.comm import init_from_env
+    from vantage6_amd.parallel.fedavg import FedAvgConfig, FedAvgTrainer
+
+    comm = init_from_env(force_backend="gloo")
+    try:
+        g = torch.Generator().manual_seed(900 + rank)
+        x = torch.randn(96, 8, generator=g)
+        dataset = {"x": x, "y": (x.sum(-1) > 0).long()}
+        mk = {"in_features": 8, "num_classes": 2}
+
+        def run(adam: bool):
+            net, loss = _build("logreg", mk, "cpu")
+            tr = FedAvgTrainer(
+                net, loss, make_batches(dataset, 32, "cpu", seed=9), comm,
+                FedAvgConfig(local_steps=1, lr=0.1, momentum=0.0,
+                             weight_decay=0.0, server_adam=adam,
+                             server_lr=0.5), n_samples=96)
+            w0 = tr.flat.master_flat.clone()
+            means = []
+            for _ in range(2):
+                # capture the plain mean by replaying aggregation math:
+                # run the round, then reconstruct from returned state
+                tr.run_round()
+                means.append(tr.flat.master_flat.clone())
+            return w0, means
+
+        w0p, plain = run(False)
+        w0a, fedadam = run(True)
+        # manual replay of FedAdam from the PLAIN means requires identical
+        # local trajectories; round 1 starts from the same w0 so mean_1 is
+        # shared. Verify round-1 closed form: d = mean1 - w0;
+        # m = 0.1*d ... b1=0.9 -> m=(1-b1)d; v=(1-b2)d^2;
+        # w1 = w0 + 0.5 * m/(sqrt(v)+tau)
+        d = plain[0] - w0p
+        m = 0.1 * d
+        v = 0.01 * d * d
+        want = w0p + 0.5 * m / (v.sqrt() + 1e-3)
+        err = float((fedadam[0] - want).abs().max())
+        results[rank] = {"err": err,
+                         "ck": float(fedadam[-1].double().sum())}
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_fedadam_gloo_two_ranks():
+    port = 29541
+    with mp.Manager() as man:
+        results = man.dict()
+        mp.spawn(_fedadam_worker, args=(2, port, results), nprocs=2, join=True)
+        r0, r1 = results[0], results[1]
+    assert r0["err"] < 1e-6, r0["err"]
+    assert r0["ck"] == r1["ck"]

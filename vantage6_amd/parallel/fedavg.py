@@ -66,6 +66,13 @@ class FedAvgConfig:
     # aggregation path applies the transform per bucket).
     server_momentum: float = 0.0
     server_lr: float = 1.0
+    # FedAdam (Reddi et al. 2021): adaptive server update over round
+    # deltas — m = b1*m + (1-b1)*d; v = b2*v + (1-b2)*d^2;
+    # w = w_old + server_lr * m / (sqrt(v) + tau). Mutually exclusive with
+    # server_momentum; also sync-path only.
+    server_adam: bool = False
+    server_adam_betas: tuple = (0.9, 0.99)
+    server_adam_tau: float = 1e-3
 
 
 class FedAvgTrainer:
@@ -125,14 +132,18 @@ class FedAvgTrainer:
             self._prox_params = [p for p in self.model.parameters()
                                  if p.requires_grad]
             self._prox_ref = [p.detach().clone() for p in self._prox_params]
-        self._srv_v = None  # server momentum buffer
+        self._srv_v = None  # server momentum / adam second-moment buffer
+        self._srv_m = None  # adam first moment
         self._srv_old = None
+        if cfg.server_adam and cfg.server_momentum:
+            raise ValueError("server_adam and server_momentum are exclusive")
 
     # ------------------------------------------------- fit/agg overlap plumbing
     def _overlap_capable(self) -> bool:
         return (self.cfg.overlap and self.comm.device.type == "cuda"
                 and self.comm.is_distributed and not self.cfg.secure_agg
-                and self.cfg.server_momentum == 0.0)
+                and self.cfg.server_momentum == 0.0
+                and not self.cfg.server_adam)
 
     def _install_overlap_hooks(self) -> None:
         """Per-module forward-pre-hooks: wait on exactly the collective
@@ -366,8 +377,9 @@ class FedAvgTrainer:
             self._aggregate_overlapped()
             return
         beta = self.cfg.server_momentum  # _srv_old snapshot: run_round start
-        if beta > 0 and self._srv_old is None:
-            beta = 0.0  # aggregate() called outside run_round: plain mean
+        adam = self.cfg.server_adam
+        if (beta > 0 or adam) and self._srv_old is None:
+            beta, adam = 0.0, False  # aggregate() outside run_round
         master.mul_(self.weight)
         bucket = max(1, self.cfg.bucket_bytes // 4)
         works = []
@@ -398,6 +410,18 @@ class FedAvgTrainer:
                 old = self._srv_old[off:hi]
                 v.mul_(beta).add_(chunk).sub_(old)
                 chunk.copy_(old).add_(v, alpha=self.cfg.server_lr)
+            elif adam:
+                hi = off + chunk.numel()
+                m = self._srv_m[off:hi]
+                v = self._srv_v[off:hi]
+                old = self._srv_old[off:hi]
+                b1, b2 = self.cfg.server_adam_betas
+                d = chunk - old
+                m.mul_(b1).add_(d, alpha=1 - b1)
+                v.mul_(b2).addcmul_(d, d, value=1 - b2)
+                chunk.copy_(old).addcdiv_(
+                    m, v.sqrt().add_(self.cfg.server_adam_tau),
+                    value=self.cfg.server_lr)
             self.flat.param_flat[off : off + chunk.numel()].copy_(
                 chunk.to(self.flat.dtype)
             )
@@ -480,13 +504,22 @@ class FedAvgTrainer:
         )
         self.comm.all_reduce(q)
         beta = self.cfg.server_momentum  # _srv_old snapshot: run_round start
-        if beta > 0 and self._srv_old is None:
-            beta = 0.0  # aggregate() called outside run_round: plain mean
+        adam = self.cfg.server_adam
+        if (beta > 0 or adam) and self._srv_old is None:
+            beta, adam = 0.0, False  # aggregate() outside run_round
         master.copy_(q.to(torch.float64).mul_(1.0 / scale).to(torch.float32))
         if beta > 0:
             self._srv_v.mul_(beta).add_(master).sub_(self._srv_old)
             master.copy_(self._srv_old).add_(self._srv_v,
                                              alpha=self.cfg.server_lr)
+        elif adam:
+            b1, b2 = self.cfg.server_adam_betas
+            d = master - self._srv_old
+            self._srv_m.mul_(b1).add_(d, alpha=1 - b1)
+            self._srv_v.mul_(b2).addcmul_(d, d, value=1 - b2)
+            master.copy_(self._srv_old).addcdiv_(
+                self._srv_m, self._srv_v.sqrt().add_(self.cfg.server_adam_tau),
+                value=self.cfg.server_lr)
         self.flat.sync_param_from_master()
 
     # ---------------------------------------------------------------- rounds
@@ -498,12 +531,14 @@ class FedAvgTrainer:
             with torch.no_grad():
                 for p, r in zip(self._prox_params, self._prox_ref):
                     r.copy_(p.detach())
-        if self.cfg.server_momentum > 0:
-            # FedAvgM's w_old is the ROUND-START global state — by
-            # aggregate time master has drifted through the local steps
+        if self.cfg.server_momentum > 0 or self.cfg.server_adam:
+            # the server optimizer's w_old is the ROUND-START global state
+            # — by aggregate time master has drifted through local steps
             if self._srv_v is None:
                 self._srv_v = torch.zeros_like(self.flat.master_flat)
                 self._srv_old = torch.empty_like(self.flat.master_flat)
+                if self.cfg.server_adam:
+                    self._srv_m = torch.zeros_like(self.flat.master_flat)
             self._srv_old.copy_(self.flat.master_flat)
         t0 = time.perf_counter()
         losses = [self.local_step() for _ in range(self.cfg.local_steps)]
