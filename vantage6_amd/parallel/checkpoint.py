@@ -26,6 +26,12 @@ def save_checkpoint(trainer, path: str | Path) -> None:
         }
         if trainer.flat.exp_avg_sq is not None:
             tensors["exp_avg_sq"] = trainer.flat.exp_avg_sq.cpu().contiguous()
+        # server-optimizer state (FedAvgM momentum / FedAdam moments) —
+        # without it a resumed run silently restarts the server recurrence
+        if getattr(trainer, "_srv_v", None) is not None:
+            tensors["srv_v"] = trainer._srv_v.cpu().contiguous()
+        if getattr(trainer, "_srv_m", None) is not None:
+            tensors["srv_m"] = trainer._srv_m.cpu().contiguous()
         save_file(tensors, str(path))
         if trainer.flat._step_dev is not None:  # device counter is authoritative
             trainer.flat._step = int(trainer.flat._step_dev.item())
@@ -51,6 +57,15 @@ def load_checkpoint(trainer, path: str | Path) -> int:
         if trainer.flat.exp_avg_sq is None:
             trainer.flat.exp_avg_sq = torch.zeros_like(trainer.flat.master_flat)
         trainer.flat.exp_avg_sq.copy_(tensors["exp_avg_sq"].to(dev))
+    if "srv_v" in tensors:
+        if getattr(trainer, "_srv_v", None) is None:
+            trainer._srv_v = torch.zeros_like(trainer.flat.master_flat)
+            trainer._srv_old = torch.empty_like(trainer.flat.master_flat)
+        trainer._srv_v.copy_(tensors["srv_v"].to(dev))
+    if "srv_m" in tensors:
+        if getattr(trainer, "_srv_m", None) is None:
+            trainer._srv_m = torch.zeros_like(trainer.flat.master_flat)
+        trainer._srv_m.copy_(tensors["srv_m"].to(dev))
     trainer.flat._step = meta.get("step", 0)
     # all ranks read the same file in tests; in production rank 0's state is
     # authoritative — broadcast ALL optimizer state (incl. second moment and
@@ -59,6 +74,10 @@ def load_checkpoint(trainer, path: str | Path) -> int:
     trainer.comm.broadcast(trainer.flat.momentum_flat, src=0)
     if trainer.flat.exp_avg_sq is not None:
         trainer.comm.broadcast(trainer.flat.exp_avg_sq, src=0)
+    if getattr(trainer, "_srv_v", None) is not None:
+        trainer.comm.broadcast(trainer._srv_v, src=0)
+    if getattr(trainer, "_srv_m", None) is not None:
+        trainer.comm.broadcast(trainer._srv_m, src=0)
     step_t = torch.tensor([trainer.flat._step], dtype=torch.long, device=dev)
     trainer.comm.broadcast(step_t, src=0)
     trainer.flat._step = int(step_t.item())
