@@ -85,3 +85,27 @@ def test_hip_gate_never_fires_for_bad_shapes(monkeypatch):
     assert v.shape == (6,)
     # small-vocab classifier logits (resnet/logreg) always fall back
     ce_mod.cross_entropy(torch.randn(4, 8), torch.randint(0, 8, (4,)))
+
+
+def test_bert_vocab_pad_loss_equivalence():
+    """bert_base(vocab_pad=...) pads only the head; with -inf masking the
+    loss is identical to the unpadded model given the same head weights
+    (pad classes get probability 0 and zero grads)."""
+    from vantage6_amd.models.factory import build_model, model_loss
+
+    torch.manual_seed(0)
+    kw = dict(hidden=32, layers=1, heads=2, ffn=64, max_seq=16, vocab=50)
+    a = build_model("bert_base", dtype=torch.float32, **kw)
+    b = build_model("bert_base", dtype=torch.float32, vocab_pad=56, **kw)
+    sd = {k: v for k, v in a.state_dict().items() if k != "head.weight"}
+    b.load_state_dict(sd, strict=False)
+    with torch.no_grad():
+        b.head.weight[:50].copy_(a.head.weight)
+    assert b.head.weight.shape[0] == 56 and b.vocab_logical == 50
+    loss_fn = model_loss("bert_base")
+    batch = {"input_ids": torch.randint(0, 50, (2, 9))}
+    la = loss_fn(a, batch)
+    lb = loss_fn(b, batch)
+    assert torch.allclose(la, lb, atol=1e-6), (la.item(), lb.item())
+    lb.backward()
+    assert b.head.weight.grad[50:].abs().max().item() == 0.0

@@ -39,6 +39,12 @@ def model_loss(name: str) -> Callable:
 
             ids = batch["input_ids"]
             logits = model(ids[:, :-1])
+            vl = getattr(model, "vocab_logical", logits.shape[-1])
+            if vl < logits.shape[-1]:
+                # padded LM head (BertBase vocab_pad): pad classes get
+                # probability exactly 0 and exactly-zero grads
+                logits = logits.clone()
+                logits[..., vl:] = float("-inf")
             return cross_entropy(logits, ids[:, 1:])
 
         return lm_loss

@@ -56,13 +56,20 @@ class BertLayer(nn.Module):
 
 class BertBase(nn.Module):
     def __init__(self, vocab: int = 30522, hidden: int = 768, layers: int = 12,
-                 heads: int = 12, ffn: int = 3072, max_seq: int = 512):
+                 heads: int = 12, ffn: int = 3072, max_seq: int = 512,
+                 vocab_pad: int | None = None):
+        """``vocab_pad`` (megatron-style) pads ONLY the LM head's output
+        to a %8 width (e.g. 30720) so the fused bf16 cross-entropy kernel
+        can engage; pad logits are masked to -inf in the loss, so the
+        distribution over real classes is mathematically identical
+        (exp(-inf)=0). Default off — a round-3 measurement lead."""
         super().__init__()
+        self.vocab_logical = vocab
         self.tok = GraphSafeEmbedding(vocab, hidden)
         self.pos = GraphSafeEmbedding(max_seq, hidden)
         self.ln = LayerNorm(hidden)
         self.blocks = nn.ModuleList(BertLayer(hidden, heads, ffn) for _ in range(layers))
-        self.head = MfmaLinear(hidden, vocab, bias=False)
+        self.head = MfmaLinear(hidden, vocab_pad or vocab, bias=False)
 
     def forward(self, input_ids):
         B, S = input_ids.shape
