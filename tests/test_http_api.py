@@ -515,3 +515,37 @@ def test_auth_rate_limit_429(net):
     assert "retry-after" in {k.lower() for k in r.headers}
     # non-auth endpoints are not limited
     assert client.get("/api/version").status_code == 200
+
+
+def test_entity_crud_over_http(server):
+    """The new PATCH/DELETE routes work over the HTTP gateway too."""
+    http = UserClient(url=server)
+    http.authenticate("root", "root")
+    org = http.organization.create(name="http-crud-org")
+    upd = http.organization.update(org["id"], name="http-crud-org2",
+                                   country="DE")
+    assert upd["name"] == "http-crud-org2" and upd["country"] == "DE"
+    c = http.collaboration.create(name="http-crud-c",
+                                  organization_ids=[org["id"]])
+    assert http.collaboration.update(c["id"], name="http-crud-c2")["name"] \
+        == "http-crud-c2"
+    st = http.study.create(name="http-st", collaboration_id=c["id"],
+                           organization_ids=[org["id"]])
+    http.study.update(st["id"], name="http-st2")
+    http.study.delete(st["id"])
+    n = http.node.create(name="http-n", collaboration_id=c["id"],
+                         organization_id=org["id"])
+    rot = http.node.reset_api_key(n["id"])
+    assert rot["api_key"] and rot["id"] == n["id"]
+    # guarded delete over HTTP: 400 until forced
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError, match="dependents"):
+        http.collaboration.delete(c["id"])
+    http.collaboration.delete(c["id"], delete_dependents=True)
+    # give the org a dependent of its own (the collab cascade removed its
+    # node) so the org guard actually fires
+    http.user.create("http-crud-u", "Str0ngPass1", organization_id=org["id"],
+                     roles=["Researcher"])
+    with _pytest.raises(RuntimeError, match="dependents"):
+        http.organization.delete(org["id"])
+    http.organization.delete(org["id"], delete_dependents=True)
