@@ -646,3 +646,42 @@ def test_federated_anova_matches_scipy(tmp_path):
     assert res["F"] == pytest.approx(ref.statistic, rel=1e-9)
     assert res["p_value"] == pytest.approx(ref.pvalue, rel=1e-8)
     assert res["df_between"] == 2
+
+
+def test_federated_chi2_matches_scipy(tmp_path):
+    """Chi-square independence on the pooled contingency table equals
+    scipy.chi2_contingency."""
+    import numpy as np
+    import pandas as pd
+    from scipy import stats
+
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+
+    rng = np.random.default_rng(71)
+    frames, paths = [], []
+    for i in range(2):
+        n = 120 + 40 * i
+        g = rng.choice(["x", "y"], n)
+        o = np.where((g == "x") & (rng.random(n) < 0.6), "hi", "lo")
+        df = pd.DataFrame({"g": g, "o": o})
+        p = tmp_path / f"chi{i}.csv"
+        df.to_csv(p, index=False)
+        frames.append(df)
+        paths.append(p)
+    pooled = pd.concat(frames)
+    ct = pd.crosstab(pooled.g, pooled.o)
+    ref = stats.chi2_contingency(ct.to_numpy(), correction=False)
+    net = DevNetwork(n_stations=2, databases=[
+        [DatabaseConfig(label="default", uri=str(p), type="csv")] for p in paths])
+    with net:
+        task = net.client.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],
+            name="chi2", image="algorithms.crosstab",
+            input_={"method": "central_chi2",
+                    "kwargs": {"row_col": "g", "col_col": "o"}})
+        (res,) = net.client.wait_for_results(task["id"])
+    assert res["chi2"] == pytest.approx(ref.statistic, rel=1e-10)
+    assert res["p_value"] == pytest.approx(ref.pvalue, rel=1e-9)
+    assert np.array(res["table"]).sum() == len(pooled)
