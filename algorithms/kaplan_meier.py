@@ -61,3 +61,77 @@ def partial_event_counts(df, time_col: str, event_col: str, min_rows: int = 1):
         "events": {float(t): int(v) for t, v in ev.items()},
         "censored": {float(t): int(v) for t, v in ce.items()},
     }
+
+
+@algorithm_client
+def central_logrank(client, time_col: str, event_col: str, group_col: str,
+                    min_rows: int = 1):
+    """Federated two-group log-rank test (matches scipy.stats.logrank).
+
+    Stations share the same per-timepoint event/censor counts as the KM
+    curve, split by group. At each pooled event time: O-E for group 1 is
+    ``d1 - d*n1/n`` with hypergeometric variance
+    ``d*(n1/n)*(n2/n)*(n-d)/(n-1)``; the statistic is
+    ``(sum(O-E))^2 / sum(V)`` ~ chi2(1)."""
+    from scipy import stats as _st
+
+    org_ids = [org["id"] for org in client.organization.list()]
+    task = client.task.create(
+        input_={"method": "partial_event_counts_grouped",
+                "kwargs": {"time_col": time_col, "event_col": event_col,
+                           "group_col": group_col, "min_rows": min_rows}},
+        organizations=org_ids,
+        name="partial log-rank counts",
+    )
+    results = client.wait_for_results(task["id"])
+    levels = sorted({lv for r in results for lv in r["groups"]})
+    if len(levels) != 2:
+        raise ValueError(f"log-rank here is two-group; got {levels}")
+    ev = {lv: {} for lv in levels}
+    ce = {lv: {} for lv in levels}
+    n_at_risk = {lv: 0 for lv in levels}
+    for r in results:
+        for lv, g in r["groups"].items():
+            n_at_risk[lv] += g["n"]
+            for t, d in g["events"].items():
+                ev[lv][float(t)] = ev[lv].get(float(t), 0) + d
+            for t, c in g["censored"].items():
+                ce[lv][float(t)] = ce[lv].get(float(t), 0) + c
+    times = sorted(set().union(*(set(ev[lv]) | set(ce[lv]) for lv in levels)))
+    a, b = levels
+    o_minus_e = 0.0
+    var = 0.0
+    for t in times:
+        d1, d2 = ev[a].get(t, 0), ev[b].get(t, 0)
+        n1, n2 = n_at_risk[a], n_at_risk[b]
+        d, n = d1 + d2, n1 + n2
+        if d and n > 1:
+            o_minus_e += d1 - d * n1 / n
+            var += d * (n1 / n) * (n2 / n) * (n - d) / (n - 1)
+        n_at_risk[a] -= d1 + ce[a].get(t, 0)
+        n_at_risk[b] -= d2 + ce[b].get(t, 0)
+    if var <= 0:
+        raise ValueError("no events in common risk sets")
+    chi2 = o_minus_e * o_minus_e / var
+    return {"chi2": float(chi2),
+            "p_value": float(_st.chi2.sf(chi2, 1)),
+            "groups": levels}
+
+
+@data(1)
+def partial_event_counts_grouped(df, time_col: str, event_col: str,
+                                 group_col: str, min_rows: int = 1):
+    if len(df) < min_rows:
+        raise PrivacyThresholdViolation(
+            f"shard has {len(df)} rows < privacy threshold {min_rows}")
+    sub = df[[time_col, event_col, group_col]].dropna()
+    out = {}
+    for lv, g in sub.groupby(group_col):
+        ev = g[g[event_col] == 1].groupby(time_col).size()
+        ce = g[g[event_col] == 0].groupby(time_col).size()
+        out[str(lv)] = {
+            "n": int(len(g)),
+            "events": {float(t): int(v) for t, v in ev.items()},
+            "censored": {float(t): int(v) for t, v in ce.items()},
+        }
+    return {"groups": out}

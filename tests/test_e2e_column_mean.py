@@ -685,3 +685,53 @@ def test_federated_chi2_matches_scipy(tmp_path):
     assert res["chi2"] == pytest.approx(ref.statistic, rel=1e-10)
     assert res["p_value"] == pytest.approx(ref.pvalue, rel=1e-9)
     assert np.array(res["table"]).sum() == len(pooled)
+
+
+def test_federated_logrank_matches_scipy(tmp_path):
+    """Two-group log-rank from pooled per-time counts equals
+    scipy.stats.logrank on the pooled rows."""
+    import numpy as np
+    import pandas as pd
+    from scipy import stats
+
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+
+    rng = np.random.default_rng(83)
+    frames, paths = [], []
+    for i in range(2):
+        n = 60 + 20 * i
+        t_a = rng.exponential(10, n).round(1) + 0.1
+        t_b = rng.exponential(14, n).round(1) + 0.1
+        cens = rng.random(2 * n) < 0.25
+        df = pd.DataFrame({
+            "t": np.concatenate([t_a, t_b]),
+            "e": (~cens).astype(int),
+            "arm": ["a"] * n + ["b"] * n,
+        })
+        p = tmp_path / f"lr{i}.csv"
+        df.to_csv(p, index=False)
+        frames.append(df)
+        paths.append(p)
+    pooled = pd.concat(frames)
+
+    def surv(arm):
+        sub = pooled[pooled.arm == arm]
+        return stats.CensoredData.right_censored(
+            sub.t.to_numpy(), (sub.e == 0).to_numpy())
+
+    ref = stats.logrank(surv("a"), surv("b"))
+    net = DevNetwork(n_stations=2, databases=[
+        [DatabaseConfig(label="default", uri=str(p), type="csv")] for p in paths])
+    with net:
+        task = net.client.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],
+            name="logrank", image="algorithms.kaplan_meier",
+            input_={"method": "central_logrank",
+                    "kwargs": {"time_col": "t", "event_col": "e",
+                               "group_col": "arm"}})
+        (res,) = net.client.wait_for_results(task["id"])
+    # scipy's statistic is the signed sqrt(chi2) (normal form)
+    assert res["chi2"] == pytest.approx(ref.statistic ** 2, rel=1e-9)
+    assert res["p_value"] == pytest.approx(ref.pvalue, rel=1e-7)
