@@ -735,3 +735,57 @@ def test_federated_logrank_matches_scipy(tmp_path):
     # scipy's statistic is the signed sqrt(chi2) (normal form)
     assert res["chi2"] == pytest.approx(ref.statistic ** 2, rel=1e-9)
     assert res["p_value"] == pytest.approx(ref.pvalue, rel=1e-7)
+
+
+def test_federated_glm_families(tmp_path):
+    """Gaussian GLM equals the pooled OLS normal equation exactly;
+    Poisson GLM matches sklearn's PoissonRegressor (alpha=0) MLE."""
+    import numpy as np
+    import pandas as pd
+    from sklearn.linear_model import PoissonRegressor
+
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+
+    rng = np.random.default_rng(91)
+    frames, paths = [], []
+    for i in range(2):
+        n = 80 + 20 * i
+        x1 = rng.normal(0, 1, n)
+        x2 = rng.normal(0, 1, n)
+        ygauss = 2.0 + 1.5 * x1 - 0.7 * x2 + rng.normal(0, 0.3, n)
+        ypois = rng.poisson(np.exp(0.4 + 0.5 * x1 - 0.3 * x2))
+        df = pd.DataFrame({"x1": x1, "x2": x2, "yg": ygauss, "yp": ypois})
+        p = tmp_path / f"glm{i}.csv"
+        df.to_csv(p, index=False)
+        frames.append(df)
+        paths.append(p)
+    pooled = pd.concat(frames)
+    X = np.column_stack([np.ones(len(pooled)), pooled.x1, pooled.x2])
+    ols = np.linalg.solve(X.T @ X, X.T @ pooled.yg.to_numpy())
+    sk = PoissonRegressor(alpha=0.0, max_iter=2000, tol=1e-12).fit(
+        pooled[["x1", "x2"]], pooled.yp)
+    net = DevNetwork(n_stations=2, databases=[
+        [DatabaseConfig(label="default", uri=str(p), type="csv")] for p in paths])
+    with net:
+        def fit(family, target):
+            task = net.client.task.create(
+                collaboration=net.collaboration["id"],
+                organizations=[net.org_ids[0]],
+                name=f"glm-{family}", image="algorithms.glm_logistic",
+                input_={"method": "central_glm",
+                        "kwargs": {"feature_cols": ["x1", "x2"],
+                                   "target_col": target, "family": family}})
+            (res,) = net.client.wait_for_results(task["id"])
+            return res
+
+        g = fit("gaussian", "yg")
+        assert g["converged"] and g["n"] == len(pooled)
+        assert g["intercept"] == pytest.approx(ols[0], abs=1e-8)
+        assert g["coefficients"]["x1"] == pytest.approx(ols[1], abs=1e-8)
+        assert g["coefficients"]["x2"] == pytest.approx(ols[2], abs=1e-8)
+        p_ = fit("poisson", "yp")
+        assert p_["converged"]
+        assert p_["intercept"] == pytest.approx(sk.intercept_, abs=2e-5)
+        assert p_["coefficients"]["x1"] == pytest.approx(sk.coef_[0], abs=2e-5)
+        assert p_["coefficients"]["x2"] == pytest.approx(sk.coef_[1], abs=2e-5)
