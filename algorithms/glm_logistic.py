@@ -137,10 +137,30 @@ def central_glm(
                         "max_step": float(np.abs(step).max())})
         if float(np.abs(step).max()) < tol:
             break
+    # Wald inference from the converged information matrix: cov = phi*H^-1
+    # (phi = 1 for poisson/binomial; gaussian dispersion = RSS/(n-p) from
+    # the stations' pooled residual sum of squares)
+    from scipy import stats as _st
+
+    cov = np.linalg.inv(H + 1e-9 * np.eye(p))
+    if family == "gaussian":
+        rss = sum(r.get("rss", 0.0) for r in results)
+        dof = max(n_total - p, 1)
+        cov = cov * (rss / dof)
+        se = np.sqrt(np.diag(cov))
+        tv = np.asarray(beta) / se
+        pv = 2.0 * _st.t.sf(np.abs(tv), dof)
+    else:
+        se = np.sqrt(np.diag(cov))
+        tv = np.asarray(beta) / se
+        pv = 2.0 * _st.norm.sf(np.abs(tv))
+    names = ["(intercept)"] + list(feature_cols)
     return {
         "family": family,
         "coefficients": dict(zip(feature_cols, beta[1:])),
         "intercept": beta[0],
+        "std_errors": dict(zip(names, se.tolist())),
+        "p_values": dict(zip(names, pv.tolist())),
         "n": n_total,
         "iterations": len(history),
         "converged": history[-1]["max_step"] < tol,
@@ -170,4 +190,5 @@ def partial_glm_irls(df, feature_cols: list[str], target_col: str,
         "hessian": (X.T * w) @ X,
         "gradient": X.T @ (y - mu),
         "n": int(len(sub)),
+        "rss": float(((y - mu) ** 2).sum()),
     }

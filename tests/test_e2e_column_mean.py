@@ -789,3 +789,45 @@ def test_federated_glm_families(tmp_path):
         assert p_["intercept"] == pytest.approx(sk.intercept_, abs=2e-5)
         assert p_["coefficients"]["x1"] == pytest.approx(sk.coef_[0], abs=2e-5)
         assert p_["coefficients"]["x2"] == pytest.approx(sk.coef_[1], abs=2e-5)
+
+
+def test_federated_glm_standard_errors(tmp_path):
+    """Gaussian GLM Wald SEs equal the OLS closed form
+    sqrt(diag((X'X)^-1) * RSS/(n-p))."""
+    import numpy as np
+    import pandas as pd
+
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+
+    rng = np.random.default_rng(97)
+    frames, paths = [], []
+    for i in range(2):
+        n = 70
+        x1 = rng.normal(0, 1, n)
+        y = 1.0 + 0.8 * x1 + rng.normal(0, 0.5, n)
+        df = pd.DataFrame({"x1": x1, "y": y})
+        p = tmp_path / f"se{i}.csv"
+        df.to_csv(p, index=False)
+        frames.append(df)
+        paths.append(p)
+    pooled = pd.concat(frames)
+    X = np.column_stack([np.ones(len(pooled)), pooled.x1])
+    bhat = np.linalg.solve(X.T @ X, X.T @ pooled.y.to_numpy())
+    rss = float(((pooled.y.to_numpy() - X @ bhat) ** 2).sum())
+    cov = np.linalg.inv(X.T @ X) * rss / (len(pooled) - 2)
+    want_se = np.sqrt(np.diag(cov))
+    net = DevNetwork(n_stations=2, databases=[
+        [DatabaseConfig(label="default", uri=str(p), type="csv")] for p in paths])
+    with net:
+        task = net.client.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],
+            name="glm-se", image="algorithms.glm_logistic",
+            input_={"method": "central_glm",
+                    "kwargs": {"feature_cols": ["x1"], "target_col": "y",
+                               "family": "gaussian"}})
+        (res,) = net.client.wait_for_results(task["id"])
+    assert res["std_errors"]["(intercept)"] == pytest.approx(want_se[0], rel=1e-6)
+    assert res["std_errors"]["x1"] == pytest.approx(want_se[1], rel=1e-6)
+    assert res["p_values"]["x1"] < 1e-10  # strong true effect
