@@ -26,6 +26,14 @@ class RoundMetrics:
                 "agg_ms": pc.Gauge("v6amd_agg_ms", "aggregation ms (last round)"),
                 "round_ms": pc.Gauge("v6amd_round_ms", "total round ms"),
                 "loss": pc.Gauge("v6amd_loss", "mean local loss (last round)"),
+                "overlap_pct": pc.Gauge(
+                    "v6amd_overlap_pct",
+                    "fraction of the aggregation collective hidden under "
+                    "the next round's forward (fit/agg overlap)"),
+                "agg_comm_ms": pc.Gauge(
+                    "v6amd_agg_comm_ms", "collective wall ms (last round)"),
+                "hbm_used": pc.Gauge(
+                    "v6amd_hbm_used_bytes", "torch HBM allocated on this rank"),
             }
             pc.start_http_server(prometheus_port)
 
@@ -37,6 +45,12 @@ class RoundMetrics:
             self._prom["agg_ms"].set(stats.get("agg_ms", 0))
             self._prom["round_ms"].set(stats.get("round_ms", 0))
             self._prom["loss"].set(stats.get("loss", 0))
+            self._prom["overlap_pct"].set(stats.get("overlap_pct", 0))
+            self._prom["agg_comm_ms"].set(stats.get("agg_comm_ms", 0))
+            import torch
+
+            if torch.cuda.is_available():
+                self._prom["hbm_used"].set(torch.cuda.memory_allocated())
 
     def table(self, last: int = 20) -> str:
         overlap = any("overlap_pct" in r for r in self.rows)
