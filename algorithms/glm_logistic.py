@@ -57,9 +57,18 @@ def central_logistic_regression(
                         "max_step": float(np.abs(step).max())})
         if float(np.abs(step).max()) < tol:
             break
+    from scipy import stats as _st
+
+    cov = np.linalg.inv(H + 1e-9 * np.eye(p))
+    se = np.sqrt(np.diag(cov))
+    zv = np.asarray(beta) / se
+    names = ["(intercept)"] + list(feature_cols)
     return {
         "coefficients": dict(zip(feature_cols, beta[1:])),
         "intercept": beta[0],
+        "std_errors": dict(zip(names, se.tolist())),
+        "p_values": dict(zip(names,
+                             (2.0 * _st.norm.sf(np.abs(zv))).tolist())),
         "n": n_total,
         "iterations": len(history),
         "converged": history[-1]["max_step"] < tol,
