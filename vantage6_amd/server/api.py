@@ -419,7 +419,11 @@ def create_app(core: ServerCore, auth_rate_limit: int = 30,
 
     @app.post("/api/recover/node")
     def node_recover(body: dict, who: Identity = Depends(ident)):
-        return wrap(core.reset_node_api_key, who, int(body["id"]))
+        try:
+            node_id = int(body["id"])
+        except (KeyError, TypeError, ValueError):
+            raise HTTPException(400, "body needs an integer 'id'")
+        return wrap(core.reset_node_api_key, who, node_id)
 
     @app.get("/api/collaboration/{collab_id}")
     def collaboration_get(collab_id: int, who: Identity = Depends(ident)):
@@ -547,17 +551,18 @@ def create_app(core: ServerCore, auth_rate_limit: int = 30,
 
     @app.post("/api/recover/lost")
     def recover_lost(body: RecoverBody):
-        mail_mod.request_password_recovery(core, mail_service, body.username)
+        wrap(mail_mod.request_password_recovery, core, mail_service,
+             body.username)
         return {"msg": "if the user exists, a recovery email was sent"}
 
     @app.post("/api/recover/reset")
     def recover_reset(body: RecoverBody):
-        mail_mod.reset_password(core, body.token, body.password)
+        wrap(mail_mod.reset_password, core, body.token, body.password)
         return {"msg": "password reset"}
 
     @app.post("/api/recover/2fa/lost")
     def recover_2fa_lost(body: RecoverBody):
-        mail_mod.request_mfa_reset(core, mail_service, body.username)
+        wrap(mail_mod.request_mfa_reset, core, mail_service, body.username)
         return {"msg": "if the user exists, an mfa-reset email was sent"}
 
     @app.post("/api/recover/2fa/reset")
@@ -572,10 +577,13 @@ def create_app(core: ServerCore, auth_rate_limit: int = 30,
     # ----------------------------------------------------------------- ops
     @app.post("/api/ping")
     def ping(body: dict, who: Identity = Depends(ident)):
-        if (body or {}).get("offline"):
-            core.node_set_offline(who.id)
+        body = body if isinstance(body, dict) else {}
+        if body.get("offline"):
+            wrap(core.node_set_offline, who.id)
         else:
-            core.node_ping(who, (body or {}).get("info") or None)
+            info = body.get("info")
+            wrap(core.node_ping, who,
+                 info if isinstance(info, dict) and info else None)
         return {"msg": "pong"}
 
     @app.get("/api/version")
