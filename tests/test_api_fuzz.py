@@ -71,3 +71,67 @@ def test_no_endpoint_returns_500():
                         resp.text[:200]))
     assert not bad, bad[:5]
     core.close()
+
+
+@pytest.mark.timeout(300)
+def test_station_survives_garbage_tasks():
+    """Malformed task inputs (junk b64, non-dict payloads, wrong kwarg
+    types, unknown methods/images) mark their runs failed — and the
+    station worker survives to run a good task afterwards."""
+    import base64 as b64
+    import time
+
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+    from vantage6_amd.common.serialization import serialize
+
+    dbs = [[DatabaseConfig(label="default",
+                           uri="synthetic://tabular?n=64&d=4&seed=0",
+                           type="synthetic")]]
+    with DevNetwork(n_stations=1, databases=dbs, name="fuzznet") as net:
+        cl = net.client
+        cases = [
+            ("not even base64 °°°", "algorithms.column_mean"),
+            (b64.b64encode(b"not json").decode(), "algorithms.column_mean"),
+            (b64.b64encode(serialize([1, 2, 3])).decode(),
+             "algorithms.column_mean"),                      # non-dict payload
+            (b64.b64encode(serialize({"no_method": 1})).decode(),
+             "algorithms.column_mean"),
+            (b64.b64encode(serialize({"method": "partial_mean",
+                                      "kwargs": {"columns": 42}})).decode(),
+             "algorithms.column_mean"),                      # wrong kwarg type
+            (b64.b64encode(serialize({"method": "nope"})).decode(),
+             "algorithms.column_mean"),
+            (b64.b64encode(serialize({"method": "partial_mean"})).decode(),
+             "algorithms.does_not_exist"),                   # bad image
+        ]
+        ids = []
+        for i, (inp, image) in enumerate(cases):
+            t = cl.task.create(collaboration=net.collaboration["id"],
+                               organizations=[{"id": net.org_ids[0],
+                                               "input": inp}],
+                               name=f"junk{i}", image=image, input_=None)
+            ids.append(t["id"])
+        deadline = time.monotonic() + 60
+        while time.monotonic() < deadline:
+            terminal = {"failed", "no image", "crashed", "not allowed",
+                        "start failed", "non-existing algorithm",
+                        "killed by user"}
+            st = {cl.task.get(tid)["status"] for tid in ids}
+            if st <= terminal:
+                break
+            time.sleep(0.1)
+        for tid in ids:
+            status = cl.task.get(tid)["status"]
+            assert status in terminal, (tid, status)
+        # worker is still alive: a good task completes
+        good = cl.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]], name="good",
+            image="algorithms.fedavg",
+            input_={"method": "partial_fit", "kwargs": {
+                "model": "logreg",
+                "model_kwargs": {"in_features": 4, "num_classes": 2},
+                "local_steps": 1, "batch_size": 16}})
+        (res,) = cl.wait_for_results(good["id"], timeout=60)
+        assert "n" in res or res

@@ -217,18 +217,27 @@ class UserClient:
             study: int | None = None,
         ) -> dict:
             p = self.parent
-            blob = serialize(input_)
             orgs = []
             encrypted = p.collaboration.get(collaboration)["encrypted"]
-            for org_id in organizations:
+            blob = serialize(input_) if input_ is not None else None
+            for org in organizations:
+                # an org entry may be a bare id (input_ is serialized and
+                # encrypted per org here) or a prepared {"id", "input"}
+                # dict (caller did its own serialization/encryption —
+                # the reference's node/proxy shape)
+                if isinstance(org, dict):
+                    orgs.append({"id": org["id"], "input": org.get("input", "")})
+                    continue
+                if blob is None:
+                    raise ValueError("input_ is required for bare org ids")
                 if encrypted:
-                    pubkey = p.organization.get(org_id)["public_key"]
+                    pubkey = p.organization.get(org)["public_key"]
                     if not pubkey:
-                        raise RuntimeError(f"organization {org_id} has no public key")
+                        raise RuntimeError(f"organization {org} has no public key")
                     enc = p.cryptor.encrypt_bytes_to_str(blob, pubkey)
                 else:
                     enc = bytes_to_base64s(blob)
-                orgs.append({"id": org_id, "input": enc})
+                orgs.append({"id": org, "input": enc})
             body = {
                 "name": name,
                 "image": image,
@@ -237,7 +246,8 @@ class UserClient:
                 "databases": databases or [],
                 "study_id": study,
                 "organizations": orgs,
-                "method": str(input_.get("method", "")),
+                "method": str(input_.get("method", ""))
+                          if isinstance(input_, dict) else "",
             }
             return p._call("create_task", "POST", "task", json=body, **body)
 
