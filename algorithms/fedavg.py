@@ -149,6 +149,7 @@ def central_fedavg(
     batch_size: int = 16,
     lr: float = 0.1,
     min_stations: int = 1,
+    fedprox_mu: float = 0.0,  # FedProx proximal term on the local steps
     fail_orgs: list | None = None,  # fault injection (tests): orgs that crash
 ):
     """Central part: per round, fan out partial_fit and weighted-average.
@@ -169,7 +170,7 @@ def central_fedavg(
                     "model": model, "model_kwargs": model_kwargs,
                     "params": global_params, "local_steps": local_steps,
                     "batch_size": batch_size, "lr": lr, "seed": 1234,
-                    "fail_orgs": fail_orgs,
+                    "fedprox_mu": fedprox_mu, "fail_orgs": fail_orgs,
                 },
             },
             organizations=org_ids,
@@ -207,6 +208,7 @@ def _partial_fit_impl(
     batch_size: int = 16,
     lr: float = 0.1,
     seed: int = 1234,
+    fedprox_mu: float = 0.0,
     fail_orgs: list | None = None,
     _v6_env: dict | None = None,
 ):
@@ -222,12 +224,20 @@ def _partial_fit_impl(
     dataset = db if isinstance(db, dict) else {"x": db}
     batches = make_batches(dataset, batch_size, device)
     opt = torch.optim.SGD(net.parameters(), lr=lr, momentum=0.9)
+    prox_ref = ([p.detach().clone() for p in net.parameters()]
+                if fedprox_mu > 0 else None)  # w_round = the received global
     losses = []
     for _ in range(local_steps):
         batch = next(batches)
         opt.zero_grad(set_to_none=True)
         loss = loss_fn(net, batch)
         loss.backward()
+        if prox_ref is not None:
+            with torch.no_grad():
+                for pp, r0 in zip(net.parameters(), prox_ref):
+                    if pp.grad is not None:
+                        pp.grad.add_((pp.detach() - r0).to(pp.grad.dtype),
+                                     alpha=fedprox_mu)
         opt.step()
         losses.append(float(loss.detach()))
     n = len(dataset[list(dataset.keys())[0]])

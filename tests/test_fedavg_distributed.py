@@ -314,3 +314,24 @@ def test_fedadam_gloo_two_ranks():
         r0, r1 = results[0], results[1]
     assert r0["err"] < 1e-6, r0["err"]
     assert r0["ck"] == r1["ck"]
+
+
+def test_partial_fit_fedprox_math():
+    """Station-mode partial_fit: with w=w_round the first prox step is
+    exactly plain SGD; a second step diverges under large mu."""
+    from algorithms.fedavg import _partial_fit_impl
+
+    x = torch.randn(48, 6, generator=torch.Generator().manual_seed(21))
+    db = {"x": x, "y": (x.sum(-1) > 0).long()}
+    mk = {"model": "logreg",
+          "model_kwargs": {"in_features": 6, "num_classes": 2},
+          "batch_size": 16, "lr": 0.1, "seed": 7}
+
+    def run(mu, steps):
+        out = _partial_fit_impl(db, local_steps=steps, fedprox_mu=mu, **mk)
+        return {k: v.clone() for k, v in out["params"].items()}
+
+    a1, b1 = run(0.0, 1), run(30.0, 1)
+    assert all(torch.equal(a1[k], b1[k]) for k in a1)
+    a2, b2 = run(0.0, 2), run(30.0, 2)
+    assert any(not torch.equal(a2[k], b2[k]) for k in a2)
