@@ -124,3 +124,22 @@ def test_crosstab_station_never_leaks_small_cells(labels, threshold):
     assert all(v >= threshold for v in out["counts"].values())
     true_cells = df["g"].value_counts()
     assert out["suppressed_cells"] == int((true_cells < threshold).sum())
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.lists(st.floats(min_value=-1e4, max_value=1e4, allow_nan=False,
+                          width=32), min_size=1, max_size=64),
+       st.sampled_from(["float32", "bfloat16", "int64"]))
+def test_tensor_roundtrip_all_dtypes(vals, dtype):
+    """Tagged tensor serialization round-trips bf16 (uint16 view), fp32
+    and int64 exactly."""
+    import torch
+
+    t = torch.tensor(vals)
+    if dtype == "int64":
+        t = t.long()
+    elif dtype == "bfloat16":
+        t = t.to(torch.bfloat16)
+    out = deserialize(serialize({"t": t}))["t"]
+    assert out.dtype == t.dtype and out.shape == t.shape
+    assert torch.equal(out, t)
