@@ -1,7 +1,7 @@
 """Property-based robustness: serialization and encryption round-trips
 hold for arbitrary payloads (hypothesis)."""
 import numpy as np
-from hypothesis import given, settings
+from hypothesis import example, given, settings
 from hypothesis import strategies as st
 
 from vantage6_amd.common.encryption import RSACryptor
@@ -24,6 +24,8 @@ payloads = st.recursive(
 
 @settings(max_examples=60, deadline=None)
 @given(payloads)
+@example(value={"__v6amd__": False})   # reserved-key collision (found by
+@example(value={"__v6amd__": "ndarray", "nested": {"__v6amd__": 1}})  # fuzz
 def test_serialize_roundtrip(value):
     assert deserialize(serialize(value)) == value
 
@@ -143,3 +145,44 @@ def test_tensor_roundtrip_all_dtypes(vals, dtype):
     out = deserialize(serialize({"t": t}))["t"]
     assert out.dtype == t.dtype and out.shape == t.shape
     assert torch.equal(out, t)
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.binary(min_size=0, max_size=200))
+def test_deserialize_garbage_fails_cleanly(blob):
+    """Adversarial bytes either decode to a value or raise a clean
+    ValueError-family error — never KeyError/AttributeError/hang (stations
+    deserialize attacker-supplied task inputs)."""
+    try:
+        deserialize(blob)
+    except (ValueError, UnicodeDecodeError):  # json/np errors subclass these
+        pass
+    except Exception as exc:  # pragma: no cover
+        from vantage6_amd.common.serialization import SerializationError
+
+        assert isinstance(exc, SerializationError), repr(exc)
+
+
+def test_deserialize_adversarial_tags_fail_cleanly():
+    """Crafted tag payloads (object dtype, absurd shapes, truncated data,
+    unknown tags) raise clean errors without huge allocations."""
+    import json
+
+    import pytest
+
+    from vantage6_amd.common.serialization import SerializationError
+
+    cases = [
+        {"__v6amd__": "ndarray", "dtype": "object", "shape": [1], "data": "AA=="},
+        {"__v6amd__": "ndarray", "dtype": "float32", "shape": [2**40],
+         "data": "AAAA"},                      # shape >> buffer
+        {"__v6amd__": "ndarray", "dtype": "not-a-dtype", "shape": [1],
+         "data": "AAAA"},
+        {"__v6amd__": "tensor", "dtype": "float32", "shape": [-3],
+         "data": "AAAA"},
+        {"__v6amd__": "totally-unknown", "data": ""},
+        {"__v6amd__": "bytes", "data": "!!!notb64!!!"},
+    ]
+    for c in cases:
+        with pytest.raises((ValueError, TypeError, SerializationError)):
+            deserialize(json.dumps(c).encode())

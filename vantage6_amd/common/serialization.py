@@ -73,6 +73,8 @@ def _decode(d: dict):
     tag = d.get("__v6amd__")
     if tag is None:
         return d
+    if tag == "escaped":  # user dict that contained the reserved key
+        return {k: v for k, v in d["items"]}
     if tag == "ndarray":
         arr = np.frombuffer(base64s_to_bytes(d["data"]), dtype=np.dtype(d["dtype"]))
         return arr.reshape(d["shape"]).copy()
@@ -100,9 +102,25 @@ def _decode(d: dict):
     raise SerializationError(f"unknown serialization tag {tag!r}")
 
 
+def _escape(o: Any) -> Any:
+    """User dicts that happen to contain the reserved tag key would be
+    mis-decoded as tagged payloads — wrap them as an 'escaped' tag whose
+    items ride as [key, value] PAIRS (a pairs list cannot collide with the
+    object_hook, which fires on every dict bottom-up)."""
+    if isinstance(o, dict):
+        walked = {k: _escape(v) for k, v in o.items()}
+        if "__v6amd__" in walked:
+            return {"__v6amd__": "escaped",
+                    "items": [[k, v] for k, v in walked.items()]}
+        return walked
+    if isinstance(o, (list, tuple)):
+        return [_escape(v) for v in o]
+    return o
+
+
 def serialize(data: Any) -> bytes:
     """Serialize an algorithm input/result to bytes (JSON, UTF-8)."""
-    return json.dumps(data, default=_encode).encode("utf-8")
+    return json.dumps(_escape(data), default=_encode).encode("utf-8")
 
 
 def deserialize(blob: bytes) -> Any:
