@@ -186,3 +186,24 @@ def test_deserialize_adversarial_tags_fail_cleanly():
     for c in cases:
         with pytest.raises((ValueError, TypeError, SerializationError)):
             deserialize(json.dumps(c).encode())
+
+
+_FUZZ_CRYPTOR = None
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.text(max_size=120))
+def test_decrypt_garbage_fails_cleanly(envelope):
+    """Random envelope strings never crash decrypt with anything but
+    ValueError (stations decrypt attacker-reachable inputs)."""
+    global _FUZZ_CRYPTOR
+    if _FUZZ_CRYPTOR is None:
+        import tempfile
+        from pathlib import Path
+
+        _FUZZ_CRYPTOR = RSACryptor(
+            str(Path(tempfile.mkdtemp()) / "fz.pem"), bits=1024)
+    try:
+        _FUZZ_CRYPTOR.decrypt_str_to_bytes(envelope)
+    except ValueError:
+        pass
