@@ -207,3 +207,10 @@ def test_decrypt_garbage_fails_cleanly(envelope):
         _FUZZ_CRYPTOR.decrypt_str_to_bytes(envelope)
     except ValueError:
         pass
+
+
+def test_totp_corrupt_secret_returns_false():
+    from vantage6_amd.common.totp import verify_code
+
+    assert verify_code("not!!base32", "123456") is False
+    assert verify_code("", "123456") is False

@@ -32,11 +32,16 @@ def totp_code(secret: str, at: float | None = None, period: int = 30,
 
 def verify_code(secret: str, code: str, at: float | None = None,
                 window: int = 1) -> bool:
-    """Accept codes from ±window periods (clock skew)."""
+    """Accept codes from ±window periods (clock skew). A corrupted
+    stored secret (invalid base32) verifies as False instead of raising —
+    login must answer 401, never 500."""
     now = at if at is not None else time.time()
-    for w in range(-window, window + 1):
-        if hmac.compare_digest(totp_code(secret, now + w * 30), str(code)):
-            return True
+    try:
+        for w in range(-window, window + 1):
+            if hmac.compare_digest(totp_code(secret, now + w * 30), str(code)):
+                return True
+    except Exception:  # noqa: BLE001 — undecodable secret == no valid code
+        return False
     return False
 
 
