@@ -46,5 +46,8 @@ def verify_code(secret: str, code: str, at: float | None = None,
 
 
 def provisioning_uri(secret: str, username: str, issuer: str = "vantage6_amd") -> str:
-    return (f"otpauth://totp/{issuer}:{username}?secret={secret}"
-            f"&issuer={issuer}&algorithm=SHA1&digits=6&period=30")
+    from urllib.parse import quote
+
+    u, i = quote(username, safe=""), quote(issuer, safe="")
+    return (f"otpauth://totp/{i}:{u}?secret={secret}"
+            f"&issuer={i}&algorithm=SHA1&digits=6&period=30")
