@@ -270,3 +270,23 @@ def test_password_policy_enforced_by_default():
     u2 = UserClient(core=core)
     u2.authenticate("pol-u", "An0therStrong")
     core.close()
+
+
+def test_username_validation():
+    """Control characters / empty / padded usernames are rejected (log
+    and UI injection hygiene)."""
+    import pytest
+
+    from vantage6_amd.server.core import BadRequest, ServerCore
+    from vantage6_amd.client import UserClient
+
+    core = ServerCore(password_policy=False)
+    core.create_root_user("root", "root")
+    cl = UserClient(core=core)
+    cl.authenticate("root", "root")
+    org = cl.organization.create(name="un-org")
+    for bad in ("", "a\nb", "user\x00", " padded", "padded ", "x" * 200):
+        with pytest.raises((BadRequest, RuntimeError)):
+            cl.user.create(bad, "pw", organization_id=org["id"], roles=[])
+    cl.user.create("fine.user-1", "pw", organization_id=org["id"], roles=[])
+    core.close()

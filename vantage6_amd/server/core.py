@@ -671,6 +671,10 @@ class ServerCore:
         **kw,
     ) -> dict:
         self._check_user_can("user", Operation.CREATE, ident, org_id=organization_id)
+        if (not username or len(username) > 128 or not username.isprintable()
+                or username.strip() != username):
+            raise BadRequest("username must be 1-128 printable characters "
+                             "with no leading/trailing whitespace")
         self.validate_password(password)
         s = self.session
         if s.query(User).filter_by(username=username).first():
