@@ -58,3 +58,22 @@ def test_shell_command_exists_and_binds_namespace(monkeypatch):
     assert {"core", "session", "client", "models", "Task", "Run"} <= set(captured)
     # the bound client is authenticated and usable
     assert captured["client"].whoami["username"] == "root"
+
+
+def test_every_cli_command_help_works():
+    """--help for every command/group imports and renders (catches broken
+    imports in command bodies without running them)."""
+    from click.testing import CliRunner
+
+    from vantage6_amd.cli.main import cli
+
+    runner = CliRunner()
+    top = runner.invoke(cli, ["--help"])
+    assert top.exit_code == 0, top.output
+    import re
+
+    cmds = re.findall(r"^  ([a-z][a-z0-9-]*)", top.output, re.M)
+    assert len(cmds) >= 8, cmds
+    for c in cmds:
+        r = runner.invoke(cli, [c, "--help"])
+        assert r.exit_code == 0, (c, r.output)
