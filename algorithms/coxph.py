@@ -82,9 +82,23 @@ def central_coxph(
                         "max_step": float(np.abs(step).max())})
         if float(np.abs(step).max()) < tol:
             break
+    # Wald inference from the converged observed information (the same
+    # Hessian the Newton step used): cov = H^-1; HR CI = exp(b ± 1.96 se)
+    from scipy import stats as _st
+
+    cov = np.linalg.inv(hess + 1e-10 * np.eye(p))
+    se = np.sqrt(np.diag(cov))
+    zv = beta / se
+    pv = 2.0 * _st.norm.sf(np.abs(zv))
+    ci_lo = np.exp(beta - 1.959963984540054 * se)
+    ci_hi = np.exp(beta + 1.959963984540054 * se)
     return {
         "coefficients": dict(zip(feature_cols, beta.tolist())),
         "hazard_ratios": dict(zip(feature_cols, np.exp(beta).tolist())),
+        "std_errors": dict(zip(feature_cols, se.tolist())),
+        "p_values": dict(zip(feature_cols, pv.tolist())),
+        "hr_ci95": {c: [lo, hi] for c, lo, hi
+                    in zip(feature_cols, ci_lo.tolist(), ci_hi.tolist())},
         "iterations": len(history),
         "converged": history[-1]["max_step"] < tol,
         "n_event_times": len(times),

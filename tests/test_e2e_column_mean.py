@@ -357,6 +357,13 @@ def test_federated_coxph_matches_pooled_newton(tmp_path):
     assert res["coefficients"]["b"] == pytest.approx(ref_beta[1], abs=1e-6)
     # directional recovery of the simulated effects
     assert res["coefficients"]["a"] > 0.3 and res["coefficients"]["b"] < -0.2
+    # Wald inference sanity: positive SEs, CI brackets the HR, strong
+    # true effects are significant
+    for c in ("a", "b"):
+        se = res["std_errors"][c]
+        lo, hi = res["hr_ci95"][c]
+        assert se > 0 and lo < res["hazard_ratios"][c] < hi
+        assert res["p_values"][c] < 0.05
 
 
 def test_federated_crosstab(tmp_path):
