@@ -37,13 +37,28 @@ def central_kaplan_meier(client, time_col: str, event_col: str,
     times = sorted(set(events) | set(censored))
     at_risk = total
     surv = 1.0
+    gw = 0.0  # Greenwood accumulator: sum d/(n(n-d))
     curve = []
     for t in times:
         d = events.get(t, 0)
         if d and at_risk:
             surv *= 1.0 - d / at_risk
+            if at_risk > d:
+                gw += d / (at_risk * (at_risk - d))
+        # Greenwood variance + log-transformed 95% CI (the plain
+        # S ± 1.96·se clips at [0,1]; exp-log keeps it inside)
+        se = surv * gw ** 0.5
+        import math
+
+        if 0.0 < surv < 1.0 and gw > 0:
+            z = 1.959963984540054
+            theta = z * (gw ** 0.5) / abs(math.log(surv))
+            lo, hi = surv ** math.exp(theta), surv ** math.exp(-theta)
+        else:
+            lo = hi = surv
         curve.append({"time": t, "at_risk": at_risk, "events": d,
-                      "censored": censored.get(t, 0), "survival": surv})
+                      "censored": censored.get(t, 0), "survival": surv,
+                      "std_err": se, "ci95": [lo, hi]})
         at_risk -= d + censored.get(t, 0)
     return {"n": total, "curve": curve}
 

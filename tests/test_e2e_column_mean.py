@@ -838,3 +838,60 @@ def test_federated_glm_standard_errors(tmp_path):
     assert res["std_errors"]["(intercept)"] == pytest.approx(want_se[0], rel=1e-6)
     assert res["std_errors"]["x1"] == pytest.approx(want_se[1], rel=1e-6)
     assert res["p_values"]["x1"] < 1e-10  # strong true effect
+
+
+def test_km_greenwood_matches_manual(tmp_path):
+    """KM std_err equals the Greenwood formula computed by hand on the
+    pooled event table; CI brackets S(t)."""
+    import numpy as np
+    import pandas as pd
+
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+
+    rng = np.random.default_rng(101)
+    frames, paths = [], []
+    for i in range(2):
+        n = 50
+        df = pd.DataFrame({
+            "t": rng.exponential(8, n).round(1) + 0.1,
+            "e": (rng.random(n) < 0.8).astype(int),
+        })
+        p = tmp_path / f"gw{i}.csv"
+        df.to_csv(p, index=False)
+        frames.append(df)
+        paths.append(p)
+    # pool from the CSVs the stations actually read (text round-trip
+    # changes float reprs vs the in-memory frames)
+    pooled = pd.concat([pd.read_csv(p) for p in paths])
+    net = DevNetwork(n_stations=2, databases=[
+        [DatabaseConfig(label="default", uri=str(p), type="csv")] for p in paths])
+    with net:
+        task = net.client.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],
+            name="km-gw", image="algorithms.kaplan_meier",
+            input_={"method": "central_kaplan_meier",
+                    "kwargs": {"time_col": "t", "event_col": "e"}})
+        (res,) = net.client.wait_for_results(task["id"])
+    # manual pooled Greenwood
+    tab = pooled.sort_values("t")
+    at_risk = len(tab)
+    surv, gw = 1.0, 0.0
+    manual = {}
+    for t, grp in tab.groupby("t"):
+        d = int((grp.e == 1).sum())
+        c = int((grp.e == 0).sum())
+        if d and at_risk:
+            surv *= 1 - d / at_risk
+            if at_risk > d:
+                gw += d / (at_risk * (at_risk - d))
+        manual[float(t)] = surv * gw ** 0.5
+        at_risk -= d + c
+    checked = 0
+    for row in res["curve"]:
+        assert row["std_err"] == pytest.approx(manual[row["time"]], abs=1e-12)
+        lo, hi = row["ci95"]
+        assert lo <= row["survival"] <= hi
+        checked += 1
+    assert checked > 20
