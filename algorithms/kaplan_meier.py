@@ -9,6 +9,8 @@ central function merges them into the product-limit estimator
 Columns: ``time_col`` (follow-up time) and ``event_col`` (1 = event,
 0 = censored). ``min_rows`` guards small shards (disclosure check).
 """
+import math
+
 from vantage6_amd.algorithm import algorithm_client, data
 from vantage6_amd.algorithm.exceptions import PrivacyThresholdViolation
 
@@ -48,8 +50,6 @@ def central_kaplan_meier(client, time_col: str, event_col: str,
         # Greenwood variance + log-transformed 95% CI (the plain
         # S ± 1.96·se clips at [0,1]; exp-log keeps it inside)
         se = surv * gw ** 0.5
-        import math
-
         if 0.0 < surv < 1.0 and gw > 0:
             z = 1.959963984540054
             theta = z * (gw ** 0.5) / abs(math.log(surv))
