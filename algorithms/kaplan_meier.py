@@ -15,10 +15,41 @@ from vantage6_amd.algorithm import algorithm_client, data
 from vantage6_amd.algorithm.exceptions import PrivacyThresholdViolation
 
 
+def _merge_counts(parts):
+    """Sum per-station {n, events{t:d}, censored{t:c}} tables."""
+    events: dict[float, int] = {}
+    censored: dict[float, int] = {}
+    total = 0
+    for res in parts:
+        total += res["n"]
+        for t, d in res["events"].items():
+            events[float(t)] = events.get(float(t), 0) + d
+        for t, c in res["censored"].items():
+            censored[float(t)] = censored.get(float(t), 0) + c
+    return total, events, censored
+
+
 @algorithm_client
 def central_kaplan_meier(client, time_col: str, event_col: str,
-                         min_rows: int = 1):
+                         group_col: str | None = None, min_rows: int = 1):
+    """One pooled curve, or (with ``group_col``) one curve per group —
+    the survival plot that pairs with ``central_logrank``."""
     org_ids = [org["id"] for org in client.organization.list()]
+    if group_col is not None:
+        task = client.task.create(
+            input_={"method": "partial_event_counts_grouped",
+                    "kwargs": {"time_col": time_col, "event_col": event_col,
+                               "group_col": group_col,
+                               "min_rows": min_rows}},
+            organizations=org_ids, name="partial KM counts by group")
+        results = client.wait_for_results(task["id"])
+        levels = sorted({lv for r in results for lv in r["groups"]})
+        out = {}
+        for lv in levels:
+            total, ev, ce = _merge_counts(
+                [r["groups"][lv] for r in results if lv in r["groups"]])
+            out[lv] = {"n": total, "curve": _build_curve(total, ev, ce)}
+        return {"groups": out}
     task = client.task.create(
         input_={"method": "partial_event_counts",
                 "kwargs": {"time_col": time_col, "event_col": event_col,
@@ -27,15 +58,11 @@ def central_kaplan_meier(client, time_col: str, event_col: str,
         name="partial KM event counts",
     )
     results = client.wait_for_results(task["id"])
-    events: dict[float, int] = {}
-    censored: dict[float, int] = {}
-    total = 0
-    for res in results:
-        total += res["n"]
-        for t, d in res["events"].items():
-            events[float(t)] = events.get(float(t), 0) + d
-        for t, c in res["censored"].items():
-            censored[float(t)] = censored.get(float(t), 0) + c
+    total, events, censored = _merge_counts(results)
+    return {"n": total, "curve": _build_curve(total, events, censored)}
+
+
+def _build_curve(total, events, censored):
     times = sorted(set(events) | set(censored))
     at_risk = total
     surv = 1.0
@@ -60,7 +87,7 @@ def central_kaplan_meier(client, time_col: str, event_col: str,
                       "censored": censored.get(t, 0), "survival": surv,
                       "std_err": se, "ci95": [lo, hi]})
         at_risk -= d + censored.get(t, 0)
-    return {"n": total, "curve": curve}
+    return curve
 
 
 @data(1)

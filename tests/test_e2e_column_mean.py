@@ -895,3 +895,53 @@ def test_km_greenwood_matches_manual(tmp_path):
         assert lo <= row["survival"] <= hi
         checked += 1
     assert checked > 20
+
+
+def test_km_by_group(tmp_path):
+    """group_col returns one KM curve per arm; each equals the pooled
+    single-arm computation."""
+    import numpy as np
+    import pandas as pd
+
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+
+    rng = np.random.default_rng(103)
+    paths = []
+    for i in range(2):
+        n = 40
+        df = pd.DataFrame({
+            "t": np.concatenate([rng.exponential(6, n),
+                                 rng.exponential(12, n)]).round(1) + 0.1,
+            "e": (rng.random(2 * n) < 0.8).astype(int),
+            "arm": ["a"] * n + ["b"] * n,
+        })
+        p = tmp_path / f"kmg{i}.csv"
+        df.to_csv(p, index=False)
+        paths.append(p)
+    pooled = pd.concat([pd.read_csv(p) for p in paths])
+    net = DevNetwork(n_stations=2, databases=[
+        [DatabaseConfig(label="default", uri=str(p), type="csv")] for p in paths])
+    with net:
+        task = net.client.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],
+            name="km-groups", image="algorithms.kaplan_meier",
+            input_={"method": "central_kaplan_meier",
+                    "kwargs": {"time_col": "t", "event_col": "e",
+                               "group_col": "arm"}})
+        (res,) = net.client.wait_for_results(task["id"])
+    assert set(res["groups"]) == {"a", "b"}
+    for arm in ("a", "b"):
+        sub = pooled[pooled.arm == arm].sort_values("t")
+        at_risk, surv = len(sub), 1.0
+        manual_last = None
+        for t, grp in sub.groupby("t"):
+            d = int((grp.e == 1).sum())
+            if d and at_risk:
+                surv *= 1 - d / at_risk
+            at_risk -= len(grp)
+            manual_last = surv
+        curve = res["groups"][arm]["curve"]
+        assert res["groups"][arm]["n"] == len(sub)
+        assert curve[-1]["survival"] == pytest.approx(manual_last, abs=1e-12)
