@@ -109,3 +109,19 @@ def test_bert_vocab_pad_loss_equivalence():
     assert torch.allclose(la, lb, atol=1e-6), (la.item(), lb.item())
     lb.backward()
     assert b.head.weight.grad[50:].abs().max().item() == 0.0
+
+
+def test_resnet50_builds_and_steps():
+    """Bottleneck ResNet-50: torchvision-exact parameter count, fwd+bwd
+    on CPU."""
+    import torch
+
+    from vantage6_amd.models.factory import build_model, model_loss
+
+    m = build_model("resnet50", dtype=torch.float32)
+    assert sum(p.numel() for p in m.parameters()) == 25_557_032
+    loss = model_loss("resnet50")(
+        m, {"x": torch.randn(2, 3, 64, 64),
+            "y": torch.randint(0, 1000, (2,))})
+    loss.backward()
+    assert torch.isfinite(loss)

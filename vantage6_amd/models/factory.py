@@ -17,6 +17,10 @@ def build_model(name: str, device="cpu", dtype=torch.bfloat16, **kw) -> torch.nn
         from .resnet import resnet18
 
         return resnet18(device=device, dtype=dtype, **kw)
+    if name in ("resnet50", "resnet-50"):
+        from .resnet import resnet50
+
+        return resnet50(device=device, dtype=dtype, **kw)
     if name in ("bert", "bert_base", "bert-base"):
         from .transformer import bert_base
 

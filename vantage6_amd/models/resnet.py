@@ -77,6 +77,38 @@ class BasicBlock(nn.Module):
         return self.bn2(self.conv2(out), residual=idt, relu=True)
 
 
+class Bottleneck(nn.Module):
+    """ResNet-50-style 1x1 -> 3x3 -> 1x1 block, conv-epilogue BN-stat
+    fusion on every conv like BasicBlock (all three shapes run on the
+    hand-written NHWC conv kernels: 1x1 and 3x3 dispatch paths)."""
+
+    expansion = 4
+
+    def __init__(self, cin, width, stride=1):
+        super().__init__()
+        cout = width * self.expansion
+        self.conv1 = HipConv2d(cin, width, 1, bias=False)
+        self.bn1 = FusedBatchNorm2d(width)
+        self.conv2 = conv3x3(width, width, stride)
+        self.bn2 = FusedBatchNorm2d(width)
+        self.conv3 = HipConv2d(width, cout, 1, bias=False)
+        self.bn3 = FusedBatchNorm2d(cout)
+        for c, b in ((self.conv1, self.bn1), (self.conv2, self.bn2),
+                     (self.conv3, self.bn3)):
+            object.__setattr__(c, "stats_bn", b)
+        self.down_conv = None
+        if stride != 1 or cin != cout:
+            self.down_conv = HipConv2d(cin, cout, 1, stride=stride, bias=False)
+            self.down_bn = FusedBatchNorm2d(cout)
+            object.__setattr__(self.down_conv, "stats_bn", self.down_bn)
+
+    def forward(self, x):
+        idt = x if self.down_conv is None else self.down_bn(self.down_conv(x))
+        out = self.bn1(self.conv1(x), relu=True)
+        out = self.bn2(self.conv2(out), relu=True)
+        return self.bn3(self.conv3(out), residual=idt, relu=True)
+
+
 class ResNet18(nn.Module):
     def __init__(self, num_classes: int = 1000, use_mfma_head: bool = True):
         super().__init__()
@@ -101,6 +133,45 @@ class ResNet18(nn.Module):
         x = self.layers(x)
         x = self.avgpool(x).flatten(1)
         return self.fc(x)
+
+
+class ResNet50(nn.Module):
+    def __init__(self, num_classes: int = 1000, use_mfma_head: bool = True):
+        super().__init__()
+        self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
+        self.bn1 = FusedBatchNorm2d(64)
+        self.maxpool = HipMaxPool2d()
+        layers = []
+        cin = 64
+        for width, blocks, stride in ((64, 3, 1), (128, 4, 2),
+                                      (256, 6, 2), (512, 3, 2)):
+            layers.append(Bottleneck(cin, width, stride))
+            cin = width * Bottleneck.expansion
+            for _ in range(blocks - 1):
+                layers.append(Bottleneck(cin, width))
+        self.layers = nn.Sequential(*layers)
+        self.avgpool = nn.AdaptiveAvgPool2d(1)
+        self.fc = (MfmaLinear(2048, num_classes) if use_mfma_head
+                   else nn.Linear(2048, num_classes))
+
+    def forward(self, x):
+        x = self.maxpool(self.bn1(self.conv1(x), relu=True))
+        x = self.layers(x)
+        x = self.avgpool(x).flatten(1)
+        return self.fc(x)
+
+
+def resnet50(num_classes: int = 1000, dtype=torch.bfloat16, device="cpu",
+             use_mfma_head: bool | None = None) -> ResNet50:
+    if use_mfma_head is None:
+        use_mfma_head = True
+    m = ResNet50(num_classes, use_mfma_head=use_mfma_head)
+    m = m.to(device=device, dtype=dtype)
+    for mod in m.modules():
+        if isinstance(mod, (BatchNorm2dBF16, FusedBatchNorm2d)):
+            mod.running_mean.data = mod.running_mean.data.float()
+            mod.running_var.data = mod.running_var.data.float()
+    return m
 
 
 def resnet18(num_classes: int = 1000, dtype=torch.bfloat16, device="cpu",
