@@ -29,6 +29,10 @@ def build_model(name: str, device="cpu", dtype=torch.bfloat16, **kw) -> torch.nn
         from .transformer import llama_1b
 
         return llama_1b(device=device, dtype=dtype, **kw)
+    if name in ("vit", "vit_b16", "vit-b16"):
+        from .transformer import vit_b16
+
+        return vit_b16(device=device, dtype=dtype, **kw)
     raise ValueError(f"unknown model {name!r}")
 
 

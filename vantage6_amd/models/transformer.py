@@ -144,6 +144,46 @@ class Llama(nn.Module):
         return self.head(self.norm(x))
 
 
+class ViT(nn.Module):
+    """Vision Transformer (ViT-B/16 default) on the same MFMA stack as
+    BERT: patch embedding is an MfmaLinear over unfolded patches (no
+    large-kernel conv needed), then pre-LN-free BertLayer encoder blocks,
+    class token + learned positions, MFMA classification head."""
+
+    def __init__(self, image_size: int = 224, patch: int = 16,
+                 hidden: int = 768, layers: int = 12, heads: int = 12,
+                 ffn: int = 3072, num_classes: int = 1000):
+        super().__init__()
+        assert image_size % patch == 0
+        self.patch = patch
+        n_tok = (image_size // patch) ** 2 + 1  # + class token
+        self.embed = MfmaLinear(3 * patch * patch, hidden)
+        self.cls = nn.Parameter(torch.zeros(1, 1, hidden))
+        self.pos = nn.Parameter(torch.zeros(1, n_tok, hidden) .normal_(0, 0.02))
+        self.blocks = nn.ModuleList(BertLayer(hidden, heads, ffn)
+                                    for _ in range(layers))
+        self.ln = LayerNorm(hidden)
+        self.head = MfmaLinear(hidden, num_classes)
+
+    def forward(self, x):
+        B, C, H, W = x.shape
+        p = self.patch
+        # [B,C,H,W] -> [B, (H/p)(W/p), C*p*p] patch unfold
+        x = (x.reshape(B, C, H // p, p, W // p, p)
+             .permute(0, 2, 4, 1, 3, 5).reshape(B, -1, C * p * p))
+        x = self.embed(x)
+        x = torch.cat([self.cls.expand(B, 1, -1).to(x.dtype), x], dim=1)
+        x = x + self.pos[:, : x.shape[1]].to(x.dtype)
+        for blk in self.blocks:
+            x = blk(x)
+        return self.head(self.ln(x[:, 0]))
+
+
+def vit_b16(dtype=torch.bfloat16, device="cpu", **kw) -> ViT:
+    m = ViT(**kw)
+    return m.to(device=device, dtype=dtype)
+
+
 def bert_base(dtype=torch.bfloat16, device="cpu", **kw) -> BertBase:
     m = BertBase(**kw)
     return m.to(device=device, dtype=dtype)
