@@ -945,3 +945,37 @@ def test_km_by_group(tmp_path):
         curve = res["groups"][arm]["curve"]
         assert res["groups"][arm]["n"] == len(sub)
         assert curve[-1]["survival"] == pytest.approx(manual_last, abs=1e-12)
+
+
+def test_sql_database_type(tmp_path):
+    """sql-type station databases (reference parity): table and query
+    fragments against a sqlalchemy URI."""
+    import numpy as np
+    import pandas as pd
+    from sqlalchemy import create_engine
+
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+
+    rng = np.random.default_rng(107)
+    uris = []
+    vals = []
+    for i in range(2):
+        db = tmp_path / f"s{i}.db"
+        eng = create_engine(f"sqlite:///{db}")
+        v = rng.normal(5, 2, 30)
+        pd.DataFrame({"v": v}).to_sql("measurements", eng, index=False)
+        eng.dispose()
+        vals.append(v)
+        uris.append(f"sqlite:///{db}#table=measurements")
+    pooled_mean = float(np.concatenate(vals).mean())
+    net = DevNetwork(n_stations=2, databases=[
+        [DatabaseConfig(label="default", uri=u, type="sql")] for u in uris])
+    with net:
+        task = net.client.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],
+            name="sql-mean", image="algorithms.column_mean",
+            input_={"method": "central_mean", "kwargs": {"columns": ["v"]}})
+        (res,) = net.client.wait_for_results(task["id"])
+    assert res["v"] == pytest.approx(pooled_mean, rel=1e-12)

@@ -48,6 +48,33 @@ def _load_database(uri: str, dbtype: str, env: dict[str, str]):
             return torch.load(uri, weights_only=True)
         if dbtype == "synthetic":
             return _synthetic_shard(uri, env)
+        if dbtype == "sql":
+            # reference: sql-type node databases; the shard is a table or
+            # query against a sqlalchemy URI. Select via URI fragment:
+            #   sqlite:///data.db#table=patients
+            #   postgresql://...#query=SELECT%20...
+            import urllib.parse
+
+            from sqlalchemy import create_engine
+
+            base, _, frag = uri.partition("#")
+            q = dict(urllib.parse.parse_qsl(frag))
+            engine = create_engine(base)
+            try:
+                if "query" in q:
+                    return pd.read_sql(urllib.parse.unquote(q["query"]),
+                                       engine)
+                if "table" in q:
+                    return pd.read_sql_table(q["table"], engine)
+                raise DataError(
+                    "sql database URI needs a #table=<name> or "
+                    "#query=<urlencoded sql> fragment")
+            finally:
+                engine.dispose()
+        if dbtype in ("sparql", "omop"):
+            raise DataError(
+                f"{dbtype} databases need client wheels not present in this "
+                f"offline image (SPARQLWrapper / OMOP CDM drivers)")
     except DataError:
         raise
     except Exception as exc:  # noqa: BLE001
