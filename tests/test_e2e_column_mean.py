@@ -978,4 +978,9 @@ def test_sql_database_type(tmp_path):
             name="sql-mean", image="algorithms.column_mean",
             input_={"method": "central_mean", "kwargs": {"columns": ["v"]}})
         (res,) = net.client.wait_for_results(task["id"])
+        # column discovery works for sql-type databases too
+        root = net.root_client
+        nodes = root.node.list(per_page=10)["data"]
+        cols = root.node.columns(nodes[0]["id"])
+        assert cols.get("default") == ["v"]
     assert res["v"] == pytest.approx(pooled_mean, rel=1e-12)

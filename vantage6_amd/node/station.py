@@ -162,7 +162,7 @@ class Station:
         out: dict[str, list[str]] = {}
         for db in self.config.databases:
             try:
-                if db.type in ("csv", "parquet", "excel"):
+                if db.type in ("csv", "parquet", "excel", "sql"):
                     from ..algorithm.decorators import _load_database
 
                     df = _load_database(db.uri, db.type, {})
