@@ -981,6 +981,6 @@ def test_sql_database_type(tmp_path):
         # column discovery works for sql-type databases too
         root = net.root_client
         nodes = root.node.list(per_page=10)["data"]
-        cols = root.node.columns(nodes[0]["id"])
-        assert cols.get("default") == ["v"]
+        cols = root.node.columns(nodes[0]["id"])  # list for label "default"
+        assert cols == ["v"]
     assert res["v"] == pytest.approx(pooled_mean, rel=1e-12)
