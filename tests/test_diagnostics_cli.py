@@ -78,3 +78,16 @@ def test_every_cli_command_help_works():
     for c in cmds:
         r = runner.invoke(cli, [c, "--help"])
         assert r.exit_code == 0, (c, r.output)
+
+
+def test_examples_run():
+    """examples/ scripts execute end-to-end on CPU."""
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    root = Path(__file__).resolve().parents[1]
+    for name in ("quickstart.py", "fedavg_training.py"):
+        r = subprocess.run([sys.executable, str(root / "examples" / name)],
+                           capture_output=True, text=True, timeout=240)
+        assert r.returncode == 0, (name, r.stderr[-1500:])
