@@ -33,7 +33,7 @@ from vantage6_amd.parallel.fedavg import FedAvgConfig, FedAvgTrainer
 
 def synthetic_dataset(model: str, n: int, seed: int, model_kwargs: dict):
     g = torch.Generator().manual_seed(seed)
-    if model in ("resnet18", "resnet50"):
+    if model in ("resnet18", "resnet50", "vit_b16"):
         return {
             "x": torch.randn(n, 3, 224, 224, generator=g),
             "y": torch.randint(0, model_kwargs.get("num_classes", 1000), (n,), generator=g),
@@ -59,8 +59,8 @@ def main() -> None:
     ap.add_argument("--steps", type=int, default=20, help="timed FedAvg rounds")
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--model", default="resnet18",
-                    choices=["resnet18", "resnet50", "logreg", "bert_base",
-                             "llama_1b"])
+                    choices=["resnet18", "resnet50", "vit_b16", "logreg",
+                             "bert_base", "llama_1b"])
     ap.add_argument("--batch-size", type=int, default=None, help="per-station batch")
     ap.add_argument("--local-steps", type=int, default=4)
     ap.add_argument("--seq-len", type=int, default=512)
@@ -92,7 +92,8 @@ def main() -> None:
         # below the random-init loss of each model (ln(1000)=6.9,
         # ln(30522)=10.3, ln(128256)=11.8) so hitting them requires real
         # optimization progress on the synthetic shards.
-        args.target_loss = {"resnet18": 6.0, "resnet50": 6.0, "logreg": 0.45,
+        args.target_loss = {"resnet18": 6.0, "resnet50": 6.0,
+                            "vit_b16": 6.5, "logreg": 0.45,
                             "bert_base": 9.5, "llama_1b": 11.0}[args.model]
 
     comm = init_from_env()
@@ -105,6 +106,7 @@ def main() -> None:
     defaults = {
         "resnet18": {"batch": 32, "kwargs": {"num_classes": 1000}, "shard": 256},
         "resnet50": {"batch": 32, "kwargs": {"num_classes": 1000}, "shard": 256},
+        "vit_b16": {"batch": 32, "kwargs": {"num_classes": 1000}, "shard": 256},
         "logreg": {"batch": 1024, "kwargs": {"in_features": 128, "num_classes": 2},
                    "shard": 16384},
         "bert_base": {"batch": 8, "kwargs": {}, "shard": 64},
@@ -139,7 +141,7 @@ def main() -> None:
         local_steps=args.local_steps,
         lr=0.1 if args.model in ("resnet18", "resnet50", "logreg") else 1e-4,
         optimizer="sgd" if args.model in ("resnet18", "resnet50", "logreg")
-        else "adamw",
+        else "adamw",  # vit trains with adamw like the transformers
         secure_agg=args.secure_agg,
         dp=dp_cfg,
         agg_algo=args.agg_algo,
