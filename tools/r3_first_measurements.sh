@@ -13,6 +13,8 @@ timeout 400 python -m pytest tests -m gpu -x -q 2>&1 | tail -3
 # 2. new model families on the HIP path (never GPU-measured in r2)
 timeout 200 python bench.py --model resnet50 --steps 15 --warmup 5 2>/dev/null | tail -1 \
   | tee gpurun_out/bench_resnet50_r3.json
+timeout 200 python bench.py --model vit_b16 --steps 10 --warmup 3 2>/dev/null | tail -1 \
+  | tee gpurun_out/bench_vit_r3.json
 # 3. BERT with the padded-vocab fused CE (r2 lead: NOTES.md item 5) —
 #    edit models/transformer.py bert_base default or pass kwargs via a
 #    micro-script, then compare against the 17.0 r/s baseline:
