@@ -984,3 +984,43 @@ def test_sql_database_type(tmp_path):
         cols = root.node.columns(nodes[0]["id"])  # list for label "default"
         assert cols == ["v"]
     assert res["v"] == pytest.approx(pooled_mean, rel=1e-12)
+
+
+def test_mock_client_agrees_with_real_network(tmp_path):
+    """MockAlgorithmClient (algorithm-dev harness) produces the SAME
+    result as the full client->orchestrator->station path for the same
+    shards — the mock is a faithful stand-in."""
+    import numpy as np
+    import pandas as pd
+
+    from vantage6_amd.algorithm.mock_client import MockAlgorithmClient
+    from vantage6_amd.cli.dev import DevNetwork
+    from vantage6_amd.common.config import DatabaseConfig
+
+    rng = np.random.default_rng(113)
+    frames, paths = [], []
+    for i in range(2):
+        df = pd.DataFrame({"v": rng.normal(3, 1, 40 + 10 * i)})
+        p = tmp_path / f"mk{i}.csv"
+        df.to_csv(p, index=False)
+        frames.append(df)
+        paths.append(p)
+
+    mock = MockAlgorithmClient(
+        datasets=[[{"database": str(p), "type": "csv"}] for p in paths],
+        module="algorithms.summary_stats")
+    from algorithms.summary_stats import central_summary
+
+    mock_res = central_summary.__wrapped__(mock, columns=["v"])
+
+    net = DevNetwork(n_stations=2, databases=[
+        [DatabaseConfig(label="default", uri=str(p), type="csv")] for p in paths])
+    with net:
+        task = net.client.task.create(
+            collaboration=net.collaboration["id"],
+            organizations=[net.org_ids[0]],
+            name="real", image="algorithms.summary_stats",
+            input_={"method": "central_summary",
+                    "kwargs": {"columns": ["v"]}})
+        (real_res,) = net.client.wait_for_results(task["id"])
+    assert mock_res == real_res
