@@ -87,7 +87,7 @@ def test_examples_run():
     from pathlib import Path
 
     root = Path(__file__).resolve().parents[1]
-    for name in ("quickstart.py", "fedavg_training.py"):
+    for name in ("quickstart.py", "federated_glm.py", "fedavg_training.py"):
         r = subprocess.run([sys.executable, str(root / "examples" / name)],
                            capture_output=True, text=True, timeout=240)
         assert r.returncode == 0, (name, r.stderr[-1500:])
